@@ -1,0 +1,331 @@
+/* bng_abi.h — the host<->device table ABI for the MI355X BNG dataplane.
+ *
+ * This is the MI355X-native analog of the reference's eBPF map contract
+ * (reference: bpf/maps.h:89-234, bpf/nat44.c:92-320, bpf/qos_ratelimit.c:24-65,
+ * bpf/antispoof.c:36-119).  Every table lives in HBM3E as a fixed-capacity
+ * open-addressing hash table (power-of-2 slots, linear probing) or a flat
+ * array.  The Python side mirrors these structs byte-for-byte in
+ * bng_amd/dataplane/abi.py; tests/test_abi.py asserts the layouts agree
+ * (the analog of the reference's test/ebpf/maps_test.go struct-ABI tests).
+ *
+ * Design departures from the reference (deliberate, GPU-first):
+ *  - One unified subscriber table replaces the reference's three maps
+ *    (subscriber_pools / vlan_subscriber_pools / circuit_id_subscribers,
+ *    bpf/maps.h:99-129,229-234): the 64-bit key space is tagged by source
+ *    (MAC / QinQ VLAN / circuit-ID FNV hash) so the fast path probes ONE
+ *    HBM table at most three times instead of three tables.
+ *  - BPF LRU maps become timeout-reclaimed open-addressing tables: a probe
+ *    that hits an expired entry may reclaim the slot with atomicCAS.
+ *  - bpf_ktime_get_ns() per packet becomes one host-written batch timestamp.
+ *  - Per-CPU stats maps become one device-global atomic counter block
+ *    (256 CUs' contention on ~10 u64 counters is negligible per batch).
+ */
+#ifndef BNG_ABI_H
+#define BNG_ABI_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---------------------------------------------------------------- sizing */
+enum {
+  BNG_MAX_SUBSCRIBERS_LOG2 = 21, /* 2M slots for 1M subscribers (load<=0.5) */
+  BNG_MAX_POOLS            = 16384,
+  BNG_MAX_NAT_SESSIONS_LOG2= 23, /* 8M slots for 4M sessions */
+  BNG_MAX_EIM_LOG2         = 22, /* 4M slots for 2M EIM mappings */
+  BNG_MAX_SUBNAT_LOG2      = 21,
+  BNG_MAX_QOS_LOG2         = 21,
+  BNG_MAX_BINDINGS_LOG2    = 21,
+  BNG_MAX_PROBE            = 128, /* linear-probe bound before declaring full */
+  BNG_MAX_PRIVATE_RANGES   = 64,  /* bpf/nat44.c:314-320 nat_private_ranges */
+  BNG_MAX_ALLOWED_RANGES   = 256, /* bpf/antispoof.c:113-119 allowed_ranges_v4 */
+  BNG_MAX_HAIRPIN_IPS      = 1024,
+  BNG_MAX_ALG_PORTS        = 64,
+  BNG_LOG_RING_LOG2        = 15,  /* 32k nat-log entries (~1.25MB; ref 1MB rb) */
+  BNG_SPOOF_RING_LOG2      = 14,
+};
+
+/* key-space tags for the unified subscriber table (bits 63:62 of the key) */
+#define BNG_KEY_MAC      0ULL   /* key = MAC as u64 (< 2^48)               */
+#define BNG_KEY_VLAN     (1ULL << 62)  /* key = tag | s_tag<<16 | c_tag    */
+#define BNG_KEY_CIRCUIT  (2ULL << 62)  /* key = tag | (fnv1a64(cid) >> 2)  */
+#define BNG_KEY_EMPTY    0ULL
+#define BNG_KEY_TOMBSTONE 0xFFFFFFFFFFFFFFFFULL
+
+/* ----------------------------------------------------- verdicts (XDP/TC) */
+enum bng_verdict {
+  BNG_PASS = 0,  /* to slow path (XDP_PASS / TC_ACT_OK untouched)          */
+  BNG_TX   = 1,  /* reply built in place, transmit (XDP_TX)                */
+  BNG_DROP = 2,  /* drop (XDP_DROP / TC_ACT_SHOT)                          */
+  BNG_FWD  = 3,  /* packet rewritten, forward (TC_ACT_OK after NAT)        */
+};
+
+/* ------------------------------------------------- subscriber fast path */
+/* Unified subscriber entry: reference pool_assignment (bpf/maps.h:89-97)
+ * plus its hash key; 32 bytes = half a cache line, 2 entries per 64B line. */
+typedef struct bng_sub_entry {
+  uint64_t key;          /* tagged key; 0 empty, ~0 tombstone              */
+  uint32_t pool_id;
+  uint32_t allocated_ip; /* network byte order, as it appears in packets   */
+  uint64_t lease_expiry; /* unix seconds (ref maps.h:94)                   */
+  uint16_t vlan_id;
+  uint8_t  client_class;
+  uint8_t  flags;
+  uint32_t _pad;
+} bng_sub_entry;  /* 32 B */
+
+/* IP pool metadata, flat array indexed by pool_id (ref maps.h:135-151). */
+typedef struct bng_ip_pool {
+  uint32_t network;      /* network byte order */
+  uint32_t gateway;
+  uint32_t dns_primary;
+  uint32_t dns_secondary;
+  uint32_t lease_time;   /* host order, seconds */
+  uint8_t  prefix_len;
+  uint8_t  valid;
+  uint16_t _pad;
+  uint32_t _pad2;
+} bng_ip_pool;  /* 28 B -> padded */
+
+typedef struct bng_server_config {   /* ref maps.h:154-159 */
+  uint8_t  server_mac[6];
+  uint16_t _pad;
+  uint32_t server_ip;    /* network byte order */
+  uint32_t if_index;
+} bng_server_config;  /* 16 B */
+
+/* DHCP fast-path counters, same order as ref maps.h:171-184. */
+enum bng_dhcp_stat {
+  BNG_ST_TOTAL_REQUESTS = 0,
+  BNG_ST_FASTPATH_HITS,
+  BNG_ST_FASTPATH_MISSES,
+  BNG_ST_ERRORS,
+  BNG_ST_CACHE_EXPIRED,
+  BNG_ST_OPTION82_PRESENT,
+  BNG_ST_OPTION82_ABSENT,
+  BNG_ST_BROADCAST_REPLIES,
+  BNG_ST_UNICAST_REPLIES,
+  BNG_ST_VLAN_PACKETS,
+  BNG_DHCP_NSTATS
+};
+
+/* --------------------------------------------------------------- NAT44 */
+/* 5-tuple (ref nat44.c:92-99). 16 B, hashed into a 64-bit slot signature. */
+typedef struct bng_nat_tuple {
+  uint32_t src_ip;   /* network order */
+  uint32_t dst_ip;
+  uint16_t src_port; /* network order */
+  uint16_t dst_port;
+  uint8_t  protocol;
+  uint8_t  _pad[3];
+} bng_nat_tuple;  /* 16 B */
+
+enum bng_nat_state {   /* ref nat44.c:65-71 */
+  BNG_NAT_NEW = 0, BNG_NAT_ESTABLISHED = 1, BNG_NAT_FIN_WAIT = 2,
+  BNG_NAT_CLOSING = 3, BNG_NAT_TIME_WAIT = 4,
+};
+
+/* Session entry: 128 B = 2 cache lines; line 0 is the hot lookup+translate
+ * path, line 1 the accounting (ref nat_session nat44.c:123-141). */
+typedef struct bng_nat_session {
+  uint64_t sig;          /* 0 empty, ~0 tombstone, else mix64(tuple)|1     */
+  bng_nat_tuple key;     /* full key for verification after sig match      */
+  uint32_t nat_ip;
+  uint16_t nat_port;     /* network order */
+  uint16_t orig_port;    /* network order */
+  uint32_t orig_ip;
+  uint8_t  state;
+  uint8_t  is_hairpin;
+  uint8_t  ready;        /* set (agent-scope release) after fields written */
+  uint8_t  _pad;
+  uint64_t last_seen;    /* ns */
+  /* ---- line 1 ---- */
+  uint64_t created;
+  uint64_t packets_out;
+  uint64_t packets_in;
+  uint64_t bytes_out;
+  uint64_t bytes_in;
+  uint64_t _pad2[3];
+} bng_nat_session;  /* 128 B */
+
+/* Reverse map entry: external tuple -> internal tuple (ref nat44.c:228-233) */
+typedef struct bng_nat_reverse {
+  uint64_t sig;
+  bng_nat_tuple key;     /* external-side tuple */
+  bng_nat_tuple orig;    /* original internal tuple */
+  uint8_t  ready;
+  uint8_t  _pad[7];
+} bng_nat_reverse;  /* 48 B */
+
+/* EIM entry (RFC 4787; ref eim_key/eim_mapping nat44.c:104-120). */
+typedef struct bng_eim_entry {
+  uint64_t sig;          /* mix of (internal_ip, internal_port, proto)     */
+  uint32_t internal_ip;
+  uint16_t internal_port; /* network order */
+  uint8_t  protocol;
+  uint8_t  ready;
+  uint32_t external_ip;
+  uint16_t external_port; /* HOST order, as in ref eim_mapping             */
+  uint16_t _pad;
+  uint64_t created;
+  uint64_t last_used;
+  uint32_t ref_count;
+  uint32_t flags;
+} bng_eim_entry;  /* 48 B */
+
+/* Per-subscriber port block + counters (ref port_block/subscriber_nat
+ * nat44.c:144-164).  Host-inserted (stream-ordered), device-updated. */
+typedef struct bng_subnat_entry {
+  uint32_t key_ip;       /* subscriber private IP (network order); 0 empty */
+  uint32_t subscriber_id;
+  uint32_t public_ip;    /* network order */
+  uint16_t port_start;   /* host order, inclusive */
+  uint16_t port_end;     /* host order, inclusive */
+  uint32_t next_port;    /* atomic rotor, host order */
+  uint32_t ports_in_use;
+  uint64_t allocated_at;
+  uint32_t sessions_active;  /* device atomics (u32; host widens to u64)   */
+  uint32_t sessions_total;
+  uint64_t bytes_out;
+  uint64_t bytes_in;
+  uint8_t  block_size_log2;
+  uint8_t  flags;
+  uint8_t  _pad[6];
+} bng_subnat_entry;  /* 64 B */
+
+/* Global NAT config (ref nat_config nat44.c:271-277). */
+#define BNG_NAT_FLAG_EIM       0x01
+#define BNG_NAT_FLAG_EIF       0x02
+#define BNG_NAT_FLAG_HAIRPIN   0x04
+#define BNG_NAT_FLAG_ALG_FTP   0x08
+#define BNG_NAT_FLAG_ALG_SIP   0x10
+#define BNG_NAT_FLAG_PARITY    0x20
+#define BNG_NAT_FLAG_CONTIG    0x40
+
+typedef struct bng_nat_config {
+  uint32_t flags;
+  uint16_t port_range_start;
+  uint16_t port_range_end;
+  uint32_t default_ports_per_sub;
+  uint32_t n_private_ranges;
+  uint32_t n_alg_ports;
+  uint32_t _pad;
+  /* private ranges as {network(be), mask(be)} pairs; ref uses an LPM trie
+   * (nat44.c:314-320) — at <=64 ranges a linear scan of an L2-resident
+   * array beats a trie on GPU. */
+  uint32_t private_net[BNG_MAX_PRIVATE_RANGES];
+  uint32_t private_mask[BNG_MAX_PRIVATE_RANGES];
+  uint32_t alg_key[BNG_MAX_ALG_PORTS];   /* port<<16 | proto (host order)  */
+} bng_nat_config;
+
+enum bng_nat_stat {  /* ref nat_stats nat44.c:176-190 */
+  BNG_NS_SNAT = 0, BNG_NS_DNAT, BNG_NS_HAIRPIN, BNG_NS_DROPPED, BNG_NS_PASSED,
+  BNG_NS_SESS_CREATED, BNG_NS_SESS_EXPIRED, BNG_NS_PORT_EXHAUSTION,
+  BNG_NS_EIM_HITS, BNG_NS_EIM_MISSES, BNG_NS_ALG_TRIGGERS,
+  BNG_NS_CT_LOOKUPS, BNG_NS_CT_HITS,
+  BNG_NAT_NSTATS
+};
+
+/* NAT compliance log record (ref nat_log_entry nat44.c:193-205); ring. */
+enum bng_nat_log_event {  /* ref nat44.c:74-82 */
+  BNG_LOG_SESSION_CREATE = 1, BNG_LOG_SESSION_DELETE = 2,
+  BNG_LOG_PB_ASSIGN = 3, BNG_LOG_PB_RELEASE = 4, BNG_LOG_PORT_EXHAUSTION = 5,
+  BNG_LOG_HAIRPIN = 6, BNG_LOG_ALG_TRIGGER = 7,
+};
+
+typedef struct bng_nat_log_entry {
+  uint64_t timestamp;
+  uint32_t event_type;
+  uint32_t subscriber_id;
+  uint32_t private_ip;
+  uint32_t public_ip;
+  uint16_t private_port;
+  uint16_t public_port;
+  uint32_t dest_ip;
+  uint16_t dest_port;
+  uint8_t  protocol;
+  uint8_t  flags;
+} bng_nat_log_entry;  /* 40 B */
+
+/* ------------------------------------------------------------------ QoS */
+/* Token bucket (ref token_bucket qos_ratelimit.c:24-31).  tokens is a
+ * SIGNED 64-bit so consume can be one atomicAdd(-len) with undo-on-negative
+ * (the reference's per-packet read-modify-write is a data race it tolerates;
+ * ours is an atomic with a bounded transient-negative race). */
+typedef struct bng_qos_bucket {
+  uint32_t key_ip;       /* subscriber IP (network order); 0 empty         */
+  uint8_t  valid;
+  uint8_t  priority;
+  uint16_t _pad;
+  uint64_t rate_bps;     /* 0 = unlimited */
+  int64_t  tokens;       /* bytes */
+  uint64_t last_update;  /* ns */
+  uint32_t burst_bytes;
+  uint32_t _pad2;
+  uint64_t _pad3[3];
+} bng_qos_bucket;  /* 64 B */
+
+enum bng_qos_stat {  /* ref qos_stats qos_ratelimit.c:53-58 */
+  BNG_QS_PKT_PASSED = 0, BNG_QS_PKT_DROPPED, BNG_QS_BYTES_PASSED,
+  BNG_QS_BYTES_DROPPED, BNG_QOS_NSTATS
+};
+
+/* ------------------------------------------------------------ antispoof */
+enum bng_antispoof_mode {  /* ref antispoof.c:30-33 */
+  BNG_AS_DISABLED = 0, BNG_AS_STRICT = 1, BNG_AS_LOOSE = 2, BNG_AS_LOG_ONLY = 3,
+};
+
+typedef struct bng_binding_entry {  /* ref subscriber_binding antispoof.c:36-43 */
+  uint64_t key_mac;      /* MAC as u64; 0 empty, ~0 tombstone              */
+  uint32_t ipv4_addr;    /* network order */
+  uint8_t  ipv4_valid;
+  uint8_t  ipv6_valid;
+  uint8_t  mode;
+  uint8_t  _pad;
+  uint8_t  ipv6_addr[16];
+} bng_binding_entry;  /* 32 B */
+
+typedef struct bng_antispoof_config {  /* ref antispoof.c:79-83 */
+  uint8_t  default_mode;
+  uint8_t  log_violations;
+  uint16_t _pad;
+  uint32_t n_allowed_ranges;
+  uint32_t allowed_net[BNG_MAX_ALLOWED_RANGES];
+  uint32_t allowed_mask[BNG_MAX_ALLOWED_RANGES];
+} bng_antispoof_config;
+
+enum bng_antispoof_stat {  /* ref antispoof_stats antispoof.c:58-65 */
+  BNG_AS_ALLOWED = 0, BNG_AS_DROPPED, BNG_AS_LOGGED,
+  BNG_AS_V4_VIOLATIONS, BNG_AS_V6_VIOLATIONS, BNG_AS_UNKNOWN_MAC,
+  BNG_AS_NSTATS
+};
+
+typedef struct bng_spoof_event {  /* ref spoof_event antispoof.c:46-55 */
+  uint64_t timestamp;
+  uint8_t  src_mac[6];
+  uint8_t  protocol;   /* 4 or 6 */
+  uint8_t  _pad;
+  uint32_t spoofed_ip;
+  uint32_t allowed_ip;
+  uint8_t  spoofed_ipv6[16];
+  uint8_t  allowed_ipv6[16];
+} bng_spoof_event;  /* 56 B */
+
+/* ----------------------------------------------------------- ring header */
+/* Device->host ordered log ring: device atomically bumps widx; host reads
+ * records [ridx, widx) after a stream-ordered copy, then advances ridx.
+ * Replaces the reference's BPF ring buffer (nat44.c:294-298). */
+typedef struct bng_ring_header {
+  uint32_t widx;     /* device atomic write index (monotonic, wraps mod 2^32) */
+  uint32_t dropped;  /* records lost to a full ring                           */
+  uint32_t capacity; /* power of two                                          */
+  uint32_t _pad;
+} bng_ring_header;  /* 16 B */
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* BNG_ABI_H */
