@@ -1,0 +1,348 @@
+#!/usr/bin/env python3
+"""bench.py — flagship BNG dataplane benchmark (driver contract).
+
+Measures the BASELINE.json headline: Mpps + p50 DHCP OFFER latency on a
+64B IPv4 mix with a 1M-subscriber table, at 1..8 MI355X GPUs.
+
+One step = ingest one fixed-size synthetic batch per GPU (restore from
+pristine = the RX-DMA analog), steer each packet to its owning shard via
+RCCL all-to-all over xGMI (world>1), and run the fused uplink pipeline
+(DHCP fast path for UDP:67; antispoof -> NAT44 SNAT -> QoS for data
+packets) on the local shard's HBM tables.  Weak scaling: per-GPU injected
+batch is fixed as N grows.
+
+Traffic mix (configurable): 90% 64-byte UDP IPv4 data packets from
+subscriber IPs (SNAT + QoS + antispoof path), 10% DHCP DISCOVER/REQUEST
+(fast-path OFFER/ACK built in place).  Tables: 1M subscribers / NAT
+blocks / QoS buckets / antispoof bindings, random-init, sharded by the
+MAC/IP hashring when N>1.
+
+vs_baseline: the reference's published Mpps figure on this metric is the
+XDP single-core packet rate it cites (24 Mpps/core, BASELINE.md "XDP
+single-core packet rate"); its DHCP-specific ceiling is 1-2 Mpps
+(BASELINE.md "DHCP packet-rate ceiling").  We divide by 24.0.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import struct
+import time
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+import sys
+sys.path.insert(0, REPO)
+
+from bng_amd.dataplane import abi
+from bng_amd.dataplane.packets import build_dhcp_request, build_ipv4, ip2u32
+
+BASELINE_MPPS = 24.0
+
+MASK64 = np.uint64(0xFFFFFFFFFFFFFFFF)
+
+
+def mix64_np(x: np.ndarray) -> np.ndarray:
+    """Vectorized splitmix64 finalizer, must match abi.mix64."""
+    with np.errstate(over="ignore"):
+        x = (x + np.uint64(0x9E3779B97F4A7C15)) & MASK64
+        z = x
+        z = ((z ^ (z >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)) & MASK64
+        z = ((z ^ (z >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)) & MASK64
+        return z ^ (z >> np.uint64(31))
+
+
+def build_tables(launcher, rank: int, world: int, n_subs: int, now_sec: int):
+    """Bulk-populate subscriber/NAT/QoS/binding tables for this shard."""
+    import torch
+    idx = np.arange(n_subs, dtype=np.uint64)
+    macs = np.uint64(0xAA0000000000) + idx
+    ips = (np.uint64(ip2u32("10.0.0.0") + 2) + idx).astype(np.uint64)
+
+    mac_owner = mix64_np(macs) % np.uint64(world)
+    ip_owner = mix64_np(ips) % np.uint64(world)
+
+    dev = launcher.device
+
+    def to_dev_u8(arr):
+        return torch.from_numpy(arr.view(np.uint8)).to(dev).flatten()
+
+    # subscriber_pools entries (MAC-owned shard)
+    sel = np.nonzero(mac_owner == rank)[0]
+    sub = np.zeros(len(sel), dtype=[("key", "<u8"), ("pool", "<u4"),
+                                    ("ip", "<u4"), ("lease", "<u8"),
+                                    ("vlan", "<u2"), ("cc", "u1"),
+                                    ("fl", "u1"), ("pad", "<u4")])
+    sub["key"] = macs[sel]
+    sub["pool"] = 1
+    sub["ip"] = ips[sel].astype(np.uint32)
+    sub["lease"] = now_sec + 86400
+    rc = torch.zeros(len(sel), dtype=torch.int32, device=dev)
+    launcher.ext.sub_upsert(launcher.subs, to_dev_u8(sub), rc)
+    n_bad = int((rc != 0).sum().item())
+    assert n_bad == 0, f"{n_bad} subscriber upserts failed (table too full)"
+
+    # subnat port blocks (IP-owned shard)
+    sel = np.nonzero(ip_owner == rank)[0]
+    pub_base = ip2u32("203.0.113.0")
+    snat = np.zeros(len(sel), dtype=[("key_ip", "<u4"), ("sid", "<u4"),
+                                     ("pub", "<u4"), ("ps", "<u2"),
+                                     ("pe", "<u2"), ("np", "<u4"),
+                                     ("piu", "<u4"), ("at", "<u8"),
+                                     ("sa", "<u4"), ("st", "<u4"),
+                                     ("bo", "<u8"), ("bi", "<u8"),
+                                     ("bl", "u1"), ("fl", "u1"),
+                                     ("pad", "6u1")])
+    snat["key_ip"] = ips[sel].astype(np.uint32)
+    snat["sid"] = sel.astype(np.uint32)
+    snat["pub"] = pub_base + (sel % 250).astype(np.uint32)
+    starts = (1024 + (sel % 63) * 1024).astype(np.uint16)
+    snat["ps"] = starts
+    snat["pe"] = starts + 1023
+    snat["np"] = starts
+    rc = torch.zeros(len(sel), dtype=torch.int32, device=dev)
+    launcher.ext.subnat_upsert(launcher.subnat, to_dev_u8(snat), rc)
+    assert int((rc != 0).sum().item()) == 0
+
+    # QoS ingress buckets (IP-owned shard): 1 Gbps / 4 MB burst => pass
+    qos = np.zeros(len(sel), dtype=[("key_ip", "<u4"), ("valid", "u1"),
+                                    ("prio", "u1"), ("pad", "<u2"),
+                                    ("rate", "<u8"), ("tokens", "<i8"),
+                                    ("last", "<u8"), ("burst", "<u4"),
+                                    ("pad2", "<u4"), ("pad3", "3<u8")])
+    qos["key_ip"] = ips[sel].astype(np.uint32)
+    qos["valid"] = 1
+    qos["rate"] = 10**9
+    qos["tokens"] = 4 << 20
+    qos["burst"] = 4 << 20
+    qos["last"] = now_sec * 10**9
+    rc = torch.zeros(len(sel), dtype=torch.int32, device=dev)
+    launcher.ext.qos_upsert(launcher.qos_ingress, to_dev_u8(qos), rc)
+    assert int((rc != 0).sum().item()) == 0
+
+    # antispoof strict bindings (packets arrive by IP shard, keyed by MAC;
+    # mode strict: src ip must equal the bound ip)
+    bind = np.zeros(len(sel), dtype=[("mac", "<u8"), ("ip", "<u4"),
+                                     ("v4", "u1"), ("v6", "u1"),
+                                     ("mode", "u1"), ("pad", "u1"),
+                                     ("v6a", "16u1")])
+    bind["mac"] = macs[sel]
+    bind["ip"] = ips[sel].astype(np.uint32)
+    bind["v4"] = 1
+    bind["mode"] = abi.AS_STRICT
+    rc = torch.zeros(len(sel), dtype=torch.int32, device=dev)
+    launcher.ext.binding_upsert(launcher.bindings, to_dev_u8(bind), rc)
+    assert int((rc != 0).sum().item()) == 0
+
+
+def gen_batch(n_pkts: int, n_subs: int, dhcp_frac: float, stride: int,
+              seed: int):
+    """Vectorized synthetic batch: [n,stride] uint8 + lens int16."""
+    rng = np.random.default_rng(seed)
+    idx = rng.integers(0, n_subs, size=n_pkts, dtype=np.uint64)
+    is_dhcp = rng.random(n_pkts) < dhcp_frac
+
+    data = np.zeros((n_pkts, stride), dtype=np.uint8)
+    lens = np.zeros(n_pkts, dtype=np.uint16)
+
+    # templates
+    udp_t = np.frombuffer(build_ipv4(
+        "aa:00:00:00:00:00", "02:00:00:00:00:01", ip2u32("10.0.0.2"),
+        ip2u32("93.184.216.34"), proto=17, sport=40000, dport=53,
+        payload=b"\x00" * 22), dtype=np.uint8)          # 64 bytes
+    assert len(udp_t) == 64
+    dhcp_t = np.frombuffer(build_dhcp_request(
+        "aa:00:00:00:00:00", 1, xid=1), dtype=np.uint8)
+    dhcp_req_t = np.frombuffer(build_dhcp_request(
+        "aa:00:00:00:00:00", 3, xid=1), dtype=np.uint8)
+
+    macs = (np.uint64(0xAA0000000000) + idx)
+    mac_b = macs.astype(">u8").view(np.uint8).reshape(n_pkts, 8)[:, 2:]
+    ips = (np.uint64(ip2u32("10.0.0.0") + 2) + idx).astype(np.uint32)
+    ip_b = ips.astype(">u4").view(np.uint8).reshape(n_pkts, 4)
+    sports = (40000 + (idx & np.uint64(0x3F))).astype(np.uint16)
+    sport_b = sports.astype(">u2").view(np.uint8).reshape(n_pkts, 2)
+
+    d = ~is_dhcp
+    nd = int(d.sum())
+    if nd:
+        data[d, :64] = udp_t
+        data[np.nonzero(d)[0][:, None], np.arange(6, 12)] = mac_b[d]
+        data[np.nonzero(d)[0][:, None], np.arange(26, 30)] = ip_b[d]
+        data[np.nonzero(d)[0][:, None], np.arange(34, 36)] = sport_b[d]
+        lens[d] = 64
+    h = is_dhcp
+    nh = int(h.sum())
+    if nh:
+        half = rng.random(n_pkts) < 0.5
+        t1 = h & half
+        t2 = h & ~half
+        for mask, t in ((t1, dhcp_t), (t2, dhcp_req_t)):
+            k = int(mask.sum())
+            if not k:
+                continue
+            rows = np.nonzero(mask)[0]
+            data[rows, :len(t)] = t
+            data[rows[:, None], np.arange(6, 12)] = mac_b[mask]      # eth src
+            data[rows[:, None], np.arange(70, 76)] = mac_b[mask]     # chaddr
+            lens[mask] = len(t)
+    return data, lens
+
+
+def log(rank, msg):
+    if rank == 0:
+        print(msg, file=sys.stderr, flush=True)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=262144,
+                    help="packets injected per GPU per step")
+    ap.add_argument("--subs", type=int, default=1_000_000)
+    ap.add_argument("--dhcp-frac", type=float, default=0.1)
+    ap.add_argument("--stride", type=int, default=512)
+    ap.add_argument("--lat-batch", type=int, default=2048)
+    ap.add_argument("--lat-reps", type=int, default=64)
+    ap.add_argument("--no-latency", action="store_true")
+    args = ap.parse_args()
+
+    import torch
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    distributed = world > 1
+    if distributed:
+        import torch.distributed as dist
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
+    device = f"cuda:{local_rank}"
+
+    now_sec = 1_700_000_000
+    from bng_amd.dataplane.launcher import HipLauncher
+    from bng_amd.parallel.sharding import exchange
+
+    t0 = time.perf_counter()
+    launcher = HipLauncher(
+        device, sub_log2=21, sess_log2=22, eim_log2=21, subnat_log2=21,
+        qos_log2=21, binding_log2=21)
+    launcher.set_server_config(b"\x02\x00\x00\x00\x00\x01",
+                               ip2u32("10.255.255.1"))
+    launcher.add_pool(1, ip2u32("10.0.0.0"), 8, ip2u32("10.255.255.1"),
+                      ip2u32("8.8.8.8"), ip2u32("1.1.1.1"), 86400)
+    launcher.set_antispoof_config(default_mode=abi.AS_DISABLED)
+    build_tables(launcher, rank, world, args.subs, now_sec)
+    log(rank, f"[bench] tables built in {time.perf_counter() - t0:.1f}s")
+
+    t0 = time.perf_counter()
+    data_np, lens_np = gen_batch(args.batch, args.subs, args.dhcp_frac,
+                                 args.stride, seed=1234 + rank)
+    pristine = torch.from_numpy(data_np).to(device)
+    work = torch.empty_like(pristine)
+    lens = torch.from_numpy(lens_np.view(np.int16)).to(device)
+    log(rank, f"[bench] batch generated in {time.perf_counter() - t0:.1f}s")
+
+    def step(now_ns):
+        work.copy_(pristine)                       # RX-DMA analog
+        d, l = work, lens
+        if distributed:
+            owner = launcher.shard_owner(d, l, world)
+            d, l = exchange(d, l, owner)
+        launcher.uplink(d, l, now_ns=now_ns, now_sec=now_sec)
+
+    base_ns = now_sec * 10**9
+    for w in range(args.warmup):
+        step(base_ns + w * 10**6)
+    if distributed:
+        import torch.distributed as dist
+        dist.barrier()
+    torch.cuda.synchronize()
+
+    t_start = time.perf_counter()
+    for k in range(args.steps):
+        step(base_ns + (args.warmup + k) * 10**6)
+    torch.cuda.synchronize()
+    if distributed:
+        import torch.distributed as dist
+        dist.barrier()
+    elapsed = time.perf_counter() - t_start
+    if distributed:
+        import torch.distributed as dist
+        e = torch.tensor([elapsed], device=device)
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    total_pkts = world * args.batch * args.steps
+    mpps = total_pkts / elapsed / 1e6
+    ms_per_step = elapsed / args.steps * 1e3
+
+    # p50 DHCP OFFER latency: small all-DHCP batches, submit -> reply ready
+    p50_us = p99_us = None
+    if not args.no_latency and rank == 0:
+        lat_np, lat_lens_np = gen_batch(args.lat_batch, args.subs, 1.0,
+                                        args.stride, seed=777)
+        lp = torch.from_numpy(lat_np).to(device)
+        lw = torch.empty_like(lp)
+        ll = torch.from_numpy(lat_lens_np.view(np.int16)).to(device)
+        lats = []
+        for r in range(args.lat_reps):
+            torch.cuda.synchronize()
+            t = time.perf_counter()
+            lw.copy_(lp)
+            launcher.dhcp_fastpath(lw, ll, now_sec=now_sec)
+            torch.cuda.synchronize()
+            lats.append((time.perf_counter() - t) * 1e6)
+        lats.sort()
+        p50_us = lats[len(lats) // 2]
+        p99_us = lats[int(len(lats) * 0.99)]
+        st = launcher.get_stats()
+        hits = st["fastpath_hits"]
+        log(rank, f"[bench] dhcp stats {st}")
+        log(rank, f"[bench] nat stats {launcher.nat_get_stats()}")
+        log(rank, f"[bench] p50 {p50_us:.1f}us p99 {p99_us:.1f}us "
+                  f"({args.lat_batch}-pkt DHCP batch)")
+
+    if rank == 0:
+        result = {
+            "metric": "mpps",
+            "value": round(mpps, 3),
+            "unit": "Mpps",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(mpps / BASELINE_MPPS, 3),
+            "dtype": "uint8",
+            "data": "synthetic",
+            "config": {
+                "model": "bng-uplink-pipeline "
+                         "(dhcp_fastpath+antispoof+nat44+qos)",
+                "benchmark": "Mpps + p50 DHCP OFFER latency, 64B mix, "
+                             "1M-sub table at 1/2/4/8 MI355X",
+                "global_batch": world * args.batch,
+                "seq_len": args.stride,
+                "parallelism": f"shard{world}-hashring-alltoall",
+                "n_subscribers": args.subs,
+                "dhcp_frac": args.dhcp_frac,
+                "p50_dhcp_offer_us": None if p50_us is None
+                else round(p50_us, 1),
+                "p99_dhcp_offer_us": None if p99_us is None
+                else round(p99_us, 1),
+                "baseline_mpps": BASELINE_MPPS,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if distributed:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

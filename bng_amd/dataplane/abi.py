@@ -138,7 +138,7 @@ class NatSession(C.Structure):
                 ("last_seen", C.c_uint64),
                 ("created", C.c_uint64), ("packets_out", C.c_uint64),
                 ("packets_in", C.c_uint64), ("bytes_out", C.c_uint64),
-                ("bytes_in", C.c_uint64), ("_pad2", C.c_uint64 * 3)]
+                ("bytes_in", C.c_uint64), ("_pad2", C.c_uint64 * 5)]
 
 
 class NatReverse(C.Structure):
@@ -230,7 +230,7 @@ EXPECTED_SIZES = {
     "bng_nat_reverse": (NatReverse, 48),
     "bng_eim_entry": (EimEntry, 48),
     "bng_subnat_entry": (SubnatEntry, 64),
-    "bng_nat_config": (NatConfig, 24 + 4 + 64 * 8 + 64 * 4),
+    "bng_nat_config": (NatConfig, 24 + 64 * 8 + 64 * 4),
     "bng_nat_log_entry": (NatLogEntry, 40),
     "bng_qos_bucket": (QosBucket, 64),
     "bng_binding_entry": (BindingEntry, 32),

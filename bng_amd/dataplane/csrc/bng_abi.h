@@ -147,7 +147,7 @@ typedef struct bng_nat_session {
   uint64_t packets_in;
   uint64_t bytes_out;
   uint64_t bytes_in;
-  uint64_t _pad2[3];
+  uint64_t _pad2[5];
 } bng_nat_session;  /* 128 B */
 
 /* Reverse map entry: external tuple -> internal tuple (ref nat44.c:228-233) */

@@ -1,0 +1,287 @@
+/* bng_ext.cpp — torch extension binding for the CDNA4 BNG dataplane.
+ *
+ * The Python-visible surface of the HIP dataplane: takes torch tensors
+ * (device table blobs + packet batches), validates shape/device, and
+ * launches the kernels on the current HIP stream so every table mutation
+ * is stream-ordered with packet processing (BPF-map consistency).
+ *
+ * Also exports the struct layout report that tests/test_abi.py checks
+ * against the Python ctypes mirrors (the analog of the reference's
+ * test/ebpf/maps_test.go struct-ABI tests).
+ */
+#include <torch/extension.h>
+#include <hip/hip_runtime.h>
+#include <c10/hip/HIPStream.h>
+
+#include "bng_abi.h"
+
+extern "C" {
+void bng_launch_dhcp(void*, const void*, void*, void*, int, int, const void*,
+                     uint32_t, const void*, uint32_t, const void*, void*,
+                     uint64_t, hipStream_t);
+void bng_launch_nat44(void*, const void*, void*, int, int, int, void*,
+                      uint32_t, void*, uint32_t, void*, uint32_t, void*,
+                      uint32_t, const void*, const void*, uint32_t, void*,
+                      void*, void*, uint64_t, hipStream_t);
+void bng_launch_qos(void*, const void*, void*, int, int, int, void*,
+                    uint32_t, void*, uint64_t, hipStream_t);
+void bng_launch_antispoof(void*, const void*, void*, int, int, const void*,
+                          uint32_t, const void*, void*, void*, void*,
+                          uint64_t, hipStream_t);
+void bng_launch_uplink(void*, const void*, void*, void*, int, int,
+                       const void*, uint32_t, const void*, uint32_t,
+                       const void*, void*, const void*, uint32_t,
+                       const void*, void*, void*, void*, void*, uint32_t,
+                       void*, uint32_t, void*, uint32_t, void*, uint32_t,
+                       const void*, const void*, uint32_t, void*, void*,
+                       void*, void*, uint32_t, void*, uint64_t, uint64_t,
+                       hipStream_t);
+void bng_launch_sub_upsert(void*, uint32_t, const void*, int, void*,
+                           hipStream_t);
+void bng_launch_sub_delete(void*, uint32_t, const void*, int, hipStream_t);
+void bng_launch_subnat_upsert(void*, uint32_t, const void*, int, void*,
+                              hipStream_t);
+void bng_launch_qos_upsert(void*, uint32_t, const void*, int, void*,
+                           hipStream_t);
+void bng_launch_binding_upsert(void*, uint32_t, const void*, int, void*,
+                               hipStream_t);
+void bng_launch_binding_delete(void*, uint32_t, const void*, int,
+                               hipStream_t);
+void bng_launch_nat_sweep(void*, uint32_t, void*, uint32_t, void*, uint32_t,
+                          uint64_t, uint64_t, uint64_t, uint64_t, uint64_t,
+                          void*, hipStream_t);
+void bng_launch_shard_owner(const void*, const void*, void*, int, int, int,
+                            hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_dev(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be a device tensor");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+uint32_t table_mask(const torch::Tensor& t, size_t entry, const char* name) {
+  size_t n = (size_t)t.numel() * t.element_size() / entry;
+  TORCH_CHECK(n > 0 && (n & (n - 1)) == 0, name,
+              " slot count must be a power of two, got ", n);
+  return (uint32_t)(n - 1);
+}
+
+void dhcp_fastpath(torch::Tensor data, torch::Tensor in_len,
+                   torch::Tensor out_len, torch::Tensor verdict,
+                   torch::Tensor subs, torch::Tensor pools,
+                   torch::Tensor cfg, torch::Tensor stats, int64_t now_sec) {
+  check_dev(data, "data"); check_dev(subs, "subs");
+  int n = in_len.numel();
+  int stride = data.size(1);
+  bng_launch_dhcp(data.data_ptr(), in_len.data_ptr(), out_len.data_ptr(),
+                  verdict.data_ptr(), n, stride, subs.data_ptr(),
+                  table_mask(subs, sizeof(bng_sub_entry), "subs"),
+                  pools.data_ptr(),
+                  (uint32_t)(pools.numel() * pools.element_size() /
+                             sizeof(bng_ip_pool)),
+                  cfg.data_ptr(), stats.data_ptr(), (uint64_t)now_sec,
+                  cur_stream());
+}
+
+void nat44(torch::Tensor data, torch::Tensor in_len, torch::Tensor verdict,
+           bool is_egress, torch::Tensor sessions, torch::Tensor reverse,
+           torch::Tensor eim, torch::Tensor subnat, torch::Tensor cfg,
+           torch::Tensor hairpin, int64_t n_hairpin, torch::Tensor stats,
+           torch::Tensor log_ring, torch::Tensor log_hdr, int64_t now_ns) {
+  check_dev(data, "data"); check_dev(sessions, "sessions");
+  int n = in_len.numel();
+  int stride = data.size(1);
+  bng_launch_nat44(
+      data.data_ptr(), in_len.data_ptr(), verdict.data_ptr(), n, stride,
+      is_egress ? 1 : 0, sessions.data_ptr(),
+      table_mask(sessions, sizeof(bng_nat_session), "sessions"),
+      reverse.data_ptr(), table_mask(reverse, sizeof(bng_nat_reverse), "reverse"),
+      eim.data_ptr(), table_mask(eim, sizeof(bng_eim_entry), "eim"),
+      subnat.data_ptr(), table_mask(subnat, sizeof(bng_subnat_entry), "subnat"),
+      cfg.data_ptr(), hairpin.data_ptr(), (uint32_t)n_hairpin,
+      stats.data_ptr(), log_ring.data_ptr(), log_hdr.data_ptr(),
+      (uint64_t)now_ns, cur_stream());
+}
+
+void qos(torch::Tensor data, torch::Tensor in_len, torch::Tensor verdict,
+         bool is_egress, torch::Tensor table, torch::Tensor stats,
+         int64_t now_ns) {
+  check_dev(data, "data"); check_dev(table, "table");
+  bng_launch_qos(data.data_ptr(), in_len.data_ptr(), verdict.data_ptr(),
+                 (int)in_len.numel(), (int)data.size(1), is_egress ? 1 : 0,
+                 table.data_ptr(),
+                 table_mask(table, sizeof(bng_qos_bucket), "qos"),
+                 stats.data_ptr(), (uint64_t)now_ns, cur_stream());
+}
+
+void antispoof(torch::Tensor data, torch::Tensor in_len,
+               torch::Tensor verdict, torch::Tensor bindings,
+               torch::Tensor cfg, torch::Tensor stats, torch::Tensor ring,
+               torch::Tensor hdr, int64_t now_ns) {
+  check_dev(data, "data"); check_dev(bindings, "bindings");
+  bng_launch_antispoof(
+      data.data_ptr(), in_len.data_ptr(), verdict.data_ptr(),
+      (int)in_len.numel(), (int)data.size(1), bindings.data_ptr(),
+      table_mask(bindings, sizeof(bng_binding_entry), "bindings"),
+      cfg.data_ptr(), stats.data_ptr(), ring.data_ptr(), hdr.data_ptr(),
+      (uint64_t)now_ns, cur_stream());
+}
+
+void uplink_pipeline(torch::Tensor data, torch::Tensor in_len,
+                     torch::Tensor out_len, torch::Tensor verdict,
+                     torch::Tensor subs, torch::Tensor pools,
+                     torch::Tensor scfg, torch::Tensor dhcp_stats,
+                     torch::Tensor bindings, torch::Tensor acfg,
+                     torch::Tensor as_stats, torch::Tensor spoof_ring,
+                     torch::Tensor spoof_hdr, torch::Tensor sessions,
+                     torch::Tensor reverse, torch::Tensor eim,
+                     torch::Tensor subnat, torch::Tensor ncfg,
+                     torch::Tensor hairpin, int64_t n_hairpin,
+                     torch::Tensor nat_stats, torch::Tensor log_ring,
+                     torch::Tensor log_hdr, torch::Tensor qos_in,
+                     torch::Tensor qos_stats, int64_t now_ns,
+                     int64_t now_sec) {
+  check_dev(data, "data");
+  bng_launch_uplink(
+      data.data_ptr(), in_len.data_ptr(), out_len.data_ptr(),
+      verdict.data_ptr(), (int)in_len.numel(), (int)data.size(1),
+      subs.data_ptr(), table_mask(subs, sizeof(bng_sub_entry), "subs"),
+      pools.data_ptr(),
+      (uint32_t)(pools.numel() * pools.element_size() / sizeof(bng_ip_pool)),
+      scfg.data_ptr(), dhcp_stats.data_ptr(), bindings.data_ptr(),
+      table_mask(bindings, sizeof(bng_binding_entry), "bindings"),
+      acfg.data_ptr(), as_stats.data_ptr(), spoof_ring.data_ptr(),
+      spoof_hdr.data_ptr(), sessions.data_ptr(),
+      table_mask(sessions, sizeof(bng_nat_session), "sessions"),
+      reverse.data_ptr(),
+      table_mask(reverse, sizeof(bng_nat_reverse), "reverse"),
+      eim.data_ptr(), table_mask(eim, sizeof(bng_eim_entry), "eim"),
+      subnat.data_ptr(),
+      table_mask(subnat, sizeof(bng_subnat_entry), "subnat"),
+      ncfg.data_ptr(), hairpin.data_ptr(), (uint32_t)n_hairpin,
+      nat_stats.data_ptr(), log_ring.data_ptr(), log_hdr.data_ptr(),
+      qos_in.data_ptr(), table_mask(qos_in, sizeof(bng_qos_bucket), "qos"),
+      qos_stats.data_ptr(), (uint64_t)now_ns, (uint64_t)now_sec,
+      cur_stream());
+}
+
+void sub_upsert(torch::Tensor table, torch::Tensor batch, torch::Tensor rc) {
+  check_dev(table, "table"); check_dev(batch, "batch");
+  int n = (int)(batch.numel() * batch.element_size() / sizeof(bng_sub_entry));
+  bng_launch_sub_upsert(table.data_ptr(),
+                        table_mask(table, sizeof(bng_sub_entry), "subs"),
+                        batch.data_ptr(), n, rc.data_ptr(), cur_stream());
+}
+void sub_delete(torch::Tensor table, torch::Tensor keys) {
+  bng_launch_sub_delete(table.data_ptr(),
+                        table_mask(table, sizeof(bng_sub_entry), "subs"),
+                        keys.data_ptr(), (int)keys.numel(), cur_stream());
+}
+void subnat_upsert(torch::Tensor table, torch::Tensor batch,
+                   torch::Tensor rc) {
+  int n = (int)(batch.numel() * batch.element_size() /
+                sizeof(bng_subnat_entry));
+  bng_launch_subnat_upsert(
+      table.data_ptr(), table_mask(table, sizeof(bng_subnat_entry), "subnat"),
+      batch.data_ptr(), n, rc.data_ptr(), cur_stream());
+}
+void qos_upsert(torch::Tensor table, torch::Tensor batch, torch::Tensor rc) {
+  int n = (int)(batch.numel() * batch.element_size() /
+                sizeof(bng_qos_bucket));
+  bng_launch_qos_upsert(table.data_ptr(),
+                        table_mask(table, sizeof(bng_qos_bucket), "qos"),
+                        batch.data_ptr(), n, rc.data_ptr(), cur_stream());
+}
+void binding_upsert(torch::Tensor table, torch::Tensor batch,
+                    torch::Tensor rc) {
+  int n = (int)(batch.numel() * batch.element_size() /
+                sizeof(bng_binding_entry));
+  bng_launch_binding_upsert(
+      table.data_ptr(),
+      table_mask(table, sizeof(bng_binding_entry), "bindings"),
+      batch.data_ptr(), n, rc.data_ptr(), cur_stream());
+}
+void binding_delete(torch::Tensor table, torch::Tensor keys) {
+  bng_launch_binding_delete(
+      table.data_ptr(),
+      table_mask(table, sizeof(bng_binding_entry), "bindings"),
+      keys.data_ptr(), (int)keys.numel(), cur_stream());
+}
+
+void nat_sweep(torch::Tensor sessions, torch::Tensor reverse,
+               torch::Tensor subnat, int64_t now_ns, int64_t udp_to,
+               int64_t tcp_est_to, int64_t tcp_tr_to, int64_t icmp_to,
+               torch::Tensor stats) {
+  uint32_t n_slots = (uint32_t)(sessions.numel() * sessions.element_size() /
+                                sizeof(bng_nat_session));
+  bng_launch_nat_sweep(
+      sessions.data_ptr(), n_slots, reverse.data_ptr(),
+      table_mask(reverse, sizeof(bng_nat_reverse), "reverse"),
+      subnat.data_ptr(),
+      table_mask(subnat, sizeof(bng_subnat_entry), "subnat"),
+      (uint64_t)now_ns, (uint64_t)udp_to, (uint64_t)tcp_est_to,
+      (uint64_t)tcp_tr_to, (uint64_t)icmp_to, stats.data_ptr(),
+      cur_stream());
+}
+
+void shard_owner(torch::Tensor data, torch::Tensor in_len,
+                 torch::Tensor owner, int64_t n_shards) {
+  bng_launch_shard_owner(data.data_ptr(), in_len.data_ptr(),
+                         owner.data_ptr(), (int)in_len.numel(),
+                         (int)data.size(1), (int)n_shards, cur_stream());
+}
+
+py::dict layout_report() {
+  py::dict d;
+#define SZ(T) d[#T] = sizeof(T)
+  SZ(bng_sub_entry); SZ(bng_ip_pool); SZ(bng_server_config);
+  SZ(bng_nat_tuple); SZ(bng_nat_session); SZ(bng_nat_reverse);
+  SZ(bng_eim_entry); SZ(bng_subnat_entry); SZ(bng_nat_config);
+  SZ(bng_nat_log_entry); SZ(bng_qos_bucket); SZ(bng_binding_entry);
+  SZ(bng_antispoof_config); SZ(bng_spoof_event); SZ(bng_ring_header);
+#undef SZ
+  py::dict off;
+  off["sub_entry.lease_expiry"] = offsetof(bng_sub_entry, lease_expiry);
+  off["nat_session.last_seen"] = offsetof(bng_nat_session, last_seen);
+  off["nat_session.created"] = offsetof(bng_nat_session, created);
+  off["nat_session.ready"] = offsetof(bng_nat_session, ready);
+  off["eim_entry.created"] = offsetof(bng_eim_entry, created);
+  off["subnat_entry.next_port"] = offsetof(bng_subnat_entry, next_port);
+  off["subnat_entry.sessions_active"] =
+      offsetof(bng_subnat_entry, sessions_active);
+  off["qos_bucket.tokens"] = offsetof(bng_qos_bucket, tokens);
+  off["qos_bucket.last_update"] = offsetof(bng_qos_bucket, last_update);
+  off["binding_entry.ipv6_addr"] = offsetof(bng_binding_entry, ipv6_addr);
+  off["nat_config.private_net"] = offsetof(bng_nat_config, private_net);
+  off["nat_config.alg_key"] = offsetof(bng_nat_config, alg_key);
+  off["antispoof_config.allowed_net"] =
+      offsetof(bng_antispoof_config, allowed_net);
+  off["spoof_event.spoofed_ip"] = offsetof(bng_spoof_event, spoofed_ip);
+  d["offsets"] = off;
+  return d;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("dhcp_fastpath", &dhcp_fastpath);
+  m.def("nat44", &nat44);
+  m.def("qos", &qos);
+  m.def("antispoof", &antispoof);
+  m.def("uplink_pipeline", &uplink_pipeline);
+  m.def("sub_upsert", &sub_upsert);
+  m.def("sub_delete", &sub_delete);
+  m.def("subnat_upsert", &subnat_upsert);
+  m.def("qos_upsert", &qos_upsert);
+  m.def("binding_upsert", &binding_upsert);
+  m.def("binding_delete", &binding_delete);
+  m.def("nat_sweep", &nat_sweep);
+  m.def("shard_owner", &shard_owner);
+  m.def("layout_report", &layout_report);
+}

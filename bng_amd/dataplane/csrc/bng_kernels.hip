@@ -1,0 +1,1500 @@
+/* bng_kernels.hip — CDNA4 (gfx950) BNG dataplane kernels.
+ *
+ * MI355X-native re-design of the reference eBPF dataplane:
+ *   dhcp_fastpath_kernel  <- bpf/dhcp_fastpath.c:619-813 (XDP)
+ *   nat44_egress_kernel   <- bpf/nat44.c:565-802 (TC egress SNAT)
+ *   nat44_ingress_kernel  <- bpf/nat44.c:805-948 (TC ingress DNAT)
+ *   qos_kernel            <- bpf/qos_ratelimit.c:126-222 (TC token bucket)
+ *   antispoof_kernel      <- bpf/antispoof.c:189-293 (TC uRPF)
+ *   uplink_pipeline_kernel — fused antispoof+NAT44+QoS+DHCP single pass
+ *                            (the per-packet program chain the reference
+ *                            runs as four separate TC/XDP hooks)
+ *
+ * Execution model: one thread per packet, grid-stride, 256-thread blocks
+ * (4 waves).  Packet batches are fixed-stride slots in HBM; every lookup
+ * table is an open-addressing HBM hash table (bng_abi.h).  Per-packet
+ * ktime becomes one host-supplied batch timestamp.  Stats are wave-
+ * aggregated before one device-scope atomic per wave (bng_device.h).
+ *
+ * Differential-tested against bng_amd/dataplane/golden.py on random
+ * batches (tests/test_kernels_gpu.py).
+ */
+#include <hip/hip_runtime.h>
+#include "bng_abi.h"
+#include "bng_device.h"
+
+/* ==================================================== table primitives */
+
+BNG_DEV const bng_sub_entry* sub_lookup(const bng_sub_entry* t, uint32_t mask,
+                                        uint64_t key) {
+  uint32_t slot = (uint32_t)bng_mix64(key) & mask;
+  for (int i = 0; i < BNG_MAX_PROBE; ++i) {
+    const bng_sub_entry* e = &t[(slot + i) & mask];
+    uint64_t k = e->key;
+    if (k == key) return e;
+    if (k == BNG_KEY_EMPTY) return nullptr;
+  }
+  return nullptr;
+}
+
+BNG_DEV bng_subnat_entry* subnat_lookup(bng_subnat_entry* t, uint32_t mask,
+                                        uint32_t ip) {
+  if (ip == 0) return nullptr;
+  uint32_t slot = (uint32_t)bng_mix64(ip) & mask;
+  for (int i = 0; i < BNG_MAX_PROBE; ++i) {
+    bng_subnat_entry* e = &t[(slot + i) & mask];
+    uint32_t k = e->key_ip;
+    if (k == ip) return e;
+    if (k == 0) return nullptr;
+  }
+  return nullptr;
+}
+
+BNG_DEV bng_qos_bucket* qos_lookup(bng_qos_bucket* t, uint32_t mask,
+                                   uint32_t ip) {
+  if (ip == 0) return nullptr;
+  uint32_t slot = (uint32_t)bng_mix64(ip) & mask;
+  for (int i = 0; i < BNG_MAX_PROBE; ++i) {
+    bng_qos_bucket* e = &t[(slot + i) & mask];
+    if (e->key_ip == ip && e->valid) return e;
+    if (e->key_ip == 0) return nullptr;
+  }
+  return nullptr;
+}
+
+BNG_DEV const bng_binding_entry* binding_lookup(const bng_binding_entry* t,
+                                                uint32_t mask, uint64_t mac) {
+  if (mac == 0) return nullptr;
+  uint32_t slot = (uint32_t)bng_mix64(mac) & mask;
+  for (int i = 0; i < BNG_MAX_PROBE; ++i) {
+    const bng_binding_entry* e = &t[(slot + i) & mask];
+    uint64_t k = e->key_mac;
+    if (k == mac) return e;
+    if (k == BNG_KEY_EMPTY) return nullptr;
+  }
+  return nullptr;
+}
+
+/* find-or-claim for device-created entries (sessions / EIM / reverse).
+ * Returns entry and sets *claimed when this thread won the slot; on a
+ * sig match the caller must bng_wait_ready() before trusting fields. */
+template <typename E>
+BNG_DEV E* sig_find_or_claim(E* t, uint32_t mask, uint64_t sig,
+                             bool* claimed, bool* found) {
+  *claimed = false; *found = false;
+  uint32_t slot = (uint32_t)sig & mask;   /* sig is already mixed */
+  for (int i = 0; i < BNG_MAX_PROBE; ++i) {
+    E* e = &t[(slot + i) & mask];
+    uint64_t k = __hip_atomic_load(&e->sig, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+    if (k == sig) { *found = true; return e; }
+    if (k == BNG_KEY_EMPTY) {
+      uint64_t old = __hip_atomic_compare_exchange_strong(
+          &e->sig, &k, sig, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+          __HIP_MEMORY_SCOPE_AGENT) ? BNG_KEY_EMPTY : k;
+      if (old == BNG_KEY_EMPTY) { *claimed = true; return e; }
+      if (old == sig) { *found = true; return e; }
+      /* lost to a different key: fall through, keep probing */
+    }
+  }
+  return nullptr;
+}
+
+template <typename E>
+BNG_DEV E* sig_lookup(E* t, uint32_t mask, uint64_t sig) {
+  uint32_t slot = (uint32_t)sig & mask;
+  for (int i = 0; i < BNG_MAX_PROBE; ++i) {
+    E* e = &t[(slot + i) & mask];
+    uint64_t k = __hip_atomic_load(&e->sig, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+    if (k == sig) return e;
+    if (k == BNG_KEY_EMPTY) return nullptr;
+  }
+  return nullptr;
+}
+
+BNG_DEV bool tuple_eq(const bng_nat_tuple& a, const bng_nat_tuple& b) {
+  return a.src_ip == b.src_ip && a.dst_ip == b.dst_ip &&
+         a.src_port == b.src_port && a.dst_port == b.dst_port &&
+         a.protocol == b.protocol;
+}
+
+/* ===================================================== parsed context */
+
+struct pktctx {
+  uint8_t* p;        /* slot base */
+  int len;
+  int ip_off;        /* -1 if not IPv4 */
+  int l4_off;
+  uint8_t proto;
+  uint32_t saddr, daddr;      /* big-endian byte order, as host ints */
+  uint16_t sport, dport;      /* host ints of the BE fields */
+  uint16_t icmp_id;
+  uint8_t tcp_flags;
+  /* vlan */
+  int vlan_offset;
+  uint16_t s_tag, c_tag;
+  bool tagged;
+  bool l4_ok;
+};
+
+/* Parse Ethernet [+VLAN/QinQ] + IPv4 + L4 ports.  NAT/QoS/antispoof paths
+ * in the reference parse untagged frames only (nat44.c:573-581); the DHCP
+ * path handles tags (dhcp_fastpath.c:352-428).  want_vlan selects. */
+BNG_DEV bool parse_pkt(pktctx& c, uint8_t* p, int len, bool want_vlan) {
+  c.p = p; c.len = len; c.ip_off = -1; c.vlan_offset = 0;
+  c.s_tag = c.c_tag = 0; c.tagged = false; c.l4_ok = false;
+  if (len < 14) return false;
+  uint16_t proto = ld_u16be(p + 12);
+  int off = 14;
+  if (want_vlan && (proto == 0x8100 || proto == 0x88A8)) {
+    if (len < off + 4) return false;
+    c.tagged = true;
+    c.s_tag = ld_u16be(p + off) & 0xFFF;
+    proto = ld_u16be(p + off + 2);
+    off += 4; c.vlan_offset = 4;
+    if (proto == 0x8100) {
+      if (len < off + 4) return false;
+      c.c_tag = ld_u16be(p + off) & 0xFFF;
+      proto = ld_u16be(p + off + 2);
+      off += 4; c.vlan_offset = 8;
+    }
+  }
+  if (proto != 0x0800 || len < off + 20) return false;
+  c.ip_off = off;
+  c.proto = p[off + 9];
+  c.saddr = ld_u32be(p + off + 12);
+  c.daddr = ld_u32be(p + off + 16);
+  int ihl = (p[off] & 0xF) * 4;
+  c.l4_off = off + ihl;
+  if (c.proto == 6 && len >= c.l4_off + 20) {
+    c.sport = ld_u16be(p + c.l4_off);
+    c.dport = ld_u16be(p + c.l4_off + 2);
+    c.tcp_flags = p[c.l4_off + 13];
+    c.l4_ok = true;
+  } else if (c.proto == 17 && len >= c.l4_off + 8) {
+    c.sport = ld_u16be(p + c.l4_off);
+    c.dport = ld_u16be(p + c.l4_off + 2);
+    c.l4_ok = true;
+  } else if (c.proto == 1 && len >= c.l4_off + 8) {
+    c.icmp_id = ld_u16be(p + c.l4_off + 4);
+    c.l4_ok = true;
+  }
+  return true;
+}
+
+/* ================================================== DHCP fast path K1 */
+
+struct dhcp_tables {
+  const bng_sub_entry* subs; uint32_t sub_mask;
+  const bng_ip_pool* pools;  uint32_t n_pools;
+  const bng_server_config* cfg;
+  unsigned long long* stats;
+  uint64_t now_sec;
+};
+
+/* stats flags gathered per packet, ballot-aggregated by the caller */
+struct dhcp_flags {
+  bool vlan, total, hit, miss, expired, error, o82, bcast, ucast;
+  BNG_DEV void clear() {
+    vlan = total = hit = miss = expired = error = o82 = bcast = ucast = false;
+  }
+};
+
+/* Scan DHCP options for msg-type (53) and option-82 circuit-id.  Full TLV
+ * scan — the GPU has no verifier, so we upgrade the reference's
+ * fixed-offset workaround (dhcp_fastpath.c:216-323) while keeping its
+ * bounds (64 options / 312 bytes, maps.h:19-22). */
+BNG_DEV void scan_dhcp_options(const uint8_t* p, int opt_off, int end,
+                               uint8_t* msg_type, const uint8_t** cid,
+                               int* cid_len) {
+  *msg_type = 0; *cid = nullptr; *cid_len = 0;
+  int i = opt_off;
+  int limit = min(end, opt_off + 312);
+  for (int iters = 0; iters < 64 && i < limit; ++iters) {
+    uint8_t code = p[i];
+    if (code == 0) { ++i; continue; }
+    if (code == 255) break;
+    if (i + 1 >= limit) break;
+    uint8_t ln = p[i + 1];
+    if (i + 2 + ln > limit) break;
+    if (code == 53 && ln == 1) *msg_type = p[i + 2];
+    else if (code == 82) {
+      int j = i + 2, sub_end = i + 2 + ln;
+      while (j + 2 <= sub_end) {
+        uint8_t sc = p[j], sl = p[j + 1];
+        if (j + 2 + sl > sub_end) break;
+        if (sc == 1 && sl > 0 && sl <= 32) { *cid = p + j + 2; *cid_len = sl; }
+        j += 2 + sl;
+      }
+    }
+    i += 2 + ln;
+  }
+}
+
+/* Per-packet DHCP fast path.  Returns verdict; *out_len set on TX. */
+BNG_DEV int dhcp_process(uint8_t* p, int len, int stride,
+                         const dhcp_tables& T, dhcp_flags& F,
+                         uint16_t* out_len) {
+  *out_len = (uint16_t)len;
+  pktctx c;
+  if (!parse_pkt(c, p, len, /*want_vlan=*/true)) {
+    if (c.tagged) F.vlan = true;
+    return BNG_PASS;
+  }
+  if (c.tagged) F.vlan = true;
+  if (c.ip_off < 0 || c.proto != 17 || !c.l4_ok) return BNG_PASS;
+  if (c.dport != 67) return BNG_PASS;
+  int dhcp_off = c.l4_off + 8;
+  if (len < dhcp_off + 240) return BNG_PASS;
+  if (p[dhcp_off] != 1) return BNG_PASS;               /* BOOTREQUEST */
+  if (ld_u32be(p + dhcp_off + 236) != 0x63825363u) return BNG_PASS;
+
+  F.total = true;
+
+  uint8_t msg_type; const uint8_t* cid; int cid_len;
+  scan_dhcp_options(p, dhcp_off + 240, len, &msg_type, &cid, &cid_len);
+  if (msg_type != 1 && msg_type != 3) { F.miss = true; return BNG_PASS; }
+
+  /* 3-way lookup: VLAN -> circuit-id -> MAC (ref :647-687) */
+  const bng_sub_entry* sub = nullptr;
+  if (c.tagged) {
+    uint64_t vk = BNG_KEY_VLAN | ((uint64_t)c.s_tag << 16) | c.c_tag;
+    sub = sub_lookup(T.subs, T.sub_mask, vk);
+  }
+  if (!sub && cid) {
+    uint8_t padded[32];
+    #pragma unroll
+    for (int i = 0; i < 32; ++i) padded[i] = (i < cid_len) ? cid[i] : 0;
+    uint64_t ck = BNG_KEY_CIRCUIT | (bng_fnv1a64(padded, 32) >> 2);
+    sub = sub_lookup(T.subs, T.sub_mask, ck);
+    if (sub) F.o82 = true;
+  }
+  if (!sub) {
+    uint64_t mac = 0;
+    #pragma unroll
+    for (int i = 0; i < 6; ++i) mac = (mac << 8) | p[dhcp_off + 28 + i];
+    sub = sub_lookup(T.subs, T.sub_mask, mac);
+  }
+  if (!sub) { F.miss = true; return BNG_PASS; }
+  if (T.now_sec > sub->lease_expiry) { F.expired = true; return BNG_PASS; }
+  if (sub->pool_id >= T.n_pools) { F.error = true; return BNG_PASS; }
+  const bng_ip_pool pool = T.pools[sub->pool_id];
+  if (!pool.valid) { F.error = true; return BNG_PASS; }
+  F.hit = true;
+
+  const bng_server_config cfg = *T.cfg;
+  uint8_t reply_type = (msg_type == 1) ? 2 : 5;     /* OFFER : ACK */
+  uint32_t giaddr = ld_u32be(p + dhcp_off + 24);
+  /* table IPs are stored as host ints equal to their wire (big-endian)
+   * interpretation — the same values ip2u32()/ld_u32be() produce */
+  uint32_t server_ip = cfg.server_ip ? cfg.server_ip : pool.gateway;
+
+  int ip_off = c.ip_off, udp_off = c.l4_off;
+  if (giaddr != 0) {   /* relayed: unicast to relay agent (ref :726-743) */
+    #pragma unroll
+    for (int i = 0; i < 6; ++i) p[i] = p[6 + i];
+    #pragma unroll
+    for (int i = 0; i < 6; ++i) p[6 + i] = cfg.server_mac[i];
+    st_u32be(p + ip_off + 12, server_ip);
+    st_u32be(p + ip_off + 16, giaddr);
+    st_u16be(p + udp_off, 67); st_u16be(p + udp_off + 2, 67);
+    F.ucast = true;
+  } else {
+    uint16_t flags = ld_u16be(p + dhcp_off + 10);
+    uint32_t ciaddr = ld_u32be(p + dhcp_off + 12);
+    bool use_bcast = (flags & 0x8000) || ciaddr == 0;
+    if (use_bcast) {
+      #pragma unroll
+      for (int i = 0; i < 6; ++i) p[i] = 0xFF;
+      F.bcast = true;
+    } else {
+      #pragma unroll
+      for (int i = 0; i < 6; ++i) p[i] = p[dhcp_off + 28 + i];
+      F.ucast = true;
+    }
+    #pragma unroll
+    for (int i = 0; i < 6; ++i) p[6 + i] = cfg.server_mac[i];
+    st_u32be(p + ip_off + 12, server_ip);
+    st_u32be(p + ip_off + 16, 0xFFFFFFFFu);
+    st_u16be(p + udp_off, 67); st_u16be(p + udp_off + 2, 68);
+  }
+  p[ip_off + 8] = 64;                 /* TTL */
+  st_u16be(p + udp_off + 6, 0);       /* UDP csum 0 (ref :741,755) */
+
+  p[dhcp_off] = 2;                    /* BOOTREPLY */
+  p[dhcp_off + 3] = 0;                /* hops */
+  st_u32be(p + dhcp_off + 16, sub->allocated_ip);
+  st_u32be(p + dhcp_off + 20, server_ip);
+  bng_zero(p + dhcp_off + 44, 192);   /* sname + file (ref :765-766) */
+
+  /* options 53/54/51/1/3/6/58/59/255 (ref build_dhcp_options :519-602) */
+  uint8_t* o = p + dhcp_off + 240;
+  int w = 0;
+  o[w++] = 53; o[w++] = 1; o[w++] = reply_type;
+  o[w++] = 54; o[w++] = 4; st_u32be(o + w, server_ip); w += 4;
+  o[w++] = 51; o[w++] = 4; st_u32be(o + w, pool.lease_time); w += 4;
+  uint32_t mask32 = (pool.prefix_len == 0) ? 0
+      : (pool.prefix_len >= 32) ? 0xFFFFFFFFu
+      : (0xFFFFFFFFu << (32 - pool.prefix_len));
+  o[w++] = 1;  o[w++] = 4; st_u32be(o + w, mask32); w += 4;
+  o[w++] = 3;  o[w++] = 4; st_u32be(o + w, pool.gateway); w += 4;
+  if (pool.dns_primary) {
+    uint8_t dl = pool.dns_secondary ? 8 : 4;
+    o[w++] = 6; o[w++] = dl;
+    st_u32be(o + w, pool.dns_primary); w += 4;
+    if (pool.dns_secondary) { st_u32be(o + w, pool.dns_secondary); w += 4; }
+  }
+  o[w++] = 58; o[w++] = 4; st_u32be(o + w, pool.lease_time / 2); w += 4;
+  o[w++] = 59; o[w++] = 4; st_u32be(o + w, (pool.lease_time * 7) / 8); w += 4;
+  o[w++] = 255;
+
+  int dhcp_len = 240 + w;
+  int udp_len = 8 + dhcp_len;
+  int ip_len = 20 + udp_len;
+  int total = 14 + c.vlan_offset + ip_len;
+  st_u16be(p + ip_off + 2, (uint16_t)ip_len);
+  st_u16be(p + udp_off + 4, (uint16_t)udp_len);
+  st_u16be(p + ip_off + 10, 0);
+  /* 10x u16 sum checksum (ref ip_checksum :488-503) */
+  uint32_t s = 0;
+  #pragma unroll
+  for (int i = 0; i < 10; ++i) s += ld_u16be(p + ip_off + 2 * i);
+  s = (s & 0xFFFF) + (s >> 16); s = (s & 0xFFFF) + (s >> 16);
+  st_u16be(p + ip_off + 10, (uint16_t)~s);
+
+  *out_len = (uint16_t)min(total, stride);
+  return BNG_TX;
+}
+
+BNG_DEV void dhcp_commit_stats(const dhcp_flags& F, unsigned long long* st) {
+  stat_inc(&st[BNG_ST_VLAN_PACKETS], F.vlan);
+  stat_inc(&st[BNG_ST_TOTAL_REQUESTS], F.total);
+  stat_inc(&st[BNG_ST_FASTPATH_HITS], F.hit);
+  stat_inc(&st[BNG_ST_FASTPATH_MISSES], F.miss);
+  stat_inc(&st[BNG_ST_CACHE_EXPIRED], F.expired);
+  stat_inc(&st[BNG_ST_ERRORS], F.error);
+  stat_inc(&st[BNG_ST_OPTION82_PRESENT], F.o82);
+  stat_inc(&st[BNG_ST_BROADCAST_REPLIES], F.bcast);
+  stat_inc(&st[BNG_ST_UNICAST_REPLIES], F.ucast);
+}
+
+__global__ void dhcp_fastpath_kernel(
+    uint8_t* __restrict__ data, const uint16_t* __restrict__ in_len,
+    uint16_t* __restrict__ out_len, uint8_t* __restrict__ verdict,
+    int n, int stride,
+    const bng_sub_entry* __restrict__ subs, uint32_t sub_mask,
+    const bng_ip_pool* __restrict__ pools, uint32_t n_pools,
+    const bng_server_config* __restrict__ cfg,
+    unsigned long long* __restrict__ stats, uint64_t now_sec) {
+  dhcp_tables T{subs, sub_mask, pools, n_pools, cfg, stats, now_sec};
+  int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  int nthreads = gridDim.x * blockDim.x;
+  for (int base = 0; base < n; base += nthreads) {
+    int pid = base + tid;
+    dhcp_flags F; F.clear();
+    if (pid < n) {
+      uint16_t ol = in_len[pid];
+      int v = dhcp_process(data + (size_t)pid * stride, in_len[pid], stride,
+                           T, F, &ol);
+      verdict[pid] = (uint8_t)v;
+      out_len[pid] = ol;
+    }
+    dhcp_commit_stats(F, stats);
+  }
+}
+
+/* ========================================================== NAT44  K2 */
+
+struct nat_tables {
+  bng_nat_session* sessions; uint32_t sess_mask;
+  bng_nat_reverse* reverse;  uint32_t rev_mask;
+  bng_eim_entry* eim;        uint32_t eim_mask;
+  bng_subnat_entry* subnat;  uint32_t subnat_mask;
+  const bng_nat_config* cfg;
+  const uint32_t* hairpin_ips; uint32_t n_hairpin;
+  unsigned long long* stats;
+  bng_nat_log_entry* log_ring; bng_ring_header* log_hdr;
+  uint64_t now_ns;
+};
+
+struct nat_flags {
+  bool snat, dnat, hairpin, dropped, passed, sess_created, sess_expired,
+       port_exh, eim_hit, eim_miss, alg;
+  BNG_DEV void clear() {
+    snat = dnat = hairpin = dropped = passed = sess_created = sess_expired =
+        port_exh = eim_hit = eim_miss = alg = false;
+  }
+};
+
+BNG_DEV bool nat_is_private(const bng_nat_config* cfg, uint32_t ip) {
+  for (uint32_t i = 0; i < cfg->n_private_ranges; ++i)
+    if ((ip & cfg->private_mask[i]) == cfg->private_net[i]) return true;
+  return false;
+}
+
+BNG_DEV bool nat_is_hairpin(const nat_tables& T, uint32_t ip) {
+  for (uint32_t i = 0; i < T.n_hairpin; ++i)
+    if (T.hairpin_ips[i] == ip) return true;
+  return false;
+}
+
+BNG_DEV bool nat_is_alg(const bng_nat_config* cfg, uint16_t port,
+                        uint8_t proto) {
+  uint32_t key = ((uint32_t)port << 16) | proto;
+  for (uint32_t i = 0; i < cfg->n_alg_ports; ++i)
+    if (cfg->alg_key[i] == key) return true;
+  return false;
+}
+
+BNG_DEV void nat_log_push(const nat_tables& T, uint32_t ev, uint32_t sub_id,
+                          uint32_t priv_ip, uint32_t pub_ip,
+                          uint16_t priv_port, uint16_t pub_port,
+                          uint32_t dest_ip, uint16_t dest_port,
+                          uint8_t proto, uint8_t flags) {
+  if (!T.log_ring) return;
+  uint32_t idx = ring_claim(T.log_hdr);
+  bng_nat_log_entry* e = &T.log_ring[idx & ((1u << BNG_LOG_RING_LOG2) - 1)];
+  e->timestamp = T.now_ns; e->event_type = ev; e->subscriber_id = sub_id;
+  e->private_ip = priv_ip; e->public_ip = pub_ip;
+  e->private_port = priv_port; e->public_port = pub_port;
+  e->dest_ip = dest_ip; e->dest_port = dest_port;
+  e->protocol = proto; e->flags = flags;
+}
+
+/* Port rotor (ref allocate_port_from_block nat44.c:408-466): atomic
+ * next_port bump, wrap, optional RTP parity, EIM-collision heuristic.
+ * The reference's wrap is a tolerated benign race; ours keeps it. */
+BNG_DEV uint16_t nat_alloc_port(const nat_tables& T, bng_subnat_entry* blk,
+                                bool parity, uint16_t orig_port,
+                                uint32_t internal_ip, uint8_t proto) {
+  uint8_t orig_parity = orig_port & 1;
+  for (int i = 0; i < 64; ++i) {
+    uint32_t r = atomicAdd(&blk->next_port, 1u);
+    uint16_t port = (uint16_t)r;
+    if (port > blk->port_end) port = blk->port_start;
+    if (r + 1 > blk->port_end)
+      atomicCAS(&blk->next_port, r + 1, (uint32_t)blk->port_start);
+    if (parity && (port & 1) != orig_parity) continue;
+    uint64_t esig = bng_eim_sig(internal_ip, port, proto);
+    bng_eim_entry* ex = sig_lookup(T.eim, T.eim_mask, esig);
+    if (ex) {
+      /* verify it really is this (ip,port,proto) */
+      if (bng_wait_ready(&ex->ready, 4096) && ex->internal_ip == internal_ip
+          && ex->internal_port == port && ex->protocol == proto)
+        continue;
+    }
+    return port;
+  }
+  return 0;
+}
+
+/* incremental checksum helpers (ref update_csum nat44.c:378-398) */
+BNG_DEV uint16_t csum_upd32(uint16_t csum, uint32_t oldv, uint32_t newv) {
+  uint32_t s = (~csum) & 0xFFFF;
+  s += (~oldv & 0xFFFF) + ((~oldv >> 16) & 0xFFFF);
+  s += (newv & 0xFFFF) + (newv >> 16);
+  s = (s & 0xFFFF) + (s >> 16);
+  s = (s & 0xFFFF) + (s >> 16);
+  return (uint16_t)~s;
+}
+BNG_DEV uint16_t csum_upd16(uint16_t csum, uint16_t oldv, uint16_t newv) {
+  uint32_t s = (~csum) & 0xFFFF;
+  s += (~oldv & 0xFFFF) + newv;
+  s = (s & 0xFFFF) + (s >> 16);
+  s = (s & 0xFFFF) + (s >> 16);
+  return (uint16_t)~s;
+}
+
+/* SNAT (ref nat44_egress nat44.c:565-802) */
+BNG_DEV int nat_egress_process(pktctx& c, const nat_tables& T, nat_flags& F) {
+  uint8_t* p = c.p;
+  if (c.ip_off < 0) return BNG_FWD;
+  const bng_nat_config* cfg = T.cfg;
+  if (!nat_is_private(cfg, c.saddr)) return BNG_FWD;
+  bng_subnat_entry* blk = subnat_lookup(T.subnat, T.subnat_mask, c.saddr);
+  if (!blk) { F.passed = true; return BNG_PASS; }
+
+  uint16_t sport, dport;
+  if (c.proto == 6) {
+    if (!c.l4_ok) return BNG_FWD;
+    sport = c.sport; dport = c.dport;
+    if (cfg->flags & (BNG_NAT_FLAG_ALG_FTP | BNG_NAT_FLAG_ALG_SIP)) {
+      if (nat_is_alg(cfg, dport, 6)) {
+        F.alg = true;
+        nat_log_push(T, BNG_LOG_ALG_TRIGGER, blk->subscriber_id, c.saddr, 0,
+                     sport, 0, c.daddr, dport, 6, 0);
+        return BNG_PASS;
+      }
+    }
+  } else if (c.proto == 17) {
+    if (!c.l4_ok) return BNG_FWD;
+    sport = c.sport; dport = c.dport;
+    if (cfg->flags & BNG_NAT_FLAG_ALG_SIP) {
+      if (nat_is_alg(cfg, dport, 17)) {
+        F.alg = true;
+        nat_log_push(T, BNG_LOG_ALG_TRIGGER, blk->subscriber_id, c.saddr, 0,
+                     sport, 0, c.daddr, dport, 17, 0);
+        return BNG_PASS;
+      }
+    }
+  } else if (c.proto == 1) {
+    if (!c.l4_ok) return BNG_FWD;
+    sport = c.icmp_id; dport = 0;
+  } else {
+    return BNG_FWD;
+  }
+
+  uint8_t is_hairpin = 0;
+  if ((cfg->flags & BNG_NAT_FLAG_HAIRPIN) && nat_is_hairpin(T, c.daddr)) {
+    is_hairpin = 1; F.hairpin = true;
+  }
+
+  uint64_t sig = bng_tuple_sig(c.saddr, c.daddr, sport, dport, c.proto);
+  bool claimed, found;
+  bng_nat_session* sess = sig_find_or_claim(T.sessions, T.sess_mask, sig,
+                                            &claimed, &found);
+  if (!sess) { F.passed = true; return BNG_PASS; }  /* table section full */
+
+  uint32_t nat_ip; uint16_t nat_port;
+  if (found) {
+    if (!bng_wait_ready(&sess->ready, 8192)) { F.passed = true; return BNG_PASS; }
+    /* sig collision with a different tuple: fall back to slow path (rare) */
+    bng_nat_tuple k{c.saddr, c.daddr, sport, dport, c.proto, {0, 0, 0}};
+    if (!tuple_eq(sess->key, k)) { F.passed = true; return BNG_PASS; }
+    nat_ip = sess->nat_ip; nat_port = sess->nat_port;
+    __hip_atomic_store(&sess->last_seen, T.now_ns, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_AGENT);
+    atomicAdd((unsigned long long*)&sess->packets_out, 1ull);
+    atomicAdd((unsigned long long*)&sess->bytes_out,
+              (unsigned long long)c.len);
+  } else {
+    /* we claimed the slot: allocate mapping (EIM or fresh port) */
+    bool parity = (cfg->flags & BNG_NAT_FLAG_PARITY) != 0;
+    bool have = false;
+    if (cfg->flags & BNG_NAT_FLAG_EIM) {
+      uint64_t esig = bng_eim_sig(c.saddr, sport, c.proto);
+      bool eclaimed, efound;
+      bng_eim_entry* eim = sig_find_or_claim(T.eim, T.eim_mask, esig,
+                                             &eclaimed, &efound);
+      if (eim && efound) {
+        if (bng_wait_ready(&eim->ready, 8192) && eim->internal_ip == c.saddr
+            && eim->internal_port == sport && eim->protocol == c.proto) {
+          __hip_atomic_store(&eim->last_used, T.now_ns, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);
+          atomicAdd(&eim->ref_count, 1u);
+          F.eim_hit = true;
+          nat_ip = eim->external_ip; nat_port = eim->external_port;
+          have = true;
+        }
+      } else if (eim && eclaimed) {
+        uint16_t ap = nat_alloc_port(T, blk, parity, sport, c.saddr, c.proto);
+        if (ap == 0) {
+          /* release the claimed EIM slot as tombstone, drop the packet */
+          __hip_atomic_store(&eim->sig, BNG_KEY_TOMBSTONE, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);
+          __hip_atomic_store(&sess->sig, BNG_KEY_TOMBSTONE, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);
+          F.port_exh = true; F.dropped = true;
+          nat_log_push(T, BNG_LOG_PORT_EXHAUSTION, blk->subscriber_id,
+                       c.saddr, blk->public_ip, sport, 0, c.daddr, dport,
+                       c.proto, 0);
+          return BNG_DROP;
+        }
+        eim->internal_ip = c.saddr; eim->internal_port = sport;
+        eim->protocol = c.proto;
+        eim->external_ip = blk->public_ip;
+        eim->external_port = ap;
+        eim->created = T.now_ns; eim->last_used = T.now_ns;
+        eim->ref_count = 1; eim->flags = 0;
+        bng_publish_ready(&eim->ready);
+        F.eim_miss = true;
+        nat_ip = eim->external_ip;
+        nat_port = ap;
+        have = true;
+      }
+    }
+    if (!have) {
+      uint16_t ap = nat_alloc_port(T, blk, parity, sport, c.saddr, c.proto);
+      if (ap == 0) {
+        __hip_atomic_store(&sess->sig, BNG_KEY_TOMBSTONE, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+        F.port_exh = true; F.dropped = true;
+        nat_log_push(T, BNG_LOG_PORT_EXHAUSTION, blk->subscriber_id, c.saddr,
+                     blk->public_ip, sport, 0, c.daddr, dport, c.proto, 0);
+        return BNG_DROP;
+      }
+      nat_ip = blk->public_ip;
+      nat_port = ap;
+    }
+    /* fill the session (ref :711-730) */
+    sess->key = bng_nat_tuple{c.saddr, c.daddr, sport, dport, c.proto, {0,0,0}};
+    sess->nat_ip = nat_ip; sess->nat_port = nat_port; sess->orig_port = sport;
+    sess->orig_ip = c.saddr; sess->state = BNG_NAT_NEW;
+    sess->is_hairpin = is_hairpin;
+    sess->last_seen = T.now_ns; sess->created = T.now_ns;
+    sess->packets_out = 1; sess->packets_in = 0;
+    sess->bytes_out = c.len; sess->bytes_in = 0;
+    bng_publish_ready(&sess->ready);
+
+    /* reverse mapping for DNAT (ref :732-740) */
+    uint64_t rsig = bng_tuple_sig(c.daddr, nat_ip, dport, nat_port, c.proto);
+    bool rclaimed, rfound;
+    bng_nat_reverse* rev = sig_find_or_claim(T.reverse, T.rev_mask, rsig,
+                                             &rclaimed, &rfound);
+    if (rev && (rclaimed || rfound)) {
+      if (rclaimed) {
+        rev->key = bng_nat_tuple{c.daddr, nat_ip, dport, nat_port, c.proto,
+                                 {0, 0, 0}};
+        rev->orig = sess->key;
+        bng_publish_ready(&rev->ready);
+      }
+    }
+    atomicAdd(&blk->sessions_active, 1u);
+    atomicAdd(&blk->sessions_total, 1u);
+    F.sess_created = true;
+    nat_log_push(T, BNG_LOG_SESSION_CREATE, blk->subscriber_id, c.saddr,
+                 nat_ip, sport, nat_port, c.daddr, dport, c.proto,
+                 is_hairpin);
+  }
+
+  /* SNAT rewrite + incremental checksums (ref :752-798) */
+  uint32_t old_ip = c.saddr;
+  st_u32be(p + c.ip_off + 12, nat_ip);
+  uint16_t ipck = ld_u16be(p + c.ip_off + 10);
+  st_u16be(p + c.ip_off + 10, csum_upd32(ipck, old_ip, nat_ip));
+  uint16_t nat_port_host = nat_port;
+  if (c.proto == 6) {
+    uint16_t old_port = ld_u16be(p + c.l4_off);
+    st_u16be(p + c.l4_off, nat_port_host);
+    uint16_t ck = ld_u16be(p + c.l4_off + 16);
+    ck = csum_upd32(ck, old_ip, nat_ip);
+    ck = csum_upd16(ck, old_port, nat_port_host);
+    st_u16be(p + c.l4_off + 16, ck);
+  } else if (c.proto == 17) {
+    uint16_t old_port = ld_u16be(p + c.l4_off);
+    st_u16be(p + c.l4_off, nat_port_host);
+    uint16_t ck = ld_u16be(p + c.l4_off + 6);
+    if (ck != 0) {
+      ck = csum_upd32(ck, old_ip, nat_ip);
+      ck = csum_upd16(ck, old_port, nat_port_host);
+      if (ck == 0) ck = 0xFFFF;
+      st_u16be(p + c.l4_off + 6, ck);
+    }
+  } else if (c.proto == 1) {
+    uint16_t old_id = ld_u16be(p + c.l4_off + 4);
+    st_u16be(p + c.l4_off + 4, nat_port_host);
+    uint16_t ck = ld_u16be(p + c.l4_off + 2);
+    st_u16be(p + c.l4_off + 2, csum_upd16(ck, old_id, nat_port_host));
+  }
+  F.snat = true;
+  return BNG_FWD;
+}
+
+/* DNAT (ref nat44_ingress nat44.c:805-948) */
+BNG_DEV int nat_ingress_process(pktctx& c, const nat_tables& T, nat_flags& F) {
+  uint8_t* p = c.p;
+  if (c.ip_off < 0) return BNG_FWD;
+  uint16_t sport, dport;
+  if (c.proto == 6 || c.proto == 17) {
+    if (!c.l4_ok) return BNG_FWD;
+    sport = c.sport; dport = c.dport;
+  } else if (c.proto == 1) {
+    if (!c.l4_ok) return BNG_FWD;
+    sport = 0; dport = c.icmp_id;
+  } else return BNG_FWD;
+
+  uint64_t rsig = bng_tuple_sig(c.saddr, c.daddr, sport, dport, c.proto);
+  bng_nat_reverse* rev = sig_lookup(T.reverse, T.rev_mask, rsig);
+  if (!rev || !bng_wait_ready(&rev->ready, 8192)) {
+    F.passed = true; return BNG_FWD;
+  }
+  bng_nat_tuple rk{c.saddr, c.daddr, sport, dport, c.proto, {0, 0, 0}};
+  if (!tuple_eq(rev->key, rk)) { F.passed = true; return BNG_FWD; }
+
+  uint64_t osig = bng_tuple_sig(rev->orig.src_ip, rev->orig.dst_ip,
+                                rev->orig.src_port, rev->orig.dst_port,
+                                rev->orig.protocol);
+  bng_nat_session* sess = sig_lookup(T.sessions, T.sess_mask, osig);
+  if (!sess || !bng_wait_ready(&sess->ready, 8192) ||
+      !tuple_eq(sess->key, rev->orig)) {
+    /* session expired: clean the reverse entry (ref :871-875) */
+    __hip_atomic_store(&rev->sig, BNG_KEY_TOMBSTONE, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_AGENT);
+    F.sess_expired = true;
+    return BNG_FWD;
+  }
+  __hip_atomic_store(&sess->last_seen, T.now_ns, __ATOMIC_RELAXED,
+                     __HIP_MEMORY_SCOPE_AGENT);
+  atomicAdd((unsigned long long*)&sess->packets_in, 1ull);
+  atomicAdd((unsigned long long*)&sess->bytes_in, (unsigned long long)c.len);
+  if (c.proto == 6) {
+    if (c.tcp_flags & 0x05)
+      sess->state = BNG_NAT_CLOSING;
+    else if (sess->state == BNG_NAT_NEW && (c.tcp_flags & 0x10))
+      sess->state = BNG_NAT_ESTABLISHED;
+  }
+
+  uint32_t old_ip = c.daddr, new_ip = sess->orig_ip;
+  st_u32be(p + c.ip_off + 16, new_ip);
+  uint16_t ipck = ld_u16be(p + c.ip_off + 10);
+  st_u16be(p + c.ip_off + 10, csum_upd32(ipck, old_ip, new_ip));
+  uint16_t new_port = sess->orig_port;   /* host-read value at egress */
+  if (c.proto == 6) {
+    uint16_t old_port = ld_u16be(p + c.l4_off + 2);
+    st_u16be(p + c.l4_off + 2, new_port);
+    uint16_t ck = ld_u16be(p + c.l4_off + 16);
+    ck = csum_upd32(ck, old_ip, new_ip);
+    ck = csum_upd16(ck, old_port, new_port);
+    st_u16be(p + c.l4_off + 16, ck);
+  } else if (c.proto == 17) {
+    uint16_t old_port = ld_u16be(p + c.l4_off + 2);
+    st_u16be(p + c.l4_off + 2, new_port);
+    uint16_t ck = ld_u16be(p + c.l4_off + 6);
+    if (ck != 0) {
+      ck = csum_upd32(ck, old_ip, new_ip);
+      ck = csum_upd16(ck, old_port, new_port);
+      if (ck == 0) ck = 0xFFFF;
+      st_u16be(p + c.l4_off + 6, ck);
+    }
+  } else if (c.proto == 1) {
+    uint16_t old_id = ld_u16be(p + c.l4_off + 4);
+    st_u16be(p + c.l4_off + 4, new_port);
+    uint16_t ck = ld_u16be(p + c.l4_off + 2);
+    st_u16be(p + c.l4_off + 2, csum_upd16(ck, old_id, new_port));
+  }
+  F.dnat = true;
+  return BNG_FWD;
+}
+
+BNG_DEV void nat_commit_stats(const nat_flags& F, unsigned long long* st) {
+  stat_inc(&st[BNG_NS_SNAT], F.snat);
+  stat_inc(&st[BNG_NS_DNAT], F.dnat);
+  stat_inc(&st[BNG_NS_HAIRPIN], F.hairpin);
+  stat_inc(&st[BNG_NS_DROPPED], F.dropped);
+  stat_inc(&st[BNG_NS_PASSED], F.passed);
+  stat_inc(&st[BNG_NS_SESS_CREATED], F.sess_created);
+  stat_inc(&st[BNG_NS_SESS_EXPIRED], F.sess_expired);
+  stat_inc(&st[BNG_NS_PORT_EXHAUSTION], F.port_exh);
+  stat_inc(&st[BNG_NS_EIM_HITS], F.eim_hit);
+  stat_inc(&st[BNG_NS_EIM_MISSES], F.eim_miss);
+  stat_inc(&st[BNG_NS_ALG_TRIGGERS], F.alg);
+}
+
+__global__ void nat44_kernel(
+    uint8_t* __restrict__ data, const uint16_t* __restrict__ in_len,
+    uint8_t* __restrict__ verdict, int n, int stride, int is_egress,
+    bng_nat_session* sessions, uint32_t sess_mask,
+    bng_nat_reverse* reverse, uint32_t rev_mask,
+    bng_eim_entry* eim, uint32_t eim_mask,
+    bng_subnat_entry* subnat, uint32_t subnat_mask,
+    const bng_nat_config* cfg,
+    const uint32_t* hairpin_ips, uint32_t n_hairpin,
+    unsigned long long* stats,
+    bng_nat_log_entry* log_ring, bng_ring_header* log_hdr,
+    uint64_t now_ns) {
+  nat_tables T{sessions, sess_mask, reverse, rev_mask, eim, eim_mask,
+               subnat, subnat_mask, cfg, hairpin_ips, n_hairpin, stats,
+               log_ring, log_hdr, now_ns};
+  int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  int nthreads = gridDim.x * blockDim.x;
+  for (int base = 0; base < n; base += nthreads) {
+    int pid = base + tid;
+    nat_flags F; F.clear();
+    if (pid < n) {
+      pktctx c;
+      parse_pkt(c, data + (size_t)pid * stride, in_len[pid], false);
+      int v = is_egress ? nat_egress_process(c, T, F)
+                        : nat_ingress_process(c, T, F);
+      verdict[pid] = (uint8_t)v;
+    }
+    nat_commit_stats(F, stats);
+  }
+}
+
+/* ============================================================ QoS  K3 */
+
+struct qos_flags { bool passed, dropped; uint32_t bytes; };
+
+/* Token-bucket check (ref token_bucket_check qos_ratelimit.c:70-104).
+ * Refill: the thread that CASes last_update old->now does one capped
+ * CAS-loop add; consume: one atomicAdd(-len) with undo-on-negative. */
+BNG_DEV bool qos_tb_check(bng_qos_bucket* tb, uint32_t pkt_len,
+                          uint64_t now_ns) {
+  uint64_t rate = tb->rate_bps;
+  if (rate == 0) return true;
+  uint64_t last = __hip_atomic_load(&tb->last_update, __ATOMIC_RELAXED,
+                                    __HIP_MEMORY_SCOPE_AGENT);
+  if (last != now_ns) {
+    if (__hip_atomic_compare_exchange_strong(
+            &tb->last_update, &last, now_ns, __ATOMIC_RELAXED,
+            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)) {
+      uint64_t add = ((now_ns - last) * (rate / 8)) / 1000000000ull;
+      int64_t burst = (int64_t)tb->burst_bytes;
+      /* capped add, bounded CAS loop */
+      for (int t = 0; t < 16; ++t) {
+        int64_t cur = (int64_t)__hip_atomic_load(
+            (uint64_t*)&tb->tokens, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        int64_t want = cur + (int64_t)add;
+        if (want > burst) want = burst;
+        if (want == cur) break;
+        uint64_t expect = (uint64_t)cur;
+        if (__hip_atomic_compare_exchange_strong(
+                (uint64_t*)&tb->tokens, &expect, (uint64_t)want,
+                __ATOMIC_RELAXED, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT))
+          break;
+      }
+    }
+  }
+  long long old = (long long)atomicAdd(
+      (unsigned long long*)&tb->tokens,
+      (unsigned long long)(-(int64_t)pkt_len));
+  if (old - (int64_t)pkt_len >= 0) return true;
+  atomicAdd((unsigned long long*)&tb->tokens, (unsigned long long)pkt_len);
+  return false;
+}
+
+BNG_DEV int qos_process(pktctx& c, bng_qos_bucket* table, uint32_t mask,
+                        bool egress, uint64_t now_ns, qos_flags& F) {
+  if (c.ip_off < 0) return BNG_FWD;
+  uint32_t key = egress ? c.daddr : c.saddr;
+  /* table keys are stored as the BE byte pattern read as host int */
+  bng_qos_bucket* tb = qos_lookup(table, mask, key);
+  if (!tb) return BNG_FWD;
+  bool ok = qos_tb_check(tb, c.len, now_ns);
+  F.bytes = c.len;
+  if (ok) { F.passed = true; return BNG_FWD; }
+  F.dropped = true;
+  return BNG_DROP;
+}
+
+__global__ void qos_kernel(
+    uint8_t* __restrict__ data, const uint16_t* __restrict__ in_len,
+    uint8_t* __restrict__ verdict, int n, int stride, int is_egress,
+    bng_qos_bucket* table, uint32_t mask,
+    unsigned long long* stats, uint64_t now_ns) {
+  int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  int nthreads = gridDim.x * blockDim.x;
+  for (int base = 0; base < n; base += nthreads) {
+    int pid = base + tid;
+    qos_flags F{false, false, 0};
+    if (pid < n) {
+      pktctx c;
+      parse_pkt(c, data + (size_t)pid * stride, in_len[pid], false);
+      verdict[pid] = (uint8_t)qos_process(c, table, mask, is_egress != 0,
+                                          now_ns, F);
+    }
+    stat_inc(&stats[BNG_QS_PKT_PASSED], F.passed);
+    stat_inc(&stats[BNG_QS_PKT_DROPPED], F.dropped);
+    stat_add(&stats[BNG_QS_BYTES_PASSED], F.bytes, F.passed);
+    stat_add(&stats[BNG_QS_BYTES_DROPPED], F.bytes, F.dropped);
+  }
+}
+
+/* ====================================================== antispoof  K4 */
+
+struct as_flags { bool allowed, dropped, logged, v4viol, v6viol; };
+
+BNG_DEV void spoof_log_push(bng_spoof_event* ring, bng_ring_header* hdr,
+                            const uint8_t* mac, uint8_t proto,
+                            uint32_t spoofed, uint32_t allowed,
+                            uint64_t now_ns) {
+  if (!ring) return;
+  uint32_t idx = ring_claim(hdr);
+  bng_spoof_event* e = &ring[idx & ((1u << BNG_SPOOF_RING_LOG2) - 1)];
+  e->timestamp = now_ns;
+  #pragma unroll
+  for (int i = 0; i < 6; ++i) e->src_mac[i] = mac[i];
+  e->protocol = proto;
+  e->spoofed_ip = spoofed; e->allowed_ip = allowed;
+  #pragma unroll
+  for (int i = 0; i < 16; ++i) { e->spoofed_ipv6[i] = 0; e->allowed_ipv6[i] = 0; }
+}
+
+/* uRPF source validation (ref antispoof_ingress antispoof.c:189-293),
+ * quirks preserved (see golden.py docstring). */
+BNG_DEV int antispoof_process(uint8_t* p, int len,
+                              const bng_binding_entry* bindings,
+                              uint32_t bmask,
+                              const bng_antispoof_config* cfg,
+                              bng_spoof_event* ring, bng_ring_header* hdr,
+                              uint64_t now_ns, as_flags& F) {
+  if (len < 14) return BNG_FWD;
+  uint64_t mac = 0;
+  #pragma unroll
+  for (int i = 0; i < 6; ++i) mac = (mac << 8) | p[6 + i];
+  const bng_binding_entry* b = binding_lookup(bindings, bmask, mac);
+  uint8_t mode = b ? b->mode : cfg->default_mode;
+  if (mode == BNG_AS_DISABLED) { F.allowed = true; return BNG_FWD; }
+  uint16_t proto = ld_u16be(p + 12);
+  if (proto == 0x0800) {
+    if (len < 34) return BNG_FWD;
+    uint32_t src_ip = ld_u32be(p + 14 + 12);
+    bool allowed = false;
+    uint32_t bound_ip = 0;
+    if (b && b->ipv4_valid) {
+      bound_ip = b->ipv4_addr;
+      if (mode == BNG_AS_STRICT || mode == BNG_AS_LOG_ONLY)
+        allowed = (src_ip == bound_ip);
+    } else if (mode == BNG_AS_LOOSE) {
+      for (uint32_t i = 0; i < cfg->n_allowed_ranges; ++i)
+        if ((src_ip & cfg->allowed_mask[i]) == cfg->allowed_net[i]) {
+          allowed = true; break;
+        }
+    }
+    if (!allowed) {
+      if (cfg->log_violations) {
+        spoof_log_push(ring, hdr, p + 6, 4, src_ip, bound_ip, now_ns);
+        F.logged = true;
+      }
+      if (mode == BNG_AS_LOG_ONLY) { F.allowed = true; return BNG_FWD; }
+      F.dropped = true; F.v4viol = true;
+      return BNG_DROP;
+    }
+    F.allowed = true; return BNG_FWD;
+  }
+  if (proto == 0x86DD) {
+    if (len < 14 + 40) return BNG_FWD;
+    bool allowed = false;
+    if (b && b->ipv6_valid) {
+      allowed = true;
+      #pragma unroll
+      for (int i = 0; i < 16; ++i)
+        if (p[14 + 8 + i] != b->ipv6_addr[i]) { allowed = false; break; }
+    } else if (mode == BNG_AS_LOOSE) {
+      allowed = true;
+    }
+    if (!allowed && mode != BNG_AS_LOG_ONLY) {
+      if (cfg->log_violations) {
+        spoof_log_push(ring, hdr, p + 6, 6, 0, 0, now_ns);
+        F.logged = true;
+      }
+      F.dropped = true; F.v6viol = true;
+      return BNG_DROP;
+    }
+    F.allowed = true; return BNG_FWD;
+  }
+  F.allowed = true;
+  return BNG_FWD;
+}
+
+__global__ void antispoof_kernel(
+    uint8_t* __restrict__ data, const uint16_t* __restrict__ in_len,
+    uint8_t* __restrict__ verdict, int n, int stride,
+    const bng_binding_entry* bindings, uint32_t bmask,
+    const bng_antispoof_config* cfg,
+    unsigned long long* stats,
+    bng_spoof_event* ring, bng_ring_header* hdr, uint64_t now_ns) {
+  int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  int nthreads = gridDim.x * blockDim.x;
+  for (int base = 0; base < n; base += nthreads) {
+    int pid = base + tid;
+    as_flags F{false, false, false, false, false};
+    if (pid < n) {
+      verdict[pid] = (uint8_t)antispoof_process(
+          data + (size_t)pid * stride, in_len[pid], bindings, bmask, cfg,
+          ring, hdr, now_ns, F);
+    }
+    stat_inc(&stats[BNG_AS_ALLOWED], F.allowed);
+    stat_inc(&stats[BNG_AS_DROPPED], F.dropped);
+    stat_inc(&stats[BNG_AS_LOGGED], F.logged);
+    stat_inc(&stats[BNG_AS_V4_VIOLATIONS], F.v4viol);
+    stat_inc(&stats[BNG_AS_V6_VIOLATIONS], F.v6viol);
+  }
+}
+
+/* ============================== fused uplink pipeline (the hot chain) */
+/* One pass per packet over the chain the reference runs as four separate
+ * kernel hooks: DHCP fast path for UDP:67, else antispoof -> NAT44 SNAT
+ * -> QoS ingress.  One parse, one packet-data round trip — the fusion the
+ * CDNA4 guide prescribes for HBM-bound pipelines. */
+__global__ void uplink_pipeline_kernel(
+    uint8_t* __restrict__ data, const uint16_t* __restrict__ in_len,
+    uint16_t* __restrict__ out_len, uint8_t* __restrict__ verdict,
+    int n, int stride,
+    /* dhcp */
+    const bng_sub_entry* subs, uint32_t sub_mask,
+    const bng_ip_pool* pools, uint32_t n_pools,
+    const bng_server_config* scfg, unsigned long long* dhcp_stats,
+    /* antispoof */
+    const bng_binding_entry* bindings, uint32_t bmask,
+    const bng_antispoof_config* acfg, unsigned long long* as_stats,
+    bng_spoof_event* spoof_ring, bng_ring_header* spoof_hdr,
+    /* nat */
+    bng_nat_session* sessions, uint32_t sess_mask,
+    bng_nat_reverse* reverse, uint32_t rev_mask,
+    bng_eim_entry* eim, uint32_t eim_mask,
+    bng_subnat_entry* subnat, uint32_t subnat_mask,
+    const bng_nat_config* ncfg, const uint32_t* hairpin_ips,
+    uint32_t n_hairpin, unsigned long long* nat_stats,
+    bng_nat_log_entry* log_ring, bng_ring_header* log_hdr,
+    /* qos (ingress: subscriber upload) */
+    bng_qos_bucket* qos_in, uint32_t qos_mask,
+    unsigned long long* qos_stats,
+    uint64_t now_ns, uint64_t now_sec) {
+  dhcp_tables DT{subs, sub_mask, pools, n_pools, scfg, dhcp_stats, now_sec};
+  nat_tables NT{sessions, sess_mask, reverse, rev_mask, eim, eim_mask,
+                subnat, subnat_mask, ncfg, hairpin_ips, n_hairpin, nat_stats,
+                log_ring, log_hdr, now_ns};
+  int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  int nthreads = gridDim.x * blockDim.x;
+  for (int base = 0; base < n; base += nthreads) {
+    int pid = base + tid;
+    dhcp_flags DF; DF.clear();
+    nat_flags NF; NF.clear();
+    as_flags AF{false, false, false, false, false};
+    qos_flags QF{false, false, 0};
+    if (pid < n) {
+      uint8_t* p = data + (size_t)pid * stride;
+      int len = in_len[pid];
+      uint16_t ol = (uint16_t)len;
+      int v;
+      pktctx c;
+      bool ip_ok = parse_pkt(c, p, len, /*want_vlan=*/true);
+      bool is_dhcp = ip_ok && c.ip_off >= 0 && c.proto == 17 && c.l4_ok &&
+                     c.dport == 67;
+      if (is_dhcp) {
+        v = dhcp_process(p, len, stride, DT, DF, &ol);
+      } else {
+        /* the reference's TC programs parse untagged frames only
+         * (nat44.c:573-581, antispoof.c:194-219): a tagged non-DHCP frame
+         * reads as non-IP there and is allowed through */
+        v = antispoof_process(p, len, bindings, bmask, acfg, spoof_ring,
+                              spoof_hdr, now_ns, AF);
+        if (v == BNG_FWD && ip_ok && !c.tagged) {
+          v = nat_egress_process(c, NT, NF);
+          if (v == BNG_FWD)
+            v = qos_process(c, qos_in, qos_mask, /*egress=*/false, now_ns, QF);
+        }
+      }
+      verdict[pid] = (uint8_t)v;
+      out_len[pid] = ol;
+    }
+    dhcp_commit_stats(DF, dhcp_stats);
+    nat_commit_stats(NF, nat_stats);
+    stat_inc(&as_stats[BNG_AS_ALLOWED], AF.allowed);
+    stat_inc(&as_stats[BNG_AS_DROPPED], AF.dropped);
+    stat_inc(&as_stats[BNG_AS_LOGGED], AF.logged);
+    stat_inc(&as_stats[BNG_AS_V4_VIOLATIONS], AF.v4viol);
+    stat_inc(&as_stats[BNG_AS_V6_VIOLATIONS], AF.v6viol);
+    stat_inc(&qos_stats[BNG_QS_PKT_PASSED], QF.passed);
+    stat_inc(&qos_stats[BNG_QS_PKT_DROPPED], QF.dropped);
+    stat_add(&qos_stats[BNG_QS_BYTES_PASSED], QF.bytes, QF.passed);
+    stat_add(&qos_stats[BNG_QS_BYTES_DROPPED], QF.bytes, QF.dropped);
+  }
+}
+
+/* ======================================= host CRUD kernels (pkg/ebpf) */
+/* Stream-ordered table upserts/deletes: the MI355X analog of BPF map
+ * update syscalls (pkg/ebpf/loader.go:349-661).  Launched on the same HIP
+ * stream as the dataplane kernels, so every batch sees a consistent table
+ * snapshot (BPF map semantics for free). */
+
+__global__ void sub_upsert_kernel(bng_sub_entry* t, uint32_t mask,
+                                  const bng_sub_entry* batch, int n,
+                                  int* rc) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  bng_sub_entry e = batch[i];
+  uint32_t slot = (uint32_t)bng_mix64(e.key) & mask;
+  int first_tomb = -1;
+  for (int k = 0; k < BNG_MAX_PROBE; ++k) {
+    bng_sub_entry* s = &t[(slot + k) & mask];
+    uint64_t cur = __hip_atomic_load(&s->key, __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+    if (cur == e.key) {   /* overwrite value */
+      s->pool_id = e.pool_id; s->allocated_ip = e.allocated_ip;
+      s->lease_expiry = e.lease_expiry; s->vlan_id = e.vlan_id;
+      s->client_class = e.client_class; s->flags = e.flags;
+      if (rc) rc[i] = 0;
+      return;
+    }
+    if (cur == BNG_KEY_TOMBSTONE && first_tomb < 0)
+      first_tomb = (int)((slot + k) & mask);
+    if (cur == BNG_KEY_EMPTY) {
+      int target = first_tomb >= 0 ? first_tomb : (int)((slot + k) & mask);
+      bng_sub_entry* d = &t[target];
+      uint64_t expect = first_tomb >= 0 ? BNG_KEY_TOMBSTONE : BNG_KEY_EMPTY;
+      if (__hip_atomic_compare_exchange_strong(
+              &d->key, &expect, e.key, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+              __HIP_MEMORY_SCOPE_AGENT)) {
+        d->pool_id = e.pool_id; d->allocated_ip = e.allocated_ip;
+        d->lease_expiry = e.lease_expiry; d->vlan_id = e.vlan_id;
+        d->client_class = e.client_class; d->flags = e.flags;
+        if (rc) rc[i] = 0;
+        return;
+      }
+      /* raced with another upsert in this batch; retry from here */
+      first_tomb = -1;
+      continue;
+    }
+  }
+  if (rc) rc[i] = -1;   /* probe bound exceeded: table too full */
+}
+
+__global__ void sub_delete_kernel(bng_sub_entry* t, uint32_t mask,
+                                  const uint64_t* keys, int n) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  uint64_t key = keys[i];
+  uint32_t slot = (uint32_t)bng_mix64(key) & mask;
+  for (int k = 0; k < BNG_MAX_PROBE; ++k) {
+    bng_sub_entry* s = &t[(slot + k) & mask];
+    uint64_t cur = s->key;
+    if (cur == key) {
+      __hip_atomic_store(&s->key, BNG_KEY_TOMBSTONE, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+      return;
+    }
+    if (cur == BNG_KEY_EMPTY) return;
+  }
+}
+
+__global__ void subnat_upsert_kernel(bng_subnat_entry* t, uint32_t mask,
+                                     const bng_subnat_entry* batch, int n,
+                                     int* rc) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  bng_subnat_entry e = batch[i];
+  uint32_t slot = (uint32_t)bng_mix64(e.key_ip) & mask;
+  for (int k = 0; k < BNG_MAX_PROBE; ++k) {
+    bng_subnat_entry* s = &t[(slot + k) & mask];
+    uint32_t cur = __hip_atomic_load(&s->key_ip, __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+    if (cur == e.key_ip || cur == 0) {
+      if (cur == 0) {
+        uint32_t expect = 0;
+        if (!__hip_atomic_compare_exchange_strong(
+                &s->key_ip, &expect, e.key_ip, __ATOMIC_RELAXED,
+                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT))
+          continue;
+      }
+      s->subscriber_id = e.subscriber_id; s->public_ip = e.public_ip;
+      s->port_start = e.port_start; s->port_end = e.port_end;
+      s->next_port = e.next_port; s->allocated_at = e.allocated_at;
+      s->block_size_log2 = e.block_size_log2; s->flags = e.flags;
+      if (rc) rc[i] = 0;
+      return;
+    }
+  }
+  if (rc) rc[i] = -1;
+}
+
+__global__ void qos_upsert_kernel(bng_qos_bucket* t, uint32_t mask,
+                                  const bng_qos_bucket* batch, int n,
+                                  int* rc) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  bng_qos_bucket e = batch[i];
+  uint32_t slot = (uint32_t)bng_mix64(e.key_ip) & mask;
+  for (int k = 0; k < BNG_MAX_PROBE; ++k) {
+    bng_qos_bucket* s = &t[(slot + k) & mask];
+    uint32_t cur = __hip_atomic_load(&s->key_ip, __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+    if (cur == e.key_ip || cur == 0) {
+      if (cur == 0) {
+        uint32_t expect = 0;
+        if (!__hip_atomic_compare_exchange_strong(
+                &s->key_ip, &expect, e.key_ip, __ATOMIC_RELAXED,
+                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT))
+          continue;
+      }
+      s->rate_bps = e.rate_bps; s->tokens = e.tokens;
+      s->last_update = e.last_update; s->burst_bytes = e.burst_bytes;
+      s->priority = e.priority;
+      s->valid = e.valid;     /* valid=0 upsert == remove policy */
+      if (rc) rc[i] = 0;
+      return;
+    }
+  }
+  if (rc) rc[i] = -1;
+}
+
+__global__ void binding_upsert_kernel(bng_binding_entry* t, uint32_t mask,
+                                      const bng_binding_entry* batch, int n,
+                                      int* rc) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  bng_binding_entry e = batch[i];
+  uint32_t slot = (uint32_t)bng_mix64(e.key_mac) & mask;
+  int first_tomb = -1;
+  for (int k = 0; k < BNG_MAX_PROBE; ++k) {
+    bng_binding_entry* s = &t[(slot + k) & mask];
+    uint64_t cur = __hip_atomic_load(&s->key_mac, __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+    if (cur == e.key_mac) {
+      s->ipv4_addr = e.ipv4_addr; s->ipv4_valid = e.ipv4_valid;
+      s->ipv6_valid = e.ipv6_valid; s->mode = e.mode;
+      #pragma unroll
+      for (int j = 0; j < 16; ++j) s->ipv6_addr[j] = e.ipv6_addr[j];
+      if (rc) rc[i] = 0;
+      return;
+    }
+    if (cur == BNG_KEY_TOMBSTONE && first_tomb < 0)
+      first_tomb = (int)((slot + k) & mask);
+    if (cur == BNG_KEY_EMPTY) {
+      int target = first_tomb >= 0 ? first_tomb : (int)((slot + k) & mask);
+      bng_binding_entry* d = &t[target];
+      uint64_t expect = first_tomb >= 0 ? BNG_KEY_TOMBSTONE : BNG_KEY_EMPTY;
+      if (__hip_atomic_compare_exchange_strong(
+              &d->key_mac, &expect, e.key_mac, __ATOMIC_RELAXED,
+              __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)) {
+        d->ipv4_addr = e.ipv4_addr; d->ipv4_valid = e.ipv4_valid;
+        d->ipv6_valid = e.ipv6_valid; d->mode = e.mode;
+        #pragma unroll
+        for (int j = 0; j < 16; ++j) d->ipv6_addr[j] = e.ipv6_addr[j];
+        if (rc) rc[i] = 0;
+        return;
+      }
+      first_tomb = -1;
+      continue;
+    }
+  }
+  if (rc) rc[i] = -1;
+}
+
+__global__ void binding_delete_kernel(bng_binding_entry* t, uint32_t mask,
+                                      const uint64_t* keys, int n) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  uint64_t key = keys[i];
+  uint32_t slot = (uint32_t)bng_mix64(key) & mask;
+  for (int k = 0; k < BNG_MAX_PROBE; ++k) {
+    bng_binding_entry* s = &t[(slot + k) & mask];
+    uint64_t cur = s->key_mac;
+    if (cur == key) {
+      __hip_atomic_store(&s->key_mac, BNG_KEY_TOMBSTONE, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+      return;
+    }
+    if (cur == BNG_KEY_EMPTY) return;
+  }
+}
+
+/* NAT session sweeper: reclaim timed-out sessions (the LRU-map analog,
+ * host-triggered; ref relies on BPF LRU eviction).  Walks the table in
+ * parallel; each thread owns a contiguous range of slots. */
+__global__ void nat_sweep_kernel(bng_nat_session* sessions, uint32_t n_slots,
+                                 bng_nat_reverse* reverse, uint32_t rev_mask,
+                                 bng_subnat_entry* subnat,
+                                 uint32_t subnat_mask,
+                                 uint64_t now_ns, uint64_t udp_to,
+                                 uint64_t tcp_est_to, uint64_t tcp_tr_to,
+                                 uint64_t icmp_to,
+                                 unsigned long long* stats) {
+  int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  int nthreads = gridDim.x * blockDim.x;
+  bool any = false;
+  for (uint32_t i = tid; i < n_slots; i += nthreads) {
+    bng_nat_session* s = &sessions[i];
+    uint64_t sig = s->sig;
+    if (sig == BNG_KEY_EMPTY || sig == BNG_KEY_TOMBSTONE || !s->ready)
+      continue;
+    uint64_t to = udp_to;
+    if (s->key.protocol == 6)
+      to = (s->state == BNG_NAT_ESTABLISHED) ? tcp_est_to : tcp_tr_to;
+    else if (s->key.protocol == 1)
+      to = icmp_to;
+    if (now_ns - s->last_seen < to && s->state != BNG_NAT_CLOSING) continue;
+    if (s->state == BNG_NAT_CLOSING && now_ns - s->last_seen < tcp_tr_to)
+      continue;
+    /* expire: tombstone session + its reverse entry, decrement counters */
+    uint64_t rsig = bng_tuple_sig(s->key.dst_ip, s->nat_ip, s->key.dst_port,
+                                  s->nat_port, s->key.protocol);
+    bng_nat_reverse* rev = sig_lookup(reverse, rev_mask, rsig);
+    if (rev)
+      __hip_atomic_store(&rev->sig, BNG_KEY_TOMBSTONE, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+    bng_subnat_entry* blk = subnat_lookup(subnat, subnat_mask, s->key.src_ip);
+    if (blk) atomicSub(&blk->sessions_active, 1u);
+    s->ready = 0;
+    __hip_atomic_store(&s->sig, BNG_KEY_TOMBSTONE, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_AGENT);
+    any = true;
+    stat_inc(&stats[BNG_NS_SESS_EXPIRED], true);
+  }
+  (void)any;
+}
+
+/* shard-owner computation for multi-GPU steering: owner by subscriber
+ * identity — source IP for upstream IPv4, chaddr MAC for DHCP.  Must match
+ * bng_amd/parallel/hashring.py owner_of_* bit-for-bit. */
+__global__ void shard_owner_kernel(const uint8_t* __restrict__ data,
+                                   const uint16_t* __restrict__ in_len,
+                                   int32_t* __restrict__ owner,
+                                   int n, int stride, int n_shards) {
+  int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  int nthreads = gridDim.x * blockDim.x;
+  for (int pid = tid; pid < n; pid += nthreads) {
+    const uint8_t* p = data + (size_t)pid * stride;
+    int len = in_len[pid];
+    uint64_t key = 0;
+    if (len >= 34 && ld_u16be(p + 12) == 0x0800) {
+      if (p[14 + 9] == 17 && len >= 42 && ld_u16be(p + 14 + 20 + 2) == 67 &&
+          len >= 42 + 240) {
+        /* DHCP: key by chaddr MAC */
+        const uint8_t* ch = p + 42 + 28;
+        #pragma unroll
+        for (int i = 0; i < 6; ++i) key = (key << 8) | ch[i];
+      } else {
+        key = ld_u32be(p + 14 + 12);   /* source IP */
+      }
+    } else if (len >= 14) {
+      #pragma unroll
+      for (int i = 0; i < 6; ++i) key = (key << 8) | p[6 + i];  /* src MAC */
+    }
+    owner[pid] = (int32_t)(bng_mix64(key) % (uint64_t)n_shards);
+  }
+}
+
+/* ============================= extern "C" launchers for the extension */
+
+static inline int pkt_grid(int n) {
+  int blocks = (n + 255) / 256;
+  /* memory-bound grid cap per CDNA4 guide G11: ~2048 blocks + grid-stride */
+  return blocks < 4096 ? blocks : 4096;
+}
+
+extern "C" {
+
+void bng_launch_dhcp(void* data, const void* in_len, void* out_len,
+                     void* verdict, int n, int stride,
+                     const void* subs, uint32_t sub_mask,
+                     const void* pools, uint32_t n_pools, const void* cfg,
+                     void* stats, uint64_t now_sec, hipStream_t s) {
+  hipLaunchKernelGGL(dhcp_fastpath_kernel, dim3(pkt_grid(n)), dim3(256), 0, s,
+      (uint8_t*)data, (const uint16_t*)in_len, (uint16_t*)out_len,
+      (uint8_t*)verdict, n, stride, (const bng_sub_entry*)subs, sub_mask,
+      (const bng_ip_pool*)pools, n_pools, (const bng_server_config*)cfg,
+      (unsigned long long*)stats, now_sec);
+}
+
+void bng_launch_nat44(void* data, const void* in_len, void* verdict, int n,
+                      int stride, int is_egress,
+                      void* sessions, uint32_t sess_mask,
+                      void* reverse, uint32_t rev_mask,
+                      void* eim, uint32_t eim_mask,
+                      void* subnat, uint32_t subnat_mask,
+                      const void* cfg, const void* hairpin,
+                      uint32_t n_hairpin, void* stats,
+                      void* log_ring, void* log_hdr, uint64_t now_ns,
+                      hipStream_t s) {
+  hipLaunchKernelGGL(nat44_kernel, dim3(pkt_grid(n)), dim3(256), 0, s,
+      (uint8_t*)data, (const uint16_t*)in_len, (uint8_t*)verdict, n, stride,
+      is_egress, (bng_nat_session*)sessions, sess_mask,
+      (bng_nat_reverse*)reverse, rev_mask, (bng_eim_entry*)eim, eim_mask,
+      (bng_subnat_entry*)subnat, subnat_mask, (const bng_nat_config*)cfg,
+      (const uint32_t*)hairpin, n_hairpin, (unsigned long long*)stats,
+      (bng_nat_log_entry*)log_ring, (bng_ring_header*)log_hdr, now_ns);
+}
+
+void bng_launch_qos(void* data, const void* in_len, void* verdict, int n,
+                    int stride, int is_egress, void* table, uint32_t mask,
+                    void* stats, uint64_t now_ns, hipStream_t s) {
+  hipLaunchKernelGGL(qos_kernel, dim3(pkt_grid(n)), dim3(256), 0, s,
+      (uint8_t*)data, (const uint16_t*)in_len, (uint8_t*)verdict, n, stride,
+      is_egress, (bng_qos_bucket*)table, mask, (unsigned long long*)stats,
+      now_ns);
+}
+
+void bng_launch_antispoof(void* data, const void* in_len, void* verdict,
+                          int n, int stride, const void* bindings,
+                          uint32_t bmask, const void* cfg, void* stats,
+                          void* ring, void* hdr, uint64_t now_ns,
+                          hipStream_t s) {
+  hipLaunchKernelGGL(antispoof_kernel, dim3(pkt_grid(n)), dim3(256), 0, s,
+      (uint8_t*)data, (const uint16_t*)in_len, (uint8_t*)verdict, n, stride,
+      (const bng_binding_entry*)bindings, bmask,
+      (const bng_antispoof_config*)cfg, (unsigned long long*)stats,
+      (bng_spoof_event*)ring, (bng_ring_header*)hdr, now_ns);
+}
+
+void bng_launch_uplink(void* data, const void* in_len, void* out_len,
+                       void* verdict, int n, int stride,
+                       const void* subs, uint32_t sub_mask,
+                       const void* pools, uint32_t n_pools,
+                       const void* scfg, void* dhcp_stats,
+                       const void* bindings, uint32_t bmask,
+                       const void* acfg, void* as_stats,
+                       void* spoof_ring, void* spoof_hdr,
+                       void* sessions, uint32_t sess_mask,
+                       void* reverse, uint32_t rev_mask,
+                       void* eim, uint32_t eim_mask,
+                       void* subnat, uint32_t subnat_mask,
+                       const void* ncfg, const void* hairpin,
+                       uint32_t n_hairpin, void* nat_stats,
+                       void* log_ring, void* log_hdr,
+                       void* qos_in, uint32_t qos_mask, void* qos_stats,
+                       uint64_t now_ns, uint64_t now_sec, hipStream_t s) {
+  hipLaunchKernelGGL(uplink_pipeline_kernel, dim3(pkt_grid(n)), dim3(256), 0,
+      s, (uint8_t*)data, (const uint16_t*)in_len, (uint16_t*)out_len,
+      (uint8_t*)verdict, n, stride,
+      (const bng_sub_entry*)subs, sub_mask, (const bng_ip_pool*)pools,
+      n_pools, (const bng_server_config*)scfg,
+      (unsigned long long*)dhcp_stats,
+      (const bng_binding_entry*)bindings, bmask,
+      (const bng_antispoof_config*)acfg, (unsigned long long*)as_stats,
+      (bng_spoof_event*)spoof_ring, (bng_ring_header*)spoof_hdr,
+      (bng_nat_session*)sessions, sess_mask, (bng_nat_reverse*)reverse,
+      rev_mask, (bng_eim_entry*)eim, eim_mask, (bng_subnat_entry*)subnat,
+      subnat_mask, (const bng_nat_config*)ncfg, (const uint32_t*)hairpin,
+      n_hairpin, (unsigned long long*)nat_stats,
+      (bng_nat_log_entry*)log_ring, (bng_ring_header*)log_hdr,
+      (bng_qos_bucket*)qos_in, qos_mask, (unsigned long long*)qos_stats,
+      now_ns, now_sec);
+}
+
+void bng_launch_sub_upsert(void* t, uint32_t mask, const void* batch, int n,
+                           void* rc, hipStream_t s) {
+  hipLaunchKernelGGL(sub_upsert_kernel, dim3((n + 255) / 256), dim3(256), 0,
+      s, (bng_sub_entry*)t, mask, (const bng_sub_entry*)batch, n, (int*)rc);
+}
+void bng_launch_sub_delete(void* t, uint32_t mask, const void* keys, int n,
+                           hipStream_t s) {
+  hipLaunchKernelGGL(sub_delete_kernel, dim3((n + 255) / 256), dim3(256), 0,
+      s, (bng_sub_entry*)t, mask, (const uint64_t*)keys, n);
+}
+void bng_launch_subnat_upsert(void* t, uint32_t mask, const void* batch,
+                              int n, void* rc, hipStream_t s) {
+  hipLaunchKernelGGL(subnat_upsert_kernel, dim3((n + 255) / 256), dim3(256),
+      0, s, (bng_subnat_entry*)t, mask, (const bng_subnat_entry*)batch, n,
+      (int*)rc);
+}
+void bng_launch_qos_upsert(void* t, uint32_t mask, const void* batch, int n,
+                           void* rc, hipStream_t s) {
+  hipLaunchKernelGGL(qos_upsert_kernel, dim3((n + 255) / 256), dim3(256), 0,
+      s, (bng_qos_bucket*)t, mask, (const bng_qos_bucket*)batch, n, (int*)rc);
+}
+void bng_launch_binding_upsert(void* t, uint32_t mask, const void* batch,
+                               int n, void* rc, hipStream_t s) {
+  hipLaunchKernelGGL(binding_upsert_kernel, dim3((n + 255) / 256), dim3(256),
+      0, s, (bng_binding_entry*)t, mask, (const bng_binding_entry*)batch, n,
+      (int*)rc);
+}
+void bng_launch_binding_delete(void* t, uint32_t mask, const void* keys,
+                               int n, hipStream_t s) {
+  hipLaunchKernelGGL(binding_delete_kernel, dim3((n + 255) / 256), dim3(256),
+      0, s, (bng_binding_entry*)t, mask, (const uint64_t*)keys, n);
+}
+
+void bng_launch_nat_sweep(void* sessions, uint32_t n_slots, void* reverse,
+                          uint32_t rev_mask, void* subnat,
+                          uint32_t subnat_mask, uint64_t now_ns,
+                          uint64_t udp_to, uint64_t tcp_est_to,
+                          uint64_t tcp_tr_to, uint64_t icmp_to, void* stats,
+                          hipStream_t s) {
+  hipLaunchKernelGGL(nat_sweep_kernel, dim3(2048), dim3(256), 0, s,
+      (bng_nat_session*)sessions, n_slots, (bng_nat_reverse*)reverse,
+      rev_mask, (bng_subnat_entry*)subnat, subnat_mask, now_ns, udp_to,
+      tcp_est_to, tcp_tr_to, icmp_to, (unsigned long long*)stats);
+}
+
+void bng_launch_shard_owner(const void* data, const void* in_len,
+                            void* owner, int n, int stride, int n_shards,
+                            hipStream_t s) {
+  hipLaunchKernelGGL(shard_owner_kernel, dim3(pkt_grid(n)), dim3(256), 0, s,
+      (const uint8_t*)data, (const uint16_t*)in_len, (int32_t*)owner, n,
+      stride, n_shards);
+}
+
+} /* extern "C" */
