@@ -42,6 +42,7 @@ def exchange(data, lens, owner, group=None) -> Tuple["object", "object"]:
     if world == 1:
         return data, lens
     stride = data.size(1)
+    assert stride % 8 == 0, "packet slot stride must be a multiple of 8"
     data_s, lens_s, send_counts, _ = bucket_by_owner(data, lens, owner, world)
 
     # exchange counts
@@ -52,14 +53,19 @@ def exchange(data, lens, owner, group=None) -> Tuple["object", "object"]:
     out_splits = recv_counts.tolist()
     m = sum(out_splits)
 
+    # ship packet bytes as int64 words (gloo rejects uint8/int16; identical
+    # bytes either way) — one contiguous all_to_all_single per tensor
+    words = stride // 8
     data_recv = data.new_empty((m, stride))
-    lens_recv = lens.new_empty((m,))
-    dist.all_to_all_single(data_recv.view(-1), data_s.contiguous().view(-1),
-                           [c * stride for c in out_splits],
-                           [c * stride for c in in_splits], group=group)
-    dist.all_to_all_single(lens_recv, lens_s.contiguous(), out_splits,
+    dist.all_to_all_single(data_recv.view(-1).view(torch.int64),
+                           data_s.contiguous().view(-1).view(torch.int64),
+                           [c * words for c in out_splits],
+                           [c * words for c in in_splits], group=group)
+    lens_recv64 = torch.empty(m, dtype=torch.int64, device=lens.device)
+    dist.all_to_all_single(lens_recv64,
+                           lens_s.contiguous().to(torch.int64), out_splits,
                            in_splits, group=group)
-    return data_recv, lens_recv
+    return data_recv, lens_recv64.to(lens.dtype)
 
 
 def allreduce_stats(*stat_tensors, group=None):

@@ -380,10 +380,16 @@ class TestUplinkPipeline:
         gpu, cpu = nat_pair()
         for l in (gpu, cpu):
             l.set_antispoof_config(default_mode=abi.AS_DISABLED)
-            l.set_qos_policy(ip2u32(PRIV), 0, 0, direction="ingress",
-                             now_ns=NOW_NS)
             l.add_subscriber(mac_bytes("aa:bb:cc:00:00:01"), 1,
                              ip2u32(PRIV), NOW_SEC + 600)
+            for k in range(64):
+                l.set_qos_policy(ip2u32(f"10.0.1.{k + 1}"), 0, 0,
+                                 direction="ingress", now_ns=NOW_NS)
+        # one flow per DISTINCT subscriber: port allocation per packet is
+        # then the deterministic first draw from that subscriber's own
+        # block (same-subscriber parallel flows draw rotor ports in
+        # nondeterministic order — covered by
+        # test_same_flow_many_packets_one_batch instead)
         frames = []
         for i in range(64):
             if i % 4 == 0:
@@ -391,7 +397,8 @@ class TestUplinkPipeline:
                                                  DHCP_REQUEST, xid=i))
             else:
                 frames.append(build_ipv4(
-                    "aa:bb:cc:00:00:01", "02:00:00:00:00:01", ip2u32(PRIV),
+                    "aa:bb:cc:00:00:01", "02:00:00:00:00:01",
+                    ip2u32(f"10.0.1.{i + 1}"),
                     ip2u32(DST), proto=17, sport=20000 + i, dport=53))
         data, lens = gpu.make_batch(frames)
         v, out_len = gpu.uplink(data, lens, now_ns=NOW_NS, now_sec=NOW_SEC)
