@@ -1,0 +1,184 @@
+"""Tests for nexus store/client/clset/http allocator (ref pkg/nexus)."""
+import threading
+import time
+
+import pytest
+
+from bng_amd.nexus.client import Client, NexusError, VLANAllocator
+from bng_amd.nexus.clset import CLSetHTTPServer, CLSetStore
+from bng_amd.nexus.http_allocator import (HTTPAllocator, NexusAllocatorServer,
+                                          NoAllocationError)
+from bng_amd.nexus.model import IPPool, ISPConfig, Subscriber
+from bng_amd.nexus.store import MemoryStore, TypedStore
+
+
+class TestMemoryStore:
+    def test_crud_and_list(self):
+        s = MemoryStore()
+        s.put("a/1", b"x")
+        s.put("a/2", b"y")
+        s.put("b/1", b"z")
+        assert s.get("a/1") == b"x"
+        assert set(s.list("a/")) == {"a/1", "a/2"}
+        s.delete("a/1")
+        assert s.get("a/1") is None
+
+    def test_watch(self):
+        s = MemoryStore()
+        events = []
+        cancel = s.watch("a/", lambda ev: events.append((ev.type, ev.key)))
+        s.put("a/1", b"x")
+        s.put("b/1", b"x")
+        s.delete("a/1")
+        assert events == [("put", "a/1"), ("delete", "a/1")]
+        cancel()
+        s.put("a/2", b"x")
+        assert len(events) == 2
+
+    def test_typed_store(self):
+        s = MemoryStore()
+        t = TypedStore(s, "subs")
+        t.put("s1", {"id": "s1", "n": 2})
+        assert t.get("s1") == {"id": "s1", "n": 2}
+        assert t.list() == {"s1": {"id": "s1", "n": 2}}
+
+
+class TestClient:
+    def make(self):
+        store = MemoryStore()
+        c = Client(store)
+        c.pools.put("pool-1", IPPool("pool-1", "10.1.0.0/24").to_dict())
+        c.isps.put("isp-1", ISPConfig("isp-1",
+                                      ipv4_pools=["pool-1"]).to_dict())
+        return store, c
+
+    def test_hashring_allocation_deterministic(self):
+        _, c = self.make()
+        c.save_subscriber(Subscriber("sub-1", isp_id="isp-1",
+                                     mac="aa:bb:cc:00:00:01"))
+        ip1 = c.allocate_ip_for_subscriber("sub-1")
+        # second call returns the stored IP (idempotent)
+        assert c.allocate_ip_for_subscriber("sub-1") == ip1
+        # the hash is deterministic: fresh client over same store agrees
+        assert Client.allocate_from_pool("10.1.0.0/24", "sub-1") == ip1
+        assert ip1.startswith("10.1.0.")
+
+    def test_dhcp_is_pure_read(self):
+        """The core invariant: allocation happens at auth time, DHCP-time
+        lookup never writes (ref README.md:19-33)."""
+        store, c = self.make()
+        c.save_subscriber(Subscriber("sub-2", isp_id="isp-1"))
+        assert c.lookup_subscriber_ip("sub-2") is None   # not authed yet
+        ip = c.allocate_ip_for_subscriber("sub-2")       # RADIUS time
+        writes_before = len(store.list("nexus/"))
+        assert c.lookup_subscriber_ip("sub-2") == ip     # DHCP time
+        assert len(store.list("nexus/")) == writes_before
+
+    def test_mac_lookup_and_watch_cache(self):
+        _, c = self.make()
+        c.start()
+        try:
+            c.save_subscriber(Subscriber("sub-3", isp_id="isp-1",
+                                         mac="AA:BB:CC:00:00:03"))
+            time.sleep(0.05)
+            sub = c.get_subscriber_by_mac("aa:bb:cc:00:00:03")
+            assert sub is not None and sub.id == "sub-3"
+        finally:
+            c.stop()
+
+    def test_missing_pool_errors(self):
+        _, c = self.make()
+        c.save_subscriber(Subscriber("sub-4"))
+        with pytest.raises(NexusError):
+            c.allocate_ip_for_subscriber("sub-4")
+
+    def test_vlan_allocator(self):
+        store, _ = self.make()
+        v = VLANAllocator(store, s_tag=100, c_tag_range=(2, 5))
+        s, c1 = v.allocate("sub-1")
+        assert (s, c1) == (100, 2)
+        assert v.allocate("sub-1") == (100, 2)       # idempotent
+        assert v.allocate("sub-2") == (100, 3)
+        v.release("sub-1")
+        assert v.allocate("sub-3") == (100, 2)       # reused
+
+
+class TestCLSet:
+    def test_lww_convergence(self):
+        a, b = CLSetStore("a"), CLSetStore("b")
+        a.add_peer(b)
+        a.put("k1", b"from-a")
+        b.put("k2", b"from-b")
+        a.sync_once()
+        assert a.get("k2") == b"from-b"
+        assert b.get("k1") == b"from-a"
+
+    def test_concurrent_writes_converge_identically(self):
+        a, b = CLSetStore("a"), CLSetStore("b")
+        a.put("k", b"va")
+        b.put("k", b"vb")            # same lamport, tie-broken by node id
+        a.merge(b.snapshot())
+        b.merge(a.snapshot())
+        assert a.get("k") == b.get("k") == b"vb"
+
+    def test_delete_wins_when_later(self):
+        a, b = CLSetStore("a"), CLSetStore("b")
+        a.put("k", b"x")
+        b.merge(a.snapshot())
+        b.delete("k")
+        a.merge(b.snapshot())
+        assert a.get("k") is None
+
+    def test_partition_then_merge(self):
+        """CRDT merge after partition (ref resilience partition_test
+        CRDTMergeAfterPartition scenario)."""
+        a, b = CLSetStore("a"), CLSetStore("b")
+        for i in range(10):
+            a.put(f"a/{i}", f"{i}".encode())
+            b.put(f"b/{i}", f"{i}".encode())
+        # partition: both keep writing independently, then heal
+        a.merge(b.snapshot())
+        b.merge(a.snapshot())
+        assert len(a.list("")) == len(b.list("")) == 20
+
+    def test_http_sync(self):
+        a, b = CLSetStore("a"), CLSetStore("b")
+        srv = CLSetHTTPServer(b).start()
+        try:
+            a.add_peer_url(srv.url)
+            a.put("k1", b"x")
+            b.put("k2", b"y")
+            a.sync_once()
+            assert a.get("k2") == b"y"
+            assert b.get("k1") == b"x"
+        finally:
+            srv.stop()
+
+
+class TestHTTPAllocator:
+    def test_allocate_lookup_release(self):
+        srv = NexusAllocatorServer().start()
+        try:
+            al = HTTPAllocator(srv.url)
+            assert al.health_check()
+            al.create_pool("p1", "10.2.0.0/24")
+            ip = al.allocate_ipv4("p1", "sub-1")
+            assert ip.startswith("10.2.0.")
+            got, pool = al.lookup_ipv4("sub-1")
+            assert got == ip and pool == "p1"
+            # idempotent allocation
+            assert al.allocate_ipv4("p1", "sub-1") == ip
+            al.release("p1", "sub-1")
+            with pytest.raises(NoAllocationError):
+                al.lookup_ipv4("sub-1")
+        finally:
+            srv.stop()
+
+    def test_no_allocation_is_walled_garden_signal(self):
+        srv = NexusAllocatorServer().start()
+        try:
+            al = HTTPAllocator(srv.url)
+            with pytest.raises(NoAllocationError):
+                al.lookup_ipv4("unknown-sub")
+        finally:
+            srv.stop()
