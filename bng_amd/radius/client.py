@@ -1,0 +1,210 @@
+"""RADIUS client: authentication + accounting with failover rotation,
+per-server rate limiting and mandatory Message-Authenticator
+(ref pkg/radius/client.go:157-427).
+"""
+from __future__ import annotations
+
+import socket
+import struct
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+from . import packet as rp
+
+
+@dataclass
+class AuthResult:
+    success: bool
+    framed_ip: str = ""
+    session_timeout: int = 0
+    idle_timeout: int = 0
+    policy_name: str = ""        # Filter-Id -> QoS policy (ref policy.go)
+    radius_class: bytes = b""
+    reply_message: str = ""
+    attributes: List[Tuple[int, bytes]] = field(default_factory=list)
+
+
+class RateLimiter:
+    """Token-bucket per server (ref client.go:378-389)."""
+
+    def __init__(self, rate_per_sec: float, burst: int):
+        self.rate = rate_per_sec
+        self.burst = burst
+        self.tokens = float(burst)
+        self.last = time.monotonic()
+        self._lock = threading.Lock()
+
+    def allow(self) -> bool:
+        with self._lock:
+            now = time.monotonic()
+            self.tokens = min(self.burst,
+                              self.tokens + (now - self.last) * self.rate)
+            self.last = now
+            if self.tokens >= 1:
+                self.tokens -= 1
+                return True
+            return False
+
+
+class RadiusTimeout(Exception):
+    pass
+
+
+class RadiusRateLimited(Exception):
+    pass
+
+
+class Client:
+    def __init__(self, servers: List[str], secret: bytes,
+                 nas_ip: str = "0.0.0.0", nas_identifier: str = "bng",
+                 timeout: float = 2.0, retries: int = 2,
+                 rate_limit: float = 0.0, rate_burst: int = 100,
+                 acct_port_offset: int = 1):
+        """servers: ['host:port', ...] — rotated on failure
+        (ref client.go:391-403)."""
+        self.servers = list(servers)
+        self.secret = secret if isinstance(secret, bytes) else secret.encode()
+        self.nas_ip = nas_ip
+        self.nas_identifier = nas_identifier
+        self.timeout = timeout
+        self.retries = retries
+        self.acct_port_offset = acct_port_offset
+        self._ident = 0
+        self._lock = threading.Lock()
+        self._limiters: Dict[str, RateLimiter] = {}
+        if rate_limit > 0:
+            for s in self.servers:
+                self._limiters[s] = RateLimiter(rate_limit, rate_burst)
+        self.stats = {"auth_ok": 0, "auth_reject": 0, "auth_timeout": 0,
+                      "acct_ok": 0, "acct_timeout": 0, "rate_limited": 0}
+
+    def _next_ident(self) -> int:
+        with self._lock:
+            self._ident = (self._ident + 1) & 0xFF
+            return self._ident
+
+    @staticmethod
+    def _addr(server: str, port_offset: int = 0) -> Tuple[str, int]:
+        host, _, port = server.rpartition(":")
+        return host or server, int(port or 1812) + port_offset
+
+    def _exchange(self, raw: bytes, server: str,
+                  port_offset: int = 0) -> Optional[bytes]:
+        lim = self._limiters.get(server)
+        if lim is not None and not lim.allow():
+            self.stats["rate_limited"] += 1
+            raise RadiusRateLimited(server)
+        addr = self._addr(server, port_offset)
+        with socket.socket(socket.AF_INET, socket.SOCK_DGRAM) as s:
+            s.settimeout(self.timeout)
+            s.sendto(raw, addr)
+            try:
+                data, _ = s.recvfrom(4096)
+                return data
+            except socket.timeout:
+                return None
+
+    def _send_with_failover(self, raw: bytes, req_auth: bytes,
+                            port_offset: int = 0) -> Optional[bytes]:
+        """Try each server in rotation, retries per server
+        (ref client.go:157-338)."""
+        for server in list(self.servers):
+            for _ in range(self.retries):
+                data = self._exchange(raw, server, port_offset)
+                if data is not None and rp.verify_response(
+                        data, req_auth, self.secret):
+                    return data
+            # rotate the failed server to the back (ref :391-403)
+            with self._lock:
+                if server in self.servers and len(self.servers) > 1:
+                    self.servers.remove(server)
+                    self.servers.append(server)
+        return None
+
+    # ----------------------------------------------------------- authn
+    def authenticate(self, username: str, password: str, mac: str = "",
+                     nas_port: int = 0,
+                     extra_attrs: Optional[List[Tuple[int, bytes]]] = None,
+                     chap: Optional[Tuple[bytes, bytes]] = None) -> AuthResult:
+        """PAP (or CHAP when chap=(challenge, ident+response)) auth
+        (ref client.go:157 Authenticate).  Raises RadiusTimeout when no
+        server answers."""
+        ident = self._next_ident()
+        req_auth = rp.random_authenticator()
+        pkt = rp.Packet(rp.ACCESS_REQUEST, ident, req_auth)
+        pkt.add(rp.USER_NAME, username)
+        if chap is not None:
+            challenge, chap_pw = chap
+            pkt.add(rp.CHAP_CHALLENGE, challenge)
+            pkt.add(rp.CHAP_PASSWORD, chap_pw)
+        else:
+            pkt.add(rp.USER_PASSWORD, rp.encrypt_user_password(
+                password.encode(), self.secret, req_auth))
+        pkt.add(rp.NAS_IP_ADDRESS,
+                struct.pack(">I", int.from_bytes(
+                    socket.inet_aton(self.nas_ip), "big")))
+        pkt.add(rp.NAS_IDENTIFIER, self.nas_identifier)
+        if nas_port:
+            pkt.add(rp.NAS_PORT, nas_port)
+        if mac:
+            pkt.add(rp.CALLING_STATION_ID, mac)
+        for t, v in (extra_attrs or []):
+            pkt.add(t, v)
+        rp.sign_message_authenticator(pkt, self.secret)
+        data = self._send_with_failover(pkt.encode(), req_auth)
+        if data is None:
+            self.stats["auth_timeout"] += 1
+            raise RadiusTimeout("no RADIUS server answered")
+        resp = rp.Packet.decode(data)
+        res = AuthResult(success=resp.code == rp.ACCESS_ACCEPT,
+                         attributes=list(resp.attributes))
+        if resp.get(rp.FRAMED_IP_ADDRESS):
+            res.framed_ip = socket.inet_ntoa(resp.get(rp.FRAMED_IP_ADDRESS))
+        res.session_timeout = resp.get_int(rp.SESSION_TIMEOUT) or 0
+        res.idle_timeout = resp.get_int(rp.IDLE_TIMEOUT) or 0
+        res.policy_name = resp.get_str(rp.FILTER_ID) or ""
+        res.radius_class = resp.get(rp.CLASS) or b""
+        res.reply_message = resp.get_str(rp.REPLY_MESSAGE) or ""
+        if res.success:
+            self.stats["auth_ok"] += 1
+        else:
+            self.stats["auth_reject"] += 1
+        return res
+
+    # ------------------------------------------------------- accounting
+    def send_accounting(self, status_type: int, session_id: str,
+                        username: str = "", framed_ip: str = "",
+                        input_octets: int = 0, output_octets: int = 0,
+                        session_time: int = 0, terminate_cause: int = 0,
+                        mac: str = "") -> bool:
+        """Acct Start/Interim/Stop (ref client.go:250 SendAccounting)."""
+        ident = self._next_ident()
+        pkt = rp.Packet(rp.ACCOUNTING_REQUEST, ident)
+        pkt.add(rp.ACCT_STATUS_TYPE, status_type)
+        pkt.add(rp.ACCT_SESSION_ID, session_id)
+        if username:
+            pkt.add(rp.USER_NAME, username)
+        if framed_ip:
+            pkt.add(rp.FRAMED_IP_ADDRESS,
+                    struct.pack(">I", int.from_bytes(
+                        socket.inet_aton(framed_ip), "big")))
+        if mac:
+            pkt.add(rp.CALLING_STATION_ID, mac)
+        pkt.add(rp.NAS_IDENTIFIER, self.nas_identifier)
+        if status_type in (rp.ACCT_INTERIM, rp.ACCT_STOP):
+            pkt.add(rp.ACCT_INPUT_OCTETS, input_octets & 0xFFFFFFFF)
+            pkt.add(rp.ACCT_OUTPUT_OCTETS, output_octets & 0xFFFFFFFF)
+            pkt.add(rp.ACCT_SESSION_TIME, session_time)
+        if status_type == rp.ACCT_STOP and terminate_cause:
+            pkt.add(rp.ACCT_TERMINATE_CAUSE, terminate_cause)
+        req_auth = rp.acct_request_authenticator(pkt, self.secret)
+        pkt.authenticator = req_auth
+        data = self._send_with_failover(pkt.encode(), req_auth,
+                                        self.acct_port_offset)
+        if data is None:
+            self.stats["acct_timeout"] += 1
+            return False
+        self.stats["acct_ok"] += 1
+        return rp.Packet.decode(data).code == rp.ACCOUNTING_RESPONSE
