@@ -1,0 +1,242 @@
+"""HA session sync — active serves HTTP + SSE, standby subscribes
+(ref pkg/ha/sync.go:25-815): sequence-numbered deltas over a
+Server-Sent-Events stream, periodic/gap-triggered full syncs, reconnect
+backoff."""
+from __future__ import annotations
+
+import json
+import queue
+import threading
+import time
+from typing import Callable, List, Optional
+
+from .protocol import (ROLE_ACTIVE, ROLE_STANDBY, SYNC_ADD, SYNC_DELETE,
+                       SYNC_FULL, SYNC_HEARTBEAT, SYNC_UPDATE,
+                       InMemorySessionStore, SessionState, SyncMessage)
+
+
+class HASyncer:
+    def __init__(self, node_id: str, role: str,
+                 store: Optional[InMemorySessionStore] = None,
+                 listen_port: int = 0, partner_url: str = "",
+                 full_sync_interval: float = 60.0,
+                 heartbeat_interval: float = 5.0,
+                 reconnect_backoff: float = 0.5,
+                 max_backoff: float = 10.0):
+        self.node_id = node_id
+        self.role = role
+        self.store = store or InMemorySessionStore()
+        self.partner_url = partner_url.rstrip("/")
+        self.full_sync_interval = full_sync_interval
+        self.heartbeat_interval = heartbeat_interval
+        self.reconnect_backoff = reconnect_backoff
+        self.max_backoff = max_backoff
+        self._seq = 0
+        self._seq_lock = threading.Lock()
+        self._subscribers: List[queue.Queue] = []
+        self._stop = threading.Event()
+        self._httpd = None
+        self._listen_port = listen_port
+        self._threads: List[threading.Thread] = []
+        self.last_partner_seq = -1
+        self.connected = False
+        self.stats = {"deltas_sent": 0, "deltas_received": 0,
+                      "full_syncs": 0, "reconnects": 0, "seq_gaps": 0}
+
+    # ------------------------------------------------------------ active
+    def _next_seq(self) -> int:
+        with self._seq_lock:
+            self._seq += 1
+            return self._seq
+
+    def _broadcast(self, msg: SyncMessage):
+        for q in list(self._subscribers):
+            try:
+                q.put_nowait(msg)
+            except queue.Full:
+                pass
+
+    def publish_add(self, s: SessionState):
+        """Call on session create (active side)."""
+        self.store.put(s)
+        self._publish(SYNC_ADD, [s.to_dict()])
+
+    def publish_update(self, s: SessionState):
+        self.store.put(s)
+        self._publish(SYNC_UPDATE, [s.to_dict()])
+
+    def publish_delete(self, session_id: str):
+        self.store.delete(session_id)
+        self._publish(SYNC_DELETE, [{"session_id": session_id}])
+
+    def _publish(self, typ: str, sessions: List[dict]):
+        if self.role != ROLE_ACTIVE:
+            return
+        msg = SyncMessage(typ, sessions, seq=self._next_seq(),
+                          node_id=self.node_id)
+        self.stats["deltas_sent"] += 1
+        self._broadcast(msg)
+
+    def _serve(self):
+        from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+        syncer = self
+
+        class Handler(BaseHTTPRequestHandler):
+            protocol_version = "HTTP/1.1"
+
+            def log_message(self, *a):
+                pass
+
+            def do_GET(self):
+                if self.path == "/health":
+                    body = json.dumps({
+                        "node_id": syncer.node_id, "role": syncer.role,
+                        "sessions": syncer.store.count()}).encode()
+                    self.send_response(200)
+                    self.send_header("Content-Length", str(len(body)))
+                    self.end_headers()
+                    self.wfile.write(body)
+                    return
+                if self.path == "/sync/full":
+                    msg = SyncMessage(
+                        SYNC_FULL,
+                        [s.to_dict() for s in syncer.store.all()],
+                        seq=syncer._seq, node_id=syncer.node_id)
+                    body = json.dumps(msg.to_dict()).encode()
+                    self.send_response(200)
+                    self.send_header("Content-Length", str(len(body)))
+                    self.end_headers()
+                    self.wfile.write(body)
+                    return
+                if self.path == "/sync/stream":
+                    self.send_response(200)
+                    self.send_header("Content-Type", "text/event-stream")
+                    self.send_header("Cache-Control", "no-cache")
+                    self.end_headers()
+                    q: queue.Queue = queue.Queue(maxsize=10000)
+                    syncer._subscribers.append(q)
+                    try:
+                        while not syncer._stop.is_set():
+                            try:
+                                msg = q.get(timeout=syncer.heartbeat_interval)
+                            except queue.Empty:
+                                msg = SyncMessage(SYNC_HEARTBEAT,
+                                                  seq=syncer._seq,
+                                                  node_id=syncer.node_id)
+                            data = json.dumps(msg.to_dict())
+                            self.wfile.write(
+                                f"data: {data}\n\n".encode())
+                            self.wfile.flush()
+                    except (BrokenPipeError, ConnectionError, OSError):
+                        pass
+                    finally:
+                        if q in syncer._subscribers:
+                            syncer._subscribers.remove(q)
+                    return
+                self.send_response(404)
+                self.send_header("Content-Length", "0")
+                self.end_headers()
+
+        self._httpd = ThreadingHTTPServer(("127.0.0.1", self._listen_port),
+                                          Handler)
+        self._listen_port = self._httpd.server_address[1]
+        t = threading.Thread(target=self._httpd.serve_forever, daemon=True)
+        t.start()
+        self._threads.append(t)
+
+    @property
+    def url(self) -> str:
+        return f"http://127.0.0.1:{self._listen_port}"
+
+    # ----------------------------------------------------------- standby
+    def _connect_loop(self):
+        """Standby: subscribe to the active's SSE stream with reconnect
+        backoff; full-sync on connect and on sequence gaps
+        (ref sync.go:77-110 connectLoop / receivedSessions :94)."""
+        import requests
+        backoff = self.reconnect_backoff
+        while not self._stop.is_set():
+            try:
+                self._full_sync()
+                self.connected = True
+                backoff = self.reconnect_backoff
+                with requests.get(f"{self.partner_url}/sync/stream",
+                                  stream=True, timeout=(3, 30)) as r:
+                    for line in r.iter_lines():
+                        if self._stop.is_set():
+                            return
+                        if not line or not line.startswith(b"data: "):
+                            continue
+                        msg = SyncMessage.from_dict(
+                            json.loads(line[6:].decode()))
+                        self._apply(msg)
+            except Exception:
+                pass
+            self.connected = False
+            self.stats["reconnects"] += 1
+            if self._stop.wait(backoff):
+                return
+            backoff = min(backoff * 2, self.max_backoff)
+
+    def _full_sync(self):
+        import requests
+        r = requests.get(f"{self.partner_url}/sync/full", timeout=5)
+        r.raise_for_status()
+        msg = SyncMessage.from_dict(r.json())
+        self.store.replace_all([SessionState.from_dict(d)
+                                for d in msg.sessions])
+        self.last_partner_seq = msg.seq
+        self.stats["full_syncs"] += 1
+
+    def _apply(self, msg: SyncMessage):
+        if msg.type == SYNC_HEARTBEAT:
+            return
+        if self.last_partner_seq >= 0 and msg.seq != self.last_partner_seq + 1:
+            self.stats["seq_gaps"] += 1
+            try:
+                self._full_sync()
+            except Exception:
+                pass
+            return
+        self.last_partner_seq = msg.seq
+        self.stats["deltas_received"] += 1
+        if msg.type in (SYNC_ADD, SYNC_UPDATE):
+            for d in msg.sessions:
+                self.store.put(SessionState.from_dict(d))
+        elif msg.type == SYNC_DELETE:
+            for d in msg.sessions:
+                self.store.delete(d["session_id"])
+
+    def _full_sync_loop(self):
+        while not self._stop.wait(self.full_sync_interval):
+            if self.role == ROLE_STANDBY and self.partner_url:
+                try:
+                    self._full_sync()
+                except Exception:
+                    pass
+
+    # --------------------------------------------------------- lifecycle
+    def start(self):
+        self._serve()
+        if self.role == ROLE_STANDBY and self.partner_url:
+            t = threading.Thread(target=self._connect_loop, daemon=True)
+            t.start()
+            self._threads.append(t)
+            t2 = threading.Thread(target=self._full_sync_loop, daemon=True)
+            t2.start()
+            self._threads.append(t2)
+        return self
+
+    def stop(self):
+        self._stop.set()
+        if self._httpd:
+            self._httpd.shutdown()
+            self._httpd.server_close()
+
+    def promote(self):
+        """Standby -> active at failover: shadow store becomes
+        authoritative and we start publishing."""
+        self.role = ROLE_ACTIVE
+
+    def demote(self):
+        self.role = ROLE_STANDBY

@@ -1,0 +1,163 @@
+"""HA tests — real active+standby HASyncer pairs on localhost HTTP
+(ref pkg/ha/sync_test.go, failover_test.go:94-193)."""
+import time
+
+import pytest
+
+from bng_amd.ha.failover import (STATE_FAILED_OVER, STATE_NORMAL,
+                                 FailoverController)
+from bng_amd.ha.health_monitor import HealthMonitor
+from bng_amd.ha.protocol import (ROLE_ACTIVE, ROLE_STANDBY, SessionState)
+from bng_amd.ha.sync import HASyncer
+
+
+def wait_for(cond, timeout=5.0, interval=0.02):
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        if cond():
+            return True
+        time.sleep(interval)
+    return False
+
+
+@pytest.fixture
+def pair():
+    active = HASyncer("node-a", ROLE_ACTIVE, heartbeat_interval=0.2).start()
+    standby = HASyncer("node-b", ROLE_STANDBY, partner_url=active.url,
+                       full_sync_interval=3600,
+                       reconnect_backoff=0.1).start()
+    yield active, standby
+    standby.stop()
+    active.stop()
+
+
+class TestSync:
+    def test_delta_stream_replication(self, pair):
+        active, standby = pair
+        assert wait_for(lambda: standby.connected)
+        active.publish_add(SessionState("s1", "sub-1", ip="10.0.1.50"))
+        active.publish_add(SessionState("s2", "sub-2", ip="10.0.1.51"))
+        assert wait_for(lambda: standby.store.count() == 2)
+        assert standby.store.get("s1").ip == "10.0.1.50"
+        active.publish_update(SessionState("s1", "sub-1", ip="10.0.1.99"))
+        assert wait_for(
+            lambda: standby.store.get("s1").ip == "10.0.1.99")
+        active.publish_delete("s2")
+        assert wait_for(lambda: standby.store.count() == 1)
+
+    def test_full_sync_on_connect(self):
+        active = HASyncer("a", ROLE_ACTIVE).start()
+        try:
+            for i in range(50):
+                active.store.put(SessionState(f"s{i}", f"sub-{i}"))
+            standby = HASyncer("b", ROLE_STANDBY, partner_url=active.url,
+                               reconnect_backoff=0.1).start()
+            try:
+                assert wait_for(lambda: standby.store.count() == 50)
+                assert standby.stats["full_syncs"] >= 1
+            finally:
+                standby.stop()
+        finally:
+            active.stop()
+
+    def test_standby_promotion_keeps_state(self, pair):
+        """At failover the shadow store becomes authoritative
+        (ref SURVEY §3.5)."""
+        active, standby = pair
+        assert wait_for(lambda: standby.connected)
+        active.publish_add(SessionState("s1", "sub-1", ip="10.0.1.50"))
+        assert wait_for(lambda: standby.store.count() == 1)
+        active.stop()
+        standby.promote()
+        assert standby.role == ROLE_ACTIVE
+        assert standby.store.get("s1").ip == "10.0.1.50"
+        # promoted node can now publish
+        standby.publish_add(SessionState("s3", "sub-3"))
+        assert standby.store.count() == 2
+
+
+class TestHealthMonitor:
+    def test_thresholds(self):
+        active = HASyncer("a", ROLE_ACTIVE).start()
+        try:
+            m = HealthMonitor(active.url, failure_threshold=2,
+                              recovery_threshold=2)
+            events = []
+            m.on_event(lambda e: events.append(e.type))
+            assert m.check_once()
+            assert m.partner_healthy
+            active.stop()
+            m.check_once()
+            assert m.partner_healthy          # 1 < threshold
+            m.check_once()
+            assert not m.partner_healthy      # threshold reached
+            assert events == ["partner_down"]
+        finally:
+            active.stop()
+
+    def test_recovery_event(self):
+        m = HealthMonitor("http://127.0.0.1:1", failure_threshold=1,
+                          recovery_threshold=1, timeout=0.2)
+        events = []
+        m.on_event(lambda e: events.append(e.type))
+        m.check_once()
+        assert not m.partner_healthy
+        active = HASyncer("a", ROLE_ACTIVE).start()
+        try:
+            m.partner_url = active.url
+            m.check_once()
+            assert m.partner_healthy
+            assert events == ["partner_down", "partner_up"]
+        finally:
+            active.stop()
+
+
+class TestFailover:
+    def test_standby_takes_over_on_partner_death(self):
+        """End-to-end: active dies -> monitor detects -> controller
+        promotes -> role callback fires (ref SURVEY §3.5 call stack)."""
+        active = HASyncer("a", ROLE_ACTIVE, heartbeat_interval=0.2).start()
+        standby = HASyncer("b", ROLE_STANDBY, partner_url=active.url,
+                           reconnect_backoff=0.1).start()
+        roles = []
+        mon = HealthMonitor(active.url, interval=0.1, timeout=0.3,
+                            failure_threshold=2)
+        ctl = FailoverController(
+            "b", ROLE_STANDBY, monitor=mon,
+            role_change_callback=lambda r: (roles.append(r),
+                                            standby.promote()))
+        try:
+            active.publish_add(SessionState("s1", "sub-1"))
+            assert wait_for(lambda: standby.store.count() == 1)
+            active.stop()
+            mon.start()
+            assert wait_for(lambda: ctl.role == ROLE_ACTIVE, timeout=10)
+            assert ctl.state == STATE_FAILED_OVER
+            assert roles == [ROLE_ACTIVE]
+            assert standby.role == ROLE_ACTIVE
+            assert standby.store.count() == 1
+        finally:
+            mon.stop()
+            standby.stop()
+
+    def test_failback_when_partner_recovers(self):
+        ctl = FailoverController("b", ROLE_STANDBY, auto_failback=True)
+        assert ctl.initiate_failover(reason="test")
+        assert ctl.role == ROLE_ACTIVE and ctl.state == STATE_FAILED_OVER
+        from bng_amd.ha.health_monitor import HealthEvent
+        ctl.handle_health_event(HealthEvent("partner_up", "x", 0, 1))
+        assert ctl.role == ROLE_STANDBY and ctl.state == STATE_NORMAL
+        assert [h["event"] for h in ctl.history] == ["failover", "failback"]
+
+    def test_forced_operations(self):
+        ctl = FailoverController("b", ROLE_STANDBY, failover_delay=30)
+        assert ctl.force_failover()       # skips the delay
+        assert ctl.role == ROLE_ACTIVE
+        assert ctl.force_failback()
+        assert ctl.role == ROLE_STANDBY
+
+    def test_active_ignores_partner_down(self):
+        ctl = FailoverController("a", ROLE_ACTIVE)
+        from bng_amd.ha.health_monitor import HealthEvent
+        ctl.handle_health_event(HealthEvent("partner_down", "x", 0, 3))
+        assert ctl.role == ROLE_ACTIVE and ctl.state == STATE_NORMAL
