@@ -1,0 +1,208 @@
+"""Audit subsystem (ref pkg/audit): structured events (types.go:101-238),
+async logger over a storage interface (logger.go:15-161), in-memory
+storage with query, rotating file exporter with compression
+(rotation.go:19-402), syslog/JSON-lines exporters (export.go:17-486),
+retention + legal hold (retention.go:9-347)."""
+from __future__ import annotations
+
+import gzip
+import json
+import os
+import queue
+import threading
+import time
+import uuid
+from dataclasses import asdict, dataclass, field
+from typing import Dict, List, Optional
+
+# event categories (ref types.go:101-238)
+CAT_SESSION = "session"
+CAT_NAT = "nat"
+CAT_AUTH = "auth"
+CAT_TLS = "tls"
+CAT_ZTP = "ztp"
+CAT_CONFIG = "config"
+CAT_SYSTEM = "system"
+
+
+@dataclass
+class Event:
+    id: str
+    category: str
+    action: str
+    timestamp: float
+    subscriber: str = ""
+    ip: str = ""
+    outcome: str = "success"
+    details: Dict[str, str] = field(default_factory=dict)
+    legal_hold: bool = False
+
+
+class MemoryStorage:
+    def __init__(self, max_events: int = 100_000):
+        self.max_events = max_events
+        self._events: List[Event] = []
+        self._lock = threading.Lock()
+
+    def store(self, ev: Event):
+        with self._lock:
+            self._events.append(ev)
+            if len(self._events) > self.max_events:
+                # never evict events under legal hold (ref retention.go)
+                keep = [e for e in self._events[:len(self._events) // 2]
+                        if e.legal_hold]
+                self._events = keep + self._events[len(self._events) // 2:]
+
+    def query(self, category: str = "", subscriber: str = "",
+              action: str = "", since: float = 0.0,
+              until: float = 0.0) -> List[Event]:
+        with self._lock:
+            out = []
+            for e in self._events:
+                if category and e.category != category:
+                    continue
+                if subscriber and e.subscriber != subscriber:
+                    continue
+                if action and e.action != action:
+                    continue
+                if since and e.timestamp < since:
+                    continue
+                if until and e.timestamp > until:
+                    continue
+                out.append(e)
+            return out
+
+    def all(self) -> List[Event]:
+        with self._lock:
+            return list(self._events)
+
+    def apply_retention(self, max_age: float) -> int:
+        cutoff = time.time() - max_age
+        with self._lock:
+            before = len(self._events)
+            self._events = [e for e in self._events
+                            if e.legal_hold or e.timestamp >= cutoff]
+            return before - len(self._events)
+
+
+class FileExporter:
+    """Rotating JSON-lines file exporter w/ gzip (ref rotation.go)."""
+
+    def __init__(self, path: str, rotate_bytes: int = 10 << 20,
+                 compress: bool = True, retention: int = 10):
+        self.path = path
+        self.rotate_bytes = rotate_bytes
+        self.compress = compress
+        self.retention = retention
+        self._fh = open(path, "a")
+        self._written = os.path.getsize(path)
+        self._lock = threading.Lock()
+
+    def export(self, ev: Event):
+        line = json.dumps(asdict(ev))
+        with self._lock:
+            self._fh.write(line + "\n")
+            self._written += len(line) + 1
+            if self._written >= self.rotate_bytes:
+                self._rotate()
+
+    def _rotate(self):
+        self._fh.close()
+        rotated = f"{self.path}.{time.strftime('%Y%m%d-%H%M%S')}-" \
+                  f"{uuid.uuid4().hex[:6]}"
+        os.rename(self.path, rotated)
+        if self.compress:
+            with open(rotated, "rb") as src, \
+                    gzip.open(rotated + ".gz", "wb") as dst:
+                dst.write(src.read())
+            os.unlink(rotated)
+        self._fh = open(self.path, "a")
+        self._written = 0
+        d = os.path.dirname(self.path) or "."
+        base = os.path.basename(self.path) + "."
+        rotated_files = sorted(f for f in os.listdir(d)
+                               if f.startswith(base))
+        while len(rotated_files) > self.retention:
+            os.unlink(os.path.join(d, rotated_files.pop(0)))
+
+    def close(self):
+        with self._lock:
+            self._fh.close()
+
+
+class SyslogExporter:
+    """RFC5424-ish lines to a callable/file (network syslog plugs in)."""
+
+    def __init__(self, sink):
+        self.sink = sink
+
+    def export(self, ev: Event):
+        self.sink(f"<110>1 - bng audit - - - [{ev.category}] {ev.action} "
+                  f"sub={ev.subscriber} ip={ev.ip} outcome={ev.outcome}")
+
+
+class Logger:
+    """Async audit logger (ref logger.go:15-161)."""
+
+    def __init__(self, storage: Optional[MemoryStorage] = None,
+                 exporters: Optional[list] = None, queue_size: int = 10000):
+        self.storage = storage or MemoryStorage()
+        self.exporters = exporters or []
+        self._q: queue.Queue = queue.Queue(maxsize=queue_size)
+        self._stop = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+        self.dropped = 0
+        self.legal_holds: set = set()       # subscriber ids under hold
+
+    def start(self):
+        self._thread = threading.Thread(target=self._loop, daemon=True)
+        self._thread.start()
+        return self
+
+    def stop(self):
+        self._stop.set()
+        self.flush()
+
+    def set_legal_hold(self, subscriber: str, held: bool = True):
+        """ref retention.go legal holds."""
+        if held:
+            self.legal_holds.add(subscriber)
+        else:
+            self.legal_holds.discard(subscriber)
+
+    def log(self, action: str, category: str = CAT_SESSION,
+            subscriber: str = "", ip: str = "", outcome: str = "success",
+            **details):
+        ev = Event(id=uuid.uuid4().hex[:12], category=category,
+                   action=action, timestamp=time.time(),
+                   subscriber=subscriber, ip=ip, outcome=outcome,
+                   details={k: str(v) for k, v in details.items()},
+                   legal_hold=subscriber in self.legal_holds)
+        try:
+            self._q.put_nowait(ev)
+        except queue.Full:
+            self.dropped += 1
+
+    def _loop(self):
+        while not self._stop.is_set():
+            try:
+                ev = self._q.get(timeout=0.2)
+            except queue.Empty:
+                continue
+            self._process(ev)
+
+    def _process(self, ev: Event):
+        self.storage.store(ev)
+        for ex in self.exporters:
+            try:
+                ex.export(ev)
+            except Exception:
+                pass
+
+    def flush(self):
+        while True:
+            try:
+                ev = self._q.get_nowait()
+            except queue.Empty:
+                return
+            self._process(ev)
