@@ -209,6 +209,8 @@ def main():
     ap.add_argument("--lat-batch", type=int, default=2048)
     ap.add_argument("--lat-reps", type=int, default=64)
     ap.add_argument("--no-latency", action="store_true")
+    ap.add_argument("--no-sort", action="store_true",
+                    help="disable on-device type-sort (wave-divergence fix)")
     args = ap.parse_args()
 
     import torch
@@ -252,7 +254,8 @@ def main():
         if distributed:
             owner = launcher.shard_owner(d, l, world)
             d, l = exchange(d, l, owner)
-        launcher.uplink(d, l, now_ns=now_ns, now_sec=now_sec)
+        launcher.uplink(d, l, now_ns=now_ns, now_sec=now_sec,
+                        sort_by_type=not args.no_sort)
 
     base_ns = now_sec * 10**9
     for w in range(args.warmup):

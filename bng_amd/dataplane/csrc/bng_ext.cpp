@@ -14,6 +14,7 @@
 #include <c10/hip/HIPStream.h>
 
 #include "bng_abi.h"
+#include "bng_params.h"
 
 extern "C" {
 void bng_launch_dhcp(void*, const void*, void*, void*, int, int, const void*,
@@ -28,14 +29,9 @@ void bng_launch_qos(void*, const void*, void*, int, int, int, void*,
 void bng_launch_antispoof(void*, const void*, void*, int, int, const void*,
                           uint32_t, const void*, void*, void*, void*,
                           uint64_t, hipStream_t);
-void bng_launch_uplink(void*, const void*, void*, void*, int, int,
-                       const void*, uint32_t, const void*, uint32_t,
-                       const void*, void*, const void*, uint32_t,
-                       const void*, void*, void*, void*, void*, uint32_t,
-                       void*, uint32_t, void*, uint32_t, void*, uint32_t,
-                       const void*, const void*, uint32_t, void*, void*,
-                       void*, void*, uint32_t, void*, uint64_t, uint64_t,
-                       hipStream_t);
+void bng_launch_uplink(bng_uplink_params*, hipStream_t);
+void bng_launch_pkt_class(const void*, const void*, void*, int, int,
+                          hipStream_t);
 void bng_launch_sub_upsert(void*, uint32_t, const void*, int, void*,
                            hipStream_t);
 void bng_launch_sub_delete(void*, uint32_t, const void*, int, hipStream_t);
@@ -146,29 +142,57 @@ void uplink_pipeline(torch::Tensor data, torch::Tensor in_len,
                      torch::Tensor nat_stats, torch::Tensor log_ring,
                      torch::Tensor log_hdr, torch::Tensor qos_in,
                      torch::Tensor qos_stats, int64_t now_ns,
-                     int64_t now_sec) {
+                     int64_t now_sec,
+                     c10::optional<torch::Tensor> order) {
   check_dev(data, "data");
-  bng_launch_uplink(
-      data.data_ptr(), in_len.data_ptr(), out_len.data_ptr(),
-      verdict.data_ptr(), (int)in_len.numel(), (int)data.size(1),
-      subs.data_ptr(), table_mask(subs, sizeof(bng_sub_entry), "subs"),
-      pools.data_ptr(),
-      (uint32_t)(pools.numel() * pools.element_size() / sizeof(bng_ip_pool)),
-      scfg.data_ptr(), dhcp_stats.data_ptr(), bindings.data_ptr(),
-      table_mask(bindings, sizeof(bng_binding_entry), "bindings"),
-      acfg.data_ptr(), as_stats.data_ptr(), spoof_ring.data_ptr(),
-      spoof_hdr.data_ptr(), sessions.data_ptr(),
-      table_mask(sessions, sizeof(bng_nat_session), "sessions"),
-      reverse.data_ptr(),
-      table_mask(reverse, sizeof(bng_nat_reverse), "reverse"),
-      eim.data_ptr(), table_mask(eim, sizeof(bng_eim_entry), "eim"),
-      subnat.data_ptr(),
-      table_mask(subnat, sizeof(bng_subnat_entry), "subnat"),
-      ncfg.data_ptr(), hairpin.data_ptr(), (uint32_t)n_hairpin,
-      nat_stats.data_ptr(), log_ring.data_ptr(), log_hdr.data_ptr(),
-      qos_in.data_ptr(), table_mask(qos_in, sizeof(bng_qos_bucket), "qos"),
-      qos_stats.data_ptr(), (uint64_t)now_ns, (uint64_t)now_sec,
-      cur_stream());
+  bng_uplink_params P{};
+  P.data = (uint8_t*)data.data_ptr();
+  P.in_len = (const uint16_t*)in_len.data_ptr();
+  P.out_len = (uint16_t*)out_len.data_ptr();
+  P.verdict = (uint8_t*)verdict.data_ptr();
+  P.order = order.has_value() ? (const int32_t*)order->data_ptr() : nullptr;
+  P.n = (int)in_len.numel();
+  P.stride = (int)data.size(1);
+  P.subs = (const bng_sub_entry*)subs.data_ptr();
+  P.sub_mask = table_mask(subs, sizeof(bng_sub_entry), "subs");
+  P.pools = (const bng_ip_pool*)pools.data_ptr();
+  P.n_pools = (uint32_t)(pools.numel() * pools.element_size() /
+                         sizeof(bng_ip_pool));
+  P.scfg = (const bng_server_config*)scfg.data_ptr();
+  P.dhcp_stats = (unsigned long long*)dhcp_stats.data_ptr();
+  P.bindings = (const bng_binding_entry*)bindings.data_ptr();
+  P.bmask = table_mask(bindings, sizeof(bng_binding_entry), "bindings");
+  P.acfg = (const bng_antispoof_config*)acfg.data_ptr();
+  P.as_stats = (unsigned long long*)as_stats.data_ptr();
+  P.spoof_ring = (bng_spoof_event*)spoof_ring.data_ptr();
+  P.spoof_hdr = (bng_ring_header*)spoof_hdr.data_ptr();
+  P.sessions = (bng_nat_session*)sessions.data_ptr();
+  P.sess_mask = table_mask(sessions, sizeof(bng_nat_session), "sessions");
+  P.reverse = (bng_nat_reverse*)reverse.data_ptr();
+  P.rev_mask = table_mask(reverse, sizeof(bng_nat_reverse), "reverse");
+  P.eim = (bng_eim_entry*)eim.data_ptr();
+  P.eim_mask = table_mask(eim, sizeof(bng_eim_entry), "eim");
+  P.subnat = (bng_subnat_entry*)subnat.data_ptr();
+  P.subnat_mask = table_mask(subnat, sizeof(bng_subnat_entry), "subnat");
+  P.ncfg = (const bng_nat_config*)ncfg.data_ptr();
+  P.hairpin_ips = (const uint32_t*)hairpin.data_ptr();
+  P.n_hairpin = (uint32_t)n_hairpin;
+  P.nat_stats = (unsigned long long*)nat_stats.data_ptr();
+  P.log_ring = (bng_nat_log_entry*)log_ring.data_ptr();
+  P.log_hdr = (bng_ring_header*)log_hdr.data_ptr();
+  P.qos_in = (bng_qos_bucket*)qos_in.data_ptr();
+  P.qos_mask = table_mask(qos_in, sizeof(bng_qos_bucket), "qos");
+  P.qos_stats = (unsigned long long*)qos_stats.data_ptr();
+  P.now_ns = (uint64_t)now_ns;
+  P.now_sec = (uint64_t)now_sec;
+  bng_launch_uplink(&P, cur_stream());
+}
+
+void pkt_class(torch::Tensor data, torch::Tensor in_len,
+               torch::Tensor cls) {
+  bng_launch_pkt_class(data.data_ptr(), in_len.data_ptr(), cls.data_ptr(),
+                       (int)in_len.numel(), (int)data.size(1),
+                       cur_stream());
 }
 
 void sub_upsert(torch::Tensor table, torch::Tensor batch, torch::Tensor rc) {
@@ -274,7 +298,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("nat44", &nat44);
   m.def("qos", &qos);
   m.def("antispoof", &antispoof);
-  m.def("uplink_pipeline", &uplink_pipeline);
+  m.def("uplink_pipeline", &uplink_pipeline,
+        py::arg("data"), py::arg("in_len"), py::arg("out_len"),
+        py::arg("verdict"), py::arg("subs"), py::arg("pools"),
+        py::arg("scfg"), py::arg("dhcp_stats"), py::arg("bindings"),
+        py::arg("acfg"), py::arg("as_stats"), py::arg("spoof_ring"),
+        py::arg("spoof_hdr"), py::arg("sessions"), py::arg("reverse"),
+        py::arg("eim"), py::arg("subnat"), py::arg("ncfg"),
+        py::arg("hairpin"), py::arg("n_hairpin"), py::arg("nat_stats"),
+        py::arg("log_ring"), py::arg("log_hdr"), py::arg("qos_in"),
+        py::arg("qos_stats"), py::arg("now_ns"), py::arg("now_sec"),
+        py::arg("order") = py::none());
+  m.def("pkt_class", &pkt_class);
   m.def("sub_upsert", &sub_upsert);
   m.def("sub_delete", &sub_delete);
   m.def("subnat_upsert", &subnat_upsert);

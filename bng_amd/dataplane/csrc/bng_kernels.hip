@@ -21,6 +21,7 @@
  */
 #include <hip/hip_runtime.h>
 #include "bng_abi.h"
+#include "bng_params.h"
 #include "bng_device.h"
 
 /* ==================================================== table primitives */
@@ -30,7 +31,10 @@ BNG_DEV const bng_sub_entry* sub_lookup(const bng_sub_entry* t, uint32_t mask,
   uint32_t slot = (uint32_t)bng_mix64(key) & mask;
   for (int i = 0; i < BNG_MAX_PROBE; ++i) {
     const bng_sub_entry* e = &t[(slot + i) & mask];
-    uint64_t k = e->key;
+    /* one 16-B vector load covers key+pool_id+allocated_ip (entry is
+     * 32-B aligned); most probes end after this single transaction */
+    uint4 v = *(const uint4*)e;
+    uint64_t k = ((uint64_t)v.y << 32) | v.x;
     if (k == key) return e;
     if (k == BNG_KEY_EMPTY) return nullptr;
   }
@@ -1008,45 +1012,29 @@ __global__ void antispoof_kernel(
  * kernel hooks: DHCP fast path for UDP:67, else antispoof -> NAT44 SNAT
  * -> QoS ingress.  One parse, one packet-data round trip — the fusion the
  * CDNA4 guide prescribes for HBM-bound pipelines. */
-__global__ void uplink_pipeline_kernel(
-    uint8_t* __restrict__ data, const uint16_t* __restrict__ in_len,
-    uint16_t* __restrict__ out_len, uint8_t* __restrict__ verdict,
-    int n, int stride,
-    /* dhcp */
-    const bng_sub_entry* subs, uint32_t sub_mask,
-    const bng_ip_pool* pools, uint32_t n_pools,
-    const bng_server_config* scfg, unsigned long long* dhcp_stats,
-    /* antispoof */
-    const bng_binding_entry* bindings, uint32_t bmask,
-    const bng_antispoof_config* acfg, unsigned long long* as_stats,
-    bng_spoof_event* spoof_ring, bng_ring_header* spoof_hdr,
-    /* nat */
-    bng_nat_session* sessions, uint32_t sess_mask,
-    bng_nat_reverse* reverse, uint32_t rev_mask,
-    bng_eim_entry* eim, uint32_t eim_mask,
-    bng_subnat_entry* subnat, uint32_t subnat_mask,
-    const bng_nat_config* ncfg, const uint32_t* hairpin_ips,
-    uint32_t n_hairpin, unsigned long long* nat_stats,
-    bng_nat_log_entry* log_ring, bng_ring_header* log_hdr,
-    /* qos (ingress: subscriber upload) */
-    bng_qos_bucket* qos_in, uint32_t qos_mask,
-    unsigned long long* qos_stats,
-    uint64_t now_ns, uint64_t now_sec) {
-  dhcp_tables DT{subs, sub_mask, pools, n_pools, scfg, dhcp_stats, now_sec};
-  nat_tables NT{sessions, sess_mask, reverse, rev_mask, eim, eim_mask,
-                subnat, subnat_mask, ncfg, hairpin_ips, n_hairpin, nat_stats,
-                log_ring, log_hdr, now_ns};
+__global__ __launch_bounds__(256, 5)  /* cap VGPRs at 96 -> 5 waves/SIMD:
+    the pipeline is latency-bound (SQ_WAIT ~95%), occupancy is the lever */
+void uplink_pipeline_kernel(bng_uplink_params P) {
+  dhcp_tables DT{P.subs, P.sub_mask, P.pools, P.n_pools, P.scfg,
+                 P.dhcp_stats, P.now_sec};
+  nat_tables NT{P.sessions, P.sess_mask, P.reverse, P.rev_mask, P.eim,
+                P.eim_mask, P.subnat, P.subnat_mask, P.ncfg, P.hairpin_ips,
+                P.n_hairpin, P.nat_stats, P.log_ring, P.log_hdr, P.now_ns};
   int tid = blockIdx.x * blockDim.x + threadIdx.x;
   int nthreads = gridDim.x * blockDim.x;
-  for (int base = 0; base < n; base += nthreads) {
-    int pid = base + tid;
+  for (int base = 0; base < P.n; base += nthreads) {
+    int i = base + tid;
     dhcp_flags DF; DF.clear();
     nat_flags NF; NF.clear();
     as_flags AF{false, false, false, false, false};
     qos_flags QF{false, false, 0};
-    if (pid < n) {
-      uint8_t* p = data + (size_t)pid * stride;
-      int len = in_len[pid];
+    if (i < P.n) {
+      /* with a type-sorted order array, a wave's 64 packets share one
+       * code path — divergence between the DHCP and data pipelines no
+       * longer serializes both per wave */
+      int pid = P.order ? P.order[i] : i;
+      uint8_t* p = P.data + (size_t)pid * P.stride;
+      int len = P.in_len[pid];
       uint16_t ol = (uint16_t)len;
       int v;
       pktctx c;
@@ -1054,33 +1042,50 @@ __global__ void uplink_pipeline_kernel(
       bool is_dhcp = ip_ok && c.ip_off >= 0 && c.proto == 17 && c.l4_ok &&
                      c.dport == 67;
       if (is_dhcp) {
-        v = dhcp_process(p, len, stride, DT, DF, &ol);
+        v = dhcp_process(p, len, P.stride, DT, DF, &ol);
       } else {
         /* the reference's TC programs parse untagged frames only
          * (nat44.c:573-581, antispoof.c:194-219): a tagged non-DHCP frame
          * reads as non-IP there and is allowed through */
-        v = antispoof_process(p, len, bindings, bmask, acfg, spoof_ring,
-                              spoof_hdr, now_ns, AF);
+        v = antispoof_process(p, len, P.bindings, P.bmask, P.acfg,
+                              P.spoof_ring, P.spoof_hdr, P.now_ns, AF);
         if (v == BNG_FWD && ip_ok && !c.tagged) {
           v = nat_egress_process(c, NT, NF);
           if (v == BNG_FWD)
-            v = qos_process(c, qos_in, qos_mask, /*egress=*/false, now_ns, QF);
+            v = qos_process(c, P.qos_in, P.qos_mask, /*egress=*/false,
+                            P.now_ns, QF);
         }
       }
-      verdict[pid] = (uint8_t)v;
-      out_len[pid] = ol;
+      P.verdict[pid] = (uint8_t)v;
+      P.out_len[pid] = ol;
     }
-    dhcp_commit_stats(DF, dhcp_stats);
-    nat_commit_stats(NF, nat_stats);
-    stat_inc(&as_stats[BNG_AS_ALLOWED], AF.allowed);
-    stat_inc(&as_stats[BNG_AS_DROPPED], AF.dropped);
-    stat_inc(&as_stats[BNG_AS_LOGGED], AF.logged);
-    stat_inc(&as_stats[BNG_AS_V4_VIOLATIONS], AF.v4viol);
-    stat_inc(&as_stats[BNG_AS_V6_VIOLATIONS], AF.v6viol);
-    stat_inc(&qos_stats[BNG_QS_PKT_PASSED], QF.passed);
-    stat_inc(&qos_stats[BNG_QS_PKT_DROPPED], QF.dropped);
-    stat_add(&qos_stats[BNG_QS_BYTES_PASSED], QF.bytes, QF.passed);
-    stat_add(&qos_stats[BNG_QS_BYTES_DROPPED], QF.bytes, QF.dropped);
+    dhcp_commit_stats(DF, P.dhcp_stats);
+    nat_commit_stats(NF, P.nat_stats);
+    stat_inc(&P.as_stats[BNG_AS_ALLOWED], AF.allowed);
+    stat_inc(&P.as_stats[BNG_AS_DROPPED], AF.dropped);
+    stat_inc(&P.as_stats[BNG_AS_LOGGED], AF.logged);
+    stat_inc(&P.as_stats[BNG_AS_V4_VIOLATIONS], AF.v4viol);
+    stat_inc(&P.as_stats[BNG_AS_V6_VIOLATIONS], AF.v6viol);
+    stat_inc(&P.qos_stats[BNG_QS_PKT_PASSED], QF.passed);
+    stat_inc(&P.qos_stats[BNG_QS_PKT_DROPPED], QF.dropped);
+    stat_add(&P.qos_stats[BNG_QS_BYTES_PASSED], QF.bytes, QF.passed);
+    stat_add(&P.qos_stats[BNG_QS_BYTES_DROPPED], QF.bytes, QF.dropped);
+  }
+}
+
+/* classify packets for type-sorting: 1 = DHCP (UDP dst 67), 0 = other */
+__global__ void pkt_class_kernel(const uint8_t* __restrict__ data,
+                                 const uint16_t* __restrict__ in_len,
+                                 uint8_t* __restrict__ cls, int n,
+                                 int stride) {
+  int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  int nthreads = gridDim.x * blockDim.x;
+  for (int pid = tid; pid < n; pid += nthreads) {
+    pktctx c;
+    bool ip_ok = parse_pkt(c, const_cast<uint8_t*>(data) +
+                           (size_t)pid * stride, in_len[pid], true);
+    cls[pid] = (ip_ok && c.ip_off >= 0 && c.proto == 17 && c.l4_ok &&
+                c.dport == 67) ? 1 : 0;
   }
 }
 
@@ -1409,39 +1414,16 @@ void bng_launch_antispoof(void* data, const void* in_len, void* verdict,
       (bng_spoof_event*)ring, (bng_ring_header*)hdr, now_ns);
 }
 
-void bng_launch_uplink(void* data, const void* in_len, void* out_len,
-                       void* verdict, int n, int stride,
-                       const void* subs, uint32_t sub_mask,
-                       const void* pools, uint32_t n_pools,
-                       const void* scfg, void* dhcp_stats,
-                       const void* bindings, uint32_t bmask,
-                       const void* acfg, void* as_stats,
-                       void* spoof_ring, void* spoof_hdr,
-                       void* sessions, uint32_t sess_mask,
-                       void* reverse, uint32_t rev_mask,
-                       void* eim, uint32_t eim_mask,
-                       void* subnat, uint32_t subnat_mask,
-                       const void* ncfg, const void* hairpin,
-                       uint32_t n_hairpin, void* nat_stats,
-                       void* log_ring, void* log_hdr,
-                       void* qos_in, uint32_t qos_mask, void* qos_stats,
-                       uint64_t now_ns, uint64_t now_sec, hipStream_t s) {
-  hipLaunchKernelGGL(uplink_pipeline_kernel, dim3(pkt_grid(n)), dim3(256), 0,
-      s, (uint8_t*)data, (const uint16_t*)in_len, (uint16_t*)out_len,
-      (uint8_t*)verdict, n, stride,
-      (const bng_sub_entry*)subs, sub_mask, (const bng_ip_pool*)pools,
-      n_pools, (const bng_server_config*)scfg,
-      (unsigned long long*)dhcp_stats,
-      (const bng_binding_entry*)bindings, bmask,
-      (const bng_antispoof_config*)acfg, (unsigned long long*)as_stats,
-      (bng_spoof_event*)spoof_ring, (bng_ring_header*)spoof_hdr,
-      (bng_nat_session*)sessions, sess_mask, (bng_nat_reverse*)reverse,
-      rev_mask, (bng_eim_entry*)eim, eim_mask, (bng_subnat_entry*)subnat,
-      subnat_mask, (const bng_nat_config*)ncfg, (const uint32_t*)hairpin,
-      n_hairpin, (unsigned long long*)nat_stats,
-      (bng_nat_log_entry*)log_ring, (bng_ring_header*)log_hdr,
-      (bng_qos_bucket*)qos_in, qos_mask, (unsigned long long*)qos_stats,
-      now_ns, now_sec);
+void bng_launch_uplink(bng_uplink_params* P, hipStream_t s) {
+  hipLaunchKernelGGL(uplink_pipeline_kernel, dim3(pkt_grid(P->n)), dim3(256),
+                     0, s, *P);
+}
+
+void bng_launch_pkt_class(const void* data, const void* in_len, void* cls,
+                          int n, int stride, hipStream_t s) {
+  hipLaunchKernelGGL(pkt_class_kernel, dim3(pkt_grid(n)), dim3(256), 0, s,
+      (const uint8_t*)data, (const uint16_t*)in_len, (uint8_t*)cls, n,
+      stride);
 }
 
 void bng_launch_sub_upsert(void* t, uint32_t mask, const void* batch, int n,

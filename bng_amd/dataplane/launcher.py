@@ -387,11 +387,22 @@ class HipLauncher:
         return verdict
 
     def uplink(self, data, lens, now_ns: Optional[int] = None,
-               now_sec: Optional[int] = None):
-        """Fused antispoof -> NAT44 SNAT -> QoS-ingress + DHCP fast path."""
+               now_sec: Optional[int] = None, sort_by_type: bool = False):
+        """Fused antispoof -> NAT44 SNAT -> QoS-ingress + DHCP fast path.
+
+        sort_by_type=True first classifies packets on-device and feeds the
+        kernel a type-sorted index order, so each wave's 64 lanes run ONE
+        of the two pipelines instead of serializing both (the wave-
+        divergence fix; packet data itself is not moved)."""
         n = lens.numel()
         verdict, out_len = self._outs(n)
         now = now_ns if now_ns is not None else time.time_ns()
+        order = None
+        if sort_by_type and n > 64:
+            cls = self.torch.empty(n, dtype=self.torch.uint8,
+                                   device=self.device)
+            self.ext.pkt_class(data, lens, cls)
+            order = self.torch.argsort(cls, stable=True).to(self.torch.int32)
         self.ext.uplink_pipeline(
             data, lens, out_len, verdict, self.subs, self.pools,
             self.server_cfg, self.dhcp_stats, self.bindings, self.as_cfg,
@@ -399,7 +410,7 @@ class HipLauncher:
             self.reverse, self.eim, self.subnat, self.nat_cfg, self.hairpin,
             self.n_hairpin, self.nat_stats, self.nat_log_ring,
             self.nat_log_hdr, self.qos_ingress, self.qos_stats, now,
-            now_sec if now_sec is not None else now // 10**9)
+            now_sec if now_sec is not None else now // 10**9, order=order)
         return verdict, out_len
 
     def shard_owner(self, data, lens, n_shards: int):
