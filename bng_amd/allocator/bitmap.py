@@ -41,16 +41,26 @@ class BitmapAllocator:
         self.unit = 1 << (bits - self.alloc_prefix)
         self.reserve_head = reserve_head
         self.reserve_tail = reserve_tail
-        self._bitmap = bytearray((self.total + 7) // 8)
+        # dense bitmap for small pools; sparse set + rotor for huge ones
+        # (an IPv6 /64 address pool has 2^64 slots)
+        self.sparse = self.total > (1 << 22)
+        self._bitmap = bytearray(0 if self.sparse
+                                 else (self.total + 7) // 8)
+        self._used: set = set()
         self._by_sub: Dict[str, int] = {}
         self._by_idx: Dict[int, str] = {}
         self._hint = 0
 
     # ----------------------------------------------------------- bit ops
     def _test(self, idx: int) -> bool:
+        if self.sparse:
+            return idx in self._used
         return bool(self._bitmap[idx >> 3] & (1 << (idx & 7)))
 
     def _set(self, idx: int, v: bool):
+        if self.sparse:
+            (self._used.add if v else self._used.discard)(idx)
+            return
         if v:
             self._bitmap[idx >> 3] |= 1 << (idx & 7)
         else:
@@ -84,7 +94,10 @@ class BitmapAllocator:
         (ref bitmap.go Allocate)."""
         if subscriber_id in self._by_sub:
             return self._idx_to_prefix(self._by_sub[subscriber_id])
-        for i in range(self.total):
+        scan = self.total if not self.sparse else \
+            min(self.total, len(self._used) + self.reserve_head +
+                self.reserve_tail + 16)
+        for i in range(scan):
             idx = (self._hint + i) % self.total
             if self._is_reserved(idx) or self._test(idx):
                 continue

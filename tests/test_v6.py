@@ -1,0 +1,158 @@
+"""DHCPv6 + SLAAC tests (ref pkg/dhcpv6, pkg/slaac test strategies)."""
+import ipaddress
+import struct
+import time
+
+import pytest
+
+from bng_amd.dhcpv6.server import (ADVERTISE, DHCPv6Message, DHCPv6Server,
+                                   OPT_CLIENTID, OPT_IA_NA, OPT_IA_PD,
+                                   OPT_IAADDR, OPT_IAPREFIX,
+                                   OPT_RAPID_COMMIT, OPT_SERVERID,
+                                   OPT_STATUS_CODE, OPT_DNS_SERVERS,
+                                   RELEASE, RENEW, REPLY, REQUEST, SOLICIT,
+                                   STATUS_NOBINDING, parse_ia)
+from bng_amd.slaac.radvd import (PrefixConfig, RAConfig, Server as RadvServer,
+                                 build_ra, parse_ra)
+
+DUID = b"\x00\x03\x00\x01\xaa\xbb\xcc\x00\x00\x01"
+
+
+def solicit(duid=DUID, iaid=1, pd=False, rapid=False, txn=0x1234):
+    m = DHCPv6Message(SOLICIT, txn)
+    m.add(OPT_CLIENTID, duid)
+    m.add(OPT_IA_NA, struct.pack(">III", iaid, 0, 0))
+    if pd:
+        m.add(OPT_IA_PD, struct.pack(">III", iaid, 0, 0))
+    if rapid:
+        m.add(OPT_RAPID_COMMIT, b"")
+    return m
+
+
+def extract_addr(resp, opt=OPT_IA_NA, sub=OPT_IAADDR):
+    body = resp.get(opt)
+    _iaid, _t1, _t2, subs = parse_ia(body)
+    for t, v in subs:
+        if t == sub:
+            if sub == OPT_IAADDR:
+                return str(ipaddress.IPv6Address(v[:16]))
+            plen = v[8]
+            return f"{ipaddress.IPv6Address(v[9:25])}/{plen}"
+    return None
+
+
+class TestDHCPv6:
+    def test_four_message_exchange(self):
+        srv = DHCPv6Server(rapid_commit=False, dns=["2001:4860:4860::8888"])
+        adv_raw = srv.handle(solicit().encode())
+        adv = DHCPv6Message.decode(adv_raw)
+        assert adv.msg_type == ADVERTISE
+        addr = extract_addr(adv)
+        assert addr.startswith("2001:db8:1:")
+        # REQUEST echoes server id
+        req = solicit(txn=0x1235)
+        req.msg_type = REQUEST
+        req.add(OPT_SERVERID, srv.server_duid)
+        rep = DHCPv6Message.decode(srv.handle(req.encode()))
+        assert rep.msg_type == REPLY
+        assert extract_addr(rep) == addr       # sticky per DUID/IAID
+        assert (DUID, 1, False) in srv.bindings
+        assert rep.get(OPT_DNS_SERVERS) is not None
+
+    def test_rapid_commit(self):
+        srv = DHCPv6Server(rapid_commit=True)
+        rep = DHCPv6Message.decode(srv.handle(
+            solicit(rapid=True).encode()))
+        assert rep.msg_type == REPLY
+        assert rep.get(OPT_RAPID_COMMIT) is not None
+        assert srv.stats["rapid_commits"] == 1
+        assert (DUID, 1, False) in srv.bindings
+
+    def test_prefix_delegation(self):
+        srv = DHCPv6Server(rapid_commit=True, pd_prefix_len=56)
+        rep = DHCPv6Message.decode(srv.handle(
+            solicit(pd=True, rapid=True).encode()))
+        pd = extract_addr(rep, OPT_IA_PD, OPT_IAPREFIX)
+        assert pd.endswith("/56")
+        # a second client gets a different prefix
+        rep2 = DHCPv6Message.decode(srv.handle(
+            solicit(duid=DUID[:-1] + b"\x02", pd=True,
+                    rapid=True).encode()))
+        assert extract_addr(rep2, OPT_IA_PD, OPT_IAPREFIX) != pd
+
+    def test_renew_extends_and_unknown_nobinding(self):
+        srv = DHCPv6Server(rapid_commit=True)
+        srv.handle(solicit(rapid=True).encode())
+        b = srv.bindings[(DUID, 1, False)]
+        old_exp = b.expiry
+        time.sleep(0.01)
+        ren = solicit(txn=0x2222)
+        ren.msg_type = RENEW
+        ren.add(OPT_SERVERID, srv.server_duid)
+        rep = DHCPv6Message.decode(srv.handle(ren.encode()))
+        assert rep.msg_type == REPLY
+        assert b.expiry > old_exp
+        # unknown binding -> NoBinding status
+        other = solicit(duid=b"\x00\x03\x00\x01xxxxxx", txn=1)
+        other.msg_type = RENEW
+        rep2 = DHCPv6Message.decode(srv.handle(other.encode()))
+        _, _, _, subs = parse_ia(rep2.get(OPT_IA_NA))
+        st = [v for t, v in subs if t == OPT_STATUS_CODE][0]
+        assert struct.unpack(">H", st[:2])[0] == STATUS_NOBINDING
+
+    def test_release_frees_address(self):
+        srv = DHCPv6Server(rapid_commit=True)
+        rep = DHCPv6Message.decode(srv.handle(
+            solicit(rapid=True).encode()))
+        addr = extract_addr(rep)
+        rel = solicit(txn=3)
+        rel.msg_type = RELEASE
+        srv.handle(rel.encode())
+        assert (DUID, 1, False) not in srv.bindings
+        # address is reusable by another DUID
+        rep2 = DHCPv6Message.decode(srv.handle(
+            solicit(duid=DUID[:-1] + b"\x09", rapid=True).encode()))
+        assert extract_addr(rep2) == addr
+
+    def test_sweep_expired(self):
+        srv = DHCPv6Server(rapid_commit=True, valid_lifetime=1)
+        srv.handle(solicit(rapid=True).encode())
+        assert srv.sweep_expired(now=time.time() + 10) == 1
+        assert not srv.bindings
+
+
+class TestSLAAC:
+    def cfg(self):
+        return RAConfig(
+            prefixes=[PrefixConfig("2001:db8:2::/64")],
+            managed=False, other_config=True, mtu=1492,
+            rdnss=["2001:4860:4860::8888", "2001:4860:4860::8844"],
+            dnssl=["isp.example.com"], source_lladdr=b"\x02\x00\x00\x00\x00\x01")
+
+    def test_ra_roundtrip(self):
+        ra = parse_ra(build_ra(self.cfg()))
+        assert ra["other"] and not ra["managed"]
+        assert ra["mtu"] == 1492
+        assert ra["prefixes"][0]["prefix"] == "2001:db8:2::/64"
+        assert ra["prefixes"][0]["autonomous"]
+        assert ra["rdnss"] == ["2001:4860:4860::8888",
+                               "2001:4860:4860::8844"]
+        assert ra["dnssl"] == ["isp.example.com"]
+
+    def test_managed_flag_for_dhcpv6_deployments(self):
+        c = self.cfg()
+        c.managed = True
+        c.prefixes[0].autonomous = False
+        ra = parse_ra(build_ra(c))
+        assert ra["managed"] and not ra["prefixes"][0]["autonomous"]
+
+    def test_rs_triggers_solicited_ra(self):
+        sent = []
+        srv = RadvServer(self.cfg(), send_fn=lambda p, d: sent.append((p, d)))
+        rs = bytes([133, 0, 0, 0, 0, 0, 0, 0])
+        payload = srv.handle_rs(rs, src="fe80::1")
+        assert payload is not None
+        assert srv.stats["rs_received"] == 1
+        assert sent[0][1] == "fe80::1"
+        assert parse_ra(sent[0][0])["prefixes"]
+        assert srv.handle_rs(b"\x00" * 8) is None
