@@ -1,0 +1,131 @@
+"""Routing tests: BGP/BFD over a fake FRR executor, policy routing over
+the memory platform, subscriber-route retry/reconcile, health hysteresis
+(ref pkg/routing bgp_test.go / bfd_test.go patterns)."""
+import pytest
+
+from bng_amd.routing.bgp import BFDManager, BGPController
+from bng_amd.routing.frr import FakeExecutor, FRRError
+from bng_amd.routing.manager import (HealthChecker, Manager, MemoryPlatform,
+                                     Route, SubscriberRouteManager)
+
+
+class TestBGP:
+    def test_start_and_neighbors(self):
+        exe = FakeExecutor()
+        bgp = BGPController(exe, 65001, router_id="10.0.0.1",
+                            ecmp_paths=4).start()
+        bgp.add_neighbor("192.0.2.1", 65002, description="upstream",
+                         bfd=True, route_map_out="EXPORT")
+        cmds = exe.all_commands()
+        assert "router bgp 65001" in cmds
+        assert "bgp router-id 10.0.0.1" in cmds
+        assert "maximum-paths 4" in cmds
+        assert "neighbor 192.0.2.1 remote-as 65002" in cmds
+        assert "neighbor 192.0.2.1 bfd" in cmds
+        assert "neighbor 192.0.2.1 route-map EXPORT out" in cmds
+        bgp.remove_neighbor("192.0.2.1")
+        assert "no neighbor 192.0.2.1" in exe.all_commands()
+        assert bgp.neighbors == {}
+
+    def test_announce_withdraw(self):
+        exe = FakeExecutor()
+        bgp = BGPController(exe, 65001)
+        bgp.announce_prefix("203.0.113.0/24")
+        assert bgp.announced_prefixes() == ["203.0.113.0/24"]
+        bgp.withdraw_prefix("203.0.113.0/24")
+        assert bgp.announced_prefixes() == []
+        cmds = exe.all_commands()
+        assert "network 203.0.113.0/24" in cmds
+        assert "no network 203.0.113.0/24" in cmds
+
+    def test_bfd_peers_and_events(self):
+        exe = FakeExecutor()
+        bfd = BFDManager(exe)
+        events = []
+        bfd.on_state_change(lambda a, up: events.append((a, up)))
+        bfd.add_peer("192.0.2.1", interval_ms=50, multiplier=3)
+        cmds = exe.all_commands()
+        assert "peer 192.0.2.1" in cmds
+        assert "receive-interval 50" in cmds
+        bfd.handle_state_change("192.0.2.1", True)
+        bfd.handle_state_change("192.0.2.1", True)   # dedup
+        bfd.handle_state_change("192.0.2.1", False)
+        assert events == [("192.0.2.1", True), ("192.0.2.1", False)]
+
+
+class TestPolicyRouting:
+    def test_isp_tables_and_rules(self):
+        plat = MemoryPlatform()
+        m = Manager(plat)
+        t1 = m.create_isp_table("isp-a", "192.0.2.1")
+        t2 = m.create_isp_table("isp-b", "192.0.2.9")
+        assert t1 != t2
+        assert m.create_isp_table("isp-a", "x") == t1    # idempotent
+        assert plat.routes(t1)[0].next_hop == "192.0.2.1"
+        m.add_subscriber_rule("10.0.1.50", "isp-a")
+        m.add_subscriber_rule("10.0.1.51", "isp-b")
+        rules = {(r.src, r.table) for r in plat.rules()}
+        assert ("10.0.1.50", t1) in rules and ("10.0.1.51", t2) in rules
+        m.remove_subscriber_rule("10.0.1.50", "isp-a")
+        assert ("10.0.1.50", t1) not in {(r.src, r.table)
+                                         for r in plat.rules()}
+        with pytest.raises(KeyError):
+            m.add_subscriber_rule("10.0.1.52", "isp-zzz")
+
+
+class TestSubscriberRoutes:
+    def test_install_and_withdraw(self):
+        bgp = BGPController(FakeExecutor(), 65001)
+        srm = SubscriberRouteManager(bgp)
+        srm.add_subscriber_route("10.0.1.50")
+        assert "10.0.1.50/32" in bgp.announced_prefixes()
+        srm.remove_subscriber_route("10.0.1.50")
+        assert bgp.announced_prefixes() == []
+
+    def test_retry_queue_on_frr_failure(self):
+        exe = FakeExecutor(fail=True)
+        bgp = BGPController(exe, 65001)
+        srm = SubscriberRouteManager(bgp)
+        srm.add_subscriber_route("10.0.1.50")
+        assert srm.retry_queue == {"10.0.1.50/32": 1}
+        exe.fail = False                      # FRR recovers
+        assert srm.retry_pending() == 1
+        assert "10.0.1.50/32" in srm.installed
+        assert srm.retry_queue == {}
+
+    def test_reconcile_repairs_drift(self):
+        bgp = BGPController(FakeExecutor(), 65001)
+        srm = SubscriberRouteManager(bgp)
+        srm.add_subscriber_route("10.0.1.50")
+        # simulated drift: FRR lost the route
+        bgp.announced.clear()
+        srm.installed.clear()
+        assert srm.reconcile() == 1
+        assert "10.0.1.50/32" in bgp.announced_prefixes()
+        # stale: installed but no longer desired
+        srm.desired.clear()
+        assert srm.reconcile() == 1
+        assert bgp.announced_prefixes() == []
+
+
+class TestHealthChecker:
+    def test_hysteresis(self):
+        state = {"ok": True}
+        events = []
+        hc = HealthChecker("x", probe=lambda: state["ok"],
+                           up_threshold=2, down_threshold=3)
+        hc.on_change(events.append)
+        for _ in range(3):
+            hc.check_once()
+        assert hc.healthy
+        state["ok"] = False
+        hc.check_once(); hc.check_once()
+        assert hc.healthy                 # 2 < down_threshold
+        hc.check_once()
+        assert not hc.healthy
+        state["ok"] = True
+        hc.check_once()
+        assert not hc.healthy             # 1 < up_threshold
+        hc.check_once()
+        assert hc.healthy
+        assert events == [False, True]
