@@ -1,0 +1,182 @@
+"""Caching DNS resolver with TTL clamps, per-client rate limiting and
+intercept rules — the walled-garden DNS (ref pkg/dns/resolver.go:16-51,
+cache.go, types.go:223).
+
+Wire-level DNS codec for A/AAAA queries; upstream is a pluggable
+callable (UDP forwarder in production, fakes in tests)."""
+from __future__ import annotations
+
+import struct
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Callable, Dict, List, Optional, Tuple
+
+QTYPE_A = 1
+QTYPE_AAAA = 28
+
+
+def encode_qname(name: str) -> bytes:
+    out = b""
+    for label in name.strip(".").split("."):
+        out += bytes([len(label)]) + label.encode()
+    return out + b"\x00"
+
+
+def decode_qname(data: bytes, off: int) -> Tuple[str, int]:
+    labels = []
+    while off < len(data):
+        ln = data[off]
+        if ln == 0:
+            off += 1
+            break
+        if ln & 0xC0:        # compression pointer
+            ptr = struct.unpack_from(">H", data, off)[0] & 0x3FFF
+            sub, _ = decode_qname(data, ptr)
+            labels.append(sub)
+            off += 2
+            return ".".join(labels), off
+        labels.append(data[off + 1:off + 1 + ln].decode(errors="replace"))
+        off += 1 + ln
+    return ".".join(labels), off
+
+
+def build_query(name: str, qtype: int = QTYPE_A, txid: int = 0x1234) -> bytes:
+    return struct.pack(">HHHHHH", txid, 0x0100, 1, 0, 0, 0) + \
+        encode_qname(name) + struct.pack(">HH", qtype, 1)
+
+
+def build_response(query: bytes, addrs: List[str], ttl: int = 300,
+                   rcode: int = 0) -> bytes:
+    txid = struct.unpack_from(">H", query, 0)[0]
+    qname, off = decode_qname(query, 12)
+    qtype, qclass = struct.unpack_from(">HH", query, off)
+    hdr = struct.pack(">HHHHHH", txid, 0x8180 | rcode, 1, len(addrs), 0, 0)
+    out = hdr + query[12:off + 4]
+    import ipaddress
+    for a in addrs:
+        ip = ipaddress.ip_address(a)
+        rd = ip.packed
+        rtype = QTYPE_A if ip.version == 4 else QTYPE_AAAA
+        out += b"\xc0\x0c" + struct.pack(">HHIH", rtype, 1, ttl, len(rd)) + rd
+    return out
+
+
+def parse_response(data: bytes) -> Tuple[str, List[str], int]:
+    """-> (qname, addresses, min_ttl)"""
+    import ipaddress
+    _txid, _flags, qd, an, _ns, _ar = struct.unpack_from(">HHHHHH", data, 0)
+    off = 12
+    qname = ""
+    for _ in range(qd):
+        qname, off = decode_qname(data, off)
+        off += 4
+    addrs, min_ttl = [], 2**31
+    for _ in range(an):
+        _, off = decode_qname(data, off)
+        rtype, _rc, ttl, rdlen = struct.unpack_from(">HHIH", data, off)
+        off += 10
+        rd = data[off:off + rdlen]
+        off += rdlen
+        if rtype in (QTYPE_A, QTYPE_AAAA):
+            addrs.append(str(ipaddress.ip_address(rd)))
+            min_ttl = min(min_ttl, ttl)
+    return qname, addrs, 0 if min_ttl == 2**31 else min_ttl
+
+
+@dataclass
+class CacheEntry:
+    addrs: List[str]
+    expires: float
+
+
+class Resolver:
+    def __init__(self, upstream: Callable[[bytes], Optional[bytes]],
+                 min_ttl: int = 30, max_ttl: int = 3600,
+                 rate_limit: float = 0.0, rate_burst: int = 50):
+        self.upstream = upstream
+        self.min_ttl = min_ttl
+        self.max_ttl = max_ttl
+        self.cache: Dict[Tuple[str, int], CacheEntry] = {}
+        self.intercepts: Dict[str, List[str]] = {}   # name -> portal IPs
+        self.intercept_all_to: Optional[List[str]] = None
+        self._lock = threading.RLock()
+        self.rate_limit = rate_limit
+        self.rate_burst = rate_burst
+        self._buckets: Dict[str, List[float]] = {}
+        self.stats = {"queries": 0, "cache_hits": 0, "intercepted": 0,
+                      "rate_limited": 0, "upstream_fail": 0}
+
+    # -------------------------------------------------------- intercepts
+    def add_intercept(self, name: str, addrs: List[str]):
+        """Walled-garden DNS rule: this name resolves to the portal."""
+        with self._lock:
+            self.intercepts[name.rstrip(".").lower()] = addrs
+
+    def set_intercept_all(self, addrs: Optional[List[str]]):
+        """Quarantined clients: EVERY name resolves to the portal."""
+        self.intercept_all_to = addrs
+
+    def remove_intercept(self, name: str):
+        with self._lock:
+            self.intercepts.pop(name.rstrip(".").lower(), None)
+
+    # ------------------------------------------------------------ resolve
+    def _allowed(self, client: str) -> bool:
+        if self.rate_limit <= 0:
+            return True
+        now = time.monotonic()
+        with self._lock:
+            b = self._buckets.setdefault(client, [float(self.rate_burst),
+                                                  now])
+            b[0] = min(self.rate_burst, b[0] + (now - b[1]) * self.rate_limit)
+            b[1] = now
+            if b[0] >= 1:
+                b[0] -= 1
+                return True
+            return False
+
+    def handle_query(self, query: bytes, client: str = "",
+                     quarantined: bool = False) -> Optional[bytes]:
+        self.stats["queries"] += 1
+        if client and not self._allowed(client):
+            self.stats["rate_limited"] += 1
+            return None
+        try:
+            qname, off = decode_qname(query, 12)
+            qtype = struct.unpack_from(">H", query, off)[0]
+        except (struct.error, IndexError):
+            return None
+        key = qname.lower()
+        # intercept rules first (walled garden)
+        if quarantined and self.intercept_all_to:
+            self.stats["intercepted"] += 1
+            return build_response(query, self.intercept_all_to, ttl=30)
+        with self._lock:
+            hit = self.intercepts.get(key)
+        if hit is not None:
+            self.stats["intercepted"] += 1
+            return build_response(query, hit, ttl=30)
+        # cache
+        with self._lock:
+            ce = self.cache.get((key, qtype))
+            if ce is not None and ce.expires > time.time():
+                self.stats["cache_hits"] += 1
+                ttl = max(1, int(ce.expires - time.time()))
+                return build_response(query, ce.addrs, ttl=ttl)
+        # upstream
+        resp = None
+        try:
+            resp = self.upstream(query)
+        except Exception:
+            resp = None
+        if resp is None:
+            self.stats["upstream_fail"] += 1
+            return None
+        _qn, addrs, ttl = parse_response(resp)
+        ttl = max(self.min_ttl, min(self.max_ttl, ttl))   # TTL clamp
+        if addrs:
+            with self._lock:
+                self.cache[(key, qtype)] = CacheEntry(addrs,
+                                                      time.time() + ttl)
+        return resp

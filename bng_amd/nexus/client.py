@@ -121,6 +121,10 @@ class Client:
     def save_subscriber(self, sub: Subscriber):
         sub.updated_at = time.time()
         self.subscribers.put(sub.id, sub.to_dict())
+        with self._lock:
+            self._cache[sub.id] = sub
+            if sub.mac:
+                self._mac_index[sub.mac.lower()] = sub.id
 
     def get_isp(self, isp_id: str) -> Optional[ISPConfig]:
         d = self.isps.get(isp_id)

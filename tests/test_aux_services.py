@@ -1,0 +1,244 @@
+"""Tests for dns / deviceauth / ztp / agent / pon / direct / wifi / qinq."""
+import time
+
+import pytest
+
+from bng_amd.agent.agent import (S_CONNECTED, S_PARTITIONED, Agent)
+from bng_amd.deviceauth.authenticator import (AuthError, MTLSAuthenticator,
+                                              PSKAuthenticator,
+                                              new_authenticator)
+from bng_amd.direct.authenticator import (Authenticator as DirectAuth,
+                                          StubBSS)
+from bng_amd.dns.resolver import (Resolver, build_query, build_response,
+                                  parse_response)
+from bng_amd.nexus.client import Client as NexusClient
+from bng_amd.nexus.model import Subscriber
+from bng_amd.nexus.store import MemoryStore
+from bng_amd.pon.manager import Manager as PONManager, QoSProfile
+from bng_amd.qinq.mapper import Mapper as QinQMapper, QinQError
+from bng_amd.wifi.gateway import Gateway
+from bng_amd.ztp.bootstrap import (BootstrapClient, ZTPServer,
+                                   discover_nexus_from_dhcp_options)
+
+
+class TestDNS:
+    def upstream(self, answers):
+        def fn(q):
+            return build_response(q, answers, ttl=500)
+        return fn
+
+    def test_resolve_and_cache(self):
+        calls = []
+        up = self.upstream(["93.184.216.34"])
+        def counting(q):
+            calls.append(1)
+            return up(q)
+        r = Resolver(counting)
+        resp = r.handle_query(build_query("example.com"))
+        _, addrs, _ = parse_response(resp)
+        assert addrs == ["93.184.216.34"]
+        r.handle_query(build_query("example.com"))
+        assert len(calls) == 1            # second from cache
+        assert r.stats["cache_hits"] == 1
+
+    def test_ttl_clamp(self):
+        r = Resolver(self.upstream(["1.2.3.4"]), min_ttl=60, max_ttl=120)
+        r.handle_query(build_query("x.com"))
+        ce = r.cache[("x.com", 1)]
+        assert 110 <= ce.expires - time.time() <= 121
+
+    def test_intercept_walled_garden(self):
+        r = Resolver(self.upstream(["1.2.3.4"]))
+        r.add_intercept("portal.isp.com", ["10.0.0.10"])
+        resp = r.handle_query(build_query("portal.isp.com"))
+        _, addrs, _ = parse_response(resp)
+        assert addrs == ["10.0.0.10"]
+        # quarantined client: EVERYTHING goes to the portal
+        r.set_intercept_all(["10.0.0.10"])
+        resp = r.handle_query(build_query("facebook.com"),
+                              quarantined=True)
+        _, addrs, _ = parse_response(resp)
+        assert addrs == ["10.0.0.10"]
+        assert r.stats["intercepted"] == 2
+
+    def test_rate_limit(self):
+        r = Resolver(self.upstream(["1.2.3.4"]), rate_limit=1,
+                     rate_burst=2)
+        q = build_query("a.com")
+        assert r.handle_query(q, client="c1") is not None
+        assert r.handle_query(build_query("b.com"), client="c1") is not None
+        assert r.handle_query(build_query("c.com"), client="c1") is None
+        assert r.stats["rate_limited"] == 1
+
+
+class TestDeviceAuth:
+    def test_psk_roundtrip_and_replay_window(self):
+        a = PSKAuthenticator(b"secret", window=300)
+        h = a.headers("olt-1")
+        assert a.verify(h) == "olt-1"
+        h2 = dict(h)
+        h2["X-Auth-Timestamp"] = str(int(time.time()) - 10_000)
+        with pytest.raises(AuthError):
+            a.verify(h2)
+        h3 = dict(h)
+        h3["X-Auth-Signature"] = "00" * 32
+        with pytest.raises(AuthError):
+            a.verify(h3)
+
+    def test_mtls_fingerprints(self):
+        a = MTLSAuthenticator()
+        a.register("olt-1", b"CERT-PEM-BYTES")
+        assert a.verify_cert("olt-1", b"CERT-PEM-BYTES") == "olt-1"
+        with pytest.raises(AuthError):
+            a.verify_cert("olt-1", b"EVIL")
+        with pytest.raises(AuthError):
+            a.verify_cert("olt-2", b"CERT-PEM-BYTES")
+
+    def test_factory(self):
+        assert new_authenticator("none").mode == "none"
+        assert new_authenticator("psk", psk=b"x").mode == "psk"
+        with pytest.raises(ValueError):
+            new_authenticator("wat")
+
+
+class TestZTP:
+    def test_bootstrap_flow(self):
+        srv = ZTPServer()
+        try:
+            c = BootstrapClient(srv.url, serial="SN123",
+                                poll_interval=0.05)
+            c.register()
+            assert c.state == "registered"
+            assert "SN123" in srv.devices
+            # operator approves with a config incl. HA partner + pool
+            srv.approve("SN123", {
+                "device_id": "bng-7", "role": "standby",
+                "ha_partner": "http://10.0.0.8:8443",
+                "pool_network": "10.7.0.0/16"})
+            cfg = c.poll_until_approved(timeout=5)
+            assert c.state == "approved"
+            assert cfg.device_id == "bng-7"
+            assert cfg.role == "standby"
+            assert cfg.ha_partner.endswith(":8443")
+            assert cfg.pool_network == "10.7.0.0/16"
+        finally:
+            srv.stop()
+
+    def test_dhcp_option_discovery(self):
+        assert discover_nexus_from_dhcp_options(
+            {224: b"https://nexus.isp:8443"}) == "https://nexus.isp:8443"
+        assert discover_nexus_from_dhcp_options(
+            {43: b"http://n"}) == "http://n"
+        assert discover_nexus_from_dhcp_options({43: b"\xff\x01x"}) is None
+        assert discover_nexus_from_dhcp_options({}) is None
+
+
+class TestAgent:
+    def test_config_watch_and_states(self):
+        store = MemoryStore()
+        a = Agent(store, "node-1", partition_after=0.0)
+        states, cfgs = [], []
+        a.on_state_change(lambda o, n: states.append(n))
+        a.on_config_change(cfgs.append)
+        a.start()
+        try:
+            assert a.state == S_CONNECTED
+            store.put("nexus/device_configs/node-1",
+                      b'{"pool": "10.9.0.0/24"}')
+            assert cfgs and cfgs[-1]["pool"] == "10.9.0.0/24"
+            # partition: store writes fail
+            orig = store.put
+            store.put = lambda *a_, **k: (_ for _ in ()).throw(OSError())
+            assert not a.heartbeat_once()
+            assert a.state == S_PARTITIONED
+            store.put = orig
+            a.heartbeat_once()
+            a.heartbeat_once()
+            assert a.state == S_CONNECTED
+            assert S_PARTITIONED in states
+        finally:
+            a.stop()
+
+
+class TestPON:
+    def test_discovery_and_provisioning(self):
+        store = MemoryStore()
+        qm = QinQMapper()
+        qm.add_range(100, 2, 100)
+        pon = PONManager(store, vlan_mapper=qm)
+        pon.add_profile(QoSProfile("residential", 1000, 200))
+        events = []
+        pon.on_event(lambda ev, n: events.append((ev, n.id)))
+        nte = pon.ont_discovered("ONT123", "pon0/1")
+        assert nte.state == "discovered"
+        p = pon.provision(nte.id, profile="residential")
+        assert p.provisioned and p.state == "active"
+        assert (p.s_tag, p.c_tag) == (100, 2)
+        pon.ont_offline(nte.id)
+        assert [e[0] for e in events] == ["discovered", "provisioned",
+                                         "offline"]
+        with pytest.raises(KeyError):
+            pon.provision(nte.id, profile="nope")
+
+
+class TestDirect:
+    def make(self):
+        store = MemoryStore()
+        c = NexusClient(store)
+        c.save_subscriber(Subscriber("sub-1", s_tag=100, c_tag=5,
+                                     mac="aa:bb:cc:00:00:01",
+                                     isp_id="isp-a"))
+        return c
+
+    def test_vlan_and_mac_identity(self):
+        c = self.make()
+        d = DirectAuth(c)
+        r = d.authenticate_by_vlan(100, 5)
+        assert r.success and r.subscriber_id == "sub-1"
+        r2 = d.authenticate_by_mac("aa:bb:cc:00:00:01")
+        assert r2.success and r2.isp_id == "isp-a"
+        assert not d.authenticate_by_vlan(1, 1).success
+
+    def test_bss_suspension(self):
+        c = self.make()
+        d = DirectAuth(c, bss=StubBSS({"sub-1": "suspended"}))
+        r = d.authenticate_by_mac("aa:bb:cc:00:00:01")
+        assert not r.success and "suspended" in r.reason
+
+
+class TestWiFi:
+    def test_guest_lifecycle(self):
+        gw = Gateway(network="192.168.100.0/28")
+        s = gw.join("AA:BB:CC:00:00:01")
+        assert gw.is_quarantined("aa:bb:cc:00:00:01")
+        assert gw.accept_terms("aa:bb:cc:00:00:01")
+        assert not gw.is_quarantined("aa:bb:cc:00:00:01")
+        # idle guest expires after grace epochs; active one renewed
+        gw.join("aa:bb:cc:00:00:02")
+        gw.advance_epoch()
+        gw.touch("aa:bb:cc:00:00:01")     # renews s1 only
+        gw.advance_epoch()
+        assert gw.session_count() == 1
+        assert gw.stats["expired"] == 1
+
+    def test_exhaustion(self):
+        gw = Gateway(network="192.168.100.0/30")   # 2 usable (net,bcast out)
+        gw.join("aa:bb:cc:00:00:01")
+        gw.join("aa:bb:cc:00:00:02")
+        from bng_amd.allocator.epoch_bitmap import PoolExhaustedError
+        with pytest.raises(PoolExhaustedError):
+            gw.join("aa:bb:cc:00:00:09")
+        assert gw.stats["exhausted"] == 1
+
+
+class TestQinQ:
+    def test_register_conflicts_and_auto_assign(self):
+        m = QinQMapper()
+        m.add_range(200, 2, 4)
+        assert m.register("sub-1", 200, 2) == (200, 2)
+        with pytest.raises(QinQError):
+            m.register("sub-2", 200, 2)
+        assert m.auto_assign("sub-2") == (200, 3)
+        assert m.lookup(200, 3) == "sub-2"
+        m.unregister("sub-2")
+        assert m.auto_assign("sub-3") == (200, 3)
