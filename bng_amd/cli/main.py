@@ -1,0 +1,467 @@
+"""bng CLI (ref cmd/bng/main.go): `run` wires every subsystem in the
+reference's order (main.go:441-1298) with reverse-order cleanup
+(:1300-1379); `demo` simulates the ONT->walled-garden->activation
+lifecycle with no dataplane (demo.go:46-60); `stats`; `version`.
+
+Flags mirror the reference's surface (~110 cobra flags; the load-bearing
+ones are implemented, YAML config supplies the rest); file values are
+applied only to flags not explicitly set on the command line
+(ref main.go:1420-1457), and secrets support --*-file indirection
+(ref resolveSecret :1567-1592).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import logging
+import os
+import signal
+import sys
+import time
+from typing import List, Optional
+
+from .. import __version__
+
+
+def build_parser() -> argparse.ArgumentParser:
+    p = argparse.ArgumentParser(prog="bng",
+                                description="MI355X-native BNG")
+    sub = p.add_subparsers(dest="command")
+
+    run = sub.add_parser("run", help="run the BNG")
+    g = run.add_argument_group("core")
+    g.add_argument("--config", help="YAML config file")
+    g.add_argument("--log-level", default="info",
+                   choices=["debug", "info", "warn", "error"])
+    g.add_argument("--node-id", default="bng-1")
+    g.add_argument("--interface", default="eth0")
+    g.add_argument("--server-ip", default="10.0.0.1")
+    g.add_argument("--server-mac", default="02:00:00:00:00:01")
+    g.add_argument("--gpu", default="auto",
+                   help="auto|off|cuda:N — dataplane device")
+    g = run.add_argument_group("dhcp")
+    g.add_argument("--pool-network", default="")
+    g.add_argument("--pool-gateway", default="")
+    g.add_argument("--pool-dns", default="")
+    g.add_argument("--lease-time", type=int, default=3600)
+    g.add_argument("--dhcp-listen", action="store_true",
+                   help="bind UDP :67 (off in tests)")
+    g = run.add_argument_group("radius")
+    g.add_argument("--radius-server", action="append", default=[])
+    g.add_argument("--radius-secret", default="")
+    g.add_argument("--radius-secret-file", default="")
+    g.add_argument("--radius-auth-mode", default="none",
+                   choices=["none", "mac"])
+    g.add_argument("--radius-coa-port", type=int, default=0)
+    g = run.add_argument_group("nexus")
+    g.add_argument("--nexus-url", default="")
+    g.add_argument("--nexus-auth", default="none",
+                   choices=["none", "psk", "mtls"])
+    g.add_argument("--nexus-psk", default="")
+    g = run.add_argument_group("peers")
+    g.add_argument("--peer", action="append", default=[],
+                   help="node_id=url")
+    g = run.add_argument_group("nat")
+    g.add_argument("--nat-enable", action="store_true")
+    g.add_argument("--nat-public-ip", action="append", default=[])
+    g.add_argument("--nat-ports-per-subscriber", type=int, default=1024)
+    g.add_argument("--nat-log-path", default="")
+    g.add_argument("--nat-log-format", default="json",
+                   choices=["json", "csv", "syslog", "nel"])
+    g.add_argument("--nat-bulk-logging", action="store_true")
+    g = run.add_argument_group("qos")
+    g.add_argument("--qos-policy", action="append", default=[],
+                   help="name:down_mbps:up_mbps")
+    g.add_argument("--qos-default-policy", default="")
+    g = run.add_argument_group("antispoof")
+    g.add_argument("--antispoof-mode", default="disabled",
+                   choices=["disabled", "strict", "loose", "log_only"])
+    g = run.add_argument_group("pppoe")
+    g.add_argument("--pppoe-enable", action="store_true")
+    g.add_argument("--pppoe-ac-name", default="bng-amd")
+    g.add_argument("--pppoe-auth", default="chap",
+                   choices=["chap", "pap", "none"])
+    g = run.add_argument_group("ipv6")
+    g.add_argument("--dhcpv6-enable", action="store_true")
+    g.add_argument("--dhcpv6-na-pool", default="2001:db8:1::/64")
+    g.add_argument("--dhcpv6-pd-pool", default="2001:db8:f000::/40")
+    g.add_argument("--slaac-enable", action="store_true")
+    g.add_argument("--slaac-prefix", default="")
+    g = run.add_argument_group("routing")
+    g.add_argument("--bgp-enable", action="store_true")
+    g.add_argument("--bgp-local-as", type=int, default=65000)
+    g.add_argument("--bgp-neighbor", action="append", default=[],
+                   help="addr:remote_as")
+    g.add_argument("--bgp-announce-subscribers", action="store_true")
+    g = run.add_argument_group("ha")
+    g.add_argument("--ha-role", default="",
+                   choices=["", "active", "standby"])
+    g.add_argument("--ha-partner-url", default="")
+    g.add_argument("--ha-listen-port", type=int, default=0)
+    g = run.add_argument_group("observability")
+    g.add_argument("--metrics-port", type=int, default=9090)
+    g.add_argument("--metrics-enable", action="store_true")
+    g.add_argument("--audit-log-path", default="")
+    g = run.add_argument_group("walledgarden")
+    g.add_argument("--walled-garden-portal", default="")
+
+    demo = sub.add_parser("demo", help="simulated subscriber lifecycle")
+    demo.add_argument("--subscribers", type=int, default=3)
+
+    sub.add_parser("stats", help="print runtime stats of a config")
+    sub.add_parser("version", help="print version")
+    return p
+
+
+def load_yaml_over_args(args: argparse.Namespace,
+                        parser: argparse.ArgumentParser,
+                        argv: List[str]) -> argparse.Namespace:
+    """Merge YAML config under explicit flags (ref main.go:1420-1457):
+    file values apply only to flags NOT set on the command line."""
+    if not getattr(args, "config", None):
+        return args
+    import yaml
+    with open(args.config) as f:
+        cfg = yaml.safe_load(f) or {}
+    explicit = set()
+    for tok in argv:
+        if tok.startswith("--"):
+            explicit.add(tok.split("=")[0][2:].replace("-", "_"))
+    for key, value in cfg.items():
+        attr = key.replace("-", "_")
+        if attr in explicit or not hasattr(args, attr):
+            continue
+        cur = getattr(args, attr)
+        if isinstance(cur, list) and not isinstance(value, list):
+            value = [value]
+        setattr(args, attr, value)
+    return args
+
+
+def resolve_secret(value: str, file_value: str) -> str:
+    """ref main.go:1567-1592 resolveSecret."""
+    if file_value:
+        with open(file_value) as f:
+            return f.read().strip()
+    return value
+
+
+class BNG:
+    """The wired application (runBNG analog).  Subsystems start in the
+    reference's order and stop in reverse."""
+
+    def __init__(self, args: argparse.Namespace):
+        self.args = args
+        self.log = logging.getLogger("bng")
+        self._cleanup: List = []
+        self.launcher = None
+        self.dhcp_server = None
+        self.metrics = None
+
+    def _defer(self, fn):
+        self._cleanup.append(fn)
+
+    def start(self):
+        a = self.args
+        from ..dataplane.launcher import GoldenLauncher, HipLauncher
+        from ..dataplane.packets import ip2u32, mac_bytes
+
+        # 1. dataplane launcher (the eBPF loader analog, main.go:498)
+        use_gpu = False
+        if a.gpu != "off":
+            try:
+                import torch
+                use_gpu = torch.cuda.is_available()
+            except Exception:
+                use_gpu = False
+        if use_gpu:
+            self.launcher = HipLauncher(
+                a.gpu if a.gpu not in ("auto",) else "cuda:0")
+            self.log.info("GPU dataplane on %s", self.launcher.device)
+        else:
+            self.launcher = GoldenLauncher()
+            self.log.info("CPU (golden-model) dataplane")
+
+        # 2. antispoof (main.go:532)
+        from ..antispoof.manager import Manager as AntispoofMgr
+        self.antispoof = AntispoofMgr(self.launcher,
+                                      default_mode=a.antispoof_mode)
+
+        # 3. walled garden (main.go:556)
+        from ..walledgarden.manager import Manager as WGMgr
+        self.walledgarden = WGMgr(portal_ip=a.walled_garden_portal).start()
+        self._defer(self.walledgarden.stop)
+
+        # 4. pools + DHCP server (main.go:567-594, :1244)
+        from ..dhcp.pool import PoolConfig, PoolManager
+        from ..dhcp.server import DHCPServer
+        self.pool_manager = PoolManager(self.launcher)
+        if a.pool_network:
+            self.pool_manager.add_pool(PoolConfig(
+                1, a.pool_network, gateway=a.pool_gateway,
+                dns=[d for d in a.pool_dns.split(",") if d],
+                lease_time=a.lease_time))
+        self.dhcp_server = DHCPServer(self.pool_manager, a.server_ip,
+                                      mac_bytes(a.server_mac),
+                                      lease_time=a.lease_time)
+        self.dhcp_server.set_launcher(self.launcher)
+        self.dhcp_server.set_walled_garden(self.walledgarden)
+
+        # 5. nexus (main.go:653-689)
+        if a.nexus_url:
+            from ..nexus.http_allocator import HTTPAllocator
+            headers = None
+            if a.nexus_auth == "psk" and a.nexus_psk:
+                from ..deviceauth.authenticator import PSKAuthenticator
+                headers = PSKAuthenticator(a.nexus_psk).headers(a.node_id)
+            self.nexus_allocator = HTTPAllocator(a.nexus_url,
+                                                 auth_headers=headers)
+            self.dhcp_server.set_nexus(allocator=self.nexus_allocator)
+
+        # 6. peer pool (main.go:719-756)
+        if a.peer and a.pool_network:
+            from ..pool.peer import PeerPool
+            peers = dict(p.split("=", 1) for p in a.peer)
+            self.peer_pool = PeerPool(a.node_id, peers,
+                                      a.pool_network).start()
+            self._defer(self.peer_pool.stop)
+            self.dhcp_server.set_peer_pool(self.peer_pool)
+
+        # 7. HA (main.go:826-881)
+        if a.ha_role:
+            from ..ha.failover import FailoverController
+            from ..ha.health_monitor import HealthMonitor
+            from ..ha.sync import HASyncer
+            self.ha = HASyncer(a.node_id, a.ha_role,
+                               listen_port=a.ha_listen_port,
+                               partner_url=a.ha_partner_url).start()
+            self._defer(self.ha.stop)
+            if a.ha_partner_url:
+                self.ha_monitor = HealthMonitor(a.ha_partner_url).start()
+                self._defer(self.ha_monitor.stop)
+                self.ha_failover = FailoverController(
+                    a.node_id, a.ha_role, monitor=self.ha_monitor,
+                    role_change_callback=lambda r: self.ha.promote()
+                    if r == "active" else self.ha.demote())
+
+        # 8. routing (main.go:901-939)
+        if a.bgp_enable:
+            from ..routing.bgp import BFDManager, BGPController
+            from ..routing.frr import FakeExecutor, VtyshExecutor
+            exe = VtyshExecutor() if os.path.exists("/usr/bin/vtysh") \
+                else FakeExecutor()
+            self.bgp = BGPController(exe, a.bgp_local_as).start()
+            for n in a.bgp_neighbor:
+                addr, _, ras = n.partition(":")
+                self.bgp.add_neighbor(addr, int(ras or 65000))
+            if a.bgp_announce_subscribers:
+                from ..routing.manager import SubscriberRouteManager
+                self.sub_routes = SubscriberRouteManager(self.bgp).start()
+                self._defer(self.sub_routes.stop)
+
+        # 9. RADIUS + policies + QoS (main.go:951-985)
+        from ..radius.policy import Policy, PolicyManager
+        self.policy_manager = PolicyManager()
+        for spec in a.qos_policy:
+            name, down, up = spec.split(":")
+            self.policy_manager.add_policy(Policy(
+                name, int(float(down) * 1e6), int(float(up) * 1e6)))
+        if a.qos_default_policy:
+            self.policy_manager.default_policy = \
+                self.policy_manager.get(a.qos_default_policy)
+        from ..qos.manager import Manager as QoSMgr
+        self.qos = QoSMgr(self.launcher, self.policy_manager)
+        self.dhcp_server.set_qos_manager(self.qos)
+        self.dhcp_server.set_policy_manager(self.policy_manager)
+        if a.radius_server:
+            from ..radius.accounting import AccountingManager
+            from ..radius.client import Client as RadiusClient
+            secret = resolve_secret(a.radius_secret, a.radius_secret_file)
+            self.radius = RadiusClient(a.radius_server, secret.encode())
+            self.dhcp_server.set_radius(self.radius, a.radius_auth_mode)
+            self.accounting = AccountingManager(self.radius).start()
+            self._defer(self.accounting.stop)
+            self.dhcp_server.set_accounting(self.accounting)
+            if a.radius_coa_port:
+                from ..radius.coa import CoAProcessor, CoAServer
+                proc = CoAProcessor(
+                    session_lookup=self._coa_lookup,
+                    terminate=self._coa_terminate,
+                    qos_updater=lambda lease, pol:
+                        self.qos.update_subscriber_policy(lease.ip, pol))
+                self.coa = CoAServer(secret.encode(),
+                                     port=a.radius_coa_port,
+                                     handler=proc).start()
+                self._defer(self.coa.stop)
+
+        # 10. NAT (main.go:1001-1040)
+        if a.nat_enable:
+            from ..nat.logging import ComplianceLogger
+            from ..nat.manager import Manager as NATMgr
+            logger = ComplianceLogger(a.nat_log_path or None,
+                                      fmt=a.nat_log_format,
+                                      bulk_mode=a.nat_bulk_logging)
+            self.nat = NATMgr(self.launcher,
+                              ports_per_subscriber=a.nat_ports_per_subscriber,
+                              logger=logger)
+            for ip in a.nat_public_ip:
+                self.nat.add_public_ip(ip)
+            self.nat.start()
+            self._defer(self.nat.stop)
+            self.dhcp_server.set_nat_manager(self.nat)
+
+        # 11. PPPoE / DHCPv6 / SLAAC (main.go:1063-1180)
+        if a.pppoe_enable:
+            from ..pppoe.server import PPPoEServer
+            from ..dataplane.packets import mac_bytes as mb
+            self.pppoe = PPPoEServer(mb(a.server_mac),
+                                     ac_name=a.pppoe_ac_name,
+                                     auth=a.pppoe_auth)
+            if getattr(self, "radius", None):
+                self.pppoe.radius = self.radius
+        if a.dhcpv6_enable:
+            from ..dhcpv6.server import DHCPv6Server
+            self.dhcpv6 = DHCPv6Server(na_pool=a.dhcpv6_na_pool,
+                                       pd_pool=a.dhcpv6_pd_pool)
+        if a.slaac_enable and a.slaac_prefix:
+            from ..slaac.radvd import PrefixConfig, RAConfig, Server
+            self.slaac = Server(RAConfig(
+                prefixes=[PrefixConfig(a.slaac_prefix)])).start()
+            self._defer(self.slaac.stop)
+
+        # 12. metrics (main.go:1214-1241)
+        if a.metrics_enable:
+            from ..metrics.metrics import Metrics
+            self.metrics = Metrics()
+            self.metrics.start_collector(self.launcher, self.dhcp_server)
+            try:
+                self.metrics.serve(port=a.metrics_port)
+                self._defer(self.metrics.stop)
+            except OSError:
+                self.log.warning("metrics port busy; collector only")
+
+        # 13. DHCP serve loop last (main.go:1244)
+        self.dhcp_server.start(serve=a.dhcp_listen)
+        self._defer(self.dhcp_server.stop)
+        return self
+
+    def _coa_lookup(self, req):
+        for lease in self.dhcp_server.leases.values():
+            from ..dataplane.packets import u32_to_ip
+            if req.framed_ip and u32_to_ip(lease.ip) == req.framed_ip:
+                return lease
+            if req.session_id and getattr(lease, "_acct_id", "") == \
+                    req.session_id:
+                return lease
+        return None
+
+    def _coa_terminate(self, lease):
+        from ..dhcp import message as dm
+        msg = dm.DHCPMessage()
+        msg.op = 1
+        msg.chaddr = lease.mac
+        msg.set_option(dm.OPT_MSG_TYPE, bytes([dm.RELEASE]))
+        self.dhcp_server.handle_release(msg)
+        return True
+
+    def stop(self):
+        for fn in reversed(self._cleanup):
+            try:
+                fn()
+            except Exception:
+                pass
+        self._cleanup.clear()
+
+    def stats(self) -> dict:
+        out = {"dhcp": dict(self.dhcp_server.stats),
+               "leases": len(self.dhcp_server.leases),
+               "fastpath": self.launcher.get_stats()}
+        if getattr(self, "nat", None):
+            out["nat"] = self.nat.get_stats()
+        return out
+
+
+def cmd_demo(args) -> int:
+    """ref cmd/bng/demo.go:46-60: full subscriber lifecycle with
+    in-memory Nexus and no dataplane."""
+    from ..nexus.client import Client
+    from ..nexus.model import IPPool, ISPConfig, Subscriber
+    from ..nexus.store import MemoryStore
+    from ..pon.manager import Manager as PONMgr, QoSProfile
+    from ..qinq.mapper import Mapper
+    from ..walledgarden.manager import Manager as WGMgr
+
+    store = MemoryStore()
+    nexus = Client(store)
+    nexus.pools.put("pool-1", IPPool("pool-1", "10.0.1.0/24").to_dict())
+    nexus.isps.put("isp-1", ISPConfig("isp-1",
+                                      ipv4_pools=["pool-1"]).to_dict())
+    qm = Mapper()
+    qm.add_range(100)
+    pon = PONMgr(store, vlan_mapper=qm)
+    pon.add_profile(QoSProfile("residential", 1000, 200))
+    wg = WGMgr(portal_ip="10.0.0.10")
+
+    print(f"bng demo — {args.subscribers} subscriber(s)")
+    for i in range(args.subscribers):
+        serial = f"ONT{i:04d}"
+        mac = f"aa:bb:cc:00:{i >> 8:02x}:{i & 0xFF:02x}"
+        nte = pon.ont_discovered(serial, f"pon0/{i}")
+        print(f"  [{serial}] ONT discovered on {nte.pon_port}")
+        wg.add(mac, "0.0.0.0", reason="new ONT")
+        print(f"  [{serial}] quarantined in walled garden")
+        nte = pon.provision(nte.id, profile="residential")
+        print(f"  [{serial}] provisioned s_tag={nte.s_tag} "
+              f"c_tag={nte.c_tag}")
+        sub = Subscriber(f"sub-{serial}", nte_id=nte.id, isp_id="isp-1",
+                         s_tag=nte.s_tag, c_tag=nte.c_tag, mac=mac)
+        nexus.save_subscriber(sub)
+        ip = nexus.allocate_ip_for_subscriber(sub.id)
+        print(f"  [{serial}] RADIUS-time allocation -> {ip}")
+        wg.activate(mac)
+        print(f"  [{serial}] activated; DHCP is now a pure read: "
+              f"{nexus.lookup_subscriber_ip(sub.id)}")
+    print("demo complete")
+    return 0
+
+
+def main(argv: Optional[List[str]] = None) -> int:
+    argv = argv if argv is not None else sys.argv[1:]
+    parser = build_parser()
+    args = parser.parse_args(argv)
+    if args.command == "version":
+        print(f"bng {__version__} (MI355X gfx950 dataplane)")
+        return 0
+    if args.command == "demo":
+        return cmd_demo(args)
+    if args.command == "run":
+        logging.basicConfig(
+            level=getattr(logging, args.log_level.upper().replace(
+                "WARN", "WARNING")),
+            format="%(asctime)s %(levelname)s %(name)s %(message)s")
+        args = load_yaml_over_args(args, parser, argv)
+        app = BNG(args).start()
+        stop = {"flag": False}
+
+        def on_sig(*_):
+            stop["flag"] = True
+        signal.signal(signal.SIGINT, on_sig)
+        signal.signal(signal.SIGTERM, on_sig)
+        print(f"bng running (node {args.node_id}); ^C to stop",
+              flush=True)
+        try:
+            while not stop["flag"]:
+                time.sleep(0.2)
+        finally:
+            app.stop()
+        return 0
+    if args.command == "stats":
+        print(json.dumps({"error": "connect to a running bng via "
+                          "metrics endpoint"}, indent=2))
+        return 1
+    parser.print_help()
+    return 2
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,111 @@
+"""CLI + full-wiring integration tests — BASELINE config 1:
+`bng run --pool-network 10.0.1.0/24` slow-path DHCP on CPU with 16
+subscribers, no GPU (ref cmd/bng runBNG wiring + demo)."""
+import json
+import time
+
+import pytest
+
+from bng_amd.cli.main import BNG, build_parser, cmd_demo, load_yaml_over_args
+from bng_amd.dataplane.packets import mac_bytes, u32_to_ip
+from bng_amd.dhcp import message as dm
+
+
+def make_app(extra=None):
+    argv = ["run", "--pool-network", "10.0.1.0/24",
+            "--pool-gateway", "10.0.1.1", "--pool-dns", "8.8.8.8",
+            "--gpu", "off"] + (extra or [])
+    args = build_parser().parse_args(argv)
+    return BNG(args).start(), argv
+
+
+class TestRunWiring:
+    def test_baseline_config1_16_subscribers(self):
+        """Standalone slow-path DHCP, 16 subscribers, no GPU."""
+        app, _ = make_app()
+        try:
+            ips = set()
+            for i in range(16):
+                mac = mac_bytes(f"aa:bb:cc:00:00:{i:02x}")
+                offer = app.dhcp_server.handle(
+                    dm.build_request(mac, dm.DISCOVER, xid=i))
+                assert offer is not None and offer.msg_type == dm.OFFER
+                ack = app.dhcp_server.handle(dm.build_request(
+                    mac, dm.REQUEST, xid=i, requested_ip=offer.yiaddr))
+                assert ack.msg_type == dm.ACK
+                ips.add(ack.yiaddr)
+            assert len(ips) == 16
+            st = app.stats()
+            assert st["dhcp"]["ack"] == 16
+            assert st["leases"] == 16
+            # fast-path mirror has all 16 subscribers
+            fp = app.launcher.dp.subscribers
+            assert len(fp) == 16
+        finally:
+            app.stop()
+
+    def test_full_stack_wiring(self):
+        app, _ = make_app([
+            "--nat-enable", "--nat-public-ip", "203.0.113.1",
+            "--qos-policy", "gold:100:20", "--qos-default-policy", "gold",
+            "--antispoof-mode", "strict", "--pppoe-enable",
+            "--dhcpv6-enable", "--metrics-enable", "--metrics-port", "0",
+            "--bgp-enable", "--bgp-announce-subscribers",
+            "--walled-garden-portal", "10.0.0.10"])
+        try:
+            mac = mac_bytes("aa:bb:cc:00:00:99")
+            offer = app.dhcp_server.handle(dm.build_request(mac, dm.DISCOVER))
+            ack = app.dhcp_server.handle(dm.build_request(
+                mac, dm.REQUEST, requested_ip=offer.yiaddr))
+            assert ack.msg_type == dm.ACK
+            ip = ack.yiaddr
+            # provisioning side effects through every manager
+            assert ip in app.launcher.dp.qos_egress          # QoS
+            assert ip in app.launcher.dp.subnat              # NAT block
+            assert app.nat.get_allocation(ip) is not None
+            assert app.pppoe.session_count() == 0            # up, idle
+            assert app.dhcpv6 is not None
+            # metrics collect runs
+            app.metrics.collect_once(app.launcher, app.dhcp_server)
+            text = app.metrics.render().decode()
+            assert "bng_pool_allocated" in text
+        finally:
+            app.stop()
+
+    def test_reverse_order_cleanup_idempotent(self):
+        app, _ = make_app(["--nat-enable", "--nat-public-ip",
+                           "203.0.113.1"])
+        app.stop()
+        app.stop()    # second stop is a no-op
+
+
+class TestYAMLMerge:
+    def test_file_values_only_fill_unset_flags(self, tmp_path):
+        cfg = tmp_path / "bng.yaml"
+        cfg.write_text("pool-network: 10.9.0.0/24\n"
+                       "lease-time: 60\n"
+                       "node-id: from-file\n")
+        argv = ["run", "--config", str(cfg), "--node-id", "from-cli"]
+        args = build_parser().parse_args(argv)
+        args = load_yaml_over_args(args, None, argv)
+        assert args.pool_network == "10.9.0.0/24"   # filled from file
+        assert args.lease_time == 60
+        assert args.node_id == "from-cli"           # CLI wins
+
+
+class TestDemo:
+    def test_demo_lifecycle(self, capsys):
+        args = build_parser().parse_args(["demo", "--subscribers", "2"])
+        assert cmd_demo(args) == 0
+        out = capsys.readouterr().out
+        assert "ONT discovered" in out
+        assert "walled garden" in out
+        assert "RADIUS-time allocation" in out
+        assert out.count("activated") == 2
+
+
+class TestVersion:
+    def test_version_command(self, capsys):
+        from bng_amd.cli.main import main
+        assert main(["version"]) == 0
+        assert "bng" in capsys.readouterr().out
