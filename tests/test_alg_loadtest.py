@@ -1,0 +1,129 @@
+"""NAT ALG + load-harness tests (ref pkg/nat/alg_test patterns and
+test/load/dhcp_benchmark_test.go:15-60 self-test with simulated
+fast/slow latency)."""
+import time
+
+import pytest
+
+from bng_amd.dataplane.packets import ip2u32, u32_to_ip
+from bng_amd.nat.alg import ALGProcessor, FTPAlg, SIPAlg
+from bng_amd.nat.manager import Manager as NATManager
+from bng_amd.utils.loadtest import (DHCPLoadTester, Result, Targets)
+
+PRIV = ip2u32("10.0.1.50")
+PUB = ip2u32("203.0.113.1")
+
+
+class TestFTPAlg:
+    def test_port_rewrite_and_pinhole(self):
+        ports = iter([2000, 2001])
+        alg = FTPAlg(lambda ip: next(ports))
+        payload = b"PORT 10,0,1,50,19,137\r\n"     # 19*256+137 = 5001
+        out, holes = alg.process_outbound(payload, PRIV, PUB)
+        assert out == b"PORT 203,0,113,1,7,208\r\n"  # 2000 = 7*256+208
+        assert len(holes) == 1
+        h = holes[0]
+        assert (h.public_ip, h.public_port) == (PUB, 2000)
+        assert (h.private_ip, h.private_port) == (PRIV, 5001)
+
+    def test_eprt_rewrite(self):
+        alg = FTPAlg(lambda ip: 3000)
+        out, holes = alg.process_outbound(
+            b"EPRT |1|10.0.1.50|5002|\r\n", PRIV, PUB)
+        assert out == b"EPRT |1|203.0.113.1|3000|\r\n"
+        assert holes[0].private_port == 5002
+
+    def test_pasv_parse(self):
+        alg = FTPAlg(lambda ip: 0)
+        got = alg.process_inbound(
+            b"227 Entering Passive Mode (93,184,216,34,195,80).\r\n")
+        assert got == (ip2u32("93.184.216.34"), 195 * 256 + 80)
+        assert alg.process_inbound(b"230 Login successful.\r\n") is None
+
+
+class TestSIPAlg:
+    def test_sdp_rewrite_rtp_parity(self):
+        ports = iter([4000, 4002])
+        alg = SIPAlg(lambda ip: next(ports))
+        sdp = (b"INVITE sip:bob@example.com SIP/2.0\r\n"
+               b"Via: SIP/2.0/UDP 10.0.1.50:5060\r\n"
+               b"Contact: <sip:alice@10.0.1.50>\r\n\r\n"
+               b"v=0\r\n"
+               b"c=IN IP4 10.0.1.50\r\n"
+               b"m=audio 16384 RTP/AVP 0\r\n")
+        out, holes = alg.process(sdp, PRIV, PUB)
+        assert b"c=IN IP4 203.0.113.1" in out
+        assert b"m=audio 4000 RTP/AVP 0" in out
+        assert b"10.0.1.50" not in out            # Via/Contact rewritten
+        # RTP even + RTCP odd pinholes
+        assert [(h.public_port, h.private_port) for h in holes] == \
+            [(4000, 16384), (4001, 16385)]
+
+
+class TestALGProcessor:
+    def test_port_allocation_within_block(self):
+        nat = NATManager()
+        nat.add_public_ip("203.0.113.1")
+        nat.allocate_nat(PRIV)
+        proc = ALGProcessor(nat)
+        ports = {proc._alloc_port(PRIV) for _ in range(10)}
+        alloc = nat.get_allocation(PRIV)
+        assert all(alloc.port_start <= p <= alloc.port_end for p in ports)
+        rtp = proc._alloc_rtp_pair(PRIV)
+        assert rtp % 2 == 0
+
+
+class TestLoadHarness:
+    def make_handler(self, fast_ratio=0.97, fast_s=0.00002, slow_s=0.003):
+        """Mock DHCP server with simulated fast/slow latency
+        (ref dhcp_benchmark_test.go:15-60)."""
+        import random
+        rng = random.Random(3)
+
+        def handler(mac, renew):
+            time.sleep(fast_s if rng.random() < fast_ratio else slow_s)
+            return True
+        return handler
+
+    def test_percentiles_and_split(self):
+        t = DHCPLoadTester(self.make_handler(), unique_macs=100,
+                           concurrency=8, warmup=10)
+        res = t.run(400)
+        assert res.total == 400 and res.errors == 0
+        assert res.p50 < 0.001 < res.p99 * 10
+        assert res.hit_rate > 0.80
+        rep = res.report()
+        assert rep["p50_us"] < rep["p99_us"]
+
+    def test_target_validation(self):
+        r = Result(total=1000, duration_s=1.0,
+                   latencies_s=[0.00005] * 990 + [0.005] * 10)
+        t = Targets(min_rps=500)
+        assert r.meets_targets(t) == []
+        assert r.meets_fastpath_target(t)
+        # violated rps
+        t2 = Targets(min_rps=10_000)
+        assert any("rps" in v for v in r.meets_targets(t2))
+        # violated hit rate
+        r2 = Result(total=10, duration_s=1.0, latencies_s=[0.01] * 10)
+        assert any("hit rate" in v for v in r2.meets_targets(Targets(
+            min_rps=1)))
+
+    def test_real_slowpath_server_throughput(self):
+        """In-process slow path only — the CPU side of config 1."""
+        from bng_amd.dhcp import message as dm
+        from bng_amd.dhcp.pool import PoolConfig, PoolManager
+        from bng_amd.dhcp.server import DHCPServer
+        pm = PoolManager()
+        pm.add_pool(PoolConfig(1, "10.0.0.0/16", gateway="10.0.0.1"))
+        srv = DHCPServer(pm, "10.0.0.1")
+
+        def handler(mac, renew):
+            mt = dm.REQUEST if renew else dm.DISCOVER
+            return srv.handle(dm.build_request(mac, mt)) is not None
+
+        t = DHCPLoadTester(handler, unique_macs=500, concurrency=4,
+                           warmup=50)
+        res = t.run(2000)
+        assert res.errors == 0
+        assert res.rps > 1000          # pure-python slow path
