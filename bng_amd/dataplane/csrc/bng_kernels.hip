@@ -26,6 +26,17 @@
 
 /* ==================================================== table primitives */
 
+typedef __attribute__((address_space(1))) uint64_t gu64_t;
+
+BNG_DEV uint64_t rlx_load64(const void* ptr) {
+  return __hip_atomic_load((const uint64_t*)ptr, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+}
+BNG_DEV void rlx_store64(void* ptr, uint64_t v) {
+  __hip_atomic_store((uint64_t*)ptr, v, __ATOMIC_RELAXED,
+                     __HIP_MEMORY_SCOPE_AGENT);
+}
+
 BNG_DEV const bng_sub_entry* sub_lookup(const bng_sub_entry* t, uint32_t mask,
                                         uint64_t key) {
   uint32_t slot = (uint32_t)bng_mix64(key) & mask;
@@ -47,21 +58,25 @@ BNG_DEV bng_subnat_entry* subnat_lookup(bng_subnat_entry* t, uint32_t mask,
   uint32_t slot = (uint32_t)bng_mix64(ip) & mask;
   for (int i = 0; i < BNG_MAX_PROBE; ++i) {
     bng_subnat_entry* e = &t[(slot + i) & mask];
-    uint32_t k = e->key_ip;
-    if (k == ip) return e;
-    if (k == 0) return nullptr;
+    uint4 v = *(const uint4*)e;    /* key_ip + sid + public_ip + ports */
+    if (v.x == ip) return e;
+    if (v.x == 0) return nullptr;
   }
   return nullptr;
 }
 
 BNG_DEV bng_qos_bucket* qos_lookup(bng_qos_bucket* t, uint32_t mask,
-                                   uint32_t ip) {
+                                   uint32_t ip, uint64_t* rate_out) {
   if (ip == 0) return nullptr;
   uint32_t slot = (uint32_t)bng_mix64(ip) & mask;
   for (int i = 0; i < BNG_MAX_PROBE; ++i) {
     bng_qos_bucket* e = &t[(slot + i) & mask];
-    if (e->key_ip == ip && e->valid) return e;
-    if (e->key_ip == 0) return nullptr;
+    uint4 v = *(const uint4*)e;    /* key_ip+valid+prio | rate in 16B */
+    if (v.x == ip && (v.y & 0xFF)) {
+      *rate_out = ((uint64_t)v.w << 32) | v.z;
+      return e;
+    }
+    if (v.x == 0) return nullptr;
   }
   return nullptr;
 }
@@ -72,7 +87,8 @@ BNG_DEV const bng_binding_entry* binding_lookup(const bng_binding_entry* t,
   uint32_t slot = (uint32_t)bng_mix64(mac) & mask;
   for (int i = 0; i < BNG_MAX_PROBE; ++i) {
     const bng_binding_entry* e = &t[(slot + i) & mask];
-    uint64_t k = e->key_mac;
+    uint4 v = *(const uint4*)e;    /* key + ipv4 + flags in one load */
+    uint64_t k = ((uint64_t)v.y << 32) | v.x;
     if (k == mac) return e;
     if (k == BNG_KEY_EMPTY) return nullptr;
   }
@@ -142,10 +158,55 @@ struct pktctx {
   bool l4_ok;
 };
 
+/* Vectorized parse fast path: untagged IPv4 with ihl=5 (the 64B-mix hot
+ * case).  Three dwordx4 loads replace ~30 dependent byte loads; fields
+ * are extracted from registers with constant shifts.  Returns false to
+ * fall back to the general byte parser (VLAN, options, short frames). */
+BNG_DEV bool parse_pkt_fast(pktctx& c, uint8_t* p, int len) {
+  if (len < 48 || ((uintptr_t)p & 15)) return false;
+  uint32_t W[12];
+  #pragma unroll
+  for (int i = 0; i < 3; ++i) {
+    uint4 v = ((const uint4*)p)[i];
+    W[i * 4 + 0] = __builtin_bswap32(v.x);
+    W[i * 4 + 1] = __builtin_bswap32(v.y);
+    W[i * 4 + 2] = __builtin_bswap32(v.z);
+    W[i * 4 + 3] = __builtin_bswap32(v.w);
+  }
+  #define BNG_B(o) ((W[(o) >> 2] >> (24 - 8 * ((o) & 3))) & 0xFFu)
+  #define BNG_H(o) ((BNG_B(o) << 8) | BNG_B((o) + 1))
+  uint32_t ethertype = BNG_H(12);
+  if (ethertype != 0x0800) return false;        /* VLAN etc: slow path */
+  if (BNG_B(14) != 0x45) return false;          /* options/ihl!=5 */
+  c.p = p; c.len = len; c.vlan_offset = 0; c.s_tag = c.c_tag = 0;
+  c.tagged = false; c.l4_ok = false;
+  c.ip_off = 14;
+  c.proto = (uint8_t)BNG_B(23);
+  c.saddr = (BNG_H(26) << 16) | BNG_H(28);
+  c.daddr = (BNG_H(30) << 16) | BNG_H(32);
+  c.l4_off = 34;
+  if (c.proto == 6 && len >= 54) {
+    c.sport = (uint16_t)BNG_H(34);
+    c.dport = (uint16_t)BNG_H(36);
+    c.tcp_flags = (uint8_t)BNG_B(47);
+    c.l4_ok = true;
+  } else if (c.proto == 17 && len >= 42) {
+    c.sport = (uint16_t)BNG_H(34);
+    c.dport = (uint16_t)BNG_H(36);
+    c.l4_ok = true;
+  } else if (c.proto == 1 && len >= 42) {
+    c.icmp_id = (uint16_t)BNG_H(38);
+    c.l4_ok = true;
+  }
+  #undef BNG_B
+  #undef BNG_H
+  return true;
+}
+
 /* Parse Ethernet [+VLAN/QinQ] + IPv4 + L4 ports.  NAT/QoS/antispoof paths
  * in the reference parse untagged frames only (nat44.c:573-581); the DHCP
  * path handles tags (dhcp_fastpath.c:352-428).  want_vlan selects. */
-BNG_DEV bool parse_pkt(pktctx& c, uint8_t* p, int len, bool want_vlan) {
+BNG_DEV bool parse_pkt_slow(pktctx& c, uint8_t* p, int len, bool want_vlan) {
   c.p = p; c.len = len; c.ip_off = -1; c.vlan_offset = 0;
   c.s_tag = c.c_tag = 0; c.tagged = false; c.l4_ok = false;
   if (len < 14) return false;
@@ -185,6 +246,11 @@ BNG_DEV bool parse_pkt(pktctx& c, uint8_t* p, int len, bool want_vlan) {
     c.l4_ok = true;
   }
   return true;
+}
+
+BNG_DEV bool parse_pkt(pktctx& c, uint8_t* p, int len, bool want_vlan) {
+  if (parse_pkt_fast(c, p, len)) return true;
+  return parse_pkt_slow(c, p, len, want_vlan);
 }
 
 /* ================================================== DHCP fast path K1 */
@@ -562,16 +628,39 @@ BNG_DEV int nat_egress_process(pktctx& c, const nat_tables& T, nat_flags& F) {
 
   uint32_t nat_ip; uint16_t nat_port;
   if (found) {
-    if (!bng_wait_ready(&sess->ready, 8192)) { F.passed = true; return BNG_PASS; }
-    /* sig collision with a different tuple: fall back to slow path (rare) */
-    bng_nat_tuple k{c.saddr, c.daddr, sport, dport, c.proto, {0, 0, 0}};
-    if (!tuple_eq(sess->key, k)) { F.passed = true; return BNG_PASS; }
-    nat_ip = sess->nat_ip; nat_port = sess->nat_port;
-    __hip_atomic_store(&sess->last_seen, T.now_ns, __ATOMIC_RELAXED,
-                       __HIP_MEMORY_SCOPE_AGENT);
-    atomicAdd((unsigned long long*)&sess->packets_out, 1ull);
-    atomicAdd((unsigned long long*)&sess->bytes_out,
-              (unsigned long long)c.len);
+    /* HIT fast path: the producer plain-stored fields then issued an
+     * agent release (L2 write-back) before the sc1 `ready` store, so
+     * relaxed agent (sc1, L2-served) loads are always fresh — no
+     * acquire fence (which would flush this CU's whole L1) needed.
+     * 4 independent u64 loads cover key + translation + ready. */
+    const uint64_t* e64 = (const uint64_t*)sess;
+    uint64_t k0 = rlx_load64(e64 + 1);   /* src_ip | dst_ip<<32 */
+    uint64_t k1 = rlx_load64(e64 + 2);   /* sport | dport<<16 | proto<<32 */
+    uint64_t t0 = rlx_load64(e64 + 3);   /* nat_ip | nat_port<<32 | orig_port<<48 */
+    uint64_t t1 = rlx_load64(e64 + 4);   /* orig_ip | state<<32 | hairpin<<40 | ready<<48 */
+    if (!((t1 >> 48) & 0xFF)) {          /* creation still in flight */
+      if (!bng_wait_ready(&sess->ready, 8192)) {
+        F.passed = true; return BNG_PASS;
+      }
+      k0 = rlx_load64(e64 + 1); k1 = rlx_load64(e64 + 2);
+      t0 = rlx_load64(e64 + 3); t1 = rlx_load64(e64 + 4);
+    }
+    uint64_t ek0 = ((uint64_t)c.daddr << 32) | c.saddr;
+    uint64_t ek1 = (uint64_t)sport | ((uint64_t)dport << 16) |
+                   ((uint64_t)c.proto << 32);
+    if (k0 != ek0 || (k1 & 0xFFFFFFFFFFull) != ek1) {
+      /* sig collision with a different tuple: slow path (rare) */
+      F.passed = true; return BNG_PASS;
+    }
+    nat_ip = (uint32_t)t0;
+    nat_port = (uint16_t)(t0 >> 32);
+    /* one fabric store per batch per session, not per packet */
+    if (rlx_load64(e64 + 5) != T.now_ns)
+      rlx_store64((void*)(e64 + 5), T.now_ns);
+    /* one packed accounting atomic: bytes<<24 | packets (folded into the
+     * u64 counters by the sweep kernel) */
+    atomicAdd((unsigned long long*)(e64 + 11),
+              ((unsigned long long)c.len << 24) | 1ull);
   } else {
     /* we claimed the slot: allocate mapping (EIM or fresh port) */
     bool parity = (cfg->flags & BNG_NAT_FLAG_PARITY) != 0;
@@ -710,28 +799,48 @@ BNG_DEV int nat_ingress_process(pktctx& c, const nat_tables& T, nat_flags& F) {
 
   uint64_t rsig = bng_tuple_sig(c.saddr, c.daddr, sport, dport, c.proto);
   bng_nat_reverse* rev = sig_lookup(T.reverse, T.rev_mask, rsig);
-  if (!rev || !bng_wait_ready(&rev->ready, 8192)) {
-    F.passed = true; return BNG_FWD;
+  if (!rev) { F.passed = true; return BNG_FWD; }
+  {
+    const uint64_t* r64 = (const uint64_t*)rev;
+    uint64_t rdy = rlx_load64(r64 + 5);      /* ready at byte 40 */
+    if (!(rdy & 0xFF) && !bng_wait_ready(&rev->ready, 8192)) {
+      F.passed = true; return BNG_FWD;
+    }
+    uint64_t k0 = rlx_load64(r64 + 1), k1 = rlx_load64(r64 + 2);
+    uint64_t ek0 = ((uint64_t)c.daddr << 32) | c.saddr;
+    uint64_t ek1 = (uint64_t)sport | ((uint64_t)dport << 16) |
+                   ((uint64_t)c.proto << 32);
+    if (k0 != ek0 || (k1 & 0xFFFFFFFFFFull) != ek1) {
+      F.passed = true; return BNG_FWD;
+    }
   }
-  bng_nat_tuple rk{c.saddr, c.daddr, sport, dport, c.proto, {0, 0, 0}};
-  if (!tuple_eq(rev->key, rk)) { F.passed = true; return BNG_FWD; }
-
-  uint64_t osig = bng_tuple_sig(rev->orig.src_ip, rev->orig.dst_ip,
-                                rev->orig.src_port, rev->orig.dst_port,
-                                rev->orig.protocol);
+  bng_nat_tuple orig;
+  {
+    const uint64_t* r64 = (const uint64_t*)rev;
+    uint64_t o0 = rlx_load64(r64 + 3), o1 = rlx_load64(r64 + 4);
+    orig.src_ip = (uint32_t)o0; orig.dst_ip = (uint32_t)(o0 >> 32);
+    orig.src_port = (uint16_t)o1; orig.dst_port = (uint16_t)(o1 >> 16);
+    orig.protocol = (uint8_t)(o1 >> 32);
+  }
+  uint64_t osig = bng_tuple_sig(orig.src_ip, orig.dst_ip, orig.src_port,
+                                orig.dst_port, orig.protocol);
   bng_nat_session* sess = sig_lookup(T.sessions, T.sess_mask, osig);
   if (!sess || !bng_wait_ready(&sess->ready, 8192) ||
-      !tuple_eq(sess->key, rev->orig)) {
+      !tuple_eq(sess->key, orig)) {
     /* session expired: clean the reverse entry (ref :871-875) */
     __hip_atomic_store(&rev->sig, BNG_KEY_TOMBSTONE, __ATOMIC_RELAXED,
                        __HIP_MEMORY_SCOPE_AGENT);
     F.sess_expired = true;
     return BNG_FWD;
   }
-  __hip_atomic_store(&sess->last_seen, T.now_ns, __ATOMIC_RELAXED,
-                     __HIP_MEMORY_SCOPE_AGENT);
-  atomicAdd((unsigned long long*)&sess->packets_in, 1ull);
-  atomicAdd((unsigned long long*)&sess->bytes_in, (unsigned long long)c.len);
+  (void)0;
+  {
+    const uint64_t* e64 = (const uint64_t*)sess;
+    if (rlx_load64(e64 + 5) != T.now_ns)
+      rlx_store64((void*)((uint64_t*)sess + 5), T.now_ns);
+    atomicAdd((unsigned long long*)((uint64_t*)sess + 12),
+              ((unsigned long long)c.len << 24) | 1ull);
+  }
   if (c.proto == 6) {
     if (c.tcp_flags & 0x05)
       sess->state = BNG_NAT_CLOSING;
@@ -820,36 +929,14 @@ __global__ void nat44_kernel(
 
 struct qos_flags { bool passed, dropped; uint32_t bytes; };
 
-/* Token-bucket check (ref token_bucket_check qos_ratelimit.c:70-104).
- * Refill: the thread that CASes last_update old->now does one capped
- * CAS-loop add; consume: one atomicAdd(-len) with undo-on-negative. */
-BNG_DEV bool qos_tb_check(bng_qos_bucket* tb, uint32_t pkt_len,
-                          uint64_t now_ns) {
-  uint64_t rate = tb->rate_bps;
-  if (rate == 0) return true;
-  uint64_t last = __hip_atomic_load(&tb->last_update, __ATOMIC_RELAXED,
-                                    __HIP_MEMORY_SCOPE_AGENT);
-  if (last != now_ns) {
-    if (__hip_atomic_compare_exchange_strong(
-            &tb->last_update, &last, now_ns, __ATOMIC_RELAXED,
-            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)) {
-      uint64_t add = ((now_ns - last) * (rate / 8)) / 1000000000ull;
-      int64_t burst = (int64_t)tb->burst_bytes;
-      /* capped add, bounded CAS loop */
-      for (int t = 0; t < 16; ++t) {
-        int64_t cur = (int64_t)__hip_atomic_load(
-            (uint64_t*)&tb->tokens, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-        int64_t want = cur + (int64_t)add;
-        if (want > burst) want = burst;
-        if (want == cur) break;
-        uint64_t expect = (uint64_t)cur;
-        if (__hip_atomic_compare_exchange_strong(
-                (uint64_t*)&tb->tokens, &expect, (uint64_t)want,
-                __ATOMIC_RELAXED, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT))
-          break;
-      }
-    }
-  }
+/* Token-bucket check (ref token_bucket_check qos_ratelimit.c:70-104),
+ * consume-first form: one atomicAdd(-len) when tokens suffice (the
+ * common case); only an INSUFFICIENT balance triggers the refill
+ * (elapsed-credit since last_update, capped at burst, refill winner by
+ * CAS on last_update) and one retry.  Verdict-equivalent to the
+ * reference's refill-then-consume: unclaimed credit stays recoverable
+ * because last_update only advances when a refill actually runs. */
+BNG_DEV bool qos_consume(bng_qos_bucket* tb, uint32_t pkt_len) {
   long long old = (long long)atomicAdd(
       (unsigned long long*)&tb->tokens,
       (unsigned long long)(-(int64_t)pkt_len));
@@ -858,14 +945,49 @@ BNG_DEV bool qos_tb_check(bng_qos_bucket* tb, uint32_t pkt_len,
   return false;
 }
 
+BNG_DEV bool qos_tb_check(bng_qos_bucket* tb, uint32_t pkt_len,
+                          uint64_t now_ns, uint64_t rate) {
+  if (rate == 0) return true;
+  if (qos_consume(tb, pkt_len)) return true;
+  /* insufficient: claim the refill, credit elapsed time, retry once */
+  uint64_t last = __hip_atomic_load(&tb->last_update, __ATOMIC_RELAXED,
+                                    __HIP_MEMORY_SCOPE_AGENT);
+  if (last != now_ns &&
+      __hip_atomic_compare_exchange_strong(
+          &tb->last_update, &last, now_ns, __ATOMIC_RELAXED,
+          __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)) {
+    uint64_t elapsed = now_ns - last;
+    int64_t burst = (int64_t)tb->burst_bytes;
+    uint64_t add = (elapsed > 100000000000ull) ? (uint64_t)burst
+        : (elapsed * (rate / 8)) / 1000000000ull;
+    /* capped add, bounded CAS loop */
+    for (int t = 0; t < 16; ++t) {
+      int64_t cur = (int64_t)__hip_atomic_load(
+          (uint64_t*)&tb->tokens, __ATOMIC_RELAXED,
+          __HIP_MEMORY_SCOPE_AGENT);
+      int64_t want = cur + (int64_t)add;
+      if (want > burst) want = burst;
+      if (want == cur) break;
+      uint64_t expect = (uint64_t)cur;
+      if (__hip_atomic_compare_exchange_strong(
+              (uint64_t*)&tb->tokens, &expect, (uint64_t)want,
+              __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+              __HIP_MEMORY_SCOPE_AGENT))
+        break;
+    }
+  }
+  return qos_consume(tb, pkt_len);
+}
+
 BNG_DEV int qos_process(pktctx& c, bng_qos_bucket* table, uint32_t mask,
                         bool egress, uint64_t now_ns, qos_flags& F) {
   if (c.ip_off < 0) return BNG_FWD;
   uint32_t key = egress ? c.daddr : c.saddr;
   /* table keys are stored as the BE byte pattern read as host int */
-  bng_qos_bucket* tb = qos_lookup(table, mask, key);
+  uint64_t rate = 0;
+  bng_qos_bucket* tb = qos_lookup(table, mask, key, &rate);
   if (!tb) return BNG_FWD;
-  bool ok = qos_tb_check(tb, c.len, now_ns);
+  bool ok = qos_tb_check(tb, c.len, now_ns, rate);
   F.bytes = c.len;
   if (ok) { F.passed = true; return BNG_FWD; }
   F.dropped = true;
@@ -1295,6 +1417,22 @@ __global__ void nat_sweep_kernel(bng_nat_session* sessions, uint32_t n_slots,
     uint64_t sig = s->sig;
     if (sig == BNG_KEY_EMPTY || sig == BNG_KEY_TOMBSTONE || !s->ready)
       continue;
+    /* fold the hot-path packed accounting deltas ({bytes:40,pkts:24} in
+     * _pad2[0/1]) into the u64 counters */
+    {
+      unsigned long long po = atomicExch(
+          (unsigned long long*)&s->_pad2[0], 0ull);
+      unsigned long long pi = atomicExch(
+          (unsigned long long*)&s->_pad2[1], 0ull);
+      if (po) {
+        s->packets_out += po & 0xFFFFFFull;
+        s->bytes_out += po >> 24;
+      }
+      if (pi) {
+        s->packets_in += pi & 0xFFFFFFull;
+        s->bytes_in += pi >> 24;
+      }
+    }
     uint64_t to = udp_to;
     if (s->key.protocol == 6)
       to = (s->state == BNG_NAT_ESTABLISHED) ? tcp_est_to : tcp_tr_to;
