@@ -236,13 +236,23 @@ class BNG:
                                listen_port=a.ha_listen_port,
                                partner_url=a.ha_partner_url).start()
             self._defer(self.ha.stop)
+            from ..ha import session_glue
+            session_glue.attach(self.dhcp_server, self.ha)
             if a.ha_partner_url:
                 self.ha_monitor = HealthMonitor(a.ha_partner_url).start()
                 self._defer(self.ha_monitor.stop)
+
+                def _role_change(r):
+                    if r == "active":
+                        session_glue.promote(
+                            self.dhcp_server, self.ha,
+                            getattr(self, "qos", None),
+                            getattr(self, "nat", None))
+                    else:
+                        self.ha.demote()
                 self.ha_failover = FailoverController(
                     a.node_id, a.ha_role, monitor=self.ha_monitor,
-                    role_change_callback=lambda r: self.ha.promote()
-                    if r == "active" else self.ha.demote())
+                    role_change_callback=_role_change)
 
         # 8. routing (main.go:901-939)
         if a.bgp_enable:

@@ -30,6 +30,7 @@ void bng_launch_antispoof(void*, const void*, void*, int, int, const void*,
                           uint32_t, const void*, void*, void*, void*,
                           uint64_t, hipStream_t);
 void bng_launch_uplink(bng_uplink_params*, hipStream_t);
+void bng_launch_downlink(bng_uplink_params*, hipStream_t);
 void bng_launch_pkt_class(const void*, const void*, void*, int, int,
                           hipStream_t);
 void bng_launch_sub_upsert(void*, uint32_t, const void*, int, void*,
@@ -143,7 +144,8 @@ void uplink_pipeline(torch::Tensor data, torch::Tensor in_len,
                      torch::Tensor log_hdr, torch::Tensor qos_in,
                      torch::Tensor qos_stats, int64_t now_ns,
                      int64_t now_sec,
-                     c10::optional<torch::Tensor> order) {
+                     c10::optional<torch::Tensor> order,
+                     bool downlink) {
   check_dev(data, "data");
   bng_uplink_params P{};
   P.data = (uint8_t*)data.data_ptr();
@@ -185,7 +187,10 @@ void uplink_pipeline(torch::Tensor data, torch::Tensor in_len,
   P.qos_stats = (unsigned long long*)qos_stats.data_ptr();
   P.now_ns = (uint64_t)now_ns;
   P.now_sec = (uint64_t)now_sec;
-  bng_launch_uplink(&P, cur_stream());
+  if (downlink)
+    bng_launch_downlink(&P, cur_stream());
+  else
+    bng_launch_uplink(&P, cur_stream());
 }
 
 void pkt_class(torch::Tensor data, torch::Tensor in_len,
@@ -308,7 +313,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("hairpin"), py::arg("n_hairpin"), py::arg("nat_stats"),
         py::arg("log_ring"), py::arg("log_hdr"), py::arg("qos_in"),
         py::arg("qos_stats"), py::arg("now_ns"), py::arg("now_sec"),
-        py::arg("order") = py::none());
+        py::arg("order") = py::none(),
+        py::arg("downlink") = false);
   m.def("pkt_class", &pkt_class);
   m.def("sub_upsert", &sub_upsert);
   m.def("sub_delete", &sub_delete);

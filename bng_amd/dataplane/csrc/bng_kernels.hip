@@ -1195,6 +1195,49 @@ void uplink_pipeline_kernel(bng_uplink_params P) {
   }
 }
 
+/* fused downlink pipeline: internet -> subscriber return path.
+ * NAT44 DNAT (reverse map -> session -> rewrite) then QoS egress
+ * (download shaping keyed by the post-DNAT destination = the
+ * subscriber's IP) in one pass — the reference's TC-ingress nat44 +
+ * TC-egress qos hook chain (nat44.c:805-948, qos_ratelimit.c:126-172). */
+__global__ __launch_bounds__(256, 5)
+void downlink_pipeline_kernel(bng_uplink_params P) {
+  nat_tables NT{P.sessions, P.sess_mask, P.reverse, P.rev_mask, P.eim,
+                P.eim_mask, P.subnat, P.subnat_mask, P.ncfg, P.hairpin_ips,
+                P.n_hairpin, P.nat_stats, P.log_ring, P.log_hdr, P.now_ns};
+  int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  int nthreads = gridDim.x * blockDim.x;
+  for (int base = 0; base < P.n; base += nthreads) {
+    int i = base + tid;
+    nat_flags NF; NF.clear();
+    qos_flags QF{false, false, 0};
+    if (i < P.n) {
+      int pid = P.order ? P.order[i] : i;
+      uint8_t* p = P.data + (size_t)pid * P.stride;
+      int len = P.in_len[pid];
+      pktctx c;
+      bool ip_ok = parse_pkt(c, p, len, false);
+      int v = BNG_FWD;
+      if (ip_ok) {
+        v = nat_ingress_process(c, NT, NF);
+        if (v == BNG_FWD && c.ip_off >= 0) {
+          /* re-read the (possibly rewritten) destination for QoS */
+          c.daddr = ld_u32be(p + c.ip_off + 16);
+          v = qos_process(c, P.qos_in, P.qos_mask, /*egress=*/true,
+                          P.now_ns, QF);
+        }
+      }
+      P.verdict[pid] = (uint8_t)v;
+      P.out_len[pid] = (uint16_t)len;
+    }
+    nat_commit_stats(NF, P.nat_stats);
+    stat_inc(&P.qos_stats[BNG_QS_PKT_PASSED], QF.passed);
+    stat_inc(&P.qos_stats[BNG_QS_PKT_DROPPED], QF.dropped);
+    stat_add(&P.qos_stats[BNG_QS_BYTES_PASSED], QF.bytes, QF.passed);
+    stat_add(&P.qos_stats[BNG_QS_BYTES_DROPPED], QF.bytes, QF.dropped);
+  }
+}
+
 /* classify packets for type-sorting: 1 = DHCP (UDP dst 67), 0 = other */
 __global__ void pkt_class_kernel(const uint8_t* __restrict__ data,
                                  const uint16_t* __restrict__ in_len,
@@ -1555,6 +1598,11 @@ void bng_launch_antispoof(void* data, const void* in_len, void* verdict,
 void bng_launch_uplink(bng_uplink_params* P, hipStream_t s) {
   hipLaunchKernelGGL(uplink_pipeline_kernel, dim3(pkt_grid(P->n)), dim3(256),
                      0, s, *P);
+}
+
+void bng_launch_downlink(bng_uplink_params* P, hipStream_t s) {
+  hipLaunchKernelGGL(downlink_pipeline_kernel, dim3(pkt_grid(P->n)),
+                     dim3(256), 0, s, *P);
 }
 
 void bng_launch_pkt_class(const void* data, const void* in_len, void* cls,

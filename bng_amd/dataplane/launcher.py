@@ -413,6 +413,61 @@ class HipLauncher:
             now_sec if now_sec is not None else now // 10**9, order=order)
         return verdict, out_len
 
+    def downlink(self, data, lens, now_ns: Optional[int] = None):
+        """Fused NAT44 DNAT -> QoS-egress return path."""
+        n = lens.numel()
+        verdict, out_len = self._outs(n)
+        now = now_ns if now_ns is not None else time.time_ns()
+        self.ext.uplink_pipeline(
+            data, lens, out_len, verdict, self.subs, self.pools,
+            self.server_cfg, self.dhcp_stats, self.bindings, self.as_cfg,
+            self.as_stats, self.spoof_ring, self.spoof_hdr, self.sessions,
+            self.reverse, self.eim, self.subnat, self.nat_cfg, self.hairpin,
+            self.n_hairpin, self.nat_stats, self.nat_log_ring,
+            self.nat_log_hdr, self.qos_egress, self.qos_stats, now,
+            now // 10**9, order=None, downlink=True)
+        return verdict
+
+    # ------------------------------------------- HA table snapshotting
+    def export_subscribers(self) -> List[dict]:
+        """Download the subscriber fast-path table for HA sync (the GPU
+        analog of the reference's session-store snapshot feeding
+        HASyncer full syncs)."""
+        import numpy as np
+        raw = self.subs.cpu().numpy()
+        arr = raw.view([("key", "<u8"), ("pool_id", "<u4"),
+                        ("ip", "<u4"), ("lease", "<u8"),
+                        ("vlan", "<u2"), ("cc", "u1"), ("fl", "u1"),
+                        ("pad", "<u4")])
+        live = arr[(arr["key"] != 0) &
+                   (arr["key"] != 0xFFFFFFFFFFFFFFFF)]
+        return [{"key": int(e["key"]), "pool_id": int(e["pool_id"]),
+                 "ip": int(e["ip"]), "lease_expiry": int(e["lease"]),
+                 "vlan_id": int(e["vlan"]),
+                 "client_class": int(e["cc"]), "flags": int(e["fl"])}
+                for e in live]
+
+    def import_subscribers(self, entries: Sequence[dict]) -> int:
+        """Bulk-load a snapshot (standby promotion: shadow state becomes
+        the live fast path)."""
+        import numpy as np
+        if not entries:
+            return 0
+        n = len(entries)
+        arr = np.zeros(n, dtype=[("key", "<u8"), ("pool_id", "<u4"),
+                                 ("ip", "<u4"), ("lease", "<u8"),
+                                 ("vlan", "<u2"), ("cc", "u1"),
+                                 ("fl", "u1"), ("pad", "<u4")])
+        for i, e in enumerate(entries):
+            arr[i] = (e["key"], e.get("pool_id", 0), e.get("ip", 0),
+                      e.get("lease_expiry", 0), e.get("vlan_id", 0),
+                      e.get("client_class", 0), e.get("flags", 0), 0)
+        batch = self.torch.from_numpy(arr.view(np.uint8)).to(self.device)
+        rc = self.torch.zeros(n, dtype=self.torch.int32,
+                              device=self.device)
+        self.ext.sub_upsert(self.subs, batch.flatten(), rc)
+        return n - int((rc != 0).sum().item())
+
     def shard_owner(self, data, lens, n_shards: int):
         n = lens.numel()
         owner = self.torch.zeros(n, dtype=self.torch.int32,
@@ -577,6 +632,22 @@ class GoldenLauncher:
             v, L = self.dp.dhcp_fastpath(fb)
             results.append((v, bytes(fb[:L])))
         return results
+
+    def export_subscribers(self):
+        from . import abi as _abi
+        return [{"key": k, "pool_id": r.pool_id, "ip": r.allocated_ip,
+                 "lease_expiry": int(r.lease_expiry),
+                 "vlan_id": r.vlan_id, "client_class": r.client_class,
+                 "flags": r.flags}
+                for k, r in self.dp.subscribers.items()]
+
+    def import_subscribers(self, entries):
+        for e in entries:
+            self.dp.subscribers[e["key"]] = SubRecord(
+                e.get("pool_id", 0), e.get("ip", 0),
+                e.get("lease_expiry", 0), e.get("vlan_id", 0),
+                e.get("client_class", 0), e.get("flags", 0))
+        return len(entries)
 
     def process_nat44(self, frames, egress=True, now_ns=None):
         if now_ns is not None:

@@ -68,6 +68,7 @@ class DHCPServer:
         self.audit = None             # audit.Logger
         self.metrics = None
         self.auth_mode = "none"       # none|mac (RADIUS auth w/ MAC creds)
+        self.on_lease_event = []      # [(event, Lease)] listeners (HA sync)
         self.stats = {k: 0 for k in (
             "discover", "request", "release", "decline", "inform",
             "offer", "ack", "nak", "auth_reject", "walled_garden",
@@ -200,6 +201,7 @@ class DHCPServer:
         lease.expiry = time.time() + self.lease_time
         self._post_ack(lease, req)
         self.stats["ack"] += 1
+        self._emit_lease("add", lease)
         return self._reply(req, dm.ACK, lease)
 
     # ----------------------------------------------------------- release
@@ -349,7 +351,24 @@ class DHCPServer:
             self.audit.log("session_start", subscriber=lease.subscriber_id,
                            ip=u32_to_ip(lease.ip))
 
+    def _emit_lease(self, event: str, lease: Lease):
+        for cb in self.on_lease_event:
+            try:
+                cb(event, lease)
+            except Exception:
+                pass
+
+    def restore_lease(self, lease: Lease):
+        """HA promotion: install a replicated lease (shadow -> live),
+        including the GPU fast-path entry."""
+        with self._lock:
+            self.leases[lease.mac] = lease
+            if lease.circuit_id:
+                self.leases_by_circuit[lease.circuit_id] = lease
+        self._update_fastpath(lease)
+
     def _teardown(self, lease: Lease):
+        self._emit_lease("delete", lease)
         self._remove_fastpath(lease)
         pool = self.pools.get_pool(lease.pool_id)
         if pool is not None:

@@ -109,12 +109,14 @@ class HASyncer:
                     self.wfile.write(body)
                     return
                 if self.path == "/sync/stream":
+                    # register BEFORE sending headers: once the client
+                    # sees headers, every later delta is guaranteed queued
+                    q: queue.Queue = queue.Queue(maxsize=10000)
+                    syncer._subscribers.append(q)
                     self.send_response(200)
                     self.send_header("Content-Type", "text/event-stream")
                     self.send_header("Cache-Control", "no-cache")
                     self.end_headers()
-                    q: queue.Queue = queue.Queue(maxsize=10000)
-                    syncer._subscribers.append(q)
                     try:
                         while not syncer._stop.is_set():
                             try:
@@ -157,15 +159,22 @@ class HASyncer:
         backoff = self.reconnect_backoff
         while not self._stop.is_set():
             try:
-                self._full_sync()
-                self.connected = True
                 backoff = self.reconnect_backoff
                 with requests.get(f"{self.partner_url}/sync/stream",
                                   stream=True, timeout=(3, 30)) as r:
-                    for line in r.iter_lines():
-                        if self._stop.is_set():
-                            return
-                        if not line or not line.startswith(b"data: "):
+                    # stream established (headers => queue registered);
+                    # full-sync now so no delta can fall in a gap
+                    self._full_sync()
+                    self.connected = True
+                    # readline on the raw stream: iter_lines buffers by
+                    # chunk and would sit on deltas until enough bytes
+                    # arrive; SSE events are newline-framed
+                    while not self._stop.is_set():
+                        line = r.raw.readline()
+                        if not line:
+                            break
+                        line = line.strip()
+                        if not line.startswith(b"data: "):
                             continue
                         msg = SyncMessage.from_dict(
                             json.loads(line[6:].decode()))
@@ -191,6 +200,8 @@ class HASyncer:
     def _apply(self, msg: SyncMessage):
         if msg.type == SYNC_HEARTBEAT:
             return
+        if msg.seq <= self.last_partner_seq:
+            return      # already covered by a full sync (dup, not a gap)
         if self.last_partner_seq >= 0 and msg.seq != self.last_partner_seq + 1:
             self.stats["seq_gaps"] += 1
             try:

@@ -161,3 +161,61 @@ class TestFailover:
         from bng_amd.ha.health_monitor import HealthEvent
         ctl.handle_health_event(HealthEvent("partner_down", "x", 0, 3))
         assert ctl.role == ROLE_ACTIVE and ctl.state == STATE_NORMAL
+
+
+class TestGPUTableSync:
+    def test_lease_replication_and_promotion(self):
+        """Active's DHCP leases replicate to the standby; at promotion the
+        standby rebuilds leases + fast-path table + QoS from the shadow
+        store."""
+        from bng_amd.cli.main import BNG, build_parser
+        from bng_amd.dhcp import message as dm
+        from bng_amd.dataplane.packets import mac_bytes
+        from bng_amd.ha import session_glue
+
+        def mk(role, partner=""):
+            argv = ["run", "--pool-network", "10.0.1.0/24", "--gpu", "off",
+                    "--ha-role", role, "--qos-policy", "gold:100:20",
+                    "--qos-default-policy", "gold"]
+            if partner:
+                argv += ["--ha-partner-url", partner]
+            return BNG(build_parser().parse_args(argv)).start()
+
+        active = mk("active")
+        standby = mk("standby", partner=active.ha.url)
+        try:
+            assert wait_for(lambda: standby.ha.connected)
+            mac = mac_bytes("aa:bb:cc:00:00:07")
+            offer = active.dhcp_server.handle(
+                dm.build_request(mac, dm.DISCOVER))
+            active.dhcp_server.handle(dm.build_request(
+                mac, dm.REQUEST, requested_ip=offer.yiaddr))
+            assert wait_for(lambda: standby.ha.store.count() == 1)
+            # promotion rebuilds live state from the shadow store
+            n = session_glue.promote(standby.dhcp_server, standby.ha,
+                                     standby.qos)
+            assert n == 1
+            assert mac in standby.dhcp_server.leases
+            lease = standby.dhcp_server.leases[mac]
+            assert lease.ip == offer.yiaddr
+            # fast-path table rebuilt on the standby
+            assert len(standby.launcher.dp.subscribers) == 1
+            # release on (old) active propagates deletes
+            active.dhcp_server.handle(dm.build_request(mac, dm.RELEASE))
+            assert active.ha.store.count() == 0
+        finally:
+            standby.stop()
+            active.stop()
+
+    def test_launcher_snapshot_roundtrip(self):
+        from bng_amd.dataplane.launcher import GoldenLauncher
+        a, b = GoldenLauncher(), GoldenLauncher()
+        a.add_subscriber(b"\xaa\xbb\xcc\x00\x00\x01", 1, 0x0A000105,
+                         2_000_000_000)
+        a.add_vlan_subscriber(100, 200, 1, 0x0A000106, 2_000_000_000)
+        snap = a.export_subscribers()
+        assert len(snap) == 2
+        assert b.import_subscribers(snap) == 2
+        assert b.export_subscribers() and \
+            {e["ip"] for e in b.export_subscribers()} == \
+            {0x0A000105, 0x0A000106}
