@@ -19,7 +19,7 @@
 extern "C" {
 void bng_launch_dhcp(void*, const void*, void*, void*, int, int, const void*,
                      uint32_t, const void*, uint32_t, const void*, void*,
-                     uint64_t, hipStream_t);
+                     uint64_t, const void*, hipStream_t);
 void bng_launch_nat44(void*, const void*, void*, int, int, int, void*,
                       uint32_t, void*, uint32_t, void*, uint32_t, void*,
                       uint32_t, const void*, const void*, uint32_t, void*,
@@ -72,7 +72,8 @@ uint32_t table_mask(const torch::Tensor& t, size_t entry, const char* name) {
 void dhcp_fastpath(torch::Tensor data, torch::Tensor in_len,
                    torch::Tensor out_len, torch::Tensor verdict,
                    torch::Tensor subs, torch::Tensor pools,
-                   torch::Tensor cfg, torch::Tensor stats, int64_t now_sec) {
+                   torch::Tensor cfg, torch::Tensor stats, int64_t now_sec,
+                   c10::optional<torch::Tensor> now_buf) {
   check_dev(data, "data"); check_dev(subs, "subs");
   int n = in_len.numel();
   int stride = data.size(1);
@@ -83,6 +84,7 @@ void dhcp_fastpath(torch::Tensor data, torch::Tensor in_len,
                   (uint32_t)(pools.numel() * pools.element_size() /
                              sizeof(bng_ip_pool)),
                   cfg.data_ptr(), stats.data_ptr(), (uint64_t)now_sec,
+                  now_buf.has_value() ? now_buf->data_ptr() : nullptr,
                   cur_stream());
 }
 
@@ -145,9 +147,12 @@ void uplink_pipeline(torch::Tensor data, torch::Tensor in_len,
                      torch::Tensor qos_stats, int64_t now_ns,
                      int64_t now_sec,
                      c10::optional<torch::Tensor> order,
-                     bool downlink) {
+                     bool downlink,
+                     c10::optional<torch::Tensor> now_buf) {
   check_dev(data, "data");
   bng_uplink_params P{};
+  P.now_ptr = now_buf.has_value()
+      ? (const uint64_t*)now_buf->data_ptr() : nullptr;
   P.data = (uint8_t*)data.data_ptr();
   P.in_len = (const uint16_t*)in_len.data_ptr();
   P.out_len = (uint16_t*)out_len.data_ptr();
@@ -299,7 +304,11 @@ py::dict layout_report() {
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("dhcp_fastpath", &dhcp_fastpath);
+  m.def("dhcp_fastpath", &dhcp_fastpath,
+        py::arg("data"), py::arg("in_len"), py::arg("out_len"),
+        py::arg("verdict"), py::arg("subs"), py::arg("pools"),
+        py::arg("cfg"), py::arg("stats"), py::arg("now_sec"),
+        py::arg("now_buf") = py::none());
   m.def("nat44", &nat44);
   m.def("qos", &qos);
   m.def("antispoof", &antispoof);
@@ -314,7 +323,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("log_ring"), py::arg("log_hdr"), py::arg("qos_in"),
         py::arg("qos_stats"), py::arg("now_ns"), py::arg("now_sec"),
         py::arg("order") = py::none(),
-        py::arg("downlink") = false);
+        py::arg("downlink") = false,
+        py::arg("now_buf") = py::none());
   m.def("pkt_class", &pkt_class);
   m.def("sub_upsert", &sub_upsert);
   m.def("sub_delete", &sub_delete);

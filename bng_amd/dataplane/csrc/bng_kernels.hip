@@ -456,7 +456,9 @@ __global__ void dhcp_fastpath_kernel(
     const bng_sub_entry* __restrict__ subs, uint32_t sub_mask,
     const bng_ip_pool* __restrict__ pools, uint32_t n_pools,
     const bng_server_config* __restrict__ cfg,
-    unsigned long long* __restrict__ stats, uint64_t now_sec) {
+    unsigned long long* __restrict__ stats, uint64_t now_sec,
+    const uint64_t* __restrict__ now_ptr) {
+  if (now_ptr) now_sec = now_ptr[1];
   dhcp_tables T{subs, sub_mask, pools, n_pools, cfg, stats, now_sec};
   int tid = blockIdx.x * blockDim.x + threadIdx.x;
   int nthreads = gridDim.x * blockDim.x;
@@ -1137,6 +1139,7 @@ __global__ void antispoof_kernel(
 __global__ __launch_bounds__(256, 5)  /* cap VGPRs at 96 -> 5 waves/SIMD:
     the pipeline is latency-bound (SQ_WAIT ~95%), occupancy is the lever */
 void uplink_pipeline_kernel(bng_uplink_params P) {
+  if (P.now_ptr) { P.now_ns = P.now_ptr[0]; P.now_sec = P.now_ptr[1]; }
   dhcp_tables DT{P.subs, P.sub_mask, P.pools, P.n_pools, P.scfg,
                  P.dhcp_stats, P.now_sec};
   nat_tables NT{P.sessions, P.sess_mask, P.reverse, P.rev_mask, P.eim,
@@ -1202,6 +1205,7 @@ void uplink_pipeline_kernel(bng_uplink_params P) {
  * TC-egress qos hook chain (nat44.c:805-948, qos_ratelimit.c:126-172). */
 __global__ __launch_bounds__(256, 5)
 void downlink_pipeline_kernel(bng_uplink_params P) {
+  if (P.now_ptr) { P.now_ns = P.now_ptr[0]; P.now_sec = P.now_ptr[1]; }
   nat_tables NT{P.sessions, P.sess_mask, P.reverse, P.rev_mask, P.eim,
                 P.eim_mask, P.subnat, P.subnat_mask, P.ncfg, P.hairpin_ips,
                 P.n_hairpin, P.nat_stats, P.log_ring, P.log_hdr, P.now_ns};
@@ -1547,12 +1551,13 @@ void bng_launch_dhcp(void* data, const void* in_len, void* out_len,
                      void* verdict, int n, int stride,
                      const void* subs, uint32_t sub_mask,
                      const void* pools, uint32_t n_pools, const void* cfg,
-                     void* stats, uint64_t now_sec, hipStream_t s) {
+                     void* stats, uint64_t now_sec, const void* now_ptr,
+                     hipStream_t s) {
   hipLaunchKernelGGL(dhcp_fastpath_kernel, dim3(pkt_grid(n)), dim3(256), 0, s,
       (uint8_t*)data, (const uint16_t*)in_len, (uint16_t*)out_len,
       (uint8_t*)verdict, n, stride, (const bng_sub_entry*)subs, sub_mask,
       (const bng_ip_pool*)pools, n_pools, (const bng_server_config*)cfg,
-      (unsigned long long*)stats, now_sec);
+      (unsigned long long*)stats, now_sec, (const uint64_t*)now_ptr);
 }
 
 void bng_launch_nat44(void* data, const void* in_len, void* verdict, int n,

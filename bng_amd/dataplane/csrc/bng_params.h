@@ -46,7 +46,10 @@ typedef struct bng_uplink_params {
   /* qos ingress */
   bng_qos_bucket* qos_in; uint32_t qos_mask;
   unsigned long long* qos_stats;
-  /* time */
+  /* time: now_ptr (device {now_ns, now_sec}) overrides the scalars when
+   * non-NULL — required under hipGraph replay, where kernel args are
+   * frozen at capture but batch time must advance */
+  const uint64_t* now_ptr;
   uint64_t now_ns;
   uint64_t now_sec;
 } bng_uplink_params;

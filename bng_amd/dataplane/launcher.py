@@ -468,6 +468,19 @@ class HipLauncher:
         self.ext.sub_upsert(self.subs, batch.flatten(), rc)
         return n - int((rc != 0).sum().item())
 
+    # ------------------------------------------- hipGraph steady state
+    def capture_uplink(self, n: int, stride: int = 512,
+                       sort_by_type: bool = True):
+        """Capture the steady-state forwarding sequence
+        [RX-copy -> classify -> type-sort -> uplink] into a hipGraph
+        (SURVEY blueprint: hipGraph-captured steady-state forwarding).
+        Returns a CapturedUplink; batch time advances through a device
+        now-buffer so replays see fresh timestamps."""
+        return CapturedUplink(self, n, stride, sort_by_type)
+
+    def capture_dhcp(self, n: int, stride: int = 512):
+        return CapturedDHCP(self, n, stride)
+
     def shard_owner(self, data, lens, n_shards: int):
         n = lens.numel()
         owner = self.torch.zeros(n, dtype=self.torch.int32,
@@ -673,3 +686,100 @@ def make_launcher(prefer_gpu: bool = True, **kw):
         if has:
             return HipLauncher(**kw)
     return GoldenLauncher(**kw)
+
+
+class CapturedUplink:
+    """hipGraph replay wrapper for the fused uplink pipeline."""
+
+    def __init__(self, launcher: HipLauncher, n: int, stride: int,
+                 sort_by_type: bool):
+        import torch
+        L = self.l = launcher
+        self.n, self.stride = n, stride
+        dev = L.device
+        self.src = torch.zeros((n, stride), dtype=torch.uint8, device=dev)
+        self.work = torch.empty_like(self.src)
+        self.lens = torch.zeros(n, dtype=torch.int16, device=dev)
+        self.verdict = torch.zeros(n, dtype=torch.uint8, device=dev)
+        self.out_len = torch.zeros(n, dtype=torch.int16, device=dev)
+        self.now_buf = torch.zeros(2, dtype=torch.int64, device=dev)
+        self.cls = torch.zeros(n, dtype=torch.uint8, device=dev)
+        self.sort = sort_by_type
+
+        def body():
+            self.work.copy_(self.src)
+            order = None
+            if self.sort:
+                L.ext.pkt_class(self.work, self.lens, self.cls)
+                order = torch.argsort(self.cls, stable=True).to(torch.int32)
+            L.ext.uplink_pipeline(
+                self.work, self.lens, self.out_len, self.verdict, L.subs,
+                L.pools, L.server_cfg, L.dhcp_stats, L.bindings, L.as_cfg,
+                L.as_stats, L.spoof_ring, L.spoof_hdr, L.sessions,
+                L.reverse, L.eim, L.subnat, L.nat_cfg, L.hairpin,
+                L.n_hairpin, L.nat_stats, L.nat_log_ring, L.nat_log_hdr,
+                L.qos_ingress, L.qos_stats, 0, 0, order=order,
+                downlink=False, now_buf=self.now_buf)
+
+        # warmup on a side stream, then capture
+        s = torch.cuda.Stream(device=dev)
+        s.wait_stream(torch.cuda.current_stream(dev))
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                body()
+        torch.cuda.current_stream(dev).wait_stream(s)
+        torch.cuda.synchronize(dev)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            body()
+
+    def run(self, data, lens, now_ns: int):
+        """Copy the batch into the captured buffers, bump time, replay."""
+        import torch
+        self.src.copy_(data)
+        self.lens.copy_(lens)
+        self.now_buf.copy_(torch.tensor(
+            [now_ns, now_ns // 10**9], dtype=torch.int64))
+        self.graph.replay()
+        return self.verdict, self.out_len
+
+
+class CapturedDHCP:
+    """hipGraph replay wrapper for the DHCP fast path (latency path)."""
+
+    def __init__(self, launcher: HipLauncher, n: int, stride: int):
+        import torch
+        L = self.l = launcher
+        dev = L.device
+        self.src = torch.zeros((n, stride), dtype=torch.uint8, device=dev)
+        self.work = torch.empty_like(self.src)
+        self.lens = torch.zeros(n, dtype=torch.int16, device=dev)
+        self.verdict = torch.zeros(n, dtype=torch.uint8, device=dev)
+        self.out_len = torch.zeros(n, dtype=torch.int16, device=dev)
+        self.now_buf = torch.zeros(2, dtype=torch.int64, device=dev)
+
+        def body():
+            self.work.copy_(self.src)
+            L.ext.dhcp_fastpath(self.work, self.lens, self.out_len,
+                                self.verdict, L.subs, L.pools,
+                                L.server_cfg, L.dhcp_stats, 0,
+                                now_buf=self.now_buf)
+
+        s = torch.cuda.Stream(device=dev)
+        s.wait_stream(torch.cuda.current_stream(dev))
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                body()
+        torch.cuda.current_stream(dev).wait_stream(s)
+        torch.cuda.synchronize(dev)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            body()
+
+    def run(self, now_sec: int):
+        """Replay on whatever is in self.src (caller filled it)."""
+        import torch
+        self.now_buf.copy_(torch.tensor(
+            [now_sec * 10**9, now_sec], dtype=torch.int64))
+        self.graph.replay()
+        return self.verdict, self.out_len

@@ -541,3 +541,50 @@ class TestSnapshotGPU:
         f = [build_dhcp_request("aa:bb:cc:00:00:01", DHCP_DISCOVER)]
         d, l = gpu2.make_batch(f)
         assert gpu2.dhcp_fastpath(d, l, NOW_SEC)[0].cpu().tolist() == [TX]
+
+
+class TestHipGraph:
+    def test_captured_dhcp_matches_eager_and_time_advances(self):
+        gpu, cpu = make_pair()
+        for l in (gpu, cpu):
+            l.add_subscriber(mac_bytes("aa:bb:cc:00:00:01"), 1,
+                             ip2u32("10.0.1.50"), NOW_SEC + 100)
+        frames = [build_dhcp_request("aa:bb:cc:00:00:01", DHCP_DISCOVER,
+                                     xid=7)] * 64
+        g = gpu.capture_dhcp(64, 512)
+        d, lns = gpu.make_batch(frames)
+        g.src.copy_(d)
+        g.lens.copy_(lns)
+        v, ol = g.run(NOW_SEC)
+        assert v.cpu().tolist() == [TX] * 64
+        import numpy as np
+        L = int(ol.cpu().numpy().view(np.uint16)[0])
+        reply = bytes(g.work[0].cpu().numpy()[:L])
+        vc, fc = cpu.process_dhcp([frames[0]], now_sec=NOW_SEC)[0]
+        assert reply == fc
+        # REPLAY with time past the lease: same graph, fresh timestamp
+        v2, _ = g.run(NOW_SEC + 1000)
+        assert v2.cpu().tolist() == [PASS] * 64     # lease expired
+        st = gpu.get_stats()
+        assert st["cache_expired"] == 64
+
+    def test_captured_uplink_replay(self):
+        gpu, _ = nat_pair()
+        gpu.set_antispoof_config(default_mode=abi.AS_DISABLED)
+        for k in range(64):
+            gpu.set_qos_policy(ip2u32(f"10.0.1.{k + 1}"), 0, 0,
+                               direction="ingress", now_ns=NOW_NS)
+        frames = [build_ipv4(
+            f"aa:00:00:00:00:{k:02x}", "02:00:00:00:00:01",
+            ip2u32(f"10.0.1.{k + 1}"), ip2u32(DST), proto=17,
+            sport=40000 + k, dport=53) for k in range(64)]
+        g = gpu.capture_uplink(64, 512, sort_by_type=True)
+        d, lns = gpu.make_batch(frames)
+        v, _ = g.run(d, lns, NOW_NS)
+        assert v.cpu().tolist() == [FWD] * 64
+        st = gpu.nat_get_stats()
+        # captured warmup (2 bodies) + replay all SNAT; sessions created once
+        assert st["sessions_created"] == 64
+        v2, _ = g.run(d, lns, NOW_NS + 10**6)
+        assert v2.cpu().tolist() == [FWD] * 64
+        assert gpu.nat_get_stats()["sessions_created"] == 64   # all hits
