@@ -108,7 +108,8 @@ def build_parser() -> argparse.ArgumentParser:
     demo = sub.add_parser("demo", help="simulated subscriber lifecycle")
     demo.add_argument("--subscribers", type=int, default=3)
 
-    sub.add_parser("stats", help="print runtime stats of a config")
+    st = sub.add_parser("stats", help="fetch stats from a running bng")
+    st.add_argument("--metrics-url", default="http://127.0.0.1:9090")
     sub.add_parser("version", help="print version")
     return p
 
@@ -323,12 +324,48 @@ class BNG:
         # 11. PPPoE / DHCPv6 / SLAAC (main.go:1063-1180)
         if a.pppoe_enable:
             from ..pppoe.server import PPPoEServer
-            from ..dataplane.packets import mac_bytes as mb
+            from ..dataplane.packets import mac_bytes as mb, u32_to_ip
             self.pppoe = PPPoEServer(mb(a.server_mac),
                                      ac_name=a.pppoe_ac_name,
                                      auth=a.pppoe_auth)
             if getattr(self, "radius", None):
                 self.pppoe.radius = self.radius
+            if a.pool_network:
+                # PPPoE IPCP addresses from the same pool family
+                from ..allocator.bitmap import BitmapAllocator
+                ppp_alloc = BitmapAllocator(a.pool_network, 32,
+                                            reserve_head=2)
+                self.pppoe.allocator = \
+                    lambda user: ppp_alloc.allocate(user).split("/")[0]
+                self.pppoe.releaser = ppp_alloc.release
+
+            # provision the dataplane for opened PPPoE sessions
+            # (antispoof binding + QoS + NAT + HA), like the DHCP ACK path
+            def _ppp_open(sess):
+                self.antispoof.add_binding(sess.client_mac,
+                                           ipv4=u32_to_ip(sess.ip))
+                self.qos.apply_policy(sess.ip, sess.policy_name)
+                if getattr(self, "nat", None):
+                    self.nat.allocate_nat(sess.ip, sess.username)
+                if getattr(self, "ha", None):
+                    from ..ha.protocol import SessionState
+                    self.ha.publish_add(SessionState(
+                        session_id=f"pppoe-{sess.session_id}",
+                        subscriber_id=sess.username,
+                        mac=sess.client_mac.hex(),
+                        ip=u32_to_ip(sess.ip), access_type="pppoe",
+                        policy_name=sess.policy_name))
+
+            def _ppp_close(sess):
+                self.antispoof.remove_binding(sess.client_mac)
+                self.qos.remove_policy(sess.ip)
+                if getattr(self, "nat", None):
+                    self.nat.release_nat(sess.ip)
+                if getattr(self, "ha", None):
+                    self.ha.publish_delete(f"pppoe-{sess.session_id}")
+
+            self.pppoe.on_session_open = _ppp_open
+            self.pppoe.on_session_close = _ppp_close
         if a.dhcpv6_enable:
             from ..dhcpv6.server import DHCPv6Server
             self.dhcpv6 = DHCPv6Server(na_pool=a.dhcpv6_na_pool,
@@ -466,9 +503,22 @@ def main(argv: Optional[List[str]] = None) -> int:
             app.stop()
         return 0
     if args.command == "stats":
-        print(json.dumps({"error": "connect to a running bng via "
-                          "metrics endpoint"}, indent=2))
-        return 1
+        import requests
+        try:
+            r = requests.get(f"{args.metrics_url}/metrics", timeout=3)
+        except Exception as e:
+            print(json.dumps({"error": str(e)}))
+            return 1
+        out = {}
+        for line in r.text.splitlines():
+            if line.startswith("bng_") and " " in line:
+                k, v = line.rsplit(" ", 1)
+                try:
+                    out[k] = float(v)
+                except ValueError:
+                    pass
+        print(json.dumps(out, indent=2, sort_keys=True))
+        return 0
     parser.print_help()
     return 2
 

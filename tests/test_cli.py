@@ -109,3 +109,51 @@ class TestVersion:
         from bng_amd.cli.main import main
         assert main(["version"]) == 0
         assert "bng" in capsys.readouterr().out
+
+
+class TestPPPoEWiring:
+    def test_pppoe_session_provisions_dataplane(self):
+        """An opened PPPoE session installs antispoof binding + QoS + NAT
+        like a DHCP ACK (the reference provisions both access types)."""
+        app, _ = make_app(["--pppoe-enable", "--pppoe-auth", "none",
+                           "--nat-enable", "--nat-public-ip",
+                           "203.0.113.1", "--qos-policy", "gold:100:20",
+                           "--qos-default-policy", "gold"])
+        try:
+            from tests.test_pppoe import SimClient
+            app.pppoe.local_users = {}
+            cli = SimClient(app.pppoe, username="u1")
+            cli.discover()
+            assert app.pppoe.stats["sessions_open"] == 1
+            sess = list(app.pppoe.sessions.values())[0]
+            assert sess.ip in app.launcher.dp.qos_egress
+            assert sess.ip in app.launcher.dp.subnat
+            from bng_amd.dataplane.abi import mac_to_u64
+            assert mac_to_u64(sess.client_mac) in app.launcher.dp.bindings
+            # teardown cleans up
+            app.pppoe.terminate_session(sess.session_id)
+            assert sess.ip not in app.launcher.dp.qos_egress
+            assert mac_to_u64(sess.client_mac) not in \
+                app.launcher.dp.bindings
+        finally:
+            app.stop()
+
+
+class TestStatsCommand:
+    def test_stats_scrapes_metrics(self, capsys):
+        from bng_amd.cli.main import main
+        from bng_amd.metrics.metrics import Metrics
+        m = Metrics().serve(port=0)
+        try:
+            rc = main(["stats", "--metrics-url",
+                       f"http://127.0.0.1:{m.port}"])
+            assert rc == 0
+            out = capsys.readouterr().out
+            assert "bng_" in out
+        finally:
+            m.stop()
+
+    def test_stats_unreachable(self, capsys):
+        from bng_amd.cli.main import main
+        assert main(["stats", "--metrics-url",
+                     "http://127.0.0.1:1"]) == 1
