@@ -387,7 +387,8 @@ class HipLauncher:
         return verdict
 
     def uplink(self, data, lens, now_ns: Optional[int] = None,
-               now_sec: Optional[int] = None, sort_by_type: bool = False):
+               now_sec: Optional[int] = None, sort_by_type: bool = False,
+               order=None):
         """Fused antispoof -> NAT44 SNAT -> QoS-ingress + DHCP fast path.
 
         sort_by_type=True first classifies packets on-device and feeds the
@@ -397,8 +398,7 @@ class HipLauncher:
         n = lens.numel()
         verdict, out_len = self._outs(n)
         now = now_ns if now_ns is not None else time.time_ns()
-        order = None
-        if sort_by_type and n > 64:
+        if order is None and sort_by_type and n > 64:
             cls = self.torch.empty(n, dtype=self.torch.uint8,
                                    device=self.device)
             self.ext.pkt_class(data, lens, cls)
