@@ -258,7 +258,12 @@ def main():
     works = [torch.empty_like(pristine) for _ in range(nbuf)]
     clss = [torch.empty(args.batch, dtype=torch.uint8, device=device)
             for _ in range(nbuf)]
-    orders = [None] * nbuf
+    # persistent cross-stream buffers: the order tensor is produced on
+    # the prep stream and consumed on the main stream, so it must NOT be
+    # a prep-stream temporary (the caching allocator would recycle it
+    # while the main-stream kernel still reads it)
+    orders = [torch.zeros(args.batch, dtype=torch.int32, device=device)
+              for _ in range(nbuf)]
     prep_stream = torch.cuda.Stream(device=device) if overlap else None
     prep_done = [torch.cuda.Event() for _ in range(nbuf)]
     work_free = [torch.cuda.Event() for _ in range(nbuf)]
@@ -274,8 +279,8 @@ def main():
             works[b].copy_(pristine)               # RX-DMA analog
             if not args.no_sort:
                 launcher.ext.pkt_class(works[b], lens, clss[b])
-                orders[b] = torch.argsort(
-                    clss[b], stable=True).to(torch.int32)
+                idx = torch.argsort(clss[b], stable=True)
+                orders[b].copy_(idx.to(torch.int32))
             prep_done[b].record(prep_stream)
 
     def step(now_ns, k=0):
@@ -285,7 +290,7 @@ def main():
             launcher.uplink(works[b], lens, now_ns=now_ns,
                             now_sec=now_sec,
                             sort_by_type=not args.no_sort,
-                            order=orders[b])
+                            order=None if args.no_sort else orders[b])
             work_free[b].record(torch.cuda.current_stream(device))
             prep(k + 1)                            # overlap next batch
             return

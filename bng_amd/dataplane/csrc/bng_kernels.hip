@@ -1158,6 +1158,7 @@ void uplink_pipeline_kernel(bng_uplink_params P) {
        * code path — divergence between the DHCP and data pipelines no
        * longer serializes both per wave */
       int pid = P.order ? P.order[i] : i;
+      if ((unsigned)pid >= (unsigned)P.n) continue;  /* defensive */
       uint8_t* p = P.data + (size_t)pid * P.stride;
       int len = P.in_len[pid];
       uint16_t ol = (uint16_t)len;
