@@ -136,6 +136,7 @@ class Client:
         pkt = rp.Packet(rp.ACCESS_REQUEST, ident, req_auth)
         pkt.add(rp.USER_NAME, username)
         if chap is not None:
+            # chap = (challenge, ident_byte + md5_response) per RFC 2865
             challenge, chap_pw = chap
             pkt.add(rp.CHAP_CHALLENGE, challenge)
             pkt.add(rp.CHAP_PASSWORD, chap_pw)

@@ -82,12 +82,22 @@ class RadiusServer:
             self.auth_requests.append(req)
         user = req.get_str(rp.USER_NAME) or ""
         enc = req.get(rp.USER_PASSWORD)
+        chap = req.get(rp.CHAP_PASSWORD)
         rec = self.users.get(user)
         ok = False
         if rec is not None and enc is not None:
             pw = rp.decrypt_user_password(enc, self.secret,
                                           req.authenticator)
             ok = pw.decode(errors="replace") == rec.get("password", "")
+        elif rec is not None and chap is not None and len(chap) == 17:
+            # CHAP: value = MD5(ident | secret | challenge) (RFC 2865 §2.2)
+            import hashlib
+            ident, value = chap[0], chap[1:]
+            challenge = req.get(rp.CHAP_CHALLENGE) or req.authenticator
+            expect = hashlib.md5(bytes([ident]) +
+                                 rec.get("password", "").encode() +
+                                 challenge).digest()
+            ok = value == expect
         resp = rp.Packet(rp.ACCESS_ACCEPT if ok else rp.ACCESS_REJECT,
                          req.identifier)
         if ok:
