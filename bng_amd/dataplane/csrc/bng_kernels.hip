@@ -52,6 +52,20 @@ BNG_DEV const bng_sub_entry* sub_lookup(const bng_sub_entry* t, uint32_t mask,
   return nullptr;
 }
 
+BNG_DEV bng_subnat_entry* subnat_lookup_hint(bng_subnat_entry* t,
+                                             uint32_t mask, uint32_t ip,
+                                             uint32_t slot, uint4 first) {
+  if (first.x == ip) return &t[slot];
+  if (first.x == 0) return nullptr;
+  for (int i = 1; i < BNG_MAX_PROBE; ++i) {
+    bng_subnat_entry* e = &t[(slot + i) & mask];
+    uint4 v = *(const uint4*)e;
+    if (v.x == ip) return e;
+    if (v.x == 0) return nullptr;
+  }
+  return nullptr;
+}
+
 BNG_DEV bng_subnat_entry* subnat_lookup(bng_subnat_entry* t, uint32_t mask,
                                         uint32_t ip) {
   if (ip == 0) return nullptr;
@@ -60,6 +74,26 @@ BNG_DEV bng_subnat_entry* subnat_lookup(bng_subnat_entry* t, uint32_t mask,
     bng_subnat_entry* e = &t[(slot + i) & mask];
     uint4 v = *(const uint4*)e;    /* key_ip + sid + public_ip + ports */
     if (v.x == ip) return e;
+    if (v.x == 0) return nullptr;
+  }
+  return nullptr;
+}
+
+BNG_DEV bng_qos_bucket* qos_lookup_hint(bng_qos_bucket* t, uint32_t mask,
+                                        uint32_t ip, uint64_t* rate_out,
+                                        uint32_t slot, uint4 first) {
+  if (first.x == ip && (first.y & 0xFF)) {
+    *rate_out = ((uint64_t)first.w << 32) | first.z;
+    return &t[slot];
+  }
+  if (first.x == 0) return nullptr;
+  for (int i = 1; i < BNG_MAX_PROBE; ++i) {
+    bng_qos_bucket* e = &t[(slot + i) & mask];
+    uint4 v = *(const uint4*)e;
+    if (v.x == ip && (v.y & 0xFF)) {
+      *rate_out = ((uint64_t)v.w << 32) | v.z;
+      return e;
+    }
     if (v.x == 0) return nullptr;
   }
   return nullptr;
@@ -77,6 +111,22 @@ BNG_DEV bng_qos_bucket* qos_lookup(bng_qos_bucket* t, uint32_t mask,
       return e;
     }
     if (v.x == 0) return nullptr;
+  }
+  return nullptr;
+}
+
+BNG_DEV const bng_binding_entry* binding_lookup_hint(
+    const bng_binding_entry* t, uint32_t mask, uint64_t mac,
+    uint32_t slot, uint4 first) {
+  uint64_t k = ((uint64_t)first.y << 32) | first.x;
+  if (k == mac) return &t[slot];
+  if (k == BNG_KEY_EMPTY) return nullptr;
+  for (int i = 1; i < BNG_MAX_PROBE; ++i) {
+    const bng_binding_entry* e = &t[(slot + i) & mask];
+    uint4 v = *(const uint4*)e;
+    k = ((uint64_t)v.y << 32) | v.x;
+    if (k == mac) return e;
+    if (k == BNG_KEY_EMPTY) return nullptr;
   }
   return nullptr;
 }
@@ -578,13 +628,17 @@ BNG_DEV uint16_t csum_upd16(uint16_t csum, uint16_t oldv, uint16_t newv) {
   return (uint16_t)~s;
 }
 
-/* SNAT (ref nat44_egress nat44.c:565-802) */
-BNG_DEV int nat_egress_process(pktctx& c, const nat_tables& T, nat_flags& F) {
+/* SNAT (ref nat44_egress nat44.c:565-802); blk_pre/blk_pre_valid let the
+ * fused pipeline supply an already-probed port block */
+BNG_DEV int nat_egress_process(pktctx& c, const nat_tables& T, nat_flags& F,
+                               bng_subnat_entry* blk_pre = nullptr,
+                               bool blk_pre_valid = false) {
   uint8_t* p = c.p;
   if (c.ip_off < 0) return BNG_FWD;
   const bng_nat_config* cfg = T.cfg;
   if (!nat_is_private(cfg, c.saddr)) return BNG_FWD;
-  bng_subnat_entry* blk = subnat_lookup(T.subnat, T.subnat_mask, c.saddr);
+  bng_subnat_entry* blk = blk_pre_valid ? blk_pre
+      : subnat_lookup(T.subnat, T.subnat_mask, c.saddr);
   if (!blk) { F.passed = true; return BNG_PASS; }
 
   uint16_t sport, dport;
@@ -981,6 +1035,16 @@ BNG_DEV bool qos_tb_check(bng_qos_bucket* tb, uint32_t pkt_len,
   return qos_consume(tb, pkt_len);
 }
 
+BNG_DEV int qos_process_with(pktctx& c, bng_qos_bucket* tb, uint64_t rate,
+                             uint64_t now_ns, qos_flags& F) {
+  if (!tb) return BNG_FWD;
+  bool ok = qos_tb_check(tb, c.len, now_ns, rate);
+  F.bytes = c.len;
+  if (ok) { F.passed = true; return BNG_FWD; }
+  F.dropped = true;
+  return BNG_DROP;
+}
+
 BNG_DEV int qos_process(pktctx& c, bng_qos_bucket* table, uint32_t mask,
                         bool egress, uint64_t now_ns, qos_flags& F) {
   if (c.ip_off < 0) return BNG_FWD;
@@ -1041,17 +1105,11 @@ BNG_DEV void spoof_log_push(bng_spoof_event* ring, bng_ring_header* hdr,
 
 /* uRPF source validation (ref antispoof_ingress antispoof.c:189-293),
  * quirks preserved (see golden.py docstring). */
-BNG_DEV int antispoof_process(uint8_t* p, int len,
-                              const bng_binding_entry* bindings,
-                              uint32_t bmask,
+BNG_DEV int antispoof_process_with(uint8_t* p, int len,
+                              const bng_binding_entry* b,
                               const bng_antispoof_config* cfg,
                               bng_spoof_event* ring, bng_ring_header* hdr,
                               uint64_t now_ns, as_flags& F) {
-  if (len < 14) return BNG_FWD;
-  uint64_t mac = 0;
-  #pragma unroll
-  for (int i = 0; i < 6; ++i) mac = (mac << 8) | p[6 + i];
-  const bng_binding_entry* b = binding_lookup(bindings, bmask, mac);
   uint8_t mode = b ? b->mode : cfg->default_mode;
   if (mode == BNG_AS_DISABLED) { F.allowed = true; return BNG_FWD; }
   uint16_t proto = ld_u16be(p + 12);
@@ -1104,6 +1162,20 @@ BNG_DEV int antispoof_process(uint8_t* p, int len,
   }
   F.allowed = true;
   return BNG_FWD;
+}
+
+BNG_DEV int antispoof_process(uint8_t* p, int len,
+                              const bng_binding_entry* bindings,
+                              uint32_t bmask,
+                              const bng_antispoof_config* cfg,
+                              bng_spoof_event* ring, bng_ring_header* hdr,
+                              uint64_t now_ns, as_flags& F) {
+  if (len < 14) return BNG_FWD;
+  uint64_t mac = 0;
+  #pragma unroll
+  for (int i = 0; i < 6; ++i) mac = (mac << 8) | p[6 + i];
+  const bng_binding_entry* b = binding_lookup(bindings, bmask, mac);
+  return antispoof_process_with(p, len, b, cfg, ring, hdr, now_ns, F);
 }
 
 __global__ void antispoof_kernel(
@@ -1169,6 +1241,35 @@ void uplink_pipeline_kernel(bng_uplink_params P) {
                      c.dport == 67;
       if (is_dhcp) {
         v = dhcp_process(p, len, P.stride, DT, DF, &ol);
+      } else if (ip_ok && !c.tagged && len >= 14) {
+        /* hot data path: the three independent first-probe loads
+         * (binding / port block / qos bucket) issue together; each is
+         * waited for at its own compare site, so their HBM latencies
+         * overlap instead of chaining */
+        uint64_t mac = 0;
+        #pragma unroll
+        for (int j = 0; j < 6; ++j) mac = (mac << 8) | p[6 + j];
+        uint32_t s0 = (uint32_t)bng_mix64(mac) & P.bmask;
+        uint32_t s1 = (uint32_t)bng_mix64(c.saddr) & P.subnat_mask;
+        uint32_t s2 = (uint32_t)bng_mix64(c.saddr) & P.qos_mask;
+        uint4 f0 = *(const uint4*)&P.bindings[s0];
+        uint4 f1 = *(const uint4*)&P.subnat[s1];
+        uint4 f2 = *(const uint4*)&P.qos_in[s2];
+        const bng_binding_entry* b =
+            binding_lookup_hint(P.bindings, P.bmask, mac, s0, f0);
+        v = antispoof_process_with(p, len, b, P.acfg, P.spoof_ring,
+                                   P.spoof_hdr, P.now_ns, AF);
+        if (v == BNG_FWD) {
+          bng_subnat_entry* blk = subnat_lookup_hint(
+              P.subnat, P.subnat_mask, c.saddr, s1, f1);
+          v = nat_egress_process(c, NT, NF, blk, true);
+          if (v == BNG_FWD) {
+            uint64_t rate = 0;
+            bng_qos_bucket* tb = qos_lookup_hint(
+                P.qos_in, P.qos_mask, c.saddr, &rate, s2, f2);
+            v = qos_process_with(c, tb, rate, P.now_ns, QF);
+          }
+        }
       } else {
         /* the reference's TC programs parse untagged frames only
          * (nat44.c:573-581, antispoof.c:194-219): a tagged non-DHCP frame
