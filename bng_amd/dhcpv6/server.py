@@ -144,6 +144,12 @@ class DHCPv6Server:
     # ------------------------------------------------------------ handle
     def handle(self, data: bytes) -> Optional[bytes]:
         try:
+            msg_type = data[0] if data else 0
+        except IndexError:
+            return None
+        if msg_type == RELAY_FORW:
+            return self._relay(data)
+        try:
             msg = DHCPv6Message.decode(data)
         except ValueError:
             return None
@@ -286,6 +292,41 @@ class DHCPv6Server:
         resp = self._base_reply(msg, REPLY, duid)
         resp.add(OPT_STATUS_CODE, struct.pack(">H", STATUS_SUCCESS) + b"ok")
         return resp
+
+    OPT_RELAY_MSG = 9
+    OPT_INTERFACE_ID = 18
+
+    def _relay(self, data: bytes) -> Optional[bytes]:
+        """RELAY-FORW: unwrap the inner message, process it, wrap the
+        answer in RELAY-REPL echoing peer/link addresses and
+        Interface-Id (RFC 8415 §19)."""
+        if len(data) < 34:
+            return None
+        hop = data[1]
+        link_peer = data[2:34]
+        inner = None
+        iface_id = None
+        i = 34
+        while i + 4 <= len(data):
+            t, ln = struct.unpack_from(">HH", data, i)
+            if i + 4 + ln > len(data):
+                break
+            if t == self.OPT_RELAY_MSG:
+                inner = data[i + 4:i + 4 + ln]
+            elif t == self.OPT_INTERFACE_ID:
+                iface_id = data[i + 4:i + 4 + ln]
+            i += 4 + ln
+        if inner is None:
+            return None
+        answer = self.handle(inner)     # recursion covers relay chains
+        if answer is None:
+            return None
+        out = bytes([RELAY_REPL, hop]) + link_peer
+        if iface_id is not None:
+            out += struct.pack(">HH", self.OPT_INTERFACE_ID,
+                               len(iface_id)) + iface_id
+        out += struct.pack(">HH", self.OPT_RELAY_MSG, len(answer)) + answer
+        return out
 
     def _inforeq(self, msg, duid):
         self.stats["info_request"] += 1

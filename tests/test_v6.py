@@ -156,3 +156,33 @@ class TestSLAAC:
         assert sent[0][1] == "fe80::1"
         assert parse_ra(sent[0][0])["prefixes"]
         assert srv.handle_rs(b"\x00" * 8) is None
+
+
+class TestDHCPv6Relay:
+    def test_relay_forw_repl_roundtrip(self):
+        import struct as st
+        from bng_amd.dhcpv6.server import RELAY_FORW, RELAY_REPL
+        srv = DHCPv6Server(rapid_commit=True)
+        inner = solicit(rapid=True).encode()
+        link = b"\x20\x01" + b"\x00" * 14
+        peer = b"\xfe\x80" + b"\x00" * 14
+        relay = bytes([RELAY_FORW, 0]) + link + peer + \
+            st.pack(">HH", 18, 5) + b"pon01" + \
+            st.pack(">HH", 9, len(inner)) + inner
+        out = srv.handle(relay)
+        assert out is not None and out[0] == RELAY_REPL
+        assert out[2:18] == link and out[18:34] == peer
+        # interface-id echoed, inner REPLY carried
+        i, inner_reply, iface = 34, None, None
+        while i + 4 <= len(out):
+            t, ln = st.unpack_from(">HH", out, i)
+            if t == 18:
+                iface = out[i + 4:i + 4 + ln]
+            if t == 9:
+                inner_reply = out[i + 4:i + 4 + ln]
+            i += 4 + ln
+        assert iface == b"pon01"
+        rep = DHCPv6Message.decode(inner_reply)
+        assert rep.msg_type == REPLY
+        assert extract_addr(rep) is not None
+        assert (DUID, 1, False) in srv.bindings
