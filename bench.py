@@ -84,42 +84,35 @@ def build_tables(launcher, rank: int, world: int, n_subs: int, now_sec: int):
     n_bad = int((rc != 0).sum().item())
     assert n_bad == 0, f"{n_bad} subscriber upserts failed (table too full)"
 
-    # subnat port blocks (IP-owned shard)
+    # merged subscriber context (IP-owned shard): NAT port block +
+    # ingress QoS bucket (1 Gbps / 4 MB burst => pass) in one 64-B entry
     sel = np.nonzero(ip_owner == rank)[0]
     pub_base = ip2u32("203.0.113.0")
-    snat = np.zeros(len(sel), dtype=[("key_ip", "<u4"), ("sid", "<u4"),
-                                     ("pub", "<u4"), ("ps", "<u2"),
-                                     ("pe", "<u2"), ("np", "<u4"),
-                                     ("piu", "<u4"), ("at", "<u8"),
-                                     ("sa", "<u4"), ("st", "<u4"),
-                                     ("bo", "<u8"), ("bi", "<u8"),
-                                     ("bl", "u1"), ("fl", "u1"),
-                                     ("pad", "6u1")])
-    snat["key_ip"] = ips[sel].astype(np.uint32)
-    snat["sid"] = sel.astype(np.uint32)
-    snat["pub"] = pub_base + (sel % 250).astype(np.uint32)
-    starts = (1024 + (sel % 63) * 1024).astype(np.uint16)
-    snat["ps"] = starts
-    snat["pe"] = starts + 1023
-    snat["np"] = starts
-    rc = torch.zeros(len(sel), dtype=torch.int32, device=dev)
-    launcher.ext.subnat_upsert(launcher.subnat, to_dev_u8(snat), rc)
-    assert int((rc != 0).sum().item()) == 0
-
-    # QoS ingress buckets (IP-owned shard): 1 Gbps / 4 MB burst => pass
-    qos = np.zeros(len(sel), dtype=[("key_ip", "<u4"), ("valid", "u1"),
-                                    ("prio", "u1"), ("pad", "<u2"),
+    ctx = np.zeros(len(sel), dtype=[("key_ip", "<u4"), ("pub", "<u4"),
+                                    ("ps", "<u2"), ("pe", "<u2"),
+                                    ("qv", "u1"), ("nv", "u1"),
+                                    ("prio", "u1"), ("fl", "u1"),
                                     ("rate", "<u8"), ("tokens", "<i8"),
                                     ("last", "<u8"), ("burst", "<u4"),
-                                    ("pad2", "<u4"), ("pad3", "3<u8")])
-    qos["key_ip"] = ips[sel].astype(np.uint32)
-    qos["valid"] = 1
-    qos["rate"] = 10**9
-    qos["tokens"] = 4 << 20
-    qos["burst"] = 4 << 20
-    qos["last"] = now_sec * 10**9
+                                    ("np", "<u4"), ("sid", "<u4"),
+                                    ("sa", "<u4"), ("st", "<u4"),
+                                    ("pad", "<u4")])
+    ctx["key_ip"] = ips[sel].astype(np.uint32)
+    ctx["sid"] = sel.astype(np.uint32)
+    ctx["pub"] = pub_base + (sel % 250).astype(np.uint32)
+    starts = (1024 + (sel % 63) * 1024).astype(np.uint16)
+    ctx["ps"] = starts
+    ctx["pe"] = starts + 1023
+    ctx["np"] = starts
+    ctx["nv"] = 1
+    ctx["qv"] = 1
+    ctx["rate"] = 10**9
+    ctx["tokens"] = 4 << 20
+    ctx["burst"] = 4 << 20
+    ctx["last"] = now_sec * 10**9
     rc = torch.zeros(len(sel), dtype=torch.int32, device=dev)
-    launcher.ext.qos_upsert(launcher.qos_ingress, to_dev_u8(qos), rc)
+    launcher.ext.subctx_upsert(launcher.subctx, to_dev_u8(ctx),
+                               abi.CTX_SET_NAT | abi.CTX_SET_QOS, rc)
     assert int((rc != 0).sum().item()) == 0
 
     # antispoof strict bindings (packets arrive by IP shard, keyed by MAC;

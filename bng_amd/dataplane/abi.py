@@ -155,15 +155,22 @@ class EimEntry(C.Structure):
                 ("ref_count", C.c_uint32), ("flags", C.c_uint32)]
 
 
-class SubnatEntry(C.Structure):
-    _fields_ = [("key_ip", C.c_uint32), ("subscriber_id", C.c_uint32),
-                ("public_ip", C.c_uint32), ("port_start", C.c_uint16),
-                ("port_end", C.c_uint16), ("next_port", C.c_uint32),
-                ("ports_in_use", C.c_uint32), ("allocated_at", C.c_uint64),
-                ("sessions_active", C.c_uint32), ("sessions_total", C.c_uint32),
-                ("bytes_out", C.c_uint64), ("bytes_in", C.c_uint64),
-                ("block_size_log2", C.c_uint8), ("flags", C.c_uint8),
-                ("_pad", C.c_uint8 * 6)]
+class SubCtx(C.Structure):
+    """Merged per-subscriber uplink context: RFC6431 port block (ref
+    subscriber_nat nat44.c:157-164) + ingress token bucket (ref
+    qos_ingress qos_ratelimit.c:44-50) in one 64-B entry."""
+    _fields_ = [("key_ip", C.c_uint32), ("public_ip", C.c_uint32),
+                ("port_start", C.c_uint16), ("port_end", C.c_uint16),
+                ("qos_valid", C.c_uint8), ("nat_valid", C.c_uint8),
+                ("priority", C.c_uint8), ("flags", C.c_uint8),
+                ("rate_bps", C.c_uint64), ("tokens", C.c_int64),
+                ("last_update", C.c_uint64), ("burst_bytes", C.c_uint32),
+                ("next_port", C.c_uint32), ("subscriber_id", C.c_uint32),
+                ("sessions_active", C.c_uint32),
+                ("sessions_total", C.c_uint32), ("_pad", C.c_uint32)]
+
+
+CTX_SET_NAT, CTX_SET_QOS, CTX_CLR_QOS, CTX_CLR_NAT = 1, 2, 4, 8
 
 
 class NatConfig(C.Structure):
@@ -229,7 +236,7 @@ EXPECTED_SIZES = {
     "bng_nat_session": (NatSession, 128),
     "bng_nat_reverse": (NatReverse, 48),
     "bng_eim_entry": (EimEntry, 48),
-    "bng_subnat_entry": (SubnatEntry, 64),
+    "bng_subctx": (SubCtx, 64),
     "bng_nat_config": (NatConfig, 24 + 64 * 8 + 64 * 4),
     "bng_nat_log_entry": (NatLogEntry, 40),
     "bng_qos_bucket": (QosBucket, 64),

@@ -36,8 +36,8 @@ void bng_launch_pkt_class(const void*, const void*, void*, int, int,
 void bng_launch_sub_upsert(void*, uint32_t, const void*, int, void*,
                            hipStream_t);
 void bng_launch_sub_delete(void*, uint32_t, const void*, int, hipStream_t);
-void bng_launch_subnat_upsert(void*, uint32_t, const void*, int, void*,
-                              hipStream_t);
+void bng_launch_subctx_upsert(void*, uint32_t, const void*, int, uint32_t,
+                              void*, hipStream_t);
 void bng_launch_qos_upsert(void*, uint32_t, const void*, int, void*,
                            hipStream_t);
 void bng_launch_binding_upsert(void*, uint32_t, const void*, int, void*,
@@ -102,7 +102,7 @@ void nat44(torch::Tensor data, torch::Tensor in_len, torch::Tensor verdict,
       table_mask(sessions, sizeof(bng_nat_session), "sessions"),
       reverse.data_ptr(), table_mask(reverse, sizeof(bng_nat_reverse), "reverse"),
       eim.data_ptr(), table_mask(eim, sizeof(bng_eim_entry), "eim"),
-      subnat.data_ptr(), table_mask(subnat, sizeof(bng_subnat_entry), "subnat"),
+      subnat.data_ptr(), table_mask(subnat, sizeof(bng_subctx), "subctx"),
       cfg.data_ptr(), hairpin.data_ptr(), (uint32_t)n_hairpin,
       stats.data_ptr(), log_ring.data_ptr(), log_hdr.data_ptr(),
       (uint64_t)now_ns, cur_stream());
@@ -115,6 +115,8 @@ void qos(torch::Tensor data, torch::Tensor in_len, torch::Tensor verdict,
   bng_launch_qos(data.data_ptr(), in_len.data_ptr(), verdict.data_ptr(),
                  (int)in_len.numel(), (int)data.size(1), is_egress ? 1 : 0,
                  table.data_ptr(),
+                 /* bng_qos_bucket (egress) and bng_subctx (ingress) are
+                    both 64 B, so one mask computation serves both */
                  table_mask(table, sizeof(bng_qos_bucket), "qos"),
                  stats.data_ptr(), (uint64_t)now_ns, cur_stream());
 }
@@ -179,16 +181,16 @@ void uplink_pipeline(torch::Tensor data, torch::Tensor in_len,
   P.rev_mask = table_mask(reverse, sizeof(bng_nat_reverse), "reverse");
   P.eim = (bng_eim_entry*)eim.data_ptr();
   P.eim_mask = table_mask(eim, sizeof(bng_eim_entry), "eim");
-  P.subnat = (bng_subnat_entry*)subnat.data_ptr();
-  P.subnat_mask = table_mask(subnat, sizeof(bng_subnat_entry), "subnat");
+  P.subctx = (bng_subctx*)subnat.data_ptr();
+  P.subctx_mask = table_mask(subnat, sizeof(bng_subctx), "subctx");
   P.ncfg = (const bng_nat_config*)ncfg.data_ptr();
   P.hairpin_ips = (const uint32_t*)hairpin.data_ptr();
   P.n_hairpin = (uint32_t)n_hairpin;
   P.nat_stats = (unsigned long long*)nat_stats.data_ptr();
   P.log_ring = (bng_nat_log_entry*)log_ring.data_ptr();
   P.log_hdr = (bng_ring_header*)log_hdr.data_ptr();
-  P.qos_in = (bng_qos_bucket*)qos_in.data_ptr();
-  P.qos_mask = table_mask(qos_in, sizeof(bng_qos_bucket), "qos");
+  P.qos_eg = (bng_qos_bucket*)qos_in.data_ptr();
+  P.qos_eg_mask = table_mask(qos_in, sizeof(bng_qos_bucket), "qos");
   P.qos_stats = (unsigned long long*)qos_stats.data_ptr();
   P.now_ns = (uint64_t)now_ns;
   P.now_sec = (uint64_t)now_sec;
@@ -217,13 +219,14 @@ void sub_delete(torch::Tensor table, torch::Tensor keys) {
                         table_mask(table, sizeof(bng_sub_entry), "subs"),
                         keys.data_ptr(), (int)keys.numel(), cur_stream());
 }
-void subnat_upsert(torch::Tensor table, torch::Tensor batch,
-                   torch::Tensor rc) {
+void subctx_upsert(torch::Tensor table, torch::Tensor batch,
+                   int64_t update_mask, torch::Tensor rc) {
   int n = (int)(batch.numel() * batch.element_size() /
-                sizeof(bng_subnat_entry));
-  bng_launch_subnat_upsert(
-      table.data_ptr(), table_mask(table, sizeof(bng_subnat_entry), "subnat"),
-      batch.data_ptr(), n, rc.data_ptr(), cur_stream());
+                sizeof(bng_subctx));
+  bng_launch_subctx_upsert(
+      table.data_ptr(), table_mask(table, sizeof(bng_subctx), "subctx"),
+      batch.data_ptr(), n, (uint32_t)update_mask, rc.data_ptr(),
+      cur_stream());
 }
 void qos_upsert(torch::Tensor table, torch::Tensor batch, torch::Tensor rc) {
   int n = (int)(batch.numel() * batch.element_size() /
@@ -258,7 +261,7 @@ void nat_sweep(torch::Tensor sessions, torch::Tensor reverse,
       sessions.data_ptr(), n_slots, reverse.data_ptr(),
       table_mask(reverse, sizeof(bng_nat_reverse), "reverse"),
       subnat.data_ptr(),
-      table_mask(subnat, sizeof(bng_subnat_entry), "subnat"),
+      table_mask(subnat, sizeof(bng_subctx), "subctx"),
       (uint64_t)now_ns, (uint64_t)udp_to, (uint64_t)tcp_est_to,
       (uint64_t)tcp_tr_to, (uint64_t)icmp_to, stats.data_ptr(),
       cur_stream());
@@ -276,7 +279,7 @@ py::dict layout_report() {
 #define SZ(T) d[#T] = sizeof(T)
   SZ(bng_sub_entry); SZ(bng_ip_pool); SZ(bng_server_config);
   SZ(bng_nat_tuple); SZ(bng_nat_session); SZ(bng_nat_reverse);
-  SZ(bng_eim_entry); SZ(bng_subnat_entry); SZ(bng_nat_config);
+  SZ(bng_eim_entry); SZ(bng_subctx); SZ(bng_nat_config);
   SZ(bng_nat_log_entry); SZ(bng_qos_bucket); SZ(bng_binding_entry);
   SZ(bng_antispoof_config); SZ(bng_spoof_event); SZ(bng_ring_header);
 #undef SZ
@@ -286,9 +289,9 @@ py::dict layout_report() {
   off["nat_session.created"] = offsetof(bng_nat_session, created);
   off["nat_session.ready"] = offsetof(bng_nat_session, ready);
   off["eim_entry.created"] = offsetof(bng_eim_entry, created);
-  off["subnat_entry.next_port"] = offsetof(bng_subnat_entry, next_port);
-  off["subnat_entry.sessions_active"] =
-      offsetof(bng_subnat_entry, sessions_active);
+  off["subctx.rate_bps"] = offsetof(bng_subctx, rate_bps);
+  off["subctx.next_port"] = offsetof(bng_subctx, next_port);
+  off["subctx.sessions_active"] = offsetof(bng_subctx, sessions_active);
   off["qos_bucket.tokens"] = offsetof(bng_qos_bucket, tokens);
   off["qos_bucket.last_update"] = offsetof(bng_qos_bucket, last_update);
   off["binding_entry.ipv6_addr"] = offsetof(bng_binding_entry, ipv6_addr);
@@ -328,7 +331,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pkt_class", &pkt_class);
   m.def("sub_upsert", &sub_upsert);
   m.def("sub_delete", &sub_delete);
-  m.def("subnat_upsert", &subnat_upsert);
+  m.def("subctx_upsert", &subctx_upsert);
   m.def("qos_upsert", &qos_upsert);
   m.def("binding_upsert", &binding_upsert);
   m.def("binding_delete", &binding_delete);

@@ -52,13 +52,17 @@ BNG_DEV const bng_sub_entry* sub_lookup(const bng_sub_entry* t, uint32_t mask,
   return nullptr;
 }
 
-BNG_DEV bng_subnat_entry* subnat_lookup_hint(bng_subnat_entry* t,
-                                             uint32_t mask, uint32_t ip,
-                                             uint32_t slot, uint4 first) {
+/* Merged subscriber-context probe: ONE table walk serves both the NAT
+ * port-block and ingress-QoS stages (the reference walks subscriber_nat
+ * nat44.c:157-164 AND qos_ingress qos_ratelimit.c:44-50 separately —
+ * two random HBM touches per packet; this is one). */
+BNG_DEV bng_subctx* subctx_lookup_hint(bng_subctx* t,
+                                       uint32_t mask, uint32_t ip,
+                                       uint32_t slot, uint4 first) {
   if (first.x == ip) return &t[slot];
   if (first.x == 0) return nullptr;
   for (int i = 1; i < BNG_MAX_PROBE; ++i) {
-    bng_subnat_entry* e = &t[(slot + i) & mask];
+    bng_subctx* e = &t[(slot + i) & mask];
     uint4 v = *(const uint4*)e;
     if (v.x == ip) return e;
     if (v.x == 0) return nullptr;
@@ -66,13 +70,13 @@ BNG_DEV bng_subnat_entry* subnat_lookup_hint(bng_subnat_entry* t,
   return nullptr;
 }
 
-BNG_DEV bng_subnat_entry* subnat_lookup(bng_subnat_entry* t, uint32_t mask,
-                                        uint32_t ip) {
+BNG_DEV bng_subctx* subctx_lookup(bng_subctx* t, uint32_t mask,
+                                  uint32_t ip) {
   if (ip == 0) return nullptr;
   uint32_t slot = (uint32_t)bng_mix64(ip) & mask;
   for (int i = 0; i < BNG_MAX_PROBE; ++i) {
-    bng_subnat_entry* e = &t[(slot + i) & mask];
-    uint4 v = *(const uint4*)e;    /* key_ip + sid + public_ip + ports */
+    bng_subctx* e = &t[(slot + i) & mask];
+    uint4 v = *(const uint4*)e;    /* key + pub_ip + ports + valid flags */
     if (v.x == ip) return e;
     if (v.x == 0) return nullptr;
   }
@@ -532,7 +536,7 @@ struct nat_tables {
   bng_nat_session* sessions; uint32_t sess_mask;
   bng_nat_reverse* reverse;  uint32_t rev_mask;
   bng_eim_entry* eim;        uint32_t eim_mask;
-  bng_subnat_entry* subnat;  uint32_t subnat_mask;
+  bng_subctx* ctx;           uint32_t ctx_mask;
   const bng_nat_config* cfg;
   const uint32_t* hairpin_ips; uint32_t n_hairpin;
   unsigned long long* stats;
@@ -587,7 +591,7 @@ BNG_DEV void nat_log_push(const nat_tables& T, uint32_t ev, uint32_t sub_id,
 /* Port rotor (ref allocate_port_from_block nat44.c:408-466): atomic
  * next_port bump, wrap, optional RTP parity, EIM-collision heuristic.
  * The reference's wrap is a tolerated benign race; ours keeps it. */
-BNG_DEV uint16_t nat_alloc_port(const nat_tables& T, bng_subnat_entry* blk,
+BNG_DEV uint16_t nat_alloc_port(const nat_tables& T, bng_subctx* blk,
                                 bool parity, uint16_t orig_port,
                                 uint32_t internal_ip, uint8_t proto) {
   uint8_t orig_parity = orig_port & 1;
@@ -631,15 +635,15 @@ BNG_DEV uint16_t csum_upd16(uint16_t csum, uint16_t oldv, uint16_t newv) {
 /* SNAT (ref nat44_egress nat44.c:565-802); blk_pre/blk_pre_valid let the
  * fused pipeline supply an already-probed port block */
 BNG_DEV int nat_egress_process(pktctx& c, const nat_tables& T, nat_flags& F,
-                               bng_subnat_entry* blk_pre = nullptr,
+                               bng_subctx* blk_pre = nullptr,
                                bool blk_pre_valid = false) {
   uint8_t* p = c.p;
   if (c.ip_off < 0) return BNG_FWD;
   const bng_nat_config* cfg = T.cfg;
   if (!nat_is_private(cfg, c.saddr)) return BNG_FWD;
-  bng_subnat_entry* blk = blk_pre_valid ? blk_pre
-      : subnat_lookup(T.subnat, T.subnat_mask, c.saddr);
-  if (!blk) { F.passed = true; return BNG_PASS; }
+  bng_subctx* blk = blk_pre_valid ? blk_pre
+      : subctx_lookup(T.ctx, T.ctx_mask, c.saddr);
+  if (!blk || !blk->nat_valid) { F.passed = true; return BNG_PASS; }
 
   uint16_t sport, dport;
   if (c.proto == 6) {
@@ -956,14 +960,14 @@ __global__ void nat44_kernel(
     bng_nat_session* sessions, uint32_t sess_mask,
     bng_nat_reverse* reverse, uint32_t rev_mask,
     bng_eim_entry* eim, uint32_t eim_mask,
-    bng_subnat_entry* subnat, uint32_t subnat_mask,
+    bng_subctx* ctx, uint32_t ctx_mask,
     const bng_nat_config* cfg,
     const uint32_t* hairpin_ips, uint32_t n_hairpin,
     unsigned long long* stats,
     bng_nat_log_entry* log_ring, bng_ring_header* log_hdr,
     uint64_t now_ns) {
   nat_tables T{sessions, sess_mask, reverse, rev_mask, eim, eim_mask,
-               subnat, subnat_mask, cfg, hairpin_ips, n_hairpin, stats,
+               ctx, ctx_mask, cfg, hairpin_ips, n_hairpin, stats,
                log_ring, log_hdr, now_ns};
   int tid = blockIdx.x * blockDim.x + threadIdx.x;
   int nthreads = gridDim.x * blockDim.x;
@@ -992,57 +996,50 @@ struct qos_flags { bool passed, dropped; uint32_t bytes; };
  * CAS on last_update) and one retry.  Verdict-equivalent to the
  * reference's refill-then-consume: unclaimed credit stays recoverable
  * because last_update only advances when a refill actually runs. */
-BNG_DEV bool qos_consume(bng_qos_bucket* tb, uint32_t pkt_len) {
+BNG_DEV bool qos_consume(int64_t* tokens, uint32_t pkt_len) {
   long long old = (long long)atomicAdd(
-      (unsigned long long*)&tb->tokens,
+      (unsigned long long*)tokens,
       (unsigned long long)(-(int64_t)pkt_len));
   if (old - (int64_t)pkt_len >= 0) return true;
-  atomicAdd((unsigned long long*)&tb->tokens, (unsigned long long)pkt_len);
+  atomicAdd((unsigned long long*)tokens, (unsigned long long)pkt_len);
   return false;
 }
 
-BNG_DEV bool qos_tb_check(bng_qos_bucket* tb, uint32_t pkt_len,
+/* field-pointer form so the standalone egress bucket (bng_qos_bucket)
+ * and the merged uplink context (bng_subctx) share one implementation */
+BNG_DEV bool qos_tb_check(int64_t* tokens, uint64_t* last_update,
+                          uint32_t burst_bytes, uint32_t pkt_len,
                           uint64_t now_ns, uint64_t rate) {
   if (rate == 0) return true;
-  if (qos_consume(tb, pkt_len)) return true;
+  if (qos_consume(tokens, pkt_len)) return true;
   /* insufficient: claim the refill, credit elapsed time, retry once */
-  uint64_t last = __hip_atomic_load(&tb->last_update, __ATOMIC_RELAXED,
+  uint64_t last = __hip_atomic_load(last_update, __ATOMIC_RELAXED,
                                     __HIP_MEMORY_SCOPE_AGENT);
   if (last != now_ns &&
       __hip_atomic_compare_exchange_strong(
-          &tb->last_update, &last, now_ns, __ATOMIC_RELAXED,
+          last_update, &last, now_ns, __ATOMIC_RELAXED,
           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)) {
     uint64_t elapsed = now_ns - last;
-    int64_t burst = (int64_t)tb->burst_bytes;
+    int64_t burst = (int64_t)burst_bytes;
     uint64_t add = (elapsed > 100000000000ull) ? (uint64_t)burst
         : (elapsed * (rate / 8)) / 1000000000ull;
     /* capped add, bounded CAS loop */
     for (int t = 0; t < 16; ++t) {
       int64_t cur = (int64_t)__hip_atomic_load(
-          (uint64_t*)&tb->tokens, __ATOMIC_RELAXED,
+          (uint64_t*)tokens, __ATOMIC_RELAXED,
           __HIP_MEMORY_SCOPE_AGENT);
       int64_t want = cur + (int64_t)add;
       if (want > burst) want = burst;
       if (want == cur) break;
       uint64_t expect = (uint64_t)cur;
       if (__hip_atomic_compare_exchange_strong(
-              (uint64_t*)&tb->tokens, &expect, (uint64_t)want,
+              (uint64_t*)tokens, &expect, (uint64_t)want,
               __ATOMIC_RELAXED, __ATOMIC_RELAXED,
               __HIP_MEMORY_SCOPE_AGENT))
         break;
     }
   }
-  return qos_consume(tb, pkt_len);
-}
-
-BNG_DEV int qos_process_with(pktctx& c, bng_qos_bucket* tb, uint64_t rate,
-                             uint64_t now_ns, qos_flags& F) {
-  if (!tb) return BNG_FWD;
-  bool ok = qos_tb_check(tb, c.len, now_ns, rate);
-  F.bytes = c.len;
-  if (ok) { F.passed = true; return BNG_FWD; }
-  F.dropped = true;
-  return BNG_DROP;
+  return qos_consume(tokens, pkt_len);
 }
 
 BNG_DEV int qos_process(pktctx& c, bng_qos_bucket* table, uint32_t mask,
@@ -1053,7 +1050,21 @@ BNG_DEV int qos_process(pktctx& c, bng_qos_bucket* table, uint32_t mask,
   uint64_t rate = 0;
   bng_qos_bucket* tb = qos_lookup(table, mask, key, &rate);
   if (!tb) return BNG_FWD;
-  bool ok = qos_tb_check(tb, c.len, now_ns, rate);
+  bool ok = qos_tb_check(&tb->tokens, &tb->last_update, tb->burst_bytes,
+                         c.len, now_ns, rate);
+  F.bytes = c.len;
+  if (ok) { F.passed = true; return BNG_FWD; }
+  F.dropped = true;
+  return BNG_DROP;
+}
+
+/* ingress QoS over the merged subscriber context — shares the entry the
+ * NAT stage already probed, so the policy check costs no extra walk */
+BNG_DEV int qos_process_ctx(pktctx& c, bng_subctx* e, uint64_t now_ns,
+                            qos_flags& F) {
+  if (c.ip_off < 0 || !e || !e->qos_valid) return BNG_FWD;
+  bool ok = qos_tb_check(&e->tokens, &e->last_update, e->burst_bytes,
+                         c.len, now_ns, e->rate_bps);
   F.bytes = c.len;
   if (ok) { F.passed = true; return BNG_FWD; }
   F.dropped = true;
@@ -1063,7 +1074,7 @@ BNG_DEV int qos_process(pktctx& c, bng_qos_bucket* table, uint32_t mask,
 __global__ void qos_kernel(
     uint8_t* __restrict__ data, const uint16_t* __restrict__ in_len,
     uint8_t* __restrict__ verdict, int n, int stride, int is_egress,
-    bng_qos_bucket* table, uint32_t mask,
+    void* table, uint32_t mask,
     unsigned long long* stats, uint64_t now_ns) {
   int tid = blockIdx.x * blockDim.x + threadIdx.x;
   int nthreads = gridDim.x * blockDim.x;
@@ -1073,8 +1084,15 @@ __global__ void qos_kernel(
     if (pid < n) {
       pktctx c;
       parse_pkt(c, data + (size_t)pid * stride, in_len[pid], false);
-      verdict[pid] = (uint8_t)qos_process(c, table, mask, is_egress != 0,
-                                          now_ns, F);
+      int v;
+      if (is_egress) {
+        v = qos_process(c, (bng_qos_bucket*)table, mask, true, now_ns, F);
+      } else {
+        bng_subctx* e = (c.ip_off >= 0)
+            ? subctx_lookup((bng_subctx*)table, mask, c.saddr) : nullptr;
+        v = qos_process_ctx(c, e, now_ns, F);
+      }
+      verdict[pid] = (uint8_t)v;
     }
     stat_inc(&stats[BNG_QS_PKT_PASSED], F.passed);
     stat_inc(&stats[BNG_QS_PKT_DROPPED], F.dropped);
@@ -1215,7 +1233,7 @@ void uplink_pipeline_kernel(bng_uplink_params P) {
   dhcp_tables DT{P.subs, P.sub_mask, P.pools, P.n_pools, P.scfg,
                  P.dhcp_stats, P.now_sec};
   nat_tables NT{P.sessions, P.sess_mask, P.reverse, P.rev_mask, P.eim,
-                P.eim_mask, P.subnat, P.subnat_mask, P.ncfg, P.hairpin_ips,
+                P.eim_mask, P.subctx, P.subctx_mask, P.ncfg, P.hairpin_ips,
                 P.n_hairpin, P.nat_stats, P.log_ring, P.log_hdr, P.now_ns};
   int tid = blockIdx.x * blockDim.x + threadIdx.x;
   int nthreads = gridDim.x * blockDim.x;
@@ -1242,33 +1260,28 @@ void uplink_pipeline_kernel(bng_uplink_params P) {
       if (is_dhcp) {
         v = dhcp_process(p, len, P.stride, DT, DF, &ol);
       } else if (ip_ok && !c.tagged && len >= 14) {
-        /* hot data path: the three independent first-probe loads
-         * (binding / port block / qos bucket) issue together; each is
-         * waited for at its own compare site, so their HBM latencies
-         * overlap instead of chaining */
+        /* hot data path: the two independent first-probe loads
+         * (binding / merged subscriber context) issue together; the
+         * context entry then serves BOTH the NAT port-block and the
+         * ingress-QoS stages — one random HBM touch where the
+         * reference's hook chain pays two */
         uint64_t mac = 0;
         #pragma unroll
         for (int j = 0; j < 6; ++j) mac = (mac << 8) | p[6 + j];
         uint32_t s0 = (uint32_t)bng_mix64(mac) & P.bmask;
-        uint32_t s1 = (uint32_t)bng_mix64(c.saddr) & P.subnat_mask;
-        uint32_t s2 = (uint32_t)bng_mix64(c.saddr) & P.qos_mask;
+        uint32_t s1 = (uint32_t)bng_mix64(c.saddr) & P.subctx_mask;
         uint4 f0 = *(const uint4*)&P.bindings[s0];
-        uint4 f1 = *(const uint4*)&P.subnat[s1];
-        uint4 f2 = *(const uint4*)&P.qos_in[s2];
+        uint4 f1 = *(const uint4*)&P.subctx[s1];
         const bng_binding_entry* b =
             binding_lookup_hint(P.bindings, P.bmask, mac, s0, f0);
         v = antispoof_process_with(p, len, b, P.acfg, P.spoof_ring,
                                    P.spoof_hdr, P.now_ns, AF);
         if (v == BNG_FWD) {
-          bng_subnat_entry* blk = subnat_lookup_hint(
-              P.subnat, P.subnat_mask, c.saddr, s1, f1);
-          v = nat_egress_process(c, NT, NF, blk, true);
-          if (v == BNG_FWD) {
-            uint64_t rate = 0;
-            bng_qos_bucket* tb = qos_lookup_hint(
-                P.qos_in, P.qos_mask, c.saddr, &rate, s2, f2);
-            v = qos_process_with(c, tb, rate, P.now_ns, QF);
-          }
+          bng_subctx* ctxe = subctx_lookup_hint(
+              P.subctx, P.subctx_mask, c.saddr, s1, f1);
+          v = nat_egress_process(c, NT, NF, ctxe, true);
+          if (v == BNG_FWD)
+            v = qos_process_ctx(c, ctxe, P.now_ns, QF);
         }
       } else {
         /* the reference's TC programs parse untagged frames only
@@ -1277,10 +1290,10 @@ void uplink_pipeline_kernel(bng_uplink_params P) {
         v = antispoof_process(p, len, P.bindings, P.bmask, P.acfg,
                               P.spoof_ring, P.spoof_hdr, P.now_ns, AF);
         if (v == BNG_FWD && ip_ok && !c.tagged) {
-          v = nat_egress_process(c, NT, NF);
+          bng_subctx* ctxe = subctx_lookup(P.subctx, P.subctx_mask, c.saddr);
+          v = nat_egress_process(c, NT, NF, ctxe, true);
           if (v == BNG_FWD)
-            v = qos_process(c, P.qos_in, P.qos_mask, /*egress=*/false,
-                            P.now_ns, QF);
+            v = qos_process_ctx(c, ctxe, P.now_ns, QF);
         }
       }
       P.verdict[pid] = (uint8_t)v;
@@ -1309,7 +1322,7 @@ __global__ __launch_bounds__(256, 5)
 void downlink_pipeline_kernel(bng_uplink_params P) {
   if (P.now_ptr) { P.now_ns = P.now_ptr[0]; P.now_sec = P.now_ptr[1]; }
   nat_tables NT{P.sessions, P.sess_mask, P.reverse, P.rev_mask, P.eim,
-                P.eim_mask, P.subnat, P.subnat_mask, P.ncfg, P.hairpin_ips,
+                P.eim_mask, P.subctx, P.subctx_mask, P.ncfg, P.hairpin_ips,
                 P.n_hairpin, P.nat_stats, P.log_ring, P.log_hdr, P.now_ns};
   int tid = blockIdx.x * blockDim.x + threadIdx.x;
   int nthreads = gridDim.x * blockDim.x;
@@ -1329,7 +1342,7 @@ void downlink_pipeline_kernel(bng_uplink_params P) {
         if (v == BNG_FWD && c.ip_off >= 0) {
           /* re-read the (possibly rewritten) destination for QoS */
           c.daddr = ld_u32be(p + c.ip_off + 16);
-          v = qos_process(c, P.qos_in, P.qos_mask, /*egress=*/true,
+          v = qos_process(c, P.qos_eg, P.qos_eg_mask, /*egress=*/true,
                           P.now_ns, QF);
         }
       }
@@ -1426,29 +1439,46 @@ __global__ void sub_delete_kernel(bng_sub_entry* t, uint32_t mask,
   }
 }
 
-__global__ void subnat_upsert_kernel(bng_subnat_entry* t, uint32_t mask,
-                                     const bng_subnat_entry* batch, int n,
-                                     int* rc) {
+/* Merge-upsert into the subscriber context.  update_mask selects which
+ * half changes (BNG_CTX_SET_NAT / SET_QOS / CLR_QOS / CLR_NAT) so the
+ * NAT manager and the QoS manager can each own their fields without
+ * clobbering the other's — the BPF-map analog of two maps sharing a key. */
+__global__ void subctx_upsert_kernel(bng_subctx* t, uint32_t mask,
+                                     const bng_subctx* batch, int n,
+                                     uint32_t um, int* rc) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
-  bng_subnat_entry e = batch[i];
+  bng_subctx e = batch[i];
   uint32_t slot = (uint32_t)bng_mix64(e.key_ip) & mask;
   for (int k = 0; k < BNG_MAX_PROBE; ++k) {
-    bng_subnat_entry* s = &t[(slot + k) & mask];
+    bng_subctx* s = &t[(slot + k) & mask];
     uint32_t cur = __hip_atomic_load(&s->key_ip, __ATOMIC_RELAXED,
                                      __HIP_MEMORY_SCOPE_AGENT);
     if (cur == e.key_ip || cur == 0) {
       if (cur == 0) {
+        if (!(um & (BNG_CTX_SET_NAT | BNG_CTX_SET_QOS))) {
+          if (rc) rc[i] = 0;   /* clear of an absent entry: no-op */
+          return;
+        }
         uint32_t expect = 0;
         if (!__hip_atomic_compare_exchange_strong(
                 &s->key_ip, &expect, e.key_ip, __ATOMIC_RELAXED,
                 __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT))
           continue;
       }
-      s->subscriber_id = e.subscriber_id; s->public_ip = e.public_ip;
-      s->port_start = e.port_start; s->port_end = e.port_end;
-      s->next_port = e.next_port; s->allocated_at = e.allocated_at;
-      s->block_size_log2 = e.block_size_log2; s->flags = e.flags;
+      if (um & BNG_CTX_SET_NAT) {
+        s->public_ip = e.public_ip;
+        s->port_start = e.port_start; s->port_end = e.port_end;
+        s->next_port = e.next_port; s->subscriber_id = e.subscriber_id;
+        s->flags = e.flags; s->nat_valid = 1;
+      }
+      if (um & BNG_CTX_SET_QOS) {
+        s->rate_bps = e.rate_bps; s->tokens = e.tokens;
+        s->last_update = e.last_update; s->burst_bytes = e.burst_bytes;
+        s->priority = e.priority; s->qos_valid = 1;
+      }
+      if (um & BNG_CTX_CLR_QOS) s->qos_valid = 0;
+      if (um & BNG_CTX_CLR_NAT) s->nat_valid = 0;
       if (rc) rc[i] = 0;
       return;
     }
@@ -1552,8 +1582,8 @@ __global__ void binding_delete_kernel(bng_binding_entry* t, uint32_t mask,
  * parallel; each thread owns a contiguous range of slots. */
 __global__ void nat_sweep_kernel(bng_nat_session* sessions, uint32_t n_slots,
                                  bng_nat_reverse* reverse, uint32_t rev_mask,
-                                 bng_subnat_entry* subnat,
-                                 uint32_t subnat_mask,
+                                 bng_subctx* ctx,
+                                 uint32_t ctx_mask,
                                  uint64_t now_ns, uint64_t udp_to,
                                  uint64_t tcp_est_to, uint64_t tcp_tr_to,
                                  uint64_t icmp_to,
@@ -1597,7 +1627,7 @@ __global__ void nat_sweep_kernel(bng_nat_session* sessions, uint32_t n_slots,
     if (rev)
       __hip_atomic_store(&rev->sig, BNG_KEY_TOMBSTONE, __ATOMIC_RELAXED,
                          __HIP_MEMORY_SCOPE_AGENT);
-    bng_subnat_entry* blk = subnat_lookup(subnat, subnat_mask, s->key.src_ip);
+    bng_subctx* blk = subctx_lookup(ctx, ctx_mask, s->key.src_ip);
     if (blk) atomicSub(&blk->sessions_active, 1u);
     s->ready = 0;
     __hip_atomic_store(&s->sig, BNG_KEY_TOMBSTONE, __ATOMIC_RELAXED,
@@ -1676,7 +1706,7 @@ void bng_launch_nat44(void* data, const void* in_len, void* verdict, int n,
       (uint8_t*)data, (const uint16_t*)in_len, (uint8_t*)verdict, n, stride,
       is_egress, (bng_nat_session*)sessions, sess_mask,
       (bng_nat_reverse*)reverse, rev_mask, (bng_eim_entry*)eim, eim_mask,
-      (bng_subnat_entry*)subnat, subnat_mask, (const bng_nat_config*)cfg,
+      (bng_subctx*)subnat, subnat_mask, (const bng_nat_config*)cfg,
       (const uint32_t*)hairpin, n_hairpin, (unsigned long long*)stats,
       (bng_nat_log_entry*)log_ring, (bng_ring_header*)log_hdr, now_ns);
 }
@@ -1686,7 +1716,7 @@ void bng_launch_qos(void* data, const void* in_len, void* verdict, int n,
                     void* stats, uint64_t now_ns, hipStream_t s) {
   hipLaunchKernelGGL(qos_kernel, dim3(pkt_grid(n)), dim3(256), 0, s,
       (uint8_t*)data, (const uint16_t*)in_len, (uint8_t*)verdict, n, stride,
-      is_egress, (bng_qos_bucket*)table, mask, (unsigned long long*)stats,
+      is_egress, table, mask, (unsigned long long*)stats,
       now_ns);
 }
 
@@ -1729,10 +1759,10 @@ void bng_launch_sub_delete(void* t, uint32_t mask, const void* keys, int n,
   hipLaunchKernelGGL(sub_delete_kernel, dim3((n + 255) / 256), dim3(256), 0,
       s, (bng_sub_entry*)t, mask, (const uint64_t*)keys, n);
 }
-void bng_launch_subnat_upsert(void* t, uint32_t mask, const void* batch,
-                              int n, void* rc, hipStream_t s) {
-  hipLaunchKernelGGL(subnat_upsert_kernel, dim3((n + 255) / 256), dim3(256),
-      0, s, (bng_subnat_entry*)t, mask, (const bng_subnat_entry*)batch, n,
+void bng_launch_subctx_upsert(void* t, uint32_t mask, const void* batch,
+                              int n, uint32_t um, void* rc, hipStream_t s) {
+  hipLaunchKernelGGL(subctx_upsert_kernel, dim3((n + 255) / 256), dim3(256),
+      0, s, (bng_subctx*)t, mask, (const bng_subctx*)batch, n, um,
       (int*)rc);
 }
 void bng_launch_qos_upsert(void* t, uint32_t mask, const void* batch, int n,
@@ -1760,7 +1790,7 @@ void bng_launch_nat_sweep(void* sessions, uint32_t n_slots, void* reverse,
                           hipStream_t s) {
   hipLaunchKernelGGL(nat_sweep_kernel, dim3(2048), dim3(256), 0, s,
       (bng_nat_session*)sessions, n_slots, (bng_nat_reverse*)reverse,
-      rev_mask, (bng_subnat_entry*)subnat, subnat_mask, now_ns, udp_to,
+      rev_mask, (bng_subctx*)subnat, subnat_mask, now_ns, udp_to,
       tcp_est_to, tcp_tr_to, icmp_to, (unsigned long long*)stats);
 }
 

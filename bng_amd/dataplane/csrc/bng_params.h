@@ -38,13 +38,13 @@ typedef struct bng_uplink_params {
   bng_nat_session* sessions; uint32_t sess_mask;
   bng_nat_reverse* reverse;  uint32_t rev_mask;
   bng_eim_entry* eim;        uint32_t eim_mask;
-  bng_subnat_entry* subnat;  uint32_t subnat_mask;
+  bng_subctx* subctx;        uint32_t subctx_mask;
   const bng_nat_config* ncfg;
   const uint32_t* hairpin_ips; uint32_t n_hairpin;
   unsigned long long* nat_stats;
   bng_nat_log_entry* log_ring; bng_ring_header* log_hdr;
-  /* qos ingress */
-  bng_qos_bucket* qos_in; uint32_t qos_mask;
+  /* qos: ingress lives in subctx; this is the downlink egress table */
+  bng_qos_bucket* qos_eg; uint32_t qos_eg_mask;
   unsigned long long* qos_stats;
   /* time: now_ptr (device {now_ns, now_sec}) overrides the scalars when
    * non-NULL — required under hipGraph replay, where kernel args are

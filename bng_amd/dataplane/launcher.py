@@ -66,7 +66,10 @@ class HipLauncher:
         self.sessions = z((1 << sess_log2) * 128)
         self.reverse = z((1 << sess_log2) * 48)
         self.eim = z((1 << eim_log2) * 48)
-        self.subnat = z((1 << subnat_log2) * 64)
+        # merged per-subscriber uplink context (NAT port block + ingress
+        # token bucket in one 64-B entry; subnat_log2 names the kwarg the
+        # reference's subscriber_nat map sizing maps to)
+        self.subctx = z((1 << subnat_log2) * 64)
         self.nat_cfg = z(ctypes.sizeof(abi.NatConfig))
         self.hairpin = torch.zeros(abi.MAX_HAIRPIN_IPS, dtype=torch.int32,
                                    device=self.device)
@@ -77,7 +80,6 @@ class HipLauncher:
         self.nat_log_hdr = z(16)
         self._nat_log_ridx = 0
         self.qos_egress = z((1 << qos_log2) * 64)
-        self.qos_ingress = z((1 << qos_log2) * 64)
         self.qos_stats = torch.zeros(abi.QOS_NSTATS, dtype=torch.int64,
                                      device=self.device)
         self.bindings = z((1 << binding_log2) * 32)
@@ -208,13 +210,12 @@ class HipLauncher:
                            port_start: int, port_end: int,
                            subscriber_id: int = 0):
         """ref nat/manager.go:398 AllocateNAT -> subscriber_nat map write."""
-        e = abi.SubnatEntry(key_ip=private_ip, subscriber_id=subscriber_id,
-                            public_ip=public_ip, port_start=port_start,
-                            port_end=port_end, next_port=port_start,
-                            allocated_at=time.time_ns())
+        e = abi.SubCtx(key_ip=private_ip, subscriber_id=subscriber_id,
+                       public_ip=public_ip, port_start=port_start,
+                       port_end=port_end, next_port=port_start)
         batch = self._to_dev(_struct_bytes(e))
         rc = self.torch.zeros(1, dtype=self.torch.int32, device=self.device)
-        self.ext.subnat_upsert(self.subnat, batch, rc)
+        self.ext.subctx_upsert(self.subctx, batch, abi.CTX_SET_NAT, rc)
 
     def set_hairpin_ips(self, ips: Sequence[int]):
         import numpy as np
@@ -257,7 +258,7 @@ class HipLauncher:
     def sweep_nat(self, now_ns: Optional[int] = None):
         """Expire timed-out sessions (the LRU/timeout sweeper; the reference
         gets eviction from BPF LRU maps)."""
-        self.ext.nat_sweep(self.sessions, self.reverse, self.subnat,
+        self.ext.nat_sweep(self.sessions, self.reverse, self.subctx,
                            now_ns or time.time_ns(), UDP_TIMEOUT_NS,
                            TCP_EST_TIMEOUT_NS, TCP_TRANSIENT_TIMEOUT_NS,
                            ICMP_TIMEOUT_NS, self.nat_stats)
@@ -268,20 +269,30 @@ class HipLauncher:
                        now_ns: Optional[int] = None):
         """ref qos/manager.go:248 SetSubscriberPolicy."""
         now = now_ns if now_ns is not None else time.time_ns()
-        b = abi.QosBucket(key_ip=ip, valid=1, priority=priority,
-                          rate_bps=rate_bps, tokens=burst_bytes,
-                          last_update=now, burst_bytes=burst_bytes)
-        table = self.qos_egress if direction == "egress" else self.qos_ingress
-        batch = self._to_dev(_struct_bytes(b))
         rc = self.torch.zeros(1, dtype=self.torch.int32, device=self.device)
-        self.ext.qos_upsert(table, batch, rc)
+        if direction == "egress":
+            b = abi.QosBucket(key_ip=ip, valid=1, priority=priority,
+                              rate_bps=rate_bps, tokens=burst_bytes,
+                              last_update=now, burst_bytes=burst_bytes)
+            self.ext.qos_upsert(self.qos_egress, self._to_dev(
+                _struct_bytes(b)), rc)
+        else:
+            e = abi.SubCtx(key_ip=ip, priority=priority, rate_bps=rate_bps,
+                           tokens=burst_bytes, last_update=now,
+                           burst_bytes=burst_bytes)
+            self.ext.subctx_upsert(self.subctx, self._to_dev(
+                _struct_bytes(e)), abi.CTX_SET_QOS, rc)
 
     def remove_qos_policy(self, ip: int, direction: str = "egress"):
-        b = abi.QosBucket(key_ip=ip, valid=0)
-        table = self.qos_egress if direction == "egress" else self.qos_ingress
-        batch = self._to_dev(_struct_bytes(b))
         rc = self.torch.zeros(1, dtype=self.torch.int32, device=self.device)
-        self.ext.qos_upsert(table, batch, rc)
+        if direction == "egress":
+            b = abi.QosBucket(key_ip=ip, valid=0)
+            self.ext.qos_upsert(self.qos_egress, self._to_dev(
+                _struct_bytes(b)), rc)
+        else:
+            e = abi.SubCtx(key_ip=ip)
+            self.ext.subctx_upsert(self.subctx, self._to_dev(
+                _struct_bytes(e)), abi.CTX_CLR_QOS, rc)
 
     def qos_get_stats(self) -> Dict[str, int]:
         return dict(zip(abi.QOS_STAT_NAMES, self.qos_stats.cpu().tolist()))
@@ -361,7 +372,7 @@ class HipLauncher:
         verdict = self.torch.zeros(n, dtype=self.torch.uint8,
                                    device=self.device)
         self.ext.nat44(data, lens, verdict, egress, self.sessions,
-                       self.reverse, self.eim, self.subnat, self.nat_cfg,
+                       self.reverse, self.eim, self.subctx, self.nat_cfg,
                        self.hairpin, self.n_hairpin, self.nat_stats,
                        self.nat_log_ring, self.nat_log_hdr,
                        now_ns if now_ns is not None else time.time_ns())
@@ -372,7 +383,8 @@ class HipLauncher:
         n = lens.numel()
         verdict = self.torch.zeros(n, dtype=self.torch.uint8,
                                    device=self.device)
-        table = self.qos_egress if egress else self.qos_ingress
+        # ingress policies live in the merged subscriber context
+        table = self.qos_egress if egress else self.subctx
         self.ext.qos(data, lens, verdict, egress, table, self.qos_stats,
                      now_ns if now_ns is not None else time.time_ns())
         return verdict
@@ -407,9 +419,9 @@ class HipLauncher:
             data, lens, out_len, verdict, self.subs, self.pools,
             self.server_cfg, self.dhcp_stats, self.bindings, self.as_cfg,
             self.as_stats, self.spoof_ring, self.spoof_hdr, self.sessions,
-            self.reverse, self.eim, self.subnat, self.nat_cfg, self.hairpin,
+            self.reverse, self.eim, self.subctx, self.nat_cfg, self.hairpin,
             self.n_hairpin, self.nat_stats, self.nat_log_ring,
-            self.nat_log_hdr, self.qos_ingress, self.qos_stats, now,
+            self.nat_log_hdr, self.qos_egress, self.qos_stats, now,
             now_sec if now_sec is not None else now // 10**9, order=order)
         return verdict, out_len
 
@@ -422,7 +434,7 @@ class HipLauncher:
             data, lens, out_len, verdict, self.subs, self.pools,
             self.server_cfg, self.dhcp_stats, self.bindings, self.as_cfg,
             self.as_stats, self.spoof_ring, self.spoof_hdr, self.sessions,
-            self.reverse, self.eim, self.subnat, self.nat_cfg, self.hairpin,
+            self.reverse, self.eim, self.subctx, self.nat_cfg, self.hairpin,
             self.n_hairpin, self.nat_stats, self.nat_log_ring,
             self.nat_log_hdr, self.qos_egress, self.qos_stats, now,
             now // 10**9, order=None, downlink=True)
@@ -716,9 +728,9 @@ class CapturedUplink:
                 self.work, self.lens, self.out_len, self.verdict, L.subs,
                 L.pools, L.server_cfg, L.dhcp_stats, L.bindings, L.as_cfg,
                 L.as_stats, L.spoof_ring, L.spoof_hdr, L.sessions,
-                L.reverse, L.eim, L.subnat, L.nat_cfg, L.hairpin,
+                L.reverse, L.eim, L.subctx, L.nat_cfg, L.hairpin,
                 L.n_hairpin, L.nat_stats, L.nat_log_ring, L.nat_log_hdr,
-                L.qos_ingress, L.qos_stats, 0, 0, order=order,
+                L.qos_egress, L.qos_stats, 0, 0, order=order,
                 downlink=False, now_buf=self.now_buf)
 
         # warmup on a side stream, then capture

@@ -44,9 +44,10 @@ def test_extension_layout_matches_ctypes():
     assert off["nat_session.last_seen"] == abi.NatSession.last_seen.offset
     assert off["nat_session.ready"] == abi.NatSession.ready.offset
     assert off["eim_entry.created"] == abi.EimEntry.created.offset
-    assert off["subnat_entry.next_port"] == abi.SubnatEntry.next_port.offset
-    assert off["subnat_entry.sessions_active"] == \
-        abi.SubnatEntry.sessions_active.offset
+    assert off["subctx.rate_bps"] == abi.SubCtx.rate_bps.offset
+    assert off["subctx.next_port"] == abi.SubCtx.next_port.offset
+    assert off["subctx.sessions_active"] == \
+        abi.SubCtx.sessions_active.offset
     assert off["qos_bucket.tokens"] == abi.QosBucket.tokens.offset
     assert off["qos_bucket.last_update"] == abi.QosBucket.last_update.offset
     assert off["binding_entry.ipv6_addr"] == abi.BindingEntry.ipv6_addr.offset
