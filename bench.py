@@ -194,7 +194,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--batch", type=int, default=262144,
+    ap.add_argument("--batch", type=int, default=1048576,
                     help="packets injected per GPU per step")
     ap.add_argument("--subs", type=int, default=1_000_000)
     ap.add_argument("--dhcp-frac", type=float, default=0.1)

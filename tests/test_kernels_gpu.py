@@ -652,3 +652,60 @@ class TestCRUDUnderTraffic:
         d, l = gpu.make_batch(frames)
         v, _ = gpu.dhcp_fastpath(d, l, NOW_SEC)
         assert all(x == TX for x in v.cpu().tolist())
+
+
+class TestSubCtxMerge:
+    """The merged 64-B subscriber context (bng_subctx) carries both the
+    NAT port block and the ingress token bucket; each manager's upsert
+    must leave the other half intact (merge-mask semantics)."""
+
+    def _data_pkt(self, sport=40001):
+        return build_ipv4("aa:00:00:00:00:01", "02:00:00:00:00:01",
+                          ip2u32(PRIV), ip2u32(DST), proto=17,
+                          sport=sport, dport=53)
+
+    def test_qos_update_preserves_nat_half(self):
+        gpu, cpu = make_pair()
+        for l in (gpu, cpu):
+            l.add_subscriber_nat(ip2u32(PRIV), ip2u32(PUB), 1024, 2047,
+                                 subscriber_id=7)
+        d, lns = gpu.make_batch([self._data_pkt()], stride=128)
+        assert gpu.nat44(d, lns, egress=True,
+                         now_ns=NOW_NS).cpu().tolist() == [FWD]
+        # install + then remove an ingress policy; the port block survives
+        gpu.set_qos_policy(ip2u32(PRIV), 10**9, 1 << 20,
+                           direction="ingress", now_ns=NOW_NS)
+        gpu.remove_qos_policy(ip2u32(PRIV), direction="ingress")
+        d, lns = gpu.make_batch([self._data_pkt(40002)], stride=128)
+        assert gpu.nat44(d, lns, egress=True,
+                         now_ns=NOW_NS).cpu().tolist() == [FWD]
+        import struct as st
+        pkt = d.cpu().numpy()[0]
+        assert st.unpack_from(">I", pkt, 26)[0] == ip2u32(PUB)  # SNAT held
+
+    def test_nat_update_preserves_qos_half(self):
+        gpu, _ = make_pair()
+        ip = ip2u32(PRIV)
+        # tiny bucket: 10-byte burst -> a 64-B packet always drops
+        gpu.set_qos_policy(ip, 8, 10, direction="ingress", now_ns=NOW_NS)
+        d, lns = gpu.make_batch([self._data_pkt()], stride=128)
+        assert gpu.qos(d, lns, egress=False,
+                       now_ns=NOW_NS).cpu().tolist() == [DROP]
+        # NAT manager writes its half; the throttle must still bite
+        gpu.add_subscriber_nat(ip, ip2u32(PUB), 1024, 2047, subscriber_id=9)
+        d, lns = gpu.make_batch([self._data_pkt(40003)], stride=128)
+        assert gpu.uplink(d, lns, now_ns=NOW_NS + 1,
+                          now_sec=NOW_SEC)[0].cpu().tolist() == [DROP]
+
+    def test_qos_only_entry_passes_nat_stage(self):
+        """An entry with only a QoS half (no port block) must behave like
+        the reference's missing subscriber_nat entry: XDP_PASS to the
+        slow path, not a bogus SNAT."""
+        gpu, _ = make_pair()
+        ip = ip2u32(PRIV)
+        gpu.set_qos_policy(ip, 0, 0, direction="ingress", now_ns=NOW_NS)
+        d, lns = gpu.make_batch([self._data_pkt()], stride=128)
+        assert gpu.nat44(d, lns, egress=True,
+                         now_ns=NOW_NS).cpu().tolist() == [PASS]
+        st_ = gpu.nat_get_stats()
+        assert st_["packets_passed"] == 1 and st_["packets_snat"] == 0
