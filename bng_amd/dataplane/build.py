@@ -34,11 +34,15 @@ def build(verbose: bool = False, force: bool = False) -> str:
     os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
     os.makedirs(BUILD_DIR, exist_ok=True)
     from torch.utils.cpp_extension import load
+    # experiment toggles (A/B on the GPU box without editing sources)
+    defs = []
+    if os.environ.get("BNG_SC1_PROBES") in ("0", "1"):
+        defs.append("-DBNG_SC1_PROBES=" + os.environ["BNG_SC1_PROBES"])
     mod = load(
         name="bng_dataplane_C",
         sources=SOURCES,
-        extra_cflags=["-O3", "-std=c++17"],
-        extra_cuda_cflags=["-O3", "-std=c++17"],
+        extra_cflags=["-O3", "-std=c++17"] + defs,
+        extra_cuda_cflags=["-O3", "-std=c++17"] + defs,
         build_directory=BUILD_DIR,
         verbose=verbose,
         is_python_module=False,

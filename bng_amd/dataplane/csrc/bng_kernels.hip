@@ -37,6 +37,26 @@ BNG_DEV void rlx_store64(void* ptr, uint64_t v) {
                      __HIP_MEMORY_SCOPE_AGENT);
 }
 
+/* Table-probe load: 16 B of a hash-table entry.  BNG_SC1_PROBES=1 issues
+ * it as two relaxed agent-scope (sc1) loads — L2-served, bypassing L1 —
+ * because random probes have ~0 L1 hit rate and only evict the packet
+ * lines the parser needs (same reasoning as the session hit path). */
+#ifndef BNG_SC1_PROBES
+#define BNG_SC1_PROBES 1
+#endif
+BNG_DEV uint4 ld_probe16(const void* p) {
+#if BNG_SC1_PROBES
+  const uint64_t* q = (const uint64_t*)p;
+  uint64_t a = rlx_load64(q), b = rlx_load64(q + 1);
+  uint4 v;
+  v.x = (uint32_t)a; v.y = (uint32_t)(a >> 32);
+  v.z = (uint32_t)b; v.w = (uint32_t)(b >> 32);
+  return v;
+#else
+  return *(const uint4*)p;
+#endif
+}
+
 BNG_DEV const bng_sub_entry* sub_lookup(const bng_sub_entry* t, uint32_t mask,
                                         uint64_t key) {
   uint32_t slot = (uint32_t)bng_mix64(key) & mask;
@@ -44,7 +64,7 @@ BNG_DEV const bng_sub_entry* sub_lookup(const bng_sub_entry* t, uint32_t mask,
     const bng_sub_entry* e = &t[(slot + i) & mask];
     /* one 16-B vector load covers key+pool_id+allocated_ip (entry is
      * 32-B aligned); most probes end after this single transaction */
-    uint4 v = *(const uint4*)e;
+    uint4 v = ld_probe16(e);
     uint64_t k = ((uint64_t)v.y << 32) | v.x;
     if (k == key) return e;
     if (k == BNG_KEY_EMPTY) return nullptr;
@@ -63,7 +83,7 @@ BNG_DEV bng_subctx* subctx_lookup_hint(bng_subctx* t,
   if (first.x == 0) return nullptr;
   for (int i = 1; i < BNG_MAX_PROBE; ++i) {
     bng_subctx* e = &t[(slot + i) & mask];
-    uint4 v = *(const uint4*)e;
+    uint4 v = ld_probe16(e);
     if (v.x == ip) return e;
     if (v.x == 0) return nullptr;
   }
@@ -76,7 +96,7 @@ BNG_DEV bng_subctx* subctx_lookup(bng_subctx* t, uint32_t mask,
   uint32_t slot = (uint32_t)bng_mix64(ip) & mask;
   for (int i = 0; i < BNG_MAX_PROBE; ++i) {
     bng_subctx* e = &t[(slot + i) & mask];
-    uint4 v = *(const uint4*)e;    /* key + pub_ip + ports + valid flags */
+    uint4 v = ld_probe16(e);    /* key + pub_ip + ports + valid flags */
     if (v.x == ip) return e;
     if (v.x == 0) return nullptr;
   }
@@ -93,7 +113,7 @@ BNG_DEV bng_qos_bucket* qos_lookup_hint(bng_qos_bucket* t, uint32_t mask,
   if (first.x == 0) return nullptr;
   for (int i = 1; i < BNG_MAX_PROBE; ++i) {
     bng_qos_bucket* e = &t[(slot + i) & mask];
-    uint4 v = *(const uint4*)e;
+    uint4 v = ld_probe16(e);
     if (v.x == ip && (v.y & 0xFF)) {
       *rate_out = ((uint64_t)v.w << 32) | v.z;
       return e;
@@ -109,7 +129,7 @@ BNG_DEV bng_qos_bucket* qos_lookup(bng_qos_bucket* t, uint32_t mask,
   uint32_t slot = (uint32_t)bng_mix64(ip) & mask;
   for (int i = 0; i < BNG_MAX_PROBE; ++i) {
     bng_qos_bucket* e = &t[(slot + i) & mask];
-    uint4 v = *(const uint4*)e;    /* key_ip+valid+prio | rate in 16B */
+    uint4 v = ld_probe16(e);    /* key_ip+valid+prio | rate in 16B */
     if (v.x == ip && (v.y & 0xFF)) {
       *rate_out = ((uint64_t)v.w << 32) | v.z;
       return e;
@@ -127,7 +147,7 @@ BNG_DEV const bng_binding_entry* binding_lookup_hint(
   if (k == BNG_KEY_EMPTY) return nullptr;
   for (int i = 1; i < BNG_MAX_PROBE; ++i) {
     const bng_binding_entry* e = &t[(slot + i) & mask];
-    uint4 v = *(const uint4*)e;
+    uint4 v = ld_probe16(e);
     k = ((uint64_t)v.y << 32) | v.x;
     if (k == mac) return e;
     if (k == BNG_KEY_EMPTY) return nullptr;
@@ -141,7 +161,7 @@ BNG_DEV const bng_binding_entry* binding_lookup(const bng_binding_entry* t,
   uint32_t slot = (uint32_t)bng_mix64(mac) & mask;
   for (int i = 0; i < BNG_MAX_PROBE; ++i) {
     const bng_binding_entry* e = &t[(slot + i) & mask];
-    uint4 v = *(const uint4*)e;    /* key + ipv4 + flags in one load */
+    uint4 v = ld_probe16(e);    /* key + ipv4 + flags in one load */
     uint64_t k = ((uint64_t)v.y << 32) | v.x;
     if (k == mac) return e;
     if (k == BNG_KEY_EMPTY) return nullptr;
@@ -1270,8 +1290,8 @@ void uplink_pipeline_kernel(bng_uplink_params P) {
         for (int j = 0; j < 6; ++j) mac = (mac << 8) | p[6 + j];
         uint32_t s0 = (uint32_t)bng_mix64(mac) & P.bmask;
         uint32_t s1 = (uint32_t)bng_mix64(c.saddr) & P.subctx_mask;
-        uint4 f0 = *(const uint4*)&P.bindings[s0];
-        uint4 f1 = *(const uint4*)&P.subctx[s1];
+        uint4 f0 = ld_probe16(&P.bindings[s0]);
+        uint4 f1 = ld_probe16(&P.subctx[s1]);
         const bng_binding_entry* b =
             binding_lookup_hint(P.bindings, P.bmask, mac, s0, f0);
         v = antispoof_process_with(p, len, b, P.acfg, P.spoof_ring,
