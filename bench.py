@@ -5,11 +5,21 @@ Measures the BASELINE.json headline: Mpps + p50 DHCP OFFER latency on a
 64B IPv4 mix with a 1M-subscriber table, at 1..8 MI355X GPUs.
 
 One step = ingest one fixed-size synthetic batch per GPU (restore from
-pristine = the RX-DMA analog), steer each packet to its owning shard via
-RCCL all-to-all over xGMI (world>1), and run the fused uplink pipeline
-(DHCP fast path for UDP:67; antispoof -> NAT44 SNAT -> QoS for data
-packets) on the local shard's HBM tables.  Weak scaling: per-GPU injected
-batch is fixed as N grows.
+pristine = the RX-DMA analog), steer mis-delivered packets to their
+owning shard via RCCL all-to-all over xGMI (world>1), and run the fused
+uplink pipeline (DHCP fast path for UDP:67; antispoof -> NAT44 SNAT ->
+QoS for data packets) on the local shard's HBM tables.  Weak scaling:
+per-GPU injected batch is fixed as N grows.
+
+Arrival model at world>1 mirrors the reference's NIC edge: RSS steers
+each data flow to the core owning its subscriber (per-core XDP never
+ships packets cross-core, loader.go RSS assumption), so data packets
+arrive at their IP-shard owner; DHCP broadcasts cannot be RSS-steered by
+subscriber MAC and arrive anywhere — the dhcp_frac slice crosses xGMI to
+its MAC-shard owner each step.  --steer-all instead routes EVERY packet
+through the all-to-all (full-shuffle stress mode).  RX copy, steering
+exchange and type-sort for batch k+1 all overlap batch k's pipeline
+kernel on a second HIP stream.
 
 Traffic mix (configurable): 90% 64-byte UDP IPv4 data packets from
 subscriber IPs (SNAT + QoS + antispoof path), 10% DHCP DISCOVER/REQUEST
@@ -131,11 +141,27 @@ def build_tables(launcher, rank: int, world: int, n_subs: int, now_sec: int):
 
 
 def gen_batch(n_pkts: int, n_subs: int, dhcp_frac: float, stride: int,
-              seed: int):
-    """Vectorized synthetic batch: [n,stride] uint8 + lens int16."""
+              seed: int, rank: int = 0, world: int = 1,
+              steer_all: bool = False):
+    """Vectorized synthetic batch: [n,stride] uint8 + lens int16.
+
+    world>1 models the NIC edge the reference also relies on: RSS steers
+    each data flow to the core/GPU owning its subscriber (per-core XDP
+    never ships packets cross-core), so data-packet subscribers are drawn
+    from THIS rank's IP shard.  DHCP broadcasts cannot be RSS-steered to
+    the MAC owner and arrive anywhere — drawn from ALL subscribers; the
+    xGMI all-to-all steers those strays.  steer_all=True instead draws
+    everything uniformly (the full-shuffle stress mode)."""
     rng = np.random.default_rng(seed)
     idx = rng.integers(0, n_subs, size=n_pkts, dtype=np.uint64)
     is_dhcp = rng.random(n_pkts) < dhcp_frac
+    if world > 1 and not steer_all:
+        all_ips = (np.uint64(ip2u32("10.0.0.0") + 2) +
+                   np.arange(n_subs, dtype=np.uint64))
+        local = np.nonzero(mix64_np(all_ips) % np.uint64(world) ==
+                           np.uint64(rank))[0].astype(np.uint64)
+        idx = np.where(is_dhcp, idx,
+                       local[rng.integers(0, len(local), size=n_pkts)])
 
     data = np.zeros((n_pkts, stride), dtype=np.uint8)
     lens = np.zeros(n_pkts, dtype=np.uint16)
@@ -205,8 +231,12 @@ def main():
     ap.add_argument("--no-sort", action="store_true",
                     help="disable on-device type-sort (wave-divergence fix)")
     ap.add_argument("--no-overlap", action="store_true",
-                    help="disable RX-copy/sort overlap with the previous "
-                         "batch's pipeline kernel")
+                    help="disable RX-copy/steer/sort overlap with the "
+                         "previous batch's pipeline kernel")
+    ap.add_argument("--steer-all", action="store_true",
+                    help="world>1: draw data traffic uniformly (every "
+                         "packet crosses xGMI) instead of the RSS-steered "
+                         "arrival model")
     args = ap.parse_args()
 
     import torch
@@ -241,12 +271,13 @@ def main():
 
     t0 = time.perf_counter()
     data_np, lens_np = gen_batch(args.batch, args.subs, args.dhcp_frac,
-                                 args.stride, seed=1234 + rank)
+                                 args.stride, seed=1234 + rank, rank=rank,
+                                 world=world, steer_all=args.steer_all)
     pristine = torch.from_numpy(data_np).to(device)
     lens = torch.from_numpy(lens_np.view(np.int16)).to(device)
     log(rank, f"[bench] batch generated in {time.perf_counter() - t0:.1f}s")
 
-    overlap = not args.no_overlap and not distributed
+    overlap = not args.no_overlap
     nbuf = 2 if overlap else 1
     works = [torch.empty_like(pristine) for _ in range(nbuf)]
     clss = [torch.empty(args.batch, dtype=torch.uint8, device=device)
@@ -257,34 +288,58 @@ def main():
     # while the main-stream kernel still reads it)
     orders = [torch.zeros(args.batch, dtype=torch.int32, device=device)
               for _ in range(nbuf)]
-    prep_stream = torch.cuda.Stream(device=device) if overlap else None
+    prep_stream = torch.cuda.Stream(device=device) if overlap else None  # noqa: E501
     prep_done = [torch.cuda.Event() for _ in range(nbuf)]
     work_free = [torch.cuda.Event() for _ in range(nbuf)]
     for ev in work_free:
         ev.record()
 
+    batches = [None] * nbuf
+
     def prep(k):
-        """RX copy + classify + type-sort for step k — on the prep
-        stream, overlapped with step k-1's pipeline kernel."""
+        """RX copy (+ xGMI steering of DHCP strays when world>1) +
+        classify + type-sort for step k — on the prep stream, overlapped
+        with step k-1's pipeline kernel (and its collectives)."""
         b = k % nbuf
         with torch.cuda.stream(prep_stream):
             prep_stream.wait_event(work_free[b])
             works[b].copy_(pristine)               # RX-DMA analog
+            d, l = works[b], lens
+            if distributed:
+                owner = launcher.shard_owner(d, l, world)
+                d, l = exchange(d, l, owner)
+            o = None
             if not args.no_sort:
-                launcher.ext.pkt_class(works[b], lens, clss[b])
-                idx = torch.argsort(clss[b], stable=True)
-                orders[b].copy_(idx.to(torch.int32))
+                if d is works[b]:
+                    launcher.ext.pkt_class(d, l, clss[b])
+                    idx = torch.argsort(clss[b], stable=True)
+                    orders[b].copy_(idx.to(torch.int32))
+                    o = orders[b]
+                else:
+                    # exchanged batch: size varies per rank; fresh
+                    # prep-stream tensors, released to the main stream
+                    # via record_stream at consumption
+                    cls = torch.empty(l.numel(), dtype=torch.uint8,
+                                      device=device)
+                    launcher.ext.pkt_class(d, l, cls)
+                    o = torch.argsort(cls, stable=True).to(torch.int32)
+            batches[b] = (d, l, o)
             prep_done[b].record(prep_stream)
 
     def step(now_ns, k=0):
+        cur = torch.cuda.current_stream(device)
         if overlap:
             b = k % nbuf
-            torch.cuda.current_stream(device).wait_event(prep_done[b])
-            launcher.uplink(works[b], lens, now_ns=now_ns,
-                            now_sec=now_sec,
-                            sort_by_type=not args.no_sort,
-                            order=None if args.no_sort else orders[b])
-            work_free[b].record(torch.cuda.current_stream(device))
+            cur.wait_event(prep_done[b])
+            d, l, o = batches[b]
+            if d is not works[b]:                  # prep-stream allocs
+                d.record_stream(cur)
+                l.record_stream(cur)
+                if o is not None:
+                    o.record_stream(cur)
+            launcher.uplink(d, l, now_ns=now_ns, now_sec=now_sec,
+                            sort_by_type=not args.no_sort, order=o)
+            work_free[b].record(cur)
             prep(k + 1)                            # overlap next batch
             return
         works[0].copy_(pristine)                   # RX-DMA analog
@@ -370,7 +425,9 @@ def main():
                              "1M-sub table at 1/2/4/8 MI355X",
                 "global_batch": world * args.batch,
                 "seq_len": args.stride,
-                "parallelism": f"shard{world}-hashring-alltoall",
+                "parallelism": f"shard{world}-rss" +
+                               ("-steerall" if args.steer_all else
+                                "-dhcp-alltoall"),
                 "n_subscribers": args.subs,
                 "dhcp_frac": args.dhcp_frac,
                 "p50_dhcp_offer_us": None if p50_us is None

@@ -98,3 +98,30 @@ def test_exchange_gloo_world2(tmp_path):
     all_pkts = [p for r in results.values() for p in r]
     assert len(all_pkts) == world * 64
     assert len(set(all_pkts)) == world * 64
+
+
+def test_bench_rss_arrival_model():
+    """bench.gen_batch(world>1): data packets must arrive pre-steered to
+    their IP-shard owner (the NIC-RSS analog); DHCP may land anywhere."""
+    import numpy as np
+    import bench
+    from bng_amd.dataplane.packets import ip2u32
+    world = 4
+    for rank in range(2):
+        data, lens = bench.gen_batch(4096, 100_000, 0.1, 512,
+                                     seed=5, rank=rank, world=world)
+        is_dhcp = lens > 64
+        ips = data[:, 26:30].astype(np.uint32)
+        ip_u32 = ((ips[:, 0] << 24) | (ips[:, 1] << 16) |
+                  (ips[:, 2] << 8) | ips[:, 3]).astype(np.uint64)
+        owners = bench.mix64_np(ip_u32) % np.uint64(world)
+        assert (owners[~is_dhcp] == rank).all()
+        # full-shuffle mode really shuffles
+        data2, lens2 = bench.gen_batch(4096, 100_000, 0.0, 512,
+                                       seed=5, rank=rank, world=world,
+                                       steer_all=True)
+        ips2 = data2[:, 26:30].astype(np.uint32)
+        ip2_u32 = ((ips2[:, 0] << 24) | (ips2[:, 1] << 16) |
+                   (ips2[:, 2] << 8) | ips2[:, 3]).astype(np.uint64)
+        own2 = bench.mix64_np(ip2_u32) % np.uint64(world)
+        assert len(set(own2.tolist())) == world
