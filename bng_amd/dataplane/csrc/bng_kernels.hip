@@ -207,6 +207,60 @@ BNG_DEV E* sig_lookup(E* t, uint32_t mask, uint64_t sig) {
   return nullptr;
 }
 
+/* Session find-or-claim with a WIDENED first probe: sig + key +
+ * translation + ready (5 independent sc1 loads of the entry's first
+ * 64-B line) issue together, so a first-slot hit — the common case at
+ * <=50% load — costs ONE L2 round trip instead of a dependent
+ * sig-then-fields pair.  Validity still gates on the ready byte and the
+ * tuple compare (the preload changes no ordering: sig is the CLAIM
+ * marker, ready the PUBLISH marker, exactly as in sig_find_or_claim). */
+struct sess_pre { uint64_t k0, k1, t0, t1; bool pre; };
+
+BNG_DEV bng_nat_session* sess_find_or_claim(bng_nat_session* t,
+                                            uint32_t mask, uint64_t sig,
+                                            bool* claimed, bool* found,
+                                            sess_pre* h) {
+  *claimed = false; *found = false; h->pre = false;
+  uint32_t slot = (uint32_t)sig & mask;
+  {
+    bng_nat_session* e = &t[slot];
+    const uint64_t* e64 = (const uint64_t*)e;
+    uint64_t k  = rlx_load64(e64 + 0);
+    uint64_t k0 = rlx_load64(e64 + 1);
+    uint64_t k1 = rlx_load64(e64 + 2);
+    uint64_t t0 = rlx_load64(e64 + 3);
+    uint64_t t1 = rlx_load64(e64 + 4);
+    if (k == sig) {
+      *found = true;
+      h->pre = true; h->k0 = k0; h->k1 = k1; h->t0 = t0; h->t1 = t1;
+      return e;
+    }
+    if (k == BNG_KEY_EMPTY) {
+      uint64_t expect = BNG_KEY_EMPTY;
+      uint64_t old = __hip_atomic_compare_exchange_strong(
+          &e->sig, &expect, sig, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+          __HIP_MEMORY_SCOPE_AGENT) ? BNG_KEY_EMPTY : expect;
+      if (old == BNG_KEY_EMPTY) { *claimed = true; return e; }
+      if (old == sig) { *found = true; return e; }
+      /* lost to a different key: keep probing */
+    }
+  }
+  for (int i = 1; i < BNG_MAX_PROBE; ++i) {
+    bng_nat_session* e = &t[(slot + i) & mask];
+    uint64_t k = __hip_atomic_load(&e->sig, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+    if (k == sig) { *found = true; return e; }
+    if (k == BNG_KEY_EMPTY) {
+      uint64_t old = __hip_atomic_compare_exchange_strong(
+          &e->sig, &k, sig, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+          __HIP_MEMORY_SCOPE_AGENT) ? BNG_KEY_EMPTY : k;
+      if (old == BNG_KEY_EMPTY) { *claimed = true; return e; }
+      if (old == sig) { *found = true; return e; }
+    }
+  }
+  return nullptr;
+}
+
 BNG_DEV bool tuple_eq(const bng_nat_tuple& a, const bng_nat_tuple& b) {
   return a.src_ip == b.src_ip && a.dst_ip == b.dst_ip &&
          a.src_port == b.src_port && a.dst_port == b.dst_port &&
@@ -702,8 +756,9 @@ BNG_DEV int nat_egress_process(pktctx& c, const nat_tables& T, nat_flags& F,
 
   uint64_t sig = bng_tuple_sig(c.saddr, c.daddr, sport, dport, c.proto);
   bool claimed, found;
-  bng_nat_session* sess = sig_find_or_claim(T.sessions, T.sess_mask, sig,
-                                            &claimed, &found);
+  sess_pre H;
+  bng_nat_session* sess = sess_find_or_claim(T.sessions, T.sess_mask, sig,
+                                             &claimed, &found, &H);
   if (!sess) { F.passed = true; return BNG_PASS; }  /* table section full */
 
   uint32_t nat_ip; uint16_t nat_port;
@@ -712,12 +767,17 @@ BNG_DEV int nat_egress_process(pktctx& c, const nat_tables& T, nat_flags& F,
      * agent release (L2 write-back) before the sc1 `ready` store, so
      * relaxed agent (sc1, L2-served) loads are always fresh — no
      * acquire fence (which would flush this CU's whole L1) needed.
-     * 4 independent u64 loads cover key + translation + ready. */
+     * On a first-slot hit the 4 words arrived WITH the sig probe. */
     const uint64_t* e64 = (const uint64_t*)sess;
-    uint64_t k0 = rlx_load64(e64 + 1);   /* src_ip | dst_ip<<32 */
-    uint64_t k1 = rlx_load64(e64 + 2);   /* sport | dport<<16 | proto<<32 */
-    uint64_t t0 = rlx_load64(e64 + 3);   /* nat_ip | nat_port<<32 | orig_port<<48 */
-    uint64_t t1 = rlx_load64(e64 + 4);   /* orig_ip | state<<32 | hairpin<<40 | ready<<48 */
+    uint64_t k0, k1, t0, t1;
+    if (H.pre) {
+      k0 = H.k0; k1 = H.k1; t0 = H.t0; t1 = H.t1;
+    } else {
+      k0 = rlx_load64(e64 + 1);   /* src_ip | dst_ip<<32 */
+      k1 = rlx_load64(e64 + 2);   /* sport | dport<<16 | proto<<32 */
+      t0 = rlx_load64(e64 + 3);   /* nat_ip | nat_port<<32 | orig_port<<48 */
+      t1 = rlx_load64(e64 + 4);   /* orig_ip | state<<32 | hairpin<<40 | ready<<48 */
+    }
     if (!((t1 >> 48) & 0xFF)) {          /* creation still in flight */
       if (!bng_wait_ready(&sess->ready, 8192)) {
         F.passed = true; return BNG_PASS;
