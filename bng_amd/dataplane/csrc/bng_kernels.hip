@@ -103,26 +103,6 @@ BNG_DEV bng_subctx* subctx_lookup(bng_subctx* t, uint32_t mask,
   return nullptr;
 }
 
-BNG_DEV bng_qos_bucket* qos_lookup_hint(bng_qos_bucket* t, uint32_t mask,
-                                        uint32_t ip, uint64_t* rate_out,
-                                        uint32_t slot, uint4 first) {
-  if (first.x == ip && (first.y & 0xFF)) {
-    *rate_out = ((uint64_t)first.w << 32) | first.z;
-    return &t[slot];
-  }
-  if (first.x == 0) return nullptr;
-  for (int i = 1; i < BNG_MAX_PROBE; ++i) {
-    bng_qos_bucket* e = &t[(slot + i) & mask];
-    uint4 v = ld_probe16(e);
-    if (v.x == ip && (v.y & 0xFF)) {
-      *rate_out = ((uint64_t)v.w << 32) | v.z;
-      return e;
-    }
-    if (v.x == 0) return nullptr;
-  }
-  return nullptr;
-}
-
 BNG_DEV bng_qos_bucket* qos_lookup(bng_qos_bucket* t, uint32_t mask,
                                    uint32_t ip, uint64_t* rate_out) {
   if (ip == 0) return nullptr;
