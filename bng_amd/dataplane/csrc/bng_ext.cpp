@@ -142,10 +142,10 @@ void uplink_pipeline(torch::Tensor data, torch::Tensor in_len,
                      torch::Tensor as_stats, torch::Tensor spoof_ring,
                      torch::Tensor spoof_hdr, torch::Tensor sessions,
                      torch::Tensor reverse, torch::Tensor eim,
-                     torch::Tensor subnat, torch::Tensor ncfg,
+                     torch::Tensor subctx, torch::Tensor ncfg,
                      torch::Tensor hairpin, int64_t n_hairpin,
                      torch::Tensor nat_stats, torch::Tensor log_ring,
-                     torch::Tensor log_hdr, torch::Tensor qos_in,
+                     torch::Tensor log_hdr, torch::Tensor qos_eg,
                      torch::Tensor qos_stats, int64_t now_ns,
                      int64_t now_sec,
                      c10::optional<torch::Tensor> order,
@@ -181,16 +181,16 @@ void uplink_pipeline(torch::Tensor data, torch::Tensor in_len,
   P.rev_mask = table_mask(reverse, sizeof(bng_nat_reverse), "reverse");
   P.eim = (bng_eim_entry*)eim.data_ptr();
   P.eim_mask = table_mask(eim, sizeof(bng_eim_entry), "eim");
-  P.subctx = (bng_subctx*)subnat.data_ptr();
-  P.subctx_mask = table_mask(subnat, sizeof(bng_subctx), "subctx");
+  P.subctx = (bng_subctx*)subctx.data_ptr();
+  P.subctx_mask = table_mask(subctx, sizeof(bng_subctx), "subctx");
   P.ncfg = (const bng_nat_config*)ncfg.data_ptr();
   P.hairpin_ips = (const uint32_t*)hairpin.data_ptr();
   P.n_hairpin = (uint32_t)n_hairpin;
   P.nat_stats = (unsigned long long*)nat_stats.data_ptr();
   P.log_ring = (bng_nat_log_entry*)log_ring.data_ptr();
   P.log_hdr = (bng_ring_header*)log_hdr.data_ptr();
-  P.qos_eg = (bng_qos_bucket*)qos_in.data_ptr();
-  P.qos_eg_mask = table_mask(qos_in, sizeof(bng_qos_bucket), "qos");
+  P.qos_eg = (bng_qos_bucket*)qos_eg.data_ptr();
+  P.qos_eg_mask = table_mask(qos_eg, sizeof(bng_qos_bucket), "qos");
   P.qos_stats = (unsigned long long*)qos_stats.data_ptr();
   P.now_ns = (uint64_t)now_ns;
   P.now_sec = (uint64_t)now_sec;
@@ -321,9 +321,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("scfg"), py::arg("dhcp_stats"), py::arg("bindings"),
         py::arg("acfg"), py::arg("as_stats"), py::arg("spoof_ring"),
         py::arg("spoof_hdr"), py::arg("sessions"), py::arg("reverse"),
-        py::arg("eim"), py::arg("subnat"), py::arg("ncfg"),
+        py::arg("eim"), py::arg("subctx"), py::arg("ncfg"),
         py::arg("hairpin"), py::arg("n_hairpin"), py::arg("nat_stats"),
-        py::arg("log_ring"), py::arg("log_hdr"), py::arg("qos_in"),
+        py::arg("log_ring"), py::arg("log_hdr"), py::arg("qos_eg"),
         py::arg("qos_stats"), py::arg("now_ns"), py::arg("now_sec"),
         py::arg("order") = py::none(),
         py::arg("downlink") = false,
