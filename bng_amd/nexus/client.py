@@ -176,32 +176,127 @@ class Client:
         return sub.ipv4_addr or None if sub else None
 
 
-class VLANAllocator:
-    """S-TAG/C-TAG allocation (ref pkg/nexus/vlan.go): sequential C-TAG
-    assignment within an S-TAG, with persistence through the store."""
+class VLANAllocation:
+    """One NTE's QinQ pair (ref pkg/nexus/vlan.go:33-43)."""
 
-    def __init__(self, store: Store, s_tag: int, c_tag_range=(2, 4094)):
-        self.store = store
-        self.s_tag = s_tag
+    def __init__(self, s_tag: int, c_tag: int, nte_id: str):
+        self.s_tag, self.c_tag, self.nte_id = s_tag, c_tag, nte_id
+
+
+class VLANAllocator:
+    """S-TAG/C-TAG allocation for QinQ NTEs (ref pkg/nexus/vlan.go).
+
+    Sequential C-TAG assignment inside the current S-TAG with rollover
+    to the next S-TAG when full (findAvailable vlan.go:170-190), sticky
+    per NTE, ISP-assigned S-TAG override (AllocateWithSTag :103), stats
+    (:212-227), and warm-start from persisted NTEs (LoadFromStore :239).
+    Optionally persists each allocation through the nexus store."""
+
+    def __init__(self, store: Optional[Store] = None, s_tag: int = None,
+                 c_tag_range=(2, 4094), s_tag_range=None):
+        if s_tag_range is None:
+            # legacy single-S-TAG form: VLANAllocator(store, s_tag=100)
+            s_tag_range = (s_tag, s_tag) if s_tag is not None else (100, 199)
+        self.s_lo, self.s_hi = s_tag_range
         self.lo, self.hi = c_tag_range
-        self.typed = TypedStore(store, f"nexus/vlans/{s_tag}")
+        self.store = store
+        self.typed = (TypedStore(store, "nexus/vlans")
+                      if store is not None else None)
+        self._alloc: Dict[str, VLANAllocation] = {}
+        self._usage: Dict[int, Dict[int, str]] = {}
+        self._cur_s = self.s_lo
         self._lock = threading.Lock()
 
-    def allocate(self, sub_id: str) -> Tuple[int, int]:
-        with self._lock:
-            existing = self.typed.list()
-            for c_tag_s, owner in existing.items():
-                if owner == sub_id:
-                    return self.s_tag, int(c_tag_s)
-            used = {int(k) for k in existing}
-            for c in range(self.lo, self.hi + 1):
-                if c not in used:
-                    self.typed.put(str(c), sub_id)
-                    return self.s_tag, c
-        raise NexusError(f"no free C-TAG under S-TAG {self.s_tag}")
+    def _find_c(self, s_tag: int) -> Optional[int]:
+        used = self._usage.get(s_tag)
+        if not used:
+            return self.lo
+        for c in range(self.lo, self.hi + 1):
+            if c not in used:
+                return c
+        return None
 
-    def release(self, sub_id: str):
+    def _record(self, nte_id: str, s: int, c: int) -> "VLANAllocation":
+        a = VLANAllocation(s, c, nte_id)
+        self._alloc[nte_id] = a
+        self._usage.setdefault(s, {})[c] = nte_id
+        if self.typed is not None:
+            self.typed.put(nte_id, f"{s}:{c}")
+        return a
+
+    def allocate(self, nte_id: str) -> Tuple[int, int]:
         with self._lock:
-            for c_tag_s, owner in self.typed.list().items():
-                if owner == sub_id:
-                    self.typed.delete(c_tag_s)
+            if nte_id in self._alloc:
+                a = self._alloc[nte_id]
+                return a.s_tag, a.c_tag
+            for s in list(range(self._cur_s, self.s_hi + 1)) +                     list(range(self.s_lo, self._cur_s)):
+                c = self._find_c(s)
+                if c is not None:
+                    self._cur_s = s
+                    a = self._record(nte_id, s, c)
+                    return a.s_tag, a.c_tag
+        raise NexusError("VLAN space exhausted")
+
+    def allocate_with_s_tag(self, nte_id: str, s_tag: int) -> Tuple[int, int]:
+        """ISP-assigned S-TAG (ref AllocateWithSTag vlan.go:103-138)."""
+        with self._lock:
+            a = self._alloc.get(nte_id)
+            if a is not None:
+                if a.s_tag == s_tag:
+                    return a.s_tag, a.c_tag
+                self._release_unlocked(nte_id)
+            c = self._find_c(s_tag)
+            if c is None:
+                raise NexusError(f"no free C-TAG under S-TAG {s_tag}")
+            a = self._record(nte_id, s_tag, c)
+            return a.s_tag, a.c_tag
+
+    def _release_unlocked(self, nte_id: str):
+        a = self._alloc.pop(nte_id, None)
+        if a is None:
+            return
+        u = self._usage.get(a.s_tag)
+        if u is not None:
+            u.pop(a.c_tag, None)
+            if not u:
+                del self._usage[a.s_tag]
+        if self.typed is not None:
+            self.typed.delete(nte_id)
+
+    def release(self, nte_id: str):
+        with self._lock:
+            self._release_unlocked(nte_id)
+
+    def get(self, nte_id: str) -> Optional["VLANAllocation"]:
+        with self._lock:
+            return self._alloc.get(nte_id)
+
+    def stats(self) -> Dict[str, int]:
+        with self._lock:
+            s_cap = self.s_hi - self.s_lo + 1
+            c_cap = self.hi - self.lo + 1
+            return {"total_allocations": len(self._alloc),
+                    "s_tags_in_use": len(self._usage),
+                    "s_tag_capacity": s_cap, "c_tag_capacity": c_cap,
+                    "total_capacity": s_cap * c_cap}
+
+    def load_from_store(self, ntes) -> None:
+        """Warm-start from persisted NTE records: iterable of
+        (nte_id, s_tag, c_tag) or dicts (ref LoadFromStore vlan.go:239)."""
+        with self._lock:
+            for n in ntes:
+                if isinstance(n, dict):
+                    nid, s, c = n["id"], n["s_tag"], n["c_tag"]
+                else:
+                    nid, s, c = n
+                a = VLANAllocation(s, c, nid)
+                self._alloc[nid] = a
+                self._usage.setdefault(s, {})[c] = nid
+
+    def sync_to_nte(self, nte: dict) -> dict:
+        """Push this NTE's pair into its record (ref SyncToNTE :265)."""
+        a = self.get(nte["id"])
+        if a is None:
+            raise NexusError(f"no VLAN allocation for {nte['id']}")
+        nte["s_tag"], nte["c_tag"] = a.s_tag, a.c_tag
+        return nte

@@ -182,3 +182,66 @@ class TestHTTPAllocator:
                 al.lookup_ipv4("unknown-sub")
         finally:
             srv.stop()
+
+
+class TestVLANAllocatorFull:
+    """Range/rollover behaviors (ref pkg/nexus/vlan_test.go)."""
+
+    def test_s_tag_rollover(self):
+        v = VLANAllocator(s_tag_range=(100, 102), c_tag_range=(100, 101))
+        v.allocate("nte-1")
+        v.allocate("nte-2")
+        s, c = v.allocate("nte-3")          # first S-TAG full
+        assert (s, c) == (101, 100)
+
+    def test_allocate_with_s_tag_and_reallocate(self):
+        v = VLANAllocator(s_tag_range=(100, 199))
+        s, c = v.allocate_with_s_tag("nte-1", 150)
+        assert (s, c) == (150, 2)
+        assert v.allocate_with_s_tag("nte-1", 150) == (150, 2)   # sticky
+        # different S-TAG requested -> reallocated there
+        s2, c2 = v.allocate_with_s_tag("nte-1", 160)
+        assert s2 == 160
+        assert v.get("nte-1").s_tag == 160
+        assert 150 not in v._usage           # old pair released
+
+    def test_exhaustion(self):
+        v = VLANAllocator(s_tag_range=(10, 10), c_tag_range=(1, 2))
+        v.allocate("a"); v.allocate("b")
+        with pytest.raises(NexusError):
+            v.allocate("c")
+
+    def test_stats(self):
+        v = VLANAllocator(s_tag_range=(100, 199), c_tag_range=(100, 199))
+        st = v.stats()
+        assert st["total_capacity"] == 10000 and st["total_allocations"] == 0
+        for n in ("nte-1", "nte-2", "nte-3"):
+            v.allocate(n)
+        st = v.stats()
+        assert st["total_allocations"] == 3 and st["s_tags_in_use"] == 1
+
+    def test_load_from_store_and_sync(self):
+        v = VLANAllocator(s_tag_range=(100, 299))
+        v.load_from_store([
+            {"id": "nte-1", "s_tag": 100, "c_tag": 100},
+            {"id": "nte-2", "s_tag": 100, "c_tag": 101},
+            {"id": "nte-3", "s_tag": 200, "c_tag": 100},
+        ])
+        st = v.stats()
+        assert st["total_allocations"] == 3 and st["s_tags_in_use"] == 2
+        a = v.get("nte-1")
+        assert (a.s_tag, a.c_tag) == (100, 100)
+        nte = v.sync_to_nte({"id": "nte-3"})
+        assert (nte["s_tag"], nte["c_tag"]) == (200, 100)
+        with pytest.raises(NexusError):
+            v.sync_to_nte({"id": "nte-9"})
+
+    def test_persistence_through_store(self):
+        from bng_amd.nexus.store import MemoryStore
+        store = MemoryStore()
+        v = VLANAllocator(store, s_tag_range=(100, 101), c_tag_range=(2, 9))
+        v.allocate("nte-1")
+        v.release("nte-1")
+        v.allocate("nte-2")
+        from bng_amd.nexus.store import TypedStore
+        assert TypedStore(store, "nexus/vlans").list() == {"nte-2": "100:2"}
