@@ -295,3 +295,115 @@ class HealthChecker:
 
     def stop(self):
         self._stop.set()
+
+
+@dataclass
+class SessionRouteConfig:
+    """ref session_integration.go:35-61 SessionRouteConfig."""
+    enable_injection: bool = True
+    enable_withdrawal: bool = True
+    injection_timeout: float = 5.0
+    withdrawal_timeout: float = 5.0
+    default_subscriber_class: str = ""
+
+
+@dataclass
+class TrackedSession:
+    session_id: str
+    subscriber_id: str
+    subscriber_class: str = ""
+    ipv4: str = ""
+    ipv6: str = ""
+    route_injected: bool = False
+    injected_at: float = 0.0
+
+
+class SessionRouteIntegration:
+    """Session lifecycle -> per-subscriber route plumbing (ref
+    session_integration.go:16-353): activate injects the /32, terminate
+    withdraws it, intermediate states are ignored, and RecoverRoutes
+    re-injects everything after an FRR restart."""
+
+    def __init__(self, route_manager: SubscriberRouteManager,
+                 config: Optional[SessionRouteConfig] = None):
+        self.rm = route_manager
+        self.config = config or SessionRouteConfig()
+        self.sessions: Dict[str, TrackedSession] = {}
+        self._lock = threading.RLock()
+        self.stats = {"activations": 0, "terminations": 0,
+                      "routes_injected": 0, "routes_withdrawn": 0,
+                      "recoveries": 0}
+
+    def on_session_activate(self, session_id: str, subscriber_id: str,
+                            ipv4: str = "", ipv6: str = "",
+                            subscriber_class: str = "") -> bool:
+        """ref :113-186; returns whether a route was injected."""
+        if not self.config.enable_injection:
+            return False
+        with self._lock:
+            ex = self.sessions.get(session_id)
+            if ex is not None and ex.route_injected:
+                return False                       # already injected
+            t = TrackedSession(
+                session_id, subscriber_id,
+                subscriber_class or self.config.default_subscriber_class,
+                ipv4, ipv6)
+            self.sessions[session_id] = t
+        self.stats["activations"] += 1
+        if ipv4:
+            self.rm.add_subscriber_route(ipv4)
+            with self._lock:
+                t.route_injected = True
+                t.injected_at = time.time()
+            self.stats["routes_injected"] += 1
+            return True
+        return False
+
+    def on_session_terminate(self, session_id: str,
+                             reason: str = "") -> bool:
+        """ref :188-237; returns whether a route was withdrawn."""
+        if not self.config.enable_withdrawal:
+            return False
+        with self._lock:
+            t = self.sessions.pop(session_id, None)
+        if t is None:
+            return False
+        self.stats["terminations"] += 1
+        if t.ipv4 and t.route_injected:
+            self.rm.remove_subscriber_route(t.ipv4)
+            self.stats["routes_withdrawn"] += 1
+            return True
+        return False
+
+    def on_session_state_change(self, session_id: str, subscriber_id: str,
+                                old_state: str, new_state: str,
+                                ipv4: str = "", ipv6: str = "",
+                                subscriber_class: str = "",
+                                reason: str = "") -> bool:
+        """ref :239-256: active -> inject, terminal states -> withdraw,
+        intermediate states -> no route change."""
+        if new_state == "active":
+            return self.on_session_activate(session_id, subscriber_id,
+                                            ipv4, ipv6, subscriber_class)
+        if new_state in ("terminated", "error", "timeout"):
+            return self.on_session_terminate(session_id, reason)
+        return False
+
+    def recover_routes(self) -> int:
+        """Re-inject every tracked session's route after an FRR restart
+        (ref :258-270)."""
+        with self._lock:
+            tracked = list(self.sessions.values())
+        n = 0
+        for t in tracked:
+            if t.ipv4:
+                self.rm.add_subscriber_route(t.ipv4)
+                with self._lock:
+                    t.route_injected = True
+                n += 1
+        self.stats["recoveries"] += 1
+        return n
+
+    def tracked_sessions(self) -> List[TrackedSession]:
+        with self._lock:
+            return list(self.sessions.values())

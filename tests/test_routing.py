@@ -129,3 +129,61 @@ class TestHealthChecker:
         hc.check_once()
         assert hc.healthy
         assert events == [False, True]
+
+
+class TestSessionRouteIntegration:
+    """Session lifecycle -> route injection bridge (ref
+    session_integration_test.go)."""
+
+    def make(self, **cfg):
+        from bng_amd.routing.manager import (SessionRouteConfig,
+                                             SessionRouteIntegration)
+        bgp = BGPController(FakeExecutor(), 65001)
+        rm = SubscriberRouteManager(bgp)
+        integ = SessionRouteIntegration(
+            rm, SessionRouteConfig(**cfg) if cfg else None)
+        return bgp, rm, integ
+
+    def test_activate_injects_and_terminate_withdraws(self):
+        _, rm, integ = self.make()
+        assert integ.on_session_activate("s1", "sub-1", "100.64.0.5")
+        assert "100.64.0.5/32" in rm.installed
+        ts = integ.tracked_sessions()
+        assert len(ts) == 1 and ts[0].route_injected
+        # duplicate activation is a no-op
+        assert not integ.on_session_activate("s1", "sub-1", "100.64.0.5")
+        assert integ.on_session_terminate("s1", reason="radius-disconnect")
+        assert "100.64.0.5/32" not in rm.installed
+        assert integ.tracked_sessions() == []
+        # unknown session: no-op
+        assert not integ.on_session_terminate("zzz")
+
+    def test_disabled_injection_and_withdrawal(self):
+        _, rm, integ = self.make(enable_injection=False)
+        assert not integ.on_session_activate("s1", "sub-1", "100.64.0.5")
+        assert rm.installed == set()
+        _, rm2, integ2 = self.make(enable_withdrawal=False)
+        integ2.on_session_activate("s2", "sub-2", "100.64.0.6")
+        assert not integ2.on_session_terminate("s2")
+        assert "100.64.0.6/32" in rm2.installed
+
+    def test_state_change_routing(self):
+        _, rm, integ = self.make()
+        assert integ.on_session_state_change(
+            "s1", "sub-1", "auth", "active", ipv4="100.64.0.7")
+        # intermediate state: no change
+        assert not integ.on_session_state_change(
+            "s1", "sub-1", "active", "rekey", ipv4="100.64.0.7")
+        assert "100.64.0.7/32" in rm.installed
+        assert integ.on_session_state_change(
+            "s1", "sub-1", "active", "timeout", reason="idle")
+        assert "100.64.0.7/32" not in rm.installed
+
+    def test_recover_routes_after_frr_restart(self):
+        bgp, rm, integ = self.make()
+        for k in range(3):
+            integ.on_session_activate(f"s{k}", f"sub-{k}", f"100.64.1.{k}")
+        # FRR restart wipes announcements; recover re-injects all
+        rm.installed.clear()
+        assert integ.recover_routes() == 3
+        assert len(rm.installed) == 3
