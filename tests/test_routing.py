@@ -187,3 +187,38 @@ class TestSessionRouteIntegration:
         rm.installed.clear()
         assert integ.recover_routes() == 3
         assert len(rm.installed) == 3
+
+
+class TestRoutingMetrics:
+    """bng_routing_* instrument surface (ref routing/metrics_test.go)."""
+
+    def test_instruments_and_collect(self):
+        from prometheus_client import generate_latest
+        from bng_amd.routing.bgp import BFDManager
+        from bng_amd.routing.metrics import RoutingMetrics
+        m = RoutingMetrics()
+        bgp = BGPController(FakeExecutor(), 65001)
+        bgp.start()
+        bgp.add_neighbor("192.0.2.1", 65002)
+        bgp.neighbors["192.0.2.1"].established = True
+        bgp.add_neighbor("192.0.2.2", 65002)
+        bgp.announce_prefix("100.64.0.0/10")
+        rm = SubscriberRouteManager(bgp)
+        rm.add_subscriber_route("100.64.0.9")
+        bfd = BFDManager(bgp.exe)
+        bfd.add_peer("192.0.2.1")
+        bfd.handle_state_change("192.0.2.1", True)
+        m.record_route_injection(0.002)
+        m.record_route_injection(ok=False)
+        m.record_route_withdrawal(0.001)
+        m.bgp_state_changes.labels("192.0.2.1", "Established").inc()
+        m.bfd_state_changes.labels("192.0.2.1", "up").inc()
+        m.collect(bgp=bgp, route_manager=rm, bfd=bfd)
+        text = generate_latest(m.registry).decode()
+        assert "bng_routing_subscriber_routes_active 1.0" in text
+        assert "bng_routing_bgp_neighbors_total 2.0" in text
+        assert "bng_routing_bgp_neighbors_established 1.0" in text
+        assert "bng_routing_bgp_prefixes_announced 1.0" in text
+        assert "bng_routing_bfd_peers_up 1.0" in text
+        assert "bng_routing_subscriber_routes_injected_total 1.0" in text
+        assert "bng_routing_route_injection_errors_total 1.0" in text
