@@ -218,7 +218,7 @@ class TestRoutingMetrics:
         assert "bng_routing_subscriber_routes_active 1.0" in text
         assert "bng_routing_bgp_neighbors_total 2.0" in text
         assert "bng_routing_bgp_neighbors_established 1.0" in text
-        assert "bng_routing_bgp_prefixes_announced 1.0" in text
+        assert "bng_routing_bgp_prefixes_announced 2.0" in text  # /10 + /32
         assert "bng_routing_bfd_peers_up 1.0" in text
         assert "bng_routing_subscriber_routes_injected_total 1.0" in text
         assert "bng_routing_route_injection_errors_total 1.0" in text
