@@ -143,3 +143,116 @@ class TestSubscriberManager:
         m = SubMgr(authenticator=Deny())
         assert m.create_session("sub-1") is None
         assert m.stats["auth_failed"] == 1
+
+
+class TestRuntimeStore:
+    """Typed operational DB behaviors (ref pkg/state/store_test.go)."""
+
+    def make(self, **kw):
+        from bng_amd.state.runtime import Config, RuntimeStore
+        return RuntimeStore(Config(**kw) if kw else None)
+
+    def test_subscriber_crud_and_indexes(self):
+        from bng_amd.state.runtime import StateError, Subscriber
+        st = self.make()
+        st.create_subscriber(Subscriber("s1", mac="aa:bb:cc:00:00:01",
+                                        nte_id="nte-1"))
+        assert st.get_subscriber_by_mac("aa:bb:cc:00:00:01").id == "s1"
+        assert st.get_subscriber_by_nte("nte-1").id == "s1"
+        with pytest.raises(StateError):      # duplicate MAC
+            st.create_subscriber(Subscriber("s2",
+                                            mac="aa:bb:cc:00:00:01"))
+        st.delete_subscriber("s1")
+        assert st.get_subscriber_by_mac("aa:bb:cc:00:00:01") is None
+
+    def test_find_pool_priority_isp_class_capacity(self):
+        from bng_amd.state.runtime import Pool, StateError, Subscriber
+        st = self.make()
+        st.create_pool(Pool("p-low", priority=1, total_addresses=10))
+        st.create_pool(Pool("p-high", priority=9, total_addresses=10))
+        st.create_pool(Pool("p-full", priority=99, total_addresses=10,
+                            allocated_addresses=10))
+        st.create_pool(Pool("p-isp", priority=50, total_addresses=10,
+                            isp_ids=["isp-2"]))
+        st.create_pool(Pool("p-class", priority=60, total_addresses=10,
+                            subscriber_class=["gold"]))
+        sub = Subscriber("s1", isp_id="isp-1", klass="silver")
+        assert st.find_pool_for_subscriber(sub).id == "p-high"
+        gold = Subscriber("s2", isp_id="isp-2", klass="gold")
+        assert st.find_pool_for_subscriber(gold).id == "p-class"
+        with pytest.raises(StateError):
+            st.find_pool_for_subscriber(sub, version=6)
+
+    def test_lease_lifecycle_and_cleanup(self):
+        from bng_amd.state.runtime import Lease
+        st = self.make()
+        now = time.time()
+        st.create_lease(Lease("l1", mac="aa:00:00:00:00:01",
+                              ipv4="10.0.0.5", expires_at=now + 100))
+        st.create_lease(Lease("l2", mac="aa:00:00:00:00:02",
+                              ipv4="10.0.0.6", expires_at=now - 1))
+        assert st.get_lease_by_ip("10.0.0.5").id == "l1"
+        assert st.get_lease_by_mac("aa:00:00:00:00:02").id == "l2"
+        st.renew_lease("l1", 3600)
+        assert st.cleanup_expired_leases(now=now) == 1
+        assert st.get_lease("l2") is None
+        assert st.get_lease_by_ip("10.0.0.6") is None
+        assert st.get_lease("l1") is not None     # renewed survives
+
+    def test_session_idle_and_hard_timeout(self):
+        from bng_amd.state.runtime import Session
+        st = self.make()
+        now = time.time()
+        st.create_session(Session("a", mac="m1", ipv4="10.0.0.1",
+                                  idle_timeout=30,
+                                  last_activity=now - 60))
+        st.create_session(Session("b", mac="m2", ipv4="10.0.0.2",
+                                  session_timeout=100,
+                                  started_at=now - 200,
+                                  last_activity=now))
+        st.create_session(Session("c", mac="m3", ipv4="10.0.0.3",
+                                  idle_timeout=30, last_activity=now))
+        st.update_session_activity("c", bytes_in=100, bytes_out=50)
+        assert st.cleanup_idle_sessions(now=now) == 2
+        assert st.get_session("c").bytes_in == 100
+        assert st.get_session_by_ip("10.0.0.1") is None
+        assert st.get_session_by_mac("m3").id == "c"
+
+    def test_nat_binding_index_and_cleanup(self):
+        from bng_amd.state.runtime import NATBinding
+        st = self.make()
+        now = time.time()
+        st.create_nat_binding(NATBinding("n1", private_ip="10.0.0.1",
+                                         private_port=4000, protocol=17,
+                                         expires_at=now + 60))
+        st.create_nat_binding(NATBinding("n2", private_ip="10.0.0.2",
+                                         private_port=4001, protocol=6,
+                                         expires_at=now - 1))
+        b = st.get_nat_binding_by_private("10.0.0.1", 4000, 17)
+        assert b.id == "n1"
+        assert st.cleanup_expired_nat(now=now) == 1
+        assert st.get_nat_binding_by_private("10.0.0.2", 4001, 6) is None
+
+    def test_max_limits(self):
+        from bng_amd.state.runtime import (Lease, LimitExceeded, Session,
+                                           Subscriber)
+        st = self.make(max_subscribers=1, max_leases=1, max_sessions=1)
+        st.create_subscriber(Subscriber("s1"))
+        with pytest.raises(LimitExceeded):
+            st.create_subscriber(Subscriber("s2"))
+        st.create_lease(Lease("l1"))
+        with pytest.raises(LimitExceeded):
+            st.create_lease(Lease("l2"))
+        st.create_session(Session("x"))
+        with pytest.raises(LimitExceeded):
+            st.create_session(Session("y"))
+        s = st.stats()
+        assert s["subscribers"] == 1 and s["sessions"] == 1
+
+    def test_start_stop_loops(self):
+        st = self.make(lease_cleanup_interval=0.01,
+                       session_cleanup_interval=0.01,
+                       nat_cleanup_interval=0.01)
+        st.start()
+        time.sleep(0.05)
+        st.stop()
