@@ -242,3 +242,85 @@ class TestQinQ:
         assert m.lookup(200, 3) == "sub-2"
         m.unregister("sub-2")
         assert m.auto_assign("sub-3") == (200, 3)
+
+
+class TestAgentTLS:
+    """TLS config builder (ref pkg/agent/tls_test.go)."""
+
+    @pytest.fixture(scope="class")
+    def certs(self, tmp_path_factory):
+        import subprocess
+        d = tmp_path_factory.mktemp("tls")
+        cert, key = str(d / "cert.pem"), str(d / "key.pem")
+        subprocess.run(
+            ["openssl", "req", "-x509", "-newkey", "rsa:2048",
+             "-keyout", key, "-out", cert, "-days", "1", "-nodes",
+             "-subj", "/CN=bng-test/O=bng"],
+            check=True, capture_output=True)
+        return cert, key
+
+    def test_disabled_returns_none(self):
+        from bng_amd.agent.tls import TLSConfig, build_ssl_context
+        assert build_ssl_context(TLSConfig(enabled=False)) is None
+
+    def test_default_and_min_versions(self):
+        import ssl
+        from bng_amd.agent.tls import (TLSConfig, TLSError,
+                                       build_ssl_context)
+        ctx = build_ssl_context(TLSConfig())
+        assert ctx.minimum_version == ssl.TLSVersion.TLSv1_2
+        ctx13 = build_ssl_context(TLSConfig(min_version="1.3"))
+        assert ctx13.minimum_version == ssl.TLSVersion.TLSv1_3
+        with pytest.raises(TLSError):
+            build_ssl_context(TLSConfig(min_version="1.1"))
+
+    def test_insecure_skip_verify(self):
+        import ssl
+        from bng_amd.agent.tls import TLSConfig, build_ssl_context
+        ctx = build_ssl_context(TLSConfig(insecure_skip_verify=True))
+        assert ctx.verify_mode == ssl.CERT_NONE
+
+    def test_ca_by_file_and_pem_and_mtls(self, certs):
+        from bng_amd.agent.tls import (TLSConfig, TLSError,
+                                       build_ssl_context,
+                                       requests_kwargs)
+        cert, key = certs
+        ctx = build_ssl_context(TLSConfig(ca_cert_file=cert,
+                                          cert_file=cert, key_file=key))
+        assert ctx.get_ca_certs()
+        pem = open(cert).read()
+        ctx2 = build_ssl_context(TLSConfig(ca_cert_pem=pem))
+        assert ctx2.get_ca_certs()
+        # half-configured mTLS rejected
+        with pytest.raises(TLSError):
+            build_ssl_context(TLSConfig(cert_file=cert))
+        with pytest.raises(TLSError):
+            build_ssl_context(TLSConfig(ca_cert_file="/nope.pem"))
+        kw = requests_kwargs(TLSConfig(ca_cert_file=cert,
+                                       cert_file=cert, key_file=key))
+        assert kw["verify"] == cert and kw["cert"] == (cert, key)
+
+    def test_fingerprint_and_pinning(self, certs):
+        import ssl as ssl_mod
+        from bng_amd.agent.tls import (TLSConfig, TLSError,
+                                       get_cert_fingerprint,
+                                       validate_tls_config, verify_pinned)
+        cert, _ = certs
+        fp = get_cert_fingerprint(cert)
+        assert len(fp) == 64
+        validate_tls_config(TLSConfig(pinned_certs=[fp]))
+        with pytest.raises(TLSError):
+            validate_tls_config(TLSConfig(pinned_certs=["zz"]))
+        der = ssl_mod.PEM_cert_to_DER_cert(open(cert).read())
+        assert verify_pinned(der, [fp])
+        assert verify_pinned(der, [fp.upper().replace("", "")])
+        assert not verify_pinned(der, ["0" * 64])
+        assert verify_pinned(der, [])        # no pins -> pass
+
+    def test_extract_cert_info(self, certs):
+        from bng_amd.agent.tls import extract_cert_info
+        cert, _ = certs
+        info = extract_cert_info(cert)
+        assert info["subject"] == "bng-test"
+        assert info["issuer"] == "bng-test"     # self-signed
+        assert info["not_after"]
