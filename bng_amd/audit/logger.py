@@ -206,3 +206,54 @@ class Logger:
             except queue.Empty:
                 return
             self._process(ev)
+
+
+class SecurityAuditor:
+    """TLS / certificate / mTLS security events (ref
+    pkg/audit/security.go + security_test.go): typed wrappers over the
+    async logger so operators get a uniform bng-security event stream."""
+
+    def __init__(self, logger: "Logger"):
+        self.logger = logger
+
+    def log_tls_handshake(self, peer: str, success: bool,
+                          tls_version: str = "", cipher: str = "",
+                          error: str = ""):
+        self.logger.log(
+            "tls_handshake", category=CAT_TLS,
+            outcome="success" if success else "failure",
+            peer=peer, tls_version=tls_version, cipher=cipher,
+            error=error)
+
+    def log_certificate_expiring(self, subject: str, not_after: str,
+                                 days_left: int):
+        self.logger.log("certificate_expiring", category=CAT_TLS,
+                        outcome="warning", subject=subject,
+                        not_after=not_after, days_left=days_left)
+
+    def log_certificate_expired(self, subject: str, not_after: str):
+        self.logger.log("certificate_expired", category=CAT_TLS,
+                        outcome="failure", subject=subject,
+                        not_after=not_after)
+
+    def log_certificate_invalid(self, subject: str, reason: str):
+        self.logger.log("certificate_invalid", category=CAT_TLS,
+                        outcome="failure", subject=subject,
+                        reason=reason)
+
+    def log_certificate_pin_failed(self, peer: str, fingerprint: str):
+        self.logger.log("certificate_pin_failed", category=CAT_TLS,
+                        outcome="failure", peer=peer,
+                        fingerprint=fingerprint)
+
+    def log_certificate_renewed(self, subject: str, not_after: str):
+        self.logger.log("certificate_renewed", category=CAT_TLS,
+                        outcome="success", subject=subject,
+                        not_after=not_after)
+
+    def log_mtls_auth(self, device_id: str, success: bool,
+                      subject: str = "", error: str = ""):
+        self.logger.log(
+            "mtls_auth", category=CAT_AUTH,
+            outcome="success" if success else "failure",
+            subscriber=device_id, subject=subject, error=error)

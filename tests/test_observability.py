@@ -256,3 +256,34 @@ class TestRuntimeStore:
         st.start()
         time.sleep(0.05)
         st.stop()
+
+
+class TestSecurityAuditor:
+    """TLS/cert/mTLS event stream (ref pkg/audit/security_test.go)."""
+
+    def test_security_events(self):
+        from bng_amd.audit.logger import (CAT_TLS, Logger, MemoryStorage,
+                                          SecurityAuditor)
+        st = MemoryStorage()
+        lg = Logger(storage=st).start()
+        sec = SecurityAuditor(lg)
+        sec.log_tls_handshake("192.0.2.9", True, tls_version="1.3",
+                              cipher="TLS_AES_128_GCM_SHA256")
+        sec.log_tls_handshake("192.0.2.9", False, error="bad cert")
+        sec.log_certificate_expiring("CN=bng", "2027-01-01", 20)
+        sec.log_certificate_expired("CN=bng", "2025-01-01")
+        sec.log_certificate_invalid("CN=bng", "hostname mismatch")
+        sec.log_certificate_pin_failed("192.0.2.9", "ab" * 32)
+        sec.log_certificate_renewed("CN=bng", "2028-01-01")
+        sec.log_mtls_auth("nte-1", True, subject="CN=nte-1")
+        sec.log_mtls_auth("nte-2", False, error="unknown CA")
+        lg.stop()
+        evs = st.query(category=CAT_TLS)
+        acts = [e.action for e in evs]
+        assert "tls_handshake" in acts and \
+            "certificate_pin_failed" in acts
+        fails = [e for e in evs if e.outcome == "failure"]
+        assert len(fails) == 4
+        auth = [e for e in st.all() if e.action == "mtls_auth"]
+        assert {e.outcome for e in auth} == {"success", "failure"}
+        assert auth[0].subscriber == "nte-1"
