@@ -56,3 +56,51 @@ def test_extension_layout_matches_ctypes():
     assert off["antispoof_config.allowed_net"] == \
         abi.AntispoofConfig.allowed_net.offset
     assert off["spoof_event.spoofed_ip"] == abi.SpoofEvent.spoofed_ip.offset
+
+
+class TestConversions:
+    """IP/MAC/key conversion consistency (ref test/ebpf/maps_test.go +
+    pkg/ebpf/loader_test.go conversion tests)."""
+
+    def test_ip_round_trip(self):
+        from bng_amd.dataplane.packets import ip2u32, u32_to_ip
+        for ip in ("0.0.0.0", "10.0.1.5", "203.0.113.255",
+                   "255.255.255.255"):
+            assert u32_to_ip(ip2u32(ip)) == ip
+        assert ip2u32("1.2.3.4") == 0x01020304   # BE wire order as host int
+
+    def test_mac_round_trip(self):
+        from bng_amd.dataplane.abi import mac_to_u64
+        from bng_amd.dataplane.packets import mac_bytes
+        m = mac_bytes("aa:bb:cc:dd:ee:ff")
+        v = mac_to_u64(m)
+        assert v == 0xAABBCCDDEEFF
+        assert bytes((v >> (8 * (5 - i))) & 0xFF for i in range(6)) == m
+
+    def test_key_spaces_disjoint(self):
+        """MAC / VLAN / circuit keys share one u64 table but must never
+        collide across types (tag bits 62-63, ref maps.h key scheme)."""
+        from bng_amd.dataplane.abi import (circuit_key, mac_to_u64,
+                                           vlan_key)
+        from bng_amd.dataplane.packets import mac_bytes
+        mk = mac_to_u64(mac_bytes("aa:bb:cc:00:00:01"))
+        vk = vlan_key(100, 200)
+        ck = circuit_key(b"pon0/1/2:100")
+        assert len({mk >> 62, vk >> 62, ck >> 62}) == 3
+        assert vlan_key(100, 200) != vlan_key(200, 100)
+
+    def test_circuit_hash_truncation_and_collisions(self):
+        """Circuit IDs hash over a 32-byte zero-padded buffer, matching
+        the reference's fixed-size map key (maps.h:216-220, ref
+        CircuitIDKeyTruncation): ids differing only beyond byte 32
+        collide BY DESIGN; within the window they stay distinct."""
+        from bng_amd.dataplane.abi import circuit_key, fnv1a64
+        assert circuit_key(b"x" * 40) == circuit_key(b"x" * 50)
+        assert circuit_key(b"a" + b"x" * 31) != circuit_key(
+            b"b" + b"x" * 31)
+        assert circuit_key(b"pad") == circuit_key(b"pad\x00\x00")
+        # 1k distinct ids -> 1k distinct keys (collision resistance)
+        keys = {circuit_key(f"olt{i}/pon{i % 7}".encode())
+                for i in range(1000)}
+        assert len(keys) == 1000
+        assert fnv1a64(b"") == 0xcbf29ce484222325   # FNV offset basis
