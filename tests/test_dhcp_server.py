@@ -229,3 +229,71 @@ class TestFuzz:
             except (ValueError, struct.error):
                 continue
             srv.handle(msg)   # must not raise
+
+
+class TestServerCoverage:
+    """Edge behaviors mirrored from the reference's coverage list
+    (ref pkg/dhcp/server_coverage_test.go)."""
+
+    def test_discover_reuses_existing_lease(self):
+        srv = make_server()
+        offer1 = srv.handle(dm.build_request(MAC, dm.DISCOVER))
+        srv.handle(dm.build_request(MAC, dm.REQUEST,
+                                    requested_ip=offer1.yiaddr))
+        offer2 = srv.handle(dm.build_request(MAC, dm.DISCOVER))
+        assert offer2.yiaddr == offer1.yiaddr
+
+    def test_request_uses_ciaddr_when_no_option50(self):
+        srv = make_server()
+        offer = srv.handle(dm.build_request(MAC, dm.DISCOVER))
+        req = dm.build_request(MAC, dm.REQUEST)
+        req.ciaddr = offer.yiaddr          # RENEWING state: ciaddr only
+        ack = srv.handle(req)
+        assert ack.msg_type == dm.ACK and ack.yiaddr == offer.yiaddr
+        # mismatched ciaddr -> NAK
+        req2 = dm.build_request(MAC, dm.REQUEST)
+        req2.ciaddr = offer.yiaddr + 1
+        assert srv.handle(req2).msg_type == dm.NAK
+
+    def test_release_nonexistent_is_noop(self):
+        srv = make_server()
+        assert srv.handle(dm.build_request(MAC, dm.RELEASE)) is None
+        assert srv.stats["release"] == 1
+
+    def test_decline_without_requested_ip_is_noop(self):
+        srv = make_server()
+        assert srv.handle(dm.build_request(MAC, dm.DECLINE)) is None
+
+    def test_unknown_message_type_ignored(self):
+        srv = make_server()
+        req = dm.build_request(MAC, dm.DISCOVER)
+        req.set_option(dm.OPT_MSG_TYPE, bytes([99]))
+        assert srv.handle(req) is None
+        # BOOTREPLY op ignored entirely
+        req2 = dm.build_request(MAC, dm.DISCOVER)
+        req2.op = 2
+        assert srv.handle(req2) is None
+
+    def test_concurrent_discover_request(self):
+        """Thread-hammered DORA: every client ends with a unique IP
+        (ref TestConcurrentDiscoverRequests)."""
+        import threading
+        srv = make_server()
+        ips, errs = {}, []
+
+        def worker(k):
+            try:
+                mac = mac_bytes(f"aa:bb:cc:00:01:{k:02x}")
+                offer = srv.handle(dm.build_request(mac, dm.DISCOVER))
+                ack = srv.handle(dm.build_request(
+                    mac, dm.REQUEST, requested_ip=offer.yiaddr))
+                assert ack.msg_type == dm.ACK
+                ips[k] = ack.yiaddr
+            except Exception as e:        # pragma: no cover
+                errs.append(e)
+        ts = [threading.Thread(target=worker, args=(k,))
+              for k in range(32)]
+        [t.start() for t in ts]
+        [t.join() for t in ts]
+        assert not errs
+        assert len(set(ips.values())) == 32
