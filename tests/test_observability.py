@@ -287,3 +287,77 @@ class TestSecurityAuditor:
         auth = [e for e in st.all() if e.action == "mtls_auth"]
         assert {e.outcome for e in auth} == {"success", "failure"}
         assert auth[0].subscriber == "nte-1"
+
+
+class TestSubscriberManagerLifecycle:
+    """Staged lifecycle + walled garden + cleanup (ref
+    pkg/subscriber/manager_test.go)."""
+
+    def test_staged_create_auth_assign_activate(self):
+        class Alloc:
+            def allocate(self, sid):
+                return "10.9.0.1"
+
+            def release(self, sid):
+                pass
+        m = SubMgr(allocator=Alloc())
+        s = m.open_session("sub-1", mac="aa:01", isp_id="isp-a",
+                           metadata={"circuit": "pon0/1"})
+        assert s.state == "created"
+        assert s.attributes["circuit"] == "pon0/1"
+        assert m.authenticate(s.id)
+        assert s.state == "authenticated"
+        assert m.assign_address(s.id) == "10.9.0.1"
+        assert s.state == "addressed"
+        assert m.activate_session(s.id)
+        assert s.state == "active"
+        assert m.get_by_mac("aa:01").id == s.id
+        assert m.get_by_ip("10.9.0.1").id == s.id
+        # idempotent open returns the same session
+        assert m.open_session("sub-1").id == s.id
+
+    def test_walled_garden_transitions(self):
+        m = SubMgr()
+        s = m.open_session("sub-1")
+        m.activate_session(s.id, walled=True)
+        assert s.state == "walled_garden"
+        assert not m.set_walled_garden(s.id)      # already walled
+        assert m.clear_walled_garden(s.id)
+        assert s.state == "active"
+        assert not m.clear_walled_garden(s.id)    # already clear
+        assert m.set_walled_garden(s.id)
+        assert not m.set_walled_garden("missing")
+
+    def test_cleanup_idle_and_session_timeout(self):
+        m = SubMgr(idle_timeout=30, session_timeout=1000)
+        a = m.create_session("sub-a", ip="10.0.0.1")
+        b = m.create_session("sub-b", ip="10.0.0.2")
+        c = m.create_session("sub-c", ip="10.0.0.3")
+        now = time.time()
+        a.last_activity = now - 60                 # idle
+        b.started_at = now - 2000                  # hard timeout
+        assert m.cleanup(now=now) == 2
+        assert m.count() == 1 and m.get(c.id) is not None
+        assert c.attributes.get("terminate_reason") is None
+
+    def test_list_by_isp_and_multiple_handlers(self):
+        got1, got2 = [], []
+        m = SubMgr()
+        m.on_event(lambda ev, s: got1.append(ev))
+        m.on_event(lambda ev, s: got2.append(ev))
+        sa = m.open_session("a", isp_id="isp-1")
+        m.open_session("b", isp_id="isp-2")
+        m.open_session("c", isp_id="isp-1")
+        assert {s.subscriber_id for s in m.list_by_isp("isp-1")} == \
+            {"a", "c"}
+        assert m.list_by_isp("isp-9") == []
+        assert len(m.list_sessions()) == 3
+        m.activate_session(sa.id)
+        m.terminate_session(sa.id)
+        assert "session_created" in got1 and "session_stop" in got2
+
+    def test_max_sessions_staged(self):
+        m = SubMgr(max_sessions=1)
+        assert m.open_session("a") is not None
+        assert m.open_session("b") is None
+        assert m.stats["rejected_capacity"] == 1
