@@ -75,3 +75,22 @@ class Mapper:
             key = self._by_sub.pop(subscriber_id, None)
             if key is not None:
                 self._by_vlan.pop(key, None)
+
+    def unregister_by_vlan(self, s_tag: int, c_tag: int) -> Optional[str]:
+        """Remove a mapping keyed by the VLAN pair (ref qinq.go
+        UnregisterByVLAN); returns the former owner."""
+        with self._lock:
+            owner = self._by_vlan.pop((s_tag, c_tag), None)
+            if owner is not None:
+                self._by_sub.pop(owner, None)
+            return owner
+
+    def stats(self) -> Dict[str, int]:
+        """ref qinq.go Stats: mappings + capacity across ranges."""
+        with self._lock:
+            cap = sum(r.c_tag_end - r.c_tag_start + 1
+                      for r in self._ranges)
+            return {"mappings": len(self._by_vlan),
+                    "ranges": len(self._ranges),
+                    "capacity": cap,
+                    "free": max(0, cap - len(self._by_vlan))}

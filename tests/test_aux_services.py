@@ -324,3 +324,38 @@ class TestAgentTLS:
         assert info["subject"] == "bng-test"
         assert info["issuer"] == "bng-test"     # self-signed
         assert info["not_after"]
+
+
+class TestQinQFull:
+    """Mapper behaviors mirrored from ref pkg/qinq/qinq_test.go."""
+
+    def test_invalid_tags_rejected(self):
+        m = QinQMapper()
+        with pytest.raises(QinQError):
+            m.register("s", 0, 5)
+        with pytest.raises(QinQError):
+            m.register("s", 100, 4095)
+        with pytest.raises(QinQError):
+            m.add_range(5000, 2, 4)
+        with pytest.raises(QinQError):
+            m.add_range(100, 10, 5)       # start > end
+
+    def test_update_moves_subscriber_mapping(self):
+        m = QinQMapper()
+        m.register("sub-1", 100, 2)
+        m.register("sub-1", 100, 9)       # re-register: frees old pair
+        assert m.lookup(100, 2) is None
+        assert m.lookup(100, 9) == "sub-1"
+        assert m.lookup_subscriber("sub-1") == (100, 9)
+
+    def test_unregister_by_vlan_and_stats(self):
+        m = QinQMapper()
+        m.add_range(300, 2, 11)
+        m.register("sub-1", 300, 2)
+        m.register("sub-2", 300, 3)
+        assert m.unregister_by_vlan(300, 2) == "sub-1"
+        assert m.unregister_by_vlan(300, 2) is None
+        assert m.lookup_subscriber("sub-1") is None
+        st = m.stats()
+        assert st == {"mappings": 1, "ranges": 1, "capacity": 10,
+                      "free": 9}
