@@ -414,7 +414,13 @@ class PPPoEServer:
         if cp.code == C.CONF_REQ:
             opts = C.decode_opts(cp.data)
             naks = []
+            rejects = []
             if proto == C.PROTO_IPCP:
+                # unsupported options (VJ IP-compression etc.) are
+                # Configure-Rejected, not silently accepted (RFC 1332;
+                # ref ipcp.go compression rejection)
+                known = {C.IPCP_OPT_IP, C.IPCP_OPT_DNS1, C.IPCP_OPT_DNS2}
+                rejects = [(t, v) for (t, v) in opts if t not in known]
                 want = C.get_opt(opts, C.IPCP_OPT_IP)
                 if want is None or struct.unpack(">I", want)[0] != s.ip:
                     naks.append((C.IPCP_OPT_IP, struct.pack(">I", s.ip)))
@@ -430,10 +436,16 @@ class PPPoEServer:
                                  struct.pack(">I", s.dns2)))
             else:
                 ifid = C.get_opt(opts, C.IPV6CP_OPT_IFID)
-                if ifid == s.ifid:   # must differ from ours
+                if ifid is None or ifid == b"\x00" * 8 or ifid == s.ifid:
+                    # zero or colliding interface-id: suggest one
+                    # derived from the client MAC (RFC 5072 NAK rule)
                     naks.append((C.IPV6CP_OPT_IFID,
                                  hashlib.md5(s.client_mac).digest()[:8]))
-            if naks:
+            if rejects:
+                out.append(self._sess_frame(s, proto, C.CPPacket(
+                    C.CONF_REJ, cp.identifier,
+                    C.encode_opts(rejects)).encode()))
+            elif naks:
                 out.append(self._sess_frame(s, proto, C.CPPacket(
                     C.CONF_NAK, cp.identifier,
                     C.encode_opts(naks)).encode()))

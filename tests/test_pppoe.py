@@ -330,3 +330,54 @@ class TestKeepaliveTeardown:
         cli2.discover()
         assert cli2.session_id != sid1
         assert srv.session_count() == 1
+
+
+class TestNCPOptionHandling:
+    """RFC 1332 / RFC 5072 option rules (ref ipcp_test.go 'reject IP
+    compression option', ipv6cp 'NAK zero interface ID')."""
+
+    def _open_session(self):
+        srv = make_server(auth="chap")
+        cli = SimClient(srv)
+        cli.discover()
+        assert srv.stats["ipcp_opened"] == 1
+        return srv, cli
+
+    def _send(self, srv, cli, proto, cp):
+        frames = srv.handle_frame(C.SessionPacket(
+            cli.session_id, proto, cp.encode(), src_mac=cli.mac,
+            dst_mac=SRV_MAC).encode())
+        out = []
+        for f in frames:
+            p = C.SessionPacket.decode(f)
+            out.append((p.ppp_proto, C.CPPacket.decode(p.payload)))
+        return out
+
+    def test_ipcp_rejects_compression_option(self):
+        srv, cli = self._open_session()
+        from bng_amd.dataplane.packets import ip2u32
+        req = C.CPPacket(C.CONF_REQ, 9, C.encode_opts(
+            [(C.IPCP_OPT_IP, struct.pack(">I", cli.got_ip)),
+             (2, b"\x00\x2d\x0f\x01")]))      # VJ compression opt
+        replies = self._send(srv, cli, C.PROTO_IPCP, req)
+        proto, cp = replies[0]
+        assert proto == C.PROTO_IPCP and cp.code == C.CONF_REJ
+        opts = C.decode_opts(cp.data)
+        assert [t for t, _ in opts] == [2]    # only the bad option
+
+    def test_ipv6cp_naks_zero_interface_id(self):
+        srv, cli = self._open_session()
+        req = C.CPPacket(C.CONF_REQ, 5, C.encode_opts(
+            [(C.IPV6CP_OPT_IFID, b"\x00" * 8)]))
+        replies = self._send(srv, cli, C.PROTO_IPV6CP, req)
+        proto, cp = replies[0]
+        assert proto == C.PROTO_IPV6CP and cp.code == C.CONF_NAK
+        sug = C.get_opt(C.decode_opts(cp.data), C.IPV6CP_OPT_IFID)
+        assert sug is not None and sug != b"\x00" * 8
+
+    def test_ipcp_acks_correct_ip(self):
+        srv, cli = self._open_session()
+        req = C.CPPacket(C.CONF_REQ, 7, C.encode_opts(
+            [(C.IPCP_OPT_IP, struct.pack(">I", cli.got_ip))]))
+        replies = self._send(srv, cli, C.PROTO_IPCP, req)
+        assert replies[0][1].code == C.CONF_ACK
