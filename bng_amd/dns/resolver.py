@@ -93,11 +93,16 @@ class CacheEntry:
 class Resolver:
     def __init__(self, upstream: Callable[[bytes], Optional[bytes]],
                  min_ttl: int = 30, max_ttl: int = 3600,
-                 rate_limit: float = 0.0, rate_burst: int = 50):
+                 rate_limit: float = 0.0, rate_burst: int = 50,
+                 max_entries: int = 100_000, negative_ttl: int = 30):
+        from collections import OrderedDict
         self.upstream = upstream
         self.min_ttl = min_ttl
         self.max_ttl = max_ttl
-        self.cache: Dict[Tuple[str, int], CacheEntry] = {}
+        self.max_entries = max_entries
+        self.negative_ttl = negative_ttl
+        self.cache: "OrderedDict[Tuple[str, int], CacheEntry]" = \
+            OrderedDict()
         self.intercepts: Dict[str, List[str]] = {}   # name -> portal IPs
         self.intercept_all_to: Optional[List[str]] = None
         self._lock = threading.RLock()
@@ -105,7 +110,8 @@ class Resolver:
         self.rate_burst = rate_burst
         self._buckets: Dict[str, List[float]] = {}
         self.stats = {"queries": 0, "cache_hits": 0, "intercepted": 0,
-                      "rate_limited": 0, "upstream_fail": 0}
+                      "rate_limited": 0, "upstream_fail": 0,
+                      "negative_hits": 0, "evicted": 0}
 
     # -------------------------------------------------------- intercepts
     def add_intercept(self, name: str, addrs: List[str]):
@@ -157,10 +163,16 @@ class Resolver:
         if hit is not None:
             self.stats["intercepted"] += 1
             return build_response(query, hit, ttl=30)
-        # cache
+        # cache (LRU: a hit moves the entry to the back)
         with self._lock:
             ce = self.cache.get((key, qtype))
             if ce is not None and ce.expires > time.time():
+                self.cache.move_to_end((key, qtype))
+                if not ce.addrs:
+                    # negative cache: answer NXDOMAIN without upstream
+                    self.stats["negative_hits"] += 1
+                    return build_response(query, [], ttl=self.negative_ttl,
+                                          rcode=3)
                 self.stats["cache_hits"] += 1
                 ttl = max(1, int(ce.expires - time.time()))
                 return build_response(query, ce.addrs, ttl=ttl)
@@ -175,8 +187,25 @@ class Resolver:
             return None
         _qn, addrs, ttl = parse_response(resp)
         ttl = max(self.min_ttl, min(self.max_ttl, ttl))   # TTL clamp
-        if addrs:
-            with self._lock:
+        with self._lock:
+            if addrs:
                 self.cache[(key, qtype)] = CacheEntry(addrs,
                                                       time.time() + ttl)
+            else:
+                # empty answer / NXDOMAIN: negative-cache it (RFC 2308)
+                self.cache[(key, qtype)] = CacheEntry(
+                    [], time.time() + self.negative_ttl)
+            self.cache.move_to_end((key, qtype))
+            while len(self.cache) > self.max_entries:   # LRU eviction
+                self.cache.popitem(last=False)
+                self.stats["evicted"] += 1
         return resp
+
+    def cleanup(self, now: Optional[float] = None) -> int:
+        """Drop expired cache entries (ref resolver cache cleanup)."""
+        now = now if now is not None else time.time()
+        with self._lock:
+            dead = [k for k, ce in self.cache.items() if ce.expires <= now]
+            for k in dead:
+                del self.cache[k]
+            return len(dead)

@@ -359,3 +359,54 @@ class TestQinQFull:
         st = m.stats()
         assert st == {"mappings": 1, "ranges": 1, "capacity": 10,
                       "free": 9}
+
+
+class TestDNSCacheBehaviors:
+    """LRU eviction / negative cache / cleanup (ref
+    pkg/dns/resolver_test.go)."""
+
+    def _mk(self, answers=None, **kw):
+        from bng_amd.dns.resolver import (Resolver, build_query,
+                                          build_response)
+        calls = []
+
+        def upstream(q):
+            calls.append(q)
+            from bng_amd.dns.resolver import decode_qname
+            name, _ = decode_qname(q, 12)
+            addrs = (answers or {}).get(name, [])
+            return build_response(q, addrs, ttl=300,
+                                  rcode=0 if addrs else 3)
+        return Resolver(upstream, **kw), calls
+
+    def test_lru_eviction(self):
+        from bng_amd.dns.resolver import build_query
+        r, calls = self._mk({f"h{i}.x": ["10.0.0.1"] for i in range(5)},
+                            max_entries=3)
+        for i in range(5):
+            r.handle_query(build_query(f"h{i}.x"))
+        assert r.stats["evicted"] == 2
+        # h0/h1 evicted; h4 cached
+        r.handle_query(build_query("h4.x"))
+        assert r.stats["cache_hits"] == 1
+        n = len(calls)
+        r.handle_query(build_query("h0.x"))
+        assert len(calls) == n + 1            # refetched
+
+    def test_negative_cache(self):
+        from bng_amd.dns.resolver import build_query
+        r, calls = self._mk({})               # every name NXDOMAIN
+        q = build_query("missing.example")
+        r.handle_query(q)
+        resp2 = r.handle_query(q)
+        assert len(calls) == 1                # second hit served locally
+        assert r.stats["negative_hits"] == 1
+        import struct as st
+        assert st.unpack_from(">H", resp2, 2)[0] & 0xF == 3   # NXDOMAIN
+
+    def test_cleanup_expired(self):
+        from bng_amd.dns.resolver import build_query
+        r, _ = self._mk({"a.x": ["10.0.0.1"]}, min_ttl=1, max_ttl=1)
+        r.handle_query(build_query("a.x"))
+        assert r.cleanup(now=time.time() + 5) == 1
+        assert r.cleanup() == 0
