@@ -219,3 +219,42 @@ class TestGPUTableSync:
         assert b.export_subscribers() and \
             {e["ip"] for e in b.export_subscribers()} == \
             {0x0A000105, 0x0A000106}
+
+
+class TestFailoverControllerCoverage:
+    """Disabled mode, multi-handler, stats, status (ref
+    pkg/ha/failover_test.go)."""
+
+    def test_disabled_ignores_health_events(self):
+        from bng_amd.ha.failover import FailoverController, STATE_NORMAL
+        from bng_amd.ha.health_monitor import (EVENT_PARTNER_DOWN,
+                                               HealthEvent)
+        fc = FailoverController("n1", ROLE_STANDBY, enabled=False)
+        fc.handle_health_event(HealthEvent(EVENT_PARTNER_DOWN, "n2",
+                                           time.time(), 3))
+        assert fc.role == ROLE_STANDBY and fc.state == STATE_NORMAL
+        # forced still works even when automatic failover is off
+        assert fc.force_failover()
+        assert fc.role == ROLE_ACTIVE and fc.stats["forced"] == 1
+
+    def test_multiple_handlers_and_stats(self):
+        from bng_amd.ha.failover import FailoverController
+        got1, got2 = [], []
+        fc = FailoverController("n1", ROLE_STANDBY,
+                                role_change_callback=got1.append)
+        fc.on_role_change(got2.append)
+        fc.force_failover()
+        fc.force_failback()
+        assert got1 == [ROLE_ACTIVE, ROLE_STANDBY]
+        assert got2 == [ROLE_ACTIVE, ROLE_STANDBY]
+        st = fc.status()
+        assert st["failovers"] == 1 and st["failbacks"] == 1
+        assert st["at_original_role"]
+
+    def test_failover_when_already_active_refused(self):
+        from bng_amd.ha.failover import FailoverController
+        fc = FailoverController("n1", ROLE_ACTIVE)
+        assert not fc.initiate_failover()
+        # failback at original (standby) role refused
+        fc2 = FailoverController("n2", ROLE_STANDBY)
+        assert not fc2.initiate_failback()
