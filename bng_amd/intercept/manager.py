@@ -83,12 +83,22 @@ class Manager:
     # ----------------------------------------------------------- warrants
     def add_warrant(self, target_subscriber: str, authority: str = "",
                     case_reference: str = "", duration: float = 0.0,
-                    intercept_type: str = "iri") -> Warrant:
+                    intercept_type: str = "iri",
+                    valid_from: float = 0.0) -> Warrant:
+        """ref AddWarrant manager.go:141-180: a future valid_from makes
+        the warrant PENDING (not matching traffic) until the sweep or a
+        lookup crosses the start time; intercept_type per ETSI:
+        iri (metadata) | cc (content) | iri+cc."""
+        if not target_subscriber:
+            raise ValueError("warrant needs a target subscriber")
+        if intercept_type not in ("iri", "cc", "iri+cc"):
+            raise ValueError(f"unknown intercept type {intercept_type}")
+        start = valid_from or time.time()
         w = Warrant(id=uuid.uuid4().hex[:12],
                     target_subscriber=target_subscriber,
                     authority=authority, case_reference=case_reference,
-                    start_time=time.time(),
-                    end_time=time.time() + duration if duration else 0.0,
+                    start_time=start,
+                    end_time=start + duration if duration else 0.0,
                     intercept_type=intercept_type)
         with self._lock:
             self.warrants[w.id] = w
@@ -113,9 +123,25 @@ class Manager:
             out = []
             for wid in self.by_target.get(subscriber, []):
                 w = self.warrants.get(wid)
-                if w and w.active and (w.end_time == 0 or w.end_time > now):
+                if w and w.active and w.start_time <= now and \
+                        (w.end_time == 0 or w.end_time > now):
                     out.append(w)
             return out
+
+    def warrant_status(self, warrant_id: str) -> str:
+        """pending | active | expired | revoked (ref WarrantStatus)."""
+        now = time.time()
+        with self._lock:
+            w = self.warrants.get(warrant_id)
+        if w is None:
+            return "unknown"
+        if not w.active:
+            return "revoked"
+        if now < w.start_time:
+            return "pending"
+        if w.end_time and now >= w.end_time:
+            return "expired"
+        return "active"
 
     def is_target(self, subscriber: str) -> bool:
         return bool(self._active_warrants(subscriber))

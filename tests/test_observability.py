@@ -361,3 +361,36 @@ class TestSubscriberManagerLifecycle:
         assert m.open_session("a") is not None
         assert m.open_session("b") is None
         assert m.stats["rejected_capacity"] == 1
+
+
+class TestInterceptWarrants:
+    """Pending warrants / validation / types (ref
+    pkg/intercept/manager_test.go)."""
+
+    def test_pending_warrant_activates_at_valid_from(self):
+        m = Intercept()
+        w = m.add_warrant("sub-1", valid_from=time.time() + 100)
+        assert m.warrant_status(w.id) == "pending"
+        assert not m.is_target("sub-1")
+        w.start_time = time.time() - 1            # time passes
+        assert m.warrant_status(w.id) == "active"
+        assert m.is_target("sub-1")
+
+    def test_validation_and_types(self):
+        m = Intercept()
+        with pytest.raises(ValueError):
+            m.add_warrant("")
+        with pytest.raises(ValueError):
+            m.add_warrant("sub-1", intercept_type="metadata")
+        for t in ("iri", "cc", "iri+cc"):
+            assert m.add_warrant("sub-t", intercept_type=t)
+
+    def test_status_transitions(self):
+        m = Intercept()
+        w = m.add_warrant("sub-1", duration=0.01)
+        assert m.warrant_status(w.id) == "active"
+        time.sleep(0.05)
+        assert m.warrant_status(w.id) == "expired"
+        m.revoke_warrant(w.id)
+        assert m.warrant_status(w.id) == "revoked"
+        assert m.warrant_status("nope") == "unknown"
