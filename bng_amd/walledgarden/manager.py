@@ -133,3 +133,18 @@ class Manager:
                 h(e)
             except Exception:
                 pass
+
+    def list_macs(self, state: Optional[str] = None):
+        """MACs currently tracked, optionally filtered by state (ref
+        manager.go ListWalledGardenMACs)."""
+        with self._lock:
+            return sorted(mac for mac, e in self.entries.items()
+                          if state is None or e.state == state)
+
+    def get_stats(self):
+        with self._lock:
+            by_state: Dict[str, int] = {}
+            for e in self.entries.values():
+                by_state[e.state] = by_state.get(e.state, 0) + 1
+            return {**self.stats, "tracked": len(self.entries),
+                    **{f"state_{k}": v for k, v in by_state.items()}}

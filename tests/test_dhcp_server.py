@@ -297,3 +297,23 @@ class TestServerCoverage:
         [t.join() for t in ts]
         assert not errs
         assert len(set(ips.values())) == 32
+
+
+class TestWalledGardenManager:
+    """MAC listing + stats (ref pkg/walledgarden/manager_test.go)."""
+
+    def test_list_macs_and_stats(self):
+        from bng_amd.walledgarden.manager import Manager as WG
+        m = WG(portal_ip="10.0.0.1")
+        m.add("aa:01", "10.0.1.5")
+        m.add("aa:02", "10.0.1.6")
+        m.activate("aa:02")
+        m.add("aa:03", "10.0.1.7")
+        m.block("aa:03", reason="fraud")
+        assert m.list_macs() == ["aa:01", "aa:02", "aa:03"]
+        assert m.list_macs("walledgarden") == ["aa:01"]
+        st = m.get_stats()
+        assert st["tracked"] == 3 and st["state_walledgarden"] == 1
+        assert st["added"] == 3 and st["blocked"] == 1
+        m.remove("aa:03")
+        assert m.get_stats()["tracked"] == 2
