@@ -60,13 +60,20 @@ class AllocationRecord:
 
 class DistributedAllocator:
     def __init__(self, store: Store, pool_id: str, cidr: str,
-                 mode: str = MODE_SESSION, prefix_length: int = 32,
+                 mode: str = MODE_SESSION, prefix_length: int = 0,
                  grace_period: int = 1, epoch_interval: float = 0.0,
                  node_id: str = "node-1"):
         self.store = store
         self.pool_id = pool_id
         self.mode = mode
         self.node_id = node_id
+        if prefix_length == 0:
+            # host route for v4, /64-per-subscriber style /56 PD default
+            # left to the caller for v6 — here: smallest sane default
+            import ipaddress as _ip
+            n = _ip.ip_network(cidr, strict=False)
+            prefix_length = 32 if n.version == 4 else \
+                min(128, n.prefixlen + 16)
         self.local = EpochBitmapAllocator(cidr, prefix_length, grace_period)
         self._lock = threading.RLock()
         self._prefix = f"alloc/{pool_id}/"

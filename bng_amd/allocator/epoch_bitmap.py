@@ -26,11 +26,14 @@ class EpochBitmapAllocator:
     def __init__(self, base_network: str, prefix_length: int = 32,
                  grace_period: int = 1):
         net = ipaddress.ip_network(base_network, strict=False)
-        if net.version != 4:
-            raise ValueError("epoch bitmap is IPv4")
-        if not (net.prefixlen <= prefix_length <= 32):
+        max_plen = 32 if net.version == 4 else 128
+        if not (net.prefixlen <= prefix_length <= max_plen):
             raise ValueError("prefix length out of range")
+        if prefix_length - net.prefixlen > 24:
+            raise ValueError("pool too large for the epoch bitmap "
+                             "(use allocator.bitmap sparse mode)")
         self.net = net
+        self._shift = max_plen - prefix_length   # v6 delegation stride
         self.prefix_length = prefix_length
         self.total = 1 << (prefix_length - net.prefixlen)
         self.generations = bytearray((self.total + 3) // 4)  # 2 bits each
@@ -59,10 +62,11 @@ class EpochBitmapAllocator:
         self.generations[idx >> 2] = (b & ~(3 << shift) & 0xFF) | (gen << shift)
 
     def _idx_to_ip(self, idx: int) -> str:
-        return str(self.net.network_address + idx)
+        return str(self.net.network_address + (idx << self._shift))
 
     def _ip_to_idx(self, ip: str) -> int:
-        off = int(ipaddress.IPv4Address(ip)) - int(self.net.network_address)
+        off = (int(ipaddress.ip_address(ip)) -
+               int(self.net.network_address)) >> self._shift
         if not (0 <= off < self.total):
             raise NotFoundError(ip)
         return off

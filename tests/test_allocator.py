@@ -152,3 +152,56 @@ class TestDistributed:
         assert b.lookup("sub-1") == p
         assert b.allocate("sub-2") != p
         b.close()
+
+
+class TestDistributedIntegration:
+    """Watch notification / concurrency / IPv6 (ref
+    distributed_integration_test.go)."""
+
+    def test_watch_notification_on_allocate_release(self):
+        store = MemoryStore()
+        events = []
+        store.watch("alloc/p/", lambda ev: events.append(
+            (ev.type, ev.key)))
+        a = DistributedAllocator(store, "p", "10.5.0.0/24", MODE_SESSION)
+        a.allocate("sub-1")
+        a.release("sub-1")
+        a.close()
+        types = [t for t, _ in events]
+        assert "put" in types and "delete" in types
+        assert all(k.startswith("alloc/p/") for _, k in events)
+
+    def test_concurrent_allocations_unique(self):
+        import threading
+        store = MemoryStore()
+        allocs = [DistributedAllocator(store, "p", "10.5.0.0/24",
+                                       MODE_SESSION, node_id=f"n{k}")
+                  for k in range(2)]
+        got, errs = {}, []
+
+        def worker(k):
+            try:
+                got[k] = allocs[k % 2].allocate(f"sub-{k}")
+            except Exception as e:      # pragma: no cover
+                errs.append(e)
+        ts = [threading.Thread(target=worker, args=(k,))
+              for k in range(40)]
+        [t.start() for t in ts]
+        [t.join() for t in ts]
+        assert not errs
+        assert len(set(got.values())) == 40   # no duplicate addresses
+        for a in allocs:
+            a.close()
+
+    def test_ipv6_prefix_mode(self):
+        store = MemoryStore()
+        a = DistributedAllocator(store, "p6", "2001:db8:100::/48",
+                                 MODE_SESSION)
+        p1 = a.allocate("sub-1")
+        p2 = a.allocate("sub-2")
+        assert p1 != p2 and p1.startswith("2001:db8:100:")
+        assert a.lookup("sub-1") == p1
+        b = DistributedAllocator(store, "p6", "2001:db8:100::/48",
+                                 MODE_SESSION, node_id="n2")
+        assert b.lookup("sub-1") == p1
+        a.close(); b.close()
