@@ -43,8 +43,26 @@ class AccountingManager:
         self._lock = threading.RLock()
         self._stop = threading.Event()
         self._threads: List[threading.Thread] = []
+        self._counter_fetcher = None
         if persist_path:
             self._recover_orphans()
+
+    def set_counter_fetcher(self, fn):
+        """Live octet-counter pull before each interim/stop record (ref
+        accounting.go SetCounterFetcher) — the dataplane integration
+        point: fn(session_record) -> (input_octets, output_octets) or
+        None to keep the pushed values."""
+        self._counter_fetcher = fn
+
+    def _refresh_counters(self, rec: "SessionRecord"):
+        if self._counter_fetcher is None:
+            return
+        try:
+            got = self._counter_fetcher(rec)
+        except Exception:
+            return
+        if got:
+            rec.input_octets, rec.output_octets = got
 
     # ---------------------------------------------------------- lifecycle
     def start(self):
@@ -80,6 +98,10 @@ class AccountingManager:
                 rec.output_octets = output_octets
 
     def stop_session(self, session_id: str, terminate_cause: int = 1):
+        with self._lock:
+            rec = self.sessions.get(session_id)
+        if rec is not None:
+            self._refresh_counters(rec)
         with self._lock:
             rec = self.sessions.pop(session_id, None)
         if rec is None:
@@ -120,6 +142,7 @@ class AccountingManager:
                         rec.last_interim = now
                         due.append(rec)
             for rec in due:
+                self._refresh_counters(rec)
                 if not self._try_send(rp.ACCT_INTERIM, rec):
                     self._queue(rp.ACCT_INTERIM, rec)
 

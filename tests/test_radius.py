@@ -218,3 +218,33 @@ class TestPolicyManager:
         pm.on_change(lambda p: seen.append(p.name))
         pm.add_policy(Policy("new", 1, 1))
         assert seen == ["new"]
+
+
+class TestCounterFetcher:
+    """Dataplane counter pull into accounting records (ref
+    accounting SetCounterFetcher tests)."""
+
+    def test_stop_record_carries_fetched_counters(self, server):
+        c = Client([server.addr], SECRET)
+        am = AccountingManager(c, interim_interval=9999)
+        am.set_counter_fetcher(lambda rec: (111_000, 222_000))
+        sid = am.start_session("alice", framed_ip="10.0.1.5")
+        am.stop_session(sid)
+        stop = server.acct_records[-1]
+        assert stop.get_int(rp.ACCT_STATUS_TYPE) == rp.ACCT_STOP
+        assert stop.get_int(rp.ACCT_INPUT_OCTETS) == 111_000
+        assert stop.get_int(rp.ACCT_OUTPUT_OCTETS) == 222_000
+
+    def test_fetcher_errors_keep_pushed_values(self, server):
+        c = Client([server.addr], SECRET)
+        am = AccountingManager(c, interim_interval=9999)
+
+        def boom(rec):
+            raise RuntimeError("dataplane gone")
+        am.set_counter_fetcher(boom)
+        sid = am.start_session("bob")
+        am.update_counters(sid, 10, 20)
+        am.stop_session(sid)
+        stop = server.acct_records[-1]
+        assert stop.get_int(rp.ACCT_INPUT_OCTETS) == 10
+        assert stop.get_int(rp.ACCT_OUTPUT_OCTETS) == 20
