@@ -266,9 +266,22 @@ class BNG:
                 addr, _, ras = n.partition(":")
                 self.bgp.add_neighbor(addr, int(ras or 65000))
             if a.bgp_announce_subscribers:
-                from ..routing.manager import SubscriberRouteManager
+                from ..dataplane.packets import u32_to_ip
+                from ..routing.manager import (SessionRouteIntegration,
+                                               SubscriberRouteManager)
                 self.sub_routes = SubscriberRouteManager(self.bgp).start()
                 self._defer(self.sub_routes.stop)
+                self.route_integ = SessionRouteIntegration(self.sub_routes)
+
+                def _lease_to_route(event, lease):
+                    sid = "dhcp-" + lease.mac.hex()
+                    if event == "add":
+                        self.route_integ.on_session_activate(
+                            sid, lease.mac.hex(), u32_to_ip(lease.ip))
+                    elif event == "delete":
+                        self.route_integ.on_session_terminate(
+                            sid, reason="lease-removed")
+                self.dhcp_server.on_lease_event.append(_lease_to_route)
 
         # 9. RADIUS + policies + QoS (main.go:951-985)
         from ..radius.policy import Policy, PolicyManager

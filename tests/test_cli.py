@@ -69,6 +69,12 @@ class TestRunWiring:
             app.metrics.collect_once(app.launcher, app.dhcp_server)
             text = app.metrics.render().decode()
             assert "bng_pool_allocated" in text
+            # lease -> /32 route injection through the BGP manager
+            from bng_amd.dataplane.packets import u32_to_ip
+            assert f"{u32_to_ip(ip)}/32" in app.sub_routes.installed
+            rel = dm.build_request(mac, dm.RELEASE)
+            app.dhcp_server.handle(rel)
+            assert f"{u32_to_ip(ip)}/32" not in app.sub_routes.installed
         finally:
             app.stop()
 
