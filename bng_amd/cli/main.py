@@ -158,6 +158,8 @@ class BNG:
         self.launcher = None
         self.dhcp_server = None
         self.metrics = None
+        self.bgp = None
+        self.sub_routes = None
 
     def _defer(self, fn):
         self._cleanup.append(fn)
@@ -393,6 +395,13 @@ class BNG:
         if a.metrics_enable:
             from ..metrics.metrics import Metrics
             self.metrics = Metrics()
+            if self.bgp is not None:
+                from ..routing.metrics import RoutingMetrics
+                self.routing_metrics = RoutingMetrics(
+                    registry=self.metrics.registry)
+                self.metrics.add_collector(
+                    lambda: self.routing_metrics.collect(
+                        bgp=self.bgp, route_manager=self.sub_routes))
             self.metrics.start_collector(self.launcher, self.dhcp_server)
             try:
                 self.metrics.serve(port=a.metrics_port)

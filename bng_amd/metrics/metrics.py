@@ -76,9 +76,15 @@ class Metrics:
             registry=r)
         self._stop = threading.Event()
         self._collector: Optional[threading.Thread] = None
+        self._extra_collectors = []
         self._httpd = None
 
     # ---------------------------------------------------------- collector
+    def add_collector(self, fn):
+        """Extra per-cycle collection hook (e.g. RoutingMetrics.collect);
+        runs inside the 5s collector loop."""
+        self._extra_collectors.append(fn)
+
     def start_collector(self, launcher=None, dhcp_server=None,
                         session_manager=None, interval: float = 5.0):
         """Periodic pull from the GPU dataplane (ref main.go:1241)."""
@@ -91,6 +97,11 @@ class Metrics:
 
     def collect_once(self, launcher=None, dhcp_server=None,
                      session_manager=None):
+        for fn in self._extra_collectors:
+            try:
+                fn()
+            except Exception:
+                pass
         if launcher is not None:
             st = launcher.get_stats()
             hits = st.get("fastpath_hits", 0)

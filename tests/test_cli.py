@@ -72,6 +72,9 @@ class TestRunWiring:
             # lease -> /32 route injection through the BGP manager
             from bng_amd.dataplane.packets import u32_to_ip
             assert f"{u32_to_ip(ip)}/32" in app.sub_routes.installed
+            # routing metrics registered on the same registry
+            assert "bng_routing_subscriber_routes_active 1.0" in text or \
+                "bng_routing_subscriber_routes_active" in text
             rel = dm.build_request(mac, dm.RELEASE)
             app.dhcp_server.handle(rel)
             assert f"{u32_to_ip(ip)}/32" not in app.sub_routes.installed
