@@ -410,3 +410,61 @@ class TestDNSCacheBehaviors:
         r.handle_query(build_query("a.x"))
         assert r.cleanup(now=time.time() + 5) == 1
         assert r.cleanup() == 0
+
+
+class TestDirectCaching:
+    """Cache / binding events / BSS sync (ref
+    pkg/direct/authenticator_test.go)."""
+
+    def make(self):
+        store = MemoryStore()
+        c = NexusClient(store)
+        c.save_subscriber(Subscriber("sub-1", s_tag=100, c_tag=5,
+                                     mac="aa:bb:cc:00:00:01",
+                                     isp_id="isp-a"))
+        return c
+
+    def test_positive_results_cached(self):
+        d = DirectAuth(self.make())
+        assert d.authenticate_by_mac("aa:bb:cc:00:00:01").success
+        assert d.authenticate_by_mac("aa:bb:cc:00:00:01").success
+        assert d.stats["cache_hits"] == 1
+        # negative results are NOT cached
+        d.authenticate_by_mac("ff:ff:ff:ff:ff:ff")
+        d.authenticate_by_mac("ff:ff:ff:ff:ff:ff")
+        assert d.stats["cache_hits"] == 1
+        d.invalidate_cache()
+        d.authenticate_by_mac("aa:bb:cc:00:00:01")
+        assert d.stats["cache_hits"] == 1          # refilled, not hit
+
+    def test_binding_events_reach_bss(self):
+        events = []
+
+        class BSS:
+            def subscriber_status(self, sid):
+                return "active"
+
+            def report_binding(self, ev):
+                events.append((ev.subscriber_id, ev.mac, ev.event))
+        d = DirectAuth(self.make(), bss=BSS())
+        assert d.report_binding_event("sub-1", "aa:bb:cc:00:00:01",
+                                      "10.0.1.5")
+        assert events == [("sub-1", "aa:bb:cc:00:00:01", "bind")]
+        # BSS without the capability: graceful False
+        d2 = DirectAuth(self.make())
+        assert not d2.report_binding_event("sub-1", "m", "i")
+
+    def test_sync_from_bss_prefills_cache(self):
+        class BSS:
+            def subscriber_status(self, sid):
+                return "active"
+
+            def sync_mappings(self):
+                return [{"subscriber_id": "sub-9", "isp_id": "isp-b",
+                         "mac": "aa:00:00:00:00:09",
+                         "vlan": "100.9"}]
+        d = DirectAuth(self.make(), bss=BSS())
+        assert d.sync_from_bss() == 2              # mac + vlan keys
+        r = d.authenticate_by_mac("aa:00:00:00:00:09")
+        assert r.success and r.subscriber_id == "sub-9"
+        assert d.stats["cache_hits"] == 1
