@@ -468,3 +468,34 @@ class TestDirectCaching:
         r = d.authenticate_by_mac("aa:00:00:00:00:09")
         assert r.success and r.subscriber_id == "sub-9"
         assert d.stats["cache_hits"] == 1
+
+
+class TestPONCoverage:
+    """Duplicate discovery / callbacks / offline-rediscovery (ref
+    pkg/pon/manager_test.go)."""
+
+    def test_duplicate_discovery_is_one_nte(self):
+        from bng_amd.pon.manager import Manager as PON
+        store = MemoryStore()
+        pon = PON(store)
+        events = []
+        pon.on_event(lambda ev, n: events.append((ev, n.id)))
+        a = pon.ont_discovered("SER001", "pon0/1")
+        b = pon.ont_discovered("SER001", "pon0/1")     # re-announce
+        assert a.id == b.id
+        assert pon.stats["discovered"] == 1            # counted once
+        assert len(pon.list_ntes()) == 1
+        assert events == [("discovered", a.id)] * 2    # both announced
+
+    def test_offline_then_rediscovered_keeps_provisioning(self):
+        from bng_amd.pon.manager import Manager as PON, QoSProfile
+        store = MemoryStore()
+        pon = PON(store)
+        pon.add_profile(QoSProfile("res-100", 100_000_000, 20_000_000))
+        nte = pon.ont_discovered("SER002", "pon0/2")
+        pon.provision(nte.id, profile="res-100", s_tag=100, c_tag=7)
+        pon.ont_offline(nte.id)
+        assert [n.state for n in pon.list_ntes()] == ["offline"]
+        back = pon.ont_discovered("SER002", "pon0/2")
+        assert back.provisioned and (back.s_tag, back.c_tag) == (100, 7)
+        assert pon.stats["offline"] == 1
