@@ -1,0 +1,123 @@
+#!/usr/bin/env python3
+"""Round-end stress evidence: (1) long soak, (2) CRUD + sweeps under
+sustained traffic (tombstone churn), (3) DHCP latency under saturating
+data load.  Prints one JSON line per phase."""
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import numpy as np
+
+
+def main():
+    import bench
+    import torch
+    from bng_amd.dataplane import abi
+    from bng_amd.dataplane.launcher import HipLauncher
+    from bng_amd.dataplane.packets import ip2u32
+    NOW = 1_700_000_000
+    n_subs = 1_000_000
+    base = 21
+    l = HipLauncher("cuda:0", sub_log2=base, sess_log2=base + 1,
+                    eim_log2=base, subnat_log2=base, qos_log2=base,
+                    binding_log2=base)
+    l.set_server_config(b"\x02\x00\x00\x00\x00\x01", ip2u32("10.255.255.1"))
+    l.add_pool(1, ip2u32("10.0.0.0"), 8, ip2u32("10.255.255.1"),
+               ip2u32("8.8.8.8"))
+    bench.build_tables(l, 0, 1, n_subs, NOW)
+    batch = 1 << 20
+    d_np, ln = bench.gen_batch(batch, n_subs, 0.1, 512, 99)
+    p = torch.from_numpy(d_np).cuda()
+    w = torch.empty_like(p)
+    lt = torch.from_numpy(ln.view(np.int16)).cuda()
+    ns = NOW * 10**9
+
+    # phase 1: 1000-step soak with a sweep every 100 steps
+    for k in range(5):
+        w.copy_(p)
+        l.uplink(w, lt, now_ns=ns + k, now_sec=NOW, sort_by_type=True)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    steps = 1000
+    for k in range(steps):
+        w.copy_(p)
+        l.uplink(w, lt, now_ns=ns + (k + 5) * 10**6, now_sec=NOW,
+                 sort_by_type=True)
+        if k % 100 == 99:
+            # periodic timeout sweep: expires everything idle > timeouts
+            l.sweep_nat(now_ns=ns + (k + 5) * 10**6)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    nstat = l.nat_get_stats()
+    print(json.dumps({"phase": "soak1000+sweeps",
+                      "mpps": round(batch * steps / dt / 1e6, 1),
+                      "sessions_created": nstat["sessions_created"],
+                      "sessions_expired": nstat["sessions_expired"]}),
+          flush=True)
+
+    # phase 2: CRUD churn under traffic — rotate qos policies + nat
+    # blocks for a 64k-subscriber slice every batch
+    ips = (np.uint64(ip2u32("10.0.0.0") + 2) +
+           np.arange(n_subs, dtype=np.uint64)).astype(np.uint32)
+    t0 = time.perf_counter()
+    steps2 = 200
+    for k in range(steps2):
+        w.copy_(p)
+        l.uplink(w, lt, now_ns=ns + (k + 3000) * 10**6, now_sec=NOW,
+                 sort_by_type=True)
+        sl = ips[(k * 64 * 1024) % n_subs:][:1024]
+        for ip in sl[:4]:
+            l.set_qos_policy(int(ip), 10**9, 4 << 20,
+                             direction="ingress",
+                             now_ns=ns + k * 10**6)
+        if k % 10 == 5:
+            l.remove_qos_policy(int(sl[5]), direction="ingress")
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    print(json.dumps({"phase": "crud-under-traffic",
+                      "mpps": round(batch * steps2 / dt / 1e6, 1)}),
+          flush=True)
+
+    # phase 3: DHCP latency while a saturating data stream runs on a
+    # second stream
+    s2 = torch.cuda.Stream()
+    d2_np, l2 = bench.gen_batch(2048, n_subs, 1.0, 512, 7)
+    lp = torch.from_numpy(d2_np).cuda()
+    lw = torch.empty_like(lp)
+    llt = torch.from_numpy(l2.view(np.int16)).cuda()
+    stop = [False]
+    lats = []
+    import threading
+
+    def flood():
+        with torch.cuda.stream(s2):
+            k = 0
+            while not stop[0]:
+                w.copy_(p)
+                l.uplink(w, lt, now_ns=ns + k, now_sec=NOW,
+                         sort_by_type=False)
+                k += 1
+                if k % 8 == 0:
+                    s2.synchronize()
+    th = threading.Thread(target=flood)
+    th.start()
+    time.sleep(0.5)
+    for r in range(100):
+        torch.cuda.synchronize()
+        t = time.perf_counter()
+        lw.copy_(lp)
+        l.dhcp_fastpath(lw, llt, now_sec=NOW)
+        torch.cuda.synchronize()
+        lats.append((time.perf_counter() - t) * 1e6)
+    stop[0] = True
+    th.join()
+    lats.sort()
+    print(json.dumps({"phase": "latency-under-saturation",
+                      "p50_us": round(lats[50], 1),
+                      "p99_us": round(lats[99], 1)}), flush=True)
+
+
+if __name__ == "__main__":
+    main()

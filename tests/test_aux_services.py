@@ -499,3 +499,46 @@ class TestPONCoverage:
         back = pon.ont_discovered("SER002", "pon0/2")
         assert back.provisioned and (back.s_tag, back.c_tag) == (100, 7)
         assert pon.stats["offline"] == 1
+
+
+class TestAuthenticatedTransport:
+    """Header-injecting HTTP transport (ref
+    pkg/deviceauth/transport_test.go)."""
+
+    def test_headers_injected_and_verified(self):
+        import threading
+        from http.server import BaseHTTPRequestHandler, HTTPServer
+        from bng_amd.deviceauth.authenticator import (AuthenticatedSession,
+                                                      PSKAuthenticator)
+        auth = PSKAuthenticator(b"shared-key")
+        seen = {}
+
+        class H(BaseHTTPRequestHandler):
+            def do_GET(self):
+                seen.update({k: v for k, v in self.headers.items()})
+                try:
+                    seen["verified"] = auth.verify(dict(self.headers))
+                except Exception as e:
+                    seen["verified"] = f"error:{e}"
+                self.send_response(200)
+                self.end_headers()
+                self.wfile.write(b"ok")
+
+            def log_message(self, *a):
+                pass
+        srv = HTTPServer(("127.0.0.1", 0), H)
+        t = threading.Thread(target=srv.serve_forever, daemon=True)
+        t.start()
+        try:
+            s = AuthenticatedSession(auth, "nte-42")
+            r = s.get(f"http://127.0.0.1:{srv.server_port}/cfg")
+            assert r.status_code == 200
+            assert seen["verified"] == "nte-42"    # server-side verify
+            assert "X-Device-Id" in seen or "X-Device-ID" in seen or \
+                any(k.lower() == "x-device-id" for k in seen)
+            # user headers preserved alongside auth headers
+            s.get(f"http://127.0.0.1:{srv.server_port}/cfg",
+                  headers={"X-Custom": "1"})
+            assert seen.get("X-Custom") == "1"
+        finally:
+            srv.shutdown()
