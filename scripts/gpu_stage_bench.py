@@ -85,6 +85,32 @@ def main():
     res["argsort"] = timeit(
         lambda: torch.argsort(cls, stable=True).to(torch.int32))
 
+    # downlink: learn each flow's SNAT (ip, port) from the rewritten
+    # uplink batch, build the return traffic, DNAT+QoS-egress it
+    w.copy_(d_data)
+    l.uplink(w, l_data, now_ns=tick(), now_sec=NOW, sort_by_type=False)
+    torch.cuda.synchronize()
+    up_host = w.cpu().numpy()
+    nat_ip = up_host[:, 26:30].copy()
+    nat_port = up_host[:, 34:36].copy()
+    ret = np.zeros_like(data_np)
+    ret[:, :64] = data_np[:, :64]
+    ret[:, 0:6] = data_np[:, 6:12]          # eth swap
+    ret[:, 6:12] = data_np[:, 0:6]
+    ret[:, 26:30] = data_np[:, 30:34]       # ip src = orig dst
+    ret[:, 30:34] = nat_ip                  # ip dst = SNAT public
+    ret[:, 34:36] = data_np[:, 36:38]       # sport = orig dport
+    ret[:, 36:38] = nat_port                # dport = SNAT port
+    ret[:, 24:26] = 0     # ip csum zeroed: kernels update incrementally,
+    ret[:, 40:42] = 0     # throughput timing does not verify checksums
+    d_ret = torch.from_numpy(ret).cuda()
+    w_ret = torch.empty_like(d_ret)
+
+    def down():
+        w_ret.copy_(d_ret)
+        l.downlink(w_ret, l_data, now_ns=tick())
+    res["copy+downlink"] = timeit(down)
+
     for k, v in res.items():
         print(f"{k:24s} {v:9.1f} us  ({N / v:.0f} pkt/us)" if v else k)
     print(f"nat44 alone ~= {res['copy+nat44'] - res['copy_268MB']:.1f} us")
