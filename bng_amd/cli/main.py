@@ -259,7 +259,7 @@ class BNG:
 
         # 8. routing (main.go:901-939)
         if a.bgp_enable:
-            from ..routing.bgp import BFDManager, BGPController
+            from ..routing.bgp import BGPController
             from ..routing.frr import FakeExecutor, VtyshExecutor
             exe = VtyshExecutor() if os.path.exists("/usr/bin/vtysh") \
                 else FakeExecutor()

@@ -4,7 +4,7 @@ implemented directly."""
 from __future__ import annotations
 
 import struct
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 DISCOVER, OFFER, REQUEST, DECLINE, ACK, NAK, RELEASE, INFORM = range(1, 9)
 MAGIC = 0x63825363

@@ -20,7 +20,7 @@ import struct
 import threading
 import time
 from dataclasses import dataclass, field
-from typing import Callable, Dict, Optional
+from typing import Dict, Optional
 
 from ..dataplane.packets import ip2u32, u32_to_ip
 from . import message as dm

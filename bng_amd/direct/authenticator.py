@@ -4,7 +4,6 @@ VLAN the frame arrived on — validated against Nexus and an optional
 BSS (business support system) client."""
 from __future__ import annotations
 
-import threading
 from dataclasses import dataclass
 from typing import Dict, Optional, Protocol
 

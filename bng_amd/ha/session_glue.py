@@ -6,7 +6,6 @@ feeding the HASyncer protocol)."""
 from __future__ import annotations
 
 import time
-from typing import Optional
 
 from ..dataplane.packets import u32_to_ip, ip2u32
 from ..dhcp.server import DHCPServer, Lease

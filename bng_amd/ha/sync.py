@@ -8,7 +8,7 @@ import json
 import queue
 import threading
 import time
-from typing import Callable, List, Optional
+from typing import List, Optional
 
 from .protocol import (ROLE_ACTIVE, ROLE_STANDBY, SYNC_ADD, SYNC_DELETE,
                        SYNC_FULL, SYNC_HEARTBEAT, SYNC_UPDATE,

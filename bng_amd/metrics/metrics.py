@@ -6,8 +6,7 @@ HTTP endpoint."""
 from __future__ import annotations
 
 import threading
-import time
-from typing import Dict, Optional
+from typing import Optional
 
 from prometheus_client import (CollectorRegistry, Counter, Gauge, Histogram,
                                generate_latest)

@@ -12,7 +12,7 @@ import time
 from typing import Callable, Dict, List, Optional, Tuple
 
 from ..dataplane.abi import fnv1a64
-from .model import Device, IPPool, ISPConfig, NTE, Subscriber
+from .model import IPPool, ISPConfig, NTE, Subscriber
 from .store import Store, TypedStore
 
 

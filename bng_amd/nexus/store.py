@@ -7,7 +7,6 @@ components against one shared MemoryStore (SURVEY.md §4 layer 5).
 """
 from __future__ import annotations
 
-import fnmatch
 import json
 import threading
 from dataclasses import dataclass, field

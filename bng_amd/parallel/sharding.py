@@ -14,7 +14,7 @@ count minimal.
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 
 def bucket_by_owner(data, lens, owner, world_size: int):

@@ -3,9 +3,8 @@ frames, PPP protocol payloads, LCP/IPCP/IPV6CP option TLVs
 (ref pkg/pppoe/protocol handling in server.go/lcp.go/ipcp.go)."""
 from __future__ import annotations
 
-import os
 import struct
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 ETH_PPPOE_DISC = 0x8863
 ETH_PPPOE_SESS = 0x8864

@@ -8,7 +8,7 @@ import hashlib
 import hmac
 import os
 import struct
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 # codes
 ACCESS_REQUEST = 1

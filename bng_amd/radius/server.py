@@ -4,9 +4,8 @@ Verifies Message-Authenticator and accounting Request Authenticators."""
 from __future__ import annotations
 
 import socket
-import struct
 import threading
-from typing import Callable, Dict, List, Optional, Tuple
+from typing import Dict, List, Optional, Tuple
 
 from . import packet as rp
 

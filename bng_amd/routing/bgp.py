@@ -7,7 +7,7 @@ import threading
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
 
-from .frr import FRRError, FRRExecutor
+from .frr import FRRExecutor
 
 
 @dataclass
