@@ -257,3 +257,93 @@ class SecurityAuditor:
             "mtls_auth", category=CAT_AUTH,
             outcome="success" if success else "failure",
             subscriber=device_id, subject=subject, error=error)
+
+
+class IPFIXExporter:
+    """RFC 7011 IPFIX export of NAT audit events (ref
+    pkg/audit/export.go:143-315).  Emits a template set (id 256:
+    sourceIPv4Address, postNATSourceIPv4Address, sourceTransportPort,
+    postNAPTSourceTransportPort, protocolIdentifier, flowStartSeconds)
+    on the first message and every template_refresh records thereafter,
+    then one data set per event; transport is a pluggable send(bytes)
+    (UDP socket in production, capture list in tests)."""
+
+    TEMPLATE_ID = 256
+    # (information element id, length)
+    FIELDS = [(8, 4), (225, 4), (7, 2), (227, 2), (4, 1), (150, 4)]
+
+    def __init__(self, send, observation_domain: int = 1,
+                 template_refresh: int = 100):
+        import struct as st
+        self._st = st
+        self.send = send
+        self.domain = observation_domain
+        self.template_refresh = template_refresh
+        self.seq = 0
+        self._since_template = None     # None => template never sent
+
+    def _msg(self, sets: bytes, now: int) -> bytes:
+        st = self._st
+        return st.pack(">HHIII", 10, 16 + len(sets), now, self.seq,
+                       self.domain) + sets
+
+    def _template_set(self) -> bytes:
+        st = self._st
+        body = st.pack(">HH", self.TEMPLATE_ID, len(self.FIELDS))
+        for ie, ln in self.FIELDS:
+            body += st.pack(">HH", ie, ln)
+        return st.pack(">HH", 2, 4 + len(body)) + body
+
+    def export(self, ev: "Event") -> bool:
+        """NAT-category events only (ref Export :210-214)."""
+        if ev.category != CAT_NAT:
+            return False
+        st = self._st
+        d = ev.details
+        rec = st.pack(
+            ">IIHHBI",
+            int(d.get("private_ip", 0) or 0),
+            int(d.get("public_ip", 0) or 0),
+            int(d.get("private_port", 0) or 0),
+            int(d.get("public_port", 0) or 0),
+            int(d.get("protocol", 0) or 0),
+            int(ev.timestamp))
+        sets = b""
+        if self._since_template is None or \
+                self._since_template >= self.template_refresh:
+            sets += self._template_set()
+            self._since_template = 0
+        sets += st.pack(">HH", self.TEMPLATE_ID, 4 + len(rec)) + rec
+        self.send(self._msg(sets, int(ev.timestamp)))
+        self.seq += 1
+        self._since_template += 1
+        return True
+
+
+class KafkaExporter:
+    """Topic-routing audit exporter with a pluggable producer (the
+    reference ships this as a stub over kafka-go, export.go:436-533;
+    ours takes produce(topic, key, value) so any client slots in)."""
+
+    def __init__(self, produce=None, topic: str = "bng-audit",
+                 topic_by_category: bool = False,
+                 topic_prefix: str = "bng-audit-",
+                 key_field: str = "subscriber"):
+        self.produce = produce or (lambda t, k, v: None)
+        self.topic = topic
+        self.topic_by_category = topic_by_category
+        self.topic_prefix = topic_prefix
+        self.key_field = key_field
+        self.exported = 0
+
+    def export(self, ev: "Event"):
+        import json as _json
+        topic = (self.topic_prefix + ev.category
+                 if self.topic_by_category else self.topic)
+        key = getattr(ev, self.key_field, "") or ev.id
+        value = _json.dumps({
+            "id": ev.id, "category": ev.category, "action": ev.action,
+            "timestamp": ev.timestamp, "subscriber": ev.subscriber,
+            "ip": ev.ip, "outcome": ev.outcome, "details": ev.details})
+        self.produce(topic, key, value)
+        self.exported += 1

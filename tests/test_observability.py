@@ -394,3 +394,63 @@ class TestInterceptWarrants:
         m.revoke_warrant(w.id)
         assert m.warrant_status(w.id) == "revoked"
         assert m.warrant_status("nope") == "unknown"
+
+
+class TestIPFIXAndKafkaExporters:
+    """Binary IPFIX framing + topic routing (ref pkg/audit/export.go)."""
+
+    def _nat_event(self, **d):
+        from bng_amd.audit.logger import CAT_NAT, Event
+        base = {"private_ip": 0x0A000105, "public_ip": 0xCB007101,
+                "private_port": 40000, "public_port": 2048,
+                "protocol": 17}
+        base.update(d)
+        return Event(id="e1", category=CAT_NAT, action="nat_mapping",
+                     timestamp=1700000000.0, subscriber="sub-1",
+                     ip="10.0.1.5", outcome="success",
+                     details={k: str(v) for k, v in base.items()})
+
+    def test_ipfix_message_framing(self):
+        import struct as st
+        from bng_amd.audit.logger import CAT_AUTH, Event, IPFIXExporter
+        msgs = []
+        x = IPFIXExporter(msgs.append, observation_domain=7)
+        assert x.export(self._nat_event())
+        # non-NAT events are skipped
+        assert not x.export(Event(id="e2", category=CAT_AUTH,
+                                  action="login", timestamp=0.0,
+                                  subscriber="", ip="", outcome="ok",
+                                  details={}))
+        ver, length, t, seq, dom = st.unpack_from(">HHIII", msgs[0], 0)
+        assert ver == 10 and dom == 7 and seq == 0
+        assert length == len(msgs[0])
+        # first message carries the template set (set id 2)
+        set_id, set_len = st.unpack_from(">HH", msgs[0], 16)
+        assert set_id == 2
+        tid, nfields = st.unpack_from(">HH", msgs[0], 20)
+        assert tid == 256 and nfields == 6
+        # data set follows, record decodes to the NAT 5-tuple
+        off = 16 + set_len
+        dsid, dlen = st.unpack_from(">HH", msgs[0], off)
+        assert dsid == 256
+        priv, pub, pport, natport, proto, ts = st.unpack_from(
+            ">IIHHBI", msgs[0], off + 4)
+        assert (priv, pub, pport, natport, proto) == \
+            (0x0A000105, 0xCB007101, 40000, 2048, 17)
+        # second export: no template, sequence advanced
+        x.export(self._nat_event())
+        set_id2, _ = st.unpack_from(">HH", msgs[1], 16)
+        assert set_id2 == 256
+        assert st.unpack_from(">HHIII", msgs[1], 0)[3] == 1
+
+    def test_kafka_topic_routing(self):
+        from bng_amd.audit.logger import KafkaExporter
+        got = []
+        x = KafkaExporter(lambda t, k, v: got.append((t, k)),
+                          topic_by_category=True)
+        x.export(self._nat_event())
+        assert got == [("bng-audit-nat", "sub-1")]
+        x2 = KafkaExporter(lambda t, k, v: got.append((t, k)))
+        x2.export(self._nat_event())
+        assert got[-1] == ("bng-audit", "sub-1")
+        assert x.exported == 1 and x2.exported == 1
