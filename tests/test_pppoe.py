@@ -381,3 +381,34 @@ class TestNCPOptionHandling:
             [(C.IPCP_OPT_IP, struct.pack(">I", cli.got_ip))]))
         replies = self._send(srv, cli, C.PROTO_IPCP, req)
         assert replies[0][1].code == C.CONF_ACK
+
+
+class TestCodecFuzz:
+    """Discovery/session decoder robustness on adversarial frames."""
+
+    def test_random_frames_never_crash(self):
+        import random
+        rng = random.Random(77)
+        srv = make_server()
+        for _ in range(2000):
+            n = rng.randrange(14, 128)
+            frame = bytearray(rng.randrange(256) for _ in range(n))
+            # half the time, make it look like PPPoE to go deeper
+            if rng.random() < 0.5:
+                struct.pack_into(">H", frame, 12,
+                                 C.ETH_PPPOE_DISC if rng.random() < 0.5
+                                 else C.ETH_PPPOE_SESS)
+            try:
+                srv.handle_frame(bytes(frame))
+            except Exception as e:      # decoder must contain errors
+                raise AssertionError(
+                    f"server crashed on fuzz frame: {e!r}") from e
+
+    def test_truncated_real_frames(self):
+        srv = make_server()
+        cli = SimClient(srv)
+        padi = C.DiscoveryPacket(C.PADI, 0, [
+            (C.TAG_SERVICE_NAME, b""), (C.TAG_HOST_UNIQ, b"H")],
+            src_mac=cli.mac).encode()
+        for cut in range(len(padi)):
+            srv.handle_frame(padi[:cut])    # must not raise

@@ -248,3 +248,38 @@ class TestCounterFetcher:
         stop = server.acct_records[-1]
         assert stop.get_int(rp.ACCT_INPUT_OCTETS) == 10
         assert stop.get_int(rp.ACCT_OUTPUT_OCTETS) == 20
+
+
+class TestPacketFuzz:
+    """Codec robustness on adversarial bytes (ref pkg/dhcp/fuzz_test.go
+    strategy applied to the RADIUS codec)."""
+
+    def test_random_bytes_never_crash(self):
+        import random
+        rng = random.Random(1234)
+        parsed = 0
+        for _ in range(2000):
+            n = rng.randrange(0, 64)
+            raw = bytes(rng.randrange(256) for _ in range(n))
+            try:
+                p = rp.Packet.decode(raw)
+                parsed += 1
+                assert 0 <= p.code <= 255
+            except (rp.RadiusError, ValueError, IndexError):
+                pass
+        # truncated/garbled real packet prefixes
+        req = rp.Packet(rp.ACCESS_REQUEST, 1, rp.random_authenticator())
+        req.add_str(rp.USER_NAME, "alice")
+        raw = req.encode(b"secret")
+        for cut in range(len(raw)):
+            try:
+                rp.Packet.decode(raw[:cut])
+            except (rp.RadiusError, ValueError, IndexError):
+                pass
+        for flip in range(0, len(raw), 3):
+            mutated = bytearray(raw)
+            mutated[flip] ^= 0xFF
+            try:
+                rp.Packet.decode(bytes(mutated))
+            except (rp.RadiusError, ValueError, IndexError):
+                pass
