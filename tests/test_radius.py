@@ -269,8 +269,8 @@ class TestPacketFuzz:
                 pass
         # truncated/garbled real packet prefixes
         req = rp.Packet(rp.ACCESS_REQUEST, 1, rp.random_authenticator())
-        req.add_str(rp.USER_NAME, "alice")
-        raw = req.encode(b"secret")
+        req.add(rp.USER_NAME, "alice")
+        raw = req.encode()
         for cut in range(len(raw)):
             try:
                 rp.Packet.decode(raw[:cut])
