@@ -709,3 +709,22 @@ class TestSubCtxMerge:
                          now_ns=NOW_NS).cpu().tolist() == [PASS]
         st_ = gpu.nat_get_stats()
         assert st_["packets_passed"] == 1 and st_["packets_snat"] == 0
+
+
+class TestMetamorphicFuzz:
+    """Compact version of scripts/gpu_fuzz.py (51k-frame run: zero
+    mismatches): sorted==unsorted and batch-split invariance on
+    adversarial frames — no routing oracle needed."""
+
+    def test_invariants_on_adversarial_frames(self):
+        import os
+        import subprocess
+        import sys
+        env = dict(os.environ, FUZZ_ROUNDS="2")
+        r = subprocess.run(
+            [sys.executable, os.path.join(
+                os.path.dirname(os.path.dirname(__file__)),
+                "scripts", "gpu_fuzz.py")],
+            env=env, capture_output=True, text=True, timeout=420)
+        assert r.returncode == 0, r.stdout + r.stderr
+        assert '"sorted_vs_unsorted": 0' in r.stdout
