@@ -264,7 +264,37 @@ struct pktctx {
   uint16_t s_tag, c_tag;
   bool tagged;
   bool l4_ok;
+  /* register window: bswapped words of bytes 0..47 when the fast parse
+   * ran (untagged ihl=5).  TA (vector-memory address) occupancy is the
+   * measured kernel bound — header rewrites happen on these registers
+   * and flush as two dwordx4 stores instead of ~10 divergent byte ops */
+  uint32_t W[12];
+  bool regs;
 };
+
+/* constant-offset register accessors (offsets fold at compile time) */
+#define RG_B(c, o) (((c).W[(o) >> 2] >> (24 - 8 * ((o) & 3))) & 0xFFu)
+BNG_DEV uint16_t rg_ld16(const pktctx& c, int o) {
+  return (uint16_t)((RG_B(c, o) << 8) | RG_B(c, o + 1));
+}
+BNG_DEV void rg_st8(pktctx& c, int o, uint32_t v) {
+  int i = o >> 2, sh = 24 - 8 * (o & 3);
+  c.W[i] = (c.W[i] & ~(0xFFu << sh)) | ((v & 0xFFu) << sh);
+}
+BNG_DEV void rg_st16(pktctx& c, int o, uint32_t v) {
+  rg_st8(c, o, v >> 8); rg_st8(c, o + 1, v);
+}
+BNG_DEV void rg_st32(pktctx& c, int o, uint32_t v) {
+  rg_st16(c, o, v >> 16); rg_st16(c, o + 2, v);
+}
+BNG_DEV void rg_flush16_47(pktctx& c) {
+  uint4* q = (uint4*)c.p;
+  q[1] = make_uint4(__builtin_bswap32(c.W[4]), __builtin_bswap32(c.W[5]),
+                    __builtin_bswap32(c.W[6]), __builtin_bswap32(c.W[7]));
+  q[2] = make_uint4(__builtin_bswap32(c.W[8]), __builtin_bswap32(c.W[9]),
+                    __builtin_bswap32(c.W[10]),
+                    __builtin_bswap32(c.W[11]));
+}
 
 /* Vectorized parse fast path: untagged IPv4 with ihl=5 (the 64B-mix hot
  * case).  Three dwordx4 loads replace ~30 dependent byte loads; fields
@@ -272,22 +302,24 @@ struct pktctx {
  * fall back to the general byte parser (VLAN, options, short frames). */
 BNG_DEV bool parse_pkt_fast(pktctx& c, uint8_t* p, int len) {
   if (len < 48 || ((uintptr_t)p & 15)) return false;
-  uint32_t W[12];
+  /* no alias pointer to c.W: an escaping pointer would force the
+   * register window into scratch memory */
   #pragma unroll
   for (int i = 0; i < 3; ++i) {
     uint4 v = ((const uint4*)p)[i];
-    W[i * 4 + 0] = __builtin_bswap32(v.x);
-    W[i * 4 + 1] = __builtin_bswap32(v.y);
-    W[i * 4 + 2] = __builtin_bswap32(v.z);
-    W[i * 4 + 3] = __builtin_bswap32(v.w);
+    c.W[i * 4 + 0] = __builtin_bswap32(v.x);
+    c.W[i * 4 + 1] = __builtin_bswap32(v.y);
+    c.W[i * 4 + 2] = __builtin_bswap32(v.z);
+    c.W[i * 4 + 3] = __builtin_bswap32(v.w);
   }
-  #define BNG_B(o) ((W[(o) >> 2] >> (24 - 8 * ((o) & 3))) & 0xFFu)
+  #define BNG_B(o) ((c.W[(o) >> 2] >> (24 - 8 * ((o) & 3))) & 0xFFu)
   #define BNG_H(o) ((BNG_B(o) << 8) | BNG_B((o) + 1))
   uint32_t ethertype = BNG_H(12);
   if (ethertype != 0x0800) return false;        /* VLAN etc: slow path */
   if (BNG_B(14) != 0x45) return false;          /* options/ihl!=5 */
   c.p = p; c.len = len; c.vlan_offset = 0; c.s_tag = c.c_tag = 0;
   c.tagged = false; c.l4_ok = false;
+  c.regs = true;
   c.ip_off = 14;
   c.proto = (uint8_t)BNG_B(23);
   c.saddr = (BNG_H(26) << 16) | BNG_H(28);
@@ -317,6 +349,7 @@ BNG_DEV bool parse_pkt_fast(pktctx& c, uint8_t* p, int len) {
 BNG_DEV bool parse_pkt_slow(pktctx& c, uint8_t* p, int len, bool want_vlan) {
   c.p = p; c.len = len; c.ip_off = -1; c.vlan_offset = 0;
   c.s_tag = c.c_tag = 0; c.tagged = false; c.l4_ok = false;
+  c.regs = false;
   if (len < 14) return false;
   uint16_t proto = ld_u16be(p + 12);
   int off = 14;
@@ -871,12 +904,48 @@ BNG_DEV int nat_egress_process(pktctx& c, const nat_tables& T, nat_flags& F,
                  is_hairpin);
   }
 
-  /* SNAT rewrite + incremental checksums (ref :752-798) */
+  /* SNAT rewrite + incremental checksums (ref :752-798).  When the
+   * fast parse ran (c.regs: untagged ihl=5, fixed offsets 14/34) the
+   * rewrite happens on the register window and flushes as two dwordx4
+   * stores — the TA unit is the measured kernel bound and this path
+   * replaces ~10 divergent byte R/W ops with 2 wide stores (the TCP
+   * checksum at byte 50 stays in memory: outside the 48-B window). */
   uint32_t old_ip = c.saddr;
+  uint16_t nat_port_host = nat_port;
+  if (c.regs) {
+    rg_st32(c, 26, nat_ip);
+    rg_st16(c, 24, csum_upd32(rg_ld16(c, 24), old_ip, nat_ip));
+    if (c.proto == 6) {
+      uint16_t old_port = rg_ld16(c, 34);
+      rg_st16(c, 34, nat_port_host);
+      uint16_t ck = ld_u16be(p + 50);
+      ck = csum_upd32(ck, old_ip, nat_ip);
+      ck = csum_upd16(ck, old_port, nat_port_host);
+      rg_flush16_47(c);
+      st_u16be(p + 50, ck);
+    } else if (c.proto == 17) {
+      uint16_t old_port = rg_ld16(c, 34);
+      rg_st16(c, 34, nat_port_host);
+      uint16_t ck = rg_ld16(c, 40);
+      if (ck != 0) {
+        ck = csum_upd32(ck, old_ip, nat_ip);
+        ck = csum_upd16(ck, old_port, nat_port_host);
+        if (ck == 0) ck = 0xFFFF;
+        rg_st16(c, 40, ck);
+      }
+      rg_flush16_47(c);
+    } else {            /* ICMP: id at 38, checksum at 36 */
+      uint16_t old_id = rg_ld16(c, 38);
+      rg_st16(c, 38, nat_port_host);
+      rg_st16(c, 36, csum_upd16(rg_ld16(c, 36), old_id, nat_port_host));
+      rg_flush16_47(c);
+    }
+    F.snat = true;
+    return BNG_FWD;
+  }
   st_u32be(p + c.ip_off + 12, nat_ip);
   uint16_t ipck = ld_u16be(p + c.ip_off + 10);
   st_u16be(p + c.ip_off + 10, csum_upd32(ipck, old_ip, nat_ip));
-  uint16_t nat_port_host = nat_port;
   if (c.proto == 6) {
     uint16_t old_port = ld_u16be(p + c.l4_off);
     st_u16be(p + c.l4_off, nat_port_host);
@@ -1325,9 +1394,15 @@ void uplink_pipeline_kernel(bng_uplink_params P) {
          * context entry then serves BOTH the NAT port-block and the
          * ingress-QoS stages — one random HBM touch where the
          * reference's hook chain pays two */
-        uint64_t mac = 0;
-        #pragma unroll
-        for (int j = 0; j < 6; ++j) mac = (mac << 8) | p[6 + j];
+        uint64_t mac;
+        if (c.regs) {
+          mac = ((uint64_t)rg_ld16(c, 6) << 32) |
+                ((uint64_t)rg_ld16(c, 8) << 16) | rg_ld16(c, 10);
+        } else {
+          mac = 0;
+          #pragma unroll
+          for (int j = 0; j < 6; ++j) mac = (mac << 8) | p[6 + j];
+        }
         uint32_t s0 = (uint32_t)bng_mix64(mac) & P.bmask;
         uint32_t s1 = (uint32_t)bng_mix64(c.saddr) & P.subctx_mask;
         uint4 f0 = ld_probe16(&P.bindings[s0]);
