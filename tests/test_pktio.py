@@ -128,3 +128,68 @@ class TestPump:
             time.sleep(0.01)
         pump.stop()
         assert sink.frames
+
+
+def _af_packet_available():
+    import socket as s
+    try:
+        sock = s.socket(s.AF_PACKET, s.SOCK_RAW, s.htons(3))
+        sock.bind(("lo", 0))
+        sock.close()
+        return True
+    except Exception:
+        return False
+
+
+@pytest.mark.skipif(not _af_packet_available(),
+                    reason="needs CAP_NET_RAW + lo")
+class TestAFPacketLoopback:
+    """Real raw-socket I/O over loopback (the NIC-edge path the
+    reference runs in XDP generic mode, loader.go fallback)."""
+
+    def test_send_and_receive_frame(self):
+        from bng_amd.dataplane.packets import build_ipv4, ip2u32
+        from bng_amd.dataplane.pktio import AFPacketIO
+        tx = AFPacketIO("lo")
+        rx = AFPacketIO("lo")
+        frame = build_ipv4("aa:bb:cc:00:00:77", "aa:bb:cc:00:00:88",
+                           ip2u32("127.0.0.1"), ip2u32("127.0.0.1"),
+                           proto=17, sport=40001, dport=40002,
+                           payload=b"bng-af-packet-test")
+        # drain anything already looping
+        rx.recv_batch(64, timeout=0.01)
+        tx.send_batch([frame])
+        got = []
+        for _ in range(50):
+            got += rx.recv_batch(16, timeout=0.02)
+            if any(b"bng-af-packet-test" in f for f in got):
+                break
+        assert any(b"bng-af-packet-test" in f for f in got)
+
+    def test_pump_over_af_packet(self):
+        """Source frames from a raw socket into the golden pipeline."""
+        from bng_amd.dataplane.launcher import GoldenLauncher
+        from bng_amd.dataplane.packets import (build_dhcp_request,
+                                               ip2u32, mac_bytes)
+        from bng_amd.dataplane.pktio import AFPacketIO, ListSink, Pump
+        tx = AFPacketIO("lo")
+        rx = AFPacketIO("lo")
+        l = GoldenLauncher()
+        l.set_server_config(mac_bytes("02:00:00:00:00:01"),
+                            ip2u32("10.0.0.1"))
+        l.add_pool(1, ip2u32("10.0.1.0"), 24, ip2u32("10.0.1.1"))
+        l.add_subscriber(mac_bytes("aa:bb:cc:00:00:09"), 1,
+                         ip2u32("10.0.1.9"), 10**12)
+        sink = ListSink()
+        rx.recv_batch(64, timeout=0.01)
+        tx.send_batch([build_dhcp_request(
+            mac_bytes("aa:bb:cc:00:00:09"), 1, xid=42)])
+        import time as _t
+        pump = Pump(l, rx, sink, max_wait=0.02)
+        for _ in range(50):
+            pump.pump_once()
+            if sink.frames:
+                break
+            _t.sleep(0.01)
+        # the DHCP OFFER built by the fast path reached the sink
+        assert any(len(f) > 240 for f in sink.frames)
