@@ -728,3 +728,45 @@ class TestMetamorphicFuzz:
             env=env, capture_output=True, text=True, timeout=420)
         assert r.returncode == 0, r.stdout + r.stderr
         assert '"sorted_vs_unsorted": 0' in r.stdout
+
+
+class TestPumpOnGPU:
+    """The serving loop end-to-end on the GPU: synthetic source ->
+    HipLauncher fused uplink -> verdict routing to sink/slow path (the
+    runtime path `bng run` drives; previously only golden-tested)."""
+
+    def test_pump_synthetic_to_sink(self):
+        from bng_amd.dataplane.pktio import ListSink, Pump, SyntheticSource
+        gpu, _ = make_pair()
+        gpu.add_subscriber(mac_bytes("aa:bb:cc:00:00:01"), 1,
+                           ip2u32("10.0.1.50"), NOW_SEC + 600)
+        gpu.add_subscriber_nat(ip2u32("10.0.1.50"), ip2u32(PUB),
+                               1024, 2047, subscriber_id=1)
+        frames = [build_dhcp_request("aa:bb:cc:00:00:01", DHCP_REQUEST,
+                                     xid=9)]
+        frames += [build_ipv4("aa:bb:cc:00:00:01", "02:00:00:00:00:01",
+                              ip2u32("10.0.1.50"), ip2u32(DST), proto=17,
+                              sport=41000 + k, dport=53)
+                   for k in range(15)]
+        queue = [list(frames)]
+
+        def gen(maxn):
+            out = queue[0][:maxn]
+            del queue[0][:maxn]
+            return out
+        src = SyntheticSource(gen)
+        sink = ListSink()
+        passed = []
+        pump = Pump(gpu, src, sink, slow_path=lambda f: passed.append(f))
+        total = 0
+        for _ in range(8):
+            total += pump.pump_once()
+            if total >= len(frames):
+                break
+        assert total == len(frames)
+        # 1 OFFER/ACK + 15 SNAT-rewritten data frames reached the sink
+        assert len(sink.frames) == 16
+        snat = [f for f in sink.frames
+                if len(f) == 64 and f[26:30] == ip2u32(PUB).to_bytes(4, "big")]
+        assert len(snat) == 15
+        assert pump.stats["tx"] >= 1
