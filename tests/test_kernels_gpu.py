@@ -738,8 +738,9 @@ class TestPumpOnGPU:
     def test_pump_synthetic_to_sink(self):
         from bng_amd.dataplane.pktio import ListSink, Pump, SyntheticSource
         gpu, _ = make_pair()
+        # Pump stamps batches with the REAL clock: lease far in future
         gpu.add_subscriber(mac_bytes("aa:bb:cc:00:00:01"), 1,
-                           ip2u32("10.0.1.50"), NOW_SEC + 600)
+                           ip2u32("10.0.1.50"), 1 << 40)
         gpu.add_subscriber_nat(ip2u32("10.0.1.50"), ip2u32(PUB),
                                1024, 2047, subscriber_id=1)
         frames = [build_dhcp_request("aa:bb:cc:00:00:01", DHCP_REQUEST,
