@@ -768,6 +768,6 @@ class TestPumpOnGPU:
         # 1 OFFER/ACK + 15 SNAT-rewritten data frames reached the sink
         assert len(sink.frames) == 16
         snat = [f for f in sink.frames
-                if len(f) == 64 and f[26:30] == ip2u32(PUB).to_bytes(4, "big")]
+                if f[26:30] == ip2u32(PUB).to_bytes(4, "big")]
         assert len(snat) == 15
         assert pump.stats["tx"] >= 1
