@@ -101,13 +101,17 @@ class MTLSAuthenticator(Authenticator):
 
 
 def new_authenticator(mode: str, **kw) -> Authenticator:
-    """ref authenticator.go:16 NewAuthenticator."""
+    """Factory (ref authenticator.go:16 NewAuthenticator); TPM mode is
+    recognized but unimplemented, exactly like the reference (:33)."""
     if mode == MODE_NONE:
         return Authenticator()
     if mode == MODE_PSK:
         return PSKAuthenticator(kw["psk"], kw.get("window", 300.0))
     if mode == MODE_MTLS:
         return MTLSAuthenticator(kw.get("registry"))
+    if mode == "tpm":
+        raise NotImplementedError(
+            "TPM authentication not yet implemented")
     raise ValueError(f"unknown auth mode {mode}")
 
 

@@ -542,3 +542,13 @@ class TestAuthenticatedTransport:
             assert seen.get("X-Custom") == "1"
         finally:
             srv.shutdown()
+
+
+def test_tpm_mode_recognized_but_unimplemented():
+    """Factory parity: TPM is a named mode that raises, exactly like
+    the reference (authenticator.go:33)."""
+    from bng_amd.deviceauth.authenticator import new_authenticator
+    with pytest.raises(NotImplementedError):
+        new_authenticator("tpm")
+    with pytest.raises(ValueError):
+        new_authenticator("retina-scan")
