@@ -771,3 +771,61 @@ class TestPumpOnGPU:
                 if f[26:30] == ip2u32(PUB).to_bytes(4, "big")]
         assert len(snat) == 15
         assert pump.stats["tx"] >= 1
+
+
+class TestALGAndHairpinGPU:
+    """ALG punts and hairpin flagging on the GPU vs golden (previously
+    golden-only; ref nat44.c:616-641 ALG, :951-991 hairpin)."""
+
+    def _pair(self):
+        gpu, cpu = make_pair()
+        for l in (gpu, cpu):
+            l.set_nat_config(flags=abi.NAT_FLAG_EIM |
+                             abi.NAT_FLAG_HAIRPIN |
+                             abi.NAT_FLAG_ALG_FTP | abi.NAT_FLAG_ALG_SIP,
+                             alg_ports=[(21, 6), (5060, 17)])
+            l.add_subscriber_nat(ip2u32(PRIV), ip2u32(PUB), 1024, 2047,
+                                 subscriber_id=5)
+            l.set_hairpin_ips([ip2u32(PUB)])
+        return gpu, cpu
+
+    def test_alg_punts_to_slow_path(self):
+        gpu, cpu = self._pair()
+        frames = [
+            build_ipv4("aa:00:00:00:00:01", "02:00:00:00:00:01",
+                       ip2u32(PRIV), ip2u32(DST), proto=6, sport=40000,
+                       dport=21),                        # FTP control
+            build_ipv4("aa:00:00:00:00:01", "02:00:00:00:00:01",
+                       ip2u32(PRIV), ip2u32(DST), proto=17, sport=40001,
+                       dport=5060),                      # SIP
+            build_ipv4("aa:00:00:00:00:01", "02:00:00:00:00:01",
+                       ip2u32(PRIV), ip2u32(DST), proto=17, sport=40002,
+                       dport=53),                        # plain data
+        ]
+        d, lns = gpu.make_batch(frames, stride=128)
+        v = gpu.nat44(d, lns, egress=True, now_ns=NOW_NS).cpu().tolist()
+        res = cpu.process_nat44(frames, egress=True, now_ns=NOW_NS)
+        assert v == [r[0] for r in res] == [PASS, PASS, FWD]
+        gs, cs = gpu.nat_get_stats(), cpu.nat_get_stats()
+        assert gs["alg_triggers"] == cs["alg_triggers"] == 2
+        # ALG punt events land in the compliance ring
+        evs = gpu.drain_nat_log()
+        assert sum(1 for e in evs
+                   if e["event_type"] == abi.LOG_ALG_TRIGGER) == 2
+
+    def test_hairpin_flagged_and_translated(self):
+        gpu, cpu = self._pair()
+        frames = [build_ipv4("aa:00:00:00:00:01", "02:00:00:00:00:01",
+                             ip2u32(PRIV), ip2u32(PUB), proto=17,
+                             sport=41000, dport=9999)]
+        d, lns = gpu.make_batch(frames, stride=128)
+        v = gpu.nat44(d, lns, egress=True, now_ns=NOW_NS).cpu().tolist()
+        res = cpu.process_nat44(frames, egress=True, now_ns=NOW_NS)
+        assert v == [r[0] for r in res] == [FWD]
+        assert d.cpu().numpy()[0][:42].tobytes() == res[0][1][:42]
+        gs, cs = gpu.nat_get_stats(), cpu.nat_get_stats()
+        assert gs["packets_hairpin"] == cs["packets_hairpin"] == 1
+        evs = gpu.drain_nat_log()
+        creates = [e for e in evs
+                   if e["event_type"] == abi.LOG_SESSION_CREATE]
+        assert creates and creates[0]["flags"] == 1     # hairpin flag
