@@ -829,3 +829,39 @@ class TestALGAndHairpinGPU:
         creates = [e for e in evs
                    if e["event_type"] == abi.LOG_SESSION_CREATE]
         assert creates and creates[0]["flags"] == 1     # hairpin flag
+
+
+class TestAntispoofV6GPU:
+    """IPv6 uRPF on the GPU vs golden (ref antispoof.c v6 branch;
+    previously golden-only)."""
+
+    def _v6_frame(self, src_mac, src_ip6: bytes):
+        import struct as st
+        eth = mac_bytes("02:00:00:00:00:01") + mac_bytes(src_mac) + \
+            st.pack(">H", 0x86DD)
+        ip6 = st.pack(">IHBB", 0x60000000, 8, 17, 64) + src_ip6 + \
+            b"\x20\x01" + b"\x00" * 14                   # dst 2001::
+        udp = st.pack(">HHHH", 4000, 53, 8, 0)
+        return eth + ip6 + udp
+
+    def test_v6_strict_binding_match_and_violation(self):
+        gpu, cpu = make_pair()
+        good6 = bytes(range(16))
+        bad6 = bytes(range(1, 17))
+        for l in (gpu, cpu):
+            l.set_antispoof_config(default_mode=abi.AS_STRICT,
+                                   log_violations=True)
+            l.add_binding(mac_bytes("aa:00:00:00:00:66"), ipv6=good6,
+                          mode=abi.AS_STRICT)
+        frames = [self._v6_frame("aa:00:00:00:00:66", good6),
+                  self._v6_frame("aa:00:00:00:00:66", bad6),
+                  self._v6_frame("aa:00:00:00:00:99", good6)]  # unbound
+        d, lns = gpu.make_batch(frames, stride=128)
+        v = gpu.antispoof(d, lns, now_ns=NOW_NS).cpu().tolist()
+        vc = [cpu.dp.antispoof(bytes(f)) for f in frames]
+        assert v == vc
+        assert v[0] == FWD and v[1] == DROP
+        gs = gpu.antispoof_get_stats()
+        cs = cpu.antispoof_get_stats()
+        assert gs["ipv6_violations"] == cs["ipv6_violations"]
+        assert gs["ipv6_violations"] >= 1
