@@ -32,6 +32,8 @@ class HealthMonitor:
         self.consecutive_failures = 0
         self.consecutive_successes = 0
         self.last_check: float = 0.0
+        self.last_response_time: float = 0.0   # seconds (ref
+        # health_monitor.go ResponseTime tracking)
         self._listeners: List[Callable[[HealthEvent], None]] = []
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
@@ -55,6 +57,7 @@ class HealthMonitor:
         """One health probe (ref health_monitor.go:232 performCheck)."""
         import requests
         self.last_check = time.time()
+        t0 = time.perf_counter()
         ok = False
         try:
             r = requests.get(f"{self.partner_url}/health",
@@ -62,6 +65,7 @@ class HealthMonitor:
             ok = r.status_code == 200
         except Exception:
             ok = False
+        self.last_response_time = time.perf_counter() - t0
         if ok:
             self.consecutive_failures = 0
             self.consecutive_successes += 1

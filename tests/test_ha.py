@@ -258,3 +258,30 @@ class TestFailoverControllerCoverage:
         # failback at original (standby) role refused
         fc2 = FailoverController("n2", ROLE_STANDBY)
         assert not fc2.initiate_failback()
+
+
+def test_health_monitor_tracks_response_time():
+    """ref TestHealthMonitor_ResponseTime: each probe records how long
+    the partner took to answer."""
+    import http.server
+    import threading as th
+    from bng_amd.ha.health_monitor import HealthMonitor
+
+    class H(http.server.BaseHTTPRequestHandler):
+        def do_GET(self):
+            time.sleep(0.02)
+            self.send_response(200)
+            self.end_headers()
+
+        def log_message(self, *a):
+            pass
+    srv = http.server.HTTPServer(("127.0.0.1", 0), H)
+    t = th.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        m = HealthMonitor(f"http://127.0.0.1:{srv.server_port}")
+        assert m.check_once()
+        assert m.last_response_time >= 0.02
+        assert m.last_check > 0
+    finally:
+        srv.shutdown()
