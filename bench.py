@@ -278,7 +278,10 @@ def main():
     log(rank, f"[bench] batch generated in {time.perf_counter() - t0:.1f}s")
 
     overlap = not args.no_overlap
-    nbuf = 2 if overlap else 1
+    # 4 in-flight batches hide prep-sync slack behind the kernel
+    # (same-box A/B: nbuf 2/3/4 -> 1537/1555/1567 Mpps); BNG_NBUF
+    # overrides for experiments
+    nbuf = (int(os.environ.get("BNG_NBUF", "4")) if overlap else 1)
     works = [torch.empty_like(pristine) for _ in range(nbuf)]
     clss = [torch.empty(args.batch, dtype=torch.uint8, device=device)
             for _ in range(nbuf)]
