@@ -53,6 +53,13 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--radius-auth-mode", default="none",
                    choices=["none", "mac"])
     g.add_argument("--radius-coa-port", type=int, default=0)
+    g.add_argument("--radius-partition-mode", default="cached",
+                   choices=["off", "reject", "cached", "allow"])
+    g = run.add_argument_group("resilience")
+    g.add_argument("--health-check-interval", type=float, default=5.0)
+    g.add_argument("--short-lease-enable", action="store_true")
+    g.add_argument("--short-lease-threshold", type=float, default=0.9)
+    g.add_argument("--short-lease-duration", type=int, default=60)
     g = run.add_argument_group("nexus")
     g.add_argument("--nexus-url", default="")
     g.add_argument("--nexus-auth", default="none",
@@ -304,7 +311,17 @@ class BNG:
             from ..radius.client import Client as RadiusClient
             secret = resolve_secret(a.radius_secret, a.radius_secret_file)
             self.radius = RadiusClient(a.radius_server, secret.encode())
-            self.dhcp_server.set_radius(self.radius, a.radius_auth_mode)
+            if a.radius_partition_mode != "off":
+                # partition degradation modes (ref resilience wiring
+                # main.go:1182-1211, radius_handler.go:52)
+                from ..resilience.radius_handler import ResilientRadius
+                self.radius_resilient = ResilientRadius(
+                    self.radius, mode=a.radius_partition_mode)
+                self.dhcp_server.set_radius(self.radius_resilient,
+                                            a.radius_auth_mode)
+            else:
+                self.dhcp_server.set_radius(self.radius,
+                                            a.radius_auth_mode)
             self.accounting = AccountingManager(self.radius).start()
             self._defer(self.accounting.stop)
             self.dhcp_server.set_accounting(self.accounting)
@@ -319,6 +336,41 @@ class BNG:
                                      port=a.radius_coa_port,
                                      handler=proc).start()
                 self._defer(self.coa.stop)
+
+        # 9b. resilience: partition FSM + short-lease under pool
+        # pressure (ref main.go:1182-1211)
+        from ..resilience.manager import Manager as ResilienceMgr
+
+        def _health() -> bool:
+            alloc = getattr(self, "nexus_allocator", None)
+            if alloc is None:
+                return True
+            try:
+                return bool(alloc.health_check())
+            except Exception:
+                return False
+        self.resilience = ResilienceMgr(
+            _health, check_interval=a.health_check_interval).start()
+        self._defer(self.resilience.stop)
+        if getattr(self, "radius_resilient", None) is not None:
+            self.resilience.on_transition(
+                lambda old, new: new == "online" and
+                self.radius_resilient.replay_buffered())
+        if a.short_lease_enable:
+            from ..resilience.pool_monitor import PoolMonitor
+
+            def _util() -> float:
+                stats = self.dhcp_server.pools.all_stats()
+                alloc = sum(x["allocated"] for x in stats)
+                total = sum(x["allocated"] + x["available"]
+                            for x in stats)
+                return alloc / total if total else 0.0
+            self.pool_monitor = PoolMonitor(
+                _util, critical=a.short_lease_threshold,
+                normal_lease=a.lease_time,
+                short_lease=a.short_lease_duration)
+            self.dhcp_server.lease_time_provider = \
+                self.pool_monitor.effective_lease_time
 
         # 10. NAT (main.go:1001-1040)
         if a.nat_enable:

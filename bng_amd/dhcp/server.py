@@ -49,6 +49,9 @@ class DHCPServer:
         self.server_ip = ip2u32(server_ip)
         self.server_mac = server_mac
         self.lease_time = lease_time
+        # optional override (resilience short-lease mode under pool
+        # pressure, ref resilience/pool_monitor + types.go:69-100)
+        self.lease_time_provider = None
         self.authoritative = authoritative
         self.leases: Dict[bytes, Lease] = {}          # by MAC
         self.leases_by_circuit: Dict[bytes, Lease] = {}
@@ -80,6 +83,11 @@ class DHCPServer:
     # ----------------------------------------------------------- wiring
     def set_launcher(self, l):
         self.launcher = l
+
+    def _lease_seconds(self) -> float:
+        if self.lease_time_provider is not None:
+            return min(self.lease_time, self.lease_time_provider())
+        return self.lease_time
 
     def set_radius(self, c, auth_mode: str = "mac"):
         self.radius = c
@@ -198,7 +206,7 @@ class DHCPServer:
             # client asks for a different IP than its lease: NAK
             self.stats["nak"] += 1
             return self._nak(req)
-        lease.expiry = time.time() + self.lease_time
+        lease.expiry = time.time() + self._lease_seconds()
         self._post_ack(lease, req)
         self.stats["ack"] += 1
         self._emit_lease("add", lease)
@@ -319,7 +327,7 @@ class DHCPServer:
             self.stats["walled_garden"] += 1
 
         lease = Lease(mac=mac, ip=ip, pool_id=pool.cfg.pool_id,
-                      expiry=time.time() + self.lease_time,
+                      expiry=time.time() + self._lease_seconds(),
                       circuit_id=req.circuit_id(), subscriber_id=sub_id,
                       policy_name=policy_name, walled_garden=walled)
         with self._lock:
@@ -444,6 +452,8 @@ class DHCPServer:
         resp.yiaddr = lease.ip
         pool = self.pools.get_pool(lease.pool_id)
         lt = pool.cfg.lease_time if pool else self.lease_time
+        if self.lease_time_provider is not None:
+            lt = min(lt, self.lease_time_provider())
         resp.set_option(dm.OPT_LEASE_TIME, struct.pack(">I", lt))
         resp.set_option(dm.OPT_RENEWAL_TIME, struct.pack(">I", lt // 2))
         resp.set_option(dm.OPT_REBIND_TIME, struct.pack(">I", lt * 7 // 8))

@@ -166,3 +166,51 @@ class TestStatsCommand:
         from bng_amd.cli.main import main
         assert main(["stats", "--metrics-url",
                      "http://127.0.0.1:1"]) == 1
+
+
+class TestResilienceWiring:
+    def test_short_lease_mode_under_pool_pressure(self):
+        """--short-lease-enable: lease time collapses when the pool runs
+        hot (ref resilience pool monitor wiring, types.go:69-100)."""
+        app, _ = make_app(["--short-lease-enable",
+                           "--short-lease-threshold", "0.5",
+                           "--short-lease-duration", "45",
+                           "--lease-time", "3600"])
+        try:
+            assert app.resilience is not None
+            # drain most of the /24 pool to cross the threshold
+            for i in range(160):
+                mac = mac_bytes(f"aa:bb:cc:01:{i >> 8:02x}:{i & 0xFF:02x}")
+                offer = app.dhcp_server.handle(
+                    dm.build_request(mac, dm.DISCOVER))
+                app.dhcp_server.handle(dm.build_request(
+                    mac, dm.REQUEST, requested_ip=offer.yiaddr))
+            app.pool_monitor.check()
+            assert app.pool_monitor.level in ("critical", "exhausted")
+            mac = mac_bytes("aa:bb:cc:02:00:01")
+            offer = app.dhcp_server.handle(dm.build_request(mac, dm.DISCOVER))
+            ack = app.dhcp_server.handle(dm.build_request(
+                mac, dm.REQUEST, requested_ip=offer.yiaddr))
+            import struct as st
+            lt = st.unpack(">I", ack.get_option(51))[0]
+            assert lt == 45                       # short lease granted
+        finally:
+            app.stop()
+
+    def test_radius_partition_cached_mode_wired(self):
+        """--radius-partition-mode cached wraps the client so DHCP auth
+        degrades instead of failing when RADIUS is unreachable."""
+        app, _ = make_app(["--radius-server", "127.0.0.1:1",
+                           "--radius-secret", "s",
+                           "--radius-auth-mode", "mac",
+                           "--radius-partition-mode", "allow"])
+        try:
+            from bng_amd.resilience.radius_handler import ResilientRadius
+            assert isinstance(app.dhcp_server.radius, ResilientRadius)
+            # RADIUS at port 1 is unreachable: allow mode still leases
+            mac = mac_bytes("aa:bb:cc:03:00:01")
+            offer = app.dhcp_server.handle(dm.build_request(mac, dm.DISCOVER))
+            assert offer is not None
+            assert app.dhcp_server.radius.stats["allow_answers"] >= 1
+        finally:
+            app.stop()
