@@ -95,7 +95,7 @@ def run(l, frames, sort):
 
 
 def main():
-    rng = random.Random(20260913)
+    rng = random.Random(int(os.environ.get("FUZZ_SEED", "20260913")))
     rounds = int(os.environ.get("FUZZ_ROUNDS", 12))
     bsz = 4096
     bad = {"sorted_vs_unsorted": 0, "split": 0, "dhcp_golden": 0}
