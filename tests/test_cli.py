@@ -214,3 +214,10 @@ class TestResilienceWiring:
             assert app.dhcp_server.radius.stats["allow_answers"] >= 1
         finally:
             app.stop()
+
+
+def test_verify_command():
+    """`bng verify` — the cmd/verify-bpf analog — passes on a healthy
+    tree and exits 0."""
+    from bng_amd.cli.main import main
+    assert main(["verify"]) == 0
