@@ -865,3 +865,65 @@ class TestAntispoofV6GPU:
         cs = cpu.antispoof_get_stats()
         assert gs["ipv6_violations"] == cs["ipv6_violations"]
         assert gs["ipv6_violations"] >= 1
+
+
+class TestCrossLayerGPU:
+    """Control-plane managers driving the GPU tables end to end."""
+
+    def test_nat_ring_to_compliance_log_pipeline(self):
+        """GPU log ring -> drain -> ComplianceLogger formats (the legal
+        pipeline, ref nat44.c ring + nat/logging.go)."""
+        import json as js
+        from bng_amd.nat.logging import ComplianceLogger
+        gpu, _ = make_pair()
+        for k in range(8):
+            gpu.add_subscriber_nat(ip2u32(f"10.0.1.{k + 1}"), ip2u32(PUB),
+                                   1024 + k * 256, 1024 + k * 256 + 255,
+                                   subscriber_id=k + 1)
+        frames = [build_ipv4(f"aa:00:00:00:00:{k:02x}",
+                             "02:00:00:00:00:01",
+                             ip2u32(f"10.0.1.{k + 1}"), ip2u32(DST),
+                             proto=17, sport=42000 + k, dport=53)
+                  for k in range(8)]
+        d, lns = gpu.make_batch(frames, stride=128)
+        gpu.nat44(d, lns, egress=True, now_ns=NOW_NS)
+        lg = ComplianceLogger(fmt="json")
+        n = sum(1 for e in gpu.drain_nat_log() if lg.log_event(e))
+        assert n == 8
+        recs = [js.loads(r) for r in lg.records]
+        assert {r["event"] for r in recs} == {"session_create"}
+        assert all(r["public_ip"] == PUB for r in recs)
+        assert lg.counters["session_create"] == 8
+
+    def test_coa_filter_id_updates_gpu_qos(self):
+        """RADIUS CoA with Filter-Id re-shapes the subscriber ON the GPU
+        (ref coa_handler.go:46-70 -> qos manager -> map write)."""
+        from bng_amd.radius.coa import CoAProcessor, CoARequest
+        from bng_amd.radius import packet as rp
+        gpu, _ = make_pair()
+        ip = ip2u32(PRIV)
+
+        class Sess:
+            pass
+        sess = Sess()
+        sess.ip = ip
+
+        def updater(session, policy):
+            rate, burst = {"throttle": (8, 10)}[policy]
+            gpu.set_qos_policy(session.ip, rate, burst,
+                               direction="ingress", now_ns=NOW_NS)
+            return True
+        proc = CoAProcessor(session_lookup=lambda r: sess,
+                            terminate=lambda s: True,
+                            qos_updater=updater)
+        req = CoARequest(code=rp.COA_REQUEST, session_id="s1",
+                         username="alice", framed_ip=PRIV,
+                         policy_name="throttle")
+        ok, code = proc(req)
+        assert ok and code == 0
+        pkt = [build_ipv4("aa:00:00:00:00:01", "02:00:00:00:00:01",
+                          ip, ip2u32(DST), proto=17, sport=5000,
+                          dport=53)]
+        d, lns = gpu.make_batch(pkt, stride=128)
+        assert gpu.qos(d, lns, egress=False,
+                       now_ns=NOW_NS).cpu().tolist() == [DROP]
