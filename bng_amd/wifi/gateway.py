@@ -19,6 +19,8 @@ class GuestSession:
     accepted_terms: bool = False
     started: float = field(default_factory=time.time)
     last_seen: float = field(default_factory=time.time)
+    bytes_in: int = 0
+    bytes_out: int = 0
 
 
 class Gateway:
@@ -104,6 +106,29 @@ class Gateway:
                 self.sessions.pop(mac, None)
         self.stats["expired"] += len(gone)
         return len(gone)
+
+    def update_traffic(self, mac: str, bytes_in: int = 0,
+                       bytes_out: int = 0) -> bool:
+        """Accumulate guest usage + keep the session warm (ref
+        wifi/gateway_test.go UpdateTrafficStats)."""
+        with self._lock:
+            s = self.sessions.get(mac)
+            if s is None:
+                return False
+            s.bytes_in += bytes_in
+            s.bytes_out += bytes_out
+            s.last_seen = time.time()
+            return True
+
+    def stats(self):
+        with self._lock:
+            return {"sessions": len(self.sessions),
+                    "accepted": sum(1 for s in self.sessions.values()
+                                    if s.accepted_terms),
+                    "bytes_in": sum(s.bytes_in
+                                    for s in self.sessions.values()),
+                    "bytes_out": sum(s.bytes_out
+                                     for s in self.sessions.values())}
 
     def session_count(self) -> int:
         with self._lock:

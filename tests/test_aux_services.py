@@ -552,3 +552,18 @@ def test_tpm_mode_recognized_but_unimplemented():
         new_authenticator("tpm")
     with pytest.raises(ValueError):
         new_authenticator("retina-scan")
+
+
+def test_wifi_traffic_stats():
+    """ref wifi/gateway_test.go UpdateTrafficStats + Stats."""
+    from bng_amd.wifi.gateway import Gateway
+    g = Gateway(network="192.168.200.0/28")
+    g.join("aa:01")
+    g.join("aa:02")
+    g.accept_terms("aa:01")
+    assert g.update_traffic("aa:01", 1000, 200)
+    assert g.update_traffic("aa:01", 500, 100)
+    assert not g.update_traffic("none", 1, 1)
+    st = g.stats()
+    assert st == {"sessions": 2, "accepted": 1,
+                  "bytes_in": 1500, "bytes_out": 300}
