@@ -120,9 +120,10 @@ class Gateway:
             s.last_seen = time.time()
             return True
 
-    def stats(self):
+    def get_stats(self):
         with self._lock:
-            return {"sessions": len(self.sessions),
+            return {**self.stats,
+                    "sessions": len(self.sessions),
                     "accepted": sum(1 for s in self.sessions.values()
                                     if s.accepted_terms),
                     "bytes_in": sum(s.bytes_in
