@@ -125,3 +125,72 @@ def test_bench_rss_arrival_model():
                    (ips2[:, 2] << 8) | ips2[:, 3]).astype(np.uint64)
         own2 = bench.mix64_np(ip2_u32) % np.uint64(world)
         assert len(set(own2.tolist())) == world
+
+
+def _pipeline_order_worker(rank, world, rendezvous_file, q):
+    """Replicates bench.py's overlapped-step CALL SEQUENCE: prep(0),
+    then for each step k the processing of batch k followed by prep(k+1)
+    — i.e., exchange() invoked one batch AHEAD of consumption, the
+    collective-order pattern the multi-GPU SCALE run relies on.  gloo
+    deadlocks (and the test times out) if any rank's order diverges."""
+    import torch.distributed as dist
+    dist.init_process_group(
+        "gloo", init_method=f"file://{rendezvous_file}",
+        rank=rank, world_size=world)
+    from bng_amd.parallel.sharding import exchange
+    import bench
+
+    steps, nbuf = 6, 4
+    n, stride = 256, 512
+    total_received = 0
+    batches = [None] * nbuf
+
+    def prep(k):
+        d_np, l_np = bench.gen_batch(n, 10_000, 0.25, stride,
+                                     seed=1000 + rank * 100 + k,
+                                     rank=rank, world=world)
+        # RSS model: DHCP frames may be foreign, data frames are local
+        from bng_amd.parallel.hashring import owner_of_ip, owner_of_mac
+        owners = []
+        for i in range(n):
+            f = d_np[i]
+            if l_np[i] > 64:      # DHCP: owner by chaddr MAC
+                owners.append(owner_of_mac(
+                    bytes(f[70:76]), world))
+            else:                 # data: owner by src ip
+                owners.append(owner_of_ip(
+                    int.from_bytes(bytes(f[26:30]), "big"), world))
+        d = torch.from_numpy(d_np.reshape(n, stride))
+        l = torch.from_numpy(l_np.view(np.int16))
+        o = torch.tensor(owners, dtype=torch.int64)
+        batches[k % nbuf] = exchange(d, l, o)
+
+    prep(0)
+    for k in range(steps):
+        d, l = batches[k % nbuf]
+        total_received += l.numel()
+        prep(k + 1)
+    dist.barrier()
+    q.put((rank, total_received))
+    dist.destroy_process_group()
+
+
+def test_bench_step_order_symmetry_gloo(tmp_path):
+    """The SCALE-run collective pattern completes without deadlock and
+    conserves packets across ranks."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    rv = str(tmp_path / "rdv2")
+    procs = [ctx.Process(target=_pipeline_order_worker,
+                         args=(r, world, rv, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    got = {}
+    for _ in range(world):
+        rank, n = q.get(timeout=180)
+        got[rank] = n
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0
+    assert sum(got.values()) == world * 6 * 256   # conservation
