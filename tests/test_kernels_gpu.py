@@ -927,3 +927,47 @@ class TestCrossLayerGPU:
         d, lns = gpu.make_batch(pkt, stride=128)
         assert gpu.qos(d, lns, egress=False,
                        now_ns=NOW_NS).cpu().tolist() == [DROP]
+
+
+class TestTCPStateGPU:
+    """TCP session state machine + sweep interplay on the GPU (ref
+    nat44.c:885-895; previously golden-only)."""
+
+    def test_syn_ack_fin_then_transient_sweep(self):
+        import struct as st
+        from bng_amd.dataplane.launcher import (TCP_EST_TIMEOUT_NS,
+                                                TCP_TRANSIENT_TIMEOUT_NS)
+        gpu, _ = make_pair()
+        gpu.add_subscriber_nat(ip2u32(PRIV), ip2u32(PUB), 1024, 2047,
+                               subscriber_id=1)
+        syn = [build_ipv4("aa:00:00:00:00:01", "02:00:00:00:00:01",
+                          ip2u32(PRIV), ip2u32(DST), proto=6, sport=5555,
+                          dport=80, tcp_flags=0x02)]
+        d, lns = gpu.make_batch(syn, stride=128)
+        assert gpu.nat44(d, lns, egress=True,
+                         now_ns=NOW_NS).cpu().tolist() == [FWD]
+        nat_port = st.unpack_from(">H", d.cpu().numpy()[0], 34)[0]
+
+        def back(flags):
+            return [build_ipv4("02:00:00:00:00:02", "02:00:00:00:00:01",
+                               ip2u32(DST), ip2u32(PUB), proto=6,
+                               sport=80, dport=nat_port,
+                               tcp_flags=flags)]
+        # SYN+ACK -> ESTABLISHED: transient sweep must NOT reclaim
+        d2, l2 = gpu.make_batch(back(0x12), stride=128)
+        assert gpu.nat44(d2, l2, egress=False,
+                         now_ns=NOW_NS).cpu().tolist() == [FWD]
+        gpu.sweep_nat(now_ns=NOW_NS + TCP_TRANSIENT_TIMEOUT_NS + 10**9)
+        assert gpu.nat_get_stats()["sessions_expired"] == 0
+        # FIN -> CLOSING: now the transient timeout applies
+        d3, l3 = gpu.make_batch(back(0x11), stride=128)
+        assert gpu.nat44(d3, l3, egress=False,
+                         now_ns=NOW_NS + 10**9).cpu().tolist() == [FWD]
+        gpu.sweep_nat(now_ns=NOW_NS + TCP_TRANSIENT_TIMEOUT_NS + 2 * 10**9)
+        st_ = gpu.nat_get_stats()
+        assert st_["sessions_expired"] == 1
+        # reverse entry tombstoned: return traffic no longer translates
+        d4, l4 = gpu.make_batch(back(0x10), stride=128)
+        gpu.nat44(d4, l4, egress=False, now_ns=NOW_NS + 3 * 10**9)
+        host = d4.cpu().numpy()[0]
+        assert st.unpack_from(">I", host, 30)[0] == ip2u32(PUB)  # untouched
