@@ -971,3 +971,27 @@ class TestTCPStateGPU:
         gpu.nat44(d4, l4, egress=False, now_ns=NOW_NS + 3 * 10**9)
         host = d4.cpu().numpy()[0]
         assert st.unpack_from(">I", host, 30)[0] == ip2u32(PUB)  # untouched
+
+
+class TestTaggedDHCPUplinkGPU:
+    """QinQ-tagged DHCP through the FUSED uplink: routed to the DHCP
+    lane (the kernel parses tags for the is_dhcp decision) and answered
+    from the VLAN-keyed subscriber entry."""
+
+    def test_tagged_discover_gets_offer(self):
+        gpu, cpu = make_pair()
+        for l in (gpu, cpu):
+            l.add_vlan_subscriber(100, 200, 1, ip2u32("10.0.1.77"),
+                                  NOW_SEC + 600)
+        frames = [build_dhcp_request("aa:bb:cc:00:00:31", DHCP_DISCOVER,
+                                     xid=31, s_tag=100, c_tag=200)]
+        d, lns = gpu.make_batch(frames)
+        v, out_len = gpu.uplink(d, lns, now_ns=NOW_NS, now_sec=NOW_SEC)
+        import numpy as np
+        ol = out_len.cpu().numpy().view(np.uint16)[0]
+        assert v.cpu().tolist() == [TX]
+        vc, out = cpu.process_dhcp(frames, now_sec=NOW_SEC)[0]
+        assert vc == TX
+        assert bytes(d.cpu().numpy()[0][:ol]) == out
+        # the OFFER keeps the QinQ tags for the return trip
+        assert out[12:14] == b"\x88\xa8" or out[12:14] == b"\x81\x00"

@@ -111,3 +111,41 @@ class TestHashProperties:
         o = mix64(key) % shards
         assert 0 <= o < shards
         assert o == mix64(key) % shards
+
+
+class TestEpochAllocatorProperties:
+    @given(st.data())
+    @settings(max_examples=40, deadline=None)
+    def test_epoch_model(self, data):
+        """Stateful model: active leases survive exactly grace_period
+        epochs without renewal; renewal extends; no address is ever
+        shared by two live subscribers."""
+        from bng_amd.allocator.epoch_bitmap import EpochBitmapAllocator
+        a = EpochBitmapAllocator("10.9.0.0/27", 32, grace_period=1)
+        live = {}          # sid -> (ip, epochs_since_renewal)
+        for _ in range(data.draw(st.integers(1, 30))):
+            op = data.draw(st.sampled_from(
+                ["alloc", "renew", "advance", "release"]))
+            if op == "alloc":
+                sid = f"s{data.draw(st.integers(0, 12))}"
+                try:
+                    ip = a.allocate(sid)
+                except Exception:
+                    continue
+                if sid not in live:
+                    assert ip not in [v[0] for v in live.values()]
+                live[sid] = (ip, 0)
+            elif op == "renew" and live:
+                sid = data.draw(st.sampled_from(sorted(live)))
+                a.renew(sid)
+                live[sid] = (live[sid][0], 0)
+            elif op == "advance":
+                a.advance_epoch()
+                live = {s: (ip, n + 1) for s, (ip, n) in live.items()
+                        if n + 1 <= 1}          # grace_period=1
+            elif live:
+                sid = data.draw(st.sampled_from(sorted(live)))
+                a.release(sid)
+                del live[sid]
+        for sid, (ip, _) in live.items():
+            assert a.lookup(sid) == ip
