@@ -995,3 +995,35 @@ class TestTaggedDHCPUplinkGPU:
         assert bytes(d.cpu().numpy()[0][:ol]) == out
         # the OFFER keeps the QinQ tags for the return trip
         assert out[12:14] == b"\x88\xa8" or out[12:14] == b"\x81\x00"
+
+
+class TestICMPNatGPU:
+    """ICMP echo-id SNAT/DNAT round trip on the GPU vs golden (ref
+    nat44.c ICMP branch; previously golden-only)."""
+
+    def test_icmp_id_round_trip(self):
+        import struct as st
+        gpu, cpu = nat_pair()
+        out = [build_ipv4("aa:00:00:00:00:01", "02:00:00:00:00:01",
+                          ip2u32(PRIV), ip2u32(DST), proto=1,
+                          icmp_id=777)]
+        d, lns = gpu.make_batch(out, stride=128)
+        assert gpu.nat44(d, lns, egress=True,
+                         now_ns=NOW_NS).cpu().tolist() == [FWD]
+        res = cpu.process_nat44(out, egress=True, now_ns=NOW_NS)
+        host = d.cpu().numpy()[0]
+        assert bytes(host[:42]) == res[0][1][:42]     # byte-identical
+        nat_id = st.unpack_from(">H", host, 38)[0]
+        assert nat_id != 777                          # id translated
+        # echo reply comes back to the translated id
+        reply = [build_ipv4("02:00:00:00:00:02", "02:00:00:00:00:01",
+                            ip2u32(DST), ip2u32(PUB), proto=1,
+                            icmp_id=nat_id)]
+        d2, l2 = gpu.make_batch(reply, stride=128)
+        assert gpu.nat44(d2, l2, egress=False,
+                         now_ns=NOW_NS).cpu().tolist() == [FWD]
+        res2 = cpu.process_nat44(reply, egress=False, now_ns=NOW_NS)
+        h2 = d2.cpu().numpy()[0]
+        assert bytes(h2[:42]) == res2[0][1][:42]
+        assert st.unpack_from(">H", h2, 38)[0] == 777  # restored
+        assert st.unpack_from(">I", h2, 30)[0] == ip2u32(PRIV)
