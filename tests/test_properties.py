@@ -149,3 +149,36 @@ class TestEpochAllocatorProperties:
                 del live[sid]
         for sid, (ip, _) in live.items():
             assert a.lookup(sid) == ip
+
+
+class TestCRDTProperties:
+    @given(st.lists(st.tuples(st.integers(0, 2),      # replica
+                              st.integers(0, 1),      # op: put/delete
+                              st.integers(0, 5),      # key id
+                              st.binary(min_size=1, max_size=8)),
+                    max_size=40))
+    @settings(max_examples=60, deadline=None)
+    def test_three_replicas_converge_identically(self, ops):
+        """LWW CRDT: any op interleaving under partition converges to
+        ONE state after full pairwise anti-entropy, regardless of sync
+        order (commutativity + idempotence of merge)."""
+        from bng_amd.nexus.clset import CLSetStore
+        reps = [CLSetStore(f"n{i}") for i in range(3)]
+        for r, op, k, v in ops:
+            if op == 0:
+                reps[r].put(f"k{k}", v)
+            else:
+                reps[r].delete(f"k{k}")
+        # full mesh anti-entropy, two rounds, arbitrary order
+        for _ in range(2):
+            for i in range(3):
+                for j in range(3):
+                    if i != j:
+                        reps[i].merge(reps[j].snapshot())
+        states = [sorted((k, bytes(v)) for k, v in r.list("").items())
+                  for r in reps]
+        assert states[0] == states[1] == states[2]
+        # merge is idempotent: re-merging changes nothing
+        before = reps[0].snapshot()
+        reps[0].merge(reps[1].snapshot())
+        assert reps[0].snapshot() == before
