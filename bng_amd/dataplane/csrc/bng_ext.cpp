@@ -45,6 +45,7 @@ void bng_launch_binding_upsert(void*, uint32_t, const void*, int, void*,
 void bng_launch_binding_delete(void*, uint32_t, const void*, int,
                                hipStream_t);
 void bng_launch_nat_sweep(void*, uint32_t, void*, uint32_t, void*, uint32_t,
+                          void*, uint32_t, uint64_t,
                           uint64_t, uint64_t, uint64_t, uint64_t, uint64_t,
                           void*, hipStream_t);
 void bng_launch_shard_owner(const void*, const void*, void*, int, int, int,
@@ -252,16 +253,20 @@ void binding_delete(torch::Tensor table, torch::Tensor keys) {
 }
 
 void nat_sweep(torch::Tensor sessions, torch::Tensor reverse,
-               torch::Tensor subnat, int64_t now_ns, int64_t udp_to,
+               torch::Tensor subnat, torch::Tensor eim, int64_t eim_to,
+               int64_t now_ns, int64_t udp_to,
                int64_t tcp_est_to, int64_t tcp_tr_to, int64_t icmp_to,
                torch::Tensor stats) {
   uint32_t n_slots = (uint32_t)(sessions.numel() * sessions.element_size() /
                                 sizeof(bng_nat_session));
+  uint32_t eim_slots = (uint32_t)(eim.numel() * eim.element_size() /
+                                  sizeof(bng_eim_entry));
   bng_launch_nat_sweep(
       sessions.data_ptr(), n_slots, reverse.data_ptr(),
       table_mask(reverse, sizeof(bng_nat_reverse), "reverse"),
       subnat.data_ptr(),
       table_mask(subnat, sizeof(bng_subctx), "subctx"),
+      eim.data_ptr(), eim_slots, (uint64_t)eim_to,
       (uint64_t)now_ns, (uint64_t)udp_to, (uint64_t)tcp_est_to,
       (uint64_t)tcp_tr_to, (uint64_t)icmp_to, stats.data_ptr(),
       cur_stream());

@@ -151,25 +151,57 @@ BNG_DEV const bng_binding_entry* binding_lookup(const bng_binding_entry* t,
 
 /* find-or-claim for device-created entries (sessions / EIM / reverse).
  * Returns entry and sets *claimed when this thread won the slot; on a
- * sig match the caller must bng_wait_ready() before trusting fields. */
+ * sig match the caller must bng_wait_ready() before trusting fields.
+ * Tombstoned slots (sweep-expired sessions) are RECLAIMED: the probe
+ * remembers the first tombstone and claims it once the full chain
+ * rules out an existing entry — without this, long-uptime churn fills
+ * the chain with tombstones (the reference gets reuse for free from
+ * BPF LRU maps). */
 template <typename E>
 BNG_DEV E* sig_find_or_claim(E* t, uint32_t mask, uint64_t sig,
                              bool* claimed, bool* found) {
   *claimed = false; *found = false;
-  uint32_t slot = (uint32_t)sig & mask;   /* sig is already mixed */
-  for (int i = 0; i < BNG_MAX_PROBE; ++i) {
-    E* e = &t[(slot + i) & mask];
-    uint64_t k = __hip_atomic_load(&e->sig, __ATOMIC_RELAXED,
-                                   __HIP_MEMORY_SCOPE_AGENT);
-    if (k == sig) { *found = true; return e; }
-    if (k == BNG_KEY_EMPTY) {
-      uint64_t old = __hip_atomic_compare_exchange_strong(
-          &e->sig, &k, sig, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
-          __HIP_MEMORY_SCOPE_AGENT) ? BNG_KEY_EMPTY : k;
-      if (old == BNG_KEY_EMPTY) { *claimed = true; return e; }
-      if (old == sig) { *found = true; return e; }
-      /* lost to a different key: fall through, keep probing */
+  uint32_t slot = (uint32_t)sig & mask;
+  for (int attempt = 0; attempt < 4; ++attempt) {
+    int first_tomb = -1;
+    for (int i = 0; i < BNG_MAX_PROBE; ++i) {
+      E* e = &t[(slot + i) & mask];
+      uint64_t k = __hip_atomic_load(&e->sig, __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+      if (k == sig) { *found = true; return e; }
+      if (k == BNG_KEY_TOMBSTONE) {
+        if (first_tomb < 0) first_tomb = (int)((slot + i) & mask);
+        continue;
+      }
+      if (k == BNG_KEY_EMPTY) {
+        E* d = first_tomb >= 0 ? &t[first_tomb] : e;
+        uint64_t expect = first_tomb >= 0 ? BNG_KEY_TOMBSTONE
+                                          : BNG_KEY_EMPTY;
+        uint64_t want = expect;
+        if (__hip_atomic_compare_exchange_strong(
+                &d->sig, &want, sig, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+                __HIP_MEMORY_SCOPE_AGENT)) {
+          *claimed = true;
+          return d;
+        }
+        if (want == sig) { *found = true; return d; }
+        break;   /* lost the slot race to a different key: rescan */
+      }
     }
+    if (first_tomb >= 0) {
+      /* chain full but reclaimable: take the tombstone directly */
+      E* d = &t[first_tomb];
+      uint64_t want = BNG_KEY_TOMBSTONE;
+      if (__hip_atomic_compare_exchange_strong(
+              &d->sig, &want, sig, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+              __HIP_MEMORY_SCOPE_AGENT)) {
+        *claimed = true;
+        return d;
+      }
+      if (want == sig) { *found = true; return d; }
+      continue;   /* raced: rescan */
+    }
+    return nullptr;   /* chain genuinely full of other keys */
   }
   return nullptr;
 }
@@ -183,60 +215,6 @@ BNG_DEV E* sig_lookup(E* t, uint32_t mask, uint64_t sig) {
                                    __HIP_MEMORY_SCOPE_AGENT);
     if (k == sig) return e;
     if (k == BNG_KEY_EMPTY) return nullptr;
-  }
-  return nullptr;
-}
-
-/* Session find-or-claim with a WIDENED first probe: sig + key +
- * translation + ready (5 independent sc1 loads of the entry's first
- * 64-B line) issue together, so a first-slot hit — the common case at
- * <=50% load — costs ONE L2 round trip instead of a dependent
- * sig-then-fields pair.  Validity still gates on the ready byte and the
- * tuple compare (the preload changes no ordering: sig is the CLAIM
- * marker, ready the PUBLISH marker, exactly as in sig_find_or_claim). */
-struct sess_pre { uint64_t k0, k1, t0, t1; bool pre; };
-
-BNG_DEV bng_nat_session* sess_find_or_claim(bng_nat_session* t,
-                                            uint32_t mask, uint64_t sig,
-                                            bool* claimed, bool* found,
-                                            sess_pre* h) {
-  *claimed = false; *found = false; h->pre = false;
-  uint32_t slot = (uint32_t)sig & mask;
-  {
-    bng_nat_session* e = &t[slot];
-    const uint64_t* e64 = (const uint64_t*)e;
-    uint64_t k  = rlx_load64(e64 + 0);
-    uint64_t k0 = rlx_load64(e64 + 1);
-    uint64_t k1 = rlx_load64(e64 + 2);
-    uint64_t t0 = rlx_load64(e64 + 3);
-    uint64_t t1 = rlx_load64(e64 + 4);
-    if (k == sig) {
-      *found = true;
-      h->pre = true; h->k0 = k0; h->k1 = k1; h->t0 = t0; h->t1 = t1;
-      return e;
-    }
-    if (k == BNG_KEY_EMPTY) {
-      uint64_t expect = BNG_KEY_EMPTY;
-      uint64_t old = __hip_atomic_compare_exchange_strong(
-          &e->sig, &expect, sig, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
-          __HIP_MEMORY_SCOPE_AGENT) ? BNG_KEY_EMPTY : expect;
-      if (old == BNG_KEY_EMPTY) { *claimed = true; return e; }
-      if (old == sig) { *found = true; return e; }
-      /* lost to a different key: keep probing */
-    }
-  }
-  for (int i = 1; i < BNG_MAX_PROBE; ++i) {
-    bng_nat_session* e = &t[(slot + i) & mask];
-    uint64_t k = __hip_atomic_load(&e->sig, __ATOMIC_RELAXED,
-                                   __HIP_MEMORY_SCOPE_AGENT);
-    if (k == sig) { *found = true; return e; }
-    if (k == BNG_KEY_EMPTY) {
-      uint64_t old = __hip_atomic_compare_exchange_strong(
-          &e->sig, &k, sig, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
-          __HIP_MEMORY_SCOPE_AGENT) ? BNG_KEY_EMPTY : k;
-      if (old == BNG_KEY_EMPTY) { *claimed = true; return e; }
-      if (old == sig) { *found = true; return e; }
-    }
   }
   return nullptr;
 }
@@ -769,9 +747,8 @@ BNG_DEV int nat_egress_process(pktctx& c, const nat_tables& T, nat_flags& F,
 
   uint64_t sig = bng_tuple_sig(c.saddr, c.daddr, sport, dport, c.proto);
   bool claimed, found;
-  sess_pre H;
-  bng_nat_session* sess = sess_find_or_claim(T.sessions, T.sess_mask, sig,
-                                             &claimed, &found, &H);
+  bng_nat_session* sess = sig_find_or_claim(T.sessions, T.sess_mask, sig,
+                                            &claimed, &found);
   if (!sess) { F.passed = true; return BNG_PASS; }  /* table section full */
 
   uint32_t nat_ip; uint16_t nat_port;
@@ -779,18 +756,12 @@ BNG_DEV int nat_egress_process(pktctx& c, const nat_tables& T, nat_flags& F,
     /* HIT fast path: the producer plain-stored fields then issued an
      * agent release (L2 write-back) before the sc1 `ready` store, so
      * relaxed agent (sc1, L2-served) loads are always fresh — no
-     * acquire fence (which would flush this CU's whole L1) needed.
-     * On a first-slot hit the 4 words arrived WITH the sig probe. */
+     * acquire fence (which would flush this CU's whole L1) needed. */
     const uint64_t* e64 = (const uint64_t*)sess;
-    uint64_t k0, k1, t0, t1;
-    if (H.pre) {
-      k0 = H.k0; k1 = H.k1; t0 = H.t0; t1 = H.t1;
-    } else {
-      k0 = rlx_load64(e64 + 1);   /* src_ip | dst_ip<<32 */
-      k1 = rlx_load64(e64 + 2);   /* sport | dport<<16 | proto<<32 */
-      t0 = rlx_load64(e64 + 3);   /* nat_ip | nat_port<<32 | orig_port<<48 */
-      t1 = rlx_load64(e64 + 4);   /* orig_ip | state<<32 | hairpin<<40 | ready<<48 */
-    }
+    uint64_t k0 = rlx_load64(e64 + 1);   /* src_ip | dst_ip<<32 */
+    uint64_t k1 = rlx_load64(e64 + 2);   /* sport | dport<<16 | proto<<32 */
+    uint64_t t0 = rlx_load64(e64 + 3);   /* nat_ip | nat_port<<32 | orig_port<<48 */
+    uint64_t t1 = rlx_load64(e64 + 4);   /* orig_ip | state<<32 | hairpin<<40 | ready<<48 */
     if (!((t1 >> 48) & 0xFF)) {          /* creation still in flight */
       if (!bng_wait_ready(&sess->ready, 8192)) {
         F.passed = true; return BNG_PASS;
@@ -1719,6 +1690,8 @@ __global__ void nat_sweep_kernel(bng_nat_session* sessions, uint32_t n_slots,
                                  bng_nat_reverse* reverse, uint32_t rev_mask,
                                  bng_subctx* ctx,
                                  uint32_t ctx_mask,
+                                 bng_eim_entry* eim, uint32_t eim_slots,
+                                 uint64_t eim_to,
                                  uint64_t now_ns, uint64_t udp_to,
                                  uint64_t tcp_est_to, uint64_t tcp_tr_to,
                                  uint64_t icmp_to,
@@ -1771,6 +1744,24 @@ __global__ void nat_sweep_kernel(bng_nat_session* sessions, uint32_t n_slots,
     stat_inc(&stats[BNG_NS_SESS_EXPIRED], true);
   }
   (void)any;
+  /* EIM idle expiry — the LRU-eviction analog for the endpoint-
+   * independent mapping table (the reference's eim_table is an LRU
+   * map, nat44.c:231-238).  Safe to expire under live sessions: the
+   * reverse map demuxes ingress by the full remote endpoint, so only
+   * endpoint-independence for FUTURE flows is (correctly) lost. */
+  if (eim != nullptr && eim_to != 0) {
+    for (uint32_t i = tid; i < eim_slots; i += nthreads) {
+      bng_eim_entry* e = &eim[i];
+      uint64_t sig = e->sig;
+      if (sig == BNG_KEY_EMPTY || sig == BNG_KEY_TOMBSTONE || !e->ready)
+        continue;
+      if (now_ns - e->last_used < eim_to)
+        continue;
+      e->ready = 0;
+      __hip_atomic_store(&e->sig, BNG_KEY_TOMBSTONE, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+    }
+  }
 }
 
 /* shard-owner computation for multi-GPU steering: owner by subscriber
@@ -1919,13 +1910,16 @@ void bng_launch_binding_delete(void* t, uint32_t mask, const void* keys,
 
 void bng_launch_nat_sweep(void* sessions, uint32_t n_slots, void* reverse,
                           uint32_t rev_mask, void* subnat,
-                          uint32_t subnat_mask, uint64_t now_ns,
+                          uint32_t subnat_mask, void* eim,
+                          uint32_t eim_slots, uint64_t eim_to,
+                          uint64_t now_ns,
                           uint64_t udp_to, uint64_t tcp_est_to,
                           uint64_t tcp_tr_to, uint64_t icmp_to, void* stats,
                           hipStream_t s) {
   hipLaunchKernelGGL(nat_sweep_kernel, dim3(2048), dim3(256), 0, s,
       (bng_nat_session*)sessions, n_slots, (bng_nat_reverse*)reverse,
-      rev_mask, (bng_subctx*)subnat, subnat_mask, now_ns, udp_to,
+      rev_mask, (bng_subctx*)subnat, subnat_mask, (bng_eim_entry*)eim,
+      eim_slots, eim_to, now_ns, udp_to,
       tcp_est_to, tcp_tr_to, icmp_to, (unsigned long long*)stats);
 }
 

@@ -25,6 +25,7 @@ from .golden import (BindingRec, GoldenDataplane, PoolRecord, QosBucketRec,
 
 # NAT timeouts (ref nat44.c:50-53)
 UDP_TIMEOUT_NS = 120 * 10**9
+EIM_TIMEOUT_NS = 2 * 3600 * 10**9     # idle EIM mapping lifetime
 TCP_TRANSIENT_TIMEOUT_NS = 240 * 10**9
 TCP_EST_TIMEOUT_NS = 7200 * 10**9
 ICMP_TIMEOUT_NS = 60 * 10**9
@@ -259,6 +260,7 @@ class HipLauncher:
         """Expire timed-out sessions (the LRU/timeout sweeper; the reference
         gets eviction from BPF LRU maps)."""
         self.ext.nat_sweep(self.sessions, self.reverse, self.subctx,
+                           self.eim, EIM_TIMEOUT_NS,
                            now_ns or time.time_ns(), UDP_TIMEOUT_NS,
                            TCP_EST_TIMEOUT_NS, TCP_TRANSIENT_TIMEOUT_NS,
                            ICMP_TIMEOUT_NS, self.nat_stats)
@@ -605,6 +607,10 @@ class GoldenLauncher:
             if blk:
                 blk.sessions_active -= 1
             self.dp.nat_stats[abi.NS_SESS_EXPIRED] += 1
+        # EIM idle expiry (mirrors the GPU sweep's LRU analog)
+        for ek in [k for k, e in self.dp.eim.items()
+                   if now - e.last_used >= EIM_TIMEOUT_NS]:
+            del self.dp.eim[ek]
 
     # QoS
     def set_qos_policy(self, ip, rate_bps, burst_bytes, priority=0,

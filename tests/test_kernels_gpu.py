@@ -1027,3 +1027,56 @@ class TestICMPNatGPU:
         assert bytes(h2[:42]) == res2[0][1][:42]
         assert st.unpack_from(">H", h2, 38)[0] == 777  # restored
         assert st.unpack_from(">I", h2, 30)[0] == ip2u32(PRIV)
+
+
+class TestTableLifecycleGPU:
+    """Tombstone reclamation under churn — the LRU-reuse property the
+    reference gets from BPF LRU maps (previously: tombstones
+    accumulated forever in device-claimed tables)."""
+
+    def _small_nat_launcher(self):
+        from bng_amd.dataplane.launcher import HipLauncher
+        gpu = HipLauncher(sub_log2=12, sess_log2=6, eim_log2=6,
+                          subnat_log2=12, qos_log2=12, binding_log2=12,
+                          n_pools=16)
+        gpu.set_server_config(mac_bytes("02:00:00:00:00:01"),
+                              ip2u32("10.0.0.1"))
+        gpu.add_subscriber_nat(ip2u32(PRIV), ip2u32(PUB), 1024, 2047,
+                               subscriber_id=1)
+        return gpu
+
+    def test_session_churn_reuses_tombstones(self):
+        from bng_amd.dataplane.launcher import UDP_TIMEOUT_NS
+        gpu = self._small_nat_launcher()
+        pkt = [build_ipv4("aa:00:00:00:00:01", "02:00:00:00:00:01",
+                          ip2u32(PRIV), ip2u32(DST), proto=17,
+                          sport=5555, dport=53)]
+        cycles = 200                   # >> the 64-slot session table
+        now = NOW_NS
+        for k in range(cycles):
+            d, lns = gpu.make_batch(pkt, stride=128)
+            assert gpu.nat44(d, lns, egress=True,
+                             now_ns=now).cpu().tolist() == [FWD], k
+            now += UDP_TIMEOUT_NS + 10**9
+            gpu.sweep_nat(now_ns=now)
+        st = gpu.nat_get_stats()
+        assert st["sessions_created"] == cycles
+        assert st["sessions_expired"] == cycles
+        assert st["packets_passed"] == 0       # chain never exhausted
+
+    def test_eim_idle_expiry(self):
+        from bng_amd.dataplane.launcher import EIM_TIMEOUT_NS
+        gpu = self._small_nat_launcher()
+        pkt = [build_ipv4("aa:00:00:00:00:01", "02:00:00:00:00:01",
+                          ip2u32(PRIV), ip2u32(DST), proto=17,
+                          sport=7777, dport=53)]
+        d, lns = gpu.make_batch(pkt, stride=128)
+        gpu.nat44(d, lns, egress=True, now_ns=NOW_NS)
+        assert gpu.nat_get_stats()["eim_misses"] == 1
+        # idle past the EIM lifetime: mapping is reclaimed
+        gpu.sweep_nat(now_ns=NOW_NS + EIM_TIMEOUT_NS + 10**9)
+        d2, l2 = gpu.make_batch(pkt, stride=128)
+        gpu.nat44(d2, l2, egress=True,
+                  now_ns=NOW_NS + EIM_TIMEOUT_NS + 2 * 10**9)
+        st = gpu.nat_get_stats()
+        assert st["eim_misses"] == 2           # fresh mapping created
