@@ -175,9 +175,8 @@ BNG_DEV E* sig_find_or_claim(E* t, uint32_t mask, uint64_t sig,
       }
       if (k == BNG_KEY_EMPTY) {
         E* d = first_tomb >= 0 ? &t[first_tomb] : e;
-        uint64_t expect = first_tomb >= 0 ? BNG_KEY_TOMBSTONE
-                                          : BNG_KEY_EMPTY;
-        uint64_t want = expect;
+        uint64_t want = first_tomb >= 0 ? BNG_KEY_TOMBSTONE
+                                        : BNG_KEY_EMPTY;
         if (__hip_atomic_compare_exchange_strong(
                 &d->sig, &want, sig, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
                 __HIP_MEMORY_SCOPE_AGENT)) {
@@ -185,23 +184,26 @@ BNG_DEV E* sig_find_or_claim(E* t, uint32_t mask, uint64_t sig,
           return d;
         }
         if (want == sig) { *found = true; return d; }
-        break;   /* lost the slot race to a different key: rescan */
+        /* lost the slot to a DIFFERENT key: keep probing.  If it was
+         * the remembered tombstone, forget it and retry this empty
+         * slot; if it was the empty slot itself, scan past it. */
+        if (first_tomb >= 0) { first_tomb = -1; --i; }
+        continue;
       }
     }
-    if (first_tomb >= 0) {
-      /* chain full but reclaimable: take the tombstone directly */
-      E* d = &t[first_tomb];
-      uint64_t want = BNG_KEY_TOMBSTONE;
-      if (__hip_atomic_compare_exchange_strong(
-              &d->sig, &want, sig, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
-              __HIP_MEMORY_SCOPE_AGENT)) {
-        *claimed = true;
-        return d;
-      }
-      if (want == sig) { *found = true; return d; }
-      continue;   /* raced: rescan */
+    if (first_tomb < 0)
+      return nullptr;            /* chain genuinely full of other keys */
+    /* chain full but reclaimable: take the tombstone directly */
+    E* d = &t[first_tomb];
+    uint64_t want = BNG_KEY_TOMBSTONE;
+    if (__hip_atomic_compare_exchange_strong(
+            &d->sig, &want, sig, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+            __HIP_MEMORY_SCOPE_AGENT)) {
+      *claimed = true;
+      return d;
     }
-    return nullptr;   /* chain genuinely full of other keys */
+    if (want == sig) { *found = true; return d; }
+    /* raced: rescan */
   }
   return nullptr;
 }
