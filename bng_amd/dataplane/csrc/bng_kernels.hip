@@ -162,6 +162,16 @@ BNG_DEV E* sig_find_or_claim(E* t, uint32_t mask, uint64_t sig,
                              bool* claimed, bool* found) {
   *claimed = false; *found = false;
   uint32_t slot = (uint32_t)sig & mask;
+  /* pass 1 — pure hit scan, identical cost to the pre-reclamation hot
+   * path (hits dominate: established flows) */
+  for (int i = 0; i < BNG_MAX_PROBE; ++i) {
+    E* e = &t[(slot + i) & mask];
+    uint64_t k = __hip_atomic_load(&e->sig, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+    if (k == sig) { *found = true; return e; }
+    if (k == BNG_KEY_EMPTY) break;
+  }
+  /* pass 2 — claim scan (session-create path only) */
   for (int attempt = 0; attempt < 4; ++attempt) {
     int first_tomb = -1;
     for (int i = 0; i < BNG_MAX_PROBE; ++i) {
