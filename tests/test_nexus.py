@@ -245,3 +245,80 @@ class TestVLANAllocatorFull:
         v.allocate("nte-2")
         from bng_amd.nexus.store import TypedStore
         assert TypedStore(store, "nexus/vlans").list() == {"nte-2": "100:2"}
+
+
+class TestHTTPAllocatorErrors:
+    """Error scenarios (ref http_allocator_test.go HTTPErrorScenarios /
+    NetworkErrors / MalformedJSONResponse)."""
+
+    def _broken_server(self, behavior):
+        import threading
+        from http.server import BaseHTTPRequestHandler, HTTPServer
+
+        class H(BaseHTTPRequestHandler):
+            def _go(self):
+                if behavior == "500":
+                    self.send_response(500)
+                    self.end_headers()
+                elif behavior == "garbage":
+                    self.send_response(200)
+                    self.send_header("Content-Type", "application/json")
+                    self.end_headers()
+                    self.wfile.write(b"{not json")
+                elif behavior == "404":
+                    self.send_response(404)
+                    self.end_headers()
+
+            do_GET = do_POST = do_DELETE = _go
+
+            def log_message(self, *a):
+                pass
+        srv = HTTPServer(("127.0.0.1", 0), H)
+        threading.Thread(target=srv.serve_forever, daemon=True).start()
+        return srv
+
+    def test_server_error_raises(self):
+        from bng_amd.nexus.http_allocator import (HTTPAllocator,
+                                                  HTTPAllocatorError)
+        srv = self._broken_server("500")
+        try:
+            a = HTTPAllocator(f"http://127.0.0.1:{srv.server_port}",
+                              timeout=2)
+            with pytest.raises(HTTPAllocatorError):
+                a.allocate_ipv4("p1", "sub-1")
+            with pytest.raises(HTTPAllocatorError):
+                a.release("p1", "sub-1")
+            assert a.health_check() is False
+        finally:
+            srv.shutdown()
+
+    def test_lookup_404_is_no_allocation(self):
+        from bng_amd.nexus.http_allocator import (HTTPAllocator,
+                                                  NoAllocationError)
+        srv = self._broken_server("404")
+        try:
+            a = HTTPAllocator(f"http://127.0.0.1:{srv.server_port}",
+                              timeout=2)
+            with pytest.raises(NoAllocationError):
+                a.lookup_ipv4("sub-1")
+        finally:
+            srv.shutdown()
+
+    def test_malformed_json_raises_not_crashes(self):
+        from bng_amd.nexus.http_allocator import (HTTPAllocator,
+                                                  HTTPAllocatorError)
+        srv = self._broken_server("garbage")
+        try:
+            a = HTTPAllocator(f"http://127.0.0.1:{srv.server_port}",
+                              timeout=2)
+            with pytest.raises((HTTPAllocatorError, ValueError)):
+                a.allocate_ipv4("p1", "sub-1")
+        finally:
+            srv.shutdown()
+
+    def test_network_error_raises(self):
+        from bng_amd.nexus.http_allocator import HTTPAllocator
+        a = HTTPAllocator("http://127.0.0.1:1", timeout=0.5)
+        with pytest.raises(Exception):
+            a.allocate_ipv4("p1", "sub-1")
+        assert a.health_check() is False
