@@ -407,6 +407,10 @@ def main():
         log(rank, f"[bench] p50 {p50_us:.1f}us p99 {p99_us:.1f}us "
                   f"({args.lat_batch}-pkt DHCP batch)")
 
+    if distributed:
+        import torch.distributed as dist
+        dist.barrier()     # hold all ranks until rank 0's latency phase
+
     if rank == 0:
         result = {
             "metric": "mpps",
