@@ -37,7 +37,6 @@ from __future__ import annotations
 import argparse
 import json
 import os
-import struct
 import time
 
 import numpy as np

@@ -14,7 +14,6 @@ import numpy as np
 def main():
     import bench
     import torch
-    from bng_amd.dataplane import abi
     from bng_amd.dataplane.launcher import HipLauncher
     from bng_amd.dataplane.packets import ip2u32
     NOW = 1_700_000_000
