@@ -445,3 +445,25 @@ class TestAntispoof:
         arp = mac_bytes("ff:ff:ff:ff:ff:ff") + mac_bytes(MAC) + \
             struct.pack(">H", 0x0806) + b"\x00" * 28
         assert dp.antispoof(arp) == FWD
+
+
+class TestQoSRefillClamp:
+    def test_long_idle_refill_capped_at_burst(self):
+        """Elapsed > 100 s credits exactly one burst, not rate*elapsed
+        (the clamp that stops counter-wrap abuse; kernel qos_tb_check)."""
+        dp = GoldenDataplane()
+        ip = ip2u32("10.0.1.50")
+        dp.qos_egress[ip] = QosBucketRec(rate_bps=8 * 10**9,  # 1 GB/s
+                                         burst_bytes=150, tokens=0,
+                                         last_update=0)
+        dp.now_ns = 200 * 10**9        # 200 s idle: naive credit = 200 GB
+        big = build_ipv4(MAC, "02:00:00:00:00:01", ip2u32(DST), ip2u32(PRIV),
+                         proto=17, sport=53, dport=5555,
+                         payload=b"x" * 100)
+        small = build_ipv4(MAC, "02:00:00:00:00:01", ip2u32(DST),
+                           ip2u32(PRIV), proto=17, sport=53, dport=5556)
+        # 142-B frame passes on the capped 150-B burst; the 8 B left
+        # cannot cover the next 42-B frame — proof the 200-GB naive
+        # credit never materialized
+        assert dp.qos(bytes(big), "egress") == FWD
+        assert dp.qos(bytes(small), "egress") == DROP
