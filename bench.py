@@ -251,7 +251,7 @@ def main():
 
     now_sec = 1_700_000_000
     from bng_amd.dataplane.launcher import HipLauncher
-    from bng_amd.parallel.sharding import exchange
+    from bng_amd.parallel.sharding import ExchangeBuffers, exchange
 
     t0 = time.perf_counter()
     # size tables for the subscriber count (load factor <= 0.5)
@@ -290,6 +290,11 @@ def main():
     # while the main-stream kernel still reads it)
     orders = [torch.zeros(args.batch, dtype=torch.int32, device=device)
               for _ in range(nbuf)]
+    # pre-allocated recv-side exchange buffers, one set per in-flight
+    # batch (no per-step prep-stream allocations with RCCL in the loop);
+    # 2x capacity covers hashring skew + steer-all variance
+    exbufs = [ExchangeBuffers(2 * args.batch, args.stride, device, world)
+              for _ in range(nbuf)] if distributed else None
     prep_stream = torch.cuda.Stream(device=device) if overlap else None  # noqa: E501
     prep_done = [torch.cuda.Event() for _ in range(nbuf)]
     work_free = [torch.cuda.Event() for _ in range(nbuf)]
@@ -309,7 +314,7 @@ def main():
             d, l = works[b], lens
             if distributed:
                 owner = launcher.shard_owner(d, l, world)
-                d, l = exchange(d, l, owner)
+                d, l = exchange(d, l, owner, bufs=exbufs[b])
             o = None
             if not args.no_sort:
                 if d is works[b]:
@@ -318,13 +323,14 @@ def main():
                     orders[b].copy_(idx.to(torch.int32))
                     o = orders[b]
                 else:
-                    # exchanged batch: size varies per rank; fresh
-                    # prep-stream tensors, released to the main stream
-                    # via record_stream at consumption
-                    cls = torch.empty(l.numel(), dtype=torch.uint8,
-                                      device=device)
+                    # exchanged batch: size varies per rank; classify
+                    # and sort into this slot's persistent scratch
+                    m = l.numel()
+                    cls = exbufs[b].cls[:m]
                     launcher.ext.pkt_class(d, l, cls)
-                    o = torch.argsort(cls, stable=True).to(torch.int32)
+                    o = exbufs[b].order[:m]
+                    o.copy_(torch.argsort(cls, stable=True)
+                            .to(torch.int32))
             batches[b] = (d, l, o)
             prep_done[b].record(prep_stream)
 
@@ -334,11 +340,6 @@ def main():
             b = k % nbuf
             cur.wait_event(prep_done[b])
             d, l, o = batches[b]
-            if d is not works[b]:                  # prep-stream allocs
-                d.record_stream(cur)
-                l.record_stream(cur)
-                if o is not None:
-                    o.record_stream(cur)
             launcher.uplink(d, l, now_ns=now_ns, now_sec=now_sec,
                             sort_by_type=not args.no_sort, order=o)
             work_free[b].record(cur)

@@ -29,11 +29,47 @@ def bucket_by_owner(data, lens, owner, world_size: int):
     return data[perm], lens[perm], counts, perm
 
 
-def exchange(data, lens, owner, group=None) -> Tuple["object", "object"]:
+class ExchangeBuffers:
+    """Pre-allocated recv-side buffers for the steering all-to-all.
+
+    The first real multi-GPU run must not allocate fresh recv tensors on
+    the prep stream every step (caching-allocator churn + cross-stream
+    lifetime hazards once RCCL is in the loop — round-1 VERDICT task 2).
+    Capacity is fixed at construction; exchange(..., bufs=) writes into
+    these tensors and returns narrow views.  Reuse across in-flight
+    batches is the caller's job (bench keeps one set per pipeline slot,
+    gated by its work-free event)."""
+
+    def __init__(self, capacity: int, stride: int, device, world: int):
+        import torch
+        assert stride % 8 == 0, "packet slot stride must be a multiple of 8"
+        self.capacity, self.stride, self.world = capacity, stride, world
+        self.data = torch.empty((capacity, stride), dtype=torch.uint8,
+                                device=device)
+        self.lens64 = torch.empty(capacity, dtype=torch.int64, device=device)
+        self.lens16 = torch.empty(capacity, dtype=torch.int16, device=device)
+        # classify/sort scratch for the type-sorted pipeline on the
+        # exchanged batch (same lifetime rules as data/lens)
+        self.cls = torch.empty(capacity, dtype=torch.uint8, device=device)
+        self.order = torch.empty(capacity, dtype=torch.int32, device=device)
+
+
+class ExchangeOverflow(RuntimeError):
+    """Received more packets than the pre-allocated exchange capacity.
+
+    Size ExchangeBuffers for the worst shard skew you admit (bench uses
+    2x the per-rank batch; the RSS arrival model keeps the expected
+    crossing at the DHCP fraction, and hashring uniformity over >=100k
+    subscribers bounds data skew far below 2x)."""
+
+
+def exchange(data, lens, owner, group=None,
+             bufs: "ExchangeBuffers" = None) -> Tuple["object", "object"]:
     """All-to-all steer packets to their owning rank.
 
     data: [N, stride] uint8, lens: [N] int16, owner: [N] int32/int64.
-    Returns (data_recv [M, stride], lens_recv [M]) on this rank.
+    Returns (data_recv [M, stride], lens_recv [M]) on this rank — views
+    into `bufs` when given (steady-state path), fresh tensors otherwise.
     Works over nccl(RCCL) with device tensors and gloo with CPU tensors.
     """
     import torch
@@ -56,15 +92,27 @@ def exchange(data, lens, owner, group=None) -> Tuple["object", "object"]:
     # ship packet bytes as int64 words (gloo rejects uint8/int16; identical
     # bytes either way) — one contiguous all_to_all_single per tensor
     words = stride // 8
-    data_recv = data.new_empty((m, stride))
+    if bufs is not None:
+        if m > bufs.capacity:
+            raise ExchangeOverflow(
+                f"recv {m} packets > exchange capacity {bufs.capacity}")
+        assert stride == bufs.stride
+        data_recv = bufs.data[:m]
+        lens_recv64 = bufs.lens64[:m]
+    else:
+        data_recv = data.new_empty((m, stride))
+        lens_recv64 = torch.empty(m, dtype=torch.int64, device=lens.device)
     dist.all_to_all_single(data_recv.view(-1).view(torch.int64),
                            data_s.contiguous().view(-1).view(torch.int64),
                            [c * words for c in out_splits],
                            [c * words for c in in_splits], group=group)
-    lens_recv64 = torch.empty(m, dtype=torch.int64, device=lens.device)
     dist.all_to_all_single(lens_recv64,
                            lens_s.contiguous().to(torch.int64), out_splits,
                            in_splits, group=group)
+    if bufs is not None and lens.dtype == torch.int16:
+        lens_recv = bufs.lens16[:m]
+        lens_recv.copy_(lens_recv64.to(torch.int16))
+        return data_recv, lens_recv
     return data_recv, lens_recv64.to(lens.dtype)
 
 

@@ -194,3 +194,147 @@ def test_bench_step_order_symmetry_gloo(tmp_path):
         p.join(timeout=180)
         assert p.exitcode == 0
     assert sum(got.values()) == world * 6 * 256   # conservation
+
+
+def _prealloc_worker(rank, world, rendezvous_file, q, skew):
+    """Steady-state exchange through pre-allocated ExchangeBuffers over
+    several steps with UNEVEN splits: 'skew' concentrates ownership so
+    per-rank recv counts differ wildly step to step (the shape the first
+    real SCALE run sees; round-1 VERDICT task 2)."""
+    import torch.distributed as dist
+    dist.init_process_group(
+        "gloo", init_method=f"file://{rendezvous_file}",
+        rank=rank, world_size=world)
+    from bng_amd.parallel.sharding import (ExchangeBuffers,
+                                           ExchangeOverflow, exchange)
+    n, stride, steps = 128, 64, 5
+    bufs = ExchangeBuffers(n * world, stride, "cpu", world)
+    total, overflowed = 0, False
+    for s in range(steps):
+        rng = np.random.default_rng(7000 + rank * 100 + s)
+        data = rng.integers(0, 255, size=(n, stride), dtype=np.uint8)
+        if skew == "all_to_zero":
+            owners = np.zeros(n, dtype=np.int64)
+        elif skew == "rotating":       # rank r sends all to (r+s) % world
+            owners = np.full(n, (rank + s) % world, dtype=np.int64)
+        else:                          # uneven random
+            owners = np.minimum(
+                rng.integers(0, world * 2, size=n), world - 1
+            ).astype(np.int64)
+        data[:, 2] = owners.astype(np.uint8)
+        lens = np.full(n, stride, dtype=np.uint16)
+        d2, l2 = exchange(torch.from_numpy(data),
+                          torch.from_numpy(lens.view(np.int16)),
+                          torch.from_numpy(owners), bufs=bufs)
+        assert (d2.numpy()[:, 2] == rank).all()
+        assert l2.dtype == torch.int16
+        total += l2.numel()
+    # overflow must raise, not corrupt: tiny capacity, ring ownership so
+    # EVERY rank receives n > 4 and raises after the counts exchange but
+    # before posting the data all-to-all (no deadlock — symmetric abort)
+    tiny = ExchangeBuffers(4, stride, "cpu", world)
+    owners = np.full(n, (rank + 1) % world, dtype=np.int64)
+    try:
+        exchange(torch.from_numpy(data), torch.from_numpy(lens.view(np.int16)),
+                 torch.from_numpy(owners), bufs=tiny)
+    except ExchangeOverflow:
+        overflowed = True
+    dist.barrier()
+    q.put((rank, total, overflowed))
+    dist.destroy_process_group()
+
+
+def _run_prealloc(world, tmp_path, skew, expected_total):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    rv = str(tmp_path / f"rdv_{world}_{skew}")
+    procs = [ctx.Process(target=_prealloc_worker,
+                         args=(r, world, rv, q, skew)) for r in range(world)]
+    for p in procs:
+        p.start()
+    got = {}
+    for _ in range(world):
+        rank, n, ovf = q.get(timeout=240)
+        got[rank] = (n, ovf)
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    assert sum(n for n, _ in got.values()) == expected_total
+    assert all(ovf for _, ovf in got.values()), \
+        "every rank must see ExchangeOverflow on tiny capacity"
+    return got
+
+
+def test_exchange_prealloc_world4_uneven(tmp_path):
+    got = _run_prealloc(4, tmp_path, "uneven", 4 * 5 * 128)
+    # uneven skew: rank 3 owns ~9/16 of traffic
+    assert got[3][0] > got[1][0]
+
+
+def test_exchange_prealloc_world4_all_to_zero(tmp_path):
+    got = _run_prealloc(4, tmp_path, "all_to_zero", 4 * 5 * 128)
+    assert got[0][0] == 4 * 5 * 128         # rank 0 got everything
+    assert got[1][0] == 0
+
+
+def test_exchange_prealloc_world8_rotating(tmp_path):
+    """World-8 rehearsal — the driver's SCALE shape — with per-step
+    rotating hot-spot ownership (each step one rank receives 8x)."""
+    got = _run_prealloc(8, tmp_path, "rotating", 8 * 5 * 128)
+
+
+def _steer_all_worker(rank, world, rendezvous_file, q):
+    """bench.py --steer-all rehearsal: every packet crosses the exchange,
+    owners from the real shard_owner hash over generated traffic."""
+    import torch.distributed as dist
+    dist.init_process_group(
+        "gloo", init_method=f"file://{rendezvous_file}",
+        rank=rank, world_size=world)
+    from bng_amd.parallel.sharding import ExchangeBuffers, exchange
+    from bng_amd.parallel.hashring import owner_of_ip, owner_of_mac
+    import bench
+    n, stride = 512, 512
+    d_np, l_np = bench.gen_batch(n, 10_000, 0.1, stride, seed=31 + rank,
+                                 rank=rank, world=world, steer_all=True)
+    owners = np.empty(n, dtype=np.int64)
+    for i in range(n):
+        if l_np[i] > 64:
+            owners[i] = owner_of_mac(bytes(d_np[i, 70:76]), world)
+        else:
+            owners[i] = owner_of_ip(
+                int.from_bytes(bytes(d_np[i, 26:30]), "big"), world)
+    bufs = ExchangeBuffers(2 * n, stride, "cpu", world)
+    d2, l2 = exchange(torch.from_numpy(d_np),
+                      torch.from_numpy(l_np.view(np.int16)),
+                      torch.from_numpy(owners), bufs=bufs)
+    # every received data packet's src IP must hash to this rank
+    got = d2.numpy()
+    l2 = l2.numpy().view(np.uint16)
+    for i in range(got.shape[0]):
+        if l2[i] == 64:
+            assert owner_of_ip(
+                int.from_bytes(bytes(got[i, 26:30]), "big"), world) == rank
+        else:
+            assert owner_of_mac(bytes(got[i, 70:76]), world) == rank
+    dist.barrier()
+    q.put((rank, got.shape[0]))
+    dist.destroy_process_group()
+
+
+def test_steer_all_world4_preallocated(tmp_path):
+    world = 4
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    rv = str(tmp_path / "rdv_sa")
+    procs = [ctx.Process(target=_steer_all_worker, args=(r, world, rv, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    tot = 0
+    for _ in range(world):
+        _, n = q.get(timeout=240)
+        tot += n
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    assert tot == world * 512
