@@ -214,6 +214,151 @@ def log(rank, msg):
         print(msg, file=sys.stderr, flush=True)
 
 
+def build_cell_map(lens_np, stride):
+    """NIC-RX-ring analog layout: each frame occupies ceil(len/64)
+    64-byte cells in a contiguous pinned buffer (DHCP frames get a full
+    8-cell slot so in-place OFFER growth never spills).  Returns
+    (n_cells_total, slot_cell_idx[int64 C]) where slot_cell_idx[c] is
+    the destination 64-B cell inside the [n, stride] slot layout."""
+    cells_per_slot = stride // 64
+    cells = np.where(lens_np > 64, cells_per_slot,
+                     (lens_np.astype(np.int64) + 63) // 64)
+    off = np.zeros(len(lens_np) + 1, dtype=np.int64)
+    np.cumsum(cells, out=off[1:])
+    C = int(off[-1])
+    pkt_of_cell = np.repeat(np.arange(len(lens_np), dtype=np.int64), cells)
+    within = np.arange(C, dtype=np.int64) - off[pkt_of_cell]
+    slot_cell_idx = pkt_of_cell * cells_per_slot + within
+    return C, slot_cell_idx
+
+
+def host_io_phase(launcher, pristine, lens, lens_np, args, now_sec,
+                  steps, warmup):
+    """Host-boundary throughput: packets cross PCIe both ways every step
+    (round-1 VERDICT task 1 — no measured number previously included the
+    host boundary).  Pipeline per slot: pinned-host packed batch
+    --H2D--> unpack to stride slots --uplink kernel--> pack --D2H-->
+    pinned host.  H2D, compute, and D2H overlap across in-flight slots
+    on three streams, the DMA-conveyor steady state.
+
+    Returns (host_mpps, per_step_s list, pipe_lat_s list)."""
+    import torch
+    device = launcher.device
+    stride = args.stride
+    n = lens.numel()
+    C, slot_idx_np = build_cell_map(lens_np, stride)
+    slot_idx = torch.from_numpy(slot_idx_np).to(device)
+
+    # packed pinned-host input: gather the pristine batch's cells once
+    dev_packed_in = torch.index_select(
+        pristine.view(-1, 64), 0, slot_idx).contiguous()
+    host_in = torch.empty((C, 64), dtype=torch.uint8, pin_memory=True)
+    host_in.copy_(dev_packed_in)
+    del dev_packed_in
+
+    nbuf = 3
+    din = [torch.empty((C, 64), dtype=torch.uint8, device=device)
+           for _ in range(nbuf)]
+    dout = [torch.empty((C, 64), dtype=torch.uint8, device=device)
+            for _ in range(nbuf)]
+    works = [torch.empty((n, stride), dtype=torch.uint8, device=device)
+             for _ in range(nbuf)]
+    hout = [torch.empty((C, 64), dtype=torch.uint8, pin_memory=True)
+            for _ in range(nbuf)]
+    clss = [torch.empty(n, dtype=torch.uint8, device=device)
+            for _ in range(nbuf)]
+    orders = [torch.zeros(n, dtype=torch.int32, device=device)
+              for _ in range(nbuf)]
+
+    h2d_stream = torch.cuda.Stream(device=device)
+    d2h_stream = torch.cuda.Stream(device=device)
+    ev_h2d = [torch.cuda.Event() for _ in range(nbuf)]
+    ev_comp = [torch.cuda.Event() for _ in range(nbuf)]
+    ev_d2h = [torch.cuda.Event() for _ in range(nbuf)]
+    for ev in ev_d2h:
+        ev.record()
+
+    base_ns = now_sec * 10**9
+
+    def step(k):
+        b = k % nbuf
+        with torch.cuda.stream(h2d_stream):
+            h2d_stream.wait_event(ev_d2h[b])       # slot free again
+            din[b].copy_(host_in, non_blocking=True)   # RX DMA (PCIe H2D)
+            ev_h2d[b].record(h2d_stream)
+        cur = torch.cuda.current_stream(device)
+        cur.wait_event(ev_h2d[b])
+        works[b].view(-1, 64).index_copy_(0, slot_idx, din[b])  # ring->slots
+        launcher.ext.pkt_class(works[b], lens, clss[b])
+        orders[b].copy_(torch.argsort(clss[b], stable=True).to(torch.int32))
+        launcher.uplink(works[b], lens, now_ns=base_ns + k * 10**6,
+                        now_sec=now_sec, sort_by_type=True, order=orders[b])
+        torch.index_select(works[b].view(-1, 64), 0, slot_idx,
+                           out=dout[b])            # slots->TX ring
+        ev_comp[b].record(cur)
+        with torch.cuda.stream(d2h_stream):
+            d2h_stream.wait_event(ev_comp[b])
+            hout[b].copy_(dout[b], non_blocking=True)  # TX DMA (PCIe D2H)
+            ev_d2h[b].record(d2h_stream)
+
+    for k in range(warmup):
+        step(k)
+    torch.cuda.synchronize()
+    # per-batch completion stamps via timing events on the D2H stream
+    # (host syncs inside the loop would serialize the conveyor)
+    t_ev = [torch.cuda.Event(enable_timing=True) for _ in range(steps + 1)]
+    t0 = time.perf_counter()
+    with torch.cuda.stream(d2h_stream):
+        t_ev[0].record(d2h_stream)
+    for k in range(steps):
+        step(warmup + k)
+        with torch.cuda.stream(d2h_stream):
+            t_ev[k + 1].record(d2h_stream)
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    per_step = [t_ev[k].elapsed_time(t_ev[k + 1]) * 1e-3
+                for k in range(steps)]
+    host_mpps = n * steps / elapsed / 1e6
+
+    # single-batch pipe latency: host write visible -> TX bytes on host
+    pipe_lat = []
+    for _ in range(16):
+        torch.cuda.synchronize()
+        t = time.perf_counter()
+        step(0)
+        ev_d2h[0].synchronize()
+        pipe_lat.append(time.perf_counter() - t)
+    return host_mpps, per_step, pipe_lat
+
+
+def host_io_dhcp_latency(launcher, args, now_sec, reps=64):
+    """Arrival->TX DHCP OFFER latency THROUGH the host boundary: pinned
+    request batch -> H2D -> dhcp_fastpath -> D2H replies -> host
+    visible.  The reference's <100us P99 target is judged on this
+    number plus batching wait (reported separately)."""
+    import torch
+    device = launcher.device
+    lat_np, lat_lens_np = gen_batch(args.lat_batch, args.subs, 1.0,
+                                    args.stride, seed=778)
+    host_req = torch.from_numpy(lat_np).pin_memory()
+    dev = torch.empty((args.lat_batch, args.stride), dtype=torch.uint8,
+                      device=device)
+    ll = torch.from_numpy(lat_lens_np.view(np.int16)).to(device)
+    host_rep = torch.empty(host_req.shape, dtype=torch.uint8,
+                           pin_memory=True)
+    lats = []
+    for _ in range(reps):
+        torch.cuda.synchronize()
+        t = time.perf_counter()
+        dev.copy_(host_req, non_blocking=True)
+        launcher.dhcp_fastpath(dev, ll, now_sec=now_sec)
+        host_rep.copy_(dev, non_blocking=True)
+        torch.cuda.synchronize()
+        lats.append((time.perf_counter() - t) * 1e6)
+    lats.sort()
+    return lats[len(lats) // 2], lats[int(len(lats) * 0.99)]
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -227,6 +372,10 @@ def main():
     ap.add_argument("--lat-batch", type=int, default=2048)
     ap.add_argument("--lat-reps", type=int, default=64)
     ap.add_argument("--no-latency", action="store_true")
+    ap.add_argument("--no-host-io", action="store_true",
+                    help="skip the host-boundary (PCIe-crossing) phase")
+    ap.add_argument("--host-io-steps", type=int, default=12)
+    ap.add_argument("--host-io-warmup", type=int, default=4)
     ap.add_argument("--no-sort", action="store_true",
                     help="disable on-device type-sort (wave-divergence fix)")
     ap.add_argument("--no-overlap", action="store_true",
@@ -407,6 +556,39 @@ def main():
         log(rank, f"[bench] p50 {p50_us:.1f}us p99 {p99_us:.1f}us "
                   f"({args.lat_batch}-pkt DHCP batch)")
 
+    # host-boundary phase: every packet crosses PCIe both ways (the
+    # number round 1 lacked: a "host-fed rate" distinct from the GPU
+    # pipeline rate).  world==1 only: each GPU has its own PCIe link, so
+    # the per-GPU host-fed rate is the scaling unit.
+    hostio = None
+    if not args.no_host_io and rank == 0 and world == 1:
+        hm, per_step, pipe_lat = host_io_phase(
+            launcher, pristine, lens, lens_np, args, now_sec,
+            args.host_io_steps, args.host_io_warmup)
+        hp50, hp99 = host_io_dhcp_latency(launcher, args, now_sec)
+        # arrival->TX distribution incl. batching wait: a packet arrives
+        # uniformly within its batch accumulation window (one period)
+        # and completes at its batch's D2H; latency = wait + residence
+        period = float(np.median(per_step))
+        res = float(np.median(pipe_lat))
+        u = np.random.default_rng(3).uniform(0, period, 20000)
+        arr = u + res
+        hostio = {
+            "host_fed_mpps": round(hm, 1),
+            "host_batch_period_us": round(period * 1e6, 1),
+            "host_pipe_residence_us_p50": round(
+                float(np.percentile(pipe_lat, 50)) * 1e6, 1),
+            "host_pipe_residence_us_p99": round(
+                float(np.percentile(pipe_lat, 99)) * 1e6, 1),
+            "arrival_to_tx_us_p50": round(
+                float(np.percentile(arr, 50)) * 1e6, 1),
+            "arrival_to_tx_us_p99": round(
+                float(np.percentile(arr, 99)) * 1e6, 1),
+            "host_dhcp_p50_us": round(hp50, 1),
+            "host_dhcp_p99_us": round(hp99, 1),
+        }
+        log(rank, f"[bench] host-io {hostio}")
+
     if distributed:
         import torch.distributed as dist
         dist.barrier()     # hold all ranks until rank 0's latency phase
@@ -442,6 +624,7 @@ def main():
                 "p99_dhcp_offer_us": None if p99_us is None
                 else round(p99_us, 1),
                 "baseline_mpps": BASELINE_MPPS,
+                "host_io": hostio,
             },
         }
         print(json.dumps(result), flush=True)

@@ -85,6 +85,57 @@ class PcapSink:
         pcap_write(self.path, self.frames)
 
 
+def _concat_aranges(counts):
+    """[3,1,2] -> [0,1,2, 0, 0,1] without a Python loop."""
+    import numpy as np
+    total = int(counts.sum())
+    starts = np.cumsum(counts) - counts
+    return np.arange(total, dtype=np.int64) - np.repeat(starts, counts)
+
+
+def pack_frames(frames, stride: int):
+    """Vectorized List[bytes] -> (data [n,stride] uint8, lens uint16).
+
+    The ingest half of the host edge: one join + fancy-index scatter,
+    no per-frame Python (round-1 VERDICT task 9)."""
+    import numpy as np
+    n = len(frames)
+    lens = np.fromiter((len(f) for f in frames), dtype=np.int64, count=n)
+    clip = np.minimum(lens, stride)
+    joined = np.frombuffer(b"".join(frames), dtype=np.uint8)
+    offs = np.cumsum(lens) - lens               # frame starts in joined
+    data = np.zeros((n, stride), dtype=np.uint8)
+    within = _concat_aranges(clip)
+    dst = np.repeat(np.arange(n, dtype=np.int64) * stride, clip) + within
+    src = np.repeat(offs, clip) + within
+    data.reshape(-1)[dst] = joined[src]
+    return data, clip.astype(np.uint16)
+
+
+def unpack_frames(data, lens) -> List[bytes]:
+    """[n,stride] + lens -> list of bytes (materialize only when a
+    consumer genuinely needs Python objects)."""
+    return [bytes(data[i, :lens[i]]) for i in range(len(lens))]
+
+
+class ArraySink:
+    """Array-native sink: accepts (data [m,stride] uint8, lens) batches
+    without per-frame materialization.  AF_XDP TX rings and pcap writers
+    consume this form directly; .frames materializes lazily for tests."""
+
+    def __init__(self):
+        self.batches: List[Tuple["object", "object"]] = []
+        self.n = 0
+
+    def send_batch_array(self, data, lens):
+        self.batches.append((data.copy(), lens.copy()))
+        self.n += len(lens)
+
+    @property
+    def frames(self) -> List[bytes]:
+        return [f for d, l in self.batches for f in unpack_frames(d, l)]
+
+
 class SyntheticSource:
     def __init__(self, generator: Callable[[int], List[bytes]]):
         self.generator = generator
@@ -203,24 +254,55 @@ class Pump:
         passed: List[bytes] = []
         if is_gpu:
             import numpy as np
-            data, lens = self.launcher.make_batch(frames, self.stride)
+            import torch
+            data_np, lens_np = pack_frames(frames, self.stride)
+            data = torch.from_numpy(data_np).to(self.launcher.device)
+            lens = torch.from_numpy(lens_np.view(np.int16)).to(
+                self.launcher.device)
             verdict, out_len = self.launcher.uplink(
                 data, lens, sort_by_type=self.sort_by_type)
             v = verdict.cpu().numpy()
             ol = out_len.cpu().numpy().view(np.uint16)
             host = data.cpu().numpy()
-            for i in range(len(frames)):
-                if v[i] == abi.TX:
-                    out_frames.append(bytes(host[i][:ol[i]]))
-                    self.stats["tx"] += 1
-                elif v[i] == abi.FWD:
-                    out_frames.append(bytes(host[i][:len(frames[i])]))
-                    self.stats["fwd"] += 1
-                elif v[i] == abi.PASS:
-                    passed.append(frames[i])
-                    self.stats["passed"] += 1
-                else:
-                    self.stats["dropped"] += 1
+            # vectorized verdict partitioning — no per-frame Python
+            tx = v == abi.TX
+            fwd = v == abi.FWD
+            pas = v == abi.PASS
+            self.stats["tx"] += int(tx.sum())
+            self.stats["fwd"] += int(fwd.sum())
+            self.stats["passed"] += int(pas.sum())
+            self.stats["dropped"] += int(
+                len(frames) - tx.sum() - fwd.sum() - pas.sum())
+            out_mask = tx | fwd
+            out_lens = np.where(tx, ol, lens_np).astype(np.uint16)[out_mask]
+            out_data = host[out_mask]
+            # PASS frames go to the slow path (few: cache misses only)
+            passed = [frames[i] for i in np.nonzero(pas)[0]]
+            slow_replies: List[bytes] = []
+            if self.slow_path is not None:
+                for fr in passed:
+                    try:
+                        reply = self.slow_path(fr)
+                    except Exception:
+                        reply = None
+                    if reply:
+                        slow_replies.append(reply)
+                        self.stats["slow_replies"] += 1
+            if self.sink is not None:
+                if hasattr(self.sink, "send_batch_array"):
+                    if len(out_lens):
+                        self.sink.send_batch_array(out_data, out_lens)
+                    if slow_replies:
+                        sd, sl = pack_frames(slow_replies, self.stride)
+                        self.sink.send_batch_array(sd, sl)
+                    return out_data, passed
+                out_frames = unpack_frames(out_data, out_lens)
+                out_frames.extend(slow_replies)
+                self.sink.send_batch(out_frames)
+                return out_frames, passed
+            out_frames = unpack_frames(out_data, out_lens)
+            out_frames.extend(slow_replies)
+            return out_frames, passed
         else:
             # golden-model launcher (CPU mode); route like the fused
             # uplink kernel: DHCP frames that miss go to the slow path,

@@ -3,6 +3,7 @@ fast+slow path loop over the golden launcher (the end-to-end RX->GPU->
 TX/slow-path wiring)."""
 import time
 
+import numpy as np
 import pytest
 
 from bng_amd.dataplane import abi
@@ -193,3 +194,82 @@ class TestAFPacketLoopback:
             _t.sleep(0.01)
         # the DHCP OFFER built by the fast path reached the sink
         assert any(len(f) > 240 for f in sink.frames)
+
+
+class _FakeGpuLauncher:
+    """GPU-shaped launcher (has make_batch/uplink/device) whose uplink is
+    a trivial header check — isolates the Pump's CPU-side ingest +
+    verdict routing for the >=1M frames/s host-edge throughput test."""
+
+    def __init__(self):
+        import torch
+        self.device = torch.device("cpu")
+
+    def make_batch(self, frames, stride):     # presence gates the GPU path
+        raise NotImplementedError
+
+    def uplink(self, data, lens, sort_by_type=True):
+        import torch
+        n = lens.numel()
+        # UDP dst 67 -> TX (fastpath reply); even index -> FWD; rest DROP
+        d = data.numpy()
+        is67 = (d[:, 36] == 0) & (d[:, 37] == 67)
+        v = np.where(is67, abi.TX,
+                     np.where(np.arange(n) % 2 == 0, abi.FWD, abi.DROP))
+        ol = np.where(is67, 300, 0).astype(np.int16)
+        return (torch.from_numpy(v.astype(np.uint8)),
+                torch.from_numpy(ol))
+
+
+class TestPumpThroughput:
+    def test_routing_rate_1m_fps(self):
+        """VERDICT r1 task 9: CPU-side Pump routing >= 1M frames/s
+        (ingest pack + verdict partition + batched array sink)."""
+        import time as _t
+        from bng_amd.dataplane.pktio import ArraySink, Pump
+        launcher = _FakeGpuLauncher()
+        sink = ArraySink()
+        pump = Pump(launcher, SyntheticSource(lambda n: []), sink,
+                    batch=8192, stride=512)
+        frames = [build_ipv4("aa:00:00:00:00:01", "02:00:00:00:00:01",
+                             ip2u32("10.0.0.2"), ip2u32("1.2.3.4"),
+                             proto=17, sport=40000, dport=53,
+                             payload=b"\x00" * 22)] * 8192
+        pump.process(frames)                   # warm
+        n_batches = 25
+        t0 = _t.perf_counter()
+        for _ in range(n_batches):
+            pump.process(frames)
+        dt = _t.perf_counter() - t0
+        fps = n_batches * len(frames) / dt
+        assert fps >= 1_000_000, f"host-edge routing {fps:,.0f} fps < 1M"
+        assert sink.n == (n_batches + 1) * 4096   # FWD half
+
+    def test_array_sink_and_tx_lengths(self):
+        """TX frames leave at out_len, FWD at original length, through
+        the array path with no per-frame materialization."""
+        from bng_amd.dataplane.pktio import ArraySink, Pump
+        launcher = _FakeGpuLauncher()
+        sink = ArraySink()
+        pump = Pump(launcher, SyntheticSource(lambda n: []), sink,
+                    batch=64, stride=512)
+        dhcp = build_dhcp_request("aa:00:00:00:00:02", 1, xid=7)
+        data_pkt = build_ipv4("aa:00:00:00:00:01", "02:00:00:00:00:01",
+                              ip2u32("10.0.0.2"), ip2u32("1.2.3.4"),
+                              proto=17, sport=40000, dport=53,
+                              payload=b"\x00" * 22)
+        out, passed = pump.process([data_pkt, dhcp])
+        assert sink.n == 2
+        lens = np.concatenate([l for _, l in sink.batches])
+        assert sorted(lens.tolist()) == [64, 300]   # FWD orig, TX out_len
+        assert pump.stats["tx"] == 1 and pump.stats["fwd"] == 1
+
+    def test_pack_frames_matches_make_batch_semantics(self):
+        from bng_amd.dataplane.pktio import pack_frames
+        frames = [b"\x01" * 60, b"\x02" * 600, b"", b"\x03" * 512]
+        data, lens = pack_frames(frames, 512)
+        assert lens.tolist() == [60, 512, 0, 512]
+        assert (data[0, :60] == 1).all() and (data[0, 60:] == 0).all()
+        assert (data[1] == 2).all()          # clipped at stride
+        assert (data[2] == 0).all()
+        assert (data[3] == 3).all()
