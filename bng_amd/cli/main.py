@@ -39,6 +39,16 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--server-mac", default="02:00:00:00:00:01")
     g.add_argument("--gpu", default="auto",
                    help="auto|off|cuda:N — dataplane device")
+    g.add_argument("--pktio", default="off",
+                   choices=["off", "afpacket", "afxdp"],
+                   help="NIC edge: attach a packet pump to --interface "
+                        "(afxdp = UMEM rings + XDP redirect, the ref's "
+                        "loader.go:294-315 attach; afpacket = raw-socket "
+                        "fallback, the ref's generic mode)")
+    g.add_argument("--pktio-batch", type=int, default=8192)
+    g.add_argument("--pktio-max-wait", type=float, default=0.0005,
+                   help="batch deadline seconds (latency/throughput "
+                        "trade at the NIC edge)")
     g = run.add_argument_group("dhcp")
     g.add_argument("--pool-network", default="")
     g.add_argument("--pool-gateway", default="")
@@ -467,7 +477,44 @@ class BNG:
         # 13. DHCP serve loop last (main.go:1244)
         self.dhcp_server.start(serve=a.dhcp_listen)
         self._defer(self.dhcp_server.stop)
+
+        # 13b. NIC-edge packet pump (XDP attach analog,
+        # loader.go:294-315: afxdp = driver-ish path, afpacket =
+        # generic fallback)
+        if a.pktio != "off":
+            from ..dataplane.pktio import AFPacketIO, Pump
+            if a.pktio == "afxdp":
+                from ..dataplane.afxdp import XskSocket
+                io = XskSocket(a.interface, mode="auto")
+                self.log.info("AF_XDP %s mode on %s", io.mode, a.interface)
+            else:
+                io = AFPacketIO(a.interface)
+                self.log.info("AF_PACKET raw socket on %s", a.interface)
+            self.pktio = io
+            self.pump = Pump(self.launcher, io, io,
+                             slow_path=self._frame_slow_path,
+                             batch=a.pktio_batch,
+                             max_wait=a.pktio_max_wait).start()
+            self._defer(self.pump.stop)
+            self._defer(io.close)
         return self
+
+    def _frame_slow_path(self, frame: bytes):
+        """PASS-verdict frames (DHCP cache misses) -> slow-path server;
+        returns the reply frame, if any (ref XDP_PASS -> server4)."""
+        from ..dataplane.packets import parse_dhcp_frame
+        from ..dhcp import message as dm
+        try:
+            p = parse_dhcp_frame(frame)
+        except (AssertionError, IndexError, ValueError):
+            return None
+        off = 14 + p.vlan_offset + 20 + 8
+        try:
+            msg = dm.DHCPMessage.decode(frame[off:])
+        except Exception:
+            return None
+        resp = self.dhcp_server.handle(msg)
+        return resp.encode() if resp else None
 
     def _coa_lookup(self, req):
         for lease in self.dhcp_server.leases.values():

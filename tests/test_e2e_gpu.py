@@ -1,0 +1,203 @@
+"""End-to-end NIC-edge test on the GPU box (round-1 VERDICT task 1
+"Done" criterion): a veth pair carries real frames into `bng run
+--gpu`'s pump (AF_XDP rings when the kernel allows, AF_PACKET
+fallback), through the HIP uplink pipeline, and replies return on the
+wire — at >100k packets/s sustained."""
+import os
+import socket
+import threading
+import time
+
+import numpy as np
+import pytest
+
+from bng_amd.dataplane import abi
+from bng_amd.dataplane.packets import (build_dhcp_request, build_ipv4,
+                                       ip2u32, mac_bytes)
+
+torch = pytest.importorskip("torch")
+pytestmark = [
+    pytest.mark.gpu,
+    pytest.mark.skipif(not torch.cuda.is_available(), reason="needs GPU"),
+    pytest.mark.skipif(os.geteuid() != 0, reason="needs root for veth"),
+]
+
+V0, V1 = "bnge0", "bnge1"
+
+
+@pytest.fixture()
+def veth():
+    from bng_amd.dataplane import afxdp
+    try:
+        try:
+            afxdp.link_del(V0)
+        except OSError:
+            pass
+        afxdp.veth_create(V0, V1)
+        afxdp.link_up(V0)
+        afxdp.link_up(V1)
+    except OSError as e:
+        pytest.skip(f"cannot create veth: {e}")
+    yield (V0, V1)
+    try:
+        afxdp.link_del(V0)
+    except OSError:
+        pass
+
+
+def _make_io(ifname):
+    """AF_XDP rings preferred; AF_PACKET raw socket fallback (the same
+    driver->generic ladder as ref loader.go:294-315)."""
+    try:
+        from bng_amd.dataplane.afxdp import XskSocket
+        return XskSocket(ifname, mode="auto", ring_size=4096), "afxdp"
+    except OSError:
+        from bng_amd.dataplane.pktio import AFPacketIO
+        return AFPacketIO(ifname), "afpacket"
+
+
+def test_bng_run_gpu_100kpps(veth):
+    """>100k pps from the wire through the GPU pipeline."""
+    from bng_amd.dataplane.launcher import HipLauncher
+    from bng_amd.dataplane.pktio import Pump
+
+    n_subs = 4096
+    now = int(time.time())
+    launcher = HipLauncher("cuda:0")
+    launcher.set_server_config(mac_bytes("02:00:00:00:00:01"),
+                               ip2u32("10.255.255.1"))
+    launcher.add_pool(1, ip2u32("10.0.0.0"), 8, ip2u32("10.255.255.1"))
+    launcher.set_antispoof_config(default_mode=abi.AS_DISABLED)
+    # subscriber + NAT/QoS context tables (bulk, like bench.build_tables)
+    idx = np.arange(n_subs, dtype=np.uint64)
+    macs = np.uint64(0xAA0000000000) + idx
+    ips = (np.uint64(ip2u32("10.0.0.0") + 2) + idx).astype(np.uint64)
+    sub = np.zeros(n_subs, dtype=[("key", "<u8"), ("pool", "<u4"),
+                                  ("ip", "<u4"), ("lease", "<u8"),
+                                  ("vlan", "<u2"), ("cc", "u1"),
+                                  ("fl", "u1"), ("pad", "<u4")])
+    sub["key"] = macs
+    sub["pool"] = 1
+    sub["ip"] = ips.astype(np.uint32)
+    sub["lease"] = now + 86400
+    rc = torch.zeros(n_subs, dtype=torch.int32, device="cuda:0")
+    launcher.ext.sub_upsert(
+        launcher.subs,
+        torch.from_numpy(sub.view(np.uint8)).cuda().flatten(), rc)
+    assert int((rc != 0).sum().item()) == 0
+    pub = ip2u32("203.0.113.0")
+    ctx = np.zeros(n_subs, dtype=[("key_ip", "<u4"), ("pub", "<u4"),
+                                  ("ps", "<u2"), ("pe", "<u2"),
+                                  ("qv", "u1"), ("nv", "u1"),
+                                  ("prio", "u1"), ("fl", "u1"),
+                                  ("rate", "<u8"), ("tokens", "<i8"),
+                                  ("last", "<u8"), ("burst", "<u4"),
+                                  ("np", "<u4"), ("sid", "<u4"),
+                                  ("sa", "<u4"), ("st", "<u4"),
+                                  ("pad", "<u4")])
+    ctx["key_ip"] = ips.astype(np.uint32)
+    ctx["pub"] = pub + (idx % 250).astype(np.uint32)
+    starts = (1024 + (idx % 63) * 1024).astype(np.uint16)
+    ctx["ps"] = starts
+    ctx["pe"] = starts + 1023
+    ctx["np"] = starts
+    ctx["nv"] = 1
+    ctx["qv"] = 1
+    ctx["rate"] = 10**9
+    ctx["tokens"] = 4 << 20
+    ctx["burst"] = 4 << 20
+    ctx["last"] = now * 10**9
+    rc = torch.zeros(n_subs, dtype=torch.int32, device="cuda:0")
+    launcher.ext.subctx_upsert(
+        launcher.subctx,
+        torch.from_numpy(ctx.view(np.uint8)).cuda().flatten(),
+        abi.CTX_SET_NAT | abi.CTX_SET_QOS, rc)
+    assert int((rc != 0).sum().item()) == 0
+
+    io, io_kind = _make_io(V1)
+    pump = Pump(launcher, io, io, batch=8192, max_wait=0.002)
+
+    # blast pre-built 64B data frames from the peer
+    tx = socket.socket(socket.AF_PACKET, socket.SOCK_RAW, socket.htons(3))
+    tx.bind((V0, 0))
+    frames = [build_ipv4("aa:00:00:00:%02x:%02x" % (i >> 8, i & 0xFF),
+                         "02:00:00:00:00:01",
+                         int(ips[i]), ip2u32("93.184.216.34"),
+                         proto=17, sport=40000 + (i % 64), dport=53,
+                         payload=b"\x00" * 22)
+              for i in range(1024)]
+    stop = threading.Event()
+    sent = [0]
+
+    def blaster():
+        while not stop.is_set():
+            for f in frames:
+                try:
+                    tx.send(f)
+                except OSError:
+                    pass
+            sent[0] += len(frames)
+
+    th = threading.Thread(target=blaster, daemon=True)
+    th.start()
+    # warm one batch (JIT caches, first-touch)
+    t_end = time.monotonic() + 0.5
+    while time.monotonic() < t_end:
+        pump.pump_once()
+    pump.stats["rx"] = 0
+    t0 = time.monotonic()
+    t_end = t0 + 3.0
+    while time.monotonic() < t_end:
+        pump.pump_once()
+    elapsed = time.monotonic() - t0
+    stop.set()
+    th.join(timeout=2)
+    rx_pps = pump.stats["rx"] / elapsed
+    print(f"[e2e] io={io_kind} rx={pump.stats['rx']} "
+          f"({rx_pps:,.0f} pps) tx={pump.stats['tx']} "
+          f"fwd={pump.stats['fwd']} drop={pump.stats['dropped']} "
+          f"blaster sent~{sent[0]}")
+    tx.close()
+    if hasattr(io, "close"):
+        io.close()
+    assert pump.stats["fwd"] > 0, "no frames traversed the GPU pipeline"
+    assert rx_pps > 100_000, (
+        f"{rx_pps:,.0f} pps < 100k through {io_kind}")
+
+
+def test_dhcp_offer_on_wire_gpu(veth):
+    """A DHCP DISCOVER on the wire returns a GPU-built OFFER frame."""
+    from bng_amd.dataplane.launcher import HipLauncher
+    from bng_amd.dataplane.pktio import Pump
+
+    now = int(time.time())
+    launcher = HipLauncher("cuda:0")
+    launcher.set_server_config(mac_bytes("02:00:00:00:00:01"),
+                               ip2u32("10.255.255.1"))
+    launcher.add_pool(1, ip2u32("10.0.1.0"), 24, ip2u32("10.0.1.1"))
+    launcher.set_antispoof_config(default_mode=abi.AS_DISABLED)
+    mac = "aa:bb:cc:00:00:07"
+    launcher.add_subscriber(mac_bytes(mac), 1, ip2u32("10.0.1.77"),
+                            now + 3600)
+    io, io_kind = _make_io(V1)
+    pump = Pump(launcher, io, io, batch=64, max_wait=0.05)
+    tx = socket.socket(socket.AF_PACKET, socket.SOCK_RAW, socket.htons(3))
+    tx.bind((V0, 0))
+    tx.send(build_dhcp_request(mac, 1, xid=0x77))
+    got = False
+    end = time.monotonic() + 3.0
+    import struct as st
+    while not got and time.monotonic() < end:
+        pump.pump_once()
+        try:
+            tx.settimeout(0.2)
+            f = tx.recv(2048)
+        except socket.timeout:
+            continue
+        if len(f) > 240 and f[6:12] == mac_bytes("02:00:00:00:00:01"):
+            if st.unpack(">I", f[58:62])[0] == ip2u32("10.0.1.77"):
+                got = True
+    tx.close()
+    if hasattr(io, "close"):
+        io.close()
+    assert got, f"no OFFER on the wire via {io_kind}"
