@@ -157,6 +157,43 @@ BNG_DEV const bng_binding_entry* binding_lookup(const bng_binding_entry* t,
  * rules out an existing entry — without this, long-uptime churn fills
  * the chain with tombstones (the reference gets reuse for free from
  * BPF LRU maps). */
+/* Post-claim duplicate check (round-1 advisor finding): if a chain slot
+ * is tombstoned concurrently (in-batch port-exhaustion release), one
+ * lane can reclaim that earlier tombstone while another lane of the
+ * SAME flow claims a later slot — leaking a duplicate session + NAT
+ * port until sweep.  After winning a claim CAS, rescan the chain for a
+ * concurrent same-sig insert and keep only the EARLIEST-position claim:
+ * the later claimant re-tombstones its slot and adopts the earlier one.
+ * The positional tie-break is what makes this safe for lockstep lanes —
+ * a symmetric release-on-any-hit rule would make two same-wave lanes
+ * release each other every retry.  Residual window: a later claimant
+ * whose rescan completed before our CAS landed keeps a shadowed
+ * duplicate; that reverts to the pre-fix behavior (sweep reclaims it,
+ * EIM collision check keeps the port safe) in a window now measured in
+ * the tens of nanoseconds instead of the whole create race. */
+template <typename E>
+BNG_DEV E* claim_dedup(E* t, uint32_t mask, uint64_t sig, uint32_t slot,
+                       int my_pos, bool* claimed, bool* found) {
+  for (int i = 0; i < BNG_MAX_PROBE; ++i) {
+    if (i == my_pos) continue;
+    E* e = &t[(slot + i) & mask];
+    uint64_t k = __hip_atomic_load(&e->sig, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+    if (k == sig && i < my_pos) {
+      /* earlier claim wins: release ours, adopt theirs (caller does
+       * bng_wait_ready before trusting fields) */
+      __hip_atomic_store(&t[(slot + my_pos) & mask].sig,
+                         BNG_KEY_TOMBSTONE, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+      *claimed = false; *found = true;
+      return e;
+    }
+    if (k == BNG_KEY_EMPTY) break;
+  }
+  *claimed = true;
+  return &t[(slot + my_pos) & mask];
+}
+
 template <typename E>
 BNG_DEV E* sig_find_or_claim(E* t, uint32_t mask, uint64_t sig,
                              bool* claimed, bool* found) {
@@ -173,31 +210,34 @@ BNG_DEV E* sig_find_or_claim(E* t, uint32_t mask, uint64_t sig,
   }
   /* pass 2 — claim scan (session-create path only) */
   for (int attempt = 0; attempt < 4; ++attempt) {
-    int first_tomb = -1;
+    int first_tomb = -1, first_tomb_pos = -1;
     for (int i = 0; i < BNG_MAX_PROBE; ++i) {
       E* e = &t[(slot + i) & mask];
       uint64_t k = __hip_atomic_load(&e->sig, __ATOMIC_RELAXED,
                                      __HIP_MEMORY_SCOPE_AGENT);
       if (k == sig) { *found = true; return e; }
       if (k == BNG_KEY_TOMBSTONE) {
-        if (first_tomb < 0) first_tomb = (int)((slot + i) & mask);
+        if (first_tomb < 0) {
+          first_tomb = (int)((slot + i) & mask);
+          first_tomb_pos = i;
+        }
         continue;
       }
       if (k == BNG_KEY_EMPTY) {
         E* d = first_tomb >= 0 ? &t[first_tomb] : e;
+        int d_pos = first_tomb >= 0 ? first_tomb_pos : i;
         uint64_t want = first_tomb >= 0 ? BNG_KEY_TOMBSTONE
                                         : BNG_KEY_EMPTY;
         if (__hip_atomic_compare_exchange_strong(
                 &d->sig, &want, sig, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
                 __HIP_MEMORY_SCOPE_AGENT)) {
-          *claimed = true;
-          return d;
+          return claim_dedup(t, mask, sig, slot, d_pos, claimed, found);
         }
         if (want == sig) { *found = true; return d; }
         /* lost the slot to a DIFFERENT key: keep probing.  If it was
          * the remembered tombstone, forget it and retry this empty
          * slot; if it was the empty slot itself, scan past it. */
-        if (first_tomb >= 0) { first_tomb = -1; --i; }
+        if (first_tomb >= 0) { first_tomb = -1; first_tomb_pos = -1; --i; }
         continue;
       }
     }
@@ -209,8 +249,8 @@ BNG_DEV E* sig_find_or_claim(E* t, uint32_t mask, uint64_t sig,
     if (__hip_atomic_compare_exchange_strong(
             &d->sig, &want, sig, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
             __HIP_MEMORY_SCOPE_AGENT)) {
-      *claimed = true;
-      return d;
+      return claim_dedup(t, mask, sig, slot, first_tomb_pos, claimed,
+                         found);
     }
     if (want == sig) { *found = true; return d; }
     /* raced: rescan */
