@@ -228,6 +228,15 @@ class RingHeader(C.Structure):
                 ("capacity", C.c_uint32), ("_pad", C.c_uint32)]
 
 
+class SvcCtrl(C.Structure):
+    """Pinned-host doorbell of the persistent DHCP service kernel."""
+    _fields_ = [("head", C.c_uint32), ("tail", C.c_uint32),
+                ("run", C.c_uint32), ("n_pkts", C.c_uint32),
+                ("now_sec", C.c_uint64), ("stride", C.c_uint32),
+                ("idle_exit_k", C.c_uint32), ("served", C.c_uint64),
+                ("batches", C.c_uint64), ("_pad", C.c_uint8 * 16)]
+
+
 EXPECTED_SIZES = {
     "bng_sub_entry": (SubEntry, 32),
     "bng_ip_pool": (IpPool, 28),
@@ -244,6 +253,7 @@ EXPECTED_SIZES = {
     "bng_antispoof_config": (AntispoofConfig, 8 + 256 * 8),
     "bng_spoof_event": (SpoofEvent, 56),
     "bng_ring_header": (RingHeader, 16),
+    "bng_svc_ctrl": (SvcCtrl, 64),
 }
 
 

@@ -339,6 +339,28 @@ typedef struct bng_ring_header {
   uint32_t _pad;
 } bng_ring_header;  /* 16 B */
 
+/* -------------------------------------------- persistent-service control */
+/* SPSC doorbell between the host and the device-resident DHCP service
+ * kernel (the persistent-kernel latency path: the service waves own
+ * their CU permanently, so a saturating data flood cannot starve the
+ * DHCP slice the way it starves a freshly launched kernel).  Lives in
+ * PINNED HOST memory: host writes head/run/now, device writes tail.
+ * No reference analog — kernel XDP gets this for free by running in
+ * the NIC IRQ path; this is the MI355X equivalent. */
+typedef struct bng_svc_ctrl {
+  uint32_t head;        /* host bumps after writing a request batch   */
+  uint32_t tail;        /* device bumps after replies are visible     */
+  uint32_t run;         /* host clears to stop the kernel             */
+  uint32_t n_pkts;      /* packets in the current batch               */
+  uint64_t now_sec;     /* host-maintained batch clock                */
+  uint32_t stride;
+  uint32_t idle_exit_k; /* idle polls (units of 1024) before self-exit
+                           — the box-safety bound                     */
+  uint64_t served;      /* device: total packets served               */
+  uint64_t batches;     /* device: total batches served               */
+  uint8_t  _pad[16];
+} bng_svc_ctrl;  /* 64 B */
+
 #ifdef __cplusplus
 }
 #endif

@@ -50,6 +50,9 @@ void bng_launch_nat_sweep(void*, uint32_t, void*, uint32_t, void*, uint32_t,
                           void*, hipStream_t);
 void bng_launch_shard_owner(const void*, const void*, void*, int, int, int,
                             hipStream_t);
+void bng_launch_dhcp_service(void*, void*, const void*, void*, void*, void*,
+                             int, const void*, uint32_t, const void*,
+                             uint32_t, const void*, void*, hipStream_t);
 }
 
 namespace {
@@ -279,6 +282,50 @@ void shard_owner(torch::Tensor data, torch::Tensor in_len,
                          (int)data.size(1), (int)n_shards, cur_stream());
 }
 
+void check_pinned(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(!t.is_cuda() && t.is_pinned(), name,
+              " must be a pinned host tensor");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+/* Persistent DHCP service: launch the device-resident kernel on its own
+ * non-blocking stream (it never returns until stopped, so it must not
+ * share a stream with ordinary work). */
+hipStream_t g_svc_stream = nullptr;
+
+void dhcp_service_start(torch::Tensor ctrl, torch::Tensor req,
+                        torch::Tensor in_len, torch::Tensor out_len,
+                        torch::Tensor verdict, torch::Tensor scratch,
+                        torch::Tensor subs, torch::Tensor pools,
+                        torch::Tensor cfg, torch::Tensor stats) {
+  check_pinned(ctrl, "ctrl"); check_pinned(req, "req");
+  check_pinned(in_len, "in_len"); check_pinned(out_len, "out_len");
+  check_pinned(verdict, "verdict");
+  check_dev(scratch, "scratch"); check_dev(subs, "subs");
+  check_dev(pools, "pools"); check_dev(cfg, "cfg");
+  check_dev(stats, "stats");
+  TORCH_CHECK(ctrl.numel() * ctrl.element_size() ==
+              (long)sizeof(bng_svc_ctrl), "ctrl must be 64 bytes");
+  int n_slots = (int)req.size(0);
+  TORCH_CHECK(in_len.numel() == n_slots && out_len.numel() == n_slots &&
+              verdict.numel() == n_slots, "slot count mismatch");
+  TORCH_CHECK(scratch.numel() == req.numel(), "scratch size mismatch");
+  if (!g_svc_stream)
+    (void)hipStreamCreateWithFlags(&g_svc_stream, hipStreamNonBlocking);
+  bng_launch_dhcp_service(
+      ctrl.data_ptr(), req.data_ptr(), in_len.data_ptr(),
+      out_len.data_ptr(), verdict.data_ptr(), scratch.data_ptr(), n_slots,
+      subs.data_ptr(), table_mask(subs, sizeof(bng_sub_entry), "subs"),
+      pools.data_ptr(),
+      (uint32_t)(pools.numel() * pools.element_size() /
+                 sizeof(bng_ip_pool)),
+      cfg.data_ptr(), stats.data_ptr(), g_svc_stream);
+}
+
+void dhcp_service_join() {
+  if (g_svc_stream) (void)hipStreamSynchronize(g_svc_stream);
+}
+
 py::dict layout_report() {
   py::dict d;
 #define SZ(T) d[#T] = sizeof(T)
@@ -287,6 +334,7 @@ py::dict layout_report() {
   SZ(bng_eim_entry); SZ(bng_subctx); SZ(bng_nat_config);
   SZ(bng_nat_log_entry); SZ(bng_qos_bucket); SZ(bng_binding_entry);
   SZ(bng_antispoof_config); SZ(bng_spoof_event); SZ(bng_ring_header);
+  SZ(bng_svc_ctrl);
 #undef SZ
   py::dict off;
   off["sub_entry.lease_expiry"] = offsetof(bng_sub_entry, lease_expiry);
@@ -342,5 +390,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("binding_delete", &binding_delete);
   m.def("nat_sweep", &nat_sweep);
   m.def("shard_owner", &shard_owner);
+  m.def("dhcp_service_start", &dhcp_service_start);
+  m.def("dhcp_service_join", &dhcp_service_join);
   m.def("layout_report", &layout_report);
 }

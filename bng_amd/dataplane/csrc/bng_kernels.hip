@@ -647,6 +647,99 @@ __global__ void dhcp_fastpath_kernel(
   }
 }
 
+/* ------------------------------------------- persistent DHCP service.
+ * One workgroup (256 threads, 1 CU of 256) stays resident and polls a
+ * pinned-host doorbell.  Because its waves already occupy their CU, a
+ * saturating 1M-packet data flood cannot starve it the way it starves
+ * a freshly launched kernel (measured r1: flood p99 453 us vs <70 us
+ * quiesced; a priority stream did not help because the launch itself
+ * must wait for CU space).  Request path: host writes packets into the
+ * pinned ring and bumps head; the kernel stages the batch to an HBM
+ * scratch with wide coalesced PCIe reads, runs dhcp_process, copies
+ * replies back, and release-stores tail.  Self-exits after
+ * idle_exit_k*1024 empty polls so a crashed host can never wedge the
+ * box. */
+__global__ void __launch_bounds__(256, 1) dhcp_service_kernel(
+    bng_svc_ctrl* __restrict__ ctrl,
+    uint8_t* __restrict__ req,            /* pinned [n_slots, stride]  */
+    const uint16_t* __restrict__ in_len,  /* pinned [n_slots]          */
+    uint16_t* __restrict__ out_len,       /* pinned [n_slots]          */
+    uint8_t* __restrict__ verdict,        /* pinned [n_slots]          */
+    uint8_t* __restrict__ scratch,        /* device [n_slots, stride]  */
+    int n_slots,
+    const bng_sub_entry* __restrict__ subs, uint32_t sub_mask,
+    const bng_ip_pool* __restrict__ pools, uint32_t n_pools,
+    const bng_server_config* __restrict__ cfg,
+    unsigned long long* __restrict__ stats) {
+  __shared__ uint32_t s_head, s_run, s_n;
+  __shared__ uint64_t s_now;
+  const int tid = threadIdx.x;
+  uint32_t done = __hip_atomic_load(&ctrl->tail, __ATOMIC_RELAXED,
+                                    __HIP_MEMORY_SCOPE_SYSTEM);
+  const int stride = (int)ctrl->stride;
+  uint64_t idle = 0;
+  const uint64_t idle_max = (uint64_t)ctrl->idle_exit_k * 1024u;
+  uint64_t served = 0, batches = 0;
+  for (;;) {
+    if (tid == 0) {
+      s_head = __hip_atomic_load(&ctrl->head, __ATOMIC_ACQUIRE,
+                                 __HIP_MEMORY_SCOPE_SYSTEM);
+      s_run = __hip_atomic_load(&ctrl->run, __ATOMIC_RELAXED,
+                                __HIP_MEMORY_SCOPE_SYSTEM);
+      s_n = ctrl->n_pkts;
+      s_now = ctrl->now_sec;
+    }
+    __syncthreads();
+    bool work = (s_head != done);
+    if (!s_run || (!work && idle >= idle_max)) break;
+    if (!work) {
+      ++idle;
+      __builtin_amdgcn_s_sleep(8);
+      __syncthreads();           /* re-converge before the next poll */
+      continue;
+    }
+    idle = 0;
+    int n = (int)s_n;
+    if (n > n_slots) n = n_slots;
+    /* stage request slots to HBM: wide coalesced PCIe reads */
+    int words = n * stride / 16;
+    const uint4* src = (const uint4*)req;
+    uint4* dst = (uint4*)scratch;
+    for (int i = tid; i < words; i += blockDim.x) dst[i] = src[i];
+    __syncthreads();
+    dhcp_tables T{subs, sub_mask, pools, n_pools, cfg, stats, s_now};
+    for (int pid = tid; pid < n; pid += blockDim.x) {
+      dhcp_flags F; F.clear();
+      uint16_t ol = in_len[pid];
+      int v = dhcp_process(scratch + (size_t)pid * stride, in_len[pid],
+                           stride, T, F, &ol);
+      verdict[pid] = (uint8_t)v;
+      out_len[pid] = ol;
+      dhcp_commit_stats(F, stats);
+    }
+    __syncthreads();
+    /* replies back to the pinned ring (posted PCIe writes) */
+    for (int i = tid; i < words; i += blockDim.x)
+      ((uint4*)req)[i] = dst[i];
+    __syncthreads();
+    ++done; ++batches; served += (uint64_t)n;
+    if (tid == 0) {
+      ctrl->served = served;
+      ctrl->batches = batches;
+      __hip_atomic_store(&ctrl->tail, done, __ATOMIC_RELEASE,
+                         __HIP_MEMORY_SCOPE_SYSTEM);
+    }
+    __syncthreads();
+  }
+  if (tid == 0) {
+    ctrl->served = served;
+    ctrl->batches = batches;
+    /* mark exit so the host can distinguish idle-reap from running */
+    __hip_atomic_store(&ctrl->run, 0u, __ATOMIC_RELEASE,
+                       __HIP_MEMORY_SCOPE_SYSTEM);
+  }
+}
+
 /* ========================================================== NAT44  K2 */
 
 struct nat_tables {
@@ -1868,6 +1961,20 @@ void bng_launch_dhcp(void* data, const void* in_len, void* out_len,
       (uint8_t*)verdict, n, stride, (const bng_sub_entry*)subs, sub_mask,
       (const bng_ip_pool*)pools, n_pools, (const bng_server_config*)cfg,
       (unsigned long long*)stats, now_sec, (const uint64_t*)now_ptr);
+}
+
+void bng_launch_dhcp_service(void* ctrl, void* req, const void* in_len,
+                             void* out_len, void* verdict, void* scratch,
+                             int n_slots, const void* subs,
+                             uint32_t sub_mask, const void* pools,
+                             uint32_t n_pools, const void* cfg, void* stats,
+                             hipStream_t s) {
+  hipLaunchKernelGGL(dhcp_service_kernel, dim3(1), dim3(256), 0, s,
+      (bng_svc_ctrl*)ctrl, (uint8_t*)req, (const uint16_t*)in_len,
+      (uint16_t*)out_len, (uint8_t*)verdict, (uint8_t*)scratch, n_slots,
+      (const bng_sub_entry*)subs, sub_mask, (const bng_ip_pool*)pools,
+      n_pools, (const bng_server_config*)cfg,
+      (unsigned long long*)stats);
 }
 
 void bng_launch_nat44(void* data, const void* in_len, void* verdict, int n,

@@ -503,6 +503,90 @@ class HipLauncher:
         return owner
 
 
+class DhcpService:
+    """Persistent-kernel DHCP fast path: one resident workgroup polls a
+    pinned-host doorbell and serves request batches with no kernel
+    launch in the loop.  Because its waves permanently own their CU, a
+    saturating data flood cannot starve it (round-1 measured flood p99
+    453 us for the launched path; the reference's in-IRQ XDP has the
+    same always-resident property, dhcp_fastpath.c:619).
+
+    Latency path only — the batched dhcp_fastpath/uplink kernels remain
+    the throughput path."""
+
+    def __init__(self, launcher: "HipLauncher", n_slots: int = 2048,
+                 stride: int = 512, idle_exit_k: int = 4_000_000):
+        import numpy as np
+        torch = launcher.torch
+        self.l = launcher
+        self.n_slots, self.stride = n_slots, stride
+        self.ctrl_t = torch.zeros(64, dtype=torch.uint8, pin_memory=True)
+        self.req = torch.zeros((n_slots, stride), dtype=torch.uint8,
+                               pin_memory=True)
+        self.in_len = torch.zeros(n_slots, dtype=torch.int16,
+                                  pin_memory=True)
+        self.out_len = torch.zeros(n_slots, dtype=torch.int16,
+                                   pin_memory=True)
+        self.verdict = torch.zeros(n_slots, dtype=torch.uint8,
+                                   pin_memory=True)
+        self.scratch = torch.empty((n_slots, stride), dtype=torch.uint8,
+                                   device=launcher.device)
+        self.c = abi.SvcCtrl.from_address(self.ctrl_t.data_ptr())
+        self.c.stride = stride
+        # idle self-exit: ~0.25us/poll -> default reaps after ~17 min
+        self.c.idle_exit_k = idle_exit_k
+        self.c.run = 1
+        launcher.ext.dhcp_service_start(
+            self.ctrl_t, self.req, self.in_len, self.out_len,
+            self.verdict, self.scratch, launcher.subs, launcher.pools,
+            launcher.server_cfg, launcher.dhcp_stats)
+        self._np = np
+
+    @property
+    def running(self) -> bool:
+        return bool(self.c.run)
+
+    def serve(self, data_np, lens_np, now_sec: int, timeout: float = 1.0):
+        """One request batch through the resident kernel; returns
+        (verdict, out_len, reply_bytes) numpy views — copy before the
+        next serve() if they must survive."""
+        np = self._np
+        n = len(lens_np)
+        assert n <= self.n_slots
+        rq = self.req.numpy()
+        rq[:n, :data_np.shape[1]] = data_np
+        self.in_len.numpy()[:n] = lens_np.view(np.int16)
+        self.c.n_pkts = n
+        self.c.now_sec = now_sec
+        target = (self.c.head + 1) & 0xFFFFFFFF
+        self.c.head = target             # doorbell (x86 TSO publishes)
+        deadline = time.perf_counter() + timeout
+        spins = 0
+        while self.c.tail != target:
+            spins += 1
+            if (spins & 0x3FFF) == 0:
+                if not self.c.run:
+                    raise RuntimeError("dhcp service kernel exited")
+                if time.perf_counter() > deadline:
+                    raise TimeoutError("dhcp service timeout")
+        return (self.verdict.numpy()[:n], self.out_len.numpy()[:n],
+                rq[:n])
+
+    def stats(self) -> Dict[str, int]:
+        return {"served": int(self.c.served),
+                "batches": int(self.c.batches)}
+
+    def stop(self):
+        self.c.run = 0
+        self.l.ext.dhcp_service_join()
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        self.stop()
+
+
 class GoldenLauncher:
     """CPU launcher over the golden model: same API, for CPU tests/demo."""
 
