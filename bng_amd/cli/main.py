@@ -259,8 +259,14 @@ class BNG:
                                listen_port=a.ha_listen_port,
                                partner_url=a.ha_partner_url).start()
             self._defer(self.ha.stop)
-            from ..ha import session_glue
-            session_glue.attach(self.dhcp_server, self.ha)
+            from ..ha import nat_glue, session_glue
+            session_glue.attach(self.dhcp_server, self.ha,
+                                nat_mgr=lambda: getattr(self, "nat", None))
+            # NAT flow replication rides the same syncer (task: a
+            # promoted standby keeps established NAT sessions)
+            self._nat_ha = nat_glue.NatHaGlue(self.launcher,
+                                              self.ha).start()
+            self._defer(self._nat_ha.stop)
             if a.ha_partner_url:
                 self.ha_monitor = HealthMonitor(a.ha_partner_url).start()
                 self._defer(self.ha_monitor.stop)
@@ -271,6 +277,7 @@ class BNG:
                             self.dhcp_server, self.ha,
                             getattr(self, "qos", None),
                             getattr(self, "nat", None))
+                        nat_glue.promote_nat(self.launcher, self.ha)
                     else:
                         self.ha.demote()
                 self.ha_failover = FailoverController(

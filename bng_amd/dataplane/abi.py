@@ -228,6 +228,26 @@ class RingHeader(C.Structure):
                 ("capacity", C.c_uint32), ("_pad", C.c_uint32)]
 
 
+class SessExport(C.Structure):
+    """Compact NAT-session record for HA snapshot/delta sync."""
+    _fields_ = [("src_ip", C.c_uint32), ("dst_ip", C.c_uint32),
+                ("src_port", C.c_uint16), ("dst_port", C.c_uint16),
+                ("protocol", C.c_uint8), ("state", C.c_uint8),
+                ("is_hairpin", C.c_uint8), ("flags", C.c_uint8),
+                ("nat_ip", C.c_uint32), ("nat_port", C.c_uint16),
+                ("eim_port", C.c_uint16), ("created", C.c_uint64),
+                ("last_seen", C.c_uint64), ("_pad", C.c_uint64)]
+
+
+SESS_EXPORT_DTYPE = [("src_ip", "<u4"), ("dst_ip", "<u4"),
+                     ("src_port", "<u2"), ("dst_port", "<u2"),
+                     ("protocol", "u1"), ("state", "u1"),
+                     ("is_hairpin", "u1"), ("flags", "u1"),
+                     ("nat_ip", "<u4"), ("nat_port", "<u2"),
+                     ("eim_port", "<u2"), ("created", "<u8"),
+                     ("last_seen", "<u8"), ("_pad", "<u8")]
+
+
 class SvcCtrl(C.Structure):
     """Pinned-host doorbell of the persistent DHCP service kernel."""
     _fields_ = [("head", C.c_uint32), ("tail", C.c_uint32),
@@ -254,6 +274,7 @@ EXPECTED_SIZES = {
     "bng_spoof_event": (SpoofEvent, 56),
     "bng_ring_header": (RingHeader, 16),
     "bng_svc_ctrl": (SvcCtrl, 64),
+    "bng_sess_export": (SessExport, 48),
 }
 
 

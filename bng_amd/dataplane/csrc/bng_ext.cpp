@@ -36,6 +36,8 @@ void bng_launch_pkt_class(const void*, const void*, void*, int, int,
 void bng_launch_sub_upsert(void*, uint32_t, const void*, int, void*,
                            hipStream_t);
 void bng_launch_sub_delete(void*, uint32_t, const void*, int, hipStream_t);
+void bng_launch_sess_import(void*, uint32_t, void*, uint32_t, void*,
+                            uint32_t, const void*, int, void*, hipStream_t);
 void bng_launch_subctx_upsert(void*, uint32_t, const void*, int, uint32_t,
                               void*, hipStream_t);
 void bng_launch_qos_upsert(void*, uint32_t, const void*, int, void*,
@@ -218,6 +220,20 @@ void sub_upsert(torch::Tensor table, torch::Tensor batch, torch::Tensor rc) {
                         table_mask(table, sizeof(bng_sub_entry), "subs"),
                         batch.data_ptr(), n, rc.data_ptr(), cur_stream());
 }
+void sess_import(torch::Tensor sessions, torch::Tensor reverse,
+                 torch::Tensor eim, torch::Tensor batch, torch::Tensor rc) {
+  check_dev(sessions, "sessions"); check_dev(reverse, "reverse");
+  check_dev(eim, "eim"); check_dev(batch, "batch");
+  int n = (int)(batch.numel() * batch.element_size() /
+                sizeof(bng_sess_export));
+  bng_launch_sess_import(
+      sessions.data_ptr(),
+      table_mask(sessions, sizeof(bng_nat_session), "sessions"),
+      reverse.data_ptr(),
+      table_mask(reverse, sizeof(bng_nat_reverse), "reverse"),
+      eim.data_ptr(), table_mask(eim, sizeof(bng_eim_entry), "eim"),
+      batch.data_ptr(), n, rc.data_ptr(), cur_stream());
+}
 void sub_delete(torch::Tensor table, torch::Tensor keys) {
   bng_launch_sub_delete(table.data_ptr(),
                         table_mask(table, sizeof(bng_sub_entry), "subs"),
@@ -334,7 +350,7 @@ py::dict layout_report() {
   SZ(bng_eim_entry); SZ(bng_subctx); SZ(bng_nat_config);
   SZ(bng_nat_log_entry); SZ(bng_qos_bucket); SZ(bng_binding_entry);
   SZ(bng_antispoof_config); SZ(bng_spoof_event); SZ(bng_ring_header);
-  SZ(bng_svc_ctrl);
+  SZ(bng_svc_ctrl); SZ(bng_sess_export);
 #undef SZ
   py::dict off;
   off["sub_entry.lease_expiry"] = offsetof(bng_sub_entry, lease_expiry);
@@ -384,6 +400,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pkt_class", &pkt_class);
   m.def("sub_upsert", &sub_upsert);
   m.def("sub_delete", &sub_delete);
+  m.def("sess_import", &sess_import);
   m.def("subctx_upsert", &subctx_upsert);
   m.def("qos_upsert", &qos_upsert);
   m.def("binding_upsert", &binding_upsert);

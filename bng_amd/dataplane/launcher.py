@@ -482,6 +482,56 @@ class HipLauncher:
         self.ext.sub_upsert(self.subs, batch.flatten(), rc)
         return n - int((rc != 0).sum().item())
 
+    def export_nat_sessions(self, since_ns: int = 0):
+        """Download live NAT sessions for HA sync as a structured numpy
+        array (SESS_EXPORT_DTYPE records).  since_ns>0 = delta export
+        (sessions seen since then); the EIM flag is set when EIM mode is
+        on so promotion restores the mapping too.  Ref ha/sync.go
+        session replication; round-1 VERDICT task 3."""
+        import numpy as np
+        blob = self.sessions.cpu().numpy()
+        arr = blob.view([("sig", "<u8"), ("src_ip", "<u4"),
+                         ("dst_ip", "<u4"), ("src_port", "<u2"),
+                         ("dst_port", "<u2"), ("protocol", "u1"),
+                         ("_kp", "3u1"), ("nat_ip", "<u4"),
+                         ("nat_port", "<u2"), ("orig_port", "<u2"),
+                         ("orig_ip", "<u4"), ("state", "u1"),
+                         ("is_hairpin", "u1"), ("ready", "u1"),
+                         ("_p", "u1"), ("last_seen", "<u8"),
+                         ("created", "<u8"), ("_acct", "9u8")])
+        live = arr[(arr["sig"] != 0) &
+                   (arr["sig"] != 0xFFFFFFFFFFFFFFFF) &
+                   (arr["ready"] != 0) &
+                   (arr["last_seen"] >= np.uint64(since_ns))]
+        out = np.zeros(len(live), dtype=abi.SESS_EXPORT_DTYPE)
+        for f in ("src_ip", "dst_ip", "src_port", "dst_port", "protocol",
+                  "state", "is_hairpin", "nat_ip", "nat_port", "created",
+                  "last_seen"):
+            out[f] = live[f]
+        # EIM restore flag: the create path stores the SAME host-order
+        # allocated port in session.nat_port and eim.external_port
+        eim_on = bool(abi.NatConfig.from_buffer_copy(
+            self.nat_cfg.cpu().numpy().tobytes()).flags & abi.NAT_FLAG_EIM)
+        if eim_on:
+            out["flags"] = 1
+            out["eim_port"] = live["nat_port"]
+        return out
+
+    def import_nat_sessions(self, records) -> int:
+        """Bulk-restore exported NAT sessions (+ reverse + EIM) on a
+        standby at promotion (sess_import_kernel)."""
+        import numpy as np
+        records = np.asarray(records, dtype=abi.SESS_EXPORT_DTYPE)
+        if len(records) == 0:
+            return 0
+        batch = self.torch.from_numpy(records.view(np.uint8)) \
+            .to(self.device)
+        rc = self.torch.zeros(len(records), dtype=self.torch.int32,
+                              device=self.device)
+        self.ext.sess_import(self.sessions, self.reverse, self.eim,
+                             batch.flatten(), rc)
+        return len(records) - int((rc != 0).sum().item())
+
     # ------------------------------------------- hipGraph steady state
     def capture_uplink(self, n: int, stride: int = 512,
                        sort_by_type: bool = True):
@@ -763,6 +813,45 @@ class GoldenLauncher:
                 e.get("lease_expiry", 0), e.get("vlan_id", 0),
                 e.get("client_class", 0), e.get("flags", 0))
         return len(entries)
+
+    def export_nat_sessions(self, since_ns: int = 0):
+        """Golden-model NAT session export, same record layout as the
+        GPU path (SESS_EXPORT_DTYPE)."""
+        import numpy as np
+        recs = [(k, s) for k, s in self.dp.nat_sessions.items()
+                if s.last_seen >= since_ns]
+        out = np.zeros(len(recs), dtype=abi.SESS_EXPORT_DTYPE)
+        eim_on = bool(self.dp.nat_flags & abi.NAT_FLAG_EIM)
+        for i, (k, s) in enumerate(recs):
+            src, dst, sp, dp_, pr = k
+            out[i] = (src, dst, sp, dp_, pr, s.state, s.is_hairpin,
+                      1 if eim_on else 0, s.nat_ip, s.nat_port,
+                      s.nat_port if eim_on else 0, s.created,
+                      s.last_seen, 0)
+        return out
+
+    def import_nat_sessions(self, records) -> int:
+        import numpy as np
+        from .golden import EimRec, NatSessionRec
+        records = np.asarray(records, dtype=abi.SESS_EXPORT_DTYPE)
+        for r in records:
+            key = (int(r["src_ip"]), int(r["dst_ip"]), int(r["src_port"]),
+                   int(r["dst_port"]), int(r["protocol"]))
+            self.dp.nat_sessions[key] = NatSessionRec(
+                int(r["nat_ip"]), int(r["nat_port"]), int(r["src_port"]),
+                int(r["src_ip"]), int(r["dst_ip"]), int(r["dst_port"]),
+                int(r["state"]), int(r["is_hairpin"]),
+                int(r["last_seen"]), int(r["created"]), 0, 0, 0, 0,
+                int(r["protocol"]))
+            self.dp.nat_reverse[(int(r["dst_ip"]), int(r["nat_ip"]),
+                                 int(r["dst_port"]), int(r["nat_port"]),
+                                 int(r["protocol"]))] = key
+            if int(r["flags"]) & 1:
+                self.dp.eim[(int(r["src_ip"]), int(r["src_port"]),
+                             int(r["protocol"]))] = EimRec(
+                    int(r["nat_ip"]), int(r["eim_port"]),
+                    int(r["created"]), int(r["last_seen"]), 1)
+        return len(records)
 
     def process_nat44(self, frames, egress=True, now_ns=None):
         if now_ns is not None:

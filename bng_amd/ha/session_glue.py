@@ -32,12 +32,26 @@ def session_to_lease(s: SessionState, pool_id: int = 1) -> Lease:
                  policy_name=s.policy_name)
 
 
-def attach(dhcp_server: DHCPServer, syncer: HASyncer) -> None:
-    """Active side: every lease add/delete becomes a sync delta."""
+def attach(dhcp_server: DHCPServer, syncer: HASyncer,
+           nat_mgr=None) -> None:
+    """Active side: every lease add/delete becomes a sync delta.  With
+    a NAT manager the delta carries the exact port-block assignment, so
+    promotion restores the SAME block (ref SessionState carries
+    NATPortStart/End, ha/protocol.go:76-111)."""
 
     def on_event(event: str, lease: Lease):
         if event == "add":
-            syncer.publish_add(lease_to_session(lease))
+            s = lease_to_session(lease)
+            # nat_mgr may be a callable for late binding (CLI wires HA
+            # before the NAT manager exists)
+            mgr = nat_mgr() if callable(nat_mgr) else nat_mgr
+            if mgr is not None:
+                alloc = mgr.allocations.get(lease.ip)
+                if alloc is not None:
+                    s.nat_public_ip = u32_to_ip(alloc.public_ip)
+                    s.nat_port_start = alloc.port_start
+                    s.nat_port_end = alloc.port_end
+            syncer.publish_add(s)
         else:
             syncer.publish_delete(f"dhcp-{lease.mac.hex()}")
 
@@ -60,7 +74,11 @@ def promote(dhcp_server: DHCPServer, syncer: HASyncer,
             qos_mgr.apply_policy(lease.ip, s.policy_name)
         if nat_mgr is not None:
             try:
-                nat_mgr.allocate_nat(lease.ip, s.subscriber_id)
+                if s.nat_public_ip:   # exact-block restore (see attach)
+                    nat_mgr.restore_nat(lease.ip, ip2u32(s.nat_public_ip),
+                                        s.nat_port_start, s.nat_port_end)
+                else:
+                    nat_mgr.allocate_nat(lease.ip, s.subscriber_id)
             except Exception:
                 pass
         n += 1

@@ -146,6 +146,37 @@ class Manager:
                 "dest_ip": 0, "dest_port": 0, "protocol": 0, "flags": 0})
         return alloc
 
+    def restore_nat(self, private_ip: int, public_ip: int,
+                    port_start: int, port_end: int,
+                    subscriber_id: int = 0) -> "NATAllocation":
+        """Exact-block restore at HA promotion: the standby must keep
+        the ACTIVE's block assignment, not allocate a fresh one —
+        otherwise restored sessions' external ports land inside other
+        subscribers' new blocks (round-1 VERDICT task 3)."""
+        with self._lock:
+            alloc = NATAllocation(
+                private_ip=private_ip, public_ip=public_ip,
+                port_start=port_start, port_end=port_end,
+                subscriber_id=subscriber_id, allocated_at=time.time())
+            self.allocations[private_ip] = alloc
+            # keep the counter ahead of restored block indices so new
+            # allocations never collide with restored ones
+            blocks_per_ip = (self.PORT_RANGE_END + 1 -
+                             self.PORT_RANGE_START) // self.ports_per_sub
+            blk = (port_start - self.PORT_RANGE_START) // self.ports_per_sub
+            try:
+                ip_idx = self.public_ips.index(public_ip)
+            except ValueError:
+                ip_idx = 0
+            idx = ip_idx * blocks_per_ip + blk
+            if idx >= self._sub_counter:
+                self._sub_counter = idx + 1
+        if self.launcher is not None:
+            self.launcher.add_subscriber_nat(
+                alloc.private_ip, alloc.public_ip, alloc.port_start,
+                alloc.port_end, alloc.subscriber_id)
+        return alloc
+
     def release_nat(self, private_ip: int):
         with self._lock:
             alloc = self.allocations.pop(private_ip, None)
