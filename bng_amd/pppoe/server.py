@@ -45,12 +45,14 @@ ST_OPENED = "opened"
 class CPState:
     state: str = ST_CLOSED
     our_ident: int = 0
-    retransmits: int = 0
+    retransmits: int = 0     # Restart counter sends (RFC1661 Max-Configure)
+    failures: int = 0        # Nak/Rej rounds (RFC1661 Max-Failure)
     last_req: bytes = b""
 
     def reset(self):
         self.state = ST_CLOSED
         self.retransmits = 0
+        self.failures = 0
 
 
 @dataclass
@@ -65,6 +67,9 @@ class Session:
     our_magic: int = 0
     peer_magic: int = 0
     peer_mru: int = 1492
+    our_mru: int = 0               # 0 = announce server default
+    magic_loops: int = 0           # looped-link detections (RFC1661 §6.4)
+    lcp_suppressed: frozenset = frozenset()   # options the peer Rejected
     auth_proto: int = C.PROTO_CHAP
     chap_challenge: bytes = b""
     chap_ident: int = 0
@@ -84,7 +89,13 @@ class Session:
 
 
 class PPPoEServer:
-    MAX_AUTH_ATTEMPTS = 3      # ref auth.go rate limiting
+    MAX_AUTH_ATTEMPTS = 3      # ref auth.go rate limiting (per session)
+    MAX_CONFIGURE = 10         # RFC1661 Max-Configure restart counter
+    MAX_FAILURE = 5            # RFC1661 Max-Failure Nak/Rej rounds
+    MAX_MAGIC_LOOPS = 3        # looped-link give-up (RFC1661 §6.4)
+    AUTH_FAIL_WINDOW = 60.0    # per-MAC throttle (ref auth.go:202-580:
+    AUTH_FAIL_LIMIT = 5        # rate limiting survives session churn)
+    AUTH_LOCKOUT = 30.0
 
     def __init__(self, server_mac: bytes, ac_name: str = "bng-amd",
                  service_name: str = "", auth: str = "chap",
@@ -114,7 +125,11 @@ class PPPoEServer:
             "padi", "pado", "padr", "pads", "padt_rx", "padt_tx",
             "lcp_opened", "auth_ok", "auth_fail", "ipcp_opened",
             "ipv6cp_opened", "sessions_open", "echo_timeout",
-            "term_rx")}
+            "term_rx", "loopback_detected", "restart_exhausted",
+            "auth_throttled")}
+        # per-MAC auth-failure timestamps (survives session teardown, so
+        # a client cannot reset the limit by re-discovering)
+        self._auth_fails: Dict[bytes, List[float]] = {}
 
     # ------------------------------------------------------------ entry
     def handle_frame(self, frame: bytes) -> List[bytes]:
@@ -153,6 +168,12 @@ class PPPoEServer:
                                       dst_mac=p.src_mac).encode()]
         if p.code == C.PADR:
             self.stats["padr"] += 1
+            if self._auth_locked(p.src_mac):
+                self.stats["auth_throttled"] += 1
+                return [C.DiscoveryPacket(
+                    C.PADS, 0,
+                    [(C.TAG_GENERIC_ERROR, b"too many auth failures")],
+                    src_mac=self.server_mac, dst_mac=p.src_mac).encode()]
             cookie = C.get_tag(p.tags, C.TAG_AC_COOKIE)
             if cookie != self._cookie(p.src_mac):
                 return [C.DiscoveryPacket(
@@ -194,15 +215,40 @@ class PPPoEServer:
         return []
 
     # ------------------------------------------------------------- LCP
+    def _auth_locked(self, mac: bytes, now: Optional[float] = None) -> bool:
+        now = now or time.time()
+        fails = self._auth_fails.get(mac)
+        if not fails:
+            return False
+        fails[:] = [t for t in fails if now - t < self.AUTH_FAIL_WINDOW]
+        if not fails:
+            del self._auth_fails[mac]
+            return False
+        return (len(fails) >= self.AUTH_FAIL_LIMIT and
+                now - fails[-1] < self.AUTH_LOCKOUT)
+
     def _send_lcp_req(self, s: Session) -> List[bytes]:
+        """(Re)send our Configure-Request, honoring options the peer
+        Rejected and values it Nak-suggested, bounded by the RFC1661
+        Max-Configure restart counter (ref lcp.go option negotiation)."""
+        s.lcp.retransmits += 1
+        if s.lcp.retransmits > self.MAX_CONFIGURE:
+            self.stats["restart_exhausted"] += 1
+            return self._teardown(s)
         s.lcp.our_ident = (s.lcp.our_ident + 1) & 0xFF
-        opts = [(C.LCP_OPT_MRU, struct.pack(">H", self.mru)),
-                (C.LCP_OPT_MAGIC, struct.pack(">I", s.our_magic))]
-        if self.auth_kind == "chap":
-            opts.append((C.LCP_OPT_AUTH,
-                         struct.pack(">HB", C.PROTO_CHAP, 5)))
-        elif self.auth_kind == "pap":
-            opts.append((C.LCP_OPT_AUTH, struct.pack(">H", C.PROTO_PAP)))
+        opts = []
+        if C.LCP_OPT_MRU not in s.lcp_suppressed:
+            opts.append((C.LCP_OPT_MRU,
+                         struct.pack(">H", s.our_mru or self.mru)))
+        if C.LCP_OPT_MAGIC not in s.lcp_suppressed:
+            opts.append((C.LCP_OPT_MAGIC, struct.pack(">I", s.our_magic)))
+        if C.LCP_OPT_AUTH not in s.lcp_suppressed:
+            if self.auth_kind == "chap":
+                opts.append((C.LCP_OPT_AUTH,
+                             struct.pack(">HB", C.PROTO_CHAP, 5)))
+            elif self.auth_kind == "pap":
+                opts.append((C.LCP_OPT_AUTH,
+                             struct.pack(">H", C.PROTO_PAP)))
         req = C.CPPacket(C.CONF_REQ, s.lcp.our_ident,
                          C.encode_opts(opts)).encode()
         s.lcp.last_req = req
@@ -247,7 +293,18 @@ class PPPoEServer:
                     else:
                         s.peer_mru = mru
                 elif t == C.LCP_OPT_MAGIC:
-                    s.peer_magic = struct.unpack(">I", v)[0]
+                    peer_magic = struct.unpack(">I", v)[0]
+                    if peer_magic == s.our_magic:
+                        # our own magic coming back: looped link
+                        # (RFC1661 §6.4) — Nak with a fresh number;
+                        # give up after MAX_MAGIC_LOOPS
+                        s.magic_loops += 1
+                        self.stats["loopback_detected"] += 1
+                        if s.magic_loops >= self.MAX_MAGIC_LOOPS:
+                            return self._teardown(s)
+                        naks.append((t, os.urandom(4)))
+                    else:
+                        s.peer_magic = peer_magic
                 elif t in (C.LCP_OPT_PFC, C.LCP_OPT_ACFC):
                     rejs.append((t, v))       # we don't compress
                 elif t == C.LCP_OPT_AUTH:
@@ -271,14 +328,48 @@ class PPPoEServer:
                     s.lcp.state = ST_OPENED
                     out += self._lcp_opened(s)
         elif cp.code == C.CONF_ACK:
+            s.lcp.retransmits = 0
+            s.lcp.failures = 0
             if s.lcp.state == ST_REQ_SENT:
                 s.lcp.state = ST_ACK_RCVD
             elif s.lcp.state == ST_ACK_SENT:
                 s.lcp.state = ST_OPENED
                 out += self._lcp_opened(s)
         elif cp.code in (C.CONF_NAK, C.CONF_REJ):
-            out += self._send_lcp_req(s)      # simplistic renegotiate
+            # real option bargaining (RFC1661; ref lcp.go): adopt Nak'd
+            # values, drop Rejected options, bounded by Max-Failure
+            s.lcp.failures += 1
+            if s.lcp.failures > self.MAX_FAILURE:
+                self.stats["restart_exhausted"] += 1
+                return self._teardown(s)
+            peer_opts = C.decode_opts(cp.data)
+            if cp.code == C.CONF_NAK:
+                for t, v in peer_opts:
+                    if t == C.LCP_OPT_MRU and len(v) == 2:
+                        sug = struct.unpack(">H", v)[0]
+                        if sug >= 576:
+                            s.our_mru = min(sug, self.mru)
+                    elif t == C.LCP_OPT_MAGIC and len(v) == 4:
+                        s.our_magic = struct.unpack(">I", v)[0]
+                    # auth-proto Nak: we do not downgrade the
+                    # configured authenticator — re-request (the
+                    # Max-Failure bound terminates a peer that will
+                    # never take it; ref auth is non-negotiable too)
+            else:
+                dropped = {t for t, _ in peer_opts}
+                if C.LCP_OPT_AUTH in dropped and self.auth_kind != "none":
+                    # peer refuses to authenticate: no service
+                    return self._teardown(s)
+                s.lcp_suppressed = frozenset(s.lcp_suppressed | dropped)
+            out += self._send_lcp_req(s)
         elif cp.code == C.ECHO_REQ:
+            if (len(cp.data) >= 4 and
+                    struct.unpack(">I", cp.data[:4])[0] == s.our_magic):
+                # echo carrying OUR magic: looped link (RFC1661 §5.8)
+                s.magic_loops += 1
+                self.stats["loopback_detected"] += 1
+                if s.magic_loops >= self.MAX_MAGIC_LOOPS:
+                    return self._teardown(s)
             out.append(self._sess_frame(s, C.PROTO_LCP, C.CPPacket(
                 C.ECHO_REP, cp.identifier,
                 struct.pack(">I", s.our_magic)).encode()))
@@ -311,6 +402,7 @@ class PPPoEServer:
                    msg: bytes) -> List[bytes]:
         self.stats["auth_fail"] += 1
         s.auth_attempts += 1
+        self._auth_fails.setdefault(s.client_mac, []).append(time.time())
         code = C.CHAP_FAILURE if proto == C.PROTO_CHAP else C.PAP_AUTH_NAK
         body = msg if proto == C.PROTO_CHAP else bytes([len(msg)]) + msg
         out = [self._sess_frame(s, proto,
@@ -534,6 +626,14 @@ class PPPoEServer:
         with self._lock:
             open_sessions = [s for s in self.sessions.values()
                              if s.phase == PH_OPEN]
+            pending_lcp = [s for s in self.sessions.values()
+                           if s.phase == PH_LCP and
+                           s.lcp.state in (ST_REQ_SENT, ST_ACK_RCVD,
+                                           ST_ACK_SENT)]
+        # restart timer: retransmit our un-Acked Configure-Request,
+        # bounded by Max-Configure (RFC1661 restart counter)
+        for s in pending_lcp:
+            out += self._send_lcp_req(s)
         for s in open_sessions:
             if s.echo_pending >= self.echo_fails:
                 self.stats["echo_timeout"] += 1

@@ -412,3 +412,151 @@ class TestCodecFuzz:
             src_mac=cli.mac).encode()
         for cut in range(len(padi)):
             srv.handle_frame(padi[:cut])    # must not raise
+
+
+def _discover_session(srv, mac=b"\xaa\xbb\xcc\x00\x00\x77"):
+    """PADI/PADR by hand; returns (session, lcp frames from PADS)."""
+    padi = C.DiscoveryPacket(C.PADI, 0, [(C.TAG_SERVICE_NAME, b"")],
+                             src_mac=mac).encode()
+    pado = C.DiscoveryPacket.decode(srv.handle_frame(padi)[0])
+    cookie = C.get_tag(pado.tags, C.TAG_AC_COOKIE)
+    padr = C.DiscoveryPacket(C.PADR, 0, [
+        (C.TAG_SERVICE_NAME, b""), (C.TAG_AC_COOKIE, cookie)],
+        src_mac=mac).encode()
+    frames = srv.handle_frame(padr)
+    pads = C.DiscoveryPacket.decode(frames[0])
+    sid = pads.session_id
+    return srv.sessions[sid], frames[1:]
+
+
+def _lcp_from(frames):
+    """Extract decoded LCP CPPackets from session frames."""
+    out = []
+    for f in frames:
+        if struct.unpack_from(">H", f, 12)[0] == C.ETH_PPPOE_SESS:
+            p = C.SessionPacket.decode(f)
+            if p.ppp_proto == C.PROTO_LCP:
+                out.append(C.CPPacket.decode(p.payload))
+    return out
+
+
+def _send_lcp(srv, s, code, ident, data=b""):
+    return srv.handle_frame(C.SessionPacket(
+        s.session_id, C.PROTO_LCP,
+        C.CPPacket(code, ident, data).encode(),
+        src_mac=s.client_mac, dst_mac=srv.server_mac).encode())
+
+
+class TestLcpBargaining:
+    """Round-1 VERDICT task 6: Configure-Nak/Reject option bargaining,
+    magic-number loop detection, restart counters, auth throttling
+    (ref lcp.go / auth.go:202-580)."""
+
+    def make(self):
+        srv = PPPoEServer(SRV_MAC, auth="chap")
+        srv.local_users["alice"] = "pw1"
+        return srv
+
+    def test_nak_adopts_suggested_mru(self):
+        srv = self.make()
+        s, frames = _discover_session(srv)
+        req = _lcp_from(frames)[0]
+        assert req.code == C.CONF_REQ
+        # client Naks our MRU, suggesting 1400
+        out = _send_lcp(srv, s, C.CONF_NAK, req.identifier,
+                        C.encode_opts([(C.LCP_OPT_MRU,
+                                        struct.pack(">H", 1400))]))
+        req2 = _lcp_from(out)[0]
+        opts = dict(C.decode_opts(req2.data))
+        assert struct.unpack(">H", opts[C.LCP_OPT_MRU])[0] == 1400
+        assert s.our_mru == 1400
+
+    def test_reject_drops_option_but_not_auth(self):
+        srv = self.make()
+        s, frames = _discover_session(srv)
+        req = _lcp_from(frames)[0]
+        # client Rejects our MAGIC option -> resent request omits it
+        out = _send_lcp(srv, s, C.CONF_REJ, req.identifier,
+                        C.encode_opts([(C.LCP_OPT_MAGIC, b"\0\0\0\0")]))
+        req2 = _lcp_from(out)[0]
+        assert req2.code == C.CONF_REQ
+        types = [t for t, _ in C.decode_opts(req2.data)]
+        assert C.LCP_OPT_MAGIC not in types
+        assert C.LCP_OPT_AUTH in types
+        # rejecting AUTH ends the session (no unauthenticated service)
+        out = _send_lcp(srv, s, C.CONF_REJ, req2.identifier,
+                        C.encode_opts([(C.LCP_OPT_AUTH,
+                                        struct.pack(">HB", C.PROTO_CHAP,
+                                                    5))]))
+        assert s.session_id not in srv.sessions
+        assert any(struct.unpack_from(">H", f, 12)[0] == C.ETH_PPPOE_DISC
+                   for f in out)   # PADT sent
+
+    def test_magic_loop_detection_naks_then_terminates(self):
+        srv = self.make()
+        s, _ = _discover_session(srv)
+        # peer's CONF_REQ carries OUR magic: looped link
+        for i in range(srv.MAX_MAGIC_LOOPS):
+            alive = s.session_id in srv.sessions
+            out = _send_lcp(srv, s, C.CONF_REQ, 10 + i,
+                            C.encode_opts([(C.LCP_OPT_MAGIC,
+                                            struct.pack(">I",
+                                                        s.our_magic))]))
+            if i < srv.MAX_MAGIC_LOOPS - 1:
+                naks = [p for p in _lcp_from(out) if p.code == C.CONF_NAK]
+                assert naks, "expected Configure-Nak with a fresh magic"
+        assert s.session_id not in srv.sessions
+        assert srv.stats["loopback_detected"] == srv.MAX_MAGIC_LOOPS
+
+    def test_nak_storm_hits_max_failure(self):
+        srv = self.make()
+        s, frames = _discover_session(srv)
+        req = _lcp_from(frames)[0]
+        ident = req.identifier
+        for _ in range(srv.MAX_FAILURE + 1):
+            if s.session_id not in srv.sessions:
+                break
+            out = _send_lcp(srv, s, C.CONF_NAK, ident,
+                            C.encode_opts([(C.LCP_OPT_MRU,
+                                            struct.pack(">H", 1400))]))
+            got = _lcp_from(out)
+            if got:
+                ident = got[0].identifier
+        assert s.session_id not in srv.sessions
+        assert srv.stats["restart_exhausted"] == 1
+
+    def test_tick_retransmits_until_max_configure(self):
+        srv = self.make()
+        s, _ = _discover_session(srv)
+        sends = 1                       # initial request at PADS
+        while s.session_id in srv.sessions and sends < 50:
+            srv.tick()
+            sends += 1
+        assert s.session_id not in srv.sessions
+        assert sends == srv.MAX_CONFIGURE + 1
+        assert srv.stats["restart_exhausted"] == 1
+
+    def test_per_mac_auth_throttle_survives_rediscovery(self):
+        srv = self.make()
+        srv.AUTH_FAIL_LIMIT = 3
+        mac = b"\xaa\xbb\xcc\x00\x00\x99"
+        # record 3 failures (as _auth_fail would)
+        import time as _t
+        srv._auth_fails[mac] = [_t.time()] * 3
+        # a fresh PADR from the same MAC is refused while locked out
+        padi = C.DiscoveryPacket(C.PADI, 0, [(C.TAG_SERVICE_NAME, b"")],
+                                 src_mac=mac).encode()
+        pado = C.DiscoveryPacket.decode(srv.handle_frame(padi)[0])
+        cookie = C.get_tag(pado.tags, C.TAG_AC_COOKIE)
+        padr = C.DiscoveryPacket(C.PADR, 0, [
+            (C.TAG_SERVICE_NAME, b""), (C.TAG_AC_COOKIE, cookie)],
+            src_mac=mac).encode()
+        out = srv.handle_frame(padr)
+        pads = C.DiscoveryPacket.decode(out[0])
+        assert pads.session_id == 0
+        assert C.get_tag(pads.tags, C.TAG_GENERIC_ERROR) is not None
+        assert srv.stats["auth_throttled"] == 1
+        # outside the window the client may try again
+        srv._auth_fails[mac] = [_t.time() - srv.AUTH_FAIL_WINDOW - 1] * 3
+        out = srv.handle_frame(padr)
+        assert C.DiscoveryPacket.decode(out[0]).session_id != 0
