@@ -1,19 +1,27 @@
-"""CLSet — CRDT-backed Store with peer sync.
+"""CLSet — CRDT-backed Store with peer sync and persistence.
 
 Python re-design of the reference's CLSetStore (pkg/nexus/clset.go:45-427
-+ crdt_backend.go): an eventually-consistent replicated K/V store that
-keeps serving reads/writes through partitions and converges on merge.
++ crdt_backend.go:1-320): an eventually-consistent replicated K/V store
+that keeps serving reads/writes through partitions and converges on
+merge.
 
 CRDT: last-writer-wins register map with tombstones — each entry carries
 (lamport, node_id); merge order is (lamport, node_id) lexicographic, so
 concurrent writes converge identically on every replica.  The reference
-uses a libp2p gossip CLSet; here sync is pull-based anti-entropy over
-HTTP (or direct peer references in tests), which fits the BNG's
-control-plane rates.
+uses a libp2p gossip CLSet over a badger datastore; here:
+
+  * persistence = snapshot file + append-only WAL in `data_dir`
+    (the badger analog: every local or merged write is durable before
+    it is acknowledged; the WAL compacts into the snapshot);
+  * sync = pull-based anti-entropy over HTTP with per-peer reconnect
+    backoff (the gossip analog at control-plane rates);
+  * discovery = snapshots advertise known peer URLs, so reachable
+    replicas learn the full mesh transitively (gossipsub analog).
 """
 from __future__ import annotations
 
 import json
+import os
 import threading
 import time
 from typing import Dict, List, Optional, Tuple
@@ -22,18 +30,99 @@ from .store import Store, WatchEvent
 
 
 class CLSetStore(Store):
-    def __init__(self, node_id: str, sync_interval: float = 1.0):
+    WAL_COMPACT_EVERY = 1024     # records before auto-compaction
+
+    def __init__(self, node_id: str, sync_interval: float = 1.0,
+                 data_dir: Optional[str] = None,
+                 advertise_url: str = "",
+                 backoff_base: float = 1.0, backoff_max: float = 30.0):
         self.node_id = node_id
         self.sync_interval = sync_interval
-        # key -> [value_b64|None, lamport, node_id, deleted]
+        self.advertise_url = advertise_url.rstrip("/")
+        self.backoff_base = backoff_base
+        self.backoff_max = backoff_max
+        # key -> [value|None, lamport, node_id, deleted]
         self._entries: Dict[str, Tuple[Optional[bytes], int, str, bool]] = {}
         self._lamport = 0
         self._lock = threading.RLock()
         self._watchers: List[tuple] = []
         self._peers: List["CLSetStore"] = []
         self._peer_urls: List[str] = []
+        # url -> {"fails": n, "next_try": ts, "last_ok": ts}
+        self._peer_state: Dict[str, dict] = {}
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
+        self.stats = {"syncs_ok": 0, "syncs_failed": 0, "adopted": 0,
+                      "peers_discovered": 0, "compactions": 0}
+        self._data_dir = data_dir
+        self._wal = None
+        self._wal_records = 0
+        if data_dir:
+            os.makedirs(data_dir, exist_ok=True)
+            self._load()
+            self._wal = open(self._wal_path, "a", encoding="utf-8")
+
+    # ------------------------------------------------------ persistence
+    @property
+    def _snap_path(self):
+        return os.path.join(self._data_dir, "clset_snapshot.json")
+
+    @property
+    def _wal_path(self):
+        return os.path.join(self._data_dir, "clset_wal.jsonl")
+
+    def _load(self):
+        import base64
+        if os.path.exists(self._snap_path):
+            with open(self._snap_path, encoding="utf-8") as f:
+                snap = json.load(f)
+            self._lamport = int(snap.get("lamport", 0))
+            for k, (v64, lam, nid, dead) in snap.get("entries",
+                                                     {}).items():
+                v = None if v64 is None else base64.b64decode(v64)
+                self._entries[k] = (v, lam, nid, dead)
+        if os.path.exists(self._wal_path):
+            with open(self._wal_path, encoding="utf-8") as f:
+                for line in f:
+                    line = line.strip()
+                    if not line:
+                        continue
+                    try:
+                        rec = json.loads(line)
+                    except ValueError:
+                        break        # torn tail record: stop replay
+                    k, v64, lam, nid, dead = rec
+                    v = None if v64 is None else base64.b64decode(v64)
+                    cur = self._entries.get(k)
+                    if cur is None or (cur[1], cur[2]) < (lam, nid):
+                        self._entries[k] = (v, lam, nid, dead)
+                    self._lamport = max(self._lamport, lam)
+                    self._wal_records += 1
+
+    def _wal_append(self, key, v, lam, nid, dead):
+        if self._wal is None:
+            return
+        import base64
+        v64 = None if v is None else base64.b64encode(v).decode()
+        self._wal.write(json.dumps([key, v64, lam, nid, dead]) + "\n")
+        self._wal.flush()
+        self._wal_records += 1
+        if self._wal_records >= self.WAL_COMPACT_EVERY:
+            self.compact()
+
+    def compact(self):
+        """Fold the WAL into the snapshot (badger-compaction analog)."""
+        if self._data_dir is None:
+            return
+        tmp = self._snap_path + ".tmp"
+        with open(tmp, "w", encoding="utf-8") as f:
+            json.dump(self.snapshot(), f)
+        os.replace(tmp, self._snap_path)
+        if self._wal is not None:
+            self._wal.close()
+        self._wal = open(self._wal_path, "w", encoding="utf-8")
+        self._wal_records = 0
+        self.stats["compactions"] += 1
 
     # ------------------------------------------------------------- Store
     def get(self, key):
@@ -48,6 +137,8 @@ class CLSetStore(Store):
             self._lamport += 1
             self._entries[key] = (bytes(value), self._lamport, self.node_id,
                                   False)
+            self._wal_append(key, bytes(value), self._lamport,
+                             self.node_id, False)
             watchers = [w for w in self._watchers if key.startswith(w[0])]
         for _, cb in watchers:
             cb(WatchEvent("put", key, bytes(value)))
@@ -56,6 +147,7 @@ class CLSetStore(Store):
         with self._lock:
             self._lamport += 1
             self._entries[key] = (None, self._lamport, self.node_id, True)
+            self._wal_append(key, None, self._lamport, self.node_id, True)
             watchers = [w for w in self._watchers if key.startswith(w[0])]
         for _, cb in watchers:
             cb(WatchEvent("delete", key))
@@ -81,9 +173,12 @@ class CLSetStore(Store):
         """Serializable replica state for anti-entropy."""
         import base64
         with self._lock:
+            known = [u for u in ([self.advertise_url] + self._peer_urls)
+                     if u]
             return {
                 "node_id": self.node_id,
                 "lamport": self._lamport,
+                "peers": known,        # gossip-style peer discovery
                 "entries": {
                     k: [None if v is None
                         else base64.b64encode(v).decode(), lam, nid, dead]
@@ -104,13 +199,22 @@ class CLSetStore(Store):
                     continue
                 v = None if v64 is None else base64.b64decode(v64)
                 self._entries[k] = (v, lam, nid, dead)
+                self._wal_append(k, v, lam, nid, dead)   # merged = durable
                 adopted += 1
                 for w in self._watchers:
                     if k.startswith(w[0]):
                         events.append((w[1], WatchEvent(
                             "delete" if dead else "put", k, v)))
+            # transitive peer discovery: adopt unknown advertised URLs
+            for u in snap.get("peers", []):
+                u = u.rstrip("/")
+                if u and u != self.advertise_url and \
+                        u not in self._peer_urls:
+                    self._peer_urls.append(u)
+                    self.stats["peers_discovered"] += 1
         for cb, ev in events:
             cb(ev)
+        self.stats["adopted"] += adopted
         return adopted
 
     # -------------------------------------------------------------- sync
@@ -120,23 +224,41 @@ class CLSetStore(Store):
     def add_peer_url(self, url: str):
         self._peer_urls.append(url.rstrip("/"))
 
-    def sync_once(self) -> int:
-        """One anti-entropy round against every reachable peer."""
+    def sync_once(self, now: Optional[float] = None) -> int:
+        """One anti-entropy round against every reachable peer, with
+        per-peer exponential reconnect backoff (the gossip transport's
+        reconnect semantics, crdt_backend.go peer management)."""
+        now = now if now is not None else time.monotonic()
         adopted = 0
         for p in self._peers:
             adopted += self.merge(p.snapshot())
             p.merge(self.snapshot())
         if self._peer_urls:
             import requests
-            for url in self._peer_urls:
+            for url in list(self._peer_urls):
+                st = self._peer_state.setdefault(
+                    url, {"fails": 0, "next_try": 0.0, "last_ok": 0.0})
+                if now < st["next_try"]:
+                    continue          # still backing off
                 try:
                     r = requests.post(f"{url}/clset/sync",
                                       json=self.snapshot(), timeout=5)
-                    if r.status_code == 200:
-                        adopted += self.merge(r.json())
+                    r.raise_for_status()
+                    adopted += self.merge(r.json())
+                    st["fails"] = 0
+                    st["next_try"] = 0.0
+                    st["last_ok"] = now
+                    self.stats["syncs_ok"] += 1
                 except Exception:
-                    continue
+                    st["fails"] += 1
+                    st["next_try"] = now + min(
+                        self.backoff_base * (2 ** (st["fails"] - 1)),
+                        self.backoff_max)
+                    self.stats["syncs_failed"] += 1
         return adopted
+
+    def peer_status(self) -> Dict[str, dict]:
+        return {u: dict(s) for u, s in self._peer_state.items()}
 
     def start(self):
         self._thread = threading.Thread(target=self._loop, daemon=True)
@@ -147,6 +269,10 @@ class CLSetStore(Store):
         self._stop.set()
         if self._thread:
             self._thread.join(timeout=2)
+        if self._wal is not None:
+            self.compact()            # durable, minimal restart replay
+            self._wal.close()
+            self._wal = None
 
     def _loop(self):
         while not self._stop.wait(self.sync_interval):

@@ -322,3 +322,116 @@ class TestHTTPAllocatorErrors:
         with pytest.raises(Exception):
             a.allocate_ipv4("p1", "sub-1")
         assert a.health_check() is False
+
+
+class TestCLSetPersistence:
+    """Round-1 VERDICT task 7: restart survival + partition/merge over
+    real HTTP peers (ref crdt_backend.go:1-320 badger persistence +
+    libp2p peer management)."""
+
+    def test_restart_survival(self, tmp_path):
+        from bng_amd.nexus.clset import CLSetStore
+        d = str(tmp_path / "a")
+        s = CLSetStore("n1", data_dir=d)
+        s.put("sub/1", b"alice")
+        s.put("sub/2", b"bob")
+        s.delete("sub/1")
+        lam = s._lamport
+        s.close()
+        s2 = CLSetStore("n1", data_dir=d)
+        assert s2.get("sub/2") == b"bob"
+        assert s2.get("sub/1") is None           # tombstone survived
+        assert s2._lamport == lam                # clock survived
+        # tombstone must still win merges after restart
+        other = CLSetStore("n0")
+        other.put("sub/1", b"stale")
+        other._entries["sub/1"] = (b"stale", 1, "n0", False)
+        s2.merge(other.snapshot())
+        assert s2.get("sub/1") is None
+        s2.close()
+
+    def test_wal_replay_and_compaction(self, tmp_path):
+        from bng_amd.nexus.clset import CLSetStore
+        d = str(tmp_path / "b")
+        s = CLSetStore("n1", data_dir=d)
+        s.WAL_COMPACT_EVERY = 50
+        for i in range(130):
+            s.put(f"k/{i}", str(i))
+        assert s.stats["compactions"] >= 2
+        # simulate crash: do NOT close (no final compaction)
+        s._wal.flush()
+        s2 = CLSetStore("n1", data_dir=d)
+        assert s2.get("k/129") == b"129"
+        assert len(s2.list("k/")) == 130
+        s2.close()
+
+    def test_torn_wal_tail_ignored(self, tmp_path):
+        from bng_amd.nexus.clset import CLSetStore
+        d = str(tmp_path / "c")
+        s = CLSetStore("n1", data_dir=d)
+        s.put("x", b"1")
+        s._wal.flush()
+        with open(s._wal_path, "a") as f:
+            f.write('["y", "zz", 99')     # torn record
+        s2 = CLSetStore("n1", data_dir=d)
+        assert s2.get("x") == b"1"
+        assert s2.get("y") is None
+        s2.close()
+
+    def test_partition_merge_over_http_with_backoff(self, tmp_path):
+        import time as _t
+        from bng_amd.nexus.clset import CLSetHTTPServer, CLSetStore
+        a = CLSetStore("a", data_dir=str(tmp_path / "pa"),
+                       backoff_base=0.05, backoff_max=0.2)
+        b = CLSetStore("b", data_dir=str(tmp_path / "pb"))
+        srv_b = CLSetHTTPServer(b).start()
+        a.add_peer_url(srv_b.url)
+        a.put("k1", b"from-a")
+        b.put("k2", b"from-b")
+        assert a.sync_once() >= 1
+        assert a.get("k2") == b"from-b"
+        assert b.get("k1") == b"from-a"
+        # partition: peer goes away; writes continue on both sides
+        srv_b.stop()
+        a.put("k3", b"a-during-partition")
+        b.put("k4", b"b-during-partition")
+        a.sync_once()
+        assert a.stats["syncs_failed"] >= 1
+        st = a.peer_status()[srv_b.url]
+        assert st["fails"] >= 1 and st["next_try"] > 0
+        # while backing off, sync_once skips the peer (no new failure)
+        fails_before = a.stats["syncs_failed"]
+        a.sync_once()
+        assert a.stats["syncs_failed"] == fails_before
+        # heal: new server, SAME store b, reachable again after backoff
+        srv_b2 = CLSetHTTPServer(b, port=0).start()
+        a._peer_urls = [srv_b2.url]
+        _t.sleep(0.25)
+        a.sync_once()
+        assert a.get("k4") == b"b-during-partition"
+        assert b.get("k3") == b"a-during-partition"
+        srv_b2.stop()
+        a.close()
+        b.close()
+
+    def test_transitive_peer_discovery(self):
+        from bng_amd.nexus.clset import CLSetHTTPServer, CLSetStore
+        b = CLSetStore("b")
+        c = CLSetStore("c")
+        srv_b = CLSetHTTPServer(b).start()
+        srv_c = CLSetHTTPServer(c).start()
+        b.advertise_url = srv_b.url
+        c.advertise_url = srv_c.url
+        b.add_peer_url(srv_c.url)       # b knows c
+        a = CLSetStore("a")
+        a.add_peer_url(srv_b.url)       # a knows only b
+        c.put("deep", b"value")
+        a.sync_once()                   # learns c's url from b's snapshot
+        assert srv_c.url in a._peer_urls
+        assert a.stats["peers_discovered"] >= 1
+        a.sync_once()                   # now reaches c directly
+        assert a.get("deep") == b"value"
+        srv_b.stop()
+        srv_c.stop()
+        for s in (a, b, c):
+            s.close()
