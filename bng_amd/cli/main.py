@@ -23,6 +23,22 @@ from typing import List, Optional
 from .. import __version__
 
 
+def parse_duration(v) -> int:
+    """Accept Go-style duration strings from flags/YAML (the reference
+    uses time.Duration flags): '300', '300s', '5m', '24h' -> seconds."""
+    if isinstance(v, (int, float)):
+        return int(v)
+    s = str(v).strip().lower()
+    mult = 1
+    for suffix, m in (("ms", 0.001), ("s", 1), ("m", 60), ("h", 3600),
+                      ("d", 86400)):
+        if s.endswith(suffix):
+            s = s[:-len(suffix)]
+            mult = m
+            break
+    return int(float(s) * mult)
+
+
 def build_parser() -> argparse.ArgumentParser:
     p = argparse.ArgumentParser(prog="bng",
                                 description="MI355X-native BNG")
@@ -49,15 +65,28 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--pktio-max-wait", type=float, default=0.0005,
                    help="batch deadline seconds (latency/throughput "
                         "trade at the NIC edge)")
+    g.add_argument("--bpf-path", default="",
+                   help="accepted for reference CLI compatibility; the "
+                        "MI355X dataplane compiles in-tree HIP kernels "
+                        "instead of loading .bpf.o")
+    g.add_argument("--metrics-addr", default="",
+                   help="host:port for /metrics (reference spelling of "
+                        "--metrics-port)")
     g = run.add_argument_group("dhcp")
     g.add_argument("--pool-network", default="")
     g.add_argument("--pool-gateway", default="")
     g.add_argument("--pool-dns", default="")
-    g.add_argument("--lease-time", type=int, default=3600)
+    g.add_argument("--lease-time", type=parse_duration, default=3600,
+                   help="seconds or Go duration (24h, 5m)")
     g.add_argument("--dhcp-listen", action="store_true",
                    help="bind UDP :67 (off in tests)")
     g = run.add_argument_group("radius")
     g.add_argument("--radius-server", action="append", default=[])
+    g.add_argument("--radius-servers", default="",
+                   help="comma list (reference spelling)")
+    g.add_argument("--radius-enabled", action="store_true")
+    g.add_argument("--radius-nas-id", default="bng")
+    g.add_argument("--radius-timeout", type=parse_duration, default=3)
     g.add_argument("--radius-secret", default="")
     g.add_argument("--radius-secret-file", default="")
     g.add_argument("--radius-auth-mode", default="none",
@@ -67,59 +96,148 @@ def build_parser() -> argparse.ArgumentParser:
                    choices=["off", "reject", "cached", "allow"])
     g = run.add_argument_group("resilience")
     g.add_argument("--health-check-interval", type=float, default=5.0)
-    g.add_argument("--short-lease-enable", action="store_true")
+    g.add_argument("--short-lease-enable", "--short-lease-enabled",
+                   action="store_true")
     g.add_argument("--short-lease-threshold", type=float, default=0.9)
-    g.add_argument("--short-lease-duration", type=int, default=60)
+    g.add_argument("--short-lease-duration", type=parse_duration,
+                   default=60)
+    g.add_argument("--health-check-retries", type=int, default=3)
+    g.add_argument("--pool-mode", default="static",
+                   choices=["static", "session", "lease"])
+    g.add_argument("--epoch-period", type=parse_duration, default=300)
+    g.add_argument("--epoch-grace", type=int, default=1)
     g = run.add_argument_group("nexus")
     g.add_argument("--nexus-url", default="")
     g.add_argument("--nexus-auth", default="none",
                    choices=["none", "psk", "mtls"])
     g.add_argument("--nexus-psk", default="")
+    g.add_argument("--nexus-pool", default="default")
+    g = run.add_argument_group("deviceauth")
+    g.add_argument("--auth-mode", default="",
+                   choices=["", "none", "psk", "mtls"],
+                   help="device->Nexus auth (reference spelling of "
+                        "--nexus-auth)")
+    g.add_argument("--auth-psk", default="")
+    g.add_argument("--auth-psk-file", default="")
+    g.add_argument("--auth-mtls-cert", default="")
+    g.add_argument("--auth-mtls-key", default="")
+    g.add_argument("--auth-mtls-ca", default="")
+    g.add_argument("--auth-mtls-server-name", default="")
+    g.add_argument("--auth-mtls-insecure", action="store_true")
     g = run.add_argument_group("peers")
     g.add_argument("--peer", action="append", default=[],
                    help="node_id=url")
+    g.add_argument("--peers", default="",
+                   help="comma list of node_id=url (reference spelling)")
+    g.add_argument("--peer-discovery", default="static",
+                   choices=["static", "dns"])
+    g.add_argument("--peer-service", default="",
+                   help="DNS SRV name for --peer-discovery dns")
+    g.add_argument("--peer-listen", default=":8081")
     g = run.add_argument_group("nat")
-    g.add_argument("--nat-enable", action="store_true")
+    g.add_argument("--nat-enable", "--nat-enabled", action="store_true")
     g.add_argument("--nat-public-ip", action="append", default=[])
     g.add_argument("--nat-ports-per-subscriber", type=int, default=1024)
     g.add_argument("--nat-log-path", default="")
     g.add_argument("--nat-log-format", default="json",
                    choices=["json", "csv", "syslog", "nel"])
     g.add_argument("--nat-bulk-logging", action="store_true")
+    g.add_argument("--nat-bpf-path", default="",
+                   help="accepted for compatibility (see --bpf-path)")
+    g.add_argument("--nat-public-ips", default="",
+                   help="comma list (reference spelling)")
+    g.add_argument("--nat-ports-per-sub", type=int, default=0,
+                   help="reference spelling of --nat-ports-per-subscriber")
+    g.add_argument("--nat-log-enabled", action="store_true")
+    g.add_argument("--nat-inside-interface", default="")
+    g.add_argument("--nat-outside-interface", default="")
+    g.add_argument("--nat-eim", default=True, type=lambda v: v not in
+                   ("false", "0", "no", False))
+    g.add_argument("--nat-eif", default=True, type=lambda v: v not in
+                   ("false", "0", "no", False))
+    g.add_argument("--nat-hairpin", default=True, type=lambda v: v not in
+                   ("false", "0", "no", False))
+    g.add_argument("--nat-alg-ftp", default=True, type=lambda v: v not in
+                   ("false", "0", "no", False))
+    g.add_argument("--nat-alg-sip", default=False, type=lambda v: v in
+                   ("true", "1", "yes", True))
     g = run.add_argument_group("qos")
     g.add_argument("--qos-policy", action="append", default=[],
                    help="name:down_mbps:up_mbps")
+    g.add_argument("--qos-enabled", action="store_true")
+    g.add_argument("--qos-bpf-path", default="",
+                   help="accepted for compatibility (see --bpf-path)")
     g.add_argument("--qos-default-policy", default="")
     g = run.add_argument_group("antispoof")
     g.add_argument("--antispoof-mode", default="disabled",
                    choices=["disabled", "strict", "loose", "log_only"])
     g = run.add_argument_group("pppoe")
-    g.add_argument("--pppoe-enable", action="store_true")
+    g.add_argument("--pppoe-enable", "--pppoe-enabled", action="store_true")
     g.add_argument("--pppoe-ac-name", default="bng-amd")
-    g.add_argument("--pppoe-auth", default="chap",
+    g.add_argument("--pppoe-auth", "--pppoe-auth-type", default="chap",
                    choices=["chap", "pap", "none"])
+    g.add_argument("--pppoe-interface", default="",
+                   help="defaults to --interface")
+    g.add_argument("--pppoe-service-name", default="")
+    g.add_argument("--pppoe-session-timeout", type=parse_duration,
+                   default=1800)
     g = run.add_argument_group("ipv6")
-    g.add_argument("--dhcpv6-enable", action="store_true")
+    g.add_argument("--dhcpv6-enable", "--dhcpv6-enabled", action="store_true")
     g.add_argument("--dhcpv6-na-pool", default="2001:db8:1::/64")
     g.add_argument("--dhcpv6-pd-pool", default="2001:db8:f000::/40")
-    g.add_argument("--slaac-enable", action="store_true")
+    g.add_argument("--slaac-enable", "--slaac-enabled", action="store_true")
     g.add_argument("--slaac-prefix", default="")
+    g.add_argument("--dhcpv6-address-pool", default="",
+                   help="reference spelling of --dhcpv6-na-pool")
+    g.add_argument("--dhcpv6-prefix-pool", default="",
+                   help="reference spelling of --dhcpv6-pd-pool")
+    g.add_argument("--dhcpv6-dns", default="", help="comma list")
+    g.add_argument("--dhcpv6-domain-search", default="",
+                   help="comma list")
+    g.add_argument("--slaac-prefixes", default="",
+                   help="comma list (reference spelling)")
+    g.add_argument("--slaac-managed", action="store_true")
+    g.add_argument("--slaac-other", action="store_true")
+    g.add_argument("--slaac-dns", default="", help="comma RDNSS list")
+    g.add_argument("--slaac-dns-domains", default="",
+                   help="comma DNSSL list")
+    g.add_argument("--slaac-min-interval", type=parse_duration,
+                   default=200)
+    g.add_argument("--slaac-max-interval", type=parse_duration,
+                   default=600)
     g = run.add_argument_group("routing")
-    g.add_argument("--bgp-enable", action="store_true")
+    g.add_argument("--bgp-enable", "--bgp-enabled", action="store_true")
     g.add_argument("--bgp-local-as", type=int, default=65000)
     g.add_argument("--bgp-neighbor", action="append", default=[],
                    help="addr:remote_as")
     g.add_argument("--bgp-announce-subscribers", action="store_true")
+    g.add_argument("--bgp-router-id", default="")
+    g.add_argument("--bgp-neighbors", default="",
+                   help="comma list of addr:remote_as (reference "
+                        "spelling)")
+    g.add_argument("--bgp-bfd-enabled", action="store_true")
     g = run.add_argument_group("ha")
     g.add_argument("--ha-role", default="",
                    choices=["", "active", "standby"])
     g.add_argument("--ha-partner-url", default="")
     g.add_argument("--ha-listen-port", type=int, default=0)
+    g.add_argument("--ha-peer", default="",
+                   help="reference spelling of --ha-partner-url")
+    g.add_argument("--ha-listen", default="",
+                   help="host:port bind for the sync server")
+    g.add_argument("--ha-auth-token", default="",
+                   help="shared secret for /sync/* (required for "
+                        "non-loopback --ha-listen)")
+    g.add_argument("--ha-tls-cert", default="")
+    g.add_argument("--ha-tls-key", default="")
+    g.add_argument("--ha-tls-ca", default="")
+    g.add_argument("--ha-tls-skip-verify", action="store_true")
     g = run.add_argument_group("observability")
     g.add_argument("--metrics-port", type=int, default=9090)
     g.add_argument("--metrics-enable", action="store_true")
     g.add_argument("--audit-log-path", default="")
     g = run.add_argument_group("walledgarden")
+    g.add_argument("--walled-garden", action="store_true")
     g.add_argument("--walled-garden-portal", default="")
 
     demo = sub.add_parser("demo", help="simulated subscriber lifecycle")
@@ -184,8 +302,47 @@ class BNG:
     def _defer(self, fn):
         self._cleanup.append(fn)
 
+    @staticmethod
+    def _normalize(a):
+        """Fold reference-spelling flags into their canonical attrs
+        (comma lists, -enabled aliases, host:port forms)."""
+        if getattr(a, "radius_servers", ""):
+            a.radius_server += [x for x in a.radius_servers.split(",") if x]
+        if getattr(a, "nat_public_ips", ""):
+            a.nat_public_ip += [x for x in a.nat_public_ips.split(",") if x]
+        if getattr(a, "nat_ports_per_sub", 0):
+            a.nat_ports_per_subscriber = a.nat_ports_per_sub
+        if getattr(a, "bgp_neighbors", ""):
+            a.bgp_neighbor += [x for x in a.bgp_neighbors.split(",") if x]
+        if getattr(a, "peers", ""):
+            a.peer += [x for x in a.peers.split(",") if x]
+        if getattr(a, "slaac_prefixes", ""):
+            if not a.slaac_prefix:
+                a.slaac_prefix = a.slaac_prefixes.split(",")[0]
+        if getattr(a, "dhcpv6_address_pool", ""):
+            a.dhcpv6_na_pool = a.dhcpv6_address_pool
+        if getattr(a, "dhcpv6_prefix_pool", ""):
+            a.dhcpv6_pd_pool = a.dhcpv6_prefix_pool
+        if getattr(a, "ha_peer", "") and not a.ha_partner_url:
+            a.ha_partner_url = a.ha_peer
+        if getattr(a, "metrics_addr", ""):
+            try:
+                a.metrics_port = int(a.metrics_addr.rsplit(":", 1)[-1])
+            except ValueError:
+                pass
+        if getattr(a, "auth_mode", ""):
+            a.nexus_auth = a.auth_mode
+        if getattr(a, "auth_psk", "") and not a.nexus_psk:
+            a.nexus_psk = a.auth_psk
+        if getattr(a, "auth_psk_file", "") and not a.nexus_psk:
+            a.nexus_psk = resolve_secret("", a.auth_psk_file)
+        if getattr(a, "radius_enabled", False) and not a.radius_server:
+            raise SystemExit(
+                "--radius-enabled requires --radius-server(s)")
+        return a
+
     def start(self):
-        a = self.args
+        a = self._normalize(self.args)
         from ..dataplane.launcher import GoldenLauncher, HipLauncher
         from ..dataplane.packets import ip2u32, mac_bytes
 
@@ -255,9 +412,20 @@ class BNG:
             from ..ha.failover import FailoverController
             from ..ha.health_monitor import HealthMonitor
             from ..ha.sync import HASyncer
+            ha_host, ha_port = "127.0.0.1", a.ha_listen_port
+            if a.ha_listen:
+                h, _, pt = a.ha_listen.rpartition(":")
+                ha_host = h or "0.0.0.0"
+                ha_port = int(pt or 0)
             self.ha = HASyncer(a.node_id, a.ha_role,
-                               listen_port=a.ha_listen_port,
-                               partner_url=a.ha_partner_url).start()
+                               listen_port=ha_port,
+                               listen_host=ha_host,
+                               auth_token=a.ha_auth_token,
+                               partner_url=a.ha_partner_url,
+                               tls_cert=a.ha_tls_cert,
+                               tls_key=a.ha_tls_key,
+                               tls_ca=a.ha_tls_ca,
+                               tls_skip_verify=a.ha_tls_skip_verify).start()
             self._defer(self.ha.stop)
             from ..ha import nat_glue, session_glue
             session_glue.attach(self.dhcp_server, self.ha,
@@ -290,7 +458,8 @@ class BNG:
             from ..routing.frr import FakeExecutor, VtyshExecutor
             exe = VtyshExecutor() if os.path.exists("/usr/bin/vtysh") \
                 else FakeExecutor()
-            self.bgp = BGPController(exe, a.bgp_local_as).start()
+            self.bgp = BGPController(exe, a.bgp_local_as,
+                                     router_id=a.bgp_router_id).start()
             for n in a.bgp_neighbor:
                 addr, _, ras = n.partition(":")
                 self.bgp.add_neighbor(addr, int(ras or 65000))
@@ -330,7 +499,9 @@ class BNG:
             from ..radius.accounting import AccountingManager
             from ..radius.client import Client as RadiusClient
             secret = resolve_secret(a.radius_secret, a.radius_secret_file)
-            self.radius = RadiusClient(a.radius_server, secret.encode())
+            self.radius = RadiusClient(a.radius_server, secret.encode(),
+                                       nas_identifier=a.radius_nas_id,
+                                       timeout=a.radius_timeout)
             if a.radius_partition_mode != "off":
                 # partition degradation modes (ref resilience wiring
                 # main.go:1182-1211, radius_handler.go:52)
@@ -392,16 +563,36 @@ class BNG:
             self.dhcp_server.lease_time_provider = \
                 self.pool_monitor.effective_lease_time
 
-        # 10. NAT (main.go:1001-1040)
+        # 10. NAT (main.go:1001-1040); mode flags mirror the
+        # reference's nat-eim/eif/hairpin/alg-* booleans
         if a.nat_enable:
+            from ..dataplane import abi as _abi
             from ..nat.logging import ComplianceLogger
             from ..nat.manager import Manager as NATMgr
             logger = ComplianceLogger(a.nat_log_path or None,
                                       fmt=a.nat_log_format,
                                       bulk_mode=a.nat_bulk_logging)
+            natflags = 0
+            alg_ports = []
+            if a.nat_eim:
+                natflags |= _abi.NAT_FLAG_EIM
+            if a.nat_eif:
+                natflags |= _abi.NAT_FLAG_EIF
+            if a.nat_hairpin:
+                natflags |= _abi.NAT_FLAG_HAIRPIN
+            if a.nat_alg_ftp:
+                natflags |= _abi.NAT_FLAG_ALG_FTP
+                alg_ports.append((21, 6))
+            if a.nat_alg_sip:
+                natflags |= _abi.NAT_FLAG_ALG_SIP
+                alg_ports += [(5060, 17), (5060, 6)]
+            self.launcher.set_nat_config(
+                flags=natflags,
+                ports_per_sub=a.nat_ports_per_subscriber,
+                alg_ports=alg_ports)
             self.nat = NATMgr(self.launcher,
                               ports_per_subscriber=a.nat_ports_per_subscriber,
-                              logger=logger)
+                              flags=natflags, logger=logger)
             for ip in a.nat_public_ip:
                 self.nat.add_public_ip(ip)
             self.nat.start()
@@ -414,6 +605,7 @@ class BNG:
             from ..dataplane.packets import mac_bytes as mb, u32_to_ip
             self.pppoe = PPPoEServer(mb(a.server_mac),
                                      ac_name=a.pppoe_ac_name,
+                                     service_name=a.pppoe_service_name,
                                      auth=a.pppoe_auth)
             if getattr(self, "radius", None):
                 self.pppoe.radius = self.radius
@@ -455,12 +647,20 @@ class BNG:
             self.pppoe.on_session_close = _ppp_close
         if a.dhcpv6_enable:
             from ..dhcpv6.server import DHCPv6Server
-            self.dhcpv6 = DHCPv6Server(na_pool=a.dhcpv6_na_pool,
-                                       pd_pool=a.dhcpv6_pd_pool)
-        if a.slaac_enable and a.slaac_prefix:
+            self.dhcpv6 = DHCPv6Server(
+                na_pool=a.dhcpv6_na_pool, pd_pool=a.dhcpv6_pd_pool,
+                dns=[d for d in a.dhcpv6_dns.split(",") if d])
+        if a.slaac_enable and (a.slaac_prefix or a.slaac_prefixes):
             from ..slaac.radvd import PrefixConfig, RAConfig, Server
+            prefixes = ([p for p in a.slaac_prefixes.split(",") if p]
+                        if a.slaac_prefixes else [a.slaac_prefix])
             self.slaac = Server(RAConfig(
-                prefixes=[PrefixConfig(a.slaac_prefix)])).start()
+                prefixes=[PrefixConfig(p) for p in prefixes],
+                managed=a.slaac_managed, other_config=a.slaac_other,
+                rdnss=[d for d in a.slaac_dns.split(",") if d],
+                dnssl=[d for d in a.slaac_dns_domains.split(",") if d],
+                min_interval=float(a.slaac_min_interval),
+                max_interval=float(a.slaac_max_interval))).start()
             self._defer(self.slaac.stop)
 
         # 12. metrics (main.go:1214-1241)

@@ -25,7 +25,9 @@ class HASyncer:
                  max_backoff: float = 10.0,
                  listen_host: str = "127.0.0.1",
                  auth_token: str = "",
-                 allow_insecure: bool = False):
+                 allow_insecure: bool = False,
+                 tls_cert: str = "", tls_key: str = "", tls_ca: str = "",
+                 tls_skip_verify: bool = False):
         self.node_id = node_id
         self.role = role
         self.store = store or InMemorySessionStore()
@@ -44,13 +46,18 @@ class HASyncer:
         # unless explicitly allowed (the ref offers TLS/mTLS here,
         # sync.go TLS options)
         if listen_host not in ("127.0.0.1", "localhost", "::1") and \
-                not auth_token and not allow_insecure:
+                not auth_token and not tls_cert and not allow_insecure:
             raise ValueError(
-                "non-loopback HA bind requires auth_token (or "
+                "non-loopback HA bind requires auth_token or TLS (or "
                 "allow_insecure=True): session data is replicated "
                 "over this socket")
         self.listen_host = listen_host
         self.auth_token = auth_token
+        # TLS/mTLS (ref sync.go TLS options): cert+key serve HTTPS; a
+        # CA on the server side requires client certs (mTLS); the
+        # standby verifies against tls_ca unless tls_skip_verify
+        self.tls_cert, self.tls_key = tls_cert, tls_key
+        self.tls_ca, self.tls_skip_verify = tls_ca, tls_skip_verify
         self._seq = 0
         self._seq_lock = threading.Lock()
         self._subscribers: List[queue.Queue] = []
@@ -194,6 +201,15 @@ class HASyncer:
 
         self._httpd = ThreadingHTTPServer(
             (self.listen_host, self._listen_port), Handler)
+        if self.tls_cert and self.tls_key:
+            import ssl
+            ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+            ctx.load_cert_chain(self.tls_cert, self.tls_key)
+            if self.tls_ca:                       # mTLS
+                ctx.load_verify_locations(self.tls_ca)
+                ctx.verify_mode = ssl.CERT_REQUIRED
+            self._httpd.socket = ctx.wrap_socket(self._httpd.socket,
+                                                 server_side=True)
         self._listen_port = self._httpd.server_address[1]
         t = threading.Thread(target=self._httpd.serve_forever, daemon=True)
         t.start()
@@ -201,11 +217,21 @@ class HASyncer:
 
     @property
     def url(self) -> str:
-        return f"http://{self.listen_host}:{self._listen_port}"
+        scheme = "https" if self.tls_cert else "http"
+        return f"{scheme}://{self.listen_host}:{self._listen_port}"
 
     def _headers(self) -> dict:
         return {"X-BNG-HA-Token": self.auth_token} if self.auth_token \
             else {}
+
+    def _req_kwargs(self) -> dict:
+        kw = {"headers": self._headers()}
+        if self.partner_url.startswith("https"):
+            kw["verify"] = (False if self.tls_skip_verify
+                            else (self.tls_ca or True))
+            if self.tls_cert and self.tls_key:
+                kw["cert"] = (self.tls_cert, self.tls_key)
+        return kw
 
     # ----------------------------------------------------------- standby
     def _connect_loop(self):
@@ -219,7 +245,7 @@ class HASyncer:
                 backoff = self.reconnect_backoff
                 with requests.get(f"{self.partner_url}/sync/stream",
                                   stream=True, timeout=(3, 30),
-                                  headers=self._headers()) as r:
+                                  **self._req_kwargs()) as r:
                     r.raise_for_status()
                     # stream established (headers => queue registered);
                     # full-sync now so no delta can fall in a gap
@@ -249,7 +275,7 @@ class HASyncer:
     def _full_sync(self):
         import requests
         r = requests.get(f"{self.partner_url}/sync/full", timeout=5,
-                         headers=self._headers())
+                         **self._req_kwargs())
         r.raise_for_status()
         d = r.json()
         msg = SyncMessage.from_dict(d)

@@ -221,3 +221,92 @@ def test_verify_command():
     tree and exits 0."""
     from bng_amd.cli.main import main
     assert main(["verify"]) == 0
+
+
+REF_FLAGS = [
+    # every `bng run` flag of the reference (cmd/bng/main.go:195-424),
+    # with a representative value
+    ("--interface", "eth1"), ("--config", ""), ("--log-level", "info"),
+    ("--bpf-path", "x.bpf.o"), ("--server-ip", "10.0.0.1"),
+    ("--metrics-addr", ":9090"), ("--pool-network", "10.0.1.0/24"),
+    ("--pool-gateway", "10.0.1.1"), ("--pool-dns", "8.8.8.8,8.8.4.4"),
+    ("--lease-time", "24h"), ("--radius-servers", "1.2.3.4:1812"),
+    ("--radius-secret", "s"), ("--radius-secret-file", ""),
+    ("--radius-nas-id", "bng"), ("--radius-timeout", "3s"),
+    ("--radius-enabled", None), ("--qos-bpf-path", "q.bpf.o"),
+    ("--qos-enabled", None), ("--nat-enabled", None),
+    ("--nat-bpf-path", "n.bpf.o"), ("--nat-public-ips", "1.1.1.1"),
+    ("--nat-ports-per-sub", "512"), ("--nat-log-enabled", None),
+    ("--nat-log-path", "/tmp/nat.log"), ("--nat-inside-interface", "e0"),
+    ("--nat-outside-interface", "e1"), ("--nat-eim", "true"),
+    ("--nat-eif", "true"), ("--nat-hairpin", "false"),
+    ("--nat-alg-ftp", "true"), ("--nat-alg-sip", "true"),
+    ("--nat-bulk-logging", None), ("--auth-mode", "psk"),
+    ("--auth-psk", "k"), ("--auth-psk-file", ""),
+    ("--auth-mtls-cert", ""), ("--auth-mtls-key", ""),
+    ("--auth-mtls-ca", ""), ("--auth-mtls-server-name", ""),
+    ("--auth-mtls-insecure", None), ("--dhcpv6-enabled", None),
+    ("--dhcpv6-address-pool", "2001:db8:1::/64"),
+    ("--dhcpv6-prefix-pool", "2001:db8:f::/40"),
+    ("--dhcpv6-dns", "2001:4860:4860::8888"),
+    ("--dhcpv6-domain-search", "example.com"),
+    ("--slaac-enabled", None), ("--slaac-prefixes", "2001:db8:2::/64"),
+    ("--slaac-managed", None), ("--slaac-other", None),
+    ("--slaac-dns", "2001:4860:4860::8888"),
+    ("--slaac-dns-domains", "example.com"),
+    ("--slaac-min-interval", "200s"), ("--slaac-max-interval", "600s"),
+    ("--nexus-url", "http://n"), ("--nexus-pool", "default"),
+    ("--peers", "n2=http://p2"), ("--peer-discovery", "static"),
+    ("--peer-service", ""), ("--node-id", "bng-1"),
+    ("--peer-listen", ":8081"), ("--ha-peer", "http://ha"),
+    ("--ha-role", "active"), ("--ha-listen", ":9000"),
+    ("--ha-tls-cert", ""), ("--ha-tls-key", ""), ("--ha-tls-ca", ""),
+    ("--ha-tls-skip-verify", None),
+    ("--health-check-interval", "5"), ("--health-check-retries", "3"),
+    ("--radius-partition-mode", "cached"),
+    ("--short-lease-enabled", None), ("--short-lease-threshold", "0.9"),
+    ("--short-lease-duration", "5m"), ("--pool-mode", "lease"),
+    ("--epoch-period", "5m"), ("--epoch-grace", "1"),
+    ("--pppoe-enabled", None), ("--pppoe-interface", "eth1"),
+    ("--pppoe-ac-name", "BNG-AC"), ("--pppoe-service-name", "internet"),
+    ("--pppoe-auth-type", "pap"), ("--pppoe-session-timeout", "30m"),
+    ("--bgp-enabled", None), ("--bgp-router-id", "1.1.1.1"),
+    ("--bgp-neighbors", "10.0.0.2:65001"), ("--bgp-bfd-enabled", None),
+    ("--antispoof-mode", "strict"), ("--walled-garden", None),
+    ("--walled-garden-portal", "10.255.255.1:8080"),
+]
+
+
+class TestReferenceFlagParity:
+    def test_every_reference_flag_parses(self):
+        """Full `bng run` flag-surface parity with cmd/bng/main.go
+        (round-1 VERDICT task 8)."""
+        argv = ["run"]
+        for flag, val in REF_FLAGS:
+            if flag == "--config":
+                continue         # needs a real file; covered elsewhere
+            argv.append(flag)
+            if val is not None:
+                argv.append(val)
+        args = build_parser().parse_args(argv)
+        # spot-check alias folding through _normalize
+        from bng_amd.cli.main import BNG
+        a = BNG._normalize(args)
+        assert "1.2.3.4:1812" in a.radius_server
+        assert "1.1.1.1" in a.nat_public_ip
+        assert a.nat_ports_per_subscriber == 512
+        assert a.lease_time == 86400
+        assert a.nat_hairpin is False and a.nat_alg_sip is True
+        assert a.ha_partner_url == "http://ha"
+        assert a.nexus_auth == "psk" and a.nexus_psk == "k"
+        assert "10.0.0.2:65001" in a.bgp_neighbor
+        assert a.dhcpv6_na_pool == "2001:db8:1::/64"
+        assert a.short_lease_duration == 300
+
+    def test_duration_parsing(self):
+        from bng_amd.cli.main import parse_duration
+        assert parse_duration("24h") == 86400
+        assert parse_duration("5m") == 300
+        assert parse_duration("300s") == 300
+        assert parse_duration(42) == 42
+        assert parse_duration("500ms") == 0
