@@ -151,12 +151,28 @@ class DHCPServer:
             except Exception:
                 continue
             if resp is not None:
-                dest = ("255.255.255.255", 68) if req.giaddr == 0 \
-                    else (u32_to_ip(req.giaddr), 67)
+                dest = self.reply_dest(req, resp)
                 try:
                     self._sock.sendto(resp.encode(), dest)
                 except OSError:
                     break
+
+    @staticmethod
+    def reply_dest(req: dm.DHCPMessage,
+                   resp: dm.DHCPMessage) -> tuple:
+        """RFC 2131 §4.1 reply addressing (round-1 advisor: renewing
+        clients were getting broadcast ACKs): relay (giaddr) wins;
+        then unicast to ciaddr for a renewing client; then honor the
+        BROADCAST flag; else unicast to the offered yiaddr (delivered
+        at L2 by chaddr — the GPU fast path makes the same decision in
+        setup_reply_l2_headers, ref dhcp_fastpath.c:436-482)."""
+        if req.giaddr != 0:
+            return (u32_to_ip(req.giaddr), 67)
+        if req.ciaddr != 0:
+            return (u32_to_ip(req.ciaddr), 68)
+        if req.flags & 0x8000 or resp.yiaddr == 0:
+            return ("255.255.255.255", 68)
+        return (u32_to_ip(resp.yiaddr), 68)
 
     # ----------------------------------------------------------- handler
     def handle(self, req: dm.DHCPMessage) -> Optional[dm.DHCPMessage]:

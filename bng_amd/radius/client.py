@@ -61,7 +61,8 @@ class Client:
                  nas_ip: str = "0.0.0.0", nas_identifier: str = "bng",
                  timeout: float = 2.0, retries: int = 2,
                  rate_limit: float = 0.0, rate_burst: int = 100,
-                 acct_port_offset: int = 1):
+                 acct_port_offset: int = 1,
+                 require_message_authenticator: bool = False):
         """servers: ['host:port', ...] — rotated on failure
         (ref client.go:391-403)."""
         self.servers = list(servers)
@@ -77,8 +78,14 @@ class Client:
         if rate_limit > 0:
             for s in self.servers:
                 self._limiters[s] = RateLimiter(rate_limit, rate_burst)
+        # round-1 advisor: the MD5 Response Authenticator alone does not
+        # stop blast-RADIUS (CVE-2024-3596); verify Message-
+        # Authenticator on Access responses when present, and require it
+        # when configured
+        self.require_message_authenticator = require_message_authenticator
         self.stats = {"auth_ok": 0, "auth_reject": 0, "auth_timeout": 0,
-                      "acct_ok": 0, "acct_timeout": 0, "rate_limited": 0}
+                      "acct_ok": 0, "acct_timeout": 0, "rate_limited": 0,
+                      "ma_invalid": 0, "ma_missing": 0}
 
     def _next_ident(self) -> int:
         with self._lock:
@@ -107,14 +114,27 @@ class Client:
                 return None
 
     def _send_with_failover(self, raw: bytes, req_auth: bytes,
-                            port_offset: int = 0) -> Optional[bytes]:
+                            port_offset: int = 0,
+                            check_ma: bool = False) -> Optional[bytes]:
         """Try each server in rotation, retries per server
-        (ref client.go:157-338)."""
+        (ref client.go:157-338).  check_ma (Access exchanges): a
+        response with an INVALID Message-Authenticator, or without one
+        when required, is treated as forged — dropped like a timeout."""
         for server in list(self.servers):
             for _ in range(self.retries):
                 data = self._exchange(raw, server, port_offset)
                 if data is not None and rp.verify_response(
                         data, req_auth, self.secret):
+                    if check_ma:
+                        resp = rp.Packet.decode(data)
+                        if resp.get(rp.MESSAGE_AUTHENTICATOR) is not None:
+                            if not rp.verify_message_authenticator(
+                                    resp, self.secret, req_auth):
+                                self.stats["ma_invalid"] += 1
+                                continue
+                        elif self.require_message_authenticator:
+                            self.stats["ma_missing"] += 1
+                            continue
                     return data
             # rotate the failed server to the back (ref :391-403)
             with self._lock:
@@ -154,7 +174,8 @@ class Client:
         for t, v in (extra_attrs or []):
             pkt.add(t, v)
         rp.sign_message_authenticator(pkt, self.secret)
-        data = self._send_with_failover(pkt.encode(), req_auth)
+        data = self._send_with_failover(pkt.encode(), req_auth,
+                                        check_ma=True)
         if data is None:
             self.stats["auth_timeout"] += 1
             raise RadiusTimeout("no RADIUS server answered")

@@ -194,6 +194,22 @@ def verify_message_authenticator(pkt: Packet, secret: bytes,
                                ma)
 
 
+def sign_response_with_ma(pkt: "Packet", req_auth: bytes,
+                          secret: bytes) -> bytes:
+    """Response with Message-Authenticator (blast-RADIUS / CVE-2024-3596
+    mitigation): MA = HMAC-MD5 over the response with the REQUEST
+    authenticator in the authenticator field and MA zeroed, computed
+    BEFORE the Response Authenticator."""
+    pkt.attributes = [(t, v) for t, v in pkt.attributes
+                      if t != MESSAGE_AUTHENTICATOR]
+    pkt.attributes.append((MESSAGE_AUTHENTICATOR, b"\x00" * 16))
+    clone = Packet(pkt.code, pkt.identifier, req_auth)
+    clone.attributes = list(pkt.attributes)
+    mac = hmac.new(secret, clone.encode(), hashlib.md5).digest()
+    pkt.attributes[-1] = (MESSAGE_AUTHENTICATOR, mac)
+    return sign_response(pkt, req_auth, secret)
+
+
 def response_authenticator(code: int, identifier: int, attrs: bytes,
                            req_auth: bytes, secret: bytes) -> bytes:
     """RFC 2865 §3: MD5(Code+ID+Length+RequestAuth+Attributes+Secret)."""

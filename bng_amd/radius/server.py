@@ -111,6 +111,10 @@ class RadiusServer:
                 resp.add(rp.CLASS, rec["class"])
         else:
             resp.add(rp.REPLY_MESSAGE, "denied")
+        if req.get(rp.MESSAGE_AUTHENTICATOR) is not None:
+            # requests carrying MA get MA'd responses (blast-RADIUS)
+            return rp.sign_response_with_ma(resp, req.authenticator,
+                                            self.secret)
         return rp.sign_response(resp, req.authenticator, self.secret)
 
     def _handle_acct(self, data: bytes) -> Optional[bytes]:
@@ -122,4 +126,8 @@ class RadiusServer:
         with self._lock:
             self.acct_records.append(req)
         resp = rp.Packet(rp.ACCOUNTING_RESPONSE, req.identifier)
+        if req.get(rp.MESSAGE_AUTHENTICATOR) is not None:
+            # requests carrying MA get MA'd responses (blast-RADIUS)
+            return rp.sign_response_with_ma(resp, req.authenticator,
+                                            self.secret)
         return rp.sign_response(resp, req.authenticator, self.secret)

@@ -317,3 +317,52 @@ class TestWalledGardenManager:
         assert st["added"] == 3 and st["blocked"] == 1
         m.remove("aa:03")
         assert m.get_stats()["tracked"] == 2
+
+
+class TestReplyAddressing:
+    """RFC 2131 §4.1 reply-addressing rules (round-1 advisor: the serve
+    loop broadcast every non-relayed reply and ignored ciaddr and the
+    BROADCAST flag)."""
+
+    def _req(self, ciaddr=0, flags=0, giaddr=0):
+        m = dm.DHCPMessage()
+        m.op = 1
+        m.ciaddr = ciaddr
+        m.flags = flags
+        m.giaddr = giaddr
+        return m
+
+    def _resp(self, yiaddr):
+        m = dm.DHCPMessage()
+        m.yiaddr = yiaddr
+        return m
+
+    def test_relay_wins(self):
+        from bng_amd.dhcp.server import DHCPServer
+        d = DHCPServer.reply_dest(
+            self._req(ciaddr=ip2u32("10.0.1.5"), giaddr=ip2u32("10.9.9.9")),
+            self._resp(ip2u32("10.0.1.5")))
+        assert d == ("10.9.9.9", 67)
+
+    def test_renewing_client_unicast_to_ciaddr(self):
+        from bng_amd.dhcp.server import DHCPServer
+        d = DHCPServer.reply_dest(self._req(ciaddr=ip2u32("10.0.1.5")),
+                                  self._resp(ip2u32("10.0.1.5")))
+        assert d == ("10.0.1.5", 68)
+
+    def test_broadcast_flag_honored(self):
+        from bng_amd.dhcp.server import DHCPServer
+        d = DHCPServer.reply_dest(self._req(flags=0x8000),
+                                  self._resp(ip2u32("10.0.1.6")))
+        assert d == ("255.255.255.255", 68)
+
+    def test_default_unicast_to_yiaddr(self):
+        from bng_amd.dhcp.server import DHCPServer
+        d = DHCPServer.reply_dest(self._req(),
+                                  self._resp(ip2u32("10.0.1.7")))
+        assert d == ("10.0.1.7", 68)
+
+    def test_nak_broadcasts(self):
+        from bng_amd.dhcp.server import DHCPServer
+        d = DHCPServer.reply_dest(self._req(), self._resp(0))
+        assert d == ("255.255.255.255", 68)

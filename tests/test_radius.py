@@ -283,3 +283,86 @@ class TestPacketFuzz:
                 rp.Packet.decode(bytes(mutated))
             except (rp.RadiusError, ValueError, IndexError):
                 pass
+
+
+class TestMessageAuthenticatorVerification:
+    """Round-1 advisor: verify Message-Authenticator on Access
+    responses (blast-RADIUS / CVE-2024-3596 client-side mitigation)."""
+
+    def _srv(self):
+        srv = RadiusServer(b"s3cret",
+                           users={"alice": {"password": "pw1"}})
+        srv.start()
+        return srv
+
+    def test_response_carries_and_verifies_ma(self):
+        srv = self._srv()
+        try:
+            c = Client([f"127.0.0.1:{srv.port}"], b"s3cret",
+                       require_message_authenticator=True)
+            res = c.authenticate("alice", "pw1")
+            assert res.success
+            assert c.stats["ma_invalid"] == 0
+            assert c.stats["ma_missing"] == 0
+        finally:
+            srv.stop()
+
+    def test_forged_response_dropped(self):
+        """A response whose MA is tampered must be treated as forged
+        (dropped), not accepted."""
+        import socket as _s
+        import threading as _t
+        from bng_amd.radius import packet as rp
+
+        sock = _s.socket(_s.AF_INET, _s.SOCK_DGRAM)
+        sock.bind(("127.0.0.1", 0))
+        port = sock.getsockname()[1]
+        secret = b"s3cret"
+
+        def forger():
+            data, addr = sock.recvfrom(4096)
+            req = rp.Packet.decode(data)
+            resp = rp.Packet(rp.ACCESS_ACCEPT, req.identifier)
+            raw = rp.sign_response_with_ma(resp, req.authenticator, secret)
+            # tamper with the MA bytes but keep the Response
+            # Authenticator valid for the tampered attrs (the attack:
+            # MD5 response-auth forgery with bogus MA)
+            resp2 = rp.Packet.decode(raw)
+            resp2.attributes = [
+                (t, b"\x00" * 16 if t == rp.MESSAGE_AUTHENTICATOR else v)
+                for t, v in resp2.attributes]
+            raw2 = rp.sign_response(resp2, req.authenticator, secret)
+            sock.sendto(raw2, addr)
+
+        th = _t.Thread(target=forger, daemon=True)
+        th.start()
+        c = Client([f"127.0.0.1:{port}"], secret, retries=1, timeout=0.5)
+        with pytest.raises(RadiusTimeout):
+            c.authenticate("alice", "pw1")
+        assert c.stats["ma_invalid"] == 1
+        sock.close()
+
+    def test_require_ma_rejects_bare_response(self):
+        import socket as _s
+        import threading as _t
+        from bng_amd.radius import packet as rp
+
+        sock = _s.socket(_s.AF_INET, _s.SOCK_DGRAM)
+        sock.bind(("127.0.0.1", 0))
+        port = sock.getsockname()[1]
+        secret = b"s3cret"
+
+        def bare():
+            data, addr = sock.recvfrom(4096)
+            req = rp.Packet.decode(data)
+            resp = rp.Packet(rp.ACCESS_ACCEPT, req.identifier)
+            sock.sendto(rp.sign_response(resp, req.authenticator, secret),
+                        addr)
+
+        _t.Thread(target=bare, daemon=True).start()
+        c = Client([f"127.0.0.1:{port}"], secret, retries=1, timeout=0.5,
+                   require_message_authenticator=True)
+        with pytest.raises(RadiusTimeout):
+            c.authenticate("alice", "pw1")
+        assert c.stats["ma_missing"] >= 1
+        sock.close()
