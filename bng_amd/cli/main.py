@@ -603,10 +603,10 @@ class BNG:
         if a.pppoe_enable:
             from ..pppoe.server import PPPoEServer
             from ..dataplane.packets import mac_bytes as mb, u32_to_ip
-            self.pppoe = PPPoEServer(mb(a.server_mac),
-                                     ac_name=a.pppoe_ac_name,
-                                     service_name=a.pppoe_service_name,
-                                     auth=a.pppoe_auth)
+            self.pppoe = PPPoEServer(
+                mb(a.server_mac), ac_name=a.pppoe_ac_name,
+                service_name=a.pppoe_service_name, auth=a.pppoe_auth,
+                session_timeout=float(a.pppoe_session_timeout))
             if getattr(self, "radius", None):
                 self.pppoe.radius = self.radius
             if a.pool_network:

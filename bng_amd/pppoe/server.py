@@ -100,7 +100,8 @@ class PPPoEServer:
     def __init__(self, server_mac: bytes, ac_name: str = "bng-amd",
                  service_name: str = "", auth: str = "chap",
                  mru: int = 1492, cookie_secret: Optional[bytes] = None,
-                 echo_interval: float = 30.0, echo_fails: int = 3):
+                 echo_interval: float = 30.0, echo_fails: int = 3,
+                 session_timeout: float = 0.0):
         self.server_mac = server_mac
         self.ac_name = ac_name
         self.service_name = service_name
@@ -109,6 +110,9 @@ class PPPoEServer:
         self.cookie_secret = cookie_secret or os.urandom(16)
         self.echo_interval = echo_interval
         self.echo_fails = echo_fails
+        # absolute session lifetime, 0 = unlimited (ref
+        # pppoe-session-timeout flag / session.go lifetime handling)
+        self.session_timeout = session_timeout
         self.sessions: Dict[int, Session] = {}
         self.by_mac: Dict[bytes, int] = {}
         self._next_sid = 1
@@ -635,6 +639,12 @@ class PPPoEServer:
         for s in pending_lcp:
             out += self._send_lcp_req(s)
         for s in open_sessions:
+            if self.session_timeout and \
+                    now - s.created >= self.session_timeout:
+                self.stats["session_timeout"] = \
+                    self.stats.get("session_timeout", 0) + 1
+                out += self._teardown(s)
+                continue
             if s.echo_pending >= self.echo_fails:
                 self.stats["echo_timeout"] += 1
                 out += self._teardown(s)

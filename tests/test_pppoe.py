@@ -560,3 +560,19 @@ class TestLcpBargaining:
         srv._auth_fails[mac] = [_t.time() - srv.AUTH_FAIL_WINDOW - 1] * 3
         out = srv.handle_frame(padr)
         assert C.DiscoveryPacket.decode(out[0]).session_id != 0
+
+
+def test_session_timeout_teardown():
+    import time
+    """--pppoe-session-timeout: open sessions past the absolute
+    lifetime are torn down on tick."""
+    srv = PPPoEServer(SRV_MAC, auth="none", session_timeout=600)
+    c = SimClient(srv)
+    c.discover()
+    sid = c.session_id
+    assert sid in srv.sessions and srv.sessions[sid].phase == "open"
+    srv.tick(now=time.time() + 300)      # inside lifetime: stays
+    assert sid in srv.sessions
+    out = srv.tick(now=time.time() + 601)
+    assert sid not in srv.sessions
+    assert srv.stats.get("session_timeout") == 1
