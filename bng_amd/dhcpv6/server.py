@@ -132,6 +132,7 @@ class DHCPv6Server:
         self.na_alloc = BitmapAllocator(na_pool, 128, reserve_head=1)
         self.pd_alloc = BitmapAllocator(pd_pool, pd_prefix_len)
         self.dns = dns or []
+        self._declined: dict = {}   # addr -> quarantine expiry
         self.preferred = preferred_lifetime
         self.valid = valid_lifetime
         self.rapid_commit = rapid_commit
@@ -159,6 +160,7 @@ class DHCPv6Server:
         h = {SOLICIT: self._solicit, REQUEST: self._request,
              RENEW: self._renew, REBIND: self._renew,
              RELEASE: self._release, CONFIRM: self._confirm,
+             DECLINE: self._decline,
              INFORMATION_REQUEST: self._inforeq}.get(msg.msg_type)
         if h is None:
             return None
@@ -183,8 +185,22 @@ class DHCPv6Server:
             iaid, _, _, _ = parse_ia(body)
             try:
                 with self._lock:
-                    prefix = self.na_alloc.allocate(f"{sub_key}/{iaid}")
-                addr = prefix.split("/")[0]
+                    # declined (conflicted) addresses are quarantined:
+                    # salt the allocation key until we land outside the
+                    # quarantine set (RFC 8415 §18.3.8; the v4 pool has
+                    # the same decline blacklist)
+                    addr = None
+                    for salt in range(4):
+                        key = (f"{sub_key}/{iaid}" if salt == 0
+                               else f"{sub_key}/{iaid}/d{salt}")
+                        prefix = self.na_alloc.allocate(key)
+                        cand = prefix.split("/")[0]
+                        if not self._is_declined(cand):
+                            addr = cand
+                            break
+                        self.na_alloc.release(key)
+                    if addr is None:
+                        raise PoolExhaustedError("quarantined")
                 if commit:
                     self.bindings[(duid, iaid, False)] = Binding(
                         duid, iaid, addr, time.time() + self.valid)
@@ -291,6 +307,45 @@ class DHCPv6Server:
         self.stats["reply"] += 1
         resp = self._base_reply(msg, REPLY, duid)
         resp.add(OPT_STATUS_CODE, struct.pack(">H", STATUS_SUCCESS) + b"ok")
+        return resp
+
+    DECLINE_QUARANTINE = 3600.0
+
+    def _is_declined(self, addr: str) -> bool:
+        exp = self._declined.get(addr)
+        if exp is None:
+            return False
+        if time.time() >= exp:
+            del self._declined[addr]
+            return False
+        return True
+
+    def _decline(self, msg, duid):
+        """Client detected an address conflict (RFC 8415 §18.3.8):
+        release the binding AND quarantine the address so the next
+        allocation doesn't hand the conflicted address straight back
+        (the reference releases only, dhcpv6/server.go:684-693; the v4
+        pool's decline blacklist is the model here)."""
+        self.stats["decline"] = self.stats.get("decline", 0) + 1
+        sub_key = duid.hex()
+        import ipaddress
+        for body in msg.get_all(OPT_IA_NA):
+            iaid, _, _, subs = parse_ia(body)
+            with self._lock:
+                b = self.bindings.pop((duid, iaid, False), None)
+                if b is not None:
+                    self._declined[b.value] = (time.time() +
+                                               self.DECLINE_QUARANTINE)
+                for t, sub in subs:
+                    if t == OPT_IAADDR and len(sub) >= 16:
+                        a = str(ipaddress.IPv6Address(sub[:16]))
+                        self._declined[a] = (time.time() +
+                                             self.DECLINE_QUARANTINE)
+                self.na_alloc.release(f"{sub_key}/{iaid}")
+        resp = self._base_reply(msg, REPLY, duid)
+        resp.add(OPT_STATUS_CODE,
+                 struct.pack(">H", STATUS_SUCCESS) + b"declined")
+        self.stats["reply"] += 1
         return resp
 
     OPT_RELAY_MSG = 9

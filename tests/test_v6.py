@@ -186,3 +186,52 @@ class TestDHCPv6Relay:
         assert rep.msg_type == REPLY
         assert extract_addr(rep) is not None
         assert (DUID, 1, False) in srv.bindings
+
+
+class TestDecline:
+    """RFC 8415 §18.3.8 Decline: conflicted addresses are quarantined
+    and the next allocation avoids them (the reference only releases,
+    dhcpv6/server.go:684-693; the v4 decline blacklist is the model)."""
+
+    def _solicit_request(self, srv, duid, iaid=1):
+        from bng_amd.dhcpv6 import server as srv_mod
+        from bng_amd.dhcpv6.server import (DHCPv6Message, OPT_CLIENTID,
+                                           OPT_IA_NA, REQUEST, SOLICIT,
+                                           parse_ia, OPT_IAADDR)
+        import struct as st
+        m = DHCPv6Message(REQUEST, 0x111)
+        m.add(OPT_CLIENTID, duid)
+        m.add(srv_mod.OPT_SERVERID, srv.server_duid)
+        m.add(OPT_IA_NA, st.pack(">III", iaid, 0, 0))
+        resp = DHCPv6Message.decode(srv.handle(m.encode()))
+        body = resp.get(OPT_IA_NA)
+        _, _, _, subs = parse_ia(body)
+        for t, sub in subs:
+            if t == OPT_IAADDR:
+                import ipaddress
+                return str(ipaddress.IPv6Address(sub[:16]))
+        return None
+
+    def test_declined_address_not_reoffered(self):
+        from bng_amd.dhcpv6.server import (DECLINE, DHCPv6Message,
+                                           DHCPv6Server, OPT_CLIENTID,
+                                           OPT_IA_NA, encode_ia_na)
+        import struct as st
+        srv = DHCPv6Server(rapid_commit=False)
+        duid = b"\x00\x01duid-x"
+        addr1 = self._solicit_request(srv, duid)
+        assert addr1 is not None
+        # client detects a conflict and declines
+        d = DHCPv6Message(DECLINE, 0x222)
+        d.add(OPT_CLIENTID, duid)
+        d.add(OPT_IA_NA, encode_ia_na(1, 0, 0, [(addr1, 0, 0)]))
+        resp = srv.handle(d.encode())
+        assert resp is not None
+        assert srv.stats["decline"] == 1
+        assert srv._is_declined(addr1)
+        # a new request must get a DIFFERENT address
+        addr2 = self._solicit_request(srv, duid)
+        assert addr2 is not None and addr2 != addr1
+        # quarantine expires
+        srv._declined[addr1] = 0.0
+        assert not srv._is_declined(addr1)
