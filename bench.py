@@ -343,7 +343,7 @@ def service_latency(launcher, args, now_sec, reps=128, batch=256,
     lat_np, lens_np = gen_batch(batch, args.subs, 1.0, args.stride,
                                 seed=881)
     svc = DhcpService(launcher, n_slots=max(256, batch),
-                      stride=args.stride)
+                      stride=args.stride, idle_exit_k=400_000)
     try:
         svc.serve(lat_np, lens_np, now_sec)          # warm
         if flood_steps and flood_fn is not None:

@@ -254,7 +254,8 @@ class SvcCtrl(C.Structure):
                 ("run", C.c_uint32), ("n_pkts", C.c_uint32),
                 ("now_sec", C.c_uint64), ("stride", C.c_uint32),
                 ("idle_exit_k", C.c_uint32), ("served", C.c_uint64),
-                ("batches", C.c_uint64), ("_pad", C.c_uint8 * 16)]
+                ("batches", C.c_uint64), ("exited", C.c_uint32),
+                ("_pad", C.c_uint8 * 12)]
 
 
 EXPECTED_SIZES = {

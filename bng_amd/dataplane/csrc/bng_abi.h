@@ -382,7 +382,10 @@ typedef struct bng_svc_ctrl {
                            — the box-safety bound                     */
   uint64_t served;      /* device: total packets served               */
   uint64_t batches;     /* device: total batches served               */
-  uint8_t  _pad[16];
+  uint32_t exited;      /* device: set on kernel exit (host watches
+                           this so stop() can never block on a dead
+                           doorbell)                                  */
+  uint8_t  _pad[12];
 } bng_svc_ctrl;  /* 64 B */
 
 #ifdef __cplusplus

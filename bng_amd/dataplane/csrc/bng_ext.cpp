@@ -298,10 +298,35 @@ void shard_owner(torch::Tensor data, torch::Tensor in_len,
                          (int)data.size(1), (int)n_shards, cur_stream());
 }
 
-void check_pinned(const torch::Tensor& t, const char* name) {
-  TORCH_CHECK(!t.is_cuda() && t.is_pinned(), name,
-              " must be a pinned host tensor");
-  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+/* Fine-grained-coherent pinned host memory for the persistent-service
+ * doorbell/rings.  torch's pin_memory allocates COARSE-grained host
+ * memory on ROCm: a running kernel caches it and never observes host
+ * stores (measured on hardware: the doorbell kernel spun forever) —
+ * the service requires hipHostMallocCoherent. */
+torch::Tensor alloc_pinned_coherent(int64_t nbytes) {
+  void* p = nullptr;
+  hipError_t e = hipHostMalloc(&p, (size_t)nbytes,
+                               hipHostMallocMapped | hipHostMallocCoherent);
+  TORCH_CHECK(e == hipSuccess, "hipHostMalloc(coherent) failed: ",
+              hipGetErrorString(e));
+  memset(p, 0, (size_t)nbytes);
+  return torch::from_blob(p, {nbytes},
+                          [](void* q) { (void)hipHostFree(q); },
+                          torch::TensorOptions().dtype(torch::kUInt8));
+}
+
+/* The device-visible alias of a pinned host pointer (same address
+ * under ROCm unified addressing, but ask the runtime rather than
+ * assume). */
+void* dev_ptr_of(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(!t.is_cuda() && t.is_contiguous(), name,
+              " must be a contiguous host tensor");
+  void* dp = nullptr;
+  hipError_t e = hipHostGetDevicePointer(&dp, t.data_ptr(), 0);
+  TORCH_CHECK(e == hipSuccess, name,
+              " is not device-mapped pinned memory (allocate via "
+              "alloc_pinned_coherent): ", hipGetErrorString(e));
+  return dp;
 }
 
 /* Persistent DHCP service: launch the device-resident kernel on its own
@@ -312,30 +337,29 @@ hipStream_t g_svc_stream = nullptr;
 void dhcp_service_start(torch::Tensor ctrl, torch::Tensor req,
                         torch::Tensor in_len, torch::Tensor out_len,
                         torch::Tensor verdict, torch::Tensor scratch,
+                        int64_t n_slots_arg,
                         torch::Tensor subs, torch::Tensor pools,
                         torch::Tensor cfg, torch::Tensor stats) {
-  check_pinned(ctrl, "ctrl"); check_pinned(req, "req");
-  check_pinned(in_len, "in_len"); check_pinned(out_len, "out_len");
-  check_pinned(verdict, "verdict");
   check_dev(scratch, "scratch"); check_dev(subs, "subs");
   check_dev(pools, "pools"); check_dev(cfg, "cfg");
   check_dev(stats, "stats");
   TORCH_CHECK(ctrl.numel() * ctrl.element_size() ==
               (long)sizeof(bng_svc_ctrl), "ctrl must be 64 bytes");
-  int n_slots = (int)req.size(0);
-  TORCH_CHECK(in_len.numel() == n_slots && out_len.numel() == n_slots &&
-              verdict.numel() == n_slots, "slot count mismatch");
-  TORCH_CHECK(scratch.numel() == req.numel(), "scratch size mismatch");
+  int n_slots = (int)n_slots_arg;
   if (!g_svc_stream)
     (void)hipStreamCreateWithFlags(&g_svc_stream, hipStreamNonBlocking);
   bng_launch_dhcp_service(
-      ctrl.data_ptr(), req.data_ptr(), in_len.data_ptr(),
-      out_len.data_ptr(), verdict.data_ptr(), scratch.data_ptr(), n_slots,
+      dev_ptr_of(ctrl, "ctrl"), dev_ptr_of(req, "req"),
+      dev_ptr_of(in_len, "in_len"), dev_ptr_of(out_len, "out_len"),
+      dev_ptr_of(verdict, "verdict"), scratch.data_ptr(), n_slots,
       subs.data_ptr(), table_mask(subs, sizeof(bng_sub_entry), "subs"),
       pools.data_ptr(),
       (uint32_t)(pools.numel() * pools.element_size() /
                  sizeof(bng_ip_pool)),
       cfg.data_ptr(), stats.data_ptr(), g_svc_stream);
+  hipError_t e = hipGetLastError();
+  TORCH_CHECK(e == hipSuccess, "dhcp_service launch failed: ",
+              hipGetErrorString(e));
 }
 
 void dhcp_service_join() {
@@ -408,6 +432,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("nat_sweep", &nat_sweep);
   m.def("shard_owner", &shard_owner);
   m.def("dhcp_service_start", &dhcp_service_start);
+  m.def("alloc_pinned_coherent", &alloc_pinned_coherent);
   m.def("dhcp_service_join", &dhcp_service_join);
   m.def("layout_report", &layout_report);
 }

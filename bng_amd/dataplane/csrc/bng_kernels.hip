@@ -734,8 +734,9 @@ __global__ void __launch_bounds__(256, 1) dhcp_service_kernel(
   if (tid == 0) {
     ctrl->served = served;
     ctrl->batches = batches;
-    /* mark exit so the host can distinguish idle-reap from running */
-    __hip_atomic_store(&ctrl->run, 0u, __ATOMIC_RELEASE,
+    /* exit ack: the host's stop() waits on this instead of a stream
+     * sync, so a dead doorbell can never block the control plane */
+    __hip_atomic_store(&ctrl->exited, 1u, __ATOMIC_RELEASE,
                        __HIP_MEMORY_SCOPE_SYSTEM);
   }
 }

@@ -565,31 +565,33 @@ class DhcpService:
     the throughput path."""
 
     def __init__(self, launcher: "HipLauncher", n_slots: int = 2048,
-                 stride: int = 512, idle_exit_k: int = 4_000_000):
+                 stride: int = 512, idle_exit_k: int = 1_000_000):
         import numpy as np
         torch = launcher.torch
         self.l = launcher
         self.n_slots, self.stride = n_slots, stride
-        self.ctrl_t = torch.zeros(64, dtype=torch.uint8, pin_memory=True)
-        self.req = torch.zeros((n_slots, stride), dtype=torch.uint8,
-                               pin_memory=True)
-        self.in_len = torch.zeros(n_slots, dtype=torch.int16,
-                                  pin_memory=True)
-        self.out_len = torch.zeros(n_slots, dtype=torch.int16,
-                                   pin_memory=True)
-        self.verdict = torch.zeros(n_slots, dtype=torch.uint8,
-                                   pin_memory=True)
+        # doorbell + rings MUST be fine-grained-coherent pinned memory
+        # (torch pin_memory is coarse-grained on ROCm: the resident
+        # kernel would cache it and never see host stores — measured as
+        # a dead doorbell on hardware)
+        alloc = launcher.ext.alloc_pinned_coherent
+        self.ctrl_t = alloc(64)
+        self.req = alloc(n_slots * stride).view(n_slots, stride)
+        self.in_len = alloc(n_slots * 2).view(torch.int16)
+        self.out_len = alloc(n_slots * 2).view(torch.int16)
+        self.verdict = alloc(n_slots)
         self.scratch = torch.empty((n_slots, stride), dtype=torch.uint8,
                                    device=launcher.device)
         self.c = abi.SvcCtrl.from_address(self.ctrl_t.data_ptr())
         self.c.stride = stride
-        # idle self-exit: ~0.25us/poll -> default reaps after ~17 min
+        # idle self-exit: ~0.25us/poll -> default reaps after ~4 min
+        # (the box-safety bound if the owner dies without stop())
         self.c.idle_exit_k = idle_exit_k
         self.c.run = 1
         launcher.ext.dhcp_service_start(
             self.ctrl_t, self.req, self.in_len, self.out_len,
-            self.verdict, self.scratch, launcher.subs, launcher.pools,
-            launcher.server_cfg, launcher.dhcp_stats)
+            self.verdict, self.scratch, n_slots, launcher.subs,
+            launcher.pools, launcher.server_cfg, launcher.dhcp_stats)
         self._np = np
 
     @property
@@ -615,7 +617,7 @@ class DhcpService:
         while self.c.tail != target:
             spins += 1
             if (spins & 0x3FFF) == 0:
-                if not self.c.run:
+                if self.c.exited:
                     raise RuntimeError("dhcp service kernel exited")
                 if time.perf_counter() > deadline:
                     raise TimeoutError("dhcp service timeout")
@@ -626,15 +628,29 @@ class DhcpService:
         return {"served": int(self.c.served),
                 "batches": int(self.c.batches)}
 
-    def stop(self):
+    def stop(self, timeout: float = 3.0):
+        """Never blocks unboundedly: wait for the kernel's exit ack,
+        and if the doorbell is dead leave the kernel to its idle
+        self-exit instead of hanging on a stream sync."""
         self.c.run = 0
+        deadline = time.perf_counter() + timeout
+        while not self.c.exited:
+            if time.perf_counter() > deadline:
+                raise RuntimeError(
+                    "dhcp service kernel did not ack stop; doorbell "
+                    "dead — kernel left to idle self-exit")
+            time.sleep(0.001)
         self.l.ext.dhcp_service_join()
 
     def __enter__(self):
         return self
 
     def __exit__(self, *exc):
-        self.stop()
+        try:
+            self.stop()
+        except RuntimeError:
+            if exc[0] is None:     # don't mask the original exception
+                raise
 
 
 class GoldenLauncher:

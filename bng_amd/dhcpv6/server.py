@@ -185,22 +185,9 @@ class DHCPv6Server:
             iaid, _, _, _ = parse_ia(body)
             try:
                 with self._lock:
-                    # declined (conflicted) addresses are quarantined:
-                    # salt the allocation key until we land outside the
-                    # quarantine set (RFC 8415 §18.3.8; the v4 pool has
-                    # the same decline blacklist)
-                    addr = None
-                    for salt in range(4):
-                        key = (f"{sub_key}/{iaid}" if salt == 0
-                               else f"{sub_key}/{iaid}/d{salt}")
-                        prefix = self.na_alloc.allocate(key)
-                        cand = prefix.split("/")[0]
-                        if not self._is_declined(cand):
-                            addr = cand
-                            break
-                        self.na_alloc.release(key)
-                    if addr is None:
-                        raise PoolExhaustedError("quarantined")
+                    self._expire_quarantine()
+                    prefix = self.na_alloc.allocate(f"{sub_key}/{iaid}")
+                    addr = prefix.split("/")[0]
                 if commit:
                     self.bindings[(duid, iaid, False)] = Binding(
                         duid, iaid, addr, time.time() + self.valid)
@@ -311,14 +298,29 @@ class DHCPv6Server:
 
     DECLINE_QUARANTINE = 3600.0
 
+    def _quarantine(self, addr: str):
+        """Hold a conflicted address out of the pool by parking it on a
+        sentinel allocation; lazily released when the quarantine
+        expires (the bitmap itself enforces the blacklist)."""
+        self._declined[addr] = time.time() + self.DECLINE_QUARANTINE
+        try:
+            self.na_alloc.allocate_specific(f"__declined__/{addr}",
+                                            f"{addr}/128")
+        except (PoolExhaustedError, ValueError, KeyError):
+            pass
+
     def _is_declined(self, addr: str) -> bool:
         exp = self._declined.get(addr)
-        if exp is None:
-            return False
-        if time.time() >= exp:
+        return exp is not None and time.time() < exp
+
+    def _expire_quarantine(self):
+        now = time.time()
+        for addr in [a for a, e in self._declined.items() if now >= e]:
             del self._declined[addr]
-            return False
-        return True
+            try:
+                self.na_alloc.release(f"__declined__/{addr}")
+            except KeyError:
+                pass
 
     def _decline(self, msg, duid):
         """Client detected an address conflict (RFC 8415 §18.3.8):
@@ -333,15 +335,14 @@ class DHCPv6Server:
             iaid, _, _, subs = parse_ia(body)
             with self._lock:
                 b = self.bindings.pop((duid, iaid, False), None)
+                self.na_alloc.release(f"{sub_key}/{iaid}")
                 if b is not None:
-                    self._declined[b.value] = (time.time() +
-                                               self.DECLINE_QUARANTINE)
+                    self._quarantine(b.value)
                 for t, sub in subs:
                     if t == OPT_IAADDR and len(sub) >= 16:
                         a = str(ipaddress.IPv6Address(sub[:16]))
-                        self._declined[a] = (time.time() +
-                                             self.DECLINE_QUARANTINE)
-                self.na_alloc.release(f"{sub_key}/{iaid}")
+                        if not self._is_declined(a):
+                            self._quarantine(a)
         resp = self._base_reply(msg, REPLY, duid)
         resp.add(OPT_STATUS_CODE,
                  struct.pack(">H", STATUS_SUCCESS) + b"declined")

@@ -113,7 +113,8 @@ def _svc_latency(l, n_subs, batch=256, reps=120):
     NOW = 1_700_000_000
     d_np, ln = bench.gen_batch(batch, n_subs, 1.0, 512, 9)
     lats = []
-    with DhcpService(l, n_slots=max(256, batch)) as svc:
+    with DhcpService(l, n_slots=max(256, batch),
+                 idle_exit_k=400_000) as svc:
         svc.serve(d_np, ln, NOW)
         for _ in range(reps):
             t = time.perf_counter()

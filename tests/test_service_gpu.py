@@ -63,7 +63,7 @@ def test_service_differential_vs_golden(launcher):
     g = _golden_twin()
     golden = g.process_dhcp([bytes(data[i, :lens[i]])
                              for i in range(len(lens))], now_sec=NOW)
-    with DhcpService(launcher, n_slots=256) as svc:
+    with DhcpService(launcher, n_slots=256, idle_exit_k=120_000) as svc:
         v, ol, rep = svc.serve(data, lens, NOW)
         for i, (gv, gfr) in enumerate(golden):
             assert v[i] == gv, f"verdict mismatch pkt {i}"
@@ -87,7 +87,7 @@ def test_service_coherence_across_batches(launcher):
     kernel must see the fresh bytes every time (PCIe-coherent pinned
     UMEM assumption validated here)."""
     from bng_amd.dataplane.launcher import DhcpService
-    with DhcpService(launcher, n_slots=64) as svc:
+    with DhcpService(launcher, n_slots=64, idle_exit_k=120_000) as svc:
         for r in range(20):
             data, lens = _batch(64, seed=100 + r)
             v, ol, rep = svc.serve(data, lens, NOW)
@@ -115,7 +115,7 @@ def test_service_latency_quiesced_and_flooded(launcher):
     fl_np, fl_lens = bench.gen_batch(1 << 20, 1 << 20, 0.0, 512, seed=8)
     fl = torch.from_numpy(fl_np).cuda()
     fll = torch.from_numpy(fl_lens.view(np.int16)).cuda()
-    with DhcpService(launcher, n_slots=256) as svc:
+    with DhcpService(launcher, n_slots=256, idle_exit_k=120_000) as svc:
         svc.serve(data, lens, NOW)
         lat_q = []
         for _ in range(64):
@@ -145,6 +145,6 @@ def test_service_stop_restarts_cleanly(launcher):
     from bng_amd.dataplane.launcher import DhcpService
     data, lens = _batch(32, seed=11)
     for _ in range(3):
-        with DhcpService(launcher, n_slots=64) as svc:
+        with DhcpService(launcher, n_slots=64, idle_exit_k=120_000) as svc:
             v, _, _ = svc.serve(data, lens, NOW)
             assert (v == abi.TX).all()
