@@ -179,8 +179,8 @@ class NatConfig(C.Structure):
                 ("default_ports_per_sub", C.c_uint32),
                 ("n_private_ranges", C.c_uint32), ("n_alg_ports", C.c_uint32),
                 ("_pad", C.c_uint32),
-                ("private_net", C.c_uint32 * MAX_PRIVATE_RANGES),
-                ("private_mask", C.c_uint32 * MAX_PRIVATE_RANGES),
+                ("priv_lo", C.c_uint32 * MAX_PRIVATE_RANGES),
+                ("priv_hi", C.c_uint32 * MAX_PRIVATE_RANGES),
                 ("alg_key", C.c_uint32 * MAX_ALG_PORTS)]
 
 
@@ -211,8 +211,8 @@ class BindingEntry(C.Structure):
 class AntispoofConfig(C.Structure):
     _fields_ = [("default_mode", C.c_uint8), ("log_violations", C.c_uint8),
                 ("_pad", C.c_uint16), ("n_allowed_ranges", C.c_uint32),
-                ("allowed_net", C.c_uint32 * MAX_ALLOWED_RANGES),
-                ("allowed_mask", C.c_uint32 * MAX_ALLOWED_RANGES)]
+                ("allowed_lo", C.c_uint32 * MAX_ALLOWED_RANGES),
+                ("allowed_hi", C.c_uint32 * MAX_ALLOWED_RANGES)]
 
 
 class SpoofEvent(C.Structure):
@@ -338,3 +338,18 @@ def eim_sig(internal_ip: int, internal_port: int, proto: int) -> int:
     if s == KEY_TOMBSTONE:
         s -= 2
     return s
+
+
+def prefixes_to_intervals(ranges):
+    """[(net, mask)] host-order prefixes -> sorted merged [lo, hi]
+    intervals (the launcher-side fold that turns the reference's LPM
+    membership maps into the GPU's binary-searchable tables)."""
+    iv = sorted((net & mask, (net & mask) | (~mask & 0xFFFFFFFF))
+                for net, mask in ranges)
+    out = []
+    for lo, hi in iv:
+        if out and lo <= out[-1][1] + 1:
+            out[-1][1] = max(out[-1][1], hi)
+        else:
+            out.append([lo, hi])
+    return out

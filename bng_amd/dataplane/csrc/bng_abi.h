@@ -227,11 +227,13 @@ typedef struct bng_nat_config {
   uint32_t n_private_ranges;
   uint32_t n_alg_ports;
   uint32_t _pad;
-  /* private ranges as {network(be), mask(be)} pairs; ref uses an LPM trie
-   * (nat44.c:314-320) — at <=64 ranges a linear scan of an L2-resident
-   * array beats a trie on GPU. */
-  uint32_t private_net[BNG_MAX_PRIVATE_RANGES];
-  uint32_t private_mask[BNG_MAX_PRIVATE_RANGES];
+  /* private ranges as SORTED, MERGED host-order intervals [lo, hi];
+   * the ref uses an LPM trie for a membership test (nat44.c:314-320) —
+   * the GPU equivalent is a binary search over an L2-resident interval
+   * table (the launcher folds {net, mask} prefixes into intervals; the
+   * cap matches the ref trie's max_entries sizing). */
+  uint32_t priv_lo[BNG_MAX_PRIVATE_RANGES];
+  uint32_t priv_hi[BNG_MAX_PRIVATE_RANGES];
   uint32_t alg_key[BNG_MAX_ALG_PORTS];   /* port<<16 | proto (host order)  */
 } bng_nat_config;
 
@@ -307,8 +309,10 @@ typedef struct bng_antispoof_config {  /* ref antispoof.c:79-83 */
   uint8_t  log_violations;
   uint16_t _pad;
   uint32_t n_allowed_ranges;
-  uint32_t allowed_net[BNG_MAX_ALLOWED_RANGES];
-  uint32_t allowed_mask[BNG_MAX_ALLOWED_RANGES];
+  /* sorted merged host-order intervals, like bng_nat_config (ref
+   * allowed_ranges_v4 LPM trie, antispoof.c:113-119) */
+  uint32_t allowed_lo[BNG_MAX_ALLOWED_RANGES];
+  uint32_t allowed_hi[BNG_MAX_ALLOWED_RANGES];
 } bng_antispoof_config;
 
 enum bng_antispoof_stat {  /* ref antispoof_stats antispoof.c:58-65 */

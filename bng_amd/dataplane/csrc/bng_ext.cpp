@@ -388,10 +388,10 @@ py::dict layout_report() {
   off["qos_bucket.tokens"] = offsetof(bng_qos_bucket, tokens);
   off["qos_bucket.last_update"] = offsetof(bng_qos_bucket, last_update);
   off["binding_entry.ipv6_addr"] = offsetof(bng_binding_entry, ipv6_addr);
-  off["nat_config.private_net"] = offsetof(bng_nat_config, private_net);
+  off["nat_config.priv_lo"] = offsetof(bng_nat_config, priv_lo);
   off["nat_config.alg_key"] = offsetof(bng_nat_config, alg_key);
-  off["antispoof_config.allowed_net"] =
-      offsetof(bng_antispoof_config, allowed_net);
+  off["antispoof_config.allowed_lo"] =
+      offsetof(bng_antispoof_config, allowed_lo);
   off["spoof_event.spoofed_ip"] = offsetof(bng_spoof_event, spoofed_ip);
   d["offsets"] = off;
   return d;

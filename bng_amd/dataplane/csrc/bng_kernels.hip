@@ -764,10 +764,23 @@ struct nat_flags {
   }
 };
 
-BNG_DEV bool nat_is_private(const bng_nat_config* cfg, uint32_t ip) {
-  for (uint32_t i = 0; i < cfg->n_private_ranges; ++i)
-    if ((ip & cfg->private_mask[i]) == cfg->private_net[i]) return true;
+/* membership over sorted disjoint intervals: binary search, O(log n)
+ * (the LPM-trie equivalent for a membership query) */
+BNG_DEV bool ip_in_intervals(const uint32_t* lo, const uint32_t* hi,
+                             uint32_t n, uint32_t ip) {
+  uint32_t l = 0, r = n;
+  while (l < r) {
+    uint32_t m = (l + r) >> 1;
+    if (ip < lo[m]) r = m;
+    else if (ip > hi[m]) l = m + 1;
+    else return true;
+  }
   return false;
+}
+
+BNG_DEV bool nat_is_private(const bng_nat_config* cfg, uint32_t ip) {
+  return ip_in_intervals(cfg->priv_lo, cfg->priv_hi,
+                         cfg->n_private_ranges, ip);
 }
 
 BNG_DEV bool nat_is_hairpin(const nat_tables& T, uint32_t ip) {
@@ -1387,10 +1400,8 @@ BNG_DEV int antispoof_process_with(uint8_t* p, int len,
       if (mode == BNG_AS_STRICT || mode == BNG_AS_LOG_ONLY)
         allowed = (src_ip == bound_ip);
     } else if (mode == BNG_AS_LOOSE) {
-      for (uint32_t i = 0; i < cfg->n_allowed_ranges; ++i)
-        if ((src_ip & cfg->allowed_mask[i]) == cfg->allowed_net[i]) {
-          allowed = true; break;
-        }
+      allowed = ip_in_intervals(cfg->allowed_lo, cfg->allowed_hi,
+                                cfg->n_allowed_ranges, src_ip);
     }
     if (!allowed) {
       if (cfg->log_violations) {

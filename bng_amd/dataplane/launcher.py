@@ -196,13 +196,14 @@ class HipLauncher:
         from .golden import DEFAULT_PRIVATE_RANGES
         pr = list(private_ranges) if private_ranges is not None \
             else list(DEFAULT_PRIVATE_RANGES)
+        iv = abi.prefixes_to_intervals(pr)
         c = abi.NatConfig(flags=flags, port_range_start=port_range[0],
                           port_range_end=port_range[1],
                           default_ports_per_sub=ports_per_sub,
-                          n_private_ranges=len(pr), n_alg_ports=len(alg_ports))
-        for i, (net, mask) in enumerate(pr):
-            c.private_net[i] = net
-            c.private_mask[i] = mask
+                          n_private_ranges=len(iv), n_alg_ports=len(alg_ports))
+        for i, (lo, hi) in enumerate(iv):
+            c.priv_lo[i] = lo
+            c.priv_hi[i] = hi
         for i, (port, proto) in enumerate(alg_ports):
             c.alg_key[i] = (port << 16) | proto
         self.nat_cfg.copy_(self._to_dev(_struct_bytes(c)))
@@ -303,12 +304,13 @@ class HipLauncher:
     def set_antispoof_config(self, default_mode: int = abi.AS_DISABLED,
                              log_violations: bool = False,
                              allowed_ranges: Sequence[Tuple[int, int]] = ()):
+        iv = abi.prefixes_to_intervals(list(allowed_ranges))
         c = abi.AntispoofConfig(default_mode=default_mode,
                                 log_violations=1 if log_violations else 0,
-                                n_allowed_ranges=len(allowed_ranges))
-        for i, (net, mask) in enumerate(allowed_ranges):
-            c.allowed_net[i] = net
-            c.allowed_mask[i] = mask
+                                n_allowed_ranges=len(iv))
+        for i, (lo, hi) in enumerate(iv):
+            c.allowed_lo[i] = lo
+            c.allowed_hi[i] = hi
         self.as_cfg.copy_(self._to_dev(_struct_bytes(c)))
 
     def add_binding(self, mac, ipv4: int = 0, ipv6: bytes = b"",

@@ -51,10 +51,10 @@ def test_extension_layout_matches_ctypes():
     assert off["qos_bucket.tokens"] == abi.QosBucket.tokens.offset
     assert off["qos_bucket.last_update"] == abi.QosBucket.last_update.offset
     assert off["binding_entry.ipv6_addr"] == abi.BindingEntry.ipv6_addr.offset
-    assert off["nat_config.private_net"] == abi.NatConfig.private_net.offset
+    assert off["nat_config.priv_lo"] == abi.NatConfig.priv_lo.offset
     assert off["nat_config.alg_key"] == abi.NatConfig.alg_key.offset
-    assert off["antispoof_config.allowed_net"] == \
-        abi.AntispoofConfig.allowed_net.offset
+    assert off["antispoof_config.allowed_lo"] == \
+        abi.AntispoofConfig.allowed_lo.offset
     assert off["spoof_event.spoofed_ip"] == abi.SpoofEvent.spoofed_ip.offset
 
 
