@@ -182,3 +182,25 @@ class TestCRDTProperties:
         before = reps[0].snapshot()
         reps[0].merge(reps[1].snapshot())
         assert reps[0].snapshot() == before
+
+
+@given(st.lists(st.tuples(st.integers(0, 0xFFFFFFFF),
+                          st.integers(0, 32)), max_size=16),
+       st.integers(0, 0xFFFFFFFF))
+@settings(max_examples=300, deadline=None)
+def test_interval_fold_equals_mask_scan(prefixes, probe_ip):
+    """The launcher's prefix->interval fold must answer membership
+    identically to the reference's per-prefix mask test (the LPM-trie
+    membership semantics the dataplane binary-searches)."""
+    from bng_amd.dataplane.abi import prefixes_to_intervals
+    ranges = [(net & (0xFFFFFFFF << (32 - plen)) & 0xFFFFFFFF
+               if plen else 0,
+               (0xFFFFFFFF << (32 - plen)) & 0xFFFFFFFF if plen else 0)
+              for net, plen in prefixes]
+    iv = prefixes_to_intervals(ranges)
+    # intervals are sorted and disjoint
+    for a, b in zip(iv, iv[1:]):
+        assert a[1] < b[0]
+    linear = any((probe_ip & m) == n for n, m in ranges)
+    binary = any(lo <= probe_ip <= hi for lo, hi in iv)
+    assert binary == linear
