@@ -356,7 +356,9 @@ def service_latency(launcher, args, now_sec, reps=128, batch=256,
             lats.append((time.perf_counter() - t) * 1e6)
         n_tx = int((v == abi.TX).sum())
         assert n_tx == batch, f"service answered {n_tx}/{batch}"
-        torch.cuda.synchronize()
+        # stream-scoped sync: a device-wide synchronize() would block
+        # on the resident service kernel
+        torch.cuda.current_stream().synchronize()
         lats.sort()
         return lats[len(lats) // 2], lats[int(len(lats) * 0.99)]
     finally:

@@ -130,7 +130,10 @@ def test_service_latency_quiesced_and_flooded(launcher):
             t = time.perf_counter()
             svc.serve(data, lens, NOW)
             lat_f.append((time.perf_counter() - t) * 1e6)
-        torch.cuda.synchronize()
+        # device-wide synchronize() would wait on the resident service
+        # kernel too (it never finishes while running) — sync only the
+        # flood stream
+        torch.cuda.current_stream().synchronize()
     lat_q.sort()
     lat_f.sort()
     q50, q99 = lat_q[32], lat_q[-1]
