@@ -53,8 +53,9 @@ void bng_launch_nat_sweep(void*, uint32_t, void*, uint32_t, void*, uint32_t,
 void bng_launch_shard_owner(const void*, const void*, void*, int, int, int,
                             hipStream_t);
 void bng_launch_dhcp_service(void*, void*, const void*, void*, void*, void*,
-                             int, const void*, uint32_t, const void*,
-                             uint32_t, const void*, void*, hipStream_t);
+                             int, int, void*, const void*, uint32_t,
+                             const void*, uint32_t, const void*, void*,
+                             hipStream_t);
 }
 
 namespace {
@@ -337,7 +338,8 @@ hipStream_t g_svc_stream = nullptr;
 void dhcp_service_start(torch::Tensor ctrl, torch::Tensor req,
                         torch::Tensor in_len, torch::Tensor out_len,
                         torch::Tensor verdict, torch::Tensor scratch,
-                        int64_t n_slots_arg,
+                        int64_t n_slots_arg, int64_t n_blocks,
+                        torch::Tensor ctrs,
                         torch::Tensor subs, torch::Tensor pools,
                         torch::Tensor cfg, torch::Tensor stats) {
   check_dev(scratch, "scratch"); check_dev(subs, "subs");
@@ -346,12 +348,17 @@ void dhcp_service_start(torch::Tensor ctrl, torch::Tensor req,
   TORCH_CHECK(ctrl.numel() * ctrl.element_size() ==
               (long)sizeof(bng_svc_ctrl), "ctrl must be 64 bytes");
   int n_slots = (int)n_slots_arg;
+  check_dev(ctrs, "ctrs");
+  TORCH_CHECK(ctrs.numel() * ctrs.element_size() >= 32,
+              "ctrs must be >= 32 bytes");
+  TORCH_CHECK(n_blocks >= 1 && n_blocks <= 16, "n_blocks in [1,16]");
   if (!g_svc_stream)
     (void)hipStreamCreateWithFlags(&g_svc_stream, hipStreamNonBlocking);
   bng_launch_dhcp_service(
       dev_ptr_of(ctrl, "ctrl"), dev_ptr_of(req, "req"),
       dev_ptr_of(in_len, "in_len"), dev_ptr_of(out_len, "out_len"),
       dev_ptr_of(verdict, "verdict"), scratch.data_ptr(), n_slots,
+      (int)n_blocks, ctrs.data_ptr(),
       subs.data_ptr(), table_mask(subs, sizeof(bng_sub_entry), "subs"),
       pools.data_ptr(),
       (uint32_t)(pools.numel() * pools.element_size() /

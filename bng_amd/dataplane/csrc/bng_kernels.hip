@@ -659,6 +659,14 @@ __global__ void dhcp_fastpath_kernel(
  * replies back, and release-stores tail.  Self-exits after
  * idle_exit_k*1024 empty polls so a crashed host can never wedge the
  * box. */
+/* Multi-block service: each block owns a fixed slice of the batch and
+ * polls the doorbell independently (no grid sync).  Spreading the
+ * slices over N CUs multiplies the outstanding PCIe reads during
+ * staging (the single-block form measured latency-bound at ~68 us for
+ * a 256-pkt batch) and divides the per-CU work a saturating data flood
+ * contends with.  Completion: the last block to finish a batch
+ * (device-scope acq_rel counter) publishes tail; the last block to
+ * exit publishes the exit ack. */
 __global__ void __launch_bounds__(256, 1) dhcp_service_kernel(
     bng_svc_ctrl* __restrict__ ctrl,
     uint8_t* __restrict__ req,            /* pinned [n_slots, stride]  */
@@ -667,6 +675,8 @@ __global__ void __launch_bounds__(256, 1) dhcp_service_kernel(
     uint8_t* __restrict__ verdict,        /* pinned [n_slots]          */
     uint8_t* __restrict__ scratch,        /* device [n_slots, stride]  */
     int n_slots,
+    uint32_t* __restrict__ ctrs,          /* device {done, served_lo,
+                                             served_hi, exits}         */
     const bng_sub_entry* __restrict__ subs, uint32_t sub_mask,
     const bng_ip_pool* __restrict__ pools, uint32_t n_pools,
     const bng_server_config* __restrict__ cfg,
@@ -674,12 +684,14 @@ __global__ void __launch_bounds__(256, 1) dhcp_service_kernel(
   __shared__ uint32_t s_head, s_run, s_n;
   __shared__ uint64_t s_now;
   const int tid = threadIdx.x;
+  const int nb = (int)gridDim.x, blk = (int)blockIdx.x;
   uint32_t done = __hip_atomic_load(&ctrl->tail, __ATOMIC_RELAXED,
                                     __HIP_MEMORY_SCOPE_SYSTEM);
   const int stride = (int)ctrl->stride;
   uint64_t idle = 0;
   const uint64_t idle_max = (uint64_t)ctrl->idle_exit_k * 1024u;
-  uint64_t served = 0, batches = 0;
+  uint64_t batches = 0;
+  unsigned long long* served64 = (unsigned long long*)(ctrs + 4);
   for (;;) {
     if (tid == 0) {
       s_head = __hip_atomic_load(&ctrl->head, __ATOMIC_ACQUIRE,
@@ -701,14 +713,17 @@ __global__ void __launch_bounds__(256, 1) dhcp_service_kernel(
     idle = 0;
     int n = (int)s_n;
     if (n > n_slots) n = n_slots;
-    /* stage request slots to HBM: wide coalesced PCIe reads */
-    int words = n * stride / 16;
-    const uint4* src = (const uint4*)req;
-    uint4* dst = (uint4*)scratch;
+    const int lo = (int)((long)n * blk / nb);
+    const int hi = (int)((long)n * (blk + 1) / nb);
+    /* stage this block's slice to HBM: wide coalesced PCIe reads */
+    const size_t w0 = (size_t)lo * stride / 16;
+    const int words = (hi - lo) * stride / 16;
+    const uint4* src = (const uint4*)req + w0;
+    uint4* dst = (uint4*)scratch + w0;
     for (int i = tid; i < words; i += blockDim.x) dst[i] = src[i];
     __syncthreads();
     dhcp_tables T{subs, sub_mask, pools, n_pools, cfg, stats, s_now};
-    for (int pid = tid; pid < n; pid += blockDim.x) {
+    for (int pid = lo + tid; pid < hi; pid += blockDim.x) {
       dhcp_flags F; F.clear();
       uint16_t ol = in_len[pid];
       int v = dhcp_process(scratch + (size_t)pid * stride, in_len[pid],
@@ -720,24 +735,34 @@ __global__ void __launch_bounds__(256, 1) dhcp_service_kernel(
     __syncthreads();
     /* replies back to the pinned ring (posted PCIe writes) */
     for (int i = tid; i < words; i += blockDim.x)
-      ((uint4*)req)[i] = dst[i];
+      ((uint4*)req)[w0 + i] = dst[i];
     __syncthreads();
-    ++done; ++batches; served += (uint64_t)n;
+    ++done; ++batches;
     if (tid == 0) {
-      ctrl->served = served;
-      ctrl->batches = batches;
-      __hip_atomic_store(&ctrl->tail, done, __ATOMIC_RELEASE,
-                         __HIP_MEMORY_SCOPE_SYSTEM);
+      atomicAdd(served64, (unsigned long long)(hi - lo));
+      /* acq_rel: the winner observes every block's served add */
+      uint32_t prev = __hip_atomic_fetch_add(&ctrs[0], 1u,
+                                             __ATOMIC_ACQ_REL,
+                                             __HIP_MEMORY_SCOPE_AGENT);
+      if ((int)(prev % (uint32_t)nb) == nb - 1) {   /* last finisher */
+        ctrl->served = *served64;
+        ctrl->batches = batches;
+        __hip_atomic_store(&ctrl->tail, done, __ATOMIC_RELEASE,
+                           __HIP_MEMORY_SCOPE_SYSTEM);
+      }
     }
     __syncthreads();
   }
   if (tid == 0) {
-    ctrl->served = served;
-    ctrl->batches = batches;
-    /* exit ack: the host's stop() waits on this instead of a stream
-     * sync, so a dead doorbell can never block the control plane */
-    __hip_atomic_store(&ctrl->exited, 1u, __ATOMIC_RELEASE,
-                       __HIP_MEMORY_SCOPE_SYSTEM);
+    uint32_t prev = __hip_atomic_fetch_add(&ctrs[1], 1u, __ATOMIC_ACQ_REL,
+                                           __HIP_MEMORY_SCOPE_AGENT);
+    if ((int)prev == nb - 1) {
+      ctrl->served = *served64;
+      /* exit ack: the host's stop() waits on this instead of a stream
+       * sync, so a dead doorbell can never block the control plane */
+      __hip_atomic_store(&ctrl->exited, 1u, __ATOMIC_RELEASE,
+                         __HIP_MEMORY_SCOPE_SYSTEM);
+    }
   }
 }
 
@@ -2037,13 +2062,15 @@ void bng_launch_dhcp(void* data, const void* in_len, void* out_len,
 
 void bng_launch_dhcp_service(void* ctrl, void* req, const void* in_len,
                              void* out_len, void* verdict, void* scratch,
-                             int n_slots, const void* subs,
+                             int n_slots, int n_blocks, void* ctrs,
+                             const void* subs,
                              uint32_t sub_mask, const void* pools,
                              uint32_t n_pools, const void* cfg, void* stats,
                              hipStream_t s) {
-  hipLaunchKernelGGL(dhcp_service_kernel, dim3(1), dim3(256), 0, s,
+  hipLaunchKernelGGL(dhcp_service_kernel, dim3(n_blocks), dim3(256), 0, s,
       (bng_svc_ctrl*)ctrl, (uint8_t*)req, (const uint16_t*)in_len,
       (uint16_t*)out_len, (uint8_t*)verdict, (uint8_t*)scratch, n_slots,
+      (uint32_t*)ctrs,
       (const bng_sub_entry*)subs, sub_mask, (const bng_ip_pool*)pools,
       n_pools, (const bng_server_config*)cfg,
       (unsigned long long*)stats);

@@ -566,34 +566,63 @@ class DhcpService:
     Latency path only — the batched dhcp_fastpath/uplink kernels remain
     the throughput path."""
 
-    def __init__(self, launcher: "HipLauncher", n_slots: int = 2048,
-                 stride: int = 512, idle_exit_k: int = 1_000_000):
-        import numpy as np
+    # Pinned-coherent buffers are allocated ONCE per (n_slots, stride)
+    # and cached for the process lifetime: hipHostMalloc/hipHostFree
+    # can implicitly synchronize the device, which deadlocks against a
+    # resident kernel (measured as a hang at service start on hardware)
+    # — never allocate or free host-mapped memory while services run.
+    _buf_cache: Dict[tuple, dict] = {}
+
+    @classmethod
+    def _buffers(cls, launcher, n_slots: int, stride: int) -> dict:
         torch = launcher.torch
+        key = (str(launcher.device), n_slots, stride)
+        b = cls._buf_cache.get(key)
+        if b is None:
+            alloc = launcher.ext.alloc_pinned_coherent
+            b = {"ctrl": alloc(64),
+                 "req": alloc(n_slots * stride).view(n_slots, stride),
+                 "in_len": alloc(n_slots * 2).view(torch.int16),
+                 "out_len": alloc(n_slots * 2).view(torch.int16),
+                 "verdict": alloc(n_slots),
+                 "scratch": torch.empty((n_slots, stride),
+                                        dtype=torch.uint8,
+                                        device=launcher.device),
+                 "ctrs": torch.zeros(8, dtype=torch.int32,
+                                     device=launcher.device)}
+            cls._buf_cache[key] = b
+        return b
+
+    def __init__(self, launcher: "HipLauncher", n_slots: int = 2048,
+                 stride: int = 512, idle_exit_k: int = 1_000_000,
+                 n_blocks: int = 4):
+        import numpy as np
         self.l = launcher
         self.n_slots, self.stride = n_slots, stride
-        # doorbell + rings MUST be fine-grained-coherent pinned memory
-        # (torch pin_memory is coarse-grained on ROCm: the resident
-        # kernel would cache it and never see host stores — measured as
-        # a dead doorbell on hardware)
-        alloc = launcher.ext.alloc_pinned_coherent
-        self.ctrl_t = alloc(64)
-        self.req = alloc(n_slots * stride).view(n_slots, stride)
-        self.in_len = alloc(n_slots * 2).view(torch.int16)
-        self.out_len = alloc(n_slots * 2).view(torch.int16)
-        self.verdict = alloc(n_slots)
-        self.scratch = torch.empty((n_slots, stride), dtype=torch.uint8,
-                                   device=launcher.device)
+        b = self._buffers(launcher, n_slots, stride)
+        self.ctrl_t = b["ctrl"]
+        self.req = b["req"]
+        self.in_len = b["in_len"]
+        self.out_len = b["out_len"]
+        self.verdict = b["verdict"]
+        self.scratch = b["scratch"]
+        self.ctrs = b["ctrs"]
+        # reset reused state (doorbell + completion counters)
+        self.ctrl_t.zero_()
+        self.ctrs.zero_()
         self.c = abi.SvcCtrl.from_address(self.ctrl_t.data_ptr())
         self.c.stride = stride
-        # idle self-exit: ~0.25us/poll -> default reaps after ~4 min
-        # (the box-safety bound if the owner dies without stop())
+        # idle self-exit: ~1us/poll (PCIe acquire) -> default reaps
+        # after ~17 min (the box-safety bound if the owner dies
+        # without stop())
         self.c.idle_exit_k = idle_exit_k
         self.c.run = 1
+        launcher.torch.cuda.synchronize(launcher.device)  # ctrs zeroed
         launcher.ext.dhcp_service_start(
             self.ctrl_t, self.req, self.in_len, self.out_len,
-            self.verdict, self.scratch, n_slots, launcher.subs,
-            launcher.pools, launcher.server_cfg, launcher.dhcp_stats)
+            self.verdict, self.scratch, n_slots, n_blocks, self.ctrs,
+            launcher.subs, launcher.pools, launcher.server_cfg,
+            launcher.dhcp_stats)
         self._np = np
 
     @property
