@@ -112,6 +112,12 @@ def build_parser() -> argparse.ArgumentParser:
                    choices=["none", "psk", "mtls"])
     g.add_argument("--nexus-psk", default="")
     g.add_argument("--nexus-pool", default="default")
+    g.add_argument("--clset-data-dir", default="",
+                   help="run an embedded CLSet CRDT replica persisted "
+                        "here (ref crdt_backend.go badger store)")
+    g.add_argument("--clset-peer", action="append", default=[],
+                   help="CLSet peer sync URL (repeatable)")
+    g.add_argument("--clset-listen-port", type=int, default=0)
     g = run.add_argument_group("deviceauth")
     g.add_argument("--auth-mode", default="",
                    choices=["", "none", "psk", "mtls"],
@@ -387,7 +393,26 @@ class BNG:
         self.dhcp_server.set_launcher(self.launcher)
         self.dhcp_server.set_walled_garden(self.walledgarden)
 
-        # 5. nexus (main.go:653-689)
+        # 5. nexus (main.go:653-689); an embedded CLSet replica is the
+        # reference's CRDT-backed store mode (crdt_backend.go)
+        if a.clset_data_dir or a.clset_peer:
+            from ..nexus.clset import CLSetHTTPServer, CLSetStore
+            from ..nexus.client import Client as NexusClient
+            self.clset = CLSetStore(
+                a.node_id, data_dir=a.clset_data_dir or None)
+            self.clset_http = CLSetHTTPServer(
+                self.clset, port=a.clset_listen_port).start()
+            self.clset.advertise_url = self.clset_http.url
+            for u in a.clset_peer:
+                self.clset.add_peer_url(u)
+            self.clset.start()
+            self._defer(self.clset_http.stop)
+            self._defer(self.clset.close)
+            self.nexus_client = NexusClient(self.clset,
+                                            node_id=a.node_id)
+            self.dhcp_server.set_nexus(client=self.nexus_client)
+            self.log.info("embedded CLSet replica at %s",
+                          self.clset_http.url)
         if a.nexus_url:
             from ..nexus.http_allocator import HTTPAllocator
             headers = None
