@@ -419,6 +419,10 @@ def main():
                     help="skip the host-boundary (PCIe-crossing) phase")
     ap.add_argument("--host-io-steps", type=int, default=12)
     ap.add_argument("--host-io-warmup", type=int, default=4)
+    ap.add_argument("--svc-cus", type=int, default=0,
+                    help="reserve N CUs for the persistent DHCP service "
+                         "and mask the pipeline off them (flood-proof "
+                         "serving latency at ~N/256 pipeline cost)")
     ap.add_argument("--no-sort", action="store_true",
                     help="disable on-device type-sort (wave-divergence fix)")
     ap.add_argument("--no-overlap", action="store_true",
@@ -451,7 +455,7 @@ def main():
     launcher = HipLauncher(
         device, sub_log2=base_log2, sess_log2=base_log2 + 1,
         eim_log2=base_log2, subnat_log2=base_log2, qos_log2=base_log2,
-        binding_log2=base_log2)
+        binding_log2=base_log2, svc_cus=args.svc_cus)
     launcher.set_server_config(b"\x02\x00\x00\x00\x00\x01",
                                ip2u32("10.255.255.1"))
     launcher.add_pool(1, ip2u32("10.0.0.0"), 8, ip2u32("10.255.255.1"),
@@ -705,6 +709,7 @@ def main():
                              "1M-sub table at 1/2/4/8 MI355X",
                 "global_batch": world * args.batch,
                 "seq_len": args.stride,
+                "svc_cus": args.svc_cus,
                 "parallelism": f"shard{world}-rss" +
                                ("-steerall" if args.steer_all else
                                 "-dhcp-alltoall"),

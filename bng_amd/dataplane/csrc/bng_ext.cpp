@@ -64,6 +64,66 @@ hipStream_t cur_stream() {
   return c10::hip::getCurrentHIPStream().stream();
 }
 
+hipStream_t g_svc_stream = nullptr;   /* persistent-service stream */
+
+/* CU partitioning (the saturated-tail lever): reserve the first
+ * `svc_cus` CUs for the resident service and launch the batched
+ * pipeline on the complementary mask, so a 100% data flood cannot
+ * co-schedule on the service CUs.  Costs the pipeline svc_cus/256 of
+ * its capacity. */
+hipStream_t g_masked_stream = nullptr;
+int g_svc_cus = 0;
+
+void set_cu_partition(int64_t svc_cus) {
+  TORCH_CHECK(svc_cus >= 0 && svc_cus <= 64, "svc_cus in [0,64]");
+  TORCH_CHECK(g_svc_stream == nullptr && g_masked_stream == nullptr,
+              "set_cu_partition must run before the first service "
+              "start / masked launch");
+  g_svc_cus = (int)svc_cus;
+}
+
+hipStream_t masked_stream() {
+  if (!g_masked_stream) {
+    if (g_svc_cus > 0) {
+      uint32_t mask[8];
+      for (int i = 0; i < 8; ++i) mask[i] = 0xFFFFFFFFu;
+      for (int c = 0; c < g_svc_cus; ++c)
+        mask[c / 32] &= ~(1u << (c % 32));
+      (void)hipExtStreamCreateWithCUMask(&g_masked_stream, 8, mask);
+    } else {
+      (void)hipStreamCreateWithFlags(&g_masked_stream,
+                                     hipStreamNonBlocking);
+    }
+  }
+  return g_masked_stream;
+}
+
+void masked_sync() {
+  if (g_masked_stream) (void)hipStreamSynchronize(g_masked_stream);
+}
+
+/* Fences that make a masked launch transparent to the caller's stream
+ * order: the masked stream waits for prior work on the current stream,
+ * and the current stream waits for the masked kernel — so bench's
+ * event machinery (prep/work_free) keeps working unchanged. */
+hipEvent_t g_fence_pre = nullptr, g_fence_post = nullptr;
+
+hipStream_t masked_entry(hipStream_t cur) {
+  hipStream_t st = masked_stream();
+  if (!g_fence_pre) {
+    (void)hipEventCreateWithFlags(&g_fence_pre, hipEventDisableTiming);
+    (void)hipEventCreateWithFlags(&g_fence_post, hipEventDisableTiming);
+  }
+  (void)hipEventRecord(g_fence_pre, cur);
+  (void)hipStreamWaitEvent(st, g_fence_pre, 0);
+  return st;
+}
+
+void masked_exit(hipStream_t st, hipStream_t cur) {
+  (void)hipEventRecord(g_fence_post, st);
+  (void)hipStreamWaitEvent(cur, g_fence_post, 0);
+}
+
 void check_dev(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda(), name, " must be a device tensor");
   TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
@@ -157,7 +217,8 @@ void uplink_pipeline(torch::Tensor data, torch::Tensor in_len,
                      int64_t now_sec,
                      c10::optional<torch::Tensor> order,
                      bool downlink,
-                     c10::optional<torch::Tensor> now_buf) {
+                     c10::optional<torch::Tensor> now_buf,
+                     bool masked) {
   check_dev(data, "data");
   bng_uplink_params P{};
   P.now_ptr = now_buf.has_value()
@@ -201,10 +262,13 @@ void uplink_pipeline(torch::Tensor data, torch::Tensor in_len,
   P.qos_stats = (unsigned long long*)qos_stats.data_ptr();
   P.now_ns = (uint64_t)now_ns;
   P.now_sec = (uint64_t)now_sec;
+  hipStream_t cur = cur_stream();
+  hipStream_t st = masked ? masked_entry(cur) : cur;
   if (downlink)
-    bng_launch_downlink(&P, cur_stream());
+    bng_launch_downlink(&P, st);
   else
-    bng_launch_uplink(&P, cur_stream());
+    bng_launch_uplink(&P, st);
+  if (masked) masked_exit(st, cur);
 }
 
 void pkt_class(torch::Tensor data, torch::Tensor in_len,
@@ -330,10 +394,10 @@ void* dev_ptr_of(const torch::Tensor& t, const char* name) {
   return dp;
 }
 
-/* Persistent DHCP service: launch the device-resident kernel on its own
- * non-blocking stream (it never returns until stopped, so it must not
- * share a stream with ordinary work). */
-hipStream_t g_svc_stream = nullptr;
+/* Persistent DHCP service: launched on its own stream (it never
+ * returns until stopped, so it must not share a stream with ordinary
+ * work); g_svc_stream declared beside the CU-partition helpers. */
+
 
 void dhcp_service_start(torch::Tensor ctrl, torch::Tensor req,
                         torch::Tensor in_len, torch::Tensor out_len,
@@ -352,8 +416,16 @@ void dhcp_service_start(torch::Tensor ctrl, torch::Tensor req,
   TORCH_CHECK(ctrs.numel() * ctrs.element_size() >= 32,
               "ctrs must be >= 32 bytes");
   TORCH_CHECK(n_blocks >= 1 && n_blocks <= 16, "n_blocks in [1,16]");
-  if (!g_svc_stream)
-    (void)hipStreamCreateWithFlags(&g_svc_stream, hipStreamNonBlocking);
+  if (!g_svc_stream) {
+    if (g_svc_cus > 0) {
+      uint32_t mask[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      for (int c = 0; c < g_svc_cus; ++c)
+        mask[c / 32] |= (1u << (c % 32));
+      (void)hipExtStreamCreateWithCUMask(&g_svc_stream, 8, mask);
+    } else {
+      (void)hipStreamCreateWithFlags(&g_svc_stream, hipStreamNonBlocking);
+    }
+  }
   bng_launch_dhcp_service(
       dev_ptr_of(ctrl, "ctrl"), dev_ptr_of(req, "req"),
       dev_ptr_of(in_len, "in_len"), dev_ptr_of(out_len, "out_len"),
@@ -427,7 +499,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("qos_stats"), py::arg("now_ns"), py::arg("now_sec"),
         py::arg("order") = py::none(),
         py::arg("downlink") = false,
-        py::arg("now_buf") = py::none());
+        py::arg("now_buf") = py::none(),
+        py::arg("masked") = false);
+  m.def("set_cu_partition", &set_cu_partition);
+  m.def("masked_sync", &masked_sync);
   m.def("pkt_class", &pkt_class);
   m.def("sub_upsert", &sub_upsert);
   m.def("sub_delete", &sub_delete);

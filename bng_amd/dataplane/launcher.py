@@ -46,11 +46,20 @@ class HipLauncher:
     def __init__(self, device: str = "cuda:0", *,
                  sub_log2: int = 18, sess_log2: int = 20, eim_log2: int = 19,
                  subnat_log2: int = 18, qos_log2: int = 18,
-                 binding_log2: int = 18, n_pools: int = 1024):
+                 binding_log2: int = 18, n_pools: int = 1024,
+                 svc_cus: int = 0):
         import torch
         from .build import get_ext
         self.torch = torch
         self.ext = get_ext(required=True)
+        # CU partition: reserve svc_cus CUs for the resident DHCP
+        # service; pipeline launches go out masked off those CUs (the
+        # saturated-flood latency lever — costs the pipeline
+        # svc_cus/256 of peak)
+        self.masked_compute = False
+        if svc_cus > 0:
+            self.ext.set_cu_partition(svc_cus)
+            self.masked_compute = True
         if self.ext is None:
             raise RuntimeError(
                 "bng dataplane extension not built — run "
@@ -426,7 +435,8 @@ class HipLauncher:
             self.reverse, self.eim, self.subctx, self.nat_cfg, self.hairpin,
             self.n_hairpin, self.nat_stats, self.nat_log_ring,
             self.nat_log_hdr, self.qos_egress, self.qos_stats, now,
-            now_sec if now_sec is not None else now // 10**9, order=order)
+            now_sec if now_sec is not None else now // 10**9, order=order,
+            masked=self.masked_compute)
         return verdict, out_len
 
     def downlink(self, data, lens, now_ns: Optional[int] = None):
@@ -441,7 +451,8 @@ class HipLauncher:
             self.reverse, self.eim, self.subctx, self.nat_cfg, self.hairpin,
             self.n_hairpin, self.nat_stats, self.nat_log_ring,
             self.nat_log_hdr, self.qos_egress, self.qos_stats, now,
-            now // 10**9, order=None, downlink=True)
+            now // 10**9, order=None, downlink=True,
+            masked=self.masked_compute)
         return verdict
 
     # ------------------------------------------- HA table snapshotting
