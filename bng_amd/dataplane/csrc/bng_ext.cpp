@@ -85,10 +85,13 @@ void set_cu_partition(int64_t svc_cus) {
 hipStream_t masked_stream() {
   if (!g_masked_stream) {
     if (g_svc_cus > 0) {
+      /* reserved CUs are spread one per 32-CU mask word so BOTH
+       * partitions keep CUs in every XCD/SE (a mask concentrated in
+       * word0 hung the masked stream on hardware) */
       uint32_t mask[8];
       for (int i = 0; i < 8; ++i) mask[i] = 0xFFFFFFFFu;
       for (int c = 0; c < g_svc_cus; ++c)
-        mask[c / 32] &= ~(1u << (c % 32));
+        mask[c % 8] &= ~(1u << (c / 8));
       (void)hipExtStreamCreateWithCUMask(&g_masked_stream, 8, mask);
     } else {
       (void)hipStreamCreateWithFlags(&g_masked_stream,
@@ -420,7 +423,7 @@ void dhcp_service_start(torch::Tensor ctrl, torch::Tensor req,
     if (g_svc_cus > 0) {
       uint32_t mask[8] = {0, 0, 0, 0, 0, 0, 0, 0};
       for (int c = 0; c < g_svc_cus; ++c)
-        mask[c / 32] |= (1u << (c % 32));
+        mask[c % 8] |= (1u << (c / 8));
       (void)hipExtStreamCreateWithCUMask(&g_svc_stream, 8, mask);
     } else {
       (void)hipStreamCreateWithFlags(&g_svc_stream, hipStreamNonBlocking);

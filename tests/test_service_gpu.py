@@ -171,7 +171,7 @@ def test_partitioned_service_flood_latency():
     from bng_amd.dataplane.build import get_ext
     from bng_amd.dataplane.launcher import DhcpService, HipLauncher
     try:
-        get_ext(required=True).set_cu_partition(4)
+        get_ext(required=True).set_cu_partition(8)
     except RuntimeError:
         pytest.skip("streams already created unpartitioned "
                     "(run this test standalone)")
