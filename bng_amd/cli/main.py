@@ -419,8 +419,10 @@ class BNG:
             if a.nexus_auth == "psk" and a.nexus_psk:
                 from ..deviceauth.authenticator import PSKAuthenticator
                 headers = PSKAuthenticator(a.nexus_psk).headers(a.node_id)
-            self.nexus_allocator = HTTPAllocator(a.nexus_url,
-                                                 auth_headers=headers)
+            self.nexus_allocator = HTTPAllocator(
+                a.nexus_url, auth_headers=headers,
+                client_cert=a.auth_mtls_cert, client_key=a.auth_mtls_key,
+                ca_cert=a.auth_mtls_ca, insecure=a.auth_mtls_insecure)
             self.dhcp_server.set_nexus(allocator=self.nexus_allocator)
 
         # 6. peer pool (main.go:719-756)

@@ -29,13 +29,24 @@ class HTTPAllocatorError(Exception):
 class HTTPAllocator:
     def __init__(self, base_url: str, timeout: float = 5.0,
                  auth_headers: Optional[Dict[str, str]] = None,
-                 session=None):
+                 session=None, client_cert: str = "",
+                 client_key: str = "", ca_cert: str = "",
+                 insecure: bool = False):
         import requests
         self.base_url = base_url.rstrip("/")
         self.timeout = timeout
         self.session = session or requests.Session()
         if auth_headers:
             self.session.headers.update(auth_headers)
+        # device->Nexus mTLS (ref deviceauth mtls.go + transport.go):
+        # client cert/key presented on every request; server verified
+        # against ca_cert unless explicitly insecure
+        if client_cert and client_key:
+            self.session.cert = (client_cert, client_key)
+        if insecure:
+            self.session.verify = False
+        elif ca_cert:
+            self.session.verify = ca_cert
 
     def _url(self, path: str) -> str:
         return self.base_url + path

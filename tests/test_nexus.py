@@ -435,3 +435,20 @@ class TestCLSetPersistence:
         srv_c.stop()
         for s in (a, b, c):
             s.close()
+
+
+def test_http_allocator_mtls_session_config(tmp_path):
+    """--auth-mtls-* flags reach the transport: client cert attached,
+    server verification pinned to the CA (ref deviceauth transport.go)."""
+    from bng_amd.nexus.http_allocator import HTTPAllocator
+    cert = tmp_path / "c.pem"
+    key = tmp_path / "k.pem"
+    ca = tmp_path / "ca.pem"
+    for f in (cert, key, ca):
+        f.write_text("pem")
+    a = HTTPAllocator("https://nexus.example", client_cert=str(cert),
+                      client_key=str(key), ca_cert=str(ca))
+    assert a.session.cert == (str(cert), str(key))
+    assert a.session.verify == str(ca)
+    b = HTTPAllocator("https://nexus.example", insecure=True)
+    assert b.session.verify is False
