@@ -140,6 +140,9 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--peer-service", default="",
                    help="DNS SRV name for --peer-discovery dns")
     g.add_argument("--peer-listen", default=":8081")
+    g.add_argument("--peer-dns-server", default="",
+                   help="DNS server for --peer-discovery dns "
+                        "(host:port)")
     g = run.add_argument_group("nat")
     g.add_argument("--nat-enable", "--nat-enabled", action="store_true")
     g.add_argument("--nat-public-ip", action="append", default=[])
@@ -425,7 +428,21 @@ class BNG:
                 ca_cert=a.auth_mtls_ca, insecure=a.auth_mtls_insecure)
             self.dhcp_server.set_nexus(allocator=self.nexus_allocator)
 
-        # 6. peer pool (main.go:719-756)
+        # 6. peer pool (main.go:719-756); DNS-SRV peer discovery
+        # (ref peer-discovery=dns + peer-service): each SRV record
+        # target:port becomes a peer URL, node id = target host
+        if a.peer_discovery == "dns" and a.peer_service:
+            from ..dns.resolver import resolve_srv
+            try:
+                srvs = resolve_srv(a.peer_service,
+                                   getattr(a, "peer_dns_server",
+                                           "") or "127.0.0.1:53")
+                for _pri, _w, port, target in sorted(srvs):
+                    if target and target != a.node_id:
+                        a.peer.append(f"{target}=http://{target}:{port}")
+                self.log.info("DNS-SRV discovered %d peer(s)", len(srvs))
+            except OSError as e:
+                self.log.warning("DNS-SRV discovery failed: %s", e)
         if a.peer and a.pool_network:
             from ..pool.peer import PeerPool
             peers = dict(p.split("=", 1) for p in a.peer)

@@ -13,6 +13,7 @@ from dataclasses import dataclass, field
 from typing import Callable, Dict, List, Optional, Tuple
 
 QTYPE_A = 1
+QTYPE_SRV = 33
 QTYPE_AAAA = 28
 
 
@@ -209,3 +210,53 @@ class Resolver:
             for k in dead:
                 del self.cache[k]
             return len(dead)
+
+
+# ------------------------------------------------------------- SRV (RFC 2782)
+def build_srv_response(query: bytes, records, ttl: int = 300) -> bytes:
+    """records: [(priority, weight, port, target_name)] — used by tests
+    and by the embedded DNS server for service records."""
+    txid = struct.unpack_from(">H", query, 0)[0]
+    _, off = decode_qname(query, 12)
+    hdr = struct.pack(">HHHHHH", txid, 0x8180, 1, len(records), 0, 0)
+    out = hdr + query[12:off + 4]
+    for pri, weight, port, target in records:
+        rdata = (struct.pack(">HHH", pri, weight, port) +
+                 encode_qname(target))
+        out += b"\xc0\x0c" + struct.pack(">HHIH", QTYPE_SRV, 1, ttl,
+                                          len(rdata)) + rdata
+    return out
+
+
+def parse_srv_response(data: bytes):
+    """-> [(priority, weight, port, target)] from a DNS SRV answer."""
+    _, flags, qd, an, _, _ = struct.unpack_from(">HHHHHH", data, 0)
+    off = 12
+    for _ in range(qd):
+        _, off = decode_qname(data, off)
+        off += 4
+    out = []
+    for _ in range(an):
+        _, off = decode_qname(data, off)
+        rtype, rclass, ttl, rdlen = struct.unpack_from(">HHIH", data, off)
+        off += 10
+        if rtype == QTYPE_SRV and rdlen >= 7:
+            pri, weight, port = struct.unpack_from(">HHH", data, off)
+            target, _ = decode_qname(data, off + 6)
+            out.append((pri, weight, port, target))
+        off += rdlen
+    return out
+
+
+def resolve_srv(name: str, server: str, timeout: float = 2.0):
+    """One-shot SRV lookup over UDP (the reference's DNS-SRV peer
+    discovery, peer-discovery=dns + peer-service flags)."""
+    import socket as _s
+    host, _, port = server.rpartition(":")
+    addr = (host or server, int(port or 53))
+    q = build_query(name, QTYPE_SRV, txid=0x5273)
+    with _s.socket(_s.AF_INET, _s.SOCK_DGRAM) as sk:
+        sk.settimeout(timeout)
+        sk.sendto(q, addr)
+        data, _ = sk.recvfrom(4096)
+    return parse_srv_response(data)

@@ -344,3 +344,38 @@ class TestClsetWiring:
         again = CLSetStore("n2", data_dir=str(tmp_path / "n2"))
         assert again.get("subscribers/s1") == b"alice"
         again.close()
+
+
+class TestDnsSrvPeerDiscovery:
+    def test_srv_records_become_peers(self, tmp_path):
+        """--peer-discovery dns resolves the SRV service into the peer
+        list before the peer pool starts."""
+        import socket
+        import threading
+        from bng_amd.dns.resolver import build_srv_response
+        sk = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+        sk.bind(("127.0.0.1", 0))
+        port = sk.getsockname()[1]
+
+        def responder():
+            q, addr = sk.recvfrom(4096)
+            sk.sendto(build_srv_response(
+                q, [(10, 5, 8081, "peer-a.pool"),
+                    (20, 5, 8082, "peer-b.pool")]), addr)
+
+        threading.Thread(target=responder, daemon=True).start()
+        from bng_amd.cli.main import BNG, build_parser
+        args = build_parser().parse_args([
+            "run", "--gpu", "off", "--node-id", "n1",
+            "--pool-network", "10.0.4.0/24",
+            "--peer-discovery", "dns",
+            "--peer-service", "_bng._tcp.pool",
+            "--peer-dns-server", f"127.0.0.1:{port}"])
+        bng = BNG(args).start()
+        try:
+            assert "peer-a.pool=http://peer-a.pool:8081" in bng.args.peer
+            assert "peer-b.pool=http://peer-b.pool:8082" in bng.args.peer
+            assert bng.peer_pool is not None
+        finally:
+            bng.stop()
+            sk.close()
