@@ -420,9 +420,10 @@ def main():
     ap.add_argument("--host-io-steps", type=int, default=12)
     ap.add_argument("--host-io-warmup", type=int, default=4)
     ap.add_argument("--svc-cus", type=int, default=0,
-                    help="reserve N CUs for the persistent DHCP service "
-                         "and mask the pipeline off them (flood-proof "
-                         "serving latency at ~N/256 pipeline cost)")
+                    help="EXPERIMENTAL: reserve N CUs for the service "
+                         "via CU-masked streams — currently hangs "
+                         "kernel completion on ROCm 7.0/gfx950 "
+                         "(R02_EVIDENCE.md); leave 0")
     ap.add_argument("--no-sort", action="store_true",
                     help="disable on-device type-sort (wave-divergence fix)")
     ap.add_argument("--no-overlap", action="store_true",
