@@ -24,13 +24,12 @@ no other changes.
 from __future__ import annotations
 
 import ctypes
-import errno
 import mmap
 import os
 import select
 import socket
 import struct
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 _libc = ctypes.CDLL(None, use_errno=True)
 _SYS_BPF = 321            # x86_64

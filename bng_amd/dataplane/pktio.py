@@ -14,7 +14,6 @@ the GPU launcher (uplink/dhcp pipelines), and routes results by verdict:
 TX/FWD frames to the sink, PASS frames to the slow-path callback."""
 from __future__ import annotations
 
-import os
 import socket
 import struct
 import threading
