@@ -263,8 +263,23 @@ class XskSocket:
 
         self.sock = socket.socket(44, socket.SOCK_RAW, 0)   # AF_XDP
         fd = self.sock.fileno()
+        try:
+            self._setup(fd, ifname, queue, frame_size, ring_size, mode,
+                        attach_prog)
+        except BaseException:
+            # leave no half-built socket/prog/link behind
+            for h in ("link_fd", "prog_fd", "map_fd"):
+                v = getattr(self, h, -1)
+                if v >= 0:
+                    os.close(v)
+            self.sock.close()
+            raise
+
+    def _setup(self, fd, ifname, queue, frame_size, ring_size, mode,
+               attach_prog):
         self.sock.setsockopt(SOL_XDP, XDP_UMEM_REG, struct.pack(
-            "<QQIIII", self.umem_addr, umem_len, frame_size, 0, 0, 0))
+            "<QQIIII", self.umem_addr, self.n_frames * frame_size,
+            frame_size, 0, 0, 0))
         for opt in (XDP_UMEM_FILL_RING, XDP_UMEM_COMPLETION_RING,
                     XDP_RX_RING, XDP_TX_RING):
             self.sock.setsockopt(SOL_XDP, opt, struct.pack("<I", ring_size))
@@ -326,7 +341,8 @@ class XskSocket:
             return out
         prod = self.rx.producer
         cons = self.rx.consumer
-        n = min(prod - cons, max_frames)
+        # u32 ring indices: subtract modulo 2^32 (the kernel wraps)
+        n = min((prod - cons) & 0xFFFFFFFF, max_frames)
         refill = []
         for i in range(n):
             addr, ln, _opts = struct.unpack("<QII",
@@ -342,7 +358,7 @@ class XskSocket:
     def _reclaim(self):
         prod = self.comp.producer
         cons = self.comp.consumer
-        for i in range(prod - cons):
+        for i in range((prod - cons) & 0xFFFFFFFF):
             addr = struct.unpack("<Q", self.comp.read_desc(cons + i))[0]
             self._tx_free.append(addr // self.frame_size)
         self.comp.consumer = prod
@@ -350,9 +366,12 @@ class XskSocket:
     def send_batch(self, frames: List[bytes]):
         self._reclaim()
         p = self.tx.producer
+        # TX ring capacity: never advance producer past consumer+size
+        room = self.tx.size - ((p - self.tx.consumer) & 0xFFFFFFFF)
         sent = 0
         for fr in frames:
-            if not self._tx_free or len(fr) > self.frame_size:
+            if sent >= room or not self._tx_free or \
+                    len(fr) > self.frame_size:
                 self.stats["tx_dropped"] += 1
                 continue
             fno = self._tx_free.pop()
@@ -375,10 +394,12 @@ class XskSocket:
         """Array-native sink API (vectorized Pump)."""
         self._reclaim()
         p = self.tx.producer
+        room = self.tx.size - ((p - self.tx.consumer) & 0xFFFFFFFF)
         sent = 0
         for i in range(len(lens)):
             ln = int(lens[i])
-            if not self._tx_free or ln > self.frame_size:
+            if sent >= room or not self._tx_free or \
+                    ln > self.frame_size:
                 self.stats["tx_dropped"] += 1
                 continue
             fno = self._tx_free.pop()
