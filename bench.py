@@ -442,6 +442,10 @@ def main():
     distributed = world > 1
     if distributed:
         import torch.distributed as dist
+        # clamp for single-GPU multi-rank rehearsals (RCCL co-located
+        # ranks); on a real N-GPU node device_count >= local_rank and
+        # this is a no-op
+        local_rank = min(local_rank, torch.cuda.device_count() - 1)
         torch.cuda.set_device(local_rank)
         dist.init_process_group("nccl")
     device = f"cuda:{local_rank}"
