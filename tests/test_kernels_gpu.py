@@ -1141,4 +1141,10 @@ def test_interval_ranges_at_scale_vs_golden():
         linear = any((ip & m) == n for n, m in privs)
         binary = any(lo <= ip <= hi for lo, hi in iv)
         assert linear == binary, f"interval fold diverges at {ip:#x}"
-    assert (gpu_v == abi.PASS).all()   # no subctx entries -> all PASS
+    # full differential: GPU NAT verdict == golden verdict per probe
+    g.dp.now_ns = NOW * 10**9
+    for i, fr in enumerate(frames):
+        fb = bytearray(fr)
+        v_cpu = g.dp.nat44_egress(fb)
+        assert gpu_v[i] == v_cpu, \
+            f"nat gate mismatch probe {i}: gpu {gpu_v[i]} cpu {v_cpu}"
