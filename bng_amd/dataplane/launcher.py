@@ -604,10 +604,19 @@ class DhcpService:
             cls._buf_cache[key] = b
         return b
 
+    _active: Optional["DhcpService"] = None
+
     def __init__(self, launcher: "HipLauncher", n_slots: int = 2048,
                  stride: int = 512, idle_exit_k: int = 1_000_000,
                  n_blocks: int = 4):
         import numpy as np
+        # one service at a time per process: the service stream is
+        # global, so a second resident kernel would queue behind the
+        # first and its doorbell would look dead
+        if DhcpService._active is not None and \
+                not DhcpService._active.c.exited:
+            raise RuntimeError("another DhcpService is still running; "
+                               "stop() it first")
         self.l = launcher
         self.n_slots, self.stride = n_slots, stride
         b = self._buffers(launcher, n_slots, stride)
@@ -634,6 +643,7 @@ class DhcpService:
             self.verdict, self.scratch, n_slots, n_blocks, self.ctrs,
             launcher.subs, launcher.pools, launcher.server_cfg,
             launcher.dhcp_stats)
+        DhcpService._active = self
         self._np = np
 
     @property
@@ -683,6 +693,8 @@ class DhcpService:
                     "dead — kernel left to idle self-exit")
             time.sleep(0.001)
         self.l.ext.dhcp_service_join()
+        if DhcpService._active is self:
+            DhcpService._active = None
 
     def __enter__(self):
         return self
