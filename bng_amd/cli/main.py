@@ -751,10 +751,16 @@ class BNG:
         return self
 
     def _frame_slow_path(self, frame: bytes):
-        """PASS-verdict frames (DHCP cache misses) -> slow-path server;
-        returns the reply frame, if any (ref XDP_PASS -> server4)."""
+        """PASS-verdict frames -> slow-path servers: PPPoE discovery/
+        session frames to the PPPoE server (which may answer with
+        several frames), DHCP cache misses to the DHCP server
+        (ref XDP_PASS -> server4 / AF_PACKET -> pppoe)."""
         from ..dataplane.packets import parse_dhcp_frame
         from ..dhcp import message as dm
+        if len(frame) >= 14 and getattr(self, "pppoe", None) is not None:
+            et = int.from_bytes(frame[12:14], "big")
+            if et in (0x8863, 0x8864):
+                return self.pppoe.handle_frame(frame) or None
         try:
             p = parse_dhcp_frame(frame)
         except (AssertionError, IndexError, ValueError):

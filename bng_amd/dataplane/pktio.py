@@ -285,8 +285,14 @@ class Pump:
                     except Exception:
                         reply = None
                     if reply:
-                        slow_replies.append(reply)
-                        self.stats["slow_replies"] += 1
+                        # a slow path may answer with several frames
+                        # (PPPoE handshakes emit PADS + LCP together)
+                        if isinstance(reply, (list, tuple)):
+                            slow_replies.extend(reply)
+                            self.stats["slow_replies"] += len(reply)
+                        else:
+                            slow_replies.append(reply)
+                            self.stats["slow_replies"] += 1
             if self.sink is not None:
                 if hasattr(self.sink, "send_batch_array"):
                     if len(out_lens):
@@ -344,8 +350,12 @@ class Pump:
                 except Exception:
                     reply = None
                 if reply:
-                    out_frames.append(reply)
-                    self.stats["slow_replies"] += 1
+                    if isinstance(reply, (list, tuple)):
+                        out_frames.extend(reply)
+                        self.stats["slow_replies"] += len(reply)
+                    else:
+                        out_frames.append(reply)
+                        self.stats["slow_replies"] += 1
         if self.sink is not None and out_frames:
             self.sink.send_batch(out_frames)
         return out_frames, passed

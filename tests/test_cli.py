@@ -379,3 +379,33 @@ class TestDnsSrvPeerDiscovery:
         finally:
             bng.stop()
             sk.close()
+
+
+class TestPppoeOverPump:
+    def test_padi_through_pump_gets_pado(self):
+        """PPPoE discovery frames arriving at the NIC edge PASS the
+        dataplane and are answered by the PPPoE server via the pump
+        slow path (`bng run --pppoe-enable --pktio ...`)."""
+        from bng_amd.cli.main import BNG, build_parser
+        from bng_amd.pppoe import codec as C
+        args = build_parser().parse_args([
+            "run", "--gpu", "off", "--pool-network", "10.0.5.0/24",
+            "--pppoe-enable"])
+        bng = BNG(args).start()
+        try:
+            from bng_amd.dataplane.pktio import ListSink, Pump, \
+                SyntheticSource
+            sink = ListSink()
+            pump = Pump(bng.launcher, SyntheticSource(lambda n: []),
+                        sink, slow_path=bng._frame_slow_path, batch=16)
+            padi = C.DiscoveryPacket(
+                C.PADI, 0, [(C.TAG_SERVICE_NAME, b"")],
+                src_mac=b"\xaa\xbb\xcc\x00\x00\x21").encode()
+            out, passed = pump.process([padi])
+            assert pump.stats["slow_replies"] >= 1
+            pados = [f for f in sink.frames
+                     if int.from_bytes(f[12:14], "big") == 0x8863
+                     and C.DiscoveryPacket.decode(f).code == C.PADO]
+            assert pados, "no PADO on the wire"
+        finally:
+            bng.stop()
