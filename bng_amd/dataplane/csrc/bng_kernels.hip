@@ -1536,6 +1536,17 @@ void uplink_pipeline_kernel(bng_uplink_params P) {
       uint16_t ol = (uint16_t)len;
       int v;
       pktctx c;
+      /* PPPoE control/session ethertypes go to the host PPPoE server:
+       * the reference's TC_ACT_OK continues the kernel stack where
+       * AF_PACKET PPPoE sockets receive them — our FWD would hairpin
+       * them back out the wire instead, so the pump equivalent of
+       * "continue the stack" is PASS (cli _frame_slow_path). */
+      uint16_t et = len >= 14 ? (uint16_t)((p[12] << 8) | p[13]) : 0;
+      if (et == 0x8863 || et == 0x8864) {
+        P.verdict[pid] = (uint8_t)BNG_PASS;
+        P.out_len[pid] = ol;
+        continue;
+      }
       bool ip_ok = parse_pkt(c, p, len, /*want_vlan=*/true);
       bool is_dhcp = ip_ok && c.ip_off >= 0 && c.proto == 17 && c.l4_ok &&
                      c.dport == 67;

@@ -1148,3 +1148,28 @@ def test_interval_ranges_at_scale_vs_golden():
         v_cpu = g.dp.nat44_egress(fb)
         assert gpu_v[i] == v_cpu, \
             f"nat gate mismatch probe {i}: gpu {gpu_v[i]} cpu {v_cpu}"
+
+
+@pytest.mark.gpu
+def test_pppoe_ethertypes_pass_to_slow_path():
+    """PPPoE discovery/session frames must PASS out of the fused
+    pipeline (to the host PPPoE server), not FWD back out the wire."""
+    from bng_amd.dataplane.launcher import HipLauncher
+    from bng_amd.pppoe import codec as C
+    l = HipLauncher("cuda:0")
+    l.set_server_config(mac_bytes("02:00:00:00:00:01"),
+                        ip2u32("10.255.255.1"))
+    padi = C.DiscoveryPacket(C.PADI, 0, [(C.TAG_SERVICE_NAME, b"")],
+                             src_mac=b"\xaa\xbb\xcc\x00\x00\x31").encode()
+    sess = C.SessionPacket(7, C.PROTO_LCP, b"\x01\x01\x00\x04",
+                           src_mac=b"\xaa\xbb\xcc\x00\x00\x31",
+                           dst_mac=b"\x02\x00\x00\x00\x00\x01").encode()
+    data_pkt = build_ipv4("aa:bb:cc:00:00:31", "02:00:00:00:00:01",
+                          ip2u32("10.0.0.9"), ip2u32("8.8.8.8"),
+                          proto=17, sport=1000, dport=53,
+                          payload=b"x" * 22)
+    d, lens = l.make_batch([padi, sess, data_pkt])
+    v, _ = l.uplink(d, lens, now_ns=10**18, now_sec=10**9)
+    v = v.cpu().numpy()
+    assert v[0] == abi.PASS and v[1] == abi.PASS
+    assert v[2] != abi.PASS     # data path unaffected
