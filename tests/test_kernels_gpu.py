@@ -1164,12 +1164,13 @@ def test_pppoe_ethertypes_pass_to_slow_path():
     sess = C.SessionPacket(7, C.PROTO_LCP, b"\x01\x01\x00\x04",
                            src_mac=b"\xaa\xbb\xcc\x00\x00\x31",
                            dst_mac=b"\x02\x00\x00\x00\x00\x01").encode()
+    # non-private source: skips the NAT gate entirely -> FWD
     data_pkt = build_ipv4("aa:bb:cc:00:00:31", "02:00:00:00:00:01",
-                          ip2u32("10.0.0.9"), ip2u32("8.8.8.8"),
+                          ip2u32("9.9.9.9"), ip2u32("8.8.8.8"),
                           proto=17, sport=1000, dport=53,
                           payload=b"x" * 22)
     d, lens = l.make_batch([padi, sess, data_pkt])
     v, _ = l.uplink(d, lens, now_ns=10**18, now_sec=10**9)
     v = v.cpu().numpy()
     assert v[0] == abi.PASS and v[1] == abi.PASS
-    assert v[2] != abi.PASS     # data path unaffected
+    assert v[2] == abi.FWD      # data path unaffected
