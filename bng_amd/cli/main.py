@@ -757,10 +757,13 @@ class BNG:
         (ref XDP_PASS -> server4 / AF_PACKET -> pppoe)."""
         from ..dataplane.packets import parse_dhcp_frame
         from ..dhcp import message as dm
-        if len(frame) >= 14 and getattr(self, "pppoe", None) is not None:
+        if len(frame) >= 14:
             et = int.from_bytes(frame[12:14], "big")
-            if et in (0x8863, 0x8864):
+            if et in (0x8863, 0x8864) and \
+                    getattr(self, "pppoe", None) is not None:
                 return self.pppoe.handle_frame(frame) or None
+            if et == 0x0806:
+                return self._arp_reply(frame)
         try:
             p = parse_dhcp_frame(frame)
         except (AssertionError, IndexError, ValueError):
@@ -772,6 +775,27 @@ class BNG:
             return None
         resp = self.dhcp_server.handle(msg)
         return resp.encode() if resp else None
+
+    def _arp_reply(self, frame: bytes):
+        """Answer who-has for our server IP (the reference leans on
+        the kernel stack for ARP; a userspace NIC edge must answer it
+        itself or no subscriber traffic ever arrives)."""
+        import struct as st
+        from ..dataplane.packets import ip2u32, mac_bytes
+        if len(frame) < 42:
+            return None
+        htype, ptype, hlen, plen, op = st.unpack_from(">HHBBH", frame, 14)
+        if (htype, ptype, hlen, plen, op) != (1, 0x0800, 6, 4, 1):
+            return None
+        sha = frame[22:28]
+        spa = frame[28:32]
+        tpa = frame[38:42]
+        if int.from_bytes(tpa, "big") != ip2u32(self.args.server_ip):
+            return None
+        our_mac = mac_bytes(self.args.server_mac)
+        return (sha + our_mac + b"\x08\x06" +
+                st.pack(">HHBBH", 1, 0x0800, 6, 4, 2) +
+                our_mac + tpa + sha + spa)
 
     def _coa_lookup(self, req):
         for lease in self.dhcp_server.leases.values():

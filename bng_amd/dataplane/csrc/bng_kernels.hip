@@ -1542,7 +1542,7 @@ void uplink_pipeline_kernel(bng_uplink_params P) {
        * them back out the wire instead, so the pump equivalent of
        * "continue the stack" is PASS (cli _frame_slow_path). */
       uint16_t et = len >= 14 ? (uint16_t)((p[12] << 8) | p[13]) : 0;
-      if (et == 0x8863 || et == 0x8864) {
+      if (et == 0x8863 || et == 0x8864 || et == 0x0806) {
         P.verdict[pid] = (uint8_t)BNG_PASS;
         P.out_len[pid] = ol;
         continue;

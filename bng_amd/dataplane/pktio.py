@@ -314,9 +314,10 @@ class Pump:
             # not down the data pipeline
             import struct as _st
             for fr in frames:
-                # PPPoE ethertypes -> slow path (see uplink kernel note)
+                # PPPoE/ARP ethertypes -> slow path (see uplink
+                # kernel note; ARP for the gateway is answered there)
                 if len(fr) >= 14 and _st.unpack_from(">H", fr, 12)[0] \
-                        in (0x8863, 0x8864):
+                        in (0x8863, 0x8864, 0x0806):
                     passed.append(fr)
                     self.stats["passed"] += 1
                     continue

@@ -409,3 +409,44 @@ class TestPppoeOverPump:
             assert pados, "no PADO on the wire"
         finally:
             bng.stop()
+
+
+class TestArpOverPump:
+    def test_who_has_gateway_gets_reply(self):
+        """ARP who-has for the BNG's server IP is answered through the
+        pump slow path (a userspace NIC edge must own ARP)."""
+        import struct as st
+        from bng_amd.cli.main import BNG, build_parser
+        from bng_amd.dataplane.packets import ip2u32, mac_bytes
+        from bng_amd.dataplane.pktio import ListSink, Pump, \
+            SyntheticSource
+        args = build_parser().parse_args([
+            "run", "--gpu", "off", "--pool-network", "10.0.6.0/24",
+            "--server-ip", "10.0.6.1"])
+        bng = BNG(args).start()
+        try:
+            sink = ListSink()
+            pump = Pump(bng.launcher, SyntheticSource(lambda n: []),
+                        sink, slow_path=bng._frame_slow_path, batch=16)
+            sha = b"\xaa\xbb\xcc\x00\x00\x41"
+            req = (b"\xff" * 6 + sha + b"\x08\x06" +
+                   st.pack(">HHBBH", 1, 0x0800, 6, 4, 1) +
+                   sha + ip2u32("10.0.6.50").to_bytes(4, "big") +
+                   b"\x00" * 6 + ip2u32("10.0.6.1").to_bytes(4, "big"))
+            pump.process([req])
+            assert sink.frames, "no ARP reply"
+            rep = sink.frames[0]
+            assert rep[0:6] == sha                         # to requester
+            assert rep[6:12] == mac_bytes("02:00:00:00:00:01")
+            assert st.unpack_from(">H", rep, 20)[0] == 2   # is-at
+            assert rep[28:32] == ip2u32("10.0.6.1").to_bytes(4, "big")
+            # who-has for some OTHER ip is ignored
+            other = (b"\xff" * 6 + sha + b"\x08\x06" +
+                     st.pack(">HHBBH", 1, 0x0800, 6, 4, 1) +
+                     sha + ip2u32("10.0.6.50").to_bytes(4, "big") +
+                     b"\x00" * 6 + ip2u32("10.0.6.99").to_bytes(4, "big"))
+            n_before = len(sink.frames)
+            pump.process([other])
+            assert len(sink.frames) == n_before
+        finally:
+            bng.stop()
