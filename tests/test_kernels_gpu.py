@@ -1169,8 +1169,12 @@ def test_pppoe_ethertypes_pass_to_slow_path():
                           ip2u32("9.9.9.9"), ip2u32("8.8.8.8"),
                           proto=17, sport=1000, dport=53,
                           payload=b"x" * 22)
-    d, lens = l.make_batch([padi, sess, data_pkt])
+    import struct as st
+    arp = (b"\xff" * 6 + b"\xaa\xbb\xcc\x00\x00\x31" + b"\x08\x06" +
+           st.pack(">HHBBH", 1, 0x0800, 6, 4, 1) + b"\x00" * 20)
+    d, lens = l.make_batch([padi, sess, data_pkt, arp])
     v, _ = l.uplink(d, lens, now_ns=10**18, now_sec=10**9)
     v = v.cpu().numpy()
     assert v[0] == abi.PASS and v[1] == abi.PASS
     assert v[2] == abi.FWD      # data path unaffected
+    assert v[3] == abi.PASS     # ARP to the host edge
