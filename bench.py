@@ -216,12 +216,16 @@ def log(rank, msg):
 
 def build_cell_map(lens_np, stride):
     """NIC-RX-ring analog layout: each frame occupies ceil(len/64)
-    64-byte cells in a contiguous pinned buffer (DHCP frames get a full
-    8-cell slot so in-place OFFER growth never spills).  Returns
-    (n_cells_total, slot_cell_idx[int64 C]) where slot_cell_idx[c] is
-    the destination 64-B cell inside the [n, stride] slot layout."""
+    64-byte cells in a contiguous pinned buffer; DHCP-sized frames
+    (len>64) carry a 64-byte growth budget so the in-place OFFER/ACK
+    (fixed option set, always < input+64) never spills its cells.
+    Returns (n_cells_total, slot_cell_idx[int64 C]) where
+    slot_cell_idx[c] is the destination 64-B cell inside the
+    [n, stride] slot layout."""
     cells_per_slot = stride // 64
-    cells = np.where(lens_np > 64, cells_per_slot,
+    grown = np.minimum((lens_np.astype(np.int64) + 64 + 63) // 64,
+                       cells_per_slot)
+    cells = np.where(lens_np > 64, grown,
                      (lens_np.astype(np.int64) + 63) // 64)
     off = np.zeros(len(lens_np) + 1, dtype=np.int64)
     np.cumsum(cells, out=off[1:])
