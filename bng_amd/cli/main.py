@@ -630,13 +630,10 @@ class BNG:
             if a.nat_alg_sip:
                 natflags |= _abi.NAT_FLAG_ALG_SIP
                 alg_ports += [(5060, 17), (5060, 6)]
-            self.launcher.set_nat_config(
-                flags=natflags,
-                ports_per_sub=a.nat_ports_per_subscriber,
-                alg_ports=alg_ports)
             self.nat = NATMgr(self.launcher,
                               ports_per_subscriber=a.nat_ports_per_subscriber,
-                              flags=natflags, logger=logger)
+                              flags=natflags, logger=logger,
+                              alg_ports=alg_ports)
             for ip in a.nat_public_ip:
                 self.nat.add_public_ip(ip)
             self.nat.start()

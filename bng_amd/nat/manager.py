@@ -35,7 +35,8 @@ class Manager:
     PORT_RANGE_END = 65535
 
     def __init__(self, launcher=None, ports_per_subscriber: int = 1024,
-                 flags: int = abi.NAT_FLAG_EIM, logger=None):
+                 flags: int = abi.NAT_FLAG_EIM, logger=None,
+                 alg_ports=None):
         self.launcher = launcher
         self.ports_per_sub = ports_per_subscriber
         self.flags = flags
@@ -44,7 +45,7 @@ class Manager:
         self.allocations: Dict[int, NATAllocation] = {}  # by private ip
         self._sub_counter = 0
         self._lock = threading.RLock()
-        self._alg_ports: List[Tuple[int, int]] = []
+        self._alg_ports: List[Tuple[int, int]] = list(alg_ports or [])
         self._hairpin = False
         self._stop = threading.Event()
         self._drain_thread: Optional[threading.Thread] = None

@@ -450,3 +450,29 @@ class TestArpOverPump:
             assert len(sink.frames) == n_before
         finally:
             bng.stop()
+
+
+class TestNatModeFlagWiring:
+    def test_flags_reach_dataplane_config(self):
+        """--nat-eim/eif/hairpin/alg-* fold into NAT_FLAG_* and the ALG
+        punt ports in the golden dataplane config."""
+        from bng_amd.cli.main import BNG, build_parser
+        from bng_amd.dataplane import abi
+        args = build_parser().parse_args([
+            "run", "--gpu", "off", "--pool-network", "10.0.7.0/24",
+            "--nat-enabled", "--nat-public-ip", "203.0.113.9",
+            "--nat-hairpin", "false", "--nat-alg-sip", "true",
+            "--nat-ports-per-sub", "512"])
+        bng = BNG(args).start()
+        try:
+            dp = bng.launcher.dp
+            assert dp.nat_flags & abi.NAT_FLAG_EIM
+            assert dp.nat_flags & abi.NAT_FLAG_EIF
+            assert not (dp.nat_flags & abi.NAT_FLAG_HAIRPIN)
+            assert dp.nat_flags & abi.NAT_FLAG_ALG_FTP
+            assert dp.nat_flags & abi.NAT_FLAG_ALG_SIP
+            assert (21, 6) in dp.alg_ports
+            assert (5060, 17) in dp.alg_ports
+            assert bng.nat.ports_per_sub == 512
+        finally:
+            bng.stop()
