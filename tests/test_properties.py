@@ -204,3 +204,44 @@ def test_interval_fold_equals_mask_scan(prefixes, probe_ip):
     linear = any((probe_ip & m) == n for n, m in ranges)
     binary = any(lo <= probe_ip <= hi for lo, hi in iv)
     assert binary == linear
+
+
+@given(st.lists(st.tuples(st.sampled_from(["put", "del", "restart",
+                                           "compact"]),
+                          st.integers(0, 7),
+                          st.binary(min_size=0, max_size=6)),
+                max_size=40))
+@settings(max_examples=60, deadline=None)
+def test_clset_wal_restart_equivalence(ops):
+    """Any interleaving of puts/deletes/compactions/restarts leaves the
+    persisted CLSet exactly equal to an in-memory reference model
+    (crash points modeled by restart-without-close: WAL replay)."""
+    import shutil
+    import tempfile
+    from bng_amd.nexus.clset import CLSetStore
+    d = tempfile.mkdtemp()
+    try:
+        s = CLSetStore("n", data_dir=d)
+        model = {}
+        for op, k, v in ops:
+            key = f"k{k}"
+            if op == "put":
+                s.put(key, v)
+                model[key] = v
+            elif op == "del":
+                s.delete(key)
+                model.pop(key, None)
+            elif op == "compact":
+                s.compact()
+            else:                      # crash + restart (no close())
+                s._wal.flush()
+                s = CLSetStore("n", data_dir=d)
+        for key in [f"k{i}" for i in range(8)]:
+            assert s.get(key) == model.get(key), key
+        s.close()
+        s2 = CLSetStore("n", data_dir=d)
+        for key in [f"k{i}" for i in range(8)]:
+            assert s2.get(key) == model.get(key), key
+        s2.close()
+    finally:
+        shutil.rmtree(d, ignore_errors=True)
