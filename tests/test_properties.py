@@ -245,3 +245,59 @@ def test_clset_wal_restart_equivalence(ops):
         s2.close()
     finally:
         shutil.rmtree(d, ignore_errors=True)
+
+
+@given(st.lists(st.tuples(st.integers(1, 250),        # sub index
+                          st.integers(1, 0xDFFFFFFF),  # dst ip
+                          st.integers(1024, 65535),    # sport
+                          st.sampled_from([6, 17])),
+                min_size=1, max_size=30, unique=True),
+       st.booleans())
+@settings(max_examples=40, deadline=None)
+def test_nat_ha_export_import_roundtrip(flows, eim_on):
+    """Golden-level NAT HA: establish arbitrary flows, export, import
+    into a FRESH dataplane — the re-export must describe the same flow
+    set, and every flow's return packet must translate identically."""
+    from bng_amd.dataplane import abi as A
+    from bng_amd.dataplane.launcher import GoldenLauncher
+    from bng_amd.dataplane.packets import build_ipv4, ip2u32
+
+    def mk():
+        l = GoldenLauncher()
+        l.set_nat_config(flags=A.NAT_FLAG_EIM if eim_on else 0)
+        for i in range(1, 251):
+            l.add_subscriber_nat(ip2u32("10.0.0.0") + i,
+                                 ip2u32("203.0.113.1"),
+                                 1024 + (i % 60) * 1024,
+                                 1024 + (i % 60) * 1024 + 1023,
+                                 subscriber_id=i)
+        return l
+
+    a, b = mk(), mk()
+    returns = []
+    for sub, dst, sport, proto in flows:
+        pkt = bytearray(build_ipv4(
+            "aa:00:00:00:00:01", "02:00:00:00:00:01",
+            ip2u32("10.0.0.0") + sub, dst, proto=proto,
+            sport=sport, dport=443, payload=b"x" * 22))
+        v = a.dp.nat44_egress(pkt)
+        if v != A.FWD:
+            continue
+        nat_port = int.from_bytes(pkt[34:36], "big")
+        ret = build_ipv4("02:00:00:00:00:01", "aa:00:00:00:00:01",
+                         dst, ip2u32("203.0.113.1"), proto=proto,
+                         sport=443, dport=nat_port, payload=b"y" * 22)
+        returns.append((ret, ip2u32("10.0.0.0") + sub, sport))
+    recs = a.export_nat_sessions()
+    assert b.import_nat_sessions(recs) == len(recs)
+    back = b.export_nat_sessions()
+    key = lambda r: (int(r["src_ip"]), int(r["dst_ip"]),
+                     int(r["src_port"]), int(r["dst_port"]),
+                     int(r["protocol"]), int(r["nat_ip"]),
+                     int(r["nat_port"]))
+    assert sorted(map(key, recs)) == sorted(map(key, back))
+    for ret, want_ip, want_port in returns:
+        fb = bytearray(ret)
+        assert b.dp.nat44_ingress(fb) == A.FWD
+        assert int.from_bytes(fb[30:34], "big") == want_ip
+        assert int.from_bytes(fb[36:38], "big") == want_port
