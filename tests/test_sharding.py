@@ -338,3 +338,73 @@ def test_steer_all_world4_preallocated(tmp_path):
         p.join(timeout=240)
         assert p.exitcode == 0
     assert tot == world * 512
+
+
+def _full_sequence_worker(rank, world, rendezvous_file, q):
+    """The COMPLETE bench collective sequence at world N over gloo:
+    overlapped prep/exchange steps, end barrier+MAX-allreduce, the
+    rank-0-only latency phase while others hold at a barrier, final
+    barrier — the exact order the 8-GPU SCALE run executes."""
+    import torch.distributed as dist
+    dist.init_process_group(
+        "gloo", init_method=f"file://{rendezvous_file}",
+        rank=rank, world_size=world)
+    from bng_amd.parallel.hashring import owner_of_ip, owner_of_mac
+    from bng_amd.parallel.sharding import ExchangeBuffers, exchange
+    import bench
+
+    steps, nbuf, n, stride = 5, 4, 128, 512
+    bufs = [ExchangeBuffers(4 * n, stride, "cpu", world)
+            for _ in range(nbuf)]
+    batches = [None] * nbuf
+
+    def prep(k):
+        d_np, l_np = bench.gen_batch(n, 5_000, 0.2, stride,
+                                     seed=4000 + rank * 100 + k,
+                                     rank=rank, world=world)
+        owners = []
+        for i in range(n):
+            if l_np[i] > 64:
+                owners.append(owner_of_mac(bytes(d_np[i, 70:76]), world))
+            else:
+                owners.append(owner_of_ip(
+                    int.from_bytes(bytes(d_np[i, 26:30]), "big"), world))
+        batches[k % nbuf] = exchange(
+            torch.from_numpy(d_np), torch.from_numpy(l_np.view(np.int16)),
+            torch.tensor(owners, dtype=torch.int64), bufs=bufs[k % nbuf])
+
+    prep(0)
+    total = 0
+    for k in range(steps):
+        d, l = batches[k % nbuf]
+        total += l.numel()
+        prep(k + 1)
+    dist.barrier()
+    e = torch.tensor([float(rank + 1)])
+    dist.all_reduce(e, op=dist.ReduceOp.MAX)     # elapsed MAX pattern
+    assert e.item() == world
+    if rank == 0:
+        pass                                      # latency phase (rank 0)
+    dist.barrier()                                # others hold here
+    dist.barrier()                                # post-svc/host-io hold
+    q.put((rank, total))
+    dist.destroy_process_group()
+
+
+def test_full_bench_sequence_world4(tmp_path):
+    world = 4
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    rv = str(tmp_path / "rdv_full4")
+    procs = [ctx.Process(target=_full_sequence_worker,
+                         args=(r, world, rv, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    tot = 0
+    for _ in range(world):
+        _, t = q.get(timeout=240)
+        tot += t
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    assert tot == world * 5 * 128
