@@ -476,3 +476,86 @@ class TestNatModeFlagWiring:
             assert bng.nat.ports_per_sub == 512
         finally:
             bng.stop()
+
+
+class TestCliHaFailover:
+    def test_health_driven_promotion_restores_state(self):
+        """Two full `bng run` instances: the active serves a lease, a
+        NAT block and a live NAT flow; when it dies, the standby's
+        health monitor drives promotion and leases, exact port blocks
+        AND the established flow are restored (VERDICT r1 task 3 at
+        the CLI level)."""
+        from bng_amd.cli.main import BNG, build_parser
+        from bng_amd.dataplane import abi
+        from bng_amd.dataplane.packets import build_ipv4, ip2u32
+        from bng_amd.dhcp import message as dm
+
+        active = BNG(build_parser().parse_args([
+            "run", "--gpu", "off", "--node-id", "act",
+            "--pool-network", "10.0.8.0/24", "--ha-role", "active",
+            "--nat-enabled", "--nat-public-ip", "203.0.113.77",
+        ])).start()
+        try:
+            mac = bytes.fromhex("aabbcc000051")
+            offer = active.dhcp_server.handle(
+                dm.build_request(mac, dm.DISCOVER))
+            ack = active.dhcp_server.handle(dm.build_request(
+                mac, dm.REQUEST, requested_ip=offer.yiaddr))
+            sub_ip = ack.yiaddr
+            active.nat.allocate_nat(sub_ip, "sub-51")
+            blk = active.nat.allocations[sub_ip]
+            # re-publish the lease so the delta carries the NAT block
+            # (ACK happened before allocate_nat in this manual flow)
+            from bng_amd.ha import session_glue
+            active.ha.publish_add(
+                session_glue.lease_to_session(
+                    active.dhcp_server.leases[mac]))
+            s = active.ha.store.get(f"dhcp-{mac.hex()}")
+            # establish a flow through the active dataplane
+            pkt = bytearray(build_ipv4(
+                "aa:bb:cc:00:00:51", "02:00:00:00:00:01", sub_ip,
+                ip2u32("93.184.216.34"), proto=17, sport=7001,
+                dport=443, payload=b"x" * 22))
+            assert active.launcher.dp.nat44_egress(pkt) == abi.FWD
+            nat_port = int.from_bytes(pkt[34:36], "big")
+            assert active._nat_ha.pump_once() == 1
+
+            standby = BNG(build_parser().parse_args([
+                "run", "--gpu", "off", "--node-id", "sby",
+                "--pool-network", "10.0.8.0/24", "--ha-role", "standby",
+                "--ha-partner-url", active.ha.url,
+                "--nat-enabled", "--nat-public-ip", "203.0.113.77",
+            ])).start()
+            try:
+                deadline = time.time() + 8
+                while (standby.ha.store.count() < 1 or
+                       not standby.ha.nat_store) and \
+                        time.time() < deadline:
+                    time.sleep(0.05)
+                assert standby.ha.store.count() >= 1
+                assert len(standby.ha.nat_store) == 1
+                # the active dies
+                active.stop()
+                deadline = time.time() + 15
+                while standby.ha.role != "active" and \
+                        time.time() < deadline:
+                    time.sleep(0.2)
+                assert standby.ha.role == "active", "no promotion"
+                # lease + exact block + live flow restored
+                assert mac in standby.dhcp_server.leases
+                blk2 = standby.nat.allocations.get(sub_ip)
+                assert blk2 is not None
+                assert (blk2.public_ip, blk2.port_start, blk2.port_end) \
+                    == (blk.public_ip, blk.port_start, blk.port_end)
+                ret = bytearray(build_ipv4(
+                    "02:00:00:00:00:01", "aa:bb:cc:00:00:51",
+                    ip2u32("93.184.216.34"), ip2u32("203.0.113.77"),
+                    proto=17, sport=443, dport=nat_port,
+                    payload=b"y" * 22))
+                assert standby.launcher.dp.nat44_ingress(ret) == abi.FWD
+                assert int.from_bytes(ret[30:34], "big") == sub_ip
+                assert int.from_bytes(ret[36:38], "big") == 7001
+            finally:
+                standby.stop()
+        finally:
+            active.stop()
