@@ -235,3 +235,52 @@ class TestDecline:
         # quarantine expires
         srv._declined[addr1] = 0.0
         assert not srv._is_declined(addr1)
+
+
+def test_nested_relay_chain():
+    """Two-level RELAY-FORW chain (RFC 8415 §19 multi-hop): the reply
+    is a matching nested RELAY-REPL with Interface-Id echoed at each
+    level and the inner REPLY intact."""
+    import struct as st
+    from bng_amd.dhcpv6.server import (DHCPv6Message, DHCPv6Server,
+                                       OPT_CLIENTID, OPT_IA_NA, RELAY_FORW,
+                                       RELAY_REPL, REPLY, SOLICIT)
+    srv = DHCPv6Server(rapid_commit=True)
+    duid = b"\x00\x01duid-r"
+    m = DHCPv6Message(SOLICIT, 0x333)
+    m.add(OPT_CLIENTID, duid)
+    m.add(14, b"")            # rapid commit
+    m.add(OPT_IA_NA, st.pack(">III", 9, 0, 0))
+    inner = m.encode()
+
+    def wrap(payload, hop, ifid):
+        out = bytes([RELAY_FORW, hop]) + b"\x00" * 32
+        out += st.pack(">HH", 18, len(ifid)) + ifid          # iface-id
+        out += st.pack(">HH", 9, len(payload)) + payload     # relay-msg
+        return out
+
+    lvl1 = wrap(inner, 0, b"eth-cust")
+    lvl2 = wrap(lvl1, 1, b"eth-aggr")
+    resp = srv.handle(lvl2)
+    assert resp is not None and resp[0] == RELAY_REPL and resp[1] == 1
+
+    def unwrap(data, want_ifid):
+        i = 34
+        msg = None
+        ifid = None
+        while i + 4 <= len(data):
+            t, ln = st.unpack_from(">HH", data, i)
+            if t == 9:
+                msg = data[i + 4:i + 4 + ln]
+            elif t == 18:
+                ifid = data[i + 4:i + 4 + ln]
+            i += 4 + ln
+        assert ifid == want_ifid
+        return msg
+
+    lvl1_rep = unwrap(resp, b"eth-aggr")
+    assert lvl1_rep[0] == RELAY_REPL and lvl1_rep[1] == 0
+    final = unwrap(lvl1_rep, b"eth-cust")
+    dec = DHCPv6Message.decode(final)
+    assert dec.msg_type == REPLY and dec.txn_id == 0x333
+    assert dec.get(OPT_IA_NA) is not None
