@@ -71,3 +71,52 @@ def test_cell_map_dhcp_growth_headroom():
     lens = np.array([70, 342, 500], dtype=np.uint16)  # all DHCP-sized
     C, idx = bench.build_cell_map(lens, 512)
     assert C == 3 + 7 + 8     # ceil((len+64)/64), capped at 8
+
+
+def test_bench_downlink_crafting_matches_golden():
+    """The bench downlink phase builds return packets from exported
+    session records; on the golden dataplane every crafted packet must
+    DNAT back to its subscriber (protects the driver-visible
+    downlink_mpps from crafting regressions)."""
+    from bng_amd.dataplane import abi
+    from bng_amd.dataplane.launcher import GoldenLauncher
+    from bng_amd.dataplane.packets import build_ipv4, ip2u32
+
+    g = GoldenLauncher()
+    g.set_nat_config()
+    for i in range(1, 40):
+        g.add_subscriber_nat(ip2u32("10.0.0.0") + i,
+                             ip2u32("203.0.113.1"),
+                             1024 + i * 1024, 1024 + i * 1024 + 1023,
+                             subscriber_id=i)
+    for i in range(1, 40):
+        pkt = bytearray(build_ipv4(
+            "aa:00:00:00:00:01", "02:00:00:00:00:01",
+            ip2u32("10.0.0.0") + i, ip2u32("93.184.216.34"),
+            proto=17, sport=40000 + i, dport=53, payload=b"x" * 22))
+        assert g.dp.nat44_egress(pkt) == abi.FWD
+    recs = g.export_nat_sessions()
+    assert len(recs) == 39
+
+    # exactly the bench crafting: template + field splice from records
+    t = np.frombuffer(build_ipv4(
+        "02:00:00:00:00:01", "aa:00:00:00:00:00",
+        ip2u32("93.184.216.34"), ip2u32("203.0.113.1"),
+        proto=17, sport=53, dport=1024, payload=b"\x00" * 22),
+        dtype=np.uint8)
+    ret = np.zeros((39, 512), dtype=np.uint8)
+    idxs = np.arange(39)
+    ret[:, :64] = t
+    ret[:, 26:30] = recs["dst_ip"][idxs].astype(">u4") \
+        .view(np.uint8).reshape(-1, 4)
+    ret[:, 30:34] = recs["nat_ip"][idxs].astype(">u4") \
+        .view(np.uint8).reshape(-1, 4)
+    ret[:, 34:36] = recs["dst_port"][idxs].astype(">u2") \
+        .view(np.uint8).reshape(-1, 2)
+    ret[:, 36:38] = recs["nat_port"][idxs].astype(">u2") \
+        .view(np.uint8).reshape(-1, 2)
+    for i in range(39):
+        fb = bytearray(ret[i, :64].tobytes())
+        assert g.dp.nat44_ingress(fb) == abi.FWD, f"row {i}"
+        assert int.from_bytes(fb[30:34], "big") == \
+            int(recs["src_ip"][i]), f"row {i} dst"
