@@ -65,6 +65,11 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--pktio-max-wait", type=float, default=0.0005,
                    help="batch deadline seconds (latency/throughput "
                         "trade at the NIC edge)")
+    g.add_argument("--dhcp-serve", default="batched",
+                   choices=["batched", "persistent"],
+                   help="persistent = route DHCP through the resident "
+                        "service kernel (~31us p50) instead of the "
+                        "batched pipeline (GPU only)")
     g.add_argument("--bpf-path", default="",
                    help="accepted for reference CLI compatibility; the "
                         "MI355X dataplane compiles in-tree HIP kernels "
@@ -739,10 +744,19 @@ class BNG:
                 io = AFPacketIO(a.interface)
                 self.log.info("AF_PACKET raw socket on %s", a.interface)
             self.pktio = io
+            svc = None
+            if a.dhcp_serve == "persistent" and use_gpu:
+                from ..dataplane.launcher import DhcpService
+                svc = DhcpService(self.launcher,
+                                  n_slots=max(256, a.pktio_batch))
+                self._defer(svc.stop)
+                self.dhcp_service = svc
+                self.log.info("persistent DHCP service attached")
             self.pump = Pump(self.launcher, io, io,
                              slow_path=self._frame_slow_path,
                              batch=a.pktio_batch,
-                             max_wait=a.pktio_max_wait).start()
+                             max_wait=a.pktio_max_wait,
+                             dhcp_service=svc).start()
             self._defer(self.pump.stop)
             self._defer(io.close)
         return self

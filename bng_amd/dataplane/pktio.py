@@ -208,11 +208,16 @@ class Pump:
     def __init__(self, launcher, source, sink=None,
                  slow_path: Optional[Callable[[bytes], Optional[bytes]]] = None,
                  batch: int = 8192, stride: int = 512,
-                 max_wait: float = 0.0005, sort_by_type: bool = True):
+                 max_wait: float = 0.0005, sort_by_type: bool = True,
+                 dhcp_service=None):
         self.launcher = launcher
         self.source = source
         self.sink = sink
         self.slow_path = slow_path
+        # optional persistent-kernel serving path: DHCP frames bypass
+        # the batched kernel and go through the resident service's
+        # doorbell (~31us p50 vs one batch residency)
+        self.dhcp_service = dhcp_service
         self.batch = batch
         self.stride = stride
         self.max_wait = max_wait
@@ -255,6 +260,30 @@ class Pump:
             import numpy as np
             import torch
             data_np, lens_np = pack_frames(frames, self.stride)
+            svc_mask = None
+            if self.dhcp_service is not None and len(frames):
+                # host-side pre-class: UDP dst 67 on untagged IPv4
+                svc_mask = ((lens_np >= 42) &
+                            (data_np[:, 12] == 0x08) &
+                            (data_np[:, 13] == 0x00) &
+                            (data_np[:, 23] == 17) &
+                            (data_np[:, 36] == 0) &
+                            (data_np[:, 37] == 67))
+                if not svc_mask.any():
+                    svc_mask = None
+            svc_frames = None
+            if svc_mask is not None:
+                rows = np.nonzero(~svc_mask)[0]
+                srows = np.nonzero(svc_mask)[0]
+                svc_frames = [frames[i] for i in srows]
+                sv, sol, srep = self.dhcp_service.serve(
+                    data_np[srows], lens_np[srows],
+                    int(time.time()))
+                sv = sv.copy()
+                sol = sol.view(np.uint16).copy()
+                srep = srep.copy()
+                data_np, lens_np = data_np[rows], lens_np[rows]
+                frames = [frames[i] for i in rows]
             data = torch.from_numpy(data_np).to(self.launcher.device)
             lens = torch.from_numpy(lens_np.view(np.int16)).to(
                 self.launcher.device)
@@ -275,6 +304,16 @@ class Pump:
             out_mask = tx | fwd
             out_lens = np.where(tx, ol, lens_np).astype(np.uint16)[out_mask]
             out_data = host[out_mask]
+            if svc_mask is not None:
+                stx = sv == abi.TX
+                self.stats["tx"] += int(stx.sum())
+                self.stats["passed"] += int((sv == abi.PASS).sum())
+                if stx.any():
+                    out_data = np.concatenate([out_data, srep[stx]])
+                    out_lens = np.concatenate([out_lens, sol[stx]])
+                # service misses (unknown subscriber) -> slow path
+                passed.extend(svc_frames[i] for i in
+                              np.nonzero(sv == abi.PASS)[0])
             # PASS frames go to the slow path (few: cache misses only)
             passed = [frames[i] for i in np.nonzero(pas)[0]]
             slow_replies: List[bytes] = []
