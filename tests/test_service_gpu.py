@@ -226,6 +226,11 @@ def test_pump_routes_dhcp_through_service(launcher):
                           ip2u32("9.9.9.9"), ip2u32("8.8.8.8"),
                           proto=17, sport=1000, dport=53,
                           payload=b"x" * 22)
+    # the pump serves at wall-clock time: give the known subscriber a
+    # live lease (the fixture's NOW constant is in the past)
+    wall = int(time.time())
+    launcher.add_subscriber(0xAA0000000005, 1, ip2u32("10.0.0.2") + 5,
+                            wall + 3600)
     slow_hits = []
     with DhcpService(launcher, n_slots=256,
                      idle_exit_k=120_000) as svc:
@@ -241,7 +246,9 @@ def test_pump_routes_dhcp_through_service(launcher):
         assert len(slow_hits) == 1            # unknown -> slow path
         # the OFFER bytes match the golden twin
         g = _golden_twin()
-        gv, gfr = g.process_dhcp([known], now_sec=int(time.time()))[0]
+        g.add_subscriber(0xAA0000000005, 1, ip2u32("10.0.0.2") + 5,
+                         wall + 3600)
+        gv, gfr = g.process_dhcp([known], now_sec=wall)[0]
         offers = [bytes(d[i, :l]) for d, ls in sink.batches
                   for i, l in enumerate(ls)]
         dhcp_offers = [o for o in offers if len(o) > 240]
