@@ -304,6 +304,8 @@ class Pump:
             out_mask = tx | fwd
             out_lens = np.where(tx, ol, lens_np).astype(np.uint16)[out_mask]
             out_data = host[out_mask]
+            # PASS frames go to the slow path (few: cache misses only)
+            passed = [frames[i] for i in np.nonzero(pas)[0]]
             if svc_mask is not None:
                 stx = sv == abi.TX
                 self.stats["tx"] += int(stx.sum())
@@ -314,8 +316,6 @@ class Pump:
                 # service misses (unknown subscriber) -> slow path
                 passed.extend(svc_frames[i] for i in
                               np.nonzero(sv == abi.PASS)[0])
-            # PASS frames go to the slow path (few: cache misses only)
-            passed = [frames[i] for i in np.nonzero(pas)[0]]
             slow_replies: List[bytes] = []
             if self.slow_path is not None:
                 for fr in passed:
