@@ -759,6 +759,25 @@ class BNG:
                              dhcp_service=svc).start()
             self._defer(self.pump.stop)
             self._defer(io.close)
+            # Optional second pump on the core-side NIC running the
+            # downlink pipeline (ref tc_egress on the NAT outside
+            # interface, tc.c / loader.go) — DNAT'd return traffic is
+            # forwarded back out the access NIC.
+            dn_if = a.nat_outside_interface
+            if dn_if and dn_if != a.interface:
+                if a.pktio == "afxdp":
+                    from ..dataplane.afxdp import XskSocket
+                    dio = XskSocket(dn_if, mode="auto")
+                else:
+                    dio = AFPacketIO(dn_if)
+                self.log.info("downlink pump on %s", dn_if)
+                self.pktio_downlink = dio
+                self.pump_downlink = Pump(
+                    self.launcher, dio, io, direction="downlink",
+                    batch=a.pktio_batch,
+                    max_wait=a.pktio_max_wait).start()
+                self._defer(self.pump_downlink.stop)
+                self._defer(dio.close)
         return self
 
     def _frame_slow_path(self, frame: bytes):

@@ -209,7 +209,12 @@ class Pump:
                  slow_path: Optional[Callable[[bytes], Optional[bytes]]] = None,
                  batch: int = 8192, stride: int = 512,
                  max_wait: float = 0.0005, sort_by_type: bool = True,
-                 dhcp_service=None):
+                 dhcp_service=None, direction: str = "uplink"):
+        if direction not in ("uplink", "downlink"):
+            raise ValueError("direction must be 'uplink' or 'downlink'")
+        # "downlink" pumps the core-side NIC: NAT44 DNAT -> QoS egress
+        # (ref tc_egress hook, tc.c) — return traffic toward subscribers
+        self.direction = direction
         self.launcher = launcher
         self.source = source
         self.sink = sink
@@ -253,6 +258,8 @@ class Pump:
         """Run one batch; returns (out_frames, passed_frames)."""
         self.stats["batches"] += 1
         self.stats["rx"] += len(frames)
+        if self.direction == "downlink":
+            return self._process_downlink(frames)
         is_gpu = hasattr(self.launcher, "make_batch")
         out_frames: List[bytes] = []
         passed: List[bytes] = []
@@ -402,6 +409,60 @@ class Pump:
                     else:
                         out_frames.append(reply)
                         self.stats["slow_replies"] += 1
+        if self.sink is not None and out_frames:
+            self.sink.send_batch(out_frames)
+        return out_frames, passed
+
+    def _process_downlink(self, frames: List[bytes]) -> Tuple[List[bytes], List[bytes]]:
+        """Core-side batch: fused NAT44 DNAT -> QoS-egress.  The
+        downlink kernel rewrites frames in place and returns a verdict
+        per frame (FWD toward the subscriber, DROP on shaping).  No
+        DHCP/ARP/PPPoE handling here — control traffic lives on the
+        access side."""
+        is_gpu = hasattr(self.launcher, "make_batch")
+        out_frames: List[bytes] = []
+        passed: List[bytes] = []
+        if is_gpu:
+            import numpy as np
+            import torch
+            data_np, lens_np = pack_frames(frames, self.stride)
+            data = torch.from_numpy(data_np).to(self.launcher.device)
+            lens = torch.from_numpy(lens_np.view(np.int16)).to(
+                self.launcher.device)
+            verdict = self.launcher.downlink(data, lens)
+            v = verdict.cpu().numpy()
+            host = data.cpu().numpy()
+            fwd = v == abi.FWD
+            pas = v == abi.PASS
+            self.stats["fwd"] += int(fwd.sum())
+            self.stats["passed"] += int(pas.sum())
+            self.stats["dropped"] += int(len(frames) - fwd.sum() - pas.sum())
+            out_mask = fwd
+            out_data = host[out_mask]
+            out_lens = lens_np[out_mask]
+            passed = [frames[i] for i in np.nonzero(pas)[0]]
+            if self.sink is not None:
+                if hasattr(self.sink, "send_batch_array"):
+                    if len(out_lens):
+                        self.sink.send_batch_array(out_data, out_lens)
+                    return out_data, passed
+                out_frames = unpack_frames(out_data, out_lens)
+                self.sink.send_batch(out_frames)
+                return out_frames, passed
+            return unpack_frames(out_data, out_lens), passed
+        for fr in frames:
+            fb = bytearray(fr)
+            vd = self.launcher.dp.nat44_ingress(fb)
+            if vd == abi.FWD:
+                vd = self.launcher.dp.qos(bytes(fb), "egress")
+            if vd == abi.FWD:
+                out_frames.append(bytes(fb))
+                self.stats["fwd"] += 1
+            elif vd == abi.PASS:
+                passed.append(fr)
+                self.stats["passed"] += 1
+            else:
+                self.stats["dropped"] += 1
         if self.sink is not None and out_frames:
             self.sink.send_batch(out_frames)
         return out_frames, passed

@@ -285,3 +285,90 @@ class TestPumpThroughput:
         assert (data[1] == 2).all()          # clipped at stride
         assert (data[2] == 0).all()
         assert (data[3] == 3).all()
+
+
+class TestDownlinkPump:
+    """Core-side pump: NAT44 DNAT -> QoS-egress (ref tc_egress, tc.c).
+    Return traffic for an established session is rewritten back to the
+    subscriber's private address and forwarded to the access sink."""
+
+    def _nat_launcher(self):
+        import struct
+        from bng_amd.dataplane.golden import SubnatRec
+        l = GoldenLauncher()
+        l.dp.subnat[ip2u32("10.0.1.50")] = SubnatRec(
+            public_ip=ip2u32("203.0.113.1"), port_start=1024,
+            port_end=2047, next_port=1024, subscriber_id=42)
+        # create the session on the uplink side
+        out = bytearray(build_ipv4(MAC, "02:00:00:00:00:01",
+                                   ip2u32("10.0.1.50"),
+                                   ip2u32("93.184.216.34"),
+                                   proto=17, sport=5555, dport=53))
+        assert l.dp.nat44_egress(out) == abi.FWD
+        nat_port = struct.unpack_from(">H", out, 34)[0]
+        return l, nat_port
+
+    def test_downlink_dnat_forwarded(self):
+        import struct
+        l, nat_port = self._nat_launcher()
+        back = build_ipv4("02:00:00:00:00:02", "02:00:00:00:00:01",
+                          ip2u32("93.184.216.34"), ip2u32("203.0.113.1"),
+                          proto=17, sport=53, dport=nat_port)
+        sink = ListSink()
+        pump = Pump(l, SyntheticSource(lambda n: []), sink,
+                    direction="downlink", batch=64)
+        out, passed = pump.process([back])
+        assert passed == []
+        assert pump.stats["fwd"] == 1 and pump.stats["dropped"] == 0
+        assert len(sink.frames) == 1
+        fwd = sink.frames[0]
+        # DNAT restored the private destination address and port
+        assert struct.unpack_from(">I", fwd, 30)[0] == ip2u32("10.0.1.50")
+        assert struct.unpack_from(">H", fwd, 36)[0] == 5555
+
+    def test_downlink_no_session_passes_through(self):
+        l, _ = self._nat_launcher()
+        stray = build_ipv4("02:00:00:00:00:02", "02:00:00:00:00:01",
+                           ip2u32("8.8.8.8"), ip2u32("203.0.113.9"),
+                           proto=17, sport=53, dport=9999)
+        sink = ListSink()
+        pump = Pump(l, SyntheticSource(lambda n: []), sink,
+                    direction="downlink", batch=64)
+        out, _ = pump.process([stray])
+        # no reverse mapping -> forwarded unmodified (golden
+        # nat44_ingress passes unmatched traffic through, nat44.c:805)
+        assert pump.stats["fwd"] == 1
+        assert bytes(sink.frames[0]) == bytes(stray)
+
+    def test_direction_validated(self):
+        with pytest.raises(ValueError):
+            Pump(GoldenLauncher(), SyntheticSource(lambda n: []),
+                 direction="sideways")
+
+    def test_gpu_branch_downlink_partition(self):
+        """The GPU-path downlink partitioner (pack -> launcher.downlink
+        -> FWD/PASS/DROP split) exercised on CPU via a GPU-shaped
+        launcher whose downlink marks even rows FWD, row 1 PASS."""
+        import torch
+
+        class _DL(_FakeGpuLauncher):
+            def downlink(self, data, lens):
+                n = lens.numel()
+                v = np.where(np.arange(n) % 2 == 0, abi.FWD, abi.DROP)
+                if n > 1:
+                    v[1] = abi.PASS
+                return torch.from_numpy(v.astype(np.uint8))
+
+        sink = ListSink()
+        pump = Pump(_DL(), SyntheticSource(lambda n: []), sink,
+                    direction="downlink", batch=64, stride=256)
+        frames = [build_ipv4("aa:00:00:00:00:01", "02:00:00:00:00:01",
+                             ip2u32("9.9.9.9"), ip2u32("203.0.113.1"),
+                             proto=17, sport=53, dport=1024 + i)
+                  for i in range(6)]
+        out, passed = pump.process(frames)
+        assert pump.stats["fwd"] == 3
+        assert pump.stats["passed"] == 1
+        assert pump.stats["dropped"] == 2
+        assert len(sink.frames) == 3
+        assert len(passed) == 1 and bytes(passed[0]) == bytes(frames[1])
