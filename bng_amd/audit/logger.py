@@ -36,6 +36,7 @@ class Event:
     outcome: str = "success"
     details: Dict[str, str] = field(default_factory=dict)
     legal_hold: bool = False
+    severity: int = 1                    # retention.INFO
 
 
 class MemoryStorage:
@@ -55,7 +56,8 @@ class MemoryStorage:
 
     def query(self, category: str = "", subscriber: str = "",
               action: str = "", since: float = 0.0,
-              until: float = 0.0) -> List[Event]:
+              until: float = 0.0, min_severity: int = 0,
+              limit: int = 0) -> List[Event]:
         with self._lock:
             out = []
             for e in self._events:
@@ -69,8 +71,24 @@ class MemoryStorage:
                     continue
                 if until and e.timestamp > until:
                     continue
+                if min_severity and e.severity < min_severity:
+                    continue
                 out.append(e)
+                if limit and len(out) >= limit:
+                    break
             return out
+
+    def delete(self, ids) -> int:
+        """ref storage.go Delete :198-230."""
+        ids = set(ids)
+        with self._lock:
+            before = len(self._events)
+            self._events = [e for e in self._events if e.id not in ids]
+            return before - len(self._events)
+
+    def count(self) -> int:
+        with self._lock:
+            return len(self._events)
 
     def all(self) -> List[Event]:
         with self._lock:
@@ -82,6 +100,17 @@ class MemoryStorage:
             before = len(self._events)
             self._events = [e for e in self._events
                             if e.legal_hold or e.timestamp >= cutoff]
+            return before - len(self._events)
+
+    def apply_retention_manager(self, rm, now: float = 0.0) -> int:
+        """Per-category/action expiry honoring legal holds (ref
+        logger.go cleanupExpired :588-610 over
+        storage.DeleteExpired)."""
+        now = now or time.time()
+        with self._lock:
+            before = len(self._events)
+            self._events = [e for e in self._events
+                            if e.legal_hold or not rm.expired(e, now)]
             return before - len(self._events)
 
 
@@ -136,16 +165,26 @@ class SyslogExporter:
     def __init__(self, sink):
         self.sink = sink
 
+    FACILITY = 13                        # log audit (RFC5424)
+
     def export(self, ev: Event):
-        self.sink(f"<110>1 - bng audit - - - [{ev.category}] {ev.action} "
-                  f"sub={ev.subscriber} ip={ev.ip} outcome={ev.outcome}")
+        # severity -> syslog level: DEBUG=7 .. EMERGENCY=0 (ref
+        # export.go formatMessage :131-141)
+        level = max(0, 7 - ev.severity)
+        pri = self.FACILITY * 8 + level
+        self.sink(f"<{pri}>1 - bng audit - - - [{ev.category}] "
+                  f"{ev.action} sub={ev.subscriber} ip={ev.ip} "
+                  f"outcome={ev.outcome}")
 
 
 class Logger:
     """Async audit logger (ref logger.go:15-161)."""
 
     def __init__(self, storage: Optional[MemoryStorage] = None,
-                 exporters: Optional[list] = None, queue_size: int = 10000):
+                 exporters: Optional[list] = None, queue_size: int = 10000,
+                 min_severity: int = 0,
+                 disabled_categories: Optional[set] = None,
+                 retention=None):
         self.storage = storage or MemoryStorage()
         self.exporters = exporters or []
         self._q: queue.Queue = queue.Queue(maxsize=queue_size)
@@ -153,6 +192,17 @@ class Logger:
         self._thread: Optional[threading.Thread] = None
         self.dropped = 0
         self.legal_holds: set = set()       # subscriber ids under hold
+        # severity floor + category kill-switch (ref logger.go
+        # shouldLog :411-434)
+        self.min_severity = min_severity
+        self.disabled_categories = disabled_categories or set()
+        # optional RetentionManager (retention.py) for per-category
+        # expiry + criteria-matched legal holds
+        self.retention = retention
+        self.filtered = 0
+        self.logged = 0
+        self.exported = 0
+        self.export_errors = 0
 
     def start(self):
         self._thread = threading.Thread(target=self._loop, daemon=True)
@@ -172,16 +222,44 @@ class Logger:
 
     def log(self, action: str, category: str = CAT_SESSION,
             subscriber: str = "", ip: str = "", outcome: str = "success",
-            **details):
+            severity: Optional[int] = None, **details):
+        from .retention import action_severity
+        sev = severity if severity is not None else \
+            action_severity(action)
+        if sev < self.min_severity or category in self.disabled_categories:
+            self.filtered += 1
+            return
         ev = Event(id=uuid.uuid4().hex[:12], category=category,
                    action=action, timestamp=time.time(),
                    subscriber=subscriber, ip=ip, outcome=outcome,
                    details={k: str(v) for k, v in details.items()},
-                   legal_hold=subscriber in self.legal_holds)
+                   legal_hold=subscriber in self.legal_holds,
+                   severity=sev)
+        if self.retention is not None and not ev.legal_hold:
+            ev.legal_hold = self.retention.is_under_hold(ev)
         try:
             self._q.put_nowait(ev)
+            self.logged += 1
         except queue.Full:
             self.dropped += 1
+
+    def stats(self) -> Dict[str, int]:
+        """ref LoggerStats logger.go:150-159."""
+        return {"logged": self.logged, "dropped": self.dropped,
+                "filtered": self.filtered,
+                "stored": self.storage.count()
+                if hasattr(self.storage, "count") else 0,
+                "exported": self.exported,
+                "export_errors": self.export_errors}
+
+    def cleanup_expired(self, now: float = 0.0) -> int:
+        """Apply the retention manager to storage (ref logger.go
+        retentionLoop/cleanupExpired :562-610)."""
+        if self.retention is None or \
+                not hasattr(self.storage, "apply_retention_manager"):
+            return 0
+        self.retention.cleanup_expired_holds()
+        return self.storage.apply_retention_manager(self.retention, now)
 
     def _loop(self):
         while not self._stop.is_set():
@@ -196,8 +274,9 @@ class Logger:
         for ex in self.exporters:
             try:
                 ex.export(ev)
+                self.exported += 1
             except Exception:
-                pass
+                self.export_errors += 1
 
     def flush(self):
         while True:

@@ -454,3 +454,131 @@ class TestIPFIXAndKafkaExporters:
         x2.export(self._nat_event())
         assert got[-1] == ("bng-audit", "sub-1")
         assert x.exported == 1 and x2.exported == 1
+
+
+class TestAuditSeverityRetention:
+    """Severity model + RetentionManager (ref pkg/audit
+    types.go:254-403, retention.go:9-357)."""
+
+    def test_action_severity_and_category(self):
+        from bng_amd.audit import retention as rt
+        assert rt.action_severity("mac_spoof") == rt.CRITICAL
+        assert rt.action_severity("brute_force_detected") == rt.ALERT
+        assert rt.action_severity("auth_failure") == rt.WARNING
+        assert rt.action_severity("session_start") == rt.INFO
+        assert rt.action_severity("nat_mapping") == rt.DEBUG
+        assert rt.action_category("mac_spoof") == "security"
+        assert rt.action_category("certificate_expired") == "tls"
+        assert rt.action_category("dhcp_ack") == "dhcp"
+        assert rt.action_category("whatever") == "other"
+        assert rt.severity_name(rt.EMERGENCY) == "EMERGENCY"
+
+    def test_logger_min_severity_and_disabled_category(self):
+        from bng_amd.audit.logger import Logger, MemoryStorage
+        from bng_amd.audit import retention as rt
+        log = Logger(MemoryStorage(), min_severity=rt.WARNING,
+                     disabled_categories={"nat"})
+        log.log("session_start", subscriber="s1")       # INFO -> filtered
+        log.log("auth_failure", category="auth")        # WARNING -> kept
+        log.log("mac_spoof", category="nat")            # category disabled
+        log.flush()
+        assert log.filtered == 2 and log.logged == 1
+        evs = log.storage.all()
+        assert len(evs) == 1 and evs[0].action == "auth_failure"
+        assert evs[0].severity == rt.WARNING
+        st = log.stats()
+        assert st["stored"] == 1 and st["filtered"] == 2
+
+    def test_retention_per_category_with_action_override(self):
+        from bng_amd.audit.retention import RetentionManager
+        rm = RetentionManager()
+        assert rm.get_retention("admin") == 730
+        assert rm.get_retention("system") == 30
+        assert rm.get_retention("unknown") == 365       # default
+        assert rm.get_retention_for_action("dhcp_ack") == 90
+        rm.set_action_retention("dhcp_ack", 7)
+        assert rm.get_retention_for_action("dhcp_ack") == 7
+        rm.set_category_retention("system", 60)
+        assert rm.policy_summary()["system"] == 60
+
+    def test_legal_hold_criteria_all_must_match(self):
+        import time as _t
+        from bng_amd.audit.logger import Event
+        from bng_amd.audit.retention import LegalHold
+        now = _t.time()
+        ev = Event(id="e1", category="session", action="session_stop",
+                   timestamp=now, subscriber="alice", ip="10.0.1.5",
+                   details={"mac": "aa:bb:cc:00:00:01",
+                            "session_id": "sess-9"})
+        assert LegalHold(subscribers=["alice"]).matches(ev)
+        assert not LegalHold(subscribers=["bob"]).matches(ev)
+        assert LegalHold(subscribers=["alice"],
+                         ips=["10.0.1.5"]).matches(ev)
+        assert not LegalHold(subscribers=["alice"],
+                             ips=["10.9.9.9"]).matches(ev)
+        assert LegalHold(macs=["aa:bb:cc:00:00:01"]).matches(ev)
+        assert LegalHold(sessions=["sess-9"]).matches(ev)
+        assert not LegalHold(actions=["auth_failure"]).matches(ev)
+        # time-window criterion
+        assert not LegalHold(start_time=now + 10).matches(ev)
+        assert LegalHold(start_time=now - 10,
+                         end_time=now + 10).matches(ev)
+
+    def test_expiry_honors_holds_and_hold_expiry(self):
+        import time as _t
+        from bng_amd.audit.logger import Event, Logger, MemoryStorage
+        from bng_amd.audit.retention import LegalHold, RetentionManager
+        rm = RetentionManager()
+        now = _t.time()
+        old = now - 200 * 86400                          # 200 days old
+        held = Event(id="h", category="dhcp", action="dhcp_ack",
+                     timestamp=old, subscriber="alice")
+        loose = Event(id="l", category="dhcp", action="dhcp_ack",
+                      timestamp=old, subscriber="bob")
+        fresh = Event(id="f", category="dhcp", action="dhcp_ack",
+                      timestamp=now, subscriber="bob")
+        rm.add_hold(LegalHold(subscribers=["alice"]))
+        # dhcp retention is 90 days: old events expired unless held
+        assert not rm.expired(held, now)
+        assert rm.expired(loose, now)
+        assert not rm.expired(fresh, now)
+        # wired through the logger + storage sweep
+        st = MemoryStorage()
+        for e in (held, loose, fresh):
+            st.store(e)
+        log = Logger(st, retention=rm)
+        assert log.cleanup_expired(now) == 1
+        assert {e.id for e in st.all()} == {"h", "f"}
+        # an expired hold stops protecting
+        hid = rm.get_holds()[0].id
+        rm.holds[hid].expires_at = now - 1
+        assert rm.cleanup_expired_holds() == 1
+        assert rm.expired(held, now)
+
+    def test_storage_query_limit_severity_delete(self):
+        from bng_amd.audit.logger import Event, MemoryStorage
+        from bng_amd.audit import retention as rt
+        st = MemoryStorage()
+        for i in range(10):
+            st.store(Event(id=f"e{i}", category="auth",
+                           action="auth_failure", timestamp=float(i),
+                           severity=rt.WARNING if i % 2 else rt.INFO))
+        assert len(st.query(min_severity=rt.WARNING)) == 5
+        assert len(st.query(limit=3)) == 3
+        assert st.count() == 10
+        assert st.delete(["e0", "e1", "nope"]) == 2
+        assert st.count() == 8
+
+    def test_syslog_priority_encodes_severity(self):
+        from bng_amd.audit.logger import Event, SyslogExporter
+        from bng_amd.audit import retention as rt
+        lines = []
+        ex = SyslogExporter(lines.append)
+        ex.export(Event(id="a", category="security", action="mac_spoof",
+                        timestamp=0.0, severity=rt.CRITICAL))
+        ex.export(Event(id="b", category="session",
+                        action="session_start", timestamp=0.0,
+                        severity=rt.INFO))
+        # facility 13: CRITICAL -> level 2 -> PRI 106; INFO -> 6 -> 110
+        assert lines[0].startswith("<106>")
+        assert lines[1].startswith("<110>")
