@@ -117,6 +117,28 @@ def extract_cert_info(cert_path: str) -> dict:
             "not_after": d.get("notAfter", "")}
 
 
+def is_certificate_expiring_soon(cert_path: str,
+                                 within_days: float = 30.0):
+    """(expiring, days_left) for renewal warnings (ref ztp/tls.go
+    IsCertificateExpiringSoon :508-522)."""
+    import time as _t
+    info = extract_cert_info(cert_path)
+    na = info.get("not_after", "")
+    if not na:
+        raise TLSError(f"no notAfter in {cert_path}")
+    expires = ssl.cert_time_to_seconds(na)
+    days_left = (expires - _t.time()) / 86400.0
+    return days_left <= within_days, days_left
+
+
+def extract_server_name_from_url(url: str) -> str:
+    """Hostname for SNI/verification (ref ztp/tls.go
+    ExtractServerNameFromURL :524-537)."""
+    from urllib.parse import urlparse
+    host = urlparse(url).hostname
+    return host or ""
+
+
 def requests_kwargs(c: TLSConfig) -> dict:
     """Map the config onto requests' verify=/cert= kwargs (the
     AuthenticatedTransport analog for our HTTP clients)."""

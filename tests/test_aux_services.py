@@ -567,3 +567,104 @@ def test_wifi_traffic_stats():
     st = g.get_stats()
     assert st["sessions"] == 2 and st["accepted"] == 1
     assert st["bytes_in"] == 1500 and st["bytes_out"] == 300
+
+
+class TestZTPFullFlow:
+    """register_and_wait pending/configured contract (ref
+    bootstrap.go:219-300) + vendor TLVs + system-info detection."""
+
+    def test_pending_then_configured(self):
+        from bng_amd.ztp.bootstrap import SystemInfo
+        srv = ZTPServer()
+        try:
+            c = BootstrapClient(srv.url, serial="SN-P1",
+                                poll_interval=0.05)
+            import threading
+
+            def approve_later():
+                import time as _t
+                while "SN-P1" not in srv.devices:
+                    _t.sleep(0.01)
+                srv.approve("SN-P1", {
+                    "node_id": "bng-42", "site_id": "pop-7",
+                    "role": "active",
+                    "pools": [{"pool_id": "p1",
+                               "cidr": "10.1.0.0/16"}],
+                    "cluster": {"peers": ["http://n1", "http://n2"]}})
+            threading.Thread(target=approve_later, daemon=True).start()
+            cfg = c.register_and_wait(
+                SystemInfo(serial="SN-P1", mac="02:00:00:00:00:aa",
+                           model="lab", firmware="1.0"),
+                initial_backoff=0.05, max_backoff=0.2, deadline=10)
+            assert c.state == "approved"
+            assert cfg.node_id == "bng-42" and cfg.site_id == "pop-7"
+            assert cfg.pools[0]["cidr"] == "10.1.0.0/16"
+            assert cfg.cluster["peers"] == ["http://n1", "http://n2"]
+            assert cfg.device_id == "bng-42"       # node_id fallback
+            # register recorded the hardware identity
+            assert srv.devices["SN-P1"]["mac"] == "02:00:00:00:00:aa"
+        finally:
+            srv.stop()
+
+    def test_pending_max_retries(self):
+        srv = ZTPServer()
+        try:
+            c = BootstrapClient(srv.url, serial="SN-P2")
+            with pytest.raises(TimeoutError):
+                c.register_and_wait(max_retries=3,
+                                    initial_backoff=0.01,
+                                    max_backoff=0.02, deadline=10)
+        finally:
+            srv.stop()
+
+    def test_vendor_tlv_discovery(self):
+        from bng_amd.ztp.bootstrap import parse_vendor_options
+        tlv = bytes([7, 2, 0, 0, 1, 8]) + b"http://n"
+        assert parse_vendor_options(tlv) == "http://n"
+        assert parse_vendor_options(b"\x01\xff") == ""   # truncated
+        assert parse_vendor_options(b"") == ""
+        # option 43 TLV beats plain decode; plain stays supported
+        assert discover_nexus_from_dhcp_options(
+            {43: tlv}) == "http://n"
+        assert discover_nexus_from_dhcp_options(
+            {43: b"http://plain"}) == "http://plain"
+        # option 224 always wins
+        assert discover_nexus_from_dhcp_options(
+            {224: b"http://x", 43: tlv}) == "http://x"
+
+    def test_detect_system_info(self):
+        from bng_amd.ztp.bootstrap import detect_system_info
+        info = detect_system_info()
+        assert info.serial            # DMI or hostname fallback
+        assert info.model
+
+    def test_plan_interface_config(self):
+        from bng_amd.ztp.bootstrap import plan_interface_config
+        cmds = plan_interface_config("eth1", "10.0.0.5", 24,
+                                     gateway="10.0.0.1",
+                                     dns=["1.1.1.1"])
+        assert cmds[0] == "ip addr add 10.0.0.5/24 dev eth1"
+        assert any("default via 10.0.0.1" in c for c in cmds)
+        assert any("1.1.1.1" in c for c in cmds)
+
+
+class TestTLSCertHelpers:
+    def test_expiring_soon_and_server_name(self, tmp_path):
+        import subprocess
+        from bng_amd.agent.tls import (extract_server_name_from_url,
+                                       is_certificate_expiring_soon)
+        cert = tmp_path / "c.pem"
+        key = tmp_path / "k.pem"
+        subprocess.run(
+            ["openssl", "req", "-x509", "-newkey", "rsa:2048",
+             "-keyout", str(key), "-out", str(cert), "-days", "10",
+             "-nodes", "-subj", "/CN=ztp-test"],
+            check=True, capture_output=True)
+        expiring, days = is_certificate_expiring_soon(str(cert), 30)
+        assert expiring and 9 < days <= 10
+        expiring2, _ = is_certificate_expiring_soon(str(cert), 5)
+        assert not expiring2
+        assert extract_server_name_from_url(
+            "https://nexus.isp:8443/api") == "nexus.isp"
+        assert extract_server_name_from_url("http://10.0.0.1") == \
+            "10.0.0.1"
