@@ -668,3 +668,87 @@ class TestTLSCertHelpers:
             "https://nexus.isp:8443/api") == "nexus.isp"
         assert extract_server_name_from_url("http://10.0.0.1") == \
             "10.0.0.1"
+
+
+class TestDeviceAuthDepth:
+    """Identity derivation, config validation, PSK lifecycle, CSR
+    renewal (ref pkg/deviceauth authenticator.go:233-308, psk.go,
+    mtls.go:362-418)."""
+
+    def test_device_id_derivation(self):
+        from bng_amd.deviceauth.authenticator import (generate_device_id,
+                                                      sanitize_id)
+        assert sanitize_id("AB-12_c!@#$") == "AB-12_c"
+        assert generate_device_id("SN 99/3") == "bng-SN993"
+        assert generate_device_id("", "02:aa:bb:cc:dd:ee") == \
+            "bng-02aabbccddee"
+        a, b = generate_device_id(), generate_device_id()
+        assert a.startswith("bng-") and a != b     # random fallback
+
+    def test_read_device_identity(self):
+        from bng_amd.deviceauth.authenticator import read_device_identity
+        ident = read_device_identity()
+        assert ident["device_id"].startswith("bng-")
+        assert ident["serial"]
+
+    def test_validate_config(self):
+        from bng_amd.deviceauth.authenticator import (AuthError,
+                                                      validate_config)
+        validate_config("none")
+        validate_config("psk", psk_key="k")
+        with pytest.raises(AuthError):
+            validate_config("psk")
+        validate_config("mtls", cert_file="c", key_file="k",
+                        ca_file="ca")
+        validate_config("mtls", cert_file="c", key_file="k",
+                        insecure_skip_verify=True)
+        with pytest.raises(AuthError):
+            validate_config("mtls", cert_file="c", key_file="k")
+        with pytest.raises(AuthError):
+            validate_config("mtls", cert_file="c")
+        with pytest.raises(NotImplementedError):
+            validate_config("tpm")
+        with pytest.raises(AuthError):
+            validate_config("wat")
+
+    def test_psk_load_and_rotate(self, tmp_path):
+        from bng_amd.deviceauth.authenticator import (AuthError,
+                                                      PSKAuthenticator,
+                                                      load_psk)
+        kf = tmp_path / "psk.key"
+        kf.write_text("  file-secret-0123456789  \n")
+        assert load_psk(key_file=str(kf)) == b"file-secret-0123456789"
+        assert load_psk(key="inline") == b"inline"
+        with pytest.raises(AuthError):
+            load_psk()
+        a = PSKAuthenticator(load_psk(key_file=str(kf)))
+        h = a.headers("dev-1")
+        assert a.verify(h) == "dev-1"
+        # rotation enforces the 16-char production floor
+        with pytest.raises(AuthError):
+            a.rotate("short")
+        a.rotate("long-enough-key-123")
+        with pytest.raises(AuthError):
+            a.verify(h)                    # old-key signature now bad
+        assert a.verify(a.headers("dev-1")) == "dev-1"
+
+    def test_csr_renewal_request(self):
+        from bng_amd.deviceauth.authenticator import MTLSAuthenticator
+        m = MTLSAuthenticator()
+        req = m.renewal_request("bng-SN7", reason="expiring")
+        assert req["csr"].startswith("-----BEGIN CERTIFICATE REQUEST")
+        assert "PRIVATE KEY" in req["_key_pem"]
+        assert req["device_id"] == "bng-SN7"
+
+    def test_cert_expires_within(self, tmp_path):
+        import subprocess
+        from bng_amd.deviceauth.authenticator import MTLSAuthenticator
+        cert = tmp_path / "c.pem"
+        subprocess.run(
+            ["openssl", "req", "-x509", "-newkey", "rsa:2048",
+             "-keyout", str(tmp_path / "k.pem"), "-out", str(cert),
+             "-days", "10", "-nodes", "-subj", "/CN=d"],
+            check=True, capture_output=True)
+        m = MTLSAuthenticator()
+        assert m.certificate_expires_within(str(cert), 30)
+        assert not m.certificate_expires_within(str(cert), 5)
