@@ -582,3 +582,102 @@ class TestAuditSeverityRetention:
         # facility 13: CRITICAL -> level 2 -> PRI 106; INFO -> 6 -> 110
         assert lines[0].startswith("<106>")
         assert lines[1].startswith("<110>")
+
+
+class TestInterceptDepth:
+    """Multi-criteria warrants, intercept sessions, filtered CC capture,
+    ETSI HI2/HI3 PDUs (ref pkg/intercept manager.go:260-460,
+    exporter.go:191-318)."""
+
+    def test_match_session_any_criterion_dedup(self):
+        m = Intercept()
+        w1 = m.add_warrant("alice", target_mac="AA:BB:CC:00:00:01",
+                           target_ipv4="10.0.1.5")
+        w2 = m.add_warrant("", target_ipv4="10.0.1.5")
+        # same warrant reachable via two criteria -> one result
+        got = m.match_session(subscriber="alice",
+                              mac="aa:bb:cc:00:00:01")
+        assert [w.id for w in got] == [w1.id]
+        got = m.match_session(ipv4="10.0.1.5")
+        assert {w.id for w in got} == {w1.id, w2.id}
+        assert m.match_session(subscriber="bob") == []
+        with pytest.raises(ValueError):
+            m.add_warrant()                     # no criterion at all
+
+    def test_intercept_session_lifecycle_and_stats(self):
+        ex = JSONExporter()
+        m = Intercept(exporters=[ex])
+        w = m.add_warrant("alice", intercept_type="iri+cc",
+                          liid="LIID-CASE-9")
+        s = m.start_intercept(w, "sess-1", "alice",
+                              ipv4="10.0.1.5")
+        assert s.liid == "LIID-CASE-9"
+        assert w.sessions_matched == 1
+        assert m.get_intercept("sess-1") is s
+        ok = m.record_cc("sess-1", "up", "10.0.1.5", "9.9.9.9",
+                         5555, 443, 6, b"x" * 100)
+        assert ok and s.cc_records == 1 and s.bytes_captured == 100
+        assert w.bytes_intercepted == 100
+        st = m.stats()
+        assert st["active_interceptions"] == 1
+        assert st["total_cc_records"] == 1
+        assert st["total_bytes_delivered"] == 100
+        m.stop_intercept("sess-1")
+        assert m.get_intercept("sess-1") is None
+        assert m.stats()["active_interceptions"] == 0
+        # session_start + session_stop IRI both exported
+        import json as _json
+        kinds = [_json.loads(l)["record_type"] for l in ex.lines]
+        assert kinds.count("session_start") == 1
+        assert kinds.count("session_stop") == 1
+
+    def test_cc_filters_every_axis(self):
+        m = Intercept()
+        w = m.add_warrant("alice", intercept_type="cc",
+                          filter_dest_ports=[443],
+                          filter_protocols=[6],
+                          filter_dest_ips=["9.9.9.9"])
+        m.start_intercept(w, "s", "alice")
+        ok = m.record_cc("s", "up", "10.0.1.5", "9.9.9.9",
+                         1, 443, 6, b"y")
+        assert ok
+        assert not m.record_cc("s", "up", "10.0.1.5", "9.9.9.9",
+                               1, 80, 6, b"y")       # port filtered
+        assert not m.record_cc("s", "up", "10.0.1.5", "9.9.9.9",
+                               1, 443, 17, b"y")     # proto filtered
+        assert not m.record_cc("s", "up", "10.0.1.5", "8.8.8.8",
+                               1, 443, 6, b"y")      # dest filtered
+        # an IRI-only warrant never captures content
+        w2 = m.add_warrant("bob", intercept_type="iri")
+        m.start_intercept(w2, "s2", "bob")
+        assert not m.record_cc("s2", "up", "1.1.1.1", "2.2.2.2",
+                               1, 2, 6, b"z")
+
+    def test_etsi_pdu_roundtrip_and_sequences(self):
+        from bng_amd.intercept.etsi import (ETSIExporter, decode_pdu,
+                                            split_stream, HI2, HI3)
+        frames = []
+        ex = ETSIExporter(send=frames.append, country_code="DE")
+        m = Intercept(exporters=[ex])
+        w = m.add_warrant("alice", intercept_type="iri+cc",
+                          liid="LIID-7")
+        m.start_intercept(w, "sess-9", "alice", ipv4="10.0.1.5")
+        m.record_cc("sess-9", "down", "9.9.9.9", "10.0.1.5",
+                    443, 5555, 6, b"PAYLOAD")
+        assert ex.sent_iri == 1 and ex.sent_cc == 1
+        iri = decode_pdu(frames[0])
+        assert iri["handover"] == HI2 and iri["liid"] == "LIID-7"
+        assert iri["seq"] == 0
+        assert iri["iri"]["event_type"] == "session_start"
+        assert iri["iri"]["session_id"] == "sess-9"
+        assert iri["iri"]["country_code"] == "DE"
+        cc = decode_pdu(frames[1])
+        assert cc["handover"] == HI3 and cc["seq"] == 1   # per-LIID seq
+        assert cc["direction"] == "down"
+        assert cc["src_ip"] == "9.9.9.9" and cc["dst_port"] == 5555
+        assert cc["protocol"] == 6 and cc["payload"] == b"PAYLOAD"
+        # stream framing: concatenated PDUs split cleanly
+        assert split_stream(frames[0] + frames[1]) == frames
+        # a second LIID starts its own sequence space
+        ex.send(ex.build_hi2("LIID-8", "session_start", "x"))
+        assert decode_pdu(frames[2])["seq"] == 0
