@@ -17,6 +17,41 @@ S_PARTITIONED = "partitioned"
 S_RECOVERING = "recovering"
 
 
+@dataclass
+class Subscriber:
+    """Wholesale subscriber: stable physical layer (NetCo) + mutable
+    service layer (ISPCo) + session state (ref agent/types.go
+    Subscriber :156-180)."""
+    subscriber_id: str
+    nte_id: str = ""
+    device_id: str = ""
+    vlan: str = ""                    # "s-tag:c-tag"
+    netco_id: str = ""
+    isp_id: str = ""
+    radius_realm: str = ""
+    service_tier: str = ""
+    qos_policy: str = ""
+    mac: str = ""
+    ipv4: str = ""
+    ipv6_prefix: str = ""
+    authenticated: bool = False
+    session_start: float = 0.0
+
+
+@dataclass
+class NTE:
+    """Discovered network-terminating equipment (ONU/ONT) (ref
+    types.go NTE :182-192)."""
+    serial: str
+    device_id: str = ""
+    port: int = 0
+    status: str = "discovered"        # discovered|provisioned|active
+    vendor: str = ""
+    model: str = ""
+    firmware: str = ""
+    discovered_at: float = 0.0
+
+
 class Agent:
     def __init__(self, store: Store, node_id: str,
                  heartbeat_interval: float = 5.0,
@@ -34,6 +69,14 @@ class Agent:
         self._stop = threading.Event()
         self._threads: List[threading.Thread] = []
         self._watch_cancel = None
+        self._started_at = 0.0
+        # wholesale registries (ref agent.go:315-440)
+        self._subs: Dict[str, Subscriber] = {}
+        self._subs_by_mac: Dict[str, str] = {}
+        self._subs_by_nte: Dict[str, str] = {}
+        self._ntes: Dict[str, NTE] = {}
+        self._churn_listeners: List[Callable[[Dict], None]] = []
+        self._reg_lock = threading.RLock()
 
     def on_state_change(self, cb):
         self._listeners.append(cb)
@@ -41,7 +84,13 @@ class Agent:
     def on_config_change(self, cb):
         self._config_listeners.append(cb)
 
+    def on_isp_churn(self, cb):
+        """cb(event_dict) when a subscriber's ISP assignment changes
+        (ref OnISPChurn agent.go:186-191)."""
+        self._churn_listeners.append(cb)
+
     def start(self):
+        self._started_at = time.time()
         cfg = self.config_store.get(self.node_id)
         if cfg:
             self.config = cfg
@@ -77,11 +126,112 @@ class Agent:
                 except Exception:
                     pass
 
+    # ------------------------------------- subscriber registry (NetCo)
+    def set_subscriber(self, sub: Subscriber):
+        """Store/update; an ISP change on an existing subscriber fires
+        the churn handlers (ref SetSubscriber agent.go:348-361 +
+        handleISPChurn :389-412)."""
+        with self._reg_lock:
+            old = self._subs.get(sub.subscriber_id)
+            self._subs[sub.subscriber_id] = sub
+            if sub.mac:
+                self._subs_by_mac[sub.mac.lower()] = sub.subscriber_id
+            if sub.nte_id:
+                self._subs_by_nte[sub.nte_id] = sub.subscriber_id
+        if old is not None and old.isp_id and sub.isp_id and \
+                old.isp_id != sub.isp_id:
+            event = {"subscriber_id": sub.subscriber_id,
+                     "old_isp_id": old.isp_id,
+                     "new_isp_id": sub.isp_id,
+                     "timestamp": time.time()}
+            for cb in self._churn_listeners:
+                try:
+                    cb(event)
+                except Exception:
+                    pass
+
+    def get_subscriber(self, subscriber_id: str) -> Optional[Subscriber]:
+        with self._reg_lock:
+            return self._subs.get(subscriber_id)
+
+    def get_subscriber_by_mac(self, mac: str) -> Optional[Subscriber]:
+        with self._reg_lock:
+            sid = self._subs_by_mac.get(mac.lower())
+            return self._subs.get(sid) if sid else None
+
+    def get_subscriber_by_nte(self, nte_serial: str) -> Optional[Subscriber]:
+        with self._reg_lock:
+            sid = self._subs_by_nte.get(nte_serial)
+            return self._subs.get(sid) if sid else None
+
+    def remove_subscriber(self, subscriber_id: str):
+        with self._reg_lock:
+            sub = self._subs.pop(subscriber_id, None)
+            if sub:
+                self._subs_by_mac.pop(sub.mac.lower(), None)
+                self._subs_by_nte.pop(sub.nte_id, None)
+
+    def subscriber_count(self) -> int:
+        with self._reg_lock:
+            return len(self._subs)
+
+    def subscriber_count_by_isp(self) -> Dict[str, int]:
+        """ref GetSubscriberCountByISP agent.go:377-387."""
+        out: Dict[str, int] = {}
+        with self._reg_lock:
+            for s in self._subs.values():
+                out[s.isp_id or "unassigned"] = \
+                    out.get(s.isp_id or "unassigned", 0) + 1
+        return out
+
+    # ------------------------------------------------- NTE registry
+    def set_nte(self, nte: NTE):
+        with self._reg_lock:
+            self._ntes[nte.serial] = nte
+
+    def get_nte(self, serial: str) -> Optional[NTE]:
+        with self._reg_lock:
+            return self._ntes.get(serial)
+
+    def remove_nte(self, serial: str):
+        with self._reg_lock:
+            self._ntes.pop(serial, None)
+
+    def nte_count(self) -> int:
+        with self._reg_lock:
+            return len(self._ntes)
+
+    # ------------------------------------------------------ queries
+    def get_isp_config(self, isp_id: str) -> Optional[Dict]:
+        """ISP block from the device config (ref GetISPConfig
+        agent.go:442-455)."""
+        for isp in self.config.get("isps", []):
+            if isp.get("isp_id") == isp_id:
+                return isp
+        return None
+
+    def uptime(self) -> float:
+        return time.time() - self._started_at if self._started_at else 0.0
+
+    def is_online(self) -> bool:
+        return self.state == S_CONNECTED
+
+    def health(self) -> Dict:
+        """ref Health agent.go:457-467."""
+        return {"status": self.state, "device_id": self.node_id,
+                "uptime_seconds": int(self.uptime()),
+                "subscribers": self.subscriber_count(),
+                "ntes": self.nte_count(), "online": self.is_online()}
+
     def heartbeat_once(self) -> bool:
         try:
             self.store.put(f"nexus/heartbeats/{self.node_id}",
                            json.dumps({"ts": time.time(),
-                                       "state": self.state}).encode())
+                                       "state": self.state,
+                                       "uptime": int(self.uptime()),
+                                       "subscribers":
+                                       self.subscriber_count(),
+                                       "ntes": self.nte_count()}).encode())
             self._last_ok = time.time()
             if self.state == S_PARTITIONED:
                 self._transition(S_RECOVERING)

@@ -752,3 +752,69 @@ class TestDeviceAuthDepth:
         m = MTLSAuthenticator()
         assert m.certificate_expires_within(str(cert), 30)
         assert not m.certificate_expires_within(str(cert), 5)
+
+
+class TestAgentRegistries:
+    """Wholesale subscriber/NTE registries + ISP churn (ref
+    pkg/agent/agent.go:315-467, types.go:156-240)."""
+
+    def _agent(self):
+        store = MemoryStore()
+        return Agent(store, "node-1", partition_after=0.0)
+
+    def test_subscriber_indexes(self):
+        from bng_amd.agent.agent import Subscriber
+        a = self._agent()
+        a.set_subscriber(Subscriber("sub-1", nte_id="NTE-7",
+                                    mac="AA:BB:CC:00:00:01",
+                                    isp_id="isp-a"))
+        assert a.get_subscriber("sub-1").isp_id == "isp-a"
+        assert a.get_subscriber_by_mac("aa:bb:cc:00:00:01") \
+            .subscriber_id == "sub-1"
+        assert a.get_subscriber_by_nte("NTE-7").subscriber_id == "sub-1"
+        assert a.subscriber_count() == 1
+        a.remove_subscriber("sub-1")
+        assert a.subscriber_count() == 0
+        assert a.get_subscriber_by_mac("aa:bb:cc:00:00:01") is None
+        assert a.get_subscriber_by_nte("NTE-7") is None
+
+    def test_isp_churn_event(self):
+        from bng_amd.agent.agent import Subscriber
+        a = self._agent()
+        events = []
+        a.on_isp_churn(events.append)
+        a.set_subscriber(Subscriber("sub-1", isp_id="isp-a"))
+        assert events == []                      # first set: no churn
+        a.set_subscriber(Subscriber("sub-1", isp_id="isp-a"))
+        assert events == []                      # same ISP: no churn
+        a.set_subscriber(Subscriber("sub-1", isp_id="isp-b"))
+        assert len(events) == 1
+        assert events[0]["old_isp_id"] == "isp-a"
+        assert events[0]["new_isp_id"] == "isp-b"
+        assert a.subscriber_count_by_isp() == {"isp-b": 1}
+
+    def test_nte_registry_and_health(self):
+        from bng_amd.agent.agent import NTE, Subscriber
+        a = self._agent()
+        a.start()
+        try:
+            a.set_nte(NTE("SER-1", port=3, status="provisioned",
+                          vendor="acme"))
+            a.set_subscriber(Subscriber("sub-1"))
+            assert a.get_nte("SER-1").vendor == "acme"
+            assert a.nte_count() == 1
+            h = a.health()
+            assert h["status"] == "connected" and h["online"]
+            assert h["subscribers"] == 1 and h["ntes"] == 1
+            assert h["uptime_seconds"] >= 0
+            a.remove_nte("SER-1")
+            assert a.nte_count() == 0
+        finally:
+            a.stop()
+
+    def test_isp_config_lookup(self):
+        a = self._agent()
+        a.config = {"isps": [{"isp_id": "isp-a", "radius_realm": "a.net"},
+                             {"isp_id": "isp-b"}]}
+        assert a.get_isp_config("isp-a")["radius_realm"] == "a.net"
+        assert a.get_isp_config("nope") is None
