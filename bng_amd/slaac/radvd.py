@@ -10,7 +10,7 @@ import struct
 import threading
 import time
 from dataclasses import dataclass, field
-from typing import List, Optional, Tuple
+from typing import Dict, List, Optional, Tuple
 
 ND_ROUTER_SOLICIT = 133
 ND_ROUTER_ADVERT = 134
@@ -156,6 +156,19 @@ class Server:
         self.send_fn(build_ra(self.cfg), dst)
         self.stats["ra_sent"] += 1
 
+    def add_prefix(self, p: "PrefixConfig"):
+        """ref AddPrefix radvd.go:491-496; takes effect next RA."""
+        self.cfg.prefixes.append(p)
+
+    def remove_prefix(self, prefix: str):
+        """ref RemovePrefix radvd.go:498-509."""
+        self.cfg.prefixes = [q for q in self.cfg.prefixes
+                             if q.prefix != prefix]
+
+    def send_immediate_ra(self):
+        """ref SendImmediateRA radvd.go:519."""
+        self.advertise()
+
     def handle_rs(self, data: bytes, src: str = "") -> Optional[bytes]:
         """Router Solicitation -> immediate unicast RA."""
         if not data or data[0] != ND_ROUTER_SOLICIT:
@@ -166,3 +179,125 @@ class Server:
         payload = build_ra(self.cfg)
         self.send_fn(payload, src or "ff02::1")
         return payload
+
+
+# ----------------------------------------------------------------------
+# address generation + classification (ref slaac/types.go:124-185)
+
+def generate_slaac_address(prefix: str, mac: str) -> str:
+    """Modified EUI-64: ff:fe spliced into the MAC, universal/local bit
+    flipped, appended to the /64 prefix (ref GenerateSLAACAddress
+    types.go:124-148)."""
+    m = bytes(int(b, 16) for b in mac.split(":"))
+    if len(m) != 6:
+        raise ValueError(f"bad MAC {mac}")
+    eui64 = bytes([m[0] ^ 0x02, m[1], m[2], 0xFF, 0xFE, m[3], m[4], m[5]])
+    net = ipaddress.IPv6Network(prefix, strict=False)
+    return str(ipaddress.IPv6Address(net.network_address.packed[:8] +
+                                     eui64))
+
+
+def generate_stable_privacy_address(prefix: str, interface_id: bytes,
+                                    secret_key: bytes,
+                                    dad_counter: int = 0) -> str:
+    """RFC 7217 stable privacy address: SHA-256 over
+    (prefix | interface_id | dad_counter | secret) -> IID (the reference
+    ships a simplified XOR and notes production should hash,
+    types.go:150-168 — we do the hash)."""
+    import hashlib
+    net = ipaddress.IPv6Network(prefix, strict=False)
+    digest = hashlib.sha256(net.network_address.packed[:8] +
+                            interface_id +
+                            dad_counter.to_bytes(4, "big") +
+                            secret_key).digest()
+    iid = bytearray(digest[:8])
+    iid[0] &= ~0x02                    # clear universal/local bit
+    return str(ipaddress.IPv6Address(net.network_address.packed[:8] +
+                                     bytes(iid)))
+
+
+def is_link_local(ip: str) -> bool:
+    a = ipaddress.ip_address(ip)
+    return a.is_link_local
+
+
+def is_global_unicast(ip: str) -> bool:
+    a = ipaddress.ip_address(ip)
+    return a.is_global and not a.is_private
+
+
+# ----------------------------------------------------------------------
+# neighbor cache (ref types.go NeighborEntry :102-122; RFC 4861 §7.3
+# reachability states)
+
+N_INCOMPLETE = "INCOMPLETE"
+N_REACHABLE = "REACHABLE"
+N_STALE = "STALE"
+N_DELAY = "DELAY"
+N_PROBE = "PROBE"
+
+
+class NeighborCache:
+    """Minimal RFC 4861 neighbor state machine: confirmations make an
+    entry REACHABLE; it decays to STALE after reachable_time; sending to
+    a STALE entry moves it DELAY, and an unanswered DELAY probes."""
+
+    def __init__(self, reachable_time: float = 30.0,
+                 delay_time: float = 5.0):
+        self.reachable_time = reachable_time
+        self.delay_time = delay_time
+        self._entries: Dict[str, dict] = {}
+
+    def confirm(self, ip: str, mac: str = "",
+                is_router: Optional[bool] = None, now: float = 0.0):
+        """Reachability confirmed (NA received / upper-layer hint);
+        is_router only changes when the confirmation says so (an NA
+        without the R flag conveys nothing about router-ness here)."""
+        now = now or time.time()
+        e = self._entries.setdefault(ip, {"is_router": False, "mac": ""})
+        if mac:
+            e["mac"] = mac
+        if is_router is not None:
+            e["is_router"] = is_router
+        e.update(state=N_REACHABLE, last_seen=now)
+
+    def incomplete(self, ip: str, now: float = 0.0):
+        """NS sent, no answer yet."""
+        self._entries[ip] = {"mac": "", "state": N_INCOMPLETE,
+                             "last_seen": now or time.time(),
+                             "is_router": False}
+
+    def state(self, ip: str, now: float = 0.0) -> str:
+        e = self._entries.get(ip)
+        if e is None:
+            return ""
+        now = now or time.time()
+        if e["state"] == N_REACHABLE and \
+                now - e["last_seen"] > self.reachable_time:
+            e["state"] = N_STALE
+        elif e["state"] == N_DELAY and \
+                now - e.get("delay_at", now) > self.delay_time:
+            e["state"] = N_PROBE
+        return e["state"]
+
+    def used(self, ip: str, now: float = 0.0):
+        """A packet was sent to the neighbor: STALE -> DELAY."""
+        now = now or time.time()
+        if self.state(ip, now) == N_STALE:
+            e = self._entries[ip]
+            e["state"] = N_DELAY
+            e["delay_at"] = now
+
+    def lookup(self, ip: str) -> Optional[dict]:
+        return self._entries.get(ip)
+
+    def routers(self):
+        return [ip for ip, e in self._entries.items() if e["is_router"]]
+
+    def purge(self, max_age: float, now: float = 0.0) -> int:
+        now = now or time.time()
+        dead = [ip for ip, e in self._entries.items()
+                if now - e["last_seen"] > max_age]
+        for ip in dead:
+            del self._entries[ip]
+        return len(dead)

@@ -284,3 +284,80 @@ def test_nested_relay_chain():
     dec = DHCPv6Message.decode(final)
     assert dec.msg_type == REPLY and dec.txn_id == 0x333
     assert dec.get(OPT_IA_NA) is not None
+
+
+class TestSLAACAddressing:
+    """EUI-64 / RFC 7217 address generation, classification, neighbor
+    cache (ref pkg/slaac/types.go:102-185)."""
+
+    def test_eui64_address(self):
+        from bng_amd.slaac.radvd import generate_slaac_address
+        # known vector: 00:1b:44:11:3a:b7 -> ::21b:44ff:fe11:3ab7
+        got = generate_slaac_address("2001:db8:1::/64",
+                                     "00:1b:44:11:3a:b7")
+        assert got == "2001:db8:1:0:21b:44ff:fe11:3ab7"
+        with pytest.raises(ValueError):
+            generate_slaac_address("2001:db8::/64", "00:1b:44")
+
+    def test_stable_privacy_address(self):
+        from bng_amd.slaac.radvd import generate_stable_privacy_address
+        a1 = generate_stable_privacy_address("2001:db8:1::/64", b"eth0",
+                                             b"secret")
+        a2 = generate_stable_privacy_address("2001:db8:1::/64", b"eth0",
+                                             b"secret")
+        assert a1 == a2                              # stable
+        a3 = generate_stable_privacy_address("2001:db8:2::/64", b"eth0",
+                                             b"secret")
+        assert a1 != a3                              # per-prefix
+        a4 = generate_stable_privacy_address("2001:db8:1::/64", b"eth0",
+                                             b"secret", dad_counter=1)
+        assert a1 != a4                              # DAD retry moves
+        assert a1.startswith("2001:db8:1:")
+        # universal/local bit cleared in the IID
+        import ipaddress
+        assert not (ipaddress.IPv6Address(a1).packed[8] & 0x02)
+
+    def test_classification(self):
+        from bng_amd.slaac.radvd import is_global_unicast, is_link_local
+        assert is_link_local("fe80::1")
+        assert is_link_local("169.254.1.1")
+        assert not is_link_local("2001:db8::1")
+        assert is_global_unicast("2600::1")
+        assert not is_global_unicast("fe80::1")
+        assert not is_global_unicast("10.0.0.1")
+
+    def test_neighbor_state_machine(self):
+        from bng_amd.slaac.radvd import (N_DELAY, N_PROBE, N_REACHABLE,
+                                         N_STALE, NeighborCache)
+        nc = NeighborCache(reachable_time=30.0, delay_time=5.0)
+        nc.confirm("fe80::1", "aa:bb:cc:00:00:01", is_router=True,
+                   now=1000.0)
+        assert nc.state("fe80::1", 1010.0) == N_REACHABLE
+        assert nc.state("fe80::1", 1031.0) == N_STALE    # decayed
+        nc.used("fe80::1", 1032.0)                       # tx to STALE
+        assert nc.state("fe80::1", 1033.0) == N_DELAY
+        assert nc.state("fe80::1", 1038.0) == N_PROBE    # unanswered
+        nc.confirm("fe80::1", now=1039.0)                # NA arrives
+        assert nc.state("fe80::1", 1040.0) == N_REACHABLE
+        assert nc.routers() == ["fe80::1"]
+        assert nc.state("fe80::9") == ""
+        nc.incomplete("fe80::9", now=1000.0)
+        assert nc.state("fe80::9") == "INCOMPLETE"
+        assert nc.purge(max_age=60.0, now=1090.0) == 1   # fe80::9 gone
+        assert nc.lookup("fe80::1") is not None
+
+    def test_server_prefix_management(self):
+        from bng_amd.slaac.radvd import parse_ra
+        sent = []
+        cfg = RAConfig(prefixes=[PrefixConfig("2001:db8:1::/64")])
+        srv = RadvServer(cfg, send_fn=lambda p, d: sent.append(p))
+        srv.add_prefix(PrefixConfig("2001:db8:2::/64"))
+        srv.send_immediate_ra()
+        got = parse_ra(sent[-1])
+        assert [p["prefix"] for p in got["prefixes"]] == \
+            ["2001:db8:1::/64", "2001:db8:2::/64"]
+        srv.remove_prefix("2001:db8:1::/64")
+        srv.send_immediate_ra()
+        got = parse_ra(sent[-1])
+        assert [p["prefix"] for p in got["prefixes"]] == \
+            ["2001:db8:2::/64"]
