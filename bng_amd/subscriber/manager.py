@@ -35,6 +35,24 @@ class Session:
     input_octets: int = 0
     output_octets: int = 0
     attributes: dict = field(default_factory=dict)
+    # access context (ref types.go SessionRequest :9-39 / Session
+    # :64-159): QinQ tags, NTE/PON identity, DHCP option-82
+    s_tag: int = 0
+    c_tag: int = 0
+    nte_id: str = ""
+    circuit_id: str = ""
+    remote_id: str = ""
+    # RADIUS-applied service attributes (ref Authenticate :229-259)
+    radius_session_id: str = ""
+    session_timeout: float = 0.0     # per-session override (s)
+    idle_timeout: float = 0.0
+    download_rate_bps: int = 0
+    upload_rate_bps: int = 0
+    qos_policy_id: str = ""
+    walled_reason: str = ""
+    state_reason: str = ""
+    packets_in: int = 0
+    packets_out: int = 0
 
 
 class Authenticator(Protocol):
@@ -147,6 +165,66 @@ class Manager:
         s.state = S_AUTHENTICATED
         return True
 
+    def authenticate_full(self, session_id: str,
+                          credentials: Optional[dict] = None) -> dict:
+        """Rich authentication applying RADIUS-style service attributes
+        to the session (ref Authenticate manager.go:179-296).  Uses the
+        authenticator's `authenticate_session(session, credentials) ->
+        dict` when available (keys: success, subscriber_id, isp_id,
+        radius_session_id, session_timeout, idle_timeout,
+        download_rate_bps, upload_rate_bps, qos_policy_id,
+        walled_garden, walled_reason, error), else falls back to the
+        boolean protocol.  A walled-garden grant lands the session in
+        StateWalledGarden instead of failing it."""
+        s = self.get(session_id)
+        if s is None:
+            return {"success": False, "error": "session not found"}
+        old_state = s.state
+        s.state = S_AUTHENTICATING
+        try:
+            if hasattr(self.auth, "authenticate_session"):
+                result = self.auth.authenticate_session(
+                    s, credentials or {})
+            else:
+                ok = self.auth.authenticate(s.subscriber_id,
+                                            credentials or {})
+                result = {"success": bool(ok)}
+        except Exception as e:
+            s.state = old_state
+            s.state_reason = f"auth error: {e}"
+            self.stats["auth_failed"] += 1
+            self._emit("session_auth_fail", s)
+            return {"success": False, "error": str(e)}
+        if not result.get("success"):
+            s.state = old_state
+            s.state_reason = result.get("error", "rejected")
+            self.stats["auth_failed"] += 1
+            self._emit("session_auth_fail", s)
+            return result
+        if result.get("subscriber_id"):
+            with self._lock:
+                self.by_subscriber.pop(s.subscriber_id, None)
+                s.subscriber_id = result["subscriber_id"]
+                self.by_subscriber[s.subscriber_id] = s.id
+        s.isp_id = result.get("isp_id", s.isp_id)
+        s.radius_session_id = result.get("radius_session_id", "")
+        # only positive attributes override (ref :234-249)
+        for key in ("session_timeout", "idle_timeout",
+                    "download_rate_bps", "upload_rate_bps"):
+            if result.get(key, 0) > 0:
+                setattr(s, key, result[key])
+        if result.get("qos_policy_id"):
+            s.qos_policy_id = result["qos_policy_id"]
+        if result.get("walled_garden"):
+            s.state = S_WALLED
+            s.walled_reason = result.get("walled_reason", "")
+            self.stats["walled"] += 1
+        else:
+            s.state = S_AUTHENTICATED
+        self.stats["auth_ok"] = self.stats.get("auth_ok", 0) + 1
+        self._emit("session_auth", s)
+        return result
+
     def assign_address(self, session_id: str) -> Optional[str]:
         s = self.get(session_id)
         if s is None or self.allocator is None:
@@ -238,21 +316,56 @@ class Manager:
                     if s.isp_id == isp_id]
 
     def cleanup(self, now: Optional[float] = None) -> int:
-        """Idle/session-timeout reaping (ref manager.go cleanup loop)."""
+        """Idle/session-timeout reaping; per-session RADIUS-applied
+        timeouts override the manager defaults (ref manager.go
+        cleanupExpiredSessions :648-690)."""
         now = now if now is not None else time.time()
         with self._lock:
             dead = []
             for s in self.sessions.values():
-                if self.idle_timeout and \
-                        now - s.last_activity > self.idle_timeout:
-                    dead.append((s.id, "idle-timeout"))
-                elif self.session_timeout and \
-                        now - s.started_at > self.session_timeout:
+                sess_to = s.session_timeout or self.session_timeout
+                idle_to = s.idle_timeout or self.idle_timeout
+                if sess_to and now - s.started_at > sess_to:
                     dead.append((s.id, "session-timeout"))
+                elif idle_to and now - s.last_activity > idle_to:
+                    dead.append((s.id, "idle-timeout"))
         for sid, why in dead:
             self.terminate_session(sid, reason=why)
         self.stats["cleaned_up"] += len(dead)
         return len(dead)
+
+    def update_activity(self, session_id: str, bytes_in: int = 0,
+                        bytes_out: int = 0, packets_in: int = 0,
+                        packets_out: int = 0) -> bool:
+        """Accumulate traffic counters + touch last_activity (ref
+        UpdateActivity manager.go:535-552)."""
+        with self._lock:
+            s = self.sessions.get(session_id)
+            if s is None:
+                return False
+            s.input_octets += bytes_in
+            s.output_octets += bytes_out
+            s.packets_in += packets_in
+            s.packets_out += packets_out
+            s.last_activity = time.time()
+        return True
+
+    def manager_stats(self) -> dict:
+        """ref ManagerStats types.go:278-288."""
+        with self._lock:
+            walled = sum(1 for s in self.sessions.values()
+                         if s.state == S_WALLED)
+            bytes_in = sum(s.input_octets for s in self.sessions.values())
+            bytes_out = sum(s.output_octets
+                            for s in self.sessions.values())
+            return {"active_sessions": len(self.sessions),
+                    "walled_garden_sessions": walled,
+                    "total_sessions_created": self.stats["created"],
+                    "total_sessions_ended": self.stats["terminated"],
+                    "auth_successes": self.stats.get("auth_ok", 0),
+                    "auth_failures": self.stats["auth_failed"],
+                    "total_bytes_in": bytes_in,
+                    "total_bytes_out": bytes_out}
 
     def count(self) -> int:
         with self._lock:

@@ -681,3 +681,95 @@ class TestInterceptDepth:
         # a second LIID starts its own sequence space
         ex.send(ex.build_hi2("LIID-8", "session_start", "x"))
         assert decode_pdu(frames[2])["seq"] == 0
+
+
+class TestSubscriberManagerDepth:
+    """Rich authentication applying RADIUS attributes, per-session
+    timeouts, traffic accounting (ref pkg/subscriber/manager.go
+    179-296, 535-552, 648-690)."""
+
+    def _mgr(self, auth=None, **kw):
+        from bng_amd.subscriber.manager import Manager
+        return Manager(authenticator=auth, **kw)
+
+    def test_authenticate_full_applies_attributes(self):
+        from bng_amd.subscriber.manager import S_AUTHENTICATED
+
+        class RadiusLike:
+            def authenticate_session(self, session, credentials):
+                assert credentials["password"] == "pw"
+                return {"success": True, "subscriber_id": "sub-real",
+                        "isp_id": "isp-a", "radius_session_id": "R1",
+                        "session_timeout": 3600,
+                        "idle_timeout": 300,
+                        "download_rate_bps": 100_000_000,
+                        "upload_rate_bps": 20_000_000,
+                        "qos_policy_id": "gold"}
+
+        m = self._mgr(RadiusLike())
+        s = m.open_session("mac-tmp", mac="aa:bb:cc:00:00:01")
+        r = m.authenticate_full(s.id, {"password": "pw"})
+        assert r["success"]
+        assert s.state == S_AUTHENTICATED
+        assert s.subscriber_id == "sub-real"       # identity rebound
+        assert m.get_by_subscriber("sub-real") is s
+        assert s.session_timeout == 3600 and s.idle_timeout == 300
+        assert s.download_rate_bps == 100_000_000
+        assert s.qos_policy_id == "gold"
+        assert m.manager_stats()["auth_successes"] == 1
+
+    def test_authenticate_full_walled_garden(self):
+        from bng_amd.subscriber.manager import S_WALLED
+
+        class Waller:
+            def authenticate_session(self, session, credentials):
+                return {"success": True, "walled_garden": True,
+                        "walled_reason": "payment_overdue"}
+
+        m = self._mgr(Waller())
+        s = m.open_session("sub-1")
+        assert m.authenticate_full(s.id)["success"]
+        assert s.state == S_WALLED
+        assert s.walled_reason == "payment_overdue"
+        assert m.manager_stats()["walled_garden_sessions"] == 1
+
+    def test_authenticate_full_failure_restores_state(self):
+        class Rejector:
+            def authenticate_session(self, session, credentials):
+                return {"success": False, "error": "bad password"}
+
+        m = self._mgr(Rejector())
+        s = m.open_session("sub-1")
+        old = s.state
+        r = m.authenticate_full(s.id, {})
+        assert not r["success"]
+        assert s.state == old and s.state_reason == "bad password"
+        assert m.manager_stats()["auth_failures"] == 1
+        # boolean-protocol fallback still works
+        m2 = self._mgr()
+        s2 = m2.open_session("sub-2")
+        assert m2.authenticate_full(s2.id)["success"]
+
+    def test_per_session_timeout_overrides_global(self):
+        import time as _t
+        m = self._mgr(idle_timeout=10_000.0)
+        a = m.open_session("sub-a")
+        b = m.open_session("sub-b")
+        a.idle_timeout = 1.0                    # RADIUS-applied
+        a.last_activity = _t.time() - 5
+        b.last_activity = _t.time() - 5
+        assert m.cleanup() == 1                 # only a reaped
+        assert m.get_by_subscriber("sub-a") is None
+        assert m.get_by_subscriber("sub-b") is not None
+
+    def test_update_activity_accumulates(self):
+        m = self._mgr()
+        s = m.open_session("sub-1")
+        assert m.update_activity(s.id, bytes_in=100, bytes_out=50,
+                                 packets_in=2, packets_out=1)
+        assert m.update_activity(s.id, bytes_in=10)
+        assert s.input_octets == 110 and s.output_octets == 50
+        assert s.packets_in == 2 and s.packets_out == 1
+        st = m.manager_stats()
+        assert st["total_bytes_in"] == 110
+        assert not m.update_activity("nope")
