@@ -818,3 +818,72 @@ class TestAgentRegistries:
                              {"isp_id": "isp-b"}]}
         assert a.get_isp_config("isp-a")["radius_realm"] == "a.net"
         assert a.get_isp_config("nope") is None
+
+
+class TestWifiManager:
+    """Captive-portal session manager with grace periods + dual
+    operating modes (ref pkg/wifi/gateway.go:27-553)."""
+
+    def test_wifi_mode_grace_period_flow(self):
+        from bng_amd.wifi.gateway import (Manager as WifiMgr, S_ACTIVE,
+                                          S_AUTHENTICATED, S_GRACE,
+                                          WifiConfig)
+        events = {"created": [], "authed": [], "expired": []}
+        m = WifiMgr(WifiConfig(lease_duration=100.0, grace_period=10.0))
+        m.on_create = lambda s: events["created"].append(s.mac)
+        m.on_auth = lambda s: events["authed"].append(s.mac)
+        m.on_expire = lambda s: events["expired"].append(s.mac)
+        s = m.create_session("AA:BB:CC:00:00:01", hostname="phone",
+                             ip="192.168.100.5")
+        assert s.state == S_GRACE
+        assert m.is_in_grace_period("aa:bb:cc:00:00:01")
+        assert m.needs_authentication("aa:bb:cc:00:00:01")
+        assert m.needs_authentication("ff:ff:ff:00:00:00")  # unknown
+        assert m.get_session_by_ip("192.168.100.5") is s
+        assert m.authenticate_session("aa:bb:cc:00:00:01",
+                                      user_identity="user@isp")
+        assert s.state == S_AUTHENTICATED and s.authenticated
+        assert not m.needs_authentication("aa:bb:cc:00:00:01")
+        m.update_traffic("aa:bb:cc:00:00:01", bytes_in=100,
+                         packets_in=1)
+        assert s.state == S_ACTIVE and s.bytes_in == 100
+        assert events == {"created": ["aa:bb:cc:00:00:01"],
+                          "authed": ["aa:bb:cc:00:00:01"],
+                          "expired": []}
+
+    def test_grace_lapse_kills_unauthenticated_only(self):
+        import time as _t
+        from bng_amd.wifi.gateway import Manager as WifiMgr, WifiConfig
+        m = WifiMgr(WifiConfig(lease_duration=1000.0, grace_period=10.0))
+        m.create_session("aa:00:00:00:00:01")
+        authed = m.create_session("aa:00:00:00:00:02")
+        m.authenticate_session("aa:00:00:00:00:02")
+        now = _t.time()
+        assert m.cleanup_expired(now + 20) == 1       # grace lapsed
+        assert m.get_session("aa:00:00:00:00:01") is None
+        assert m.get_session("aa:00:00:00:00:02") is authed
+        # lease expiry takes the authenticated one too
+        assert m.cleanup_expired(now + 2000) == 1
+        assert m.manager_stats()["active_sessions"] == 0
+
+    def test_renewal_extends_lease(self):
+        import time as _t
+        from bng_amd.wifi.gateway import Manager as WifiMgr, WifiConfig
+        m = WifiMgr(WifiConfig(lease_duration=100.0,
+                               captive_portal_enabled=False))
+        s = m.create_session("aa:00:00:00:00:03")
+        assert s.authenticated                       # portal off
+        assert not m.needs_authentication("aa:00:00:00:00:03")
+        first_expiry = s.lease_expiry
+        _t.sleep(0.01)
+        assert m.renew_session("aa:00:00:00:00:03")
+        assert s.lease_expiry > first_expiry
+        assert not m.renew_session("none")
+
+    def test_olt_bng_preset(self):
+        from bng_amd.wifi.gateway import MODE_OLT_BNG, WifiConfig
+        c = WifiConfig.olt_bng()
+        assert c.mode == MODE_OLT_BNG
+        assert c.allocation_trigger == "radius_auth"
+        assert c.deallocation_trigger == "session_termination"
+        assert not c.captive_portal_enabled
