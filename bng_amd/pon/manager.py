@@ -92,3 +92,102 @@ class Manager:
 
     def list_ntes(self) -> List[NTE]:
         return [NTE.from_dict(d) for d in self.ntes.list().values()]
+
+
+# NTE reachability states (ref manager.go:16-39)
+ST_UNKNOWN = "UNKNOWN"
+ST_CONNECTED = "CONNECTED"
+ST_DISCONNECTED = "DISCONNECTED"
+ST_UNCONFIGURED = "UNCONFIGURED"
+
+
+class AutoProvisioner:
+    """Discovery -> provision pipeline with retries (ref
+    handleDiscoveryEvent manager.go:216-277): a discovered NTE goes
+    UNCONFIGURED + pending, provisioning is retried `retries` times
+    with `retry_delay` between attempts, success promotes it to
+    CONNECTED and fires the provisioned callback with the assigned
+    VLANs; an NTE that was already provisioned just reconnects with its
+    existing tags (provisionNTE :279-330)."""
+
+    def __init__(self, manager: Manager, retries: int = 3,
+                 retry_delay: float = 5.0, default_profile: str = ""):
+        self.manager = manager
+        self.retries = retries
+        self.retry_delay = retry_delay
+        self.default_profile = default_profile
+        self.states: Dict[str, str] = {}
+        self.pending: Dict[str, dict] = {}
+        self.on_discovered = None
+        self.on_provisioned = None
+        self.on_disconnected = None
+        self.stats = {"discovered": 0, "provisioned": 0, "failed": 0,
+                      "reconnected": 0, "disconnected": 0}
+
+    def handle_discovery(self, serial: str, pon_port: str,
+                         device_id: str = "") -> dict:
+        """Returns the ProvisioningResult-shaped dict."""
+        t0 = time.time()
+        self.stats["discovered"] += 1
+        if self.on_discovered:
+            self.on_discovered({"serial": serial, "pon_port": pon_port})
+        nte = self.manager.ont_discovered(serial, pon_port, device_id)
+        # reconnection of an already-provisioned NTE keeps its tags
+        if nte.provisioned:
+            self.states[serial] = ST_CONNECTED
+            self.pending.pop(serial, None)
+            self.stats["reconnected"] += 1
+            result = {"nte_id": nte.id, "success": True,
+                      "s_tag": nte.s_tag, "c_tag": nte.c_tag,
+                      "duration": time.time() - t0}
+            if self.on_provisioned:
+                self.on_provisioned(result)
+            return result
+        self.states[serial] = ST_UNCONFIGURED
+        self.pending[serial] = {"serial": serial, "pon_port": pon_port,
+                                "ts": t0}
+        last_err = None
+        for attempt in range(self.retries + 1):
+            if attempt and self.retry_delay:
+                time.sleep(self.retry_delay)
+            try:
+                nte = self.manager.provision(
+                    nte.id, profile=self.default_profile)
+            except Exception as e:
+                last_err = e
+                continue
+            self.states[serial] = ST_CONNECTED
+            self.pending.pop(serial, None)
+            self.stats["provisioned"] += 1
+            result = {"nte_id": nte.id, "success": True,
+                      "s_tag": nte.s_tag, "c_tag": nte.c_tag,
+                      "duration": time.time() - t0}
+            if self.on_provisioned:
+                self.on_provisioned(result)
+            return result
+        self.stats["failed"] += 1
+        result = {"nte_id": f"nte-{serial}", "success": False,
+                  "error": str(last_err),
+                  "duration": time.time() - t0}
+        if self.on_provisioned:
+            self.on_provisioned(result)
+        return result
+
+    def handle_disconnect(self, serial: str):
+        """ref HandleDisconnect manager.go:398-427."""
+        self.states[serial] = ST_DISCONNECTED
+        self.pending.pop(serial, None)
+        self.stats["disconnected"] += 1
+        self.manager.ont_offline(f"nte-{serial}")
+        if self.on_disconnected:
+            self.on_disconnected(serial)
+
+    def nte_state(self, serial: str) -> str:
+        return self.states.get(serial, ST_UNKNOWN)
+
+    def list_connected(self) -> List[str]:
+        return sorted(s for s, st in self.states.items()
+                      if st == ST_CONNECTED)
+
+    def list_pending(self) -> List[dict]:
+        return list(self.pending.values())

@@ -887,3 +887,58 @@ class TestWifiManager:
         assert c.allocation_trigger == "radius_auth"
         assert c.deallocation_trigger == "session_termination"
         assert not c.captive_portal_enabled
+
+
+class TestPONAutoProvisioner:
+    """Discovery -> provision pipeline with retries and state machine
+    (ref pkg/pon/manager.go:216-330, 398-486)."""
+
+    def _mk(self, **kw):
+        from bng_amd.pon.manager import AutoProvisioner, Manager as PON
+        from bng_amd.qinq.mapper import Mapper
+        store = MemoryStore()
+        vm = Mapper()
+        vm.add_range(100)
+        pon = PON(store, vlan_mapper=vm)
+        return pon, AutoProvisioner(pon, retry_delay=0.0, **kw)
+
+    def test_discovery_provisions_with_vlans(self):
+        pon, ap = self._mk()
+        discovered, provisioned = [], []
+        ap.on_discovered = discovered.append
+        ap.on_provisioned = provisioned.append
+        r = ap.handle_discovery("SER100", "pon0/1")
+        assert r["success"] and r["s_tag"] and r["c_tag"]
+        assert ap.nte_state("SER100") == "CONNECTED"
+        assert ap.list_connected() == ["SER100"]
+        assert ap.list_pending() == []
+        assert discovered[0]["serial"] == "SER100"
+        assert provisioned[0]["nte_id"] == "nte-SER100"
+
+    def test_reconnect_keeps_existing_tags(self):
+        pon, ap = self._mk()
+        r1 = ap.handle_discovery("SER101", "pon0/1")
+        ap.handle_disconnect("SER101")
+        assert ap.nte_state("SER101") == "DISCONNECTED"
+        assert ap.stats["disconnected"] == 1
+        r2 = ap.handle_discovery("SER101", "pon0/1")
+        assert r2["success"]
+        assert (r2["s_tag"], r2["c_tag"]) == (r1["s_tag"], r1["c_tag"])
+        assert ap.stats["reconnected"] == 1
+
+    def test_provisioning_retries_then_fails(self):
+        pon, ap = self._mk(retries=2, default_profile="missing-profile")
+        results = []
+        ap.on_provisioned = results.append
+        r = ap.handle_discovery("SER102", "pon0/1")
+        assert not r["success"]
+        assert "missing-profile" in r["error"]
+        assert ap.nte_state("SER102") == "UNCONFIGURED"
+        assert len(ap.list_pending()) == 1
+        assert ap.stats["failed"] == 1
+        # operator adds the profile; rediscovery succeeds
+        from bng_amd.pon.manager import QoSProfile
+        pon.add_profile(QoSProfile("missing-profile", 10, 5))
+        r2 = ap.handle_discovery("SER102", "pon0/1")
+        assert r2["success"]
+        assert ap.list_pending() == []
