@@ -942,3 +942,89 @@ class TestPONAutoProvisioner:
         r2 = ap.handle_discovery("SER102", "pon0/1")
         assert r2["success"]
         assert ap.list_pending() == []
+
+
+class TestONTAuthenticator:
+    """ONT-mapping-driven direct auth feeding the subscriber manager's
+    rich protocol (ref pkg/direct/authenticator.go:93-470)."""
+
+    def _auth(self):
+        from bng_amd.direct.authenticator import (MappingBSS, ONTMapping,
+                                                  ONTAuthenticator)
+        bss = MappingBSS()
+        bss.add_mapping(ONTMapping(
+            ont_serial="SER-1", subscriber_id="sub-1",
+            circuit_id="olt1/pon0/1", isp_id="isp-a",
+            qos_policy="gold", download_bps=500_000_000,
+            upload_bps=100_000_000))
+        bss.add_mapping(ONTMapping(
+            ont_serial="SER-2", subscriber_id="sub-2",
+            status="suspended"))
+        bss.add_mapping(ONTMapping(
+            ont_serial="SER-3", subscriber_id="sub-3",
+            status="disconnected"))
+        return bss, ONTAuthenticator(bss, default_isp="isp-def",
+                                     session_timeout=3600)
+
+    def test_full_flow_through_subscriber_manager(self):
+        from bng_amd.subscriber.manager import Manager, S_AUTHENTICATED
+        bss, auth = self._auth()
+        m = Manager(authenticator=auth)
+        s = m.open_session("pending", mac="aa:00:00:00:00:01")
+        s.circuit_id = "olt1/pon0/1"
+        r = m.authenticate_full(s.id)
+        assert r["success"]
+        assert s.state == S_AUTHENTICATED
+        assert s.subscriber_id == "sub-1" and s.isp_id == "isp-a"
+        assert s.qos_policy_id == "gold"
+        assert s.download_rate_bps == 500_000_000
+        assert s.session_timeout == 3600
+
+    def test_suspended_goes_walled(self):
+        from bng_amd.subscriber.manager import Manager, S_WALLED
+        bss, auth = self._auth()
+        m = Manager(authenticator=auth)
+        s = m.open_session("x")
+        s.nte_id = "SER-2"
+        assert m.authenticate_full(s.id)["success"]
+        assert s.state == S_WALLED
+        assert "suspended" in s.walled_reason.lower()
+
+    def test_disconnected_and_unknown_fail(self):
+        bss, auth = self._auth()
+
+        class S:
+            circuit_id = ""
+            nte_id = "SER-3"
+        r = auth.authenticate_session(S(), {})
+        assert not r["success"] and "disconnected" in r["error"]
+        S.nte_id = "SER-404"
+        assert not auth.authenticate_session(S(), {})["success"]
+        assert auth.stats["not_found"] == 1
+
+    def test_cache_and_invalidate(self):
+        bss, auth = self._auth()
+
+        class S:
+            circuit_id = ""
+            nte_id = "SER-1"
+        auth.authenticate_session(S(), {})
+        auth.authenticate_session(S(), {})
+        assert auth.stats["cache_hits"] == 1
+        # BSS change invisible until invalidation
+        bss.by_serial["SER-1"].status = "suspended"
+        assert "walled_garden" not in auth.authenticate_session(S(), {})
+        auth.invalidate_cache(ont_serial="SER-1")
+        assert auth.authenticate_session(S(), {}).get("walled_garden")
+        assert auth.sync_from_bss() == 3
+
+    def test_binding_events(self):
+        import pytest as _pt
+        bss, auth = self._auth()
+        auth.report_binding("assign", "SER-1", "sub-1",
+                            mac="aa:00:00:00:00:01", ipv4="10.0.1.5")
+        auth.report_binding("release", "SER-1", "sub-1")
+        assert [b["event_type"] for b in bss.bindings] == \
+            ["assign", "release"]
+        with _pt.raises(ValueError):
+            auth.report_binding("wat", "SER-1", "sub-1")
