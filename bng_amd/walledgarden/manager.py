@@ -27,21 +27,43 @@ class Entry:
     reason: str = ""
 
 
+# allowed-destination reason codes (ref AllowedDestination.Reason
+# manager.go:56-63 / initAllowedDestinations :187-235)
+REASON_DNS = 1
+REASON_PORTAL = 2
+REASON_CUSTOM = 3
+
+# classification verdicts for quarantined traffic
+V_FORWARD = "forward"
+V_REDIRECT = "redirect"
+V_DROP = "drop"
+
+
 class Manager:
     def __init__(self, portal_ip: str = "", dns_servers: Optional[List[str]] = None,
-                 default_ttl: float = 3600.0):
+                 default_ttl: float = 3600.0, portal_port: int = 8080):
         self.portal_ip = portal_ip
+        self.portal_port = portal_port
         self.dns_servers = dns_servers or []
         self.default_ttl = default_ttl
         self.entries: Dict[str, Entry] = {}
         self.allowed_destinations: List[str] = (
             ([portal_ip] if portal_ip else []) + self.dns_servers)
+        # keyed (ip, port, proto) -> reason (ref allowedDestKey :237-242)
+        self.allowed_dests: Dict[tuple, int] = {}
+        for dns in self.dns_servers:
+            self.allowed_dests[(dns, 53, 17)] = REASON_DNS
+            self.allowed_dests[(dns, 53, 6)] = REASON_DNS
+        if portal_ip:
+            self.allowed_dests[(portal_ip, portal_port, 6)] = REASON_PORTAL
         self._lock = threading.RLock()
         self._hooks: List[Callable[[Entry], None]] = []
+        self._redirect_cbs: List[Callable[[str, str], None]] = []
         self._stop = threading.Event()
         self._checker: Optional[threading.Thread] = None
         self.stats = {"added": 0, "activated": 0, "blocked": 0,
-                      "expired": 0}
+                      "expired": 0, "redirects": 0, "dropped": 0,
+                      "allowed": 0}
 
     def set_dataplane_hooks(self, on_change: Callable[[Entry], None]):
         """Optional enforcement hook (ref SetEBPFMaps manager.go:173-180)."""
@@ -104,9 +126,57 @@ class Manager:
     def is_quarantined(self, mac: str) -> bool:
         return self.state_of(mac) == STATE_WALLED
 
-    def is_destination_allowed(self, ip: str) -> bool:
-        """For quarantined clients only DNS + portal are reachable."""
-        return ip in self.allowed_destinations
+    def is_destination_allowed(self, ip: str, port: int = 0,
+                               proto: int = 0) -> bool:
+        """For quarantined clients only DNS + portal (+ configured
+        extras) are reachable.  Bare-IP calls keep the legacy
+        any-port semantics; (ip, port, proto) checks the keyed table
+        (ref allowedDestKey manager.go:237-242)."""
+        if port == 0 and proto == 0:
+            return ip in self.allowed_destinations
+        return (ip, port, proto) in self.allowed_dests
+
+    def allow_destination(self, ip: str, port: int, proto: int,
+                          reason: int = REASON_CUSTOM):
+        """ref Config.AllowedDestinations / initAllowedDestinations
+        :216-230."""
+        self.allowed_dests[(ip, port, proto)] = reason
+        if ip not in self.allowed_destinations:
+            self.allowed_destinations.append(ip)
+
+    def on_redirect(self, cb: Callable[[str, str], None]):
+        """cb(mac, dst_ip) whenever quarantined HTTP gets redirected
+        to the portal (ref OnRedirect manager.go:182-185)."""
+        self._redirect_cbs.append(cb)
+
+    def classify(self, mac: str, dst_ip: str, dst_port: int = 0,
+                 proto: int = 6) -> str:
+        """Quarantine-time verdict for one flow: active subscribers
+        and unknown MACs forward; blocked MACs drop; quarantined
+        traffic forwards to allowed destinations, HTTP redirects to
+        the captive portal (firing redirect callbacks), everything
+        else drops — the walled-garden decision table the reference
+        encodes in its eBPF program."""
+        st = self.state_of(mac)
+        if st == STATE_BLOCKED:
+            self.stats["dropped"] += 1
+            return V_DROP
+        if st != STATE_WALLED:
+            return V_FORWARD
+        if self.is_destination_allowed(dst_ip, dst_port, proto) or \
+                (dst_port == 0 and dst_ip in self.allowed_destinations):
+            self.stats["allowed"] += 1
+            return V_FORWARD
+        if proto == 6 and dst_port in (80, 8080):
+            self.stats["redirects"] += 1
+            for cb in self._redirect_cbs:
+                try:
+                    cb(mac.lower(), dst_ip)
+                except Exception:
+                    pass
+            return V_REDIRECT
+        self.stats["dropped"] += 1
+        return V_DROP
 
     # ------------------------------------------------------------ expiry
     def _check_loop(self, interval: float):

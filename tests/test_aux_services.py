@@ -1028,3 +1028,59 @@ class TestONTAuthenticator:
             ["assign", "release"]
         with _pt.raises(ValueError):
             auth.report_binding("wat", "SER-1", "sub-1")
+
+
+class TestWalledGardenClassify:
+    """Keyed allowed destinations + flow classification with portal
+    redirect (ref pkg/walledgarden/manager.go:56-242)."""
+
+    def _mgr(self):
+        from bng_amd.walledgarden.manager import Manager as WG
+        return WG(portal_ip="10.255.255.1", portal_port=8080,
+                  dns_servers=["8.8.8.8"])
+
+    def test_default_allowed_dests(self):
+        from bng_amd.walledgarden.manager import (REASON_DNS,
+                                                  REASON_PORTAL)
+        m = self._mgr()
+        assert m.is_destination_allowed("8.8.8.8", 53, 17)
+        assert m.is_destination_allowed("8.8.8.8", 53, 6)
+        assert m.is_destination_allowed("10.255.255.1", 8080, 6)
+        assert not m.is_destination_allowed("10.255.255.1", 443, 6)
+        assert m.allowed_dests[("8.8.8.8", 53, 17)] == REASON_DNS
+        assert m.allowed_dests[("10.255.255.1", 8080, 6)] == \
+            REASON_PORTAL
+        # legacy bare-IP call still works
+        assert m.is_destination_allowed("8.8.8.8")
+
+    def test_classification_table(self):
+        from bng_amd.walledgarden.manager import (V_DROP, V_FORWARD,
+                                                  V_REDIRECT)
+        m = self._mgr()
+        redirects = []
+        m.on_redirect(lambda mac, ip: redirects.append((mac, ip)))
+        m.add("AA:00:00:00:00:01", "10.0.1.5")
+        # quarantined: DNS ok, HTTP redirects, HTTPS drops
+        assert m.classify("aa:00:00:00:00:01", "8.8.8.8", 53, 17) == \
+            V_FORWARD
+        assert m.classify("aa:00:00:00:00:01", "93.184.216.34", 80,
+                          6) == V_REDIRECT
+        assert redirects == [("aa:00:00:00:00:01", "93.184.216.34")]
+        assert m.classify("aa:00:00:00:00:01", "93.184.216.34", 443,
+                          6) == V_DROP
+        # custom allowed destination opens up
+        m.allow_destination("93.184.216.34", 443, 6)
+        assert m.classify("aa:00:00:00:00:01", "93.184.216.34", 443,
+                          6) == V_FORWARD
+        # activation ends quarantine entirely
+        m.activate("aa:00:00:00:00:01")
+        assert m.classify("aa:00:00:00:00:01", "1.2.3.4", 443, 6) == \
+            V_FORWARD
+        # blocked drops everything, allowed or not
+        m.block("aa:00:00:00:00:01")
+        assert m.classify("aa:00:00:00:00:01", "8.8.8.8", 53, 17) == \
+            V_DROP
+        # unknown MAC is not the garden's business
+        assert m.classify("ff:ff:ff:00:00:00", "1.2.3.4", 443, 6) == \
+            V_FORWARD
+        assert m.stats["redirects"] == 1
