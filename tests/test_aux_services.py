@@ -1084,3 +1084,61 @@ class TestWalledGardenClassify:
         assert m.classify("ff:ff:ff:00:00:00", "1.2.3.4", 443, 6) == \
             V_FORWARD
         assert m.stats["redirects"] == 1
+
+
+class TestQinQConfiguredMapper:
+    """Range-validated VLAN mapping + lookup priority (ref
+    pkg/qinq/qinq.go:18-212)."""
+
+    def test_pair_semantics(self):
+        from bng_amd.qinq.mapper import VLANPair
+        assert str(VLANPair(100, 7)) == "s100.c7"
+        assert str(VLANPair(0, 7)) == "c7"
+        assert VLANPair(100, 7).is_double_tagged
+        assert VLANPair(0, 7).is_single_tagged
+        assert VLANPair().is_untagged
+
+    def test_range_validation_and_moves(self):
+        from bng_amd.qinq.mapper import (ConfiguredMapper, QinQConfig,
+                                         QinQError, VLANPair)
+        m = ConfiguredMapper(QinQConfig(
+            s_tag_ranges=[(100, 199, "isp-a"), (300, 399, "isp-b")],
+            c_tag_range=(10, 20)))
+        m.register(VLANPair(100, 10), "sub-1")
+        m.register(VLANPair(300, 20), "sub-2")
+        with pytest.raises(QinQError):
+            m.register(VLANPair(200, 10), "sub-3")   # S out of range
+        with pytest.raises(QinQError):
+            m.register(VLANPair(100, 50), "sub-3")   # C out of range
+        with pytest.raises(QinQError):
+            m.register(VLANPair(100, 10), "sub-9")   # owned
+        # re-registering moves the subscriber, freeing the old pair
+        m.register(VLANPair(100, 11), "sub-1")
+        assert m.get_subscriber(VLANPair(100, 10)) is None
+        assert m.get_vlan("sub-1") == VLANPair(100, 11)
+        m.unregister_subscriber("sub-2")
+        assert m.get_subscriber(VLANPair(300, 20)) is None
+        assert m.stats()["total_mappings"] == 1
+
+    def test_lookup_priority(self):
+        from bng_amd.qinq.mapper import (ConfiguredMapper, QinQConfig,
+                                         VLANPair)
+        mac_table = {"aa:bb": "sub-mac"}
+        pair = VLANPair(100, 110)
+
+        def mk(prio):
+            m = ConfiguredMapper(QinQConfig(lookup_priority=prio))
+            m.register(pair, "sub-vlan")
+            return m
+
+        look = lambda mac: mac_table.get(mac)
+        assert mk("vlan_first").lookup(pair, look, "aa:bb") == "sub-vlan"
+        assert mk("mac_first").lookup(pair, look, "aa:bb") == "sub-mac"
+        assert mk("vlan_only").lookup(pair, look, "aa:bb") == "sub-vlan"
+        # miss on the preferred path falls through
+        m = mk("vlan_first")
+        assert m.lookup(VLANPair(100, 99), look, "aa:bb") == "sub-mac"
+        assert mk("vlan_only").lookup(VLANPair(100, 99), look,
+                                      "aa:bb") is None
+        with pytest.raises(Exception):
+            QinQConfig(lookup_priority="wat")
