@@ -94,3 +94,84 @@ class Manager:
         with self._lock:
             return time.time() - self.partitioned_at \
                 if self.partitioned_at else 0.0
+
+
+class Reconciler:
+    """Partition-heal pipeline (ref manager.go performReconciliation
+    :342-427): (1) detect allocation conflicts against the remote view
+    and resolve them with the site policy, (2) queue re-auths for every
+    session admitted on a degraded answer, (3) process them rate-
+    limited, (4) replay buffered accounting, (5) drain the deferred
+    request queue.  Returns the ReconciliationResult dict."""
+
+    def __init__(self, detector=None, radius=None, request_queue=None,
+                 reauth_rate_limit: int = 0,
+                 on_conflict: Optional[Callable] = None):
+        self.detector = detector
+        self.radius = radius
+        self.request_queue = request_queue
+        self.reauth_rate_limit = reauth_rate_limit
+        self.on_conflict = on_conflict
+
+    def reconcile(self, remote_allocations=None,
+                  credentials=None) -> dict:
+        t0 = time.time()
+        result = {"started_at": t0, "conflicts_found": 0,
+                  "conflicts_resolved": 0, "reauths_queued": 0,
+                  "reauths_completed": 0, "reauths_failed": 0,
+                  "acct_records_synced": 0, "requests_drained": 0,
+                  "errors": []}
+        if self.detector is not None and remote_allocations is not None:
+            conflicts = self.detector.detect(remote_allocations)
+            result["conflicts_found"] = len(conflicts)
+            for c in conflicts:
+                try:
+                    self.detector.resolve(c)
+                    result["conflicts_resolved"] += 1
+                    if self.on_conflict:
+                        self.on_conflict(c)
+                except Exception as e:
+                    result["errors"].append(f"resolve {c.ip}: {e}")
+            self.detector.clear_partition_flags()
+        if self.radius is not None:
+            degraded = self.radius.degraded_sessions()
+            result["reauths_queued"] = len(degraded)
+            for user in degraded:
+                self.radius.queue_reauth(user)
+            done, failed = self.radius.process_reauths(
+                self.reauth_rate_limit, credentials)
+            result["reauths_completed"] = done
+            result["reauths_failed"] = failed
+            result["acct_records_synced"] = self.radius.replay_buffered()
+        if self.request_queue is not None:
+            result["requests_drained"] = self.request_queue.drain()
+        result["duration"] = time.time() - t0
+        return result
+
+
+class ShortLeasePolicy:
+    """While partitioned, hand out short DHCP leases so address churn
+    stays reconcilable after heal (ref manager.go ShouldUseShortLease
+    :620-641 — driven by partition state + pool pressure)."""
+
+    def __init__(self, manager: Manager, short_lease: float = 300.0,
+                 normal_lease: float = 86400.0,
+                 pool_monitor=None):
+        self.manager = manager
+        self.short_lease = short_lease
+        self.normal_lease = normal_lease
+        self.pool_monitor = pool_monitor
+        self.short_leases_issued = 0
+
+    def should_use_short_lease(self) -> bool:
+        if self.manager.is_partitioned:
+            return True
+        if self.pool_monitor is not None:
+            return self.pool_monitor.check() in ("warning", "critical")
+        return False
+
+    def lease_time(self) -> float:
+        if self.should_use_short_lease():
+            self.short_leases_issued += 1
+            return self.short_lease
+        return self.normal_lease

@@ -37,7 +37,13 @@ class ResilientRadius:
         self._lock = threading.RLock()
         self.stats = {"cache_hits": 0, "cache_answers": 0,
                       "allow_answers": 0, "rejects": 0,
-                      "acct_buffered": 0, "acct_replayed": 0}
+                      "acct_buffered": 0, "acct_replayed": 0,
+                      "reauths_ok": 0, "reauths_failed": 0}
+        # usernames admitted on degraded answers -> owed a real
+        # authentication once the partition heals (ref
+        # radius_handler.go GetDegradedSessions/ProcessReauths)
+        self._degraded: Dict[str, dict] = {}
+        self._reauth_queue: List[str] = []
 
     def authenticate(self, username: str, password: str, **kw) -> AuthResult:
         try:
@@ -61,11 +67,17 @@ class ResilientRadius:
             if prof is not None and \
                     time.time() - prof.cached_at <= self.cache_ttl:
                 self.stats["cache_answers"] += 1
+                with self._lock:
+                    self._degraded[username] = {"mode": "cached",
+                                                "at": time.time()}
                 return prof.result
             self.stats["rejects"] += 1
             return AuthResult(False, reply_message="no cached profile")
         # MODE_ALLOW: admit with defaults (degraded service)
         self.stats["allow_answers"] += 1
+        with self._lock:
+            self._degraded[username] = {"mode": "allow",
+                                        "at": time.time()}
         return AuthResult(True, policy_name="")
 
     # ------------------------------------------------------- accounting
@@ -98,3 +110,48 @@ class ResilientRadius:
                 self.buffered_acct.append((args, kw))
         self.stats["acct_replayed"] += done
         return done
+
+
+    # -------------------------------------------- recovery re-auth
+    def degraded_sessions(self) -> List[str]:
+        """Usernames admitted on cached/allow answers during the
+        partition (ref GetDegradedSessions)."""
+        with self._lock:
+            return list(self._degraded)
+
+    def queue_reauth(self, username: str):
+        with self._lock:
+            if username not in self._reauth_queue:
+                self._reauth_queue.append(username)
+
+    def process_reauths(self, rate_limit: int = 0,
+                        credentials=None) -> tuple:
+        """Re-authenticate queued degraded sessions against the real
+        RADIUS, at most rate_limit per call (0 = all).  credentials:
+        optional username -> (password, kwargs) provider; without it
+        the cached profile is revalidated by a bare authenticate.
+        Returns (completed, failed); failures stay degraded (ref
+        ProcessReauths rate-limited loop)."""
+        with self._lock:
+            batch = self._reauth_queue[:rate_limit or None]
+            self._reauth_queue = self._reauth_queue[len(batch):]
+        done = failed = 0
+        for user in batch:
+            try:
+                if credentials is not None:
+                    pw, kw = credentials(user)
+                    res = self.client.authenticate(user, pw, **kw)
+                else:
+                    res = self.client.authenticate(user, "")
+                ok = bool(res.success)
+            except Exception:
+                ok = False
+            if ok:
+                done += 1
+                with self._lock:
+                    self._degraded.pop(user, None)
+            else:
+                failed += 1
+        self.stats["reauths_ok"] += done
+        self.stats["reauths_failed"] += failed
+        return done, failed

@@ -187,3 +187,144 @@ class TestSplitBrain:
         assert c.keeper.subscriber_id == "sub-a"       # oldest wins
         assert [a.subscriber_id for a in c.evicted] == ["sub-b"]
         assert seen == ["10.0.1.5"]
+
+
+class TestSiteConflictDetector:
+    """Site-aware conflict detection + the reference's resolution
+    policy (ref conflict_detector.go:121-330, manager.go:430-527)."""
+
+    def _d(self):
+        from bng_amd.resilience.conflict_detector import (
+            IPAllocation, SiteConflictDetector)
+        d = SiteConflictDetector("site-a")
+        return d, IPAllocation
+
+    def test_same_subscriber_is_no_conflict(self):
+        d, A = self._d()
+        d.record("10.0.1.5", "aa:01", "sub-1", allocated_at=100)
+        remote = [A("10.0.1.5", "AA:01", "sub-1", "site-b", 90)]
+        assert d.detect(remote) == []
+
+    def test_cross_site_conflict_and_policy(self):
+        from bng_amd.resilience.conflict_detector import (
+            R_LOCAL_WINS, R_REMOTE_WINS, SiteConflictDetector)
+        d, A = self._d()
+        # local pre-partition vs remote during-partition -> local wins
+        d.record("10.0.1.5", "aa:01", "sub-1", allocated_at=100)
+        c = d.detect([A("10.0.1.5", "bb:02", "sub-2", "site-b", 200,
+                        is_partition=True)])[0]
+        d.resolve(c)
+        assert c.resolution == R_LOCAL_WINS
+        assert c.affected_mac == "bb:02"
+        # local during-partition vs remote pre-partition -> remote wins
+        d2 = SiteConflictDetector("site-a")
+        d2.record("10.0.2.5", "aa:01", "sub-1", allocated_at=300,
+                  is_partition=True)
+        c2 = d2.detect([A("10.0.2.5", "bb:02", "sub-2", "site-b",
+                          100)])[0]
+        d2.resolve(c2)
+        assert c2.resolution == R_REMOTE_WINS
+        assert c2.affected_mac == "aa:01"
+        # both partition-era -> most recent wins
+        d3 = SiteConflictDetector("site-a")
+        d3.record("10.0.3.5", "aa:01", "sub-1", allocated_at=500,
+                  is_partition=True)
+        c3 = d3.detect([A("10.0.3.5", "bb:02", "sub-2", "site-b", 400,
+                          is_partition=True)])[0]
+        d3.resolve(c3)
+        assert c3.resolution == R_LOCAL_WINS
+
+    def test_validate_raises_on_foreign_holder(self):
+        import pytest as _pt
+        from bng_amd.resilience.conflict_detector import ConflictError
+        d, A = self._d()
+        d.record("10.0.1.5", "aa:01", "sub-1")
+        d.validate("10.0.1.5", "AA:01")        # same MAC ok
+        with _pt.raises(ConflictError):
+            d.validate("10.0.1.5", "bb:02")
+        d.remove("10.0.1.5")
+        d.validate("10.0.1.5", "bb:02")        # freed
+        # export/import roundtrip
+        d.record("10.0.9.9", "cc:03", "sub-3", is_partition=True)
+        from bng_amd.resilience.conflict_detector import \
+            SiteConflictDetector
+        d2 = SiteConflictDetector("site-a")
+        d2.import_allocations(d.export_allocations())
+        assert d2.get("10.0.9.9").mac == "cc:03"
+        assert len(d2.partition_allocations()) == 1
+        d2.clear_partition_flags()
+        assert d2.partition_allocations() == []
+
+
+class TestReconciler:
+    """Partition-heal pipeline (ref manager.go:342-427)."""
+
+    def test_full_reconciliation(self):
+        from bng_amd.radius.client import AuthResult, RadiusTimeout
+        from bng_amd.resilience.conflict_detector import (
+            IPAllocation, SiteConflictDetector)
+        from bng_amd.resilience.manager import Reconciler
+        from bng_amd.resilience.radius_handler import ResilientRadius
+        from bng_amd.resilience.request_queue import RequestQueue
+
+        class FlappyClient:
+            def __init__(self):
+                self.down = True
+                self.acct = []
+
+            def authenticate(self, user, pw, **kw):
+                if self.down:
+                    raise RadiusTimeout()
+                return AuthResult(True, policy_name="gold")
+
+            def send_accounting(self, *a, **kw):
+                if self.down:
+                    raise RadiusTimeout()
+                self.acct.append(a)
+                return True
+
+        client = FlappyClient()
+        rr = ResilientRadius(client, mode="allow")
+        # partition: degraded admits + buffered accounting
+        assert rr.authenticate("alice", "pw").success
+        assert rr.authenticate("bob", "pw").success
+        rr.send_accounting("alice", 100)
+        assert rr.stats["acct_buffered"] == 1
+        assert sorted(rr.degraded_sessions()) == ["alice", "bob"]
+        det = SiteConflictDetector("site-a")
+        det.record("10.0.1.5", "aa:01", "alice", allocated_at=10,
+                   is_partition=True)
+        q = RequestQueue()
+        drained = []
+        q.enqueue("update", lambda: drained.append(1) or True)
+        conflicts_seen = []
+        rec = Reconciler(det, rr, q,
+                         on_conflict=conflicts_seen.append)
+        # heal
+        client.down = False
+        remote = [IPAllocation("10.0.1.5", "bb:02", "carol", "site-b",
+                               5)]
+        result = rec.reconcile(remote,
+                               credentials=lambda u: ("pw", {}))
+        assert result["conflicts_found"] == 1
+        assert result["conflicts_resolved"] == 1
+        assert conflicts_seen[0].resolution == "remote_wins"
+        assert result["reauths_queued"] == 2
+        assert result["reauths_completed"] == 2
+        assert result["acct_records_synced"] == 1
+        assert result["requests_drained"] == 1
+        assert drained == [1]
+        assert rr.degraded_sessions() == []
+        assert det.partition_allocations() == []
+
+    def test_short_lease_policy(self):
+        from bng_amd.resilience.manager import (Manager, ShortLeasePolicy,
+                                                STATE_PARTITIONED)
+        m = Manager(health_check=lambda: False, failure_threshold=1)
+        pol = ShortLeasePolicy(m, short_lease=300, normal_lease=86400)
+        assert pol.lease_time() == 86400
+        m.check_once()                          # -> partitioned
+        assert m.state == STATE_PARTITIONED
+        assert pol.should_use_short_lease()
+        assert pol.lease_time() == 300
+        assert pol.short_leases_issued == 1
