@@ -13,6 +13,7 @@ from dataclasses import dataclass, field
 from typing import Callable, Dict, List, Optional, Tuple
 
 QTYPE_A = 1
+QTYPE_CNAME = 5
 QTYPE_SRV = 33
 QTYPE_AAAA = 28
 
@@ -63,6 +64,17 @@ def build_response(query: bytes, addrs: List[str], ttl: int = 300,
     return out
 
 
+def build_cname_response(query: bytes, target: str,
+                         ttl: int = 300) -> bytes:
+    """CNAME answer (ref createCNAMEResponse resolver.go:515-530)."""
+    txid = struct.unpack_from(">H", query, 0)[0]
+    _qname, off = decode_qname(query, 12)
+    rd = encode_qname(target)
+    hdr = struct.pack(">HHHHHH", txid, 0x8180, 1, 1, 0, 0)
+    return hdr + query[12:off + 4] + b"\xc0\x0c" + \
+        struct.pack(">HHIH", QTYPE_CNAME, 1, ttl, len(rd)) + rd
+
+
 def parse_response(data: bytes) -> Tuple[str, List[str], int]:
     """-> (qname, addresses, min_ttl)"""
     import ipaddress
@@ -106,6 +118,8 @@ class Resolver:
             OrderedDict()
         self.intercepts: Dict[str, List[str]] = {}   # name -> portal IPs
         self.intercept_all_to: Optional[List[str]] = None
+        self.rules: List[dict] = []          # typed interception rules
+        self.walled_clients: set = set()     # client IPs in quarantine
         self._lock = threading.RLock()
         self.rate_limit = rate_limit
         self.rate_burst = rate_burst
@@ -127,6 +141,71 @@ class Resolver:
     def remove_intercept(self, name: str):
         with self._lock:
             self.intercepts.pop(name.rstrip(".").lower(), None)
+
+    # typed interception rules (ref checkInterceptionRules
+    # resolver.go:444-530; actions allow/block/redirect/cname, match
+    # modes exact / suffix / wildcard-subdomain :468-490)
+    def add_rule(self, domain: str = "", action: str = "block",
+                 redirect: Optional[List[str]] = None, cname: str = "",
+                 exact: bool = False, suffix: str = ""):
+        if action not in ("allow", "block", "redirect", "cname"):
+            raise ValueError(f"unknown intercept action {action}")
+        with self._lock:
+            self.rules.append({"domain": domain.rstrip(".").lower(),
+                               "suffix": suffix.lower(),
+                               "action": action,
+                               "redirect": redirect or [],
+                               "cname": cname, "exact": exact})
+
+    def remove_rule(self, domain: str) -> bool:
+        d = domain.rstrip(".").lower()
+        with self._lock:
+            before = len(self.rules)
+            self.rules = [r for r in self.rules if r["domain"] != d]
+            return len(self.rules) != before
+
+    @staticmethod
+    def _rule_matches(rule: dict, qname: str) -> bool:
+        if rule["exact"]:
+            return qname == rule["domain"]
+        if rule["suffix"]:
+            return qname.endswith(rule["suffix"])
+        d = rule["domain"]
+        return bool(d) and (qname == d or qname.endswith("." + d))
+
+    def _check_rules(self, query: bytes, qname: str):
+        """-> response bytes or None (allow)."""
+        with self._lock:
+            rules = list(self.rules)
+        for r in rules:
+            if not self._rule_matches(r, qname):
+                continue
+            if r["action"] == "allow":
+                return None
+            self.stats["intercepted"] += 1
+            if r["action"] == "block":
+                return build_response(query, [], ttl=30, rcode=3)
+            if r["action"] == "redirect":
+                return build_response(query, r["redirect"], ttl=300)
+            return build_cname_response(query, r["cname"])
+        return None
+
+    # per-client walled-garden registry (ref AddWalledGardenClient
+    # resolver.go:238-270): queries from these IPs resolve everything
+    # to the portal
+    def add_walled_client(self, ip: str):
+        with self._lock:
+            self.walled_clients.add(ip)
+
+    def remove_walled_client(self, ip: str) -> bool:
+        with self._lock:
+            had = ip in self.walled_clients
+            self.walled_clients.discard(ip)
+            return had
+
+    def is_walled(self, ip: str) -> bool:
+        with self._lock:
+            return ip in self.walled_clients
 
     # ------------------------------------------------------------ resolve
     def _allowed(self, client: str) -> bool:
@@ -155,7 +234,13 @@ class Resolver:
         except (struct.error, IndexError):
             return None
         key = qname.lower()
-        # intercept rules first (walled garden)
+        # typed rules first (block/redirect/cname)
+        ruled = self._check_rules(query, key)
+        if ruled is not None:
+            return ruled
+        # walled garden: flagged caller or registered client IP
+        if client and not quarantined:
+            quarantined = self.is_walled(client)
         if quarantined and self.intercept_all_to:
             self.stats["intercepted"] += 1
             return build_response(query, self.intercept_all_to, ttl=30)

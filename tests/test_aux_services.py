@@ -1142,3 +1142,88 @@ class TestQinQConfiguredMapper:
                                       "aa:bb") is None
         with pytest.raises(Exception):
             QinQConfig(lookup_priority="wat")
+
+
+class TestDNSInterceptRules:
+    """Typed interception rules + per-client walled garden (ref
+    pkg/dns/resolver.go:212-530)."""
+
+    def _resolver(self):
+        upstream_log = []
+
+        def upstream(q):
+            upstream_log.append(q)
+            return build_response(q, ["93.184.216.34"], ttl=300)
+        r = Resolver(upstream)
+        return r, upstream_log
+
+    def test_block_redirect_cname_actions(self):
+        import struct as _st
+        from bng_amd.dns.resolver import (build_cname_response,
+                                          decode_qname)
+        r, log = self._resolver()
+        r.add_rule("ads.example", action="block")
+        r.add_rule("old.example", action="redirect",
+                   redirect=["10.9.9.9"])
+        r.add_rule("www.legacy", action="cname",
+                   cname="portal.isp.net")
+        # block -> NXDOMAIN, no upstream
+        resp = r.handle_query(build_query("ads.example"))
+        assert _st.unpack_from(">H", resp, 2)[0] & 0xF == 3
+        assert log == []
+        # subdomain wildcard matches too
+        resp = r.handle_query(build_query("x.ads.example"))
+        assert _st.unpack_from(">H", resp, 2)[0] & 0xF == 3
+        # redirect -> the configured IP
+        _, addrs, _ = parse_response(
+            r.handle_query(build_query("old.example")))
+        assert addrs == ["10.9.9.9"]
+        # cname -> CNAME record with the target
+        resp = r.handle_query(build_query("www.legacy"))
+        an = _st.unpack_from(">H", resp, 6)[0]
+        assert an == 1
+        _qn, off = decode_qname(resp, 12)
+        off += 4 + 2          # question + name pointer
+        rtype = _st.unpack_from(">H", resp, off)[0]
+        assert rtype == 5     # CNAME
+        target, _ = decode_qname(resp, off + 10)
+        assert target == "portal.isp.net"
+        # non-matching name goes upstream
+        r.handle_query(build_query("fine.example"))
+        assert len(log) == 1
+
+    def test_match_modes(self):
+        r, log = self._resolver()
+        r.add_rule("exact.example", action="block", exact=True)
+        r.add_rule(action="block", suffix=".tracker.net")
+        # exact does not match subdomains
+        import struct as _st
+        assert _st.unpack_from(
+            ">H", r.handle_query(build_query("exact.example")), 2
+        )[0] & 0xF == 3
+        r.handle_query(build_query("sub.exact.example"))
+        assert len(log) == 1                  # went upstream
+        assert _st.unpack_from(
+            ">H", r.handle_query(build_query("x.tracker.net")), 2
+        )[0] & 0xF == 3
+        assert r.remove_rule("exact.example")
+        r.handle_query(build_query("exact.example"))
+        assert len(log) == 2
+
+    def test_walled_client_registry(self):
+        r, log = self._resolver()
+        r.set_intercept_all(["10.255.255.1"])
+        r.add_walled_client("10.0.1.5")
+        assert r.is_walled("10.0.1.5")
+        _, addrs, _ = parse_response(
+            r.handle_query(build_query("anything.example"),
+                           client="10.0.1.5"))
+        assert addrs == ["10.255.255.1"]      # portal for walled client
+        r.handle_query(build_query("anything.example"),
+                       client="10.0.2.2")
+        assert len(log) == 1                  # other clients normal
+        assert r.remove_walled_client("10.0.1.5")
+        assert not r.remove_walled_client("10.0.1.5")
+        r.handle_query(build_query("other.example"),
+                       client="10.0.1.5")
+        assert len(log) == 2
