@@ -73,6 +73,55 @@ class Metrics:
         self.antispoof_violations = Gauge(
             "bng_antispoof_violations_total", "uRPF violations",
             registry=r)
+        # remaining reference families (ref metrics.go:16-84)
+        self.pool_available = Gauge(
+            "bng_pool_available", "free addresses", ["pool"], registry=r)
+        self.circuit_id_collisions = Counter(
+            "bng_circuit_id_collisions_total",
+            "option-82 circuit-id hash collisions", registry=r)
+        self.session_duration = Histogram(
+            "bng_session_duration_seconds", "session lifetime",
+            buckets=(60, 300, 1800, 3600, 14400, 86400, 604800),
+            registry=r)
+        self.session_bytes_in = Counter(
+            "bng_session_bytes_in_total", "subscriber ingress bytes",
+            ["access_type"], registry=r)
+        self.session_bytes_out = Counter(
+            "bng_session_bytes_out_total", "subscriber egress bytes",
+            ["access_type"], registry=r)
+        self.nat_translations = Counter(
+            "bng_nat_translations_total", "NAT translations",
+            ["direction"], registry=r)
+        self.nat_ports_used = Gauge(
+            "bng_nat_ports_used", "allocated NAT ports",
+            ["public_ip"], registry=r)
+        self.radius_latency = Histogram(
+            "bng_radius_request_duration_seconds", "RADIUS round-trip",
+            buckets=(1e-3, 5e-3, 1e-2, 5e-2, 1e-1, 5e-1, 1, 5),
+            registry=r)
+        self.radius_timeouts = Counter(
+            "bng_radius_timeouts_total", "RADIUS timeouts", ["server"],
+            registry=r)
+        self.pppoe_sessions = Gauge(
+            "bng_pppoe_sessions_active", "established PPPoE sessions",
+            registry=r)
+        self.pppoe_negotiations = Counter(
+            "bng_pppoe_negotiations_total", "PPPoE phase outcomes",
+            ["phase", "result"], registry=r)
+        self.routes_active = Gauge(
+            "bng_routes_active", "installed routes", ["proto"],
+            registry=r)
+        self.bgp_peers_up = Gauge(
+            "bng_bgp_peers_up", "established BGP peers", registry=r)
+        self.subscriber_total = Gauge(
+            "bng_subscribers_total", "known subscribers", registry=r)
+        self.subscriber_by_isp = Gauge(
+            "bng_subscribers_by_isp", "subscribers per ISP", ["isp"],
+            registry=r)
+        self.table_entries = Gauge(
+            "bng_dataplane_table_entries",
+            "GPU table occupancy (the eBPF map-entries analog)",
+            ["table"], registry=r)
         self._stop = threading.Event()
         self._collector: Optional[threading.Thread] = None
         self._extra_collectors = []

@@ -773,3 +773,44 @@ class TestSubscriberManagerDepth:
         st = m.manager_stats()
         assert st["total_bytes_in"] == 110
         assert not m.update_activity("nope")
+
+
+class TestMetricsFamilies:
+    """The full reference metric inventory renders (ref
+    pkg/metrics/metrics.go:16-84)."""
+
+    def test_all_reference_families_present(self):
+        from bng_amd.metrics.metrics import Metrics
+        m = Metrics()
+        m.session_duration.observe(120)
+        m.session_bytes_in.labels("dhcp").inc(1000)
+        m.nat_translations.labels("egress").inc()
+        m.nat_ports_used.labels("203.0.113.1").set(512)
+        m.radius_latency.observe(0.01)
+        m.radius_timeouts.labels("10.0.0.9").inc()
+        m.pppoe_sessions.set(3)
+        m.pppoe_negotiations.labels("lcp", "ok").inc()
+        m.routes_active.labels("bgp").set(100)
+        m.bgp_peers_up.set(2)
+        m.subscriber_total.set(5000)
+        m.subscriber_by_isp.labels("isp-a").set(4000)
+        m.circuit_id_collisions.inc()
+        m.pool_available.labels("p1").set(200)
+        m.table_entries.labels("subscribers").set(65536)
+        text = m.render().decode()
+        for family in ("bng_session_duration_seconds",
+                       "bng_session_bytes_in_total",
+                       "bng_nat_translations_total",
+                       "bng_nat_ports_used",
+                       "bng_radius_request_duration_seconds",
+                       "bng_radius_timeouts_total",
+                       "bng_pppoe_sessions_active",
+                       "bng_pppoe_negotiations_total",
+                       "bng_routes_active", "bng_bgp_peers_up",
+                       "bng_subscribers_total",
+                       "bng_subscribers_by_isp",
+                       "bng_circuit_id_collisions_total",
+                       "bng_pool_available",
+                       "bng_dataplane_table_entries"):
+            assert family in text, family
+        assert 'bng_subscribers_by_isp{isp="isp-a"} 4000' in text
