@@ -301,3 +301,81 @@ def test_nat_ha_export_import_roundtrip(flows, eim_on):
         assert b.dp.nat44_ingress(fb) == A.FWD
         assert int.from_bytes(fb[30:34], "big") == want_ip
         assert int.from_bytes(fb[36:38], "big") == want_port
+
+
+class TestEtsiPduProperties:
+    """ETSI HI2/HI3 codec: build/decode roundtrip and stream framing
+    hold for arbitrary field values."""
+
+    @given(liid=st.text(alphabet=st.characters(
+               whitelist_categories=("Lu", "Ll", "Nd"),
+               whitelist_characters="-_"), min_size=1, max_size=32),
+           direction=st.sampled_from(["up", "down"]),
+           sport=st.integers(0, 65535), dport=st.integers(0, 65535),
+           proto=st.integers(0, 255),
+           payload=st.binary(max_size=512),
+           ts=st.floats(0, 4e9))
+    @settings(max_examples=150, deadline=None)
+    def test_hi3_roundtrip(self, liid, direction, sport, dport, proto,
+                           payload, ts):
+        from bng_amd.intercept.etsi import ETSIExporter, decode_pdu
+        ex = ETSIExporter()
+        pdu = ex.build_hi3(liid, direction, "10.0.1.5", "9.9.9.9",
+                           sport, dport, proto, payload, ts=ts)
+        d = decode_pdu(pdu)
+        assert d["liid"] == liid and d["direction"] == direction
+        assert d["src_port"] == sport and d["dst_port"] == dport
+        assert d["protocol"] == proto and d["payload"] == payload
+        assert abs(d["timestamp"] - ts) < 0.002
+
+    @given(st.lists(st.tuples(st.sampled_from(["hi2", "hi3"]),
+                              st.binary(max_size=64)),
+                    min_size=1, max_size=8))
+    @settings(max_examples=100, deadline=None)
+    def test_stream_framing(self, kinds):
+        from bng_amd.intercept.etsi import ETSIExporter, split_stream
+        ex = ETSIExporter()
+        frames = []
+        for kind, payload in kinds:
+            if kind == "hi2":
+                frames.append(ex.build_hi2("L1", "ev", "s", ts=1.0))
+            else:
+                frames.append(ex.build_hi3("L1", "up", "1.2.3.4",
+                                           "5.6.7.8", 1, 2, 6,
+                                           payload, ts=1.0))
+        assert split_stream(b"".join(frames)) == frames
+
+
+class TestAuditSeverityProperties:
+    @given(st.text(max_size=40))
+    @settings(max_examples=200, deadline=None)
+    def test_severity_and_category_total(self, action):
+        """Every action string maps to a valid severity and category."""
+        from bng_amd.audit import retention as rt
+        sev = rt.action_severity(action)
+        assert 0 <= sev < 8
+        assert rt.severity_name(sev) != "UNKNOWN"
+        assert isinstance(rt.action_category(action), str)
+
+    @given(st.integers(-5, 12))
+    @settings(max_examples=50, deadline=None)
+    def test_severity_name_never_raises(self, sev):
+        from bng_amd.audit.retention import severity_name
+        assert isinstance(severity_name(sev), str)
+
+
+class TestDnsCnameProperty:
+    @given(st.lists(st.text(alphabet="abcdefghijklmnopqrstuvwxyz0123456789-",
+                            min_size=1, max_size=20),
+                    min_size=1, max_size=5))
+    @settings(max_examples=150, deadline=None)
+    def test_cname_target_roundtrip(self, labels):
+        from bng_amd.dns.resolver import (build_cname_response,
+                                          build_query, decode_qname)
+        target = ".".join(labels)
+        q = build_query("www.example.com")
+        resp = build_cname_response(q, target)
+        _qn, off = decode_qname(resp, 12)
+        off += 4 + 2
+        got, _ = decode_qname(resp, off + 10)
+        assert got == target
