@@ -310,6 +310,7 @@ class BNG:
         self.launcher = None
         self.dhcp_server = None
         self.metrics = None
+        self.audit = None
         self.bgp = None
         self.sub_routes = None
 
@@ -726,6 +727,18 @@ class BNG:
                 self._defer(self.metrics.stop)
             except OSError:
                 self.log.warning("metrics port busy; collector only")
+
+        # 12b. audit trail: rotating JSON-lines file + the DHCP
+        # server's session events (severity/retention defaults apply)
+        if a.audit_log_path:
+            from ..audit.logger import FileExporter, Logger as AuditLog
+            from ..audit.retention import RetentionManager
+            self.audit = AuditLog(
+                exporters=[FileExporter(a.audit_log_path)],
+                retention=RetentionManager()).start()
+            self.dhcp_server.audit = self.audit
+            self._defer(self.audit.stop)
+            self.log.info("audit trail at %s", a.audit_log_path)
 
         # 13. DHCP serve loop last (main.go:1244)
         self.dhcp_server.start(serve=a.dhcp_listen)

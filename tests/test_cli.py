@@ -559,3 +559,28 @@ class TestCliHaFailover:
                 standby.stop()
         finally:
             active.stop()
+
+
+def test_audit_log_path_wires_trail(tmp_path):
+    """--audit-log-path creates the audit logger, hooks DHCP session
+    events, and writes JSON lines to the file."""
+    import json as _json
+    from bng_amd.cli.main import build_parser, BNG
+    path = tmp_path / "audit.jsonl"
+    args = build_parser().parse_args(
+        ["run", "--interface", "lo", "--pool-network", "10.9.0.0/24",
+         "--audit-log-path", str(path)])
+    bng = BNG(args).start()
+    try:
+        assert bng.audit is not None
+        assert bng.dhcp_server.audit is bng.audit
+        bng.audit.log("session_start", subscriber="sub-1",
+                      ip="10.9.0.5")
+        bng.audit.flush()
+        for ex in bng.audit.exporters:
+            ex._fh.flush()
+        rec = _json.loads(path.read_text().splitlines()[-1])
+        assert rec["action"] == "session_start"
+        assert rec["subscriber"] == "sub-1"
+    finally:
+        bng.stop()
