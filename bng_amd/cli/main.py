@@ -384,7 +384,12 @@ class BNG:
 
         # 3. walled garden (main.go:556)
         from ..walledgarden.manager import Manager as WGMgr
-        self.walledgarden = WGMgr(portal_ip=a.walled_garden_portal).start()
+        portal = a.walled_garden_portal or \
+            ("10.255.255.1:8080" if a.walled_garden else "")
+        p_ip, _, p_port = portal.partition(":")
+        self.walledgarden = WGMgr(
+            portal_ip=p_ip,
+            portal_port=int(p_port) if p_port else 8080).start()
         self._defer(self.walledgarden.stop)
 
         # 4. pools + DHCP server (main.go:567-594, :1244)
@@ -452,8 +457,14 @@ class BNG:
         if a.peer and a.pool_network:
             from ..pool.peer import PeerPool
             peers = dict(p.split("=", 1) for p in a.peer)
-            self.peer_pool = PeerPool(a.node_id, peers,
-                                      a.pool_network).start()
+            listen_port = 0
+            if a.peer_listen:
+                try:
+                    listen_port = int(a.peer_listen.rsplit(":", 1)[-1])
+                except ValueError:
+                    pass
+            self.peer_pool = PeerPool(a.node_id, peers, a.pool_network,
+                                      listen_port=listen_port).start()
             self._defer(self.peer_pool.stop)
             self.dhcp_server.set_peer_pool(self.peer_pool)
 
@@ -591,7 +602,8 @@ class BNG:
             except Exception:
                 return False
         self.resilience = ResilienceMgr(
-            _health, check_interval=a.health_check_interval).start()
+            _health, check_interval=a.health_check_interval,
+            failure_threshold=a.health_check_retries).start()
         self._defer(self.resilience.stop)
         if getattr(self, "radius_resilient", None) is not None:
             self.resilience.on_transition(
@@ -696,7 +708,9 @@ class BNG:
             from ..dhcpv6.server import DHCPv6Server
             self.dhcpv6 = DHCPv6Server(
                 na_pool=a.dhcpv6_na_pool, pd_pool=a.dhcpv6_pd_pool,
-                dns=[d for d in a.dhcpv6_dns.split(",") if d])
+                dns=[d for d in a.dhcpv6_dns.split(",") if d],
+                domains=[d for d in a.dhcpv6_domain_search.split(",")
+                         if d])
         if a.slaac_enable and (a.slaac_prefix or a.slaac_prefixes):
             from ..slaac.radvd import PrefixConfig, RAConfig, Server
             prefixes = ([p for p in a.slaac_prefixes.split(",") if p]

@@ -126,12 +126,14 @@ class DHCPv6Server:
                  pd_pool: str = "2001:db8:f000::/40",
                  pd_prefix_len: int = 56,
                  dns: Optional[List[str]] = None,
+                 domains: Optional[List[str]] = None,
                  preferred_lifetime: int = 1800, valid_lifetime: int = 3600,
                  rapid_commit: bool = True):
         self.server_duid = duid
         self.na_alloc = BitmapAllocator(na_pool, 128, reserve_head=1)
         self.pd_alloc = BitmapAllocator(pd_pool, pd_prefix_len)
         self.dns = dns or []
+        self.domains = domains or []      # RFC 8415 option 24
         self._declined: dict = {}   # addr -> quarantine expiry
         self.preferred = preferred_lifetime
         self.valid = valid_lifetime
@@ -176,6 +178,13 @@ class DHCPv6Server:
         if self.dns:
             r.add(OPT_DNS_SERVERS, b"".join(
                 ipaddress.IPv6Address(d).packed for d in self.dns))
+        if self.domains:
+            enc = b""
+            for name in self.domains:
+                for label in name.strip(".").split("."):
+                    enc += bytes([len(label)]) + label.encode()
+                enc += b"\x00"
+            r.add(OPT_DOMAIN_LIST, enc)
         return r
 
     def _fill_ias(self, msg: DHCPv6Message, resp: DHCPv6Message,

@@ -584,3 +584,40 @@ def test_audit_log_path_wires_trail(tmp_path):
         assert rec["subscriber"] == "sub-1"
     finally:
         bng.stop()
+
+
+def test_flag_wiring_reaches_subsystems(tmp_path):
+    """Flags that used to parse-only now reach their subsystems:
+    walled-garden portal host:port, health-check retries, DHCPv6
+    domain search."""
+    from bng_amd.cli.main import build_parser, BNG
+    args = build_parser().parse_args(
+        ["run", "--interface", "lo", "--pool-network", "10.9.0.0/24",
+         "--walled-garden", "--walled-garden-portal", "10.1.2.3:9999",
+         "--health-check-retries", "7",
+         "--dhcpv6-enable", "--dhcpv6-domain-search", "isp.net,lab"])
+    bng = BNG(args).start()
+    try:
+        assert bng.walledgarden.portal_ip == "10.1.2.3"
+        assert bng.walledgarden.portal_port == 9999
+        assert ("10.1.2.3", 9999, 6) in bng.walledgarden.allowed_dests
+        assert bng.resilience.failure_threshold == 7
+        assert bng.dhcpv6.domains == ["isp.net", "lab"]
+    finally:
+        bng.stop()
+
+
+def test_dhcpv6_domain_search_option_on_wire():
+    """Option 24 carries the encoded search list in ADVERTISE."""
+    from bng_amd.dhcpv6.server import (DHCPv6Message, DHCPv6Server,
+                                       OPT_CLIENTID, OPT_DOMAIN_LIST,
+                                       OPT_IA_NA, SOLICIT)
+    import struct as _st
+    srv = DHCPv6Server(domains=["isp.net"], rapid_commit=False)
+    sol = DHCPv6Message(SOLICIT, 0x123456)
+    sol.add(OPT_CLIENTID, b"\x00\x01duid-x")
+    sol.add(OPT_IA_NA, _st.pack(">III", 1, 0, 0))
+    resp = srv.handle(sol.encode())
+    msg = DHCPv6Message.decode(resp)
+    enc = msg.get(OPT_DOMAIN_LIST)
+    assert enc == b"\x03isp\x03net\x00"
