@@ -468,6 +468,25 @@ class BNG:
             self._defer(self.peer_pool.stop)
             self.dhcp_server.set_peer_pool(self.peer_pool)
 
+        # 6b. store-replicated allocator modes (ref modes.go:14-92):
+        # pool-mode session|lease runs a DistributedAllocator over the
+        # embedded CLSet (or a process-local store without one) with
+        # the configured epoch period/grace
+        if a.pool_mode != "static" and a.pool_network:
+            from ..allocator.distributed import DistributedAllocator
+            from ..nexus.store import MemoryStore
+            store = getattr(self, "clset", None) or MemoryStore()
+            self.distributed_alloc = DistributedAllocator(
+                store, a.nexus_pool, a.pool_network, mode=a.pool_mode,
+                grace_period=a.epoch_grace,
+                epoch_interval=(float(a.epoch_period)
+                                if a.pool_mode == "lease" else 0.0),
+                node_id=a.node_id)
+            self._defer(self.distributed_alloc.close)
+            self.dhcp_server.set_distributed(self.distributed_alloc)
+            self.log.info("pool-mode %s (epoch %ss, grace %d)",
+                          a.pool_mode, a.epoch_period, a.epoch_grace)
+
         # 7. HA (main.go:826-881)
         if a.ha_role:
             from ..ha.failover import FailoverController
@@ -521,9 +540,14 @@ class BNG:
                 else FakeExecutor()
             self.bgp = BGPController(exe, a.bgp_local_as,
                                      router_id=a.bgp_router_id).start()
+            if a.bgp_bfd_enabled:
+                from ..routing.bgp import BFDManager
+                self.bfd = BFDManager(exe)
             for n in a.bgp_neighbor:
                 addr, _, ras = n.partition(":")
                 self.bgp.add_neighbor(addr, int(ras or 65000))
+                if a.bgp_bfd_enabled:
+                    self.bfd.add_peer(addr)
             if a.bgp_announce_subscribers:
                 from ..dataplane.packets import u32_to_ip
                 from ..routing.manager import (SessionRouteIntegration,

@@ -64,6 +64,7 @@ class DHCPServer:
         self.nexus_allocator = None   # nexus.HTTPAllocator
         self.nexus_client = None      # nexus.Client
         self.peer_pool = None         # pool.PeerPool
+        self.distributed = None       # allocator.DistributedAllocator
         self.walled_garden = None     # walledgarden.Manager
         self.accounting = None        # radius.AccountingManager
         self.policy_mgr = None        # radius.PolicyManager
@@ -105,6 +106,12 @@ class DHCPServer:
 
     def set_peer_pool(self, p):
         self.peer_pool = p
+
+    def set_distributed(self, alloc):
+        """Store-replicated allocator (ref modes.go framework): slots
+        into the chain after the peer pool; pool-mode session/lease
+        semantics live in the allocator itself."""
+        self.distributed = alloc
 
     def set_walled_garden(self, w):
         self.walled_garden = w
@@ -323,6 +330,14 @@ class DHCPServer:
                 ip = ip2u32(self.peer_pool.allocate(sub_id))
             except Exception:
                 pass
+        # 3b. store-replicated distributed allocator (pool-mode
+        # session|lease)
+        if not ip and self.distributed is not None:
+            try:
+                ip = ip2u32(
+                    self.distributed.allocate(sub_id).split("/")[0])
+            except Exception:
+                pass
         # 4. local pool
         if not ip:
             pool = self.pools.classify_client(mac, req.vendor_class)
@@ -400,6 +415,11 @@ class DHCPServer:
         if self.peer_pool is not None:
             try:
                 self.peer_pool.release(lease.subscriber_id)
+            except Exception:
+                pass
+        if self.distributed is not None:
+            try:
+                self.distributed.release(lease.subscriber_id)
             except Exception:
                 pass
         if self.qos_mgr is not None:

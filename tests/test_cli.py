@@ -621,3 +621,42 @@ def test_dhcpv6_domain_search_option_on_wire():
     msg = DHCPv6Message.decode(resp)
     enc = msg.get(OPT_DOMAIN_LIST)
     assert enc == b"\x03isp\x03net\x00"
+
+
+def test_bgp_bfd_enabled_wires_peers():
+    """--bgp-bfd-enabled creates the BFD manager and registers a BFD
+    peer per BGP neighbor (ref bfd.go peer-per-neighbor wiring)."""
+    from bng_amd.cli.main import build_parser, BNG
+    args = build_parser().parse_args(
+        ["run", "--interface", "lo", "--pool-network", "10.9.0.0/24",
+         "--bgp-enable", "--bgp-local-as", "65001",
+         "--bgp-neighbor", "192.0.2.1:65002", "--bgp-bfd-enabled"])
+    bng = BNG(args).start()
+    try:
+        assert "192.0.2.1" in bng.bfd.peers
+        assert bng.bfd.peers["192.0.2.1"].interval_ms == 50
+    finally:
+        bng.stop()
+
+
+def test_pool_mode_lease_wires_distributed_allocator():
+    """--pool-mode lease runs a store-replicated allocator in the DHCP
+    chain with the configured epoch grace; the same subscriber gets a
+    stable address and releases propagate."""
+    from bng_amd.cli.main import build_parser, BNG
+    args = build_parser().parse_args(
+        ["run", "--interface", "lo", "--pool-network", "10.9.0.0/24",
+         "--pool-mode", "lease", "--epoch-period", "300",
+         "--epoch-grace", "2"])
+    bng = BNG(args).start()
+    try:
+        d = bng.distributed_alloc
+        assert d.mode == "lease"
+        assert bng.dhcp_server.distributed is d
+        ip1 = d.allocate("sub-1").split("/")[0]
+        assert d.allocate("sub-1").split("/")[0] == ip1   # sticky
+        assert d.lookup("sub-1") is not None
+        d.release("sub-1")
+        assert d.lookup("sub-1") is None
+    finally:
+        bng.stop()
