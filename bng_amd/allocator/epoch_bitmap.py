@@ -39,18 +39,20 @@ class EpochBitmapAllocator:
         self.generations = bytearray((self.total + 3) // 4)  # 2 bits each
         self.subscribers: Dict[str, int] = {}
         self.ip_to_sub: Dict[int, str] = {}
-        # start at epoch 2 so generation 0 reads as expired (ref :92)
+        # generation 0 is reserved as "never allocated"; live slots
+        # cycle through gens 1..3, so at most 2 epochs of grace fit in
+        # the 2-bit field (current + two previous) (ref :92)
         self.current_epoch = 2
-        self.grace_period = max(1, grace_period)
+        self.grace_period = min(2, max(1, grace_period))
         self.next_free_hint = 1
         self._lock = threading.RLock()
 
     # ------------------------------------------------------- generations
     def _cur_gen(self) -> int:
-        return self.current_epoch % 4
+        return (self.current_epoch % 3) + 1
 
     def _active_gens(self) -> set:
-        return {(self.current_epoch - k) % 4
+        return {((self.current_epoch - k) % 3) + 1
                 for k in range(self.grace_period + 1)}
 
     def _get_gen(self, idx: int) -> int:
@@ -108,7 +110,7 @@ class EpochBitmapAllocator:
             idx = self.subscribers.pop(subscriber_id, None)
             if idx is None:
                 return
-            self._set_gen(idx, (self._cur_gen() + 2) % 4)  # oldest gen
+            self._set_gen(idx, 0)                 # back to never-used
             self.ip_to_sub.pop(idx, None)
             if idx < self.next_free_hint:
                 self.next_free_hint = idx

@@ -60,9 +60,10 @@ class TestEpochBitmap:
         assert a.lookup("s1") == ip
         a.advance_epoch()                # expired
         assert a.lookup("s1") is None
-        # slot is reallocatable
-        ip2 = a.allocate("s2")
-        assert ip2 == ip
+        # the expired slot is reallocatable: filling the whole pool
+        # succeeds only if s1's old slot is reclaimed
+        ips = {a.allocate(f"s{i}") for i in range(2, 16)}
+        assert len(ips) == 14 and ip in ips
 
     def test_renew_extends(self):
         a = EpochBitmapAllocator("10.0.0.0/28", 32)
