@@ -808,8 +808,8 @@ class BNG:
                              batch=a.pktio_batch,
                              max_wait=a.pktio_max_wait,
                              dhcp_service=svc).start()
-            self._defer(self.pump.stop)
             self._defer(io.close)
+            self._defer(self.pump.stop)
             # Optional second pump on the core-side NIC running the
             # downlink pipeline (ref tc_egress on the NAT outside
             # interface, tc.c / loader.go) — DNAT'd return traffic is
@@ -827,8 +827,23 @@ class BNG:
                     self.launcher, dio, io, direction="downlink",
                     batch=a.pktio_batch,
                     max_wait=a.pktio_max_wait).start()
-                self._defer(self.pump_downlink.stop)
                 self._defer(dio.close)
+                self._defer(self.pump_downlink.stop)
+            # PPPoE on its own access NIC (ref --pppoe-interface,
+            # main.go pppoe raw socket on a dedicated interface):
+            # a second uplink pump sharing the launcher + slow path
+            pe_if = a.pppoe_interface
+            if pe_if and pe_if != a.interface and a.pppoe_enable:
+                pio = AFPacketIO(pe_if)
+                self.log.info("PPPoE pump on %s", pe_if)
+                self.pktio_pppoe = pio
+                self.pump_pppoe = Pump(
+                    self.launcher, pio, pio,
+                    slow_path=self._frame_slow_path,
+                    batch=a.pktio_batch,
+                    max_wait=a.pktio_max_wait).start()
+                self._defer(pio.close)
+                self._defer(self.pump_pppoe.stop)
         return self
 
     def _frame_slow_path(self, frame: bytes):

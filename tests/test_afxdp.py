@@ -236,3 +236,61 @@ class TestCliPktio:
             tx.close()
         finally:
             bng.stop()
+
+
+class TestPPPoEInterfacePump:
+    def test_dedicated_pppoe_interface(self, veth):
+        """--pppoe-interface on a second NIC attaches its own pump: a
+        PADI on that wire gets a PADO back while the main pump owns
+        --interface (ref main.go pppoe raw socket on a dedicated
+        interface)."""
+        from bng_amd.dataplane import afxdp
+        from bng_amd.cli.main import BNG, build_parser
+        P0, P1 = "bngp0", "bngp1"
+        try:
+            try:
+                afxdp.link_del(P0)
+            except OSError:
+                pass
+            afxdp.veth_create(P0, P1)
+            afxdp.link_up(P0)
+            afxdp.link_up(P1)
+        except OSError as e:
+            pytest.skip(f"cannot create second veth: {e}")
+        try:
+            args = build_parser().parse_args([
+                "run", "--gpu", "off", "--pktio", "afpacket",
+                "--interface", VETH1,
+                "--pppoe-enable", "--pppoe-interface", P1,
+                "--pktio-batch", "64", "--pktio-max-wait", "0.02",
+                "--pool-network", "10.0.3.0/24"])
+            bng = BNG(args).start()
+            try:
+                assert bng.pump_pppoe is not None
+                tx = _raw_sock(P0)
+                tx.settimeout(0.2)
+                # PADI: dst broadcast, ethertype 0x8863, code 0x09
+                src = bytes.fromhex("aabbcc000077")
+                padi = (b"\xff" * 6 + src +
+                        b"\x88\x63" + b"\x11\x09\x00\x00\x00\x04" +
+                        b"\x01\x01\x00\x00")
+                got_pado = False
+                end = time.monotonic() + 5.0
+                while not got_pado and time.monotonic() < end:
+                    tx.send(padi)
+                    try:
+                        f = tx.recv(2048)
+                    except socket.timeout:
+                        continue
+                    if len(f) >= 20 and f[12:14] == b"\x88\x63" and \
+                            f[15] == 0x07:          # PADO
+                        got_pado = True
+                assert got_pado, "no PADO on the dedicated PPPoE NIC"
+                tx.close()
+            finally:
+                bng.stop()
+        finally:
+            try:
+                afxdp.link_del(P0)
+            except OSError:
+                pass
