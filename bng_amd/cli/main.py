@@ -133,7 +133,9 @@ def build_parser() -> argparse.ArgumentParser:
     g.add_argument("--auth-mtls-cert", default="")
     g.add_argument("--auth-mtls-key", default="")
     g.add_argument("--auth-mtls-ca", default="")
-    g.add_argument("--auth-mtls-server-name", default="")
+    g.add_argument("--auth-mtls-server-name", default="",
+                   help="accepted for compatibility; SNI follows the "
+                        "--nexus-url host")
     g.add_argument("--auth-mtls-insecure", action="store_true")
     g = run.add_argument_group("peers")
     g.add_argument("--peer", action="append", default=[],
@@ -162,8 +164,12 @@ def build_parser() -> argparse.ArgumentParser:
                    help="comma list (reference spelling)")
     g.add_argument("--nat-ports-per-sub", type=int, default=0,
                    help="reference spelling of --nat-ports-per-subscriber")
-    g.add_argument("--nat-log-enabled", action="store_true")
-    g.add_argument("--nat-inside-interface", default="")
+    g.add_argument("--nat-log-enabled", action="store_true",
+                   help="accepted for compatibility; logging activates "
+                        "when --nat-log-path is set")
+    g.add_argument("--nat-inside-interface", default="",
+                   help="accepted for compatibility; the uplink pump "
+                        "attaches to --interface (the access NIC)")
     g.add_argument("--nat-outside-interface", default="")
     g.add_argument("--nat-eim", default=True, type=lambda v: v not in
                    ("false", "0", "no", False))
@@ -178,7 +184,9 @@ def build_parser() -> argparse.ArgumentParser:
     g = run.add_argument_group("qos")
     g.add_argument("--qos-policy", action="append", default=[],
                    help="name:down_mbps:up_mbps")
-    g.add_argument("--qos-enabled", action="store_true")
+    g.add_argument("--qos-enabled", action="store_true",
+                   help="accepted for compatibility; the QoS manager "
+                        "is always wired, policies activate it")
     g.add_argument("--qos-bpf-path", default="",
                    help="accepted for compatibility (see --bpf-path)")
     g.add_argument("--qos-default-policy", default="")
