@@ -218,3 +218,28 @@ class Manager:
                 by_state[e.state] = by_state.get(e.state, 0) + 1
             return {**self.stats, "tracked": len(self.entries),
                     **{f"state_{k}": v for k, v in by_state.items()}}
+
+
+def attach_dns(manager: Manager, resolver, portal_ips: List[str]):
+    """Keep the DNS resolver's walled-client registry in sync with the
+    garden (the reference couples pkg/dns WalledGardenClients to the
+    walled-garden state the same way): entries in STATE_WALLED resolve
+    everything to the portal; activation/removal/expiry releases them.
+    Returns the hook (already registered) for tests."""
+    resolver.set_intercept_all(portal_ips)
+
+    def hook(entry: Entry):
+        tracked = manager.state_of(entry.mac) is not None
+        if tracked and entry.state == STATE_WALLED:
+            if entry.ip:
+                resolver.add_walled_client(entry.ip)
+        elif entry.ip:
+            resolver.remove_walled_client(entry.ip)
+
+    manager.set_dataplane_hooks(hook)
+    # adopt anything already quarantined
+    with manager._lock:
+        for e in manager.entries.values():
+            if e.state == STATE_WALLED and e.ip:
+                resolver.add_walled_client(e.ip)
+    return hook

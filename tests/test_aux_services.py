@@ -1227,3 +1227,41 @@ class TestDNSInterceptRules:
         r.handle_query(build_query("other.example"),
                        client="10.0.1.5")
         assert len(log) == 2
+
+
+class TestWalledGardenDnsGlue:
+    def test_quarantine_drives_dns_interception(self):
+        """Garden state changes flow into the resolver's walled-client
+        registry: quarantined IPs resolve everything to the portal,
+        activation restores normal resolution."""
+        from bng_amd.walledgarden.manager import Manager as WG, attach_dns
+        upstream_hits = []
+
+        def upstream(q):
+            upstream_hits.append(q)
+            return build_response(q, ["93.184.216.34"])
+        r = Resolver(upstream)
+        wg = WG(portal_ip="10.255.255.1", dns_servers=["8.8.8.8"])
+        attach_dns(wg, r, ["10.255.255.1"])
+        wg.add("aa:00:00:00:00:01", "10.0.1.5")
+        assert r.is_walled("10.0.1.5")
+        _, addrs, _ = parse_response(
+            r.handle_query(build_query("x.example"), client="10.0.1.5"))
+        assert addrs == ["10.255.255.1"]
+        assert upstream_hits == []
+        wg.activate("aa:00:00:00:00:01")
+        assert not r.is_walled("10.0.1.5")
+        r.handle_query(build_query("x.example"), client="10.0.1.5")
+        assert len(upstream_hits) == 1
+        # expiry also releases
+        import time as _t
+        wg.add("aa:00:00:00:00:02", "10.0.1.6", ttl=5.0)
+        assert r.is_walled("10.0.1.6")
+        wg.expire_stale(now=_t.time() + 10)
+        assert not r.is_walled("10.0.1.6")
+        # pre-existing quarantine adopted at attach time
+        wg2 = WG(portal_ip="10.255.255.1")
+        wg2.add("aa:00:00:00:00:03", "10.0.1.7")
+        r2 = Resolver(upstream)
+        attach_dns(wg2, r2, ["10.255.255.1"])
+        assert r2.is_walled("10.0.1.7")
