@@ -934,6 +934,24 @@ class BNG:
                "fastpath": self.launcher.get_stats()}
         if getattr(self, "nat", None):
             out["nat"] = self.nat.get_stats()
+        if getattr(self, "walledgarden", None):
+            out["walledgarden"] = self.walledgarden.get_stats()
+        if getattr(self, "resilience", None):
+            out["resilience"] = {
+                "state": self.resilience.state,
+                "partition_duration":
+                    self.resilience.partition_duration()}
+        if getattr(self, "audit", None):
+            out["audit"] = self.audit.stats()
+        if getattr(self, "pump", None):
+            out["pktio"] = dict(self.pump.stats)
+            if getattr(self, "pump_downlink", None):
+                out["pktio_downlink"] = dict(self.pump_downlink.stats)
+        if getattr(self, "distributed_alloc", None):
+            n, total, util = self.distributed_alloc.local.stats()
+            out["pool_mode"] = {"mode": self.distributed_alloc.mode,
+                                "allocated": n, "usable": total,
+                                "utilization": util}
         return out
 
 

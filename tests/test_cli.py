@@ -660,3 +660,25 @@ def test_pool_mode_lease_wires_distributed_allocator():
         assert d.lookup("sub-1") is None
     finally:
         bng.stop()
+
+
+def test_stats_covers_new_subsystems(tmp_path):
+    """bng stats surfaces walled garden, resilience, audit, and
+    pool-mode alongside the dataplane counters."""
+    from bng_amd.cli.main import build_parser, BNG
+    args = build_parser().parse_args(
+        ["run", "--interface", "lo", "--pool-network", "10.9.0.0/24",
+         "--walled-garden", "--pool-mode", "session",
+         "--audit-log-path", str(tmp_path / "a.jsonl")])
+    bng = BNG(args).start()
+    try:
+        bng.distributed_alloc.allocate("sub-1")
+        st = bng.stats()
+        assert "walledgarden" in st and "tracked" in st["walledgarden"]
+        assert st["resilience"]["state"] in ("online", "partitioned",
+                                             "recovering")
+        assert "logged" in st["audit"]
+        assert st["pool_mode"]["mode"] == "session"
+        assert st["pool_mode"]["allocated"] == 1
+    finally:
+        bng.stop()
