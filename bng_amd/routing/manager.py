@@ -22,6 +22,7 @@ class Route:
     next_hop: str = ""
     table: int = 254
     metric: int = 0
+    device: str = ""                  # output interface (RTA_OIF)
 
 
 @dataclass
@@ -29,6 +30,7 @@ class Rule:
     src: str
     table: int
     priority: int = 1000
+    fwmark: int = 0                   # FRA_FWMARK match
 
 
 class RoutingPlatform:
@@ -93,15 +95,20 @@ class Manager:
         self._next_table = self.ISP_TABLE_BASE
         self._lock = threading.RLock()
 
-    def create_isp_table(self, isp_id: str, default_next_hop: str) -> int:
-        """ref manager.go:521 CreateISPTable."""
+    def create_isp_table(self, isp_id: str, default_next_hop: str,
+                         device: str = "") -> int:
+        """ref manager.go:521 CreateISPTable; the default route needs
+        a gateway or an output device (a gatewayless, deviceless
+        default is invalid on a real kernel)."""
         with self._lock:
             if isp_id in self.isp_tables:
                 return self.isp_tables[isp_id]
             table = self._next_table
             self._next_table += 1
             self.isp_tables[isp_id] = table
-        self.platform.add_route(Route("0.0.0.0/0", default_next_hop, table))
+        if default_next_hop or device:
+            self.platform.add_route(Route("0.0.0.0/0", default_next_hop,
+                                          table, device=device))
         return table
 
     def remove_isp_table(self, isp_id: str):

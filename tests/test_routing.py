@@ -1,6 +1,8 @@
 """Routing tests: BGP/BFD over a fake FRR executor, policy routing over
 the memory platform, subscriber-route retry/reconcile, health hysteresis
 (ref pkg/routing bgp_test.go / bfd_test.go patterns)."""
+import os
+
 import pytest
 
 from bng_amd.routing.bgp import BFDManager, BGPController
@@ -222,3 +224,99 @@ class TestRoutingMetrics:
         assert "bng_routing_bfd_peers_up 1.0" in text
         assert "bng_routing_subscriber_routes_injected_total 1.0" in text
         assert "bng_routing_route_injection_errors_total 1.0" in text
+
+
+@pytest.mark.skipif(os.geteuid() != 0, reason="netlink needs root")
+class TestNetlinkPlatform:
+    """Raw-rtnetlink platform against the live kernel (ref
+    netlink_linux.go:20-235); runs where CAP_NET_ADMIN exists (build
+    container), skips on the GPU pool."""
+
+    IFACE0, IFACE1 = "bngr0", "bngr1"
+    TABLE = 177
+
+    @pytest.fixture()
+    def dev(self):
+        from bng_amd.dataplane import afxdp
+        try:
+            try:
+                afxdp.link_del(self.IFACE0)
+            except OSError:
+                pass
+            afxdp.veth_create(self.IFACE0, self.IFACE1)
+            afxdp.link_up(self.IFACE0)
+            afxdp.link_up(self.IFACE1)
+        except OSError as e:
+            pytest.skip(f"no CAP_NET_ADMIN: {e}")
+        yield self.IFACE0
+        from bng_amd.routing.netlink import NetlinkPlatform
+        try:
+            NetlinkPlatform().flush_table(self.TABLE)
+        except OSError:
+            pass
+        try:
+            afxdp.link_del(self.IFACE0)
+        except OSError:
+            pass
+
+    def test_route_add_dump_delete(self, dev):
+        from bng_amd.routing.manager import Route
+        from bng_amd.routing.netlink import NetlinkPlatform
+        p = NetlinkPlatform()
+        r = Route(prefix="203.0.113.0/24", table=self.TABLE,
+                  device=dev, metric=50)
+        p.add_route(r)
+        got = p.routes(self.TABLE)
+        assert any(x.prefix == "203.0.113.0/24" and x.device == dev and
+                   x.metric == 50 for x in got), got
+        # main table not polluted
+        assert not any(x.prefix == "203.0.113.0/24"
+                       for x in p.routes(254))
+        p.del_route(r)
+        assert not any(x.prefix == "203.0.113.0/24"
+                       for x in p.routes(self.TABLE))
+
+    def test_flush_table(self, dev):
+        from bng_amd.routing.manager import Route
+        from bng_amd.routing.netlink import NetlinkPlatform
+        p = NetlinkPlatform()
+        for i in range(3):
+            p.add_route(Route(prefix=f"198.51.{100 + i}.0/24",
+                              table=self.TABLE, device=dev))
+        assert p.flush_table(self.TABLE) == 3
+        assert p.routes(self.TABLE) == []
+
+    def test_policy_rules(self, dev):
+        from bng_amd.routing.manager import Rule
+        from bng_amd.routing.netlink import NetlinkPlatform
+        p = NetlinkPlatform()
+        r = Rule(src="10.77.0.0/16", table=self.TABLE, priority=1177)
+        p.add_rule(r)
+        try:
+            got = p.rules()
+            assert any(x.src == "10.77.0.0/16" and
+                       x.table == self.TABLE and x.priority == 1177
+                       for x in got), got
+        finally:
+            p.del_rule(r)
+        assert not any(x.priority == 1177 for x in p.rules())
+
+    def test_manager_on_live_kernel(self, dev):
+        """The per-ISP table manager drives real kernel state."""
+        from bng_amd.routing.manager import Manager, Route
+        from bng_amd.routing.netlink import NetlinkPlatform
+        p = NetlinkPlatform()
+        m = Manager(platform=p)
+        table = m.create_isp_table("isp-x", default_next_hop="",
+                                   device=dev)
+        try:
+            p.add_route(Route(prefix="192.0.2.0/24", table=table,
+                              device=dev))
+            m.add_subscriber_rule("10.88.0.5", "isp-x")
+            assert any(x.table == table for x in p.rules())
+            m.remove_subscriber_rule("10.88.0.5", "isp-x")
+            assert not any(x.src.startswith("10.88.0.5")
+                           for x in p.rules())
+        finally:
+            p.flush_table(table)
+            m.remove_isp_table("isp-x")
