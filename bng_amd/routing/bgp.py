@@ -94,6 +94,53 @@ class BGPController:
         with self._lock:
             return sorted(self.announced)
 
+    def announce_prefix_with_options(self, prefix: str,
+                                     next_hop: str = "",
+                                     community: str = "",
+                                     local_pref: int = 0,
+                                     med: int = 0) -> dict:
+        """Announcement with path attributes via a per-prefix
+        route-map (ref AnnouncePrefixWithOptions bgp.go:328-361 —
+        the reference caches the options; applying them in FRR needs
+        a route-map, which we emit too)."""
+        rm = f"BNG-{prefix.replace('/', '-').replace('.', '-')}"
+        lines = []
+        if community or local_pref or med:
+            lines += [f"route-map {rm} permit 10"]
+            if community:
+                lines += [f"set community {community}"]
+            if local_pref:
+                lines += [f"set local-preference {local_pref}"]
+            if med:
+                lines += [f"set metric {med}"]
+            lines += ["exit"]
+        self.exe.run(["configure terminal"] + lines + [
+            f"router bgp {self.local_as}",
+            "address-family ipv4 unicast",
+            f"network {prefix}" + (f" route-map {rm}"
+                                   if lines else ""),
+            "exit-address-family", "end"])
+        ann = {"prefix": prefix, "next_hop": next_hop,
+               "community": community, "local_pref": local_pref,
+               "med": med}
+        with self._lock:
+            self.announced[prefix] = ann
+        return ann
+
+    def enable_max_paths(self, max_paths: int):
+        """ECMP across equal BGP paths (ref EnableMaxPaths
+        bgp.go:431-448)."""
+        if not (1 <= max_paths <= 128):
+            raise ValueError("max_paths out of range")
+        self._cfg([f"maximum-paths {max_paths}"])
+
+    def configure_bfd_for_neighbor(self, address: str):
+        """Tie the neighbor to BFD liveness (ref ConfigureBFD
+        bgp.go:451-470)."""
+        self.exe.run(["configure terminal",
+                      f"router bgp {self.local_as}",
+                      f"neighbor {address} bfd", "end"])
+
 
 @dataclass
 class BFDPeer:

@@ -320,3 +320,39 @@ class TestNetlinkPlatform:
         finally:
             p.flush_table(table)
             m.remove_isp_table("isp-x")
+
+
+class TestBGPAnnouncementOptions:
+    """Path attributes, ECMP, neighbor BFD (ref bgp.go:328-470)."""
+
+    def test_announce_with_options_emits_route_map(self):
+        exe = FakeExecutor()
+        b = BGPController(exe, 65001).start()
+        ann = b.announce_prefix_with_options(
+            "203.0.113.0/24", community="65001:100",
+            local_pref=200, med=50)
+        joined = "\n".join(c for b_ in exe.batches for c in b_)
+        assert "set community 65001:100" in joined
+        assert "set local-preference 200" in joined
+        assert "set metric 50" in joined
+        assert "network 203.0.113.0/24 route-map" in joined
+        assert b.announced["203.0.113.0/24"]["local_pref"] == 200
+        # plain announcement: no route-map emitted
+        exe2 = FakeExecutor()
+        b2 = BGPController(exe2, 65001).start()
+        b2.announce_prefix_with_options("198.51.100.0/24")
+        j2 = "\n".join(c for b_ in exe2.batches for c in b_)
+        assert "route-map" not in j2.split("network")[-1]
+
+    def test_max_paths_and_neighbor_bfd(self):
+        exe = FakeExecutor()
+        b = BGPController(exe, 65001).start()
+        b.enable_max_paths(8)
+        assert any("maximum-paths 8" in c
+                   for b_ in exe.batches for c in b_)
+        with pytest.raises(ValueError):
+            b.enable_max_paths(0)
+        b.add_neighbor("192.0.2.1", 65002)
+        b.configure_bfd_for_neighbor("192.0.2.1")
+        assert any("neighbor 192.0.2.1 bfd" in c
+                   for b_ in exe.batches for c in b_)
