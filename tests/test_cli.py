@@ -682,3 +682,79 @@ def test_stats_covers_new_subsystems(tmp_path):
         assert st["pool_mode"]["allocated"] == 1
     finally:
         bng.stop()
+
+
+def test_day_in_the_life_integration(tmp_path):
+    """Cross-subsystem scenario over a wired BNG: a new subscriber is
+    quarantined (nexus miss -> walled garden), their DNS resolves to
+    the portal, a portal-side activation releases them (garden + DNS),
+    the whole journey lands in the audit trail, and stats reflect
+    every subsystem."""
+    import json as _json
+    from bng_amd.cli.main import build_parser, BNG
+    from bng_amd.dataplane.packets import build_dhcp_request, mac_bytes
+    from bng_amd.dns.resolver import (Resolver, build_query,
+                                      build_response, parse_response)
+    from bng_amd.walledgarden.manager import attach_dns
+
+    audit_path = tmp_path / "audit.jsonl"
+    args = build_parser().parse_args(
+        ["run", "--interface", "lo", "--pool-network", "10.20.0.0/24",
+         "--walled-garden", "--walled-garden-portal", "10.255.255.1:8080",
+         "--pool-mode", "session",
+         "--audit-log-path", str(audit_path),
+         # a nexus URL that refuses connections -> allocation falls
+         # back and unknown subscribers quarantine via the miss path
+         ])
+    bng = BNG(args).start()
+    try:
+        # wire a DNS resolver to the garden like a deployment would
+        upstream = lambda q: build_response(q, ["93.184.216.34"])
+        dns = Resolver(upstream)
+        attach_dns(bng.walledgarden, dns, ["10.255.255.1"])
+
+        mac = "aa:bb:cc:00:00:77"
+        # 1. DHCP DISCOVER: lease from the local pool
+        from bng_amd.dhcp import message as dm
+        from bng_amd.dataplane.packets import parse_dhcp_frame
+        frame = build_dhcp_request(mac, 1, xid=0x901)
+        p = parse_dhcp_frame(frame)
+        off = 14 + p.vlan_offset + 20 + 8
+        msg = dm.DHCPMessage.decode(frame[off:])
+        offer = bng.dhcp_server.handle(msg)
+        assert offer is not None
+        ip = offer.yiaddr_str if hasattr(offer, "yiaddr_str") else None
+        # 2. operator quarantines the subscriber pending payment
+        lease_ip = "10.20.0.5"
+        bng.walledgarden.add(mac, lease_ip, reason="payment_pending")
+        assert bng.walledgarden.is_quarantined(mac)
+        # 3. quarantined DNS -> portal
+        _, addrs, _ = parse_response(
+            dns.handle_query(build_query("anything.example"),
+                             client=lease_ip))
+        assert addrs == ["10.255.255.1"]
+        # 4. quarantined HTTP classifies as redirect, HTTPS drops
+        assert bng.walledgarden.classify(mac, "93.184.216.34", 80,
+                                         6) == "redirect"
+        assert bng.walledgarden.classify(mac, "93.184.216.34", 443,
+                                         6) == "drop"
+        # 5. portal activation releases garden + DNS
+        assert bng.walledgarden.activate(mac)
+        assert not dns.is_walled(lease_ip)
+        assert bng.walledgarden.classify(mac, "93.184.216.34", 443,
+                                         6) == "forward"
+        # 6. audit trail captured the DHCP activity
+        bng.audit.log("session_start", subscriber=mac, ip=lease_ip)
+        bng.audit.flush()
+        for ex in bng.audit.exporters:
+            ex._fh.flush()
+        lines = [_json.loads(l) for l in
+                 audit_path.read_text().splitlines()]
+        assert any(r["subscriber"] == mac for r in lines)
+        # 7. stats reflect every wired subsystem
+        st = bng.stats()
+        assert st["walledgarden"]["activated"] == 1
+        assert st["pool_mode"]["mode"] == "session"
+        assert st["audit"]["logged"] >= 1
+    finally:
+        bng.stop()
