@@ -366,3 +366,40 @@ class TestReplyAddressing:
         from bng_amd.dhcp.server import DHCPServer
         d = DHCPServer.reply_dest(self._req(), self._resp(0))
         assert d == ("255.255.255.255", 68)
+
+
+class TestRequestEdgeBehaviors:
+    """Reference coverage-test scenarios (server_coverage_test.go):
+    ciaddr-only renewals and option-82 preservation."""
+
+    def test_request_without_requested_ip_uses_ciaddr(self):
+        srv = make_server()
+        mac = mac_bytes("aa:bb:cc:00:00:14")
+        offer = srv.handle(dm.build_request(mac, dm.DISCOVER, xid=1))
+        ip = offer.yiaddr
+        # RENEWING state: no option 50, ciaddr carries the address
+        ack = srv.handle(dm.build_request(mac, dm.REQUEST, xid=2,
+                                          ciaddr=ip))
+        assert ack is not None and ack.msg_type == dm.ACK
+        assert ack.yiaddr == ip
+        # a ciaddr that disagrees with the lease still NAKs
+        nak = srv.handle(dm.build_request(mac, dm.REQUEST, xid=3,
+                                          ciaddr=ip + 1))
+        assert nak.msg_type == dm.NAK
+
+    def test_renewal_without_option82_preserves_it(self):
+        srv = make_server()
+        mac = mac_bytes("aa:bb:cc:00:00:16")
+        srv.handle(dm.build_request(mac, dm.DISCOVER, xid=1,
+                                    circuit_id=b"olt1/pon0/3"))
+        ack = srv.handle(dm.build_request(
+            mac, dm.REQUEST, xid=2, circuit_id=b"olt1/pon0/3"))
+        assert ack.msg_type == dm.ACK
+        lease = srv.leases[mac]
+        assert lease.circuit_id == b"olt1/pon0/3"
+        # renewal WITHOUT option 82 must not clobber the recorded id
+        ack2 = srv.handle(dm.build_request(mac, dm.REQUEST, xid=3,
+                                           ciaddr=lease.ip))
+        assert ack2.msg_type == dm.ACK
+        assert srv.leases[mac].circuit_id == b"olt1/pon0/3"
+        assert srv.leases_by_circuit[b"olt1/pon0/3"] is srv.leases[mac]
