@@ -129,3 +129,97 @@ class ComplianceLogger:
             if self._fh is not None:
                 self._fh.close()
                 self._fh = None
+
+    # ----------------------------------------- law-enforcement queries
+    def _iter_json_events(self):
+        """Every JSON event: rotated files (incl. gzipped) oldest
+        first, then the live file, then the in-memory tail for unsynced
+        lines.  Only meaningful for fmt=json — compliance queries need
+        the structured format (ref logging.go notes the same)."""
+        if self.fmt != "json":
+            raise ValueError("compliance queries require fmt=json")
+        seen = set()
+        paths = []
+        if self.path:
+            d = os.path.dirname(self.path) or "."
+            base = os.path.basename(self.path)
+            paths = [os.path.join(d, f) for f in sorted(
+                f for f in os.listdir(d) if f.startswith(base + "."))]
+            paths.append(self.path)
+        for p in paths:
+            try:
+                opener = gzip.open if p.endswith(".gz") else open
+                with opener(p, "rt") as f:
+                    for line in f:
+                        line = line.strip()
+                        if line:
+                            seen.add(line)
+                            yield json.loads(line)
+            except (OSError, ValueError):
+                continue
+        with self._lock:
+            tail = list(self.records)
+        for line in tail:
+            if line not in seen:
+                try:
+                    yield json.loads(line)
+                except ValueError:
+                    continue
+
+    def query_by_public_endpoint(self, public_ip: str, public_port: int,
+                                 at_time: float = 0.0) -> List[dict]:
+        """Who was behind public ip:port (optionally at a moment in
+        time)?  The CGNAT legal question (ref QueryByPublicEndpoint):
+        returns the matching mapping events — with at_time, only
+        mappings whose assign..release interval covers it (a port-block
+        event matches any port inside its block)."""
+        sessions: Dict[tuple, dict] = {}
+        out = []
+        for e in self._iter_json_events():
+            if e.get("public_ip") != public_ip:
+                continue
+            ev = e.get("event")
+            if ev in ("session_create", "session_delete"):
+                if e.get("public_port") != public_port:
+                    continue
+                key = ("s", e.get("subscriber_id"),
+                       e.get("private_ip"), e.get("private_port"),
+                       e.get("public_port"), e.get("protocol"))
+            elif ev in ("port_block_assign", "port_block_release"):
+                # block events encode [start, end] in the private/
+                # public port fields (manager.py PB_ASSIGN shape)
+                lo = e.get("private_port", 0)
+                hi = e.get("public_port", 0) or lo
+                if not (lo <= public_port <= (hi if hi >= lo else lo)):
+                    continue
+                key = ("b", e.get("subscriber_id"), e.get("private_ip"),
+                       lo)
+            else:
+                continue
+            if ev.endswith("create") or ev.endswith("assign"):
+                sessions[key] = {"start": e, "end": None}
+            else:
+                if key in sessions:
+                    sessions[key]["end"] = e
+                else:
+                    sessions[key] = {"start": None, "end": e}
+        for rec in sessions.values():
+            if rec["start"] is None:
+                continue
+            t0 = rec["start"].get("ts", 0)
+            t1 = rec["end"].get("ts", float("inf")) if rec["end"] \
+                else float("inf")
+            if at_time and not (t0 <= at_time <= t1):
+                continue
+            out.append({**rec["start"],
+                        "released_ts": None if rec["end"] is None
+                        else rec["end"].get("ts")})
+        return out
+
+    def export_for_compliance(self, start: float = 0.0,
+                              end: float = 0.0) -> List[dict]:
+        """All events in [start, end] across rotated + live + buffered
+        storage (ref ExportForCompliance)."""
+        end = end or float("inf")
+        return [e for e in self._iter_json_events()
+                if start <= e.get("ts", 0) <= end]

@@ -199,3 +199,67 @@ class TestComplianceLogger:
         lg = ComplianceLogger()
         assert lg.log_event(drained)
         assert "port_exhaustion" in lg.records[0]
+
+
+class TestComplianceQueries:
+    """Law-enforcement queries over the compliance log (ref
+    logging.go QueryByPublicEndpoint / ExportForCompliance)."""
+
+    def _ev(self, et, ts, sub=1, priv="10.0.1.50", pp=5555,
+            pub="203.0.113.1", pubp=1024):
+        from bng_amd.dataplane import abi
+        from bng_amd.dataplane.packets import ip2u32
+        return {"timestamp": ts, "ts": ts, "event_type": et,
+                "subscriber_id": sub, "private_ip": ip2u32(priv),
+                "private_port": pp, "public_ip": ip2u32(pub),
+                "public_port": pubp, "dest_ip": 0, "dest_port": 0,
+                "protocol": 6}
+
+    def test_query_by_public_endpoint_session_window(self, tmp_path):
+        from bng_amd.dataplane import abi
+        from bng_amd.nat.logging import ComplianceLogger
+        lg = ComplianceLogger(str(tmp_path / "nat.log"), fmt="json")
+        # subscriber 1 held 203.0.113.1:1024 during [100, 200]
+        e = self._ev(abi.LOG_SESSION_CREATE, 100)
+        e["timestamp"] = 100
+        lg.log_event({**e, "timestamp": 100})
+        lg.log_event({**self._ev(abi.LOG_SESSION_DELETE, 200)})
+        # subscriber 2 reused the same endpoint from t=300 (open)
+        lg.log_event({**self._ev(abi.LOG_SESSION_CREATE, 300, sub=2,
+                                 priv="10.0.1.60", pp=7777)})
+        got = lg.query_by_public_endpoint("203.0.113.1", 1024,
+                                          at_time=150)
+        assert len(got) == 1 and got[0]["subscriber_id"] == 1
+        assert got[0]["released_ts"] == 200
+        got = lg.query_by_public_endpoint("203.0.113.1", 1024,
+                                          at_time=400)
+        assert len(got) == 1 and got[0]["subscriber_id"] == 2
+        assert got[0]["released_ts"] is None
+        # no match at a gap moment or for a foreign port
+        assert lg.query_by_public_endpoint("203.0.113.1", 1024,
+                                           at_time=250) == []
+        assert lg.query_by_public_endpoint("203.0.113.1", 9999) == []
+        lg.close()
+
+    def test_query_covers_port_blocks_and_rotated_files(self, tmp_path):
+        from bng_amd.dataplane import abi
+        from bng_amd.nat.logging import ComplianceLogger
+        path = tmp_path / "nat.log"
+        lg = ComplianceLogger(str(path), fmt="json", bulk_mode=True,
+                              rotate_bytes=200, compress=True)
+        # block [1024, 2047] assigned at t=100 (bulk mode logs blocks)
+        pb = self._ev(abi.LOG_PB_ASSIGN, 100, sub=7)
+        pb["private_port"], pb["public_port"] = 1024, 2047
+        lg.log_event(pb)
+        # enough traffic to force rotation (events land in .gz files)
+        for i in range(20):
+            pad = self._ev(abi.LOG_PORT_EXHAUSTION, 110 + i, sub=9)
+            lg.log_event(pad)
+        got = lg.query_by_public_endpoint("203.0.113.1", 1500,
+                                          at_time=120)
+        assert len(got) == 1 and got[0]["subscriber_id"] == 7
+        assert lg.query_by_public_endpoint("203.0.113.1", 3000) == []
+        # export window spans rotated + live storage
+        exp = lg.export_for_compliance(105, 115)
+        assert {e["ts"] for e in exp} == set(range(110, 116))
+        lg.close()
