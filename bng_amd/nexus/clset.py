@@ -54,6 +54,8 @@ class CLSetStore(Store):
         self._thread: Optional[threading.Thread] = None
         self.stats = {"syncs_ok": 0, "syncs_failed": 0, "adopted": 0,
                       "peers_discovered": 0, "compactions": 0}
+        self._membership_cbs: List = []
+        self.peer_ttl = 300.0
         self._data_dir = data_dir
         self._wal = None
         self._wal_records = 0
@@ -223,6 +225,52 @@ class CLSetStore(Store):
 
     def add_peer_url(self, url: str):
         self._peer_urls.append(url.rstrip("/"))
+        self._fire_membership()
+
+    def on_membership(self, cb):
+        """cb(members: dict url -> {last_ok, fails, alive}) whenever
+        the peer set or a peer's reachability changes (ref clset.go
+        WithMembershipHook :57-60 + peer TTL :62-65)."""
+        self._membership_cbs.append(cb)
+        self._fire_membership()
+
+    def members(self) -> dict:
+        now = time.monotonic()
+        out = {}
+        for u in self._peer_urls:
+            st = self._peer_state.get(u, {})
+            last_ok = st.get("last_ok", 0.0)
+            out[u] = {"last_ok": last_ok, "fails": st.get("fails", 0),
+                      "alive": st.get("fails", 0) == 0 or
+                      (last_ok and now - last_ok < self.peer_ttl)}
+        return out
+
+    def _fire_membership(self):
+        for cb in getattr(self, "_membership_cbs", []):
+            try:
+                cb(self.members())
+            except Exception:
+                pass
+
+    def expire_peers(self, now: Optional[float] = None) -> int:
+        """Drop peers silent past peer_ttl (ref WithPeerTTL); they
+        re-enter via gossip if they come back."""
+        now = now if now is not None else time.monotonic()
+        dead = []
+        for u in list(self._peer_urls):
+            st = self._peer_state.get(u)
+            if st is None:
+                continue
+            last = st.get("last_ok", 0.0)
+            ref = last or st.get("first_seen", 0.0)
+            if st.get("fails", 0) > 0 and ref and \
+                    now - ref > self.peer_ttl:
+                self._peer_urls.remove(u)
+                self._peer_state.pop(u, None)
+                dead.append(u)
+        if dead:
+            self._fire_membership()
+        return len(dead)
 
     def sync_once(self, now: Optional[float] = None) -> int:
         """One anti-entropy round against every reachable peer, with
@@ -237,7 +285,8 @@ class CLSetStore(Store):
             import requests
             for url in list(self._peer_urls):
                 st = self._peer_state.setdefault(
-                    url, {"fails": 0, "next_try": 0.0, "last_ok": 0.0})
+                    url, {"fails": 0, "next_try": 0.0, "last_ok": 0.0,
+                          "first_seen": now})
                 if now < st["next_try"]:
                     continue          # still backing off
                 try:
@@ -249,8 +298,14 @@ class CLSetStore(Store):
                     st["next_try"] = 0.0
                     st["last_ok"] = now
                     self.stats["syncs_ok"] += 1
+                    if st.get("was_down"):
+                        st["was_down"] = False
+                        self._fire_membership()
                 except Exception:
                     st["fails"] += 1
+                    if st["fails"] == 1:
+                        st["was_down"] = True
+                        self._fire_membership()
                     st["next_try"] = now + min(
                         self.backoff_base * (2 ** (st["fails"] - 1)),
                         self.backoff_max)

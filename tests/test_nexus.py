@@ -452,3 +452,34 @@ def test_http_allocator_mtls_session_config(tmp_path):
     assert a.session.verify == str(ca)
     b = HTTPAllocator("https://nexus.example", insecure=True)
     assert b.session.verify is False
+
+
+class TestCLSetMembership:
+    """Membership hooks + peer TTL (ref nexus/clset/clset.go
+    WithMembershipHook :57-65)."""
+
+    def test_membership_events_and_ttl(self):
+        a = CLSetStore("a")
+        events = []
+        a.on_membership(events.append)
+        assert events == [{}]                      # fired on register
+        a.add_peer_url("http://127.0.0.1:1")       # unreachable port
+        assert len(events) == 2
+        assert "http://127.0.0.1:1" in events[-1]
+        # a failed sync round marks the peer down (fires once)
+        a.sync_once(now=100.0)
+        assert len(events) == 3
+        m = a.members()
+        assert m["http://127.0.0.1:1"]["fails"] >= 1
+        assert not m["http://127.0.0.1:1"]["alive"]
+        # silent past peer_ttl -> expired from the set
+        a.peer_ttl = 50.0
+        assert a.expire_peers(now=200.0) == 1
+        assert a.members() == {}
+        assert len(events) == 4
+        # a healthy in-process peer pair keeps membership quiet
+        b = CLSetStore("b")
+        a.add_peer(b)
+        a.put("k", b"v")
+        a.sync_once()
+        assert b.get("k") == b"v"
