@@ -442,3 +442,32 @@ class TestHAListenHostAuth:
         finally:
             active.stop()
             standby.stop()
+
+
+def test_failover_canceled_when_partner_recovers_in_grace():
+    """The grace delay re-checks partner health; a recovery during it
+    cancels the failover (ref failover_test.go CanceledFailover)."""
+    from bng_amd.ha.failover import (FailoverController, ROLE_STANDBY,
+                                     STATE_NORMAL)
+
+    class Mon:
+        partner_healthy = False
+
+        def on_event(self, cb):
+            pass
+
+    mon = Mon()
+    c = FailoverController("node-b", ROLE_STANDBY, monitor=mon,
+                           failover_delay=0.05)
+    # partner comes back while we wait out the grace period
+    import threading
+    threading.Timer(0.01, lambda: setattr(mon, "partner_healthy",
+                                          True)).start()
+    assert c.initiate_failover("partner down") is False
+    assert c.role == ROLE_STANDBY
+    assert c.state == STATE_NORMAL
+    assert c.stats["canceled"] == 1
+    # forced failover skips the grace re-check entirely
+    mon.partner_healthy = True
+    assert c.initiate_failover("operator", forced=True) is True
+    assert c.role == "active"
