@@ -292,13 +292,20 @@ class PPPoEServer:
             for t, v in opts:
                 if t == C.LCP_OPT_MRU:
                     mru = struct.unpack(">H", v)[0]
-                    if mru < 576:
+                    # PPPoE caps MRU at 1492 (RFC 2516 §7); too-small
+                    # MRUs are also bargained up (ref lcp.go NAK rules)
+                    if mru < 576 or mru > 1492:
                         naks.append((t, struct.pack(">H", self.mru)))
                     else:
                         s.peer_mru = mru
                 elif t == C.LCP_OPT_MAGIC:
                     peer_magic = struct.unpack(">I", v)[0]
-                    if peer_magic == s.our_magic:
+                    if peer_magic == 0:
+                        # zero magic defeats loop detection: NAK with
+                        # a proper random number (ref lcp_test "NAK
+                        # zero magic number")
+                        naks.append((t, os.urandom(4)))
+                    elif peer_magic == s.our_magic:
                         # our own magic coming back: looped link
                         # (RFC1661 §6.4) — Nak with a fresh number;
                         # give up after MAX_MAGIC_LOOPS

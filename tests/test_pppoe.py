@@ -576,3 +576,37 @@ def test_session_timeout_teardown():
     out = srv.tick(now=time.time() + 601)
     assert sid not in srv.sessions
     assert srv.stats.get("session_timeout") == 1
+
+
+class TestLcpOptionBounds:
+    """MRU upper bound (RFC 2516) and zero-magic NAK (ref lcp_test
+    scenarios 'NAK MRU greater than 1492' / 'NAK zero magic')."""
+
+    def _req(self, srv, s, opts):
+        return _send_lcp(srv, s, C.CONF_REQ, 9, C.encode_opts(opts))
+
+    def test_mru_over_1492_naked(self):
+        srv = PPPoEServer(SRV_MAC, auth="none")
+        s, _ = _discover_session(srv)
+        out = self._req(srv, s, [(C.LCP_OPT_MRU,
+                                  struct.pack(">H", 1500))])
+        naks = [c for c in _lcp_from(out) if c.code == C.CONF_NAK]
+        assert naks, "oversized MRU must be Naked"
+        opts = dict(C.decode_opts(naks[0].data))
+        assert struct.unpack(">H", opts[C.LCP_OPT_MRU])[0] <= 1492
+        # valid MRU is accepted
+        out = self._req(srv, s, [(C.LCP_OPT_MRU,
+                                  struct.pack(">H", 1492))])
+        assert any(c.code == C.CONF_ACK for c in _lcp_from(out))
+        assert s.peer_mru == 1492
+
+    def test_zero_magic_naked_with_real_number(self):
+        srv = PPPoEServer(SRV_MAC, auth="none")
+        s, _ = _discover_session(srv)
+        out = self._req(srv, s, [(C.LCP_OPT_MAGIC, b"\x00" * 4)])
+        naks = [c for c in _lcp_from(out) if c.code == C.CONF_NAK]
+        assert naks, "zero magic must be Naked"
+        opts = dict(C.decode_opts(naks[0].data))
+        assert opts[C.LCP_OPT_MAGIC] != b"\x00" * 4
+        # and it is not mistaken for a loop
+        assert s.magic_loops == 0
