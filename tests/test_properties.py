@@ -2,6 +2,8 @@
 allocators — the randomized complement to the example-based suites."""
 import ipaddress
 
+import struct
+
 from hypothesis import given, settings, strategies as st
 
 from bng_amd.dhcp import message as dm
@@ -379,3 +381,38 @@ class TestDnsCnameProperty:
         off += 4 + 2
         got, _ = decode_qname(resp, off + 10)
         assert got == target
+
+
+class TestParserRobustness:
+    """New binary parsers never hang or crash on garbage."""
+
+    @given(st.binary(max_size=256))
+    @settings(max_examples=300, deadline=None)
+    def test_netlink_attr_parser_total(self, blob):
+        from bng_amd.routing.netlink import _parse_attrs
+        out = _parse_attrs(blob)
+        assert isinstance(out, dict)
+
+    @given(st.binary(max_size=128))
+    @settings(max_examples=300, deadline=None)
+    def test_etsi_decode_rejects_garbage(self, blob):
+        from bng_amd.intercept.etsi import decode_pdu
+        try:
+            decode_pdu(blob)
+        except (ValueError, IndexError, KeyError, struct.error):
+            pass
+
+    @given(st.binary(max_size=200))
+    @settings(max_examples=300, deadline=None)
+    def test_vendor_tlv_parser_total(self, blob):
+        from bng_amd.ztp.bootstrap import parse_vendor_options
+        assert isinstance(parse_vendor_options(blob), str)
+
+    @given(st.binary(max_size=300))
+    @settings(max_examples=200, deadline=None)
+    def test_dns_query_handler_survives_garbage(self, blob):
+        from bng_amd.dns.resolver import Resolver
+        r = Resolver(lambda q: None)
+        r.add_rule("x.example", action="block")
+        out = r.handle_query(blob)
+        assert out is None or isinstance(out, bytes)
