@@ -221,6 +221,10 @@ class Manager:
         with self._lock:
             s = self.sessions.get(mac)
             if s is not None:
+                # re-create on a known MAC renews the lease (ref
+                # gateway_test.go SessionRenewalOnCreate)
+                s.last_renewal = now
+                s.lease_expiry = now + self.config.lease_duration
                 return s
             s = WifiSession(id=uuid.uuid4().hex[:12], mac=mac, ip=ip,
                             hostname=hostname, pool_id=pool_id,

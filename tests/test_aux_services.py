@@ -1265,3 +1265,17 @@ class TestWalledGardenDnsGlue:
         r2 = Resolver(upstream)
         attach_dns(wg2, r2, ["10.255.255.1"])
         assert r2.is_walled("10.0.1.7")
+
+
+def test_wifi_recreate_renews_lease():
+    """CreateSession on a known MAC returns the same session and
+    extends its lease (ref gateway_test.go SessionRenewalOnCreate)."""
+    import time as _t
+    from bng_amd.wifi.gateway import Manager as WifiMgr, WifiConfig
+    m = WifiMgr(WifiConfig(lease_duration=100.0))
+    s1 = m.create_session("aa:00:00:00:00:09")
+    first_expiry = s1.lease_expiry
+    _t.sleep(0.01)
+    s2 = m.create_session("aa:00:00:00:00:09")
+    assert s2 is s1
+    assert s2.lease_expiry > first_expiry
