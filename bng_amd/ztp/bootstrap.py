@@ -179,6 +179,16 @@ class BootstrapClient:
         self.register()
         return self.poll_until_approved(timeout)
 
+    def healthcheck(self) -> bool:
+        """GET /health on the Nexus (ref Healthcheck
+        bootstrap.go:466-485)."""
+        try:
+            r = self.session.get(f"{self.nexus_url}/health",
+                                 headers=self._headers(), timeout=5)
+            return r.status_code == 200
+        except Exception:
+            return False
+
     def register_and_wait(self, system_info: Optional[SystemInfo] = None,
                           max_retries: int = 0,
                           initial_backoff: float = 5.0,
@@ -294,6 +304,8 @@ class ZTPServer:
                 self._send(404, {})
 
             def do_GET(self):
+                if self.path == "/health":
+                    return self._send(200, {"status": "ok"})
                 if self.path.startswith("/api/v1/devices/") and \
                         self.path.endswith("/config"):
                     serial = self.path.split("/")[4]

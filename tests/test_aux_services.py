@@ -1279,3 +1279,16 @@ def test_wifi_recreate_renews_lease():
     s2 = m.create_session("aa:00:00:00:00:09")
     assert s2 is s1
     assert s2.lease_expiry > first_expiry
+
+
+def test_ztp_healthcheck():
+    """/health probe against the Nexus (ref bootstrap.go Healthcheck)."""
+    srv = ZTPServer()
+    try:
+        c = BootstrapClient(srv.url, serial="SN-H")
+        assert c.healthcheck() is True
+        # a dead endpoint is unhealthy, not an exception
+        c2 = BootstrapClient("http://127.0.0.1:1", serial="SN-H")
+        assert c2.healthcheck() is False
+    finally:
+        srv.stop()
