@@ -928,6 +928,52 @@ class BNG:
                 pass
         self._cleanup.clear()
 
+    def reload(self, argv: Optional[List[str]] = None) -> dict:
+        """SIGHUP hot reload (ref FEATURES.md Hot Reload): re-read the
+        YAML config and apply the session-safe subset without
+        restarting — log level, lease time, QoS policy definitions and
+        the default policy.  The file is parsed and validated before
+        anything is applied; a bad config changes nothing."""
+        argv = argv if argv is not None else []
+        parser = build_parser()
+        base = parser.parse_args(["run"] + argv)
+        base.config = self.args.config
+        if not base.config:
+            return {"reloaded": False, "reason": "no config file"}
+        try:
+            fresh = load_yaml_over_args(base, parser, argv)
+            policies = []
+            for spec in fresh.qos_policy:
+                name, down, up = spec.split(":")
+                policies.append((name, int(float(down) * 1e6),
+                                 int(float(up) * 1e6)))
+        except Exception as e:
+            self.log.warning("reload rejected: %s", e)
+            return {"reloaded": False, "reason": str(e)}
+        from ..radius.policy import Policy
+        applied = {"reloaded": True, "changed": []}
+        if fresh.log_level != self.args.log_level:
+            logging.getLogger().setLevel(getattr(
+                logging, fresh.log_level.upper().replace(
+                    "WARN", "WARNING")))
+            self.args.log_level = fresh.log_level
+            applied["changed"].append("log_level")
+        if fresh.lease_time != self.args.lease_time:
+            self.args.lease_time = fresh.lease_time
+            self.dhcp_server.lease_time = fresh.lease_time
+            applied["changed"].append("lease_time")
+        for name, down, up in policies:
+            cur = self.policy_manager.get(name)
+            if cur is None or cur.download_rate_bps != down or \
+                    cur.upload_rate_bps != up:
+                self.policy_manager.add_policy(Policy(name, down, up))
+                applied["changed"].append(f"qos_policy:{name}")
+        if fresh.qos_default_policy:
+            self.policy_manager.default_policy = \
+                self.policy_manager.get(fresh.qos_default_policy)
+        self.log.info("config reloaded: %s", applied["changed"])
+        return applied
+
     def stats(self) -> dict:
         out = {"dhcp": dict(self.dhcp_server.stats),
                "leases": len(self.dhcp_server.leases),
@@ -1086,6 +1132,8 @@ def main(argv: Optional[List[str]] = None) -> int:
             stop["flag"] = True
         signal.signal(signal.SIGINT, on_sig)
         signal.signal(signal.SIGTERM, on_sig)
+        # SIGHUP = hot reload of the YAML config (ref FEATURES.md)
+        signal.signal(signal.SIGHUP, lambda *_: app.reload(argv[1:]))
         print(f"bng running (node {args.node_id}); ^C to stop",
               flush=True)
         try:

@@ -758,3 +758,52 @@ def test_day_in_the_life_integration(tmp_path):
         assert st["audit"]["logged"] >= 1
     finally:
         bng.stop()
+
+
+def test_sighup_hot_reload(tmp_path):
+    """Config hot reload without restart (ref FEATURES.md Hot Reload):
+    session-safe settings apply, sessions survive, a bad file changes
+    nothing."""
+    import yaml
+    from bng_amd.cli.main import (BNG, build_parser,
+                                  load_yaml_over_args)
+    cfg = tmp_path / "bng.yaml"
+    cfg.write_text(yaml.safe_dump({
+        "pool-network": "10.9.0.0/24", "lease-time": 3600,
+        "qos-policy": ["gold:100:20"]}))
+    argv = ["run", "--interface", "lo", "--config", str(cfg)]
+    parser = build_parser()
+    args = load_yaml_over_args(parser.parse_args(argv), parser, argv)
+    bng = BNG(args).start()
+    try:
+        assert bng.dhcp_server.lease_time == 3600
+        assert bng.policy_manager.get("gold").download_rate_bps == \
+            100_000_000
+        # an existing lease must survive the reload
+        from bng_amd.dhcp import message as dm
+        from bng_amd.dataplane.packets import mac_bytes
+        mac = mac_bytes("aa:bb:cc:00:00:99")
+        bng.dhcp_server.handle(dm.build_request(mac, dm.DISCOVER))
+        bng.dhcp_server.handle(dm.build_request(mac, dm.REQUEST))
+        assert mac in bng.dhcp_server.leases
+        # operator edits the file and SIGHUPs
+        cfg.write_text(yaml.safe_dump({
+            "pool-network": "10.9.0.0/24", "lease-time": 7200,
+            "qos-policy": ["gold:200:40", "silver:50:10"],
+            "qos-default-policy": "silver"}))
+        r = bng.reload()
+        assert r["reloaded"]
+        assert "lease_time" in r["changed"]
+        assert bng.dhcp_server.lease_time == 7200
+        assert bng.policy_manager.get("gold").download_rate_bps == \
+            200_000_000
+        assert bng.policy_manager.get("silver") is not None
+        assert bng.policy_manager.default_policy.name == "silver"
+        assert mac in bng.dhcp_server.leases      # session survived
+        # invalid file: rejected, nothing changes
+        cfg.write_text("qos-policy: [notaspec]")
+        r2 = bng.reload()
+        assert not r2["reloaded"]
+        assert bng.dhcp_server.lease_time == 7200
+    finally:
+        bng.stop()
