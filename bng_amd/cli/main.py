@@ -299,6 +299,41 @@ def load_yaml_over_args(args: argparse.Namespace,
     return args
 
 
+def apply_env_overrides(args: argparse.Namespace,
+                        argv: List[str],
+                        environ: Optional[dict] = None
+                        ) -> argparse.Namespace:
+    """BNG_<FLAG> environment overrides (ref FEATURES.md config
+    sources: flags > env > YAML): applied to any flag not given
+    explicitly on the command line, coerced to the current attr's
+    type; lists split on commas."""
+    env = environ if environ is not None else os.environ
+    explicit = {tok.split("=")[0][2:].replace("-", "_")
+                for tok in argv if tok.startswith("--")}
+    for name, raw in env.items():
+        if not name.startswith("BNG_"):
+            continue
+        attr = name[4:].lower()
+        if attr in explicit or not hasattr(args, attr):
+            continue
+        cur = getattr(args, attr)
+        try:
+            if isinstance(cur, bool):
+                value = raw.lower() in ("1", "true", "yes", "on")
+            elif isinstance(cur, int):
+                value = parse_duration(raw)
+            elif isinstance(cur, float):
+                value = float(raw)
+            elif isinstance(cur, list):
+                value = [x for x in raw.split(",") if x]
+            else:
+                value = raw
+        except (ValueError, TypeError):
+            continue
+        setattr(args, attr, value)
+    return args
+
+
 def resolve_secret(value: str, file_value: str) -> str:
     """ref main.go:1567-1592 resolveSecret."""
     if file_value:
@@ -1125,6 +1160,7 @@ def main(argv: Optional[List[str]] = None) -> int:
                 "WARN", "WARNING")),
             format="%(asctime)s %(levelname)s %(name)s %(message)s")
         args = load_yaml_over_args(args, parser, argv)
+        args = apply_env_overrides(args, argv)
         app = BNG(args).start()
         stop = {"flag": False}
 

@@ -807,3 +807,30 @@ def test_sighup_hot_reload(tmp_path):
         assert bng.dhcp_server.lease_time == 7200
     finally:
         bng.stop()
+
+
+def test_env_var_overrides(tmp_path):
+    """BNG_* env overrides sit between flags and YAML (ref FEATURES.md
+    config sources): explicit flags win, env beats the file, types are
+    coerced."""
+    import yaml
+    from bng_amd.cli.main import (apply_env_overrides, build_parser,
+                                  load_yaml_over_args)
+    cfg = tmp_path / "bng.yaml"
+    cfg.write_text(yaml.safe_dump({"lease-time": 3600,
+                                   "node-id": "from-file"}))
+    argv = ["run", "--interface", "lo", "--config", str(cfg),
+            "--node-id", "from-flag"]
+    parser = build_parser()
+    args = parser.parse_args(argv)
+    args = load_yaml_over_args(args, parser, argv)
+    env = {"BNG_LEASE_TIME": "2h", "BNG_NODE_ID": "from-env",
+           "BNG_DHCP_LISTEN": "true", "BNG_RADIUS_SERVER": "a:1,b:2",
+           "BNG_HEALTH_CHECK_INTERVAL": "2.5",
+           "BNG_NOT_A_FLAG": "x", "OTHER": "y"}
+    args = apply_env_overrides(args, argv, env)
+    assert args.lease_time == 7200             # env beats file, Go dur
+    assert args.node_id == "from-flag"         # explicit flag wins
+    assert args.dhcp_listen is True            # bool coercion
+    assert args.radius_server == ["a:1", "b:2"]
+    assert args.health_check_interval == 2.5
