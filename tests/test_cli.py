@@ -834,3 +834,22 @@ def test_env_var_overrides(tmp_path):
     assert args.dhcp_listen is True            # bool coercion
     assert args.radius_server == ["a:1", "b:2"]
     assert args.health_check_interval == 2.5
+
+
+def test_every_module_imports():
+    """Import sweep over the whole package: any module with a syntax
+    error or missing dependency fails here rather than at deploy time
+    (the _C extension is exercised separately — it needs torch loaded
+    first)."""
+    import importlib
+    import pkgutil
+    import bng_amd
+    failures = []
+    for m in pkgutil.walk_packages(bng_amd.__path__, "bng_amd."):
+        if "csrc" in m.name or m.name.endswith(("__main__", "._C")):
+            continue
+        try:
+            importlib.import_module(m.name)
+        except Exception as e:
+            failures.append((m.name, repr(e)))
+    assert failures == []
